@@ -1,0 +1,64 @@
+"""In-tree extension build for spacy_ray_amd.
+
+Build everything in place (the built .so files travel to the GPU box with the
+repo snapshot):
+
+    python setup.py build_ext --inplace
+
+Two extensions:
+  * spacy_ray_amd._srx_cpu — pybind11/C++ core (murmur hashing, transition
+    systems).  Plain g++, no GPU needed.
+  * spacy_ray_amd._srx_hip — CDNA4 (gfx950) HIP kernels as a torch extension;
+    hipcc cross-compiles on a GPU-less box (PYTORCH_ROCM_ARCH=gfx950).
+"""
+import os
+import sys
+from pathlib import Path
+
+from setuptools import setup, Extension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+ROOT = Path(__file__).parent
+CSRC = ROOT / "spacy_ray_amd" / "ops" / "csrc"
+KERNELS = ROOT / "spacy_ray_amd" / "ops" / "kernels"
+
+import pybind11
+
+ext_modules = [
+    Extension(
+        "spacy_ray_amd._srx_cpu",
+        sources=[
+            str(CSRC / "cpu_ext.cpp"),
+            str(CSRC / "transitions.cpp"),
+        ],
+        include_dirs=[pybind11.get_include()],
+        extra_compile_args=["-O3", "-std=c++17", "-fvisibility=hidden"],
+        language="c++",
+    )
+]
+
+cmdclass = {}
+hip_sources = sorted(KERNELS.glob("*.hip")) + sorted(KERNELS.glob("*_hip.cpp"))
+if hip_sources and "SRX_SKIP_HIP" not in os.environ:
+    from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+    ext_modules.append(
+        CUDAExtension(
+            "spacy_ray_amd._srx_hip",
+            sources=[str(p) for p in hip_sources],
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    )
+    cmdclass["build_ext"] = BuildExtension
+
+setup(
+    name="spacy_ray_amd",
+    version="0.1.0",
+    packages=["spacy_ray_amd"],
+    ext_modules=ext_modules,
+    cmdclass=cmdclass,
+)

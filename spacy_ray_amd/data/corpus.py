@@ -1,0 +1,156 @@
+"""Corpus readers: DocBin files + deterministic synthetic corpora.
+
+Registered under ``@readers`` and resolved from ``[corpora.*]`` blocks via
+dotted names (`/root/reference/spacy_ray/worker.py:94-95`).  A corpus is a
+callable ``corpus(nlp) -> Iterator[Example]``.
+
+The synthetic corpus is the BASELINE data source (no network for datasets —
+BASELINE.md): deterministic per seed, Zipf-distributed synthetic vocabulary,
+random projective dependency trees, word-correlated tags/deps and BILUO
+entity spans, so models have learnable structure for convergence tests.
+"""
+from __future__ import annotations
+
+import random
+from pathlib import Path
+from typing import Iterator, List, Optional
+
+import numpy as np
+
+from spacy_ray_amd.config.registry import registry
+from spacy_ray_amd.vocab.doc import Doc, Example, Vocab
+from .docbin import DocBin
+
+
+@registry.readers("spacy.Corpus.v1")
+def create_docbin_reader(
+    path: Optional[str] = None,
+    gold_preproc: bool = False,
+    max_length: int = 0,
+    limit: int = 0,
+    augmenter=None,
+):
+    def corpus(nlp) -> Iterator[Example]:
+        if not path:
+            return
+        docbin = DocBin.from_disk(path, nlp.vocab)
+        n = 0
+        for doc in docbin.get_docs(nlp.vocab):
+            if max_length and len(doc) > max_length:
+                continue
+            yield Example.from_doc(doc)
+            n += 1
+            if limit and n >= limit:
+                break
+
+    return corpus
+
+
+def _random_projective_heads(n: int, rng: random.Random) -> List[int]:
+    """Random projective tree over [0, n): recursive root splitting."""
+    heads = [-1] * n
+
+    def build(lo: int, hi: int, head: int) -> None:
+        if lo >= hi:
+            return
+        r = rng.randrange(lo, hi)
+        heads[r] = head
+        build(lo, r, r)
+        build(r + 1, hi, r)
+
+    build(0, n, -1)
+    return heads
+
+
+def make_synthetic_docs(
+    vocab: Vocab,
+    *,
+    n_docs: int,
+    words_per_doc: int = 20,
+    vocab_size: int = 5000,
+    n_tags: int = 50,
+    n_deps: int = 40,
+    n_ent_types: int = 4,
+    seed: int = 0,
+) -> List[Doc]:
+    rng = random.Random(seed)
+    lexicon = [f"w{i}" for i in range(vocab_size)]
+    # Zipf sampling over the lexicon
+    ranks = np.arange(1, vocab_size + 1, dtype=np.float64)
+    probs = 1.0 / ranks
+    probs /= probs.sum()
+    np_rng = np.random.RandomState(seed)
+    tag_of_word = np_rng.randint(0, n_tags, size=vocab_size)
+    dep_of_word = np_rng.randint(0, n_deps, size=vocab_size)
+    ent_of_word = np_rng.randint(0, n_ent_types, size=vocab_size)
+
+    docs: List[Doc] = []
+    for _ in range(n_docs):
+        n = max(2, int(np_rng.poisson(words_per_doc)))
+        word_ids = np_rng.choice(vocab_size, size=n, p=probs)
+        words = [lexicon[i] for i in word_ids]
+        # tags: word-correlated with 10% noise
+        tags = []
+        for wid in word_ids:
+            t = tag_of_word[wid] if rng.random() > 0.1 else rng.randrange(n_tags)
+            tags.append(f"TAG{t}")
+        heads = _random_projective_heads(n, rng)
+        deps = [f"dep{dep_of_word[wid]}" for wid in word_ids]
+        for i, h in enumerate(heads):
+            if h == -1:
+                deps[i] = "ROOT"
+        # BILUO entity spans: ~1 entity per 8 tokens, len 1-3, non-overlapping
+        ents = ["O"] * n
+        i = 0
+        while i < n:
+            if rng.random() < 0.125:
+                length = min(rng.randint(1, 3), n - i)
+                etype = f"ENT{ent_of_word[word_ids[i]]}"
+                if length == 1:
+                    ents[i] = f"U-{etype}"
+                else:
+                    ents[i] = f"B-{etype}"
+                    for j in range(i + 1, i + length - 1):
+                        ents[j] = f"I-{etype}"
+                    ents[i + length - 1] = f"L-{etype}"
+                i += length
+            else:
+                i += 1
+        docs.append(Doc(vocab, words, tags=tags, heads=heads, deps=deps, ents=ents))
+    return docs
+
+
+@registry.readers("spacy-mi.SyntheticCorpus.v1")
+def create_synthetic_corpus(
+    n_docs: int = 1000,
+    words_per_doc: int = 20,
+    vocab_size: int = 5000,
+    n_tags: int = 50,
+    n_deps: int = 40,
+    n_ent_types: int = 4,
+    seed: int = 0,
+    shuffle: bool = True,
+):
+    cache: dict = {}
+
+    def corpus(nlp) -> Iterator[Example]:
+        if "docs" not in cache:
+            cache["docs"] = make_synthetic_docs(
+                nlp.vocab,
+                n_docs=n_docs,
+                words_per_doc=words_per_doc,
+                vocab_size=vocab_size,
+                n_tags=n_tags,
+                n_deps=n_deps,
+                n_ent_types=n_ent_types,
+                seed=seed,
+            )
+            cache["epoch"] = 0
+        docs = list(cache["docs"])
+        if shuffle:
+            random.Random(seed + cache["epoch"]).shuffle(docs)
+        cache["epoch"] += 1
+        for doc in docs:
+            yield Example.from_doc(doc)
+
+    return corpus

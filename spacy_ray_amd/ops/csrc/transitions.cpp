@@ -1,0 +1,465 @@
+// Batched transition systems for the transition-based parser and NER.
+//
+// Behavioral contract of spaCy's Cython parser internals (SURVEY.md §2.2 N7:
+// upstream spacy/pipeline/_parser_internals/{arc_eager.pyx, ner.pyx,
+// _state.pxd}) — re-designed, not translated: one C++ object holds ALL states
+// of a batch in struct-of-arrays form, and every API call (features / valid /
+// costs / advance) operates on the whole batch so the Python-side per-step
+// loop does O(1) native calls per transition step instead of per-state ones.
+//
+// Arc-eager with the Goldberg & Nivre (2012) dynamic oracle; BILUO NER with
+// per-token gold-action costs.  Labeled actions: cost +1 when the arc matches
+// gold but the label does not.
+#include <pybind11/pybind11.h>
+#include <pybind11/numpy.h>
+
+#include <algorithm>
+#include <cstdint>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+constexpr float KInvalid = 1e9f;
+
+// ------------------------------------------------------------------ parser
+// Actions: 0=SHIFT, 1=REDUCE, 2..2+L-1 = LEFT-ARC(l), 2+L..2+2L-1 = RIGHT-ARC(l)
+struct ParserState {
+  std::vector<int32_t> stack;
+  int32_t buf = 0;   // index of buffer front
+  int32_t len = 0;
+  std::vector<int32_t> head;    // -1 = none
+  std::vector<int32_t> label;   // -1 = none
+  // children bookkeeping for features (two leftmost / two rightmost)
+  std::vector<int32_t> l1, l2, r1, r2;
+  std::vector<int32_t> n_head_in_stack_cache;  // unused; placeholder
+
+  void init(int32_t n) {
+    len = n;
+    buf = 0;
+    stack.clear();
+    head.assign(n, -1);
+    label.assign(n, -1);
+    l1.assign(n, -1);
+    l2.assign(n, -1);
+    r1.assign(n, -1);
+    r2.assign(n, -1);
+  }
+  bool final_state() const { return buf >= len && stack.size() <= 1; }
+  int32_t s0() const { return stack.empty() ? -1 : stack.back(); }
+  int32_t s1() const { return stack.size() < 2 ? -1 : stack[stack.size() - 2]; }
+  int32_t s2() const { return stack.size() < 3 ? -1 : stack[stack.size() - 3]; }
+
+  void add_arc(int32_t h, int32_t d, int32_t lab) {
+    head[d] = h;
+    label[d] = lab;
+    if (d < h) {
+      if (l1[h] == -1 || d < l1[h]) { l2[h] = l1[h]; l1[h] = d; }
+      else if (l2[h] == -1 || d < l2[h]) { l2[h] = d; }
+    } else {
+      if (r1[h] == -1 || d > r1[h]) { r2[h] = r1[h]; r1[h] = d; }
+      else if (r2[h] == -1 || d > r2[h]) { r2[h] = d; }
+    }
+  }
+};
+
+struct ArcEagerBatch {
+  int32_t n_labels;
+  std::vector<ParserState> states;
+  std::vector<int32_t> offsets;               // doc start offset in flat arrays
+  std::vector<std::vector<int32_t>> gold_head;   // per doc (empty if no gold)
+  std::vector<std::vector<int32_t>> gold_label;
+  bool has_gold = false;
+
+  ArcEagerBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
+                int32_t n_labels_)
+      : n_labels(n_labels_) {
+    auto L = lengths.unchecked<1>();
+    int32_t off = 0;
+    states.resize(L.shape(0));
+    offsets.resize(L.shape(0));
+    for (py::ssize_t i = 0; i < L.shape(0); i++) {
+      states[i].init(L(i));
+      offsets[i] = off;
+      off += L(i);
+    }
+  }
+
+  void set_gold(py::array_t<int32_t, py::array::c_style | py::array::forcecast> heads,
+                py::array_t<int32_t, py::array::c_style | py::array::forcecast> labels) {
+    auto H = heads.unchecked<1>();
+    auto Lb = labels.unchecked<1>();
+    gold_head.resize(states.size());
+    gold_label.resize(states.size());
+    for (size_t d = 0; d < states.size(); d++) {
+      int32_t off = offsets[d], n = states[d].len;
+      gold_head[d].assign(n, -1);
+      gold_label[d].assign(n, -1);
+      for (int32_t i = 0; i < n; i++) {
+        gold_head[d][i] = H(off + i);
+        gold_label[d][i] = Lb(off + i);
+      }
+    }
+    has_gold = true;
+  }
+
+  int32_t n_actions() const { return 2 + 2 * n_labels; }
+  size_t size() const { return states.size(); }
+
+  py::array_t<uint8_t> is_final() const {
+    py::array_t<uint8_t> out((py::ssize_t)states.size());
+    auto r = out.mutable_unchecked<1>();
+    for (size_t i = 0; i < states.size(); i++) r(i) = states[i].final_state() ? 1 : 0;
+    return out;
+  }
+
+  // 13 context tokens per state, as batch-flat indices (-1 = missing):
+  // [S0,S1,S2, B0,B1,B2, L1(S0),L2(S0), R1(S0),R2(S0), L1(S1), R1(S1), head(S0)]
+  py::array_t<int32_t> features() const {
+    py::ssize_t S = (py::ssize_t)states.size();
+    py::array_t<int32_t> out({S, (py::ssize_t)13});
+    auto r = out.mutable_unchecked<2>();
+    for (py::ssize_t i = 0; i < S; i++) {
+      const ParserState& st = states[i];
+      int32_t off = offsets[i];
+      int32_t f[13];
+      int32_t s0 = st.s0(), s1 = st.s1(), s2 = st.s2();
+      f[0] = s0; f[1] = s1; f[2] = s2;
+      f[3] = st.buf < st.len ? st.buf : -1;
+      f[4] = st.buf + 1 < st.len ? st.buf + 1 : -1;
+      f[5] = st.buf + 2 < st.len ? st.buf + 2 : -1;
+      f[6] = s0 >= 0 ? st.l1[s0] : -1;
+      f[7] = s0 >= 0 ? st.l2[s0] : -1;
+      f[8] = s0 >= 0 ? st.r1[s0] : -1;
+      f[9] = s0 >= 0 ? st.r2[s0] : -1;
+      f[10] = s1 >= 0 ? st.l1[s1] : -1;
+      f[11] = s1 >= 0 ? st.r1[s1] : -1;
+      f[12] = s0 >= 0 ? st.head[s0] : -1;
+      for (int k = 0; k < 13; k++) r(i, k) = f[k] >= 0 ? off + f[k] : -1;
+    }
+    return out;
+  }
+
+  void fill_valid(uint8_t* v, const ParserState& st) const {
+    const int32_t A = n_actions();
+    bool has_buf = st.buf < st.len;
+    bool has_s0 = !st.stack.empty();
+    bool s0_has_head = has_s0 && st.head[st.s0()] != -1;
+    std::fill(v, v + A, 0);
+    if (has_buf) v[0] = 1;                                    // SHIFT
+    // REDUCE: s0 has a head; or forced cleanup when the buffer is exhausted
+    // (headless pops attach to root), so every non-final state has >=1 valid
+    // action and the step loop always terminates.
+    if (has_s0 && (s0_has_head || !has_buf)) v[1] = 1;
+    bool la_ok = has_s0 && has_buf && !s0_has_head;
+    bool ra_ok = has_s0 && has_buf;
+    for (int32_t l = 0; l < n_labels; l++) {
+      v[2 + l] = la_ok ? 1 : 0;
+      v[2 + n_labels + l] = ra_ok ? 1 : 0;
+    }
+  }
+
+  py::array_t<uint8_t> valid() const {
+    py::ssize_t S = (py::ssize_t)states.size(), A = n_actions();
+    py::array_t<uint8_t> out({S, A});
+    auto r = out.mutable_unchecked<2>();
+    for (py::ssize_t i = 0; i < S; i++) fill_valid(r.mutable_data(i, 0), states[i]);
+    return out;
+  }
+
+  // Goldberg&Nivre dynamic-oracle costs; invalid actions get KInvalid.
+  py::array_t<float> costs() const {
+    if (!has_gold) throw std::runtime_error("costs() requires set_gold()");
+    py::ssize_t S = (py::ssize_t)states.size(), A = n_actions();
+    py::array_t<float> out({S, A});
+    auto r = out.mutable_unchecked<2>();
+    std::vector<uint8_t> v((size_t)A);
+    for (py::ssize_t i = 0; i < S; i++) {
+      const ParserState& st = states[i];
+      const auto& gh = gold_head[i];
+      const auto& gl = gold_label[i];
+      fill_valid(v.data(), st);
+      int32_t b = st.buf < st.len ? st.buf : -1;
+      int32_t s0 = st.s0();
+      // membership helpers
+      auto in_stack = [&](int32_t t) {
+        for (int32_t s : st.stack) if (s == t) return true;
+        return false;
+      };
+      // SHIFT cost: gold head of b in stack (recoverable only via RA now —
+      // i.e. only if gold head == s0, which SHIFT forgoes), plus headless
+      // stack items whose gold head is b.
+      float c_shift = 0, c_reduce = 0, c_la = 0, c_ra = 0;
+      if (b >= 0) {
+        if (gh[b] >= 0 && in_stack(gh[b])) c_shift += 1;
+        for (int32_t s : st.stack)
+          if (st.head[s] == -1 && gh[s] == b) c_shift += 1;
+      }
+      if (s0 >= 0) {
+        // REDUCE: lose gold dependents of s0 in buffer.
+        for (int32_t d = st.buf; d < st.len; d++)
+          if (gh[d] == s0) c_reduce += 1;
+        if (b >= 0) {
+          // LEFT-ARC(b -> s0): lose gold deps of s0 in buffer, and gold head
+          // of s0 if it is in buffer beyond b (head==b is the gold arc).
+          for (int32_t d = st.buf; d < st.len; d++)
+            if (gh[d] == s0) c_la += 1;
+          if (gh[s0] >= 0 && gh[s0] > b) c_la += 1;
+          if (gh[s0] >= 0 && in_stack(gh[s0])) c_la += 1;
+          // RIGHT-ARC(s0 -> b): lose gold head of b elsewhere (stack != s0 or
+          // buffer), and headless stack items whose gold head is b.
+          if (gh[b] >= 0 && gh[b] != s0 && (in_stack(gh[b]) || gh[b] > b)) c_ra += 1;
+          for (int32_t s : st.stack)
+            if (s != s0 && st.head[s] == -1 && gh[s] == b) c_ra += 1;
+        }
+      }
+      r(i, 0) = v[0] ? c_shift : KInvalid;
+      r(i, 1) = v[1] ? c_reduce : KInvalid;
+      for (int32_t l = 0; l < n_labels; l++) {
+        float la = c_la, ra = c_ra;
+        if (b >= 0 && s0 >= 0) {
+          if (gh[s0] == b && gl[s0] != l) la += 1;  // right arc, wrong label
+          if (gh[b] == s0 && gl[b] != l) ra += 1;
+        }
+        r(i, 2 + l) = v[2 + l] ? la : KInvalid;
+        r(i, 2 + n_labels + l) = v[2 + n_labels + l] ? ra : KInvalid;
+      }
+    }
+    return out;
+  }
+
+  void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
+    auto a = actions.unchecked<1>();
+    for (py::ssize_t i = 0; i < a.shape(0); i++) {
+      ParserState& st = states[i];
+      if (st.final_state()) continue;
+      int32_t act = a(i);
+      if (act < 0) continue;  // explicit no-op (already-final slot)
+      if (act == 0) {  // SHIFT
+        st.stack.push_back(st.buf);
+        st.buf += 1;
+      } else if (act == 1) {  // REDUCE
+        st.stack.pop_back();
+      } else if (act < 2 + n_labels) {  // LEFT-ARC
+        int32_t l = act - 2;
+        int32_t s0 = st.stack.back();
+        st.add_arc(st.buf, s0, l);
+        st.stack.pop_back();
+      } else {  // RIGHT-ARC
+        int32_t l = act - 2 - n_labels;
+        int32_t s0 = st.stack.back();
+        st.add_arc(s0, st.buf, l);
+        st.stack.push_back(st.buf);
+        st.buf += 1;
+      }
+      // Degenerate-state guard: buffer exhausted with stack >1 -> force pops.
+      if (st.buf >= st.len) {
+        // states are final when stack <=1; remaining stack entries keep
+        // head -1 (attached to nothing = root), like spaCy's unattached
+        // tokens defaulting to root.
+      }
+    }
+  }
+
+  py::array_t<int32_t> heads() const {
+    int32_t total = 0;
+    for (auto& st : states) total += st.len;
+    py::array_t<int32_t> out((py::ssize_t)total);
+    auto r = out.mutable_unchecked<1>();
+    int32_t k = 0;
+    for (auto& st : states)
+      for (int32_t i = 0; i < st.len; i++) r(k++) = st.head[i];
+    return out;
+  }
+
+  py::array_t<int32_t> labels() const {
+    int32_t total = 0;
+    for (auto& st : states) total += st.len;
+    py::array_t<int32_t> out((py::ssize_t)total);
+    auto r = out.mutable_unchecked<1>();
+    int32_t k = 0;
+    for (auto& st : states)
+      for (int32_t i = 0; i < st.len; i++) r(k++) = st.label[i];
+    return out;
+  }
+};
+
+// --------------------------------------------------------------------- NER
+// Per-token gold codes: 0 = O; for type t: 1+4t=B, 2+4t=I, 3+4t=L, 4+4t=U.
+// Actions: 0 = OUT; for type t: 1+4t=BEGIN, 2+4t=IN, 3+4t=LAST, 4+4t=UNIT.
+struct NerState {
+  int32_t i = 0;      // current token
+  int32_t len = 0;
+  int32_t open = -1;  // open entity type or -1
+  int32_t open_start = -1;
+  std::vector<int32_t> tags;  // emitted per-token action codes
+
+  void init(int32_t n) { i = 0; len = n; open = -1; open_start = -1; tags.assign(n, 0); }
+  bool final_state() const { return i >= len; }
+};
+
+struct BiluoBatch {
+  int32_t n_types;
+  std::vector<NerState> states;
+  std::vector<int32_t> offsets;
+  std::vector<std::vector<int32_t>> gold;  // per-token gold codes
+  bool has_gold = false;
+
+  BiluoBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
+             int32_t n_types_)
+      : n_types(n_types_) {
+    auto L = lengths.unchecked<1>();
+    states.resize(L.shape(0));
+    offsets.resize(L.shape(0));
+    int32_t off = 0;
+    for (py::ssize_t i = 0; i < L.shape(0); i++) {
+      states[i].init(L(i));
+      offsets[i] = off;
+      off += L(i);
+    }
+  }
+
+  void set_gold(py::array_t<int32_t, py::array::c_style | py::array::forcecast> codes) {
+    auto G = codes.unchecked<1>();
+    gold.resize(states.size());
+    for (size_t d = 0; d < states.size(); d++) {
+      int32_t off = offsets[d], n = states[d].len;
+      gold[d].assign(n, 0);
+      for (int32_t i = 0; i < n; i++) gold[d][i] = G(off + i);
+    }
+    has_gold = true;
+  }
+
+  int32_t n_actions() const { return 1 + 4 * n_types; }
+  size_t size() const { return states.size(); }
+
+  py::array_t<uint8_t> is_final() const {
+    py::array_t<uint8_t> out((py::ssize_t)states.size());
+    auto r = out.mutable_unchecked<1>();
+    for (size_t i = 0; i < states.size(); i++) r(i) = states[i].final_state() ? 1 : 0;
+    return out;
+  }
+
+  // 6 context tokens: [i-2, i-1, i, i+1, i+2, open_start]  (batch-flat, -1 pad)
+  py::array_t<int32_t> features() const {
+    py::ssize_t S = (py::ssize_t)states.size();
+    py::array_t<int32_t> out({S, (py::ssize_t)6});
+    auto r = out.mutable_unchecked<2>();
+    for (py::ssize_t s = 0; s < S; s++) {
+      const NerState& st = states[s];
+      int32_t off = offsets[s];
+      int32_t f[6] = {st.i - 2, st.i - 1, st.i, st.i + 1, st.i + 2, st.open_start};
+      for (int k = 0; k < 6; k++) {
+        r(s, k) = (f[k] >= 0 && f[k] < st.len) ? off + f[k] : -1;
+      }
+    }
+    return out;
+  }
+
+  void fill_valid(uint8_t* v, const NerState& st) const {
+    const int32_t A = n_actions();
+    std::fill(v, v + A, 0);
+    if (st.final_state()) return;
+    bool last_tok = st.i == st.len - 1;
+    if (st.open < 0) {
+      v[0] = 1;  // OUT
+      for (int32_t t = 0; t < n_types; t++) {
+        if (!last_tok) v[1 + 4 * t] = 1;  // BEGIN needs a following token
+        v[4 + 4 * t] = 1;                 // UNIT
+      }
+    } else {
+      if (!last_tok) v[2 + 4 * st.open] = 1;  // IN
+      v[3 + 4 * st.open] = 1;                 // LAST
+    }
+  }
+
+  py::array_t<uint8_t> valid() const {
+    py::ssize_t S = (py::ssize_t)states.size(), A = n_actions();
+    py::array_t<uint8_t> out({S, A});
+    auto r = out.mutable_unchecked<2>();
+    for (py::ssize_t i = 0; i < S; i++) fill_valid(r.mutable_data(i, 0), states[i]);
+    return out;
+  }
+
+  py::array_t<float> costs() const {
+    if (!has_gold) throw std::runtime_error("costs() requires set_gold()");
+    py::ssize_t S = (py::ssize_t)states.size(), A = n_actions();
+    py::array_t<float> out({S, A});
+    auto r = out.mutable_unchecked<2>();
+    std::vector<uint8_t> v((size_t)A);
+    for (py::ssize_t s = 0; s < S; s++) {
+      const NerState& st = states[s];
+      fill_valid(v.data(), st);
+      for (py::ssize_t a = 0; a < A; a++) {
+        if (!v[a]) { r(s, a) = KInvalid; continue; }
+        int32_t g = st.final_state() ? -1 : gold[s][st.i];
+        r(s, a) = ((int32_t)a == g) ? 0.0f : 1.0f;
+      }
+    }
+    return out;
+  }
+
+  void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
+    auto a = actions.unchecked<1>();
+    for (py::ssize_t s = 0; s < a.shape(0); s++) {
+      NerState& st = states[s];
+      if (st.final_state()) continue;
+      int32_t act = a(s);
+      if (act < 0) continue;
+      st.tags[st.i] = act;
+      if (act == 0) {
+        st.open = -1; st.open_start = -1;
+      } else {
+        int32_t t = (act - 1) / 4;
+        int32_t kind = (act - 1) % 4;  // 0=B,1=I,2=L,3=U
+        if (kind == 0) { st.open = t; st.open_start = st.i; }
+        else if (kind == 1) { /* stays open */ }
+        else { st.open = -1; st.open_start = -1; }
+      }
+      st.i += 1;
+    }
+  }
+
+  py::array_t<int32_t> tags() const {
+    int32_t total = 0;
+    for (auto& st : states) total += st.len;
+    py::array_t<int32_t> out((py::ssize_t)total);
+    auto r = out.mutable_unchecked<1>();
+    int32_t k = 0;
+    for (auto& st : states)
+      for (int32_t i = 0; i < st.len; i++) r(k++) = st.tags[i];
+    return out;
+  }
+};
+
+}  // namespace
+
+void init_transitions(py::module_& m) {
+  py::class_<ArcEagerBatch>(m, "ArcEagerBatch")
+      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t>(),
+           py::arg("lengths"), py::arg("n_labels"))
+      .def("set_gold", &ArcEagerBatch::set_gold, py::arg("heads"), py::arg("labels"))
+      .def_property_readonly("n_actions", &ArcEagerBatch::n_actions)
+      .def("__len__", &ArcEagerBatch::size)
+      .def("is_final", &ArcEagerBatch::is_final)
+      .def("features", &ArcEagerBatch::features)
+      .def("valid", &ArcEagerBatch::valid)
+      .def("costs", &ArcEagerBatch::costs)
+      .def("advance", &ArcEagerBatch::advance)
+      .def("heads", &ArcEagerBatch::heads)
+      .def("labels", &ArcEagerBatch::labels);
+
+  py::class_<BiluoBatch>(m, "BiluoBatch")
+      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t>(),
+           py::arg("lengths"), py::arg("n_types"))
+      .def("set_gold", &BiluoBatch::set_gold, py::arg("codes"))
+      .def_property_readonly("n_actions", &BiluoBatch::n_actions)
+      .def("__len__", &BiluoBatch::size)
+      .def("is_final", &BiluoBatch::is_final)
+      .def("features", &BiluoBatch::features)
+      .def("valid", &BiluoBatch::valid)
+      .def("costs", &BiluoBatch::costs)
+      .def("advance", &BiluoBatch::advance)
+      .def("tags", &BiluoBatch::tags);
+}

@@ -1,0 +1,25 @@
+"""Pipe factories (the ``factory = "..."`` names in [components.*] blocks)."""
+from __future__ import annotations
+
+from spacy_ray_amd.config.registry import registry
+from .pipes import NerPipe, ParserPipe, TaggerPipe, Tok2VecPipe
+
+
+@registry.factories("tok2vec")
+def make_tok2vec_pipe(name: str, model):
+    return Tok2VecPipe(name, model)
+
+
+@registry.factories("tagger")
+def make_tagger_pipe(name: str, model):
+    return TaggerPipe(name, model)
+
+
+@registry.factories("parser")
+def make_parser_pipe(name: str, model):
+    return ParserPipe(name, model)
+
+
+@registry.factories("ner")
+def make_ner_pipe(name: str, model):
+    return NerPipe(name, model)

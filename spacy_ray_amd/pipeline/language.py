@@ -1,0 +1,212 @@
+"""Language: the pipeline container (spaCy ``nlp`` contract).
+
+Holds vocab + ordered pipes, runs the shared-tok2vec single-graph training
+step (SURVEY.md §3.2 disposition), scores on dev data, and round-trips to a
+spaCy-style on-disk layout: config.cfg + meta.json + vocab/ + one directory
+per component with its params (`/root/reference/spacy_ray/worker.py:219-222`
+``nlp.to_disk`` contract).
+"""
+from __future__ import annotations
+
+import json
+from pathlib import Path
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.nn as nn
+
+from spacy_ray_amd.config.config import Config, resolve, resolve_dot_names
+from spacy_ray_amd.config.registry import registry
+from spacy_ray_amd.models.batch import TokenBatch
+from spacy_ray_amd.vocab.doc import Doc, Example, Vocab
+from .pipes import TrainablePipe, Tok2VecPipe
+
+
+class Language:
+    def __init__(self, vocab: Vocab, config: Config, device: str = "cpu") -> None:
+        self.vocab = vocab
+        self.config = config
+        self.device = torch.device(device)
+        self.pipeline: List[Tuple[str, TrainablePipe]] = []
+        self.meta: Dict = {"lang": vocab.lang, "name": "pipeline", "version": "0.0.0"}
+        self._frozen: List[str] = []
+
+    # ------------------------------------------------------------- pipeline
+    @property
+    def pipe_names(self) -> List[str]:
+        return [n for n, _ in self.pipeline]
+
+    def get_pipe(self, name: str) -> TrainablePipe:
+        for n, p in self.pipeline:
+            if n == name:
+                return p
+        raise KeyError(name)
+
+    def add_pipe(self, name: str, pipe: TrainablePipe) -> None:
+        self.pipeline.append((name, pipe))
+
+    @property
+    def tok2vec(self) -> Optional[Tok2VecPipe]:
+        for _, p in self.pipeline:
+            if isinstance(p, Tok2VecPipe):
+                return p
+        return None
+
+    def torch_module(self) -> nn.ModuleDict:
+        """All trainable parameters as one module (flat-buffer optimizer input).
+        Iteration order == pipeline order == identical on every rank (the
+        cross-rank key invariant)."""
+        d = nn.ModuleDict()
+        for name, pipe in self.pipeline:
+            if pipe.module is not None:
+                d[name.replace(".", "_")] = pipe.module
+        return d
+
+    # ------------------------------------------------------------- training
+    def forward_loss(self, examples: Sequence[Example], losses: Optional[Dict[str, float]] = None,
+                     drop: float = 0.0):
+        """One forward pass over all trainable pipes -> (total_loss, losses)."""
+        if losses is None:
+            losses = {}
+        docs = [eg.predicted for eg in examples]
+        batch = TokenBatch(docs, self.device)
+        t2v_pipe = self.tok2vec
+        t2v = t2v_pipe.forward(batch, drop=drop) if t2v_pipe is not None else None
+        total = None
+        for name, pipe in self.pipeline:
+            if isinstance(pipe, Tok2VecPipe) or name in self._frozen:
+                continue
+            loss, _count = pipe.get_loss(examples, t2v, batch)
+            losses[name] = losses.get(name, 0.0) + float(loss.detach())
+            total = loss if total is None else total + loss
+        if total is None:
+            total = torch.zeros((), device=self.device)
+        return total, losses
+
+    def update(self, examples: Sequence[Example], *, sgd=None, losses=None, drop: float = 0.0):
+        """Convenience single-process update (tests / CPU smoke)."""
+        total, losses = self.forward_loss(examples, losses, drop=drop)
+        total.backward()
+        if sgd is not None:
+            sgd.step()
+            sgd.zero_grad()
+        return losses
+
+    # ------------------------------------------------------------ inference
+    def predict_docs(self, docs: Sequence[Doc]) -> Sequence[Doc]:
+        if not docs:
+            return docs
+        with torch.no_grad():
+            batch = TokenBatch(docs, self.device)
+            t2v_pipe = self.tok2vec
+            t2v = t2v_pipe.forward(batch) if t2v_pipe is not None else None
+        for name, pipe in self.pipeline:
+            if isinstance(pipe, Tok2VecPipe):
+                continue
+            pipe.predict_and_set(docs, t2v, batch)
+        return docs
+
+    def evaluate(self, examples: Sequence[Example], batch_size: int = 256) -> Dict[str, float]:
+        from spacy_ray_amd.train.scorer import score_examples
+
+        for i in range(0, len(examples), batch_size):
+            chunk = examples[i : i + batch_size]
+            self.predict_docs([eg.predicted for eg in chunk])
+        return score_examples(examples, self.pipe_names)
+
+    def __call__(self, text: str) -> Doc:
+        from spacy_ray_amd.vocab.doc import simple_tokenize
+
+        doc = simple_tokenize(self.vocab, text)
+        self.predict_docs([doc])
+        return doc
+
+    # ---------------------------------------------------------- persistence
+    def to_disk(self, path) -> None:
+        path = Path(path)
+        path.mkdir(parents=True, exist_ok=True)
+        self.config.to_disk(path / "config.cfg")
+        (path / "meta.json").write_text(json.dumps(self.meta, indent=2))
+        vocab_dir = path / "vocab"
+        vocab_dir.mkdir(exist_ok=True)
+        (vocab_dir / "strings.json").write_text(json.dumps(self.vocab.strings.to_list()))
+        for name, pipe in self.pipeline:
+            pdir = path / name
+            pdir.mkdir(exist_ok=True)
+            (pdir / "cfg.json").write_text(json.dumps(pipe.state_cfg()))
+            if pipe.module is not None:
+                from safetensors.torch import save_file
+
+                state = {k: v.detach().cpu().contiguous() for k, v in pipe.module.state_dict().items()}
+                save_file(state, str(pdir / "model.safetensors"))
+
+    def from_disk(self, path) -> "Language":
+        path = Path(path)
+        self.meta = json.loads((path / "meta.json").read_text())
+        strings = json.loads((path / "vocab" / "strings.json").read_text())
+        for s in strings:
+            self.vocab.strings.add(s)
+        for name, pipe in self.pipeline:
+            pdir = path / name
+            if (pdir / "cfg.json").exists():
+                pipe.load_cfg(json.loads((pdir / "cfg.json").read_text()), self.device)
+            mfile = pdir / "model.safetensors"
+            if mfile.exists() and pipe.module is not None:
+                from safetensors.torch import load_file
+
+                state = load_file(str(mfile))
+                pipe.module.load_state_dict(state)
+                pipe.module.to(self.device)
+        return self
+
+    def select_pipes(self, disable: Sequence[str]):
+        lang = self
+
+        class _Ctx:
+            def __enter__(self):
+                lang._frozen = list(disable)
+                return lang
+
+            def __exit__(self, *a):
+                lang._frozen = []
+
+        return _Ctx()
+
+
+def build_nlp(config: Config, device: str = "cpu") -> Language:
+    """Build the pipeline skeleton from [nlp]/[components] (uninitialized)."""
+    registry.ensure_populated()
+    cfg = config.interpolate()
+    lang = cfg.get("nlp", {}).get("lang", "xx")
+    pipe_names = cfg.get("nlp", {}).get("pipeline", [])
+    nlp = Language(Vocab(lang), config, device=device)
+    components = cfg.get("components", {})
+    for name in pipe_names:
+        comp_cfg = dict(components.get(name, {}))
+        factory_name = comp_cfg.pop("factory", name)
+        model_cfg = comp_cfg.pop("model", None)
+        model_spec = resolve(model_cfg) if model_cfg else None
+        factory = registry.factories.get(factory_name)
+        pipe = factory(name=name, model=model_spec, **comp_cfg)
+        nlp.add_pipe(name, pipe)
+    return nlp
+
+
+def init_nlp(config: Config, device: str = "cpu", sample_size: int = 128) -> Language:
+    """Build + initialize (label discovery on a corpus sample + param init) —
+    the contract of spaCy's init_nlp at `/root/reference/spacy_ray/worker.py:91`.
+    Deterministic: seeds torch RNG from [training.seed] BEFORE building so
+    every rank gets identical initial params (SURVEY.md §3.4 invariant)."""
+    cfg = config.interpolate()
+    seed = int(cfg.get("training", {}).get("seed", 0) or 0)
+    torch.manual_seed(seed)
+    nlp = build_nlp(config, device=device)
+    (train_corpus,) = resolve_dot_names(cfg, [cfg["training"]["train_corpus"]])
+    sample: List[Example] = []
+    for eg in train_corpus(nlp):
+        sample.append(eg)
+        if len(sample) >= sample_size:
+            break
+    for name, pipe in nlp.pipeline:
+        pipe.initialize(sample, nlp.device)
+    return nlp

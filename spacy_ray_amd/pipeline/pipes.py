@@ -1,0 +1,341 @@
+"""Trainable pipes: tok2vec, tagger, transition parser, NER.
+
+Re-design of spaCy's pipeline components for a single-autograd-graph step
+(SURVEY.md §3.2 disposition): the tok2vec pipe runs ONCE per batch; tagger/
+parser/NER are listeners consuming the shared [T, W] tensor; all losses sum
+into one backward so the distributed engine sees a single gradient pass to
+overlap with.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from spacy_ray_amd import _srx_cpu
+from spacy_ray_amd.models.batch import TokenBatch
+from spacy_ray_amd.models.parser_model import TransitionModel
+from spacy_ray_amd.vocab.doc import Doc, Example, biluo_to_codes, codes_to_biluo
+
+NEG_INF = -1e30
+
+
+class TrainablePipe:
+    name: str
+    listens_to: Optional[str] = None
+
+    def __init__(self) -> None:
+        self.module: Optional[nn.Module] = None
+        self.cfg: Dict = {}
+
+    def initialize(self, examples: Sequence[Example], device) -> None:
+        raise NotImplementedError
+
+    def get_loss(self, examples, t2v, batch) -> Tuple[torch.Tensor, float]:
+        """-> (loss tensor in the autograd graph, float count for logging)"""
+        raise NotImplementedError
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        raise NotImplementedError
+
+    # ------- serialization hooks (labels etc.); params live in module
+    def state_cfg(self) -> Dict:
+        return dict(self.cfg)
+
+    def load_cfg(self, cfg: Dict, device) -> None:
+        self.cfg = dict(cfg)
+
+
+class Tok2VecPipe(TrainablePipe):
+    name = "tok2vec"
+
+    def __init__(self, name: str, spec) -> None:
+        super().__init__()
+        self.name = name
+        self.spec = spec
+        self.width = spec.width
+
+    def initialize(self, examples, device) -> None:
+        if self.module is None:
+            self.module = self.spec.build().to(device)
+
+    def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
+        return self.module(batch, drop=drop)
+
+    def get_loss(self, examples, t2v, batch):
+        return t2v.new_zeros(()), 0.0
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        pass  # downstream pipes consume t2v directly
+
+
+class TaggerHead(nn.Module):
+    def __init__(self, width: int, n_tags: int):
+        super().__init__()
+        self.output = nn.Linear(width, n_tags)
+        nn.init.zeros_(self.output.weight)
+        nn.init.zeros_(self.output.bias)
+
+    def forward(self, X):
+        return self.output(X)
+
+
+class TaggerPipe(TrainablePipe):
+    name = "tagger"
+    listens_to = "tok2vec"
+
+    def __init__(self, name: str, spec) -> None:
+        super().__init__()
+        self.name = name
+        self.width = spec.width
+        self.labels: List[str] = []
+        self.label2id: Dict[str, int] = {}
+
+    def initialize(self, examples, device) -> None:
+        if not self.labels:
+            labels = set()
+            for eg in examples:
+                if eg.reference.tags:
+                    labels.update(eg.reference.tags)
+            self.labels = sorted(labels)
+            self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.cfg["labels"] = self.labels
+        if self.module is None:
+            self.module = TaggerHead(self.width, max(1, len(self.labels))).to(device)
+
+    def load_cfg(self, cfg, device) -> None:
+        super().load_cfg(cfg, device)
+        self.labels = list(cfg.get("labels", []))
+        self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.module = TaggerHead(self.width, max(1, len(self.labels))).to(device)
+
+    def _gold_ids(self, examples) -> np.ndarray:
+        ids = []
+        for eg in examples:
+            tags = eg.reference.tags or ["" for _ in range(len(eg.reference))]
+            ids.extend(self.label2id.get(t, -1) for t in tags)
+        return np.asarray(ids, dtype=np.int64)
+
+    def get_loss(self, examples, t2v, batch):
+        scores = self.module(t2v)  # [T, nT]
+        gold = torch.from_numpy(self._gold_ids(examples)).to(scores.device)
+        n = int((gold >= 0).sum())
+        loss = torch.nn.functional.cross_entropy(
+            scores.float(), gold, ignore_index=-1, reduction="sum"
+        ) / max(1, n)
+        return loss, float(n)
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        with torch.no_grad():
+            pred = self.module(t2v).argmax(dim=-1).cpu().numpy()
+        off = 0
+        for doc in docs:
+            n = len(doc)
+            doc.tags = [self.labels[i] if self.labels else "" for i in pred[off:off + n]]
+            off += n
+
+
+class _TransitionPipeBase(TrainablePipe):
+    """Shared greedy transition loop for parser and NER: per step, the C++
+    batch object yields features/valid/costs for ALL states; the GPU scores
+    them (fused gather+maxout + upper GEMM); training follows the
+    best-scoring min-cost action; loss = CE of softmax-over-valid against the
+    uniform min-cost target (contract of spaCy's parser loss,
+    SURVEY.md §2.2 N8)."""
+
+    listens_to = "tok2vec"
+
+    def __init__(self, name: str, spec) -> None:
+        super().__init__()
+        self.name = name
+        self.spec = spec
+        self.width = spec.width
+        self.hidden_width = getattr(spec, "hidden_width", 64)
+        self.nF = getattr(spec, "nF", 13)
+        self.labels: List[str] = []
+        self.label2id: Dict[str, int] = {}
+
+    def _build_module(self, device) -> None:
+        if self.module is None:
+            self.module = TransitionModel(
+                self.width, hidden_width=self.hidden_width, nF=self.nF
+            ).to(device)
+            self.module.initialize_output(self._n_actions())
+            self.module.to(device)
+
+    def load_cfg(self, cfg, device) -> None:
+        super().load_cfg(cfg, device)
+        self.labels = list(cfg.get("labels", []))
+        self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self._build_module(device)
+
+    # ---- subclass hooks
+    def _n_actions(self) -> int:
+        raise NotImplementedError
+
+    def _make_states(self, lengths: np.ndarray):
+        raise NotImplementedError
+
+    def _set_gold(self, states, examples) -> None:
+        raise NotImplementedError
+
+    def _annotate(self, docs, states) -> None:
+        raise NotImplementedError
+
+    # ---- shared machinery
+    def _step_loop(self, states, t2v, train: bool):
+        device = t2v.device
+        T = t2v.shape[0]
+        pre = self.module.precompute(t2v)
+        loss_terms: List[torch.Tensor] = []
+        n_states_total = 0
+        max_steps = 4 * T + 16
+        for _ in range(max_steps):
+            final = states.is_final()
+            if final.all():
+                break
+            active = final == 0
+            feats = states.features()
+            valid = states.valid().astype(bool)
+            feats_t = torch.from_numpy(
+                np.where(feats < 0, T, feats).astype(np.int64)
+            ).to(device)
+            scores = self.module.score(pre, feats_t)  # [S, A]
+            valid_t = torch.from_numpy(valid).to(device)
+            if train:
+                costs = states.costs()
+                cmin = costs.min(axis=1, keepdims=True)
+                is_gold = (costs <= cmin + 1e-6) & valid
+                counts = is_gold.sum(axis=1, keepdims=True)
+                ok = (counts[:, 0] > 0) & active
+                target = is_gold.astype(np.float32) / np.maximum(counts, 1)
+                target_t = torch.from_numpy(target).to(device)
+                logp = torch.log_softmax(
+                    scores.float().masked_fill(~valid_t, NEG_INF), dim=-1
+                )
+                row_loss = -(target_t * logp).sum(dim=-1)
+                ok_t = torch.from_numpy(ok).to(device)
+                loss_terms.append(row_loss.masked_fill(~ok_t, 0).sum())
+                n_states_total += int(active.sum())
+                with torch.no_grad():
+                    s_np = scores.detach().float().cpu().numpy()
+                choose_from = np.where(is_gold, s_np, NEG_INF)
+                # states where no valid min-cost action exists: any valid one
+                fallback = np.where(valid, s_np, NEG_INF)
+                choose_from = np.where(counts > 0, choose_from, fallback)
+            else:
+                with torch.no_grad():
+                    s_np = scores.float().cpu().numpy()
+                choose_from = np.where(valid, s_np, NEG_INF)
+            actions = choose_from.argmax(axis=1).astype(np.int32)
+            actions[~active] = -1
+            states.advance(actions)
+        if train:
+            total = (
+                torch.stack(loss_terms).sum() / max(1, n_states_total)
+                if loss_terms
+                else t2v.new_zeros(())
+            )
+            return total, float(n_states_total)
+        return None, 0.0
+
+    def get_loss(self, examples, t2v, batch):
+        lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
+        states = self._make_states(lengths)
+        self._set_gold(states, examples)
+        return self._step_loop(states, t2v, train=True)
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
+        states = self._make_states(lengths)
+        self._step_loop(states, t2v, train=False)
+        self._annotate(docs, states)
+
+
+class ParserPipe(_TransitionPipeBase):
+    name = "parser"
+
+    def initialize(self, examples, device) -> None:
+        if not self.labels:
+            labels = set()
+            for eg in examples:
+                if eg.reference.deps:
+                    labels.update(d for d in eg.reference.deps if d != "ROOT")
+            self.labels = sorted(labels) or ["dep"]
+            self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.cfg["labels"] = self.labels
+        self._build_module(device)
+
+    def _n_actions(self) -> int:
+        return 2 + 2 * len(self.labels)
+
+    def _make_states(self, lengths):
+        return _srx_cpu.ArcEagerBatch(lengths, len(self.labels))
+
+    def _set_gold(self, states, examples) -> None:
+        heads = np.concatenate([
+            eg.reference.heads if eg.reference.heads is not None
+            else np.full(len(eg.reference), -1, dtype=np.int32)
+            for eg in examples
+        ]).astype(np.int32)
+        labs = []
+        for eg in examples:
+            deps = eg.reference.deps or ["dep"] * len(eg.reference)
+            labs.extend(self.label2id.get(d, 0) for d in deps)
+        states.set_gold(heads, np.asarray(labs, dtype=np.int32))
+
+    def _annotate(self, docs, states) -> None:
+        heads = states.heads()
+        labels = states.labels()
+        off = 0
+        for doc in docs:
+            n = len(doc)
+            doc.heads = heads[off:off + n].copy()
+            doc.deps = [
+                self.labels[l] if 0 <= l < len(self.labels) else "ROOT"
+                for l in labels[off:off + n]
+            ]
+            for i in range(n):
+                if doc.heads[i] == -1:
+                    doc.deps[i] = "ROOT"
+            off += n
+
+
+class NerPipe(_TransitionPipeBase):
+    name = "ner"
+
+    def initialize(self, examples, device) -> None:
+        if not self.labels:
+            labels = set()
+            for eg in examples:
+                if eg.reference.ents:
+                    for tag in eg.reference.ents:
+                        if tag not in ("O", "-", None, ""):
+                            labels.add(tag.partition("-")[2])
+            self.labels = sorted(labels) or ["ENT"]
+            self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.cfg["labels"] = self.labels
+        self._build_module(device)
+
+    def _n_actions(self) -> int:
+        return 1 + 4 * len(self.labels)
+
+    def _make_states(self, lengths):
+        return _srx_cpu.BiluoBatch(lengths, len(self.labels))
+
+    def _set_gold(self, states, examples) -> None:
+        codes = np.concatenate([
+            biluo_to_codes(eg.reference.ents or ["O"] * len(eg.reference), self.label2id)
+            for eg in examples
+        ]).astype(np.int32)
+        states.set_gold(codes)
+
+    def _annotate(self, docs, states) -> None:
+        tags = states.tags()
+        off = 0
+        for doc in docs:
+            n = len(doc)
+            doc.ents = codes_to_biluo(tags[off:off + n], self.labels)
+            off += n

@@ -1,0 +1,83 @@
+"""Scoring: tag accuracy, UAS/LAS, NER P/R/F (spaCy Scorer contract for the
+score keys the reference's loggers/score_weights consume)."""
+from __future__ import annotations
+
+from typing import Dict, List, Sequence, Set, Tuple
+
+from spacy_ray_amd.vocab.doc import Example
+
+
+def _ents_to_spans(ents: List[str]) -> Set[Tuple[int, int, str]]:
+    spans = set()
+    if not ents:
+        return spans
+    start, label = None, None
+    for i, tag in enumerate(ents):
+        if tag is None or tag == "O" or tag == "-" or tag == "":
+            start, label = None, None
+            continue
+        kind, _, lab = tag.partition("-")
+        if kind == "U":
+            spans.add((i, i + 1, lab))
+            start, label = None, None
+        elif kind == "B":
+            start, label = i, lab
+        elif kind == "L" and start is not None and lab == label:
+            spans.add((start, i + 1, lab))
+            start, label = None, None
+        elif kind == "I" and start is not None and lab == label:
+            continue
+        else:  # inconsistent sequence — drop the open span
+            start, label = None, None
+    return spans
+
+
+def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict[str, float]:
+    scores: Dict[str, float] = {}
+    if "tagger" in pipe_names:
+        correct = total = 0
+        for eg in examples:
+            gold, pred = eg.reference.tags, eg.predicted.tags
+            if gold is None or pred is None:
+                continue
+            for g, p in zip(gold, pred):
+                total += 1
+                correct += int(g == p)
+        scores["tag_acc"] = correct / total if total else 0.0
+    if "parser" in pipe_names:
+        uas = las = total = 0
+        for eg in examples:
+            gh, ph = eg.reference.heads, eg.predicted.heads
+            gd, pd = eg.reference.deps, eg.predicted.deps
+            if gh is None or ph is None:
+                continue
+            for i in range(len(eg.reference)):
+                total += 1
+                if int(gh[i]) == int(ph[i]):
+                    uas += 1
+                    if gd and pd and gd[i] == pd[i]:
+                        las += 1
+        scores["dep_uas"] = uas / total if total else 0.0
+        scores["dep_las"] = las / total if total else 0.0
+    if "ner" in pipe_names:
+        tp = fp = fn = 0
+        for eg in examples:
+            gold = _ents_to_spans(eg.reference.ents or [])
+            pred = _ents_to_spans(eg.predicted.ents or [])
+            tp += len(gold & pred)
+            fp += len(pred - gold)
+            fn += len(gold - pred)
+        p = tp / (tp + fp) if tp + fp else 0.0
+        r = tp / (tp + fn) if tp + fn else 0.0
+        scores["ents_p"] = p
+        scores["ents_r"] = r
+        scores["ents_f"] = 2 * p * r / (p + r) if p + r else 0.0
+    return scores
+
+
+def weighted_score(scores: Dict[str, float], weights: Dict[str, float]) -> float:
+    total = 0.0
+    for key, w in (weights or {}).items():
+        if w and key in scores and scores[key] is not None:
+            total += w * scores[key]
+    return total

@@ -1,0 +1,161 @@
+"""Doc / Vocab / Example.
+
+MI355X-first re-design of spaCy's Cython Doc/Vocab: a Doc
+is a contiguous struct-of-arrays — token texts plus a precomputed
+(n_tokens, 4) uint64 attr-hash matrix that ships to the GPU as one int64
+tensor.  Annotations (tags / heads / deps / BILUO ents) are plain arrays.
+Examples pair a predicted Doc with a reference (gold) Doc, the contract of
+spaCy's training Example consumed by nlp.update.
+"""
+from __future__ import annotations
+
+import re
+from typing import Dict, Iterator, List, Optional, Sequence
+
+import numpy as np
+
+from .attrs import extract_attr_hashes
+from .strings import StringStore
+
+_TOKEN_RE = re.compile(r"\w+|[^\w\s]", re.UNICODE)
+
+
+class Vocab:
+    def __init__(self, lang: str = "xx") -> None:
+        self.lang = lang
+        self.strings = StringStore()
+
+    def __repr__(self) -> str:
+        return f"Vocab(lang={self.lang!r}, strings={len(self.strings)})"
+
+
+class Doc:
+    __slots__ = (
+        "vocab", "words", "spaces", "attr_hashes",
+        "tags", "heads", "deps", "ents",
+        "tensor", "user_data",
+    )
+
+    def __init__(
+        self,
+        vocab: Vocab,
+        words: Sequence[str],
+        *,
+        spaces: Optional[Sequence[bool]] = None,
+        tags: Optional[Sequence[str]] = None,
+        heads: Optional[Sequence[int]] = None,
+        deps: Optional[Sequence[str]] = None,
+        ents: Optional[Sequence[str]] = None,  # per-token BILUO strings, e.g. "B-ORG"/"O"
+    ) -> None:
+        self.vocab = vocab
+        self.words = list(words)
+        self.spaces = list(spaces) if spaces is not None else [True] * len(self.words)
+        self.attr_hashes = extract_attr_hashes(self.words)
+        self.tags = list(tags) if tags is not None else None
+        self.heads = np.asarray(heads, dtype=np.int32) if heads is not None else None
+        self.deps = list(deps) if deps is not None else None
+        self.ents = list(ents) if ents is not None else None
+        self.tensor: Optional[np.ndarray] = None
+        self.user_data: Dict = {}
+
+    def __len__(self) -> int:
+        return len(self.words)
+
+    def __iter__(self) -> Iterator[str]:
+        return iter(self.words)
+
+    @property
+    def text(self) -> str:
+        out = []
+        for w, sp in zip(self.words, self.spaces):
+            out.append(w)
+            if sp:
+                out.append(" ")
+        return "".join(out).rstrip()
+
+    def copy_unannotated(self) -> "Doc":
+        return Doc(self.vocab, self.words, spaces=self.spaces)
+
+    def to_dict(self) -> Dict:
+        return {
+            "words": self.words,
+            "spaces": self.spaces,
+            "tags": self.tags,
+            "heads": self.heads.tolist() if self.heads is not None else None,
+            "deps": self.deps,
+            "ents": self.ents,
+        }
+
+    @classmethod
+    def from_dict(cls, vocab: Vocab, data: Dict) -> "Doc":
+        return cls(
+            vocab,
+            data["words"],
+            spaces=data.get("spaces"),
+            tags=data.get("tags"),
+            heads=data.get("heads"),
+            deps=data.get("deps"),
+            ents=data.get("ents"),
+        )
+
+
+class Example:
+    """predicted + reference Doc pair (contract of spaCy's Example)."""
+
+    __slots__ = ("predicted", "reference")
+
+    def __init__(self, predicted: Doc, reference: Doc) -> None:
+        self.predicted = predicted
+        self.reference = reference
+
+    @classmethod
+    def from_doc(cls, doc: Doc) -> "Example":
+        return cls(doc.copy_unannotated(), doc)
+
+    def __len__(self) -> int:
+        return len(self.reference)
+
+    @property
+    def x(self) -> Doc:
+        return self.predicted
+
+    @property
+    def y(self) -> Doc:
+        return self.reference
+
+
+def simple_tokenize(vocab: Vocab, text: str) -> Doc:
+    """Rule-based fallback tokenizer (word chars / single punct).  Training
+    from pre-annotated corpora never calls this (SURVEY.md §2.2 N5 note:
+    tokenization is off the hot path)."""
+    words = _TOKEN_RE.findall(text)
+    return Doc(vocab, words)
+
+
+def biluo_to_codes(ents: Optional[Sequence[str]], label2id: Dict[str, int]) -> np.ndarray:
+    """Per-token BILUO strings -> int codes: 0=O; for type t: 1+4t=B, 2+4t=I,
+    3+4t=L, 4+4t=U (layout shared with ops/csrc/transitions.cpp BiluoBatch)."""
+    if ents is None:
+        return np.zeros(0, dtype=np.int32)
+    kinds = {"B": 0, "I": 1, "L": 2, "U": 3}
+    out = np.zeros(len(ents), dtype=np.int32)
+    for i, tag in enumerate(ents):
+        if tag in (None, "O", "-", ""):
+            out[i] = 0
+        else:
+            kind, _, label = tag.partition("-")
+            t = label2id[label]
+            out[i] = 1 + 4 * t + kinds[kind]
+    return out
+
+
+def codes_to_biluo(codes: np.ndarray, id2label: List[str]) -> List[str]:
+    kinds = "BILU"
+    out = []
+    for c in codes.tolist():
+        if c <= 0:
+            out.append("O")
+        else:
+            t, k = (c - 1) // 4, (c - 1) % 4
+            out.append(f"{kinds[k]}-{id2label[t]}")
+    return out
