@@ -1,0 +1,163 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: en_core CNN (tok2vec+tagger+parser+NER) training
+words/sec, whole node — the BASELINE.json metric.
+
+    python bench.py --gpus N --steps K --warmup W
+
+Driver contract: for N>1 this is launched under torch.distributed.run with
+one rank per GPU (RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* in env).  W untimed
+warmup steps, then exactly K timed steps bracketed by barrier +
+torch.cuda.synchronize on both sides; time is MAX over ranks; rank 0 prints
+one JSON line.  Synthetic data (no network for datasets), random-init
+weights, bf16 on GPU.  Weak scaling: per-GPU work is fixed as N grows.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def build_batches(nlp, *, batch_words: int, n_batches: int, seed: int,
+                  words_per_doc: int, vocab_size: int):
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.vocab.doc import Example
+
+    docs = make_synthetic_docs(
+        nlp.vocab,
+        n_docs=max(64, (batch_words * n_batches) // max(1, words_per_doc)),
+        words_per_doc=words_per_doc,
+        vocab_size=vocab_size,
+        n_tags=50,
+        n_deps=40,
+        n_ent_types=4,
+        seed=seed,
+        world_seed=0,
+    )
+    batches = []
+    cur, n = [], 0
+    for d in docs:
+        cur.append(Example.from_doc(d))
+        n += len(d)
+        if n >= batch_words:
+            batches.append(cur)
+            cur, n = [], 0
+            if len(batches) >= n_batches:
+                break
+    if not batches:
+        batches = [[Example.from_doc(d) for d in docs]]
+    return batches
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch-words", type=int, default=8000,
+                    help="words per step per GPU (weak scaling)")
+    ap.add_argument("--words-per-doc", type=int, default=20)
+    ap.add_argument("--vocab-size", type=int, default=5000)
+    ap.add_argument("--config", type=str, default="examples/configs/en_core_cnn.cfg")
+    ap.add_argument("--profile-steps", type=int, default=0,
+                    help="if >0, run only this many unsynchronized steps (for rocprof)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.parallel.comm import init_comm_from_env
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        device = "cpu"
+    comm = init_comm_from_env()
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    cfg_path = args.config if os.path.isabs(args.config) else os.path.join(here, args.config)
+    config = Config.from_disk(cfg_path)
+    nlp = init_nlp(config, device=device, sample_size=64)
+    T = resolve(config.interpolate()["training"], validate=False)
+    engine = ZeRO1Engine(nlp, T["optimizer"], comm)
+    torch.manual_seed(1234 + rank)
+
+    batches = build_batches(
+        nlp, batch_words=args.batch_words, n_batches=8,
+        seed=100 + rank, words_per_doc=args.words_per_doc,
+        vocab_size=args.vocab_size,
+    )
+    words_per_step = sum(len(eg) for eg in batches[0])
+
+    def step(i: int) -> None:
+        engine.accumulate(batches[i % len(batches)], drop=0.1)
+        engine.apply_step()
+
+    if args.profile_steps:
+        for i in range(args.profile_steps):
+            step(i)
+        if use_cuda:
+            torch.cuda.synchronize()
+        return
+
+    for i in range(args.warmup):
+        step(i)
+
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    comm.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        import torch.distributed as dist
+
+        t_dev = t.to(device) if comm.backend == "nccl" else t
+        dist.all_reduce(t_dev, op=dist.ReduceOp.MAX)
+        elapsed = float(t_dev.item())
+
+    if rank == 0:
+        total_words = words_per_step * world * args.steps
+        value = total_words / elapsed
+        out = {
+            "metric": "words/sec (whole node) en_core CNN tagger+parser+NER train",
+            "value": value,
+            "unit": "words/s",
+            "n_gpus": world if use_cuda else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "en_core_web_cnn (MultiHashEmbed+MaxoutWindowEncoder w96d4 + tagger + parser + ner)",
+                "global_batch": words_per_step * world,
+                "seq_len": args.words_per_doc,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

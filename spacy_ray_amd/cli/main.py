@@ -1,0 +1,102 @@
+"""CLI: ``spacy-mi ray train config.cfg --n-workers N [...]``.
+
+Flag surface of the reference command (`/root/reference/spacy_ray/
+train_cli.py:26-53`): config_path positional; --code, --output, --n-workers,
+--address, --gpu-id, --verbose; trailing ``--dotted.key value`` config
+overrides.  ``--output`` is actually wired here (the reference parses it and
+drops it, train_cli.py:41).  The command is the launcher (L5): it spawns one
+worker process per GPU/rank and supervises them.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import sys
+from pathlib import Path
+from typing import List, Optional
+
+import typer
+
+from spacy_ray_amd.config.config import Config, parse_config_overrides
+
+app = typer.Typer(name="spacy-mi", no_args_is_help=True)
+ray_app = typer.Typer(name="ray", no_args_is_help=True,
+                      help="Distributed/parallel training (reference-compatible command group)")
+app.add_typer(ray_app)
+
+
+@ray_app.command(
+    "train",
+    context_settings={"allow_extra_args": True, "ignore_unknown_options": True},
+)
+def ray_train_cli(
+    ctx: typer.Context,
+    config_path: Path = typer.Argument(..., help="Path to config file"),
+    code_path: Optional[Path] = typer.Option(None, "--code", "-c", help="Path to Python file with additional code to be imported"),
+    output_path: Optional[Path] = typer.Option(None, "--output", "-o", help="Output directory for checkpoints"),
+    n_workers: int = typer.Option(1, "--n-workers", "-n", help="Number of workers (1 process per GPU)"),
+    address: Optional[str] = typer.Option(None, "--address", "-a", help="Rendezvous address host[:port] (multi-node)"),
+    use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
+    verbose: bool = typer.Option(False, "--verbose", "-V", help="Display more information"),
+):
+    """Train a pipeline with N data-parallel workers over RCCL/xGMI."""
+    logging.basicConfig(level=logging.DEBUG if verbose else logging.ERROR)
+    overrides = parse_config_overrides(list(ctx.args))
+    config = Config.from_disk(config_path, overrides=overrides)
+    raise SystemExit(
+        ray_train(config, config_path=config_path, output_path=output_path,
+                  code_path=code_path, n_workers=n_workers, address=address,
+                  use_gpu=use_gpu, overrides=overrides)
+    )
+
+
+def ray_train(
+    config: Config,
+    *,
+    config_path: Path,
+    output_path: Optional[Path] = None,
+    code_path: Optional[Path] = None,
+    n_workers: int = 1,
+    address: Optional[str] = None,
+    use_gpu: int = -1,
+    overrides: Optional[dict] = None,
+) -> int:
+    """Launcher (contract of `/root/reference/spacy_ray/train_cli.py:56-91`)."""
+    from spacy_ray_amd.parallel.launcher import launch_workers
+
+    if n_workers <= 1 and not address:
+        # single process: run in-process, no process group
+        from spacy_ray_amd.train.worker import distributed_train
+
+        distributed_train(
+            config,
+            output_path=output_path,
+            use_gpu=use_gpu,
+            code_path=code_path,
+            metrics_path=(Path(output_path) / "metrics.jsonl") if output_path else None,
+        )
+        return 0
+    worker_cmd = [sys.executable, "-m", "spacy_ray_amd.cli.worker", str(config_path)]
+    if output_path:
+        worker_cmd += ["--output", str(output_path)]
+    if code_path:
+        worker_cmd += ["--code", str(code_path)]
+    worker_cmd += ["--gpu-id", str(use_gpu)]
+    if overrides:
+        worker_cmd += ["--overrides-json", json.dumps(overrides)]
+    master_addr, master_port = "127.0.0.1", None
+    if address:
+        host, _, port = address.partition(":")
+        master_addr = host or "127.0.0.1"
+        master_port = int(port) if port else None
+    return launch_workers(worker_cmd, n_workers, master_addr=master_addr,
+                          master_port=master_port)
+
+
+def main() -> None:
+    app()
+
+
+if __name__ == "__main__":
+    main()

@@ -72,17 +72,22 @@ def make_synthetic_docs(
     n_deps: int = 40,
     n_ent_types: int = 4,
     seed: int = 0,
+    world_seed: int = 0,
 ) -> List[Doc]:
+    """`world_seed` fixes the synthetic language itself (word->tag/dep/ent
+    mappings) so train/dev corpora with different `seed`s sample different
+    docs from the SAME learnable world."""
     rng = random.Random(seed)
     lexicon = [f"w{i}" for i in range(vocab_size)]
     # Zipf sampling over the lexicon
     ranks = np.arange(1, vocab_size + 1, dtype=np.float64)
     probs = 1.0 / ranks
     probs /= probs.sum()
+    world_rng = np.random.RandomState(world_seed)
+    tag_of_word = world_rng.randint(0, n_tags, size=vocab_size)
+    dep_of_word = world_rng.randint(0, n_deps, size=vocab_size)
+    ent_of_word = world_rng.randint(0, n_ent_types, size=vocab_size)
     np_rng = np.random.RandomState(seed)
-    tag_of_word = np_rng.randint(0, n_tags, size=vocab_size)
-    dep_of_word = np_rng.randint(0, n_deps, size=vocab_size)
-    ent_of_word = np_rng.randint(0, n_ent_types, size=vocab_size)
 
     docs: List[Doc] = []
     for _ in range(n_docs):
@@ -129,6 +134,7 @@ def create_synthetic_corpus(
     n_deps: int = 40,
     n_ent_types: int = 4,
     seed: int = 0,
+    world_seed: int = 0,
     shuffle: bool = True,
 ):
     cache: dict = {}
@@ -144,6 +150,7 @@ def create_synthetic_corpus(
                 n_deps=n_deps,
                 n_ent_types=n_ent_types,
                 seed=seed,
+                world_seed=world_seed,
             )
             cache["epoch"] = 0
         docs = list(cache["docs"])

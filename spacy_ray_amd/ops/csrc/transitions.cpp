@@ -202,16 +202,19 @@ struct ArcEagerBatch {
           if (gh[d] == s0) c_reduce += 1;
         if (b >= 0) {
           // LEFT-ARC(b -> s0): lose gold deps of s0 in buffer, and gold head
-          // of s0 if it is in buffer beyond b (head==b is the gold arc).
+          // of s0 if it is in buffer beyond b (head==b is the gold arc; a
+          // gold head already in the stack was lost earlier — additional
+          // cost only).
           for (int32_t d = st.buf; d < st.len; d++)
             if (gh[d] == s0) c_la += 1;
           if (gh[s0] >= 0 && gh[s0] > b) c_la += 1;
-          if (gh[s0] >= 0 && in_stack(gh[s0])) c_la += 1;
           // RIGHT-ARC(s0 -> b): lose gold head of b elsewhere (stack != s0 or
-          // buffer), and headless stack items whose gold head is b.
+          // buffer), and ALL headless stack items (s0 included) whose gold
+          // head is b — after RA, b is no longer in the buffer so none of
+          // them can take b as head via LEFT-ARC.
           if (gh[b] >= 0 && gh[b] != s0 && (in_stack(gh[b]) || gh[b] > b)) c_ra += 1;
           for (int32_t s : st.stack)
-            if (s != s0 && st.head[s] == -1 && gh[s] == b) c_ra += 1;
+            if (st.head[s] == -1 && gh[s] == b) c_ra += 1;
         }
       }
       r(i, 0) = v[0] ? c_shift : KInvalid;

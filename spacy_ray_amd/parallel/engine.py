@@ -1,0 +1,232 @@
+"""ZeRO-1 engine: flat-buffer data parallelism with sharded Adam.
+
+The MI355X re-architecture of the reference's parameter-sharding scheme
+(SURVEY.md §2.3): where spacy-ray partitions Thinc param keys across Ray
+actors and pushes version-stamped grads/params asynchronously
+(`/root/reference/spacy_ray/proxies.py:9-133`, `util.py:57-75`), this engine
+keeps the same math — each rank owns a shard of the parameters and runs the
+optimizer only on it — expressed synchronously as collectives over xGMI:
+
+  backward  -> per-bucket gradient reduce-scatter (RCCL) launched from
+               post-accumulate-grad hooks on a side HIP stream, overlapping
+               the rest of backward;
+  step      -> global-norm clip (one scalar all-reduce) + Adam on the
+               rank's contiguous fp32 master shard (bf16 params on GPU);
+  publish   -> per-bucket parameter all-gather.
+
+Layout: all trainable params live in ONE flat device buffer, partitioned
+into buckets of ~bucket_bytes at parameter boundaries; each bucket is padded
+so `world` divides it, and rank r owns slice r of every bucket.  Optimizer
+state (master/m/v) is one contiguous fp32 tensor per kind covering the
+rank's slices — so the whole Adam step is a handful of elementwise kernels
+over single contiguous tensors (one fused HIP kernel in ops/kernels).
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import torch
+
+from spacy_ray_amd.train.optimizer import AdamSpec
+from .comm import Comm, LocalComm
+
+
+class _Bucket:
+    __slots__ = ("index", "start", "end", "per", "shard_off", "params", "ready", "launched")
+
+    def __init__(self, index: int, start: int) -> None:
+        self.index = index
+        self.start = start
+        self.end = start
+        self.per = 0
+        self.shard_off = 0
+        self.params: List[torch.nn.Parameter] = []
+        self.ready = 0
+        self.launched = False
+
+
+class ZeRO1Engine:
+    def __init__(
+        self,
+        nlp,
+        spec: AdamSpec,
+        comm: Optional[Comm] = None,
+        *,
+        bucket_bytes: int = 16 << 20,
+        overlap: bool = True,
+        compute_dtype: Optional[torch.dtype] = None,
+    ) -> None:
+        self.nlp = nlp
+        self.spec = spec
+        self.comm = comm or LocalComm()
+        self.device = nlp.device
+        self.is_cuda = self.device.type == "cuda"
+        if compute_dtype is None:
+            compute_dtype = torch.bfloat16 if self.is_cuda else torch.float32
+        self.dtype = compute_dtype
+        self.overlap = overlap and self.is_cuda
+        self.module = nlp.torch_module()
+        self.step_count = 0
+        self._sync = False
+
+        world = self.comm.world
+        align = 64 * world
+        named = [(n, p) for n, p in self.module.named_parameters() if p.requires_grad]
+        self.param_names = [n for n, _ in named]
+        params = [p for _, p in named]
+
+        # ---- bucket construction at parameter boundaries
+        self.buckets: List[_Bucket] = []
+        b = _Bucket(0, 0)
+        offset = 0
+        self._offsets: List[int] = []
+        for p in params:
+            n = p.numel()
+            self._offsets.append(offset)
+            b.params.append(p)
+            offset += n
+            if (offset - b.start) * self.dtype.itemsize >= bucket_bytes:
+                offset = -(-offset // align) * align  # pad bucket end
+                b.end = offset
+                self.buckets.append(b)
+                b = _Bucket(len(self.buckets), offset)
+        if b.params:
+            offset = -(-offset // align) * align
+            b.end = offset
+            self.buckets.append(b)
+        total = offset
+        shard_off = 0
+        for b in self.buckets:
+            b.per = (b.end - b.start) // world
+            b.shard_off = shard_off
+            shard_off += b.per
+        self.shard_elems = shard_off
+
+        # ---- flat buffers
+        self.flat_param = torch.zeros(total, dtype=self.dtype, device=self.device)
+        self.flat_grad = torch.zeros(total, dtype=self.dtype, device=self.device)
+        for p, off in zip(params, self._offsets):
+            self.flat_param[off : off + p.numel()].copy_(p.data.reshape(-1).to(self.dtype))
+            p.data = self.flat_param[off : off + p.numel()].view(p.shape)
+            p.grad = self.flat_grad[off : off + p.numel()].view(p.shape)
+
+        # ---- sharded optimizer state (fp32)
+        r = self.comm.rank
+        self.master = torch.zeros(self.shard_elems, dtype=torch.float32, device=self.device)
+        for b in self.buckets:
+            src = self.flat_param[b.start + r * b.per : b.start + (r + 1) * b.per]
+            self.master[b.shard_off : b.shard_off + b.per].copy_(src.float())
+        self.exp_avg = torch.zeros_like(self.master)
+        self.exp_avg_sq = torch.zeros_like(self.master)
+        self.grad_shard = torch.zeros(self.shard_elems, dtype=self.dtype, device=self.device)
+        self._g32 = torch.zeros_like(self.master)
+
+        # ---- overlap machinery
+        self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
+        self._param_bucket: Dict[int, _Bucket] = {}
+        for bkt in self.buckets:
+            for p in bkt.params:
+                self._param_bucket[id(p)] = bkt
+        for p in params:
+            p.register_post_accumulate_grad_hook(self._grad_ready_hook)
+
+    # ------------------------------------------------------------ internals
+    def _grad_ready_hook(self, p: torch.nn.Parameter) -> None:
+        if not (self._sync and self.overlap):
+            return
+        bkt = self._param_bucket[id(p)]
+        bkt.ready += 1
+        if bkt.ready >= len(bkt.params) and not bkt.launched:
+            self._launch_bucket(bkt)
+
+    def _launch_bucket(self, bkt: _Bucket) -> None:
+        bkt.launched = True
+        r = self.comm.rank
+        seg = self.flat_grad[bkt.start : bkt.end]
+        out = self.grad_shard[bkt.shard_off : bkt.shard_off + bkt.per]
+        if self.is_cuda:
+            ev = torch.cuda.Event()
+            ev.record()
+            with torch.cuda.stream(self.comm_stream):
+                self.comm_stream.wait_event(ev)
+                self.comm.reduce_scatter_flat(seg, out)
+        else:
+            self.comm.reduce_scatter_flat(seg, out)
+
+    def _gather_bucket(self, bkt: _Bucket) -> None:
+        r = self.comm.rank
+        shard = self.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
+        self.comm.all_gather_flat(self.flat_param[bkt.start : bkt.end], shard)
+
+    # ------------------------------------------------------------- stepper
+    def accumulate(self, examples, drop: float = 0.0, losses: Optional[Dict] = None,
+                   sync: bool = True) -> None:
+        self._sync = sync
+        total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop)
+        total.backward()
+        self._sync = False
+
+    def apply_step(self) -> None:
+        # launch any bucket the hooks didn't (grad-less params, overlap off)
+        for bkt in self.buckets:
+            if not bkt.launched:
+                self._launch_bucket(bkt)
+        if self.is_cuda:
+            torch.cuda.current_stream().wait_stream(self.comm_stream)
+
+        s = self.spec
+        self._g32.copy_(self.grad_shard)
+        # global-norm clip: one scalar all-reduce over shard norms
+        if s.grad_clip:
+            sq = self._g32.pow(2).sum()
+            self.comm.all_reduce_(sq)
+            norm = sq.sqrt()
+            scale = torch.clamp(s.grad_clip / (norm + 1e-12), max=1.0)
+            self._g32.mul_(scale)
+        lr = s.lr(self.step_count)
+        t = self.step_count + 1
+        if s.L2:
+            if s.L2_is_weight_decay:
+                self.master.mul_(1.0 - lr * s.L2)
+            else:
+                self._g32.add_(self.master, alpha=s.L2)
+        self.exp_avg.mul_(s.beta1).add_(self._g32, alpha=1 - s.beta1)
+        self.exp_avg_sq.mul_(s.beta2).addcmul_(self._g32, self._g32, value=1 - s.beta2)
+        bc1 = 1 - s.beta1 ** t
+        bc2 = 1 - s.beta2 ** t
+        denom = (self.exp_avg_sq / bc2).sqrt_().add_(s.eps)
+        self.master.addcdiv_(self.exp_avg, denom, value=-lr / bc1)
+        # write back + republish
+        r = self.comm.rank
+        for bkt in self.buckets:
+            dst = self.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
+            dst.copy_(self.master[bkt.shard_off : bkt.shard_off + bkt.per].to(self.dtype))
+            bkt.ready = 0
+            bkt.launched = False
+        for bkt in self.buckets:
+            self._gather_bucket(bkt)
+        self.flat_grad.zero_()
+        self.grad_shard.zero_()
+        self.step_count += 1
+
+    # ------------------------------------------------- checkpoint interface
+    def state_dict(self) -> Dict:
+        return {
+            "step": self.step_count,
+            "master": self.master,
+            "exp_avg": self.exp_avg,
+            "exp_avg_sq": self.exp_avg_sq,
+        }
+
+    def load_state_dict(self, state: Dict) -> None:
+        self.step_count = int(state["step"])
+        self.master.copy_(state["master"])
+        self.exp_avg.copy_(state["exp_avg"])
+        self.exp_avg_sq.copy_(state["exp_avg_sq"])
+        r = self.comm.rank
+        for bkt in self.buckets:
+            dst = self.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
+            dst.copy_(self.master[bkt.shard_off : bkt.shard_off + bkt.per].to(self.dtype))
+        for bkt in self.buckets:
+            self._gather_bucket(bkt)

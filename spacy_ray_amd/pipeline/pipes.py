@@ -61,6 +61,11 @@ class Tok2VecPipe(TrainablePipe):
         if self.module is None:
             self.module = self.spec.build().to(device)
 
+    def load_cfg(self, cfg, device) -> None:
+        super().load_cfg(cfg, device)
+        if self.module is None:
+            self.module = self.spec.build().to(device)
+
     def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
         return self.module(batch, drop=drop)
 

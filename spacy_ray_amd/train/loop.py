@@ -70,8 +70,9 @@ def train_while_improving(
             before_update(nlp, {"step": step, "epoch": epoch})
         n_words = sum(len(eg) for eg in batch)
         words_seen += n_words
-        for sub in _subdivide(batch, accumulate_gradient):
-            stepper.accumulate(sub, drop=dropout, losses=losses)
+        subs = _subdivide(batch, accumulate_gradient)
+        for i, sub in enumerate(subs):
+            stepper.accumulate(sub, drop=dropout, losses=losses, sync=(i == len(subs) - 1))
         stepper.apply_step()
         if (step % eval_frequency) == 0 and step > 0 or (
             eval_frequency == 1 and step == 0

@@ -20,7 +20,8 @@ class SimpleStepper:
         self.module = nlp.torch_module()
         self.opt = SimpleAdam(self.module, spec)
 
-    def accumulate(self, examples, drop: float = 0.0, losses: Optional[Dict] = None) -> None:
+    def accumulate(self, examples, drop: float = 0.0, losses: Optional[Dict] = None,
+                   sync: bool = True) -> None:
         total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop)
         total.backward()
 

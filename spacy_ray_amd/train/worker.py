@@ -1,0 +1,192 @@
+"""Per-rank training worker.
+
+The MI355X re-architecture of the reference Worker actor
+(`/root/reference/spacy_ray/worker.py:23-262`): one OS process per GPU
+(launched by parallel/launcher.py or torchrun), RCCL process group instead of
+Ray RPC endpoints, synchronous ZeRO-1 steps instead of the async proxy.
+Responsibilities kept 1:1: build nlp from config, resolve
+the training schema, run the train_while_improving iterator, rank-0 eval +
+score broadcast (the Evaluator actor's role, worker.py:281-300 -> C3
+broadcast), rank-0 checkpointing (wired, unlike the reference's TODO at
+train_cli.py:41), cluster-scaled words logging (worker.py:308-311).
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+from pathlib import Path
+from typing import Dict, Iterator, Optional
+
+import torch
+
+from spacy_ray_amd.config.config import Config, resolve, resolve_dot_names
+from spacy_ray_amd.parallel.comm import Comm, LocalComm, init_comm_from_env
+from spacy_ray_amd.parallel.engine import ZeRO1Engine
+from spacy_ray_amd.pipeline.language import init_nlp
+from spacy_ray_amd.train.loop import create_train_batches, train_while_improving
+from spacy_ray_amd.train.scorer import weighted_score
+
+
+def _shard_corpus(corpus, rank: int, world: int):
+    """Explicit rank::world interleave so an epoch is a true partition
+    (improves on the reference, where every worker iterates the full corpus —
+    SURVEY.md §2.3 DP row)."""
+    if world <= 1:
+        return corpus
+
+    def sharded(nlp) -> Iterator:
+        for i, eg in enumerate(corpus(nlp)):
+            if i % world == rank:
+                yield eg
+
+    return sharded
+
+
+def _check_param_manifest(nlp, comm: Comm) -> None:
+    """Assert identical parameter (name, shape) tables across ranks — the
+    cross-rank key invariant."""
+    manifest = [
+        (n, tuple(p.shape))
+        for n, p in nlp.torch_module().named_parameters()
+    ]
+    ref = comm.broadcast_obj(manifest, src=0)
+    if manifest != ref:
+        raise RuntimeError(
+            f"rank {comm.rank}: parameter manifest differs from rank 0 — "
+            "non-deterministic model build"
+        )
+
+
+def distributed_train(
+    config: Config,
+    *,
+    output_path: Optional[Path] = None,
+    use_gpu: int = -1,
+    code_path: Optional[Path] = None,
+    resume: bool = False,
+    metrics_path: Optional[Path] = None,
+) -> Dict:
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    if code_path:
+        import importlib.util
+
+        spec = importlib.util.spec_from_file_location("srx_user_code", code_path)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+
+    if use_gpu >= 0 and torch.cuda.is_available():
+        device = f"cuda:{local_rank}"
+        torch.cuda.set_device(local_rank)
+    else:
+        device = "cpu"
+
+    comm = init_comm_from_env()
+    icfg = config.interpolate()
+    T = resolve(icfg["training"], validate=False)
+
+    nlp = init_nlp(config, device=device)
+    _check_param_manifest(nlp, comm)
+    # params are now identical on all ranks; diverge the RNG for dropout
+    seed = int(icfg.get("training", {}).get("seed", 0) or 0)
+    torch.manual_seed(seed * 1000 + 17 * rank + 1)
+
+    train_corpus, dev_corpus = resolve_dot_names(
+        icfg, [T["train_corpus"], T["dev_corpus"]]
+    )
+    train_corpus = _shard_corpus(train_corpus, rank, world)
+    engine = ZeRO1Engine(nlp, T["optimizer"], comm)
+    if resume and output_path and (Path(output_path) / "model-last").exists():
+        nlp.from_disk(Path(output_path) / "model-last")
+        opt_state = Path(output_path) / "model-last" / f"optim.rank{rank}.pt"
+        if opt_state.exists():
+            engine.load_state_dict(torch.load(opt_state, map_location=device))
+
+    dev_examples = None
+
+    def evaluate():
+        nonlocal dev_examples
+        if rank == 0:
+            if dev_examples is None:
+                dev_examples = list(dev_corpus(nlp))
+            scores = nlp.evaluate(dev_examples)
+            score = weighted_score(scores, T.get("score_weights") or {})
+            payload = (score, scores)
+        else:
+            payload = None
+        if world > 1:
+            payload = comm.broadcast_obj(payload, src=0)
+        return payload
+
+    batches = create_train_batches(nlp, train_corpus, T["batcher"], T.get("max_epochs", 0) or 0)
+    if rank == 0:
+        logger_setup = T.get("logger")
+        if logger_setup is None:
+            from spacy_ray_amd.train.loggers import console_logger
+
+            logger_setup = console_logger()
+        print_row, finalize_logger = logger_setup(nlp)
+    else:
+        print_row, finalize_logger = (lambda info: None), (lambda: None)
+
+    metrics_fh = open(metrics_path, "a") if (metrics_path and rank == 0) else None
+    words_cum = 0
+    t_start = time.time()
+    t_last = t_start
+    final_info: Dict = {}
+
+    step_iter = train_while_improving(
+        nlp,
+        engine,
+        batches,
+        evaluate=evaluate,
+        dropout=T.get("dropout", 0.1),
+        accumulate_gradient=int(T.get("accumulate_gradient", 1) or 1),
+        patience=int(T.get("patience", 0) or 0),
+        max_steps=int(T.get("max_steps", 0) or 0),
+        eval_frequency=int(T.get("eval_frequency", 200) or 200),
+        exclude=T.get("frozen_components") or [],
+        annotating_components=T.get("annotating_components") or [],
+        before_update=T.get("before_update"),
+    )
+    best_score = None
+    for batch, info, is_best_checkpoint in step_iter:
+        words_cum += info["words"] * world
+        final_info = info
+        if rank == 0:
+            now = time.time()
+            info["words_scaled"] = info["words"] * world
+            info["wps"] = info["words"] * world / max(1e-9, now - t_last)
+            t_last = now
+            if metrics_fh is not None:
+                metrics_fh.write(json.dumps({
+                    "step": info["step"], "epoch": info["epoch"],
+                    "losses": info["losses"], "score": info["score"],
+                    "other_scores": info["other_scores"],
+                    "words": info["words_scaled"], "words_cum": words_cum,
+                    "wps": info["wps"], "time": now - t_start,
+                }) + "\n")
+                metrics_fh.flush()
+            if is_best_checkpoint is not None:
+                print_row(info)
+        if is_best_checkpoint and output_path:
+            # all ranks participate: params are replicated post-all-gather;
+            # rank 0 writes the pipeline, every rank its optimizer shard.
+            if rank == 0:
+                best = info["score"]
+                nlp.meta["performance"] = info["other_scores"]
+                nlp.to_disk(Path(output_path) / "model-best")
+            comm.barrier()
+    if output_path:
+        if rank == 0:
+            nlp.to_disk(Path(output_path) / "model-last")
+        comm.barrier()
+        torch.save(engine.state_dict(), Path(output_path) / "model-last" / f"optim.rank{rank}.pt")
+    if rank == 0:
+        finalize_logger()
+        if metrics_fh is not None:
+            metrics_fh.close()
+    return final_info

@@ -1,0 +1,171 @@
+"""End-to-end pipeline tests on CPU: build from config, train, converge,
+checkpoint round-trip."""
+import numpy as np
+import pytest
+import torch
+
+from spacy_ray_amd.config.config import Config, resolve, resolve_dot_names
+from spacy_ray_amd.pipeline.language import build_nlp, init_nlp
+from spacy_ray_amd.train.loop import create_train_batches, train_while_improving
+from spacy_ray_amd.train.stepper import SimpleStepper
+from spacy_ray_amd.train.scorer import score_examples, weighted_score
+
+TAGGER_CFG = """
+[nlp]
+lang = "en"
+pipeline = ["tok2vec", "tagger"]
+
+[components]
+
+[components.tok2vec]
+factory = "tok2vec"
+
+[components.tok2vec.model]
+@architectures = "spacy.HashEmbedCNN.v2"
+width = 32
+depth = 2
+embed_size = 500
+
+[components.tagger]
+factory = "tagger"
+
+[components.tagger.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.tagger.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 32
+
+[corpora]
+
+[corpora.train]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 200
+words_per_doc = 10
+vocab_size = 100
+n_tags = 5
+seed = 0
+
+[corpora.dev]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 50
+words_per_doc = 10
+vocab_size = 100
+n_tags = 5
+seed = 1
+shuffle = false
+
+[training]
+seed = 0
+dropout = 0.0
+max_steps = 60
+eval_frequency = 30
+train_corpus = "corpora.train"
+dev_corpus = "corpora.dev"
+
+[training.batcher]
+@batchers = "spacy.batch_by_words.v1"
+size = 500
+
+[training.optimizer]
+@optimizers = "Adam.v1"
+learn_rate = 0.005
+
+[training.score_weights]
+tag_acc = 1.0
+"""
+
+
+def _train(cfg_text, max_steps=60):
+    cfg = Config.from_str(cfg_text)
+    nlp = init_nlp(cfg)
+    icfg = cfg.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    train_corpus, dev_corpus = resolve_dot_names(icfg, [T["train_corpus"], T["dev_corpus"]])
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    batches = create_train_batches(nlp, train_corpus, T["batcher"], 0)
+    dev = list(dev_corpus(nlp))
+
+    def evaluate():
+        scores = nlp.evaluate(dev)
+        return weighted_score(scores, icfg["training"]["score_weights"]), scores
+
+    first_losses, last_losses = None, None
+    for batch, info, is_best in train_while_improving(
+        nlp, stepper, batches, evaluate=evaluate, dropout=0.0,
+        max_steps=max_steps, eval_frequency=max_steps // 2,
+    ):
+        if info["step"] == 2:
+            first_losses = dict(info["losses"])
+        last_losses = info
+    return nlp, first_losses, last_losses, evaluate
+
+
+def test_tagger_learns():
+    nlp, first, last, evaluate = _train(TAGGER_CFG)
+    score, scores = evaluate()
+    # deterministic word->tag mapping with 10% noise: accuracy must beat
+    # majority-class baseline decisively after 60 steps
+    assert scores["tag_acc"] > 0.5, scores
+
+
+def test_deterministic_init():
+    cfg = Config.from_str(TAGGER_CFG)
+    nlp1 = init_nlp(cfg)
+    nlp2 = init_nlp(Config.from_str(TAGGER_CFG))
+    p1 = dict(nlp1.torch_module().named_parameters())
+    p2 = dict(nlp2.torch_module().named_parameters())
+    assert p1.keys() == p2.keys()
+    for k in p1:
+        assert torch.equal(p1[k], p2[k]), k
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    nlp, _, _, _ = _train(TAGGER_CFG, max_steps=20)
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+
+    docs = make_synthetic_docs(nlp.vocab, n_docs=5, words_per_doc=8, vocab_size=100,
+                               n_tags=5, n_deps=3, n_ent_types=2, seed=9)
+    preds1 = [list(d.tags) for d in nlp.predict_docs([d.copy_unannotated() for d in docs])]
+    nlp.to_disk(tmp_path / "model")
+    cfg = Config.from_disk(tmp_path / "model" / "config.cfg")
+    nlp2 = build_nlp(cfg)
+    nlp2.from_disk(tmp_path / "model")
+    preds2 = [list(d.tags) for d in nlp2.predict_docs([d.copy_unannotated() for d in docs])]
+    assert preds1 == preds2
+
+
+def test_parser_ner_train_step():
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    nlp = init_nlp(cfg, sample_size=32)
+    icfg = cfg.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    train_corpus, _ = resolve_dot_names(icfg, [T["train_corpus"], T["dev_corpus"]])
+    examples = []
+    for eg in train_corpus(nlp):
+        examples.append(eg)
+        if len(examples) >= 16:
+            break
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    losses = {}
+    stepper.accumulate(examples, drop=0.0, losses=losses)
+    stepper.apply_step()
+    assert set(losses) == {"tagger", "parser", "ner"}
+    assert all(np.isfinite(v) and v > 0 for v in losses.values()), losses
+
+
+def test_predict_sets_annotations():
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    nlp = init_nlp(cfg, sample_size=16)
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+
+    docs = make_synthetic_docs(nlp.vocab, n_docs=3, words_per_doc=10, vocab_size=100,
+                               n_tags=5, n_deps=3, n_ent_types=2, seed=3)
+    outs = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    for d in outs:
+        assert d.tags is not None and len(d.tags) == len(d)
+        assert d.heads is not None and len(d.heads) == len(d)
+        assert d.deps is not None
+        assert d.ents is not None
+        # heads are in-range or -1 (root)
+        assert all(-1 <= int(h) < len(d) for h in d.heads)
