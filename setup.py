@@ -39,7 +39,8 @@ ext_modules = [
 ]
 
 cmdclass = {}
-hip_sources = sorted(KERNELS.glob("*.hip")) + sorted(KERNELS.glob("*_hip.cpp"))
+# exclude hipify-generated *_hip.hip artifacts from previous builds
+hip_sources = [p for p in sorted(KERNELS.glob("*.hip")) if not p.name.endswith("_hip.hip")]
 if hip_sources and "SRX_SKIP_HIP" not in os.environ:
     from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 

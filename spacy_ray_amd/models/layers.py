@@ -20,19 +20,21 @@ def glorot_uniform_(w: torch.Tensor) -> torch.Tensor:
 
 
 class Maxout(nn.Module):
-    """Linear (nI -> nO*P) followed by max over P pieces, the Thinc Maxout
-    layer (SURVEY.md §2.5 maxout_fwd/bwd)."""
+    """Linear (nI -> P*nO, pieces-major) followed by max over P pieces — the
+    Thinc Maxout layer (SURVEY.md §2.5 maxout_fwd/bwd).  Weight rows are laid
+    out pieces-major so each piece is a contiguous nO block in the GEMM
+    output (coalesced maxout kernel)."""
 
     def __init__(self, nI: int, nO: int, pieces: int = 3, normalize: bool = False):
         super().__init__()
         self.nI, self.nO, self.pieces = nI, nO, pieces
-        self.weight = nn.Parameter(glorot_uniform_(torch.empty(nO * pieces, nI)))
-        self.bias = nn.Parameter(torch.zeros(nO * pieces))
+        self.weight = nn.Parameter(glorot_uniform_(torch.empty(pieces * nO, nI)))
+        self.bias = nn.Parameter(torch.zeros(pieces * nO))
         self.norm = LayerNorm(nO) if normalize else None
 
     def forward(self, X: torch.Tensor) -> torch.Tensor:
         Y = torch.nn.functional.linear(X, self.weight, self.bias)
-        Y = ops.maxout(Y.view(*Y.shape[:-1], self.nO, self.pieces))
+        Y = ops.maxout(Y.view(*Y.shape[:-1], self.pieces, self.nO))
         if self.norm is not None:
             Y = self.norm(Y)
         return Y

@@ -68,6 +68,8 @@ def seq2col(X: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
 
 # ---------------------------------------------------------------- maxout
 class _Maxout(torch.autograd.Function):
+    """Pieces-major maxout: [..., P, W] -> [..., W] (see ref.maxout)."""
+
     @staticmethod
     def forward(ctx, X: torch.Tensor) -> torch.Tensor:
         if _want_hip(X):
@@ -75,7 +77,7 @@ class _Maxout(torch.autograd.Function):
         else:
             Y, which = ref.maxout(X)
         ctx.save_for_backward(which)
-        ctx.P = X.shape[-1]
+        ctx.P = X.shape[-2]
         return Y
 
     @staticmethod
@@ -87,7 +89,7 @@ class _Maxout(torch.autograd.Function):
 
 
 def maxout(X: torch.Tensor) -> torch.Tensor:
-    """[..., W, P] -> [..., W]"""
+    """[..., P, W] -> [..., W]"""
     return _Maxout.apply(X)
 
 
@@ -183,7 +185,7 @@ class _ParserStepScore(torch.autograd.Function):
         else:
             slot = torch.arange(nF, device=feats.device).unsqueeze(0)
             summed = precomputed[feats.long(), slot].sum(dim=1) + bias
-            hidden, which = ref.maxout(summed.view(S, H, P))
+            hidden, which = ref.maxout(summed.view(S, P, H))
         ctx.save_for_backward(feats, which)
         ctx.shape = (precomputed.shape[0], nF, HP)
         return hidden
@@ -196,7 +198,7 @@ class _ParserStepScore(torch.autograd.Function):
         if _want_hip(dHidden):
             dPre, dBias = hip_ext().parser_step_bwd(dHidden.contiguous(), feats, which, T1, nF, HP)
             return dPre, None, dBias
-        dSummed = ref.maxout_backward(dHidden, which, P).reshape(dHidden.shape[0], HP)
+        dSummed = ref.maxout_backward(dHidden, which, P).reshape(dHidden.shape[0], HP)  # [S, P*H]
         dBias = dSummed.sum(dim=0)
         dPre = dHidden.new_zeros(T1, nF, HP)
         # scatter-add into the gathered rows

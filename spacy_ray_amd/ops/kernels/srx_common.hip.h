@@ -1,0 +1,83 @@
+// Common device helpers for the gfx950 kernels.
+// Written CDNA4-first: wave64 everywhere (no 32-wide warp idioms), bf16
+// stored as raw ushort bits (torch bf16 layout), fp32 accumulation.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define SRX_WAVE 64
+
+typedef unsigned short bf16_t;
+
+__device__ __forceinline__ float bf2f(bf16_t b) {
+  uint32_t u = ((uint32_t)b) << 16;
+  return __uint_as_float(u);
+}
+
+__device__ __forceinline__ bf16_t f2bf(float f) {
+  uint32_t u = __float_as_uint(f);
+  uint32_t r = 0x7fffu + ((u >> 16) & 1u);  // round-to-nearest-even
+  return (bf16_t)((u + r) >> 16);
+}
+
+template <typename T>
+struct Elem;
+
+template <>
+struct Elem<float> {
+  static __device__ __forceinline__ float ld(const float* p) { return *p; }
+  static __device__ __forceinline__ void st(float* p, float v) { *p = v; }
+};
+
+template <>
+struct Elem<bf16_t> {
+  static __device__ __forceinline__ float ld(const bf16_t* p) { return bf2f(*p); }
+  static __device__ __forceinline__ void st(bf16_t* p, float v) { *p = f2bf(v); }
+};
+
+// Wave-wide (64-lane) sum reduction; result valid in all lanes.
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, SRX_WAVE);
+  return v;
+}
+
+// ---- MurmurHash3 x64_128 of one 8-byte key: device twin of
+// ops/csrc/murmur3.h::murmur3_hash4_u64 — MUST stay bit-identical (the
+// CPU/GPU HashEmbed row-assignment contract).
+__device__ __forceinline__ uint64_t srx_rotl64(uint64_t x, int r) {
+  return (x << r) | (x >> (64 - r));
+}
+
+__device__ __forceinline__ uint64_t srx_fmix64(uint64_t k) {
+  k ^= k >> 33;
+  k *= 0xff51afd7ed558ccdULL;
+  k ^= k >> 33;
+  k *= 0xc4ceb9fe1a85ec53ULL;
+  k ^= k >> 33;
+  return k;
+}
+
+__device__ __forceinline__ void murmur3_hash4_u64_dev(uint64_t key, uint32_t seed,
+                                                      uint32_t out[4]) {
+  const uint64_t c1 = 0x87c37b91114253d5ULL;
+  const uint64_t c2 = 0x4cf5ad432745937fULL;
+  uint64_t h1 = seed, h2 = seed;
+  uint64_t k1 = key;
+  k1 *= c1;
+  k1 = srx_rotl64(k1, 31);
+  k1 *= c2;
+  h1 ^= k1;
+  h1 ^= 8ULL;
+  h2 ^= 8ULL;
+  h1 += h2;
+  h2 += h1;
+  h1 = srx_fmix64(h1);
+  h2 = srx_fmix64(h2);
+  h1 += h2;
+  h2 += h1;
+  out[0] = (uint32_t)(h1 & 0xffffffffULL);
+  out[1] = (uint32_t)(h1 >> 32);
+  out[2] = (uint32_t)(h2 & 0xffffffffULL);
+  out[3] = (uint32_t)(h2 >> 32);
+}

@@ -1,0 +1,250 @@
+// Bandwidth-bound kernels: seq2col, maxout, layernorm, fused Adam.
+// Semantics defined by ops/torch_ref.py (the fp32 torch reference these are
+// tested against).  All are memory-bound: vectorized loads (V elems/lane:
+// 8 for bf16 = 16B, 4 for f32 = 16B per guide G13), grid-stride loops,
+// fp32 accumulation.
+#pragma once
+#include "srx_common.hip.h"
+
+// ------------------------------------------------------------- seq2col
+// Y[t, s*W + w] = X[t-1+s, w] for s in {0,1,2}, zeroed across doc bounds.
+// is_start/is_end: per-token doc-boundary bytes (computed once per batch).
+template <typename T, int V>
+__global__ void seq2col_fwd_kernel(const T* __restrict__ X, T* __restrict__ Y,
+                                   const uint8_t* __restrict__ is_start,
+                                   const uint8_t* __restrict__ is_end,
+                                   long nT, int W) {
+  const long chunks_per_sec = W / V;
+  const long total = nT * 3 * chunks_per_sec;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long t = i / (3 * chunks_per_sec);
+    long rem = i % (3 * chunks_per_sec);
+    int s = (int)(rem / chunks_per_sec);
+    int w = (int)(rem % chunks_per_sec) * V;
+    long src_t = t + s - 1;
+    bool zero = (s == 0 && (t == 0 || is_start[t])) ||
+                (s == 2 && (t == nT - 1 || is_end[t]));
+    T* out = Y + (t * 3 + s) * (long)W + w;
+    if (zero || src_t < 0 || src_t >= nT) {
+#pragma unroll
+      for (int k = 0; k < V; k++) Elem<T>::st(out + k, 0.0f);
+    } else {
+      const T* in = X + src_t * (long)W + w;
+#pragma unroll
+      for (int k = 0; k < V; k++) Elem<T>::st(out + k, Elem<T>::ld(in + k));
+    }
+  }
+}
+
+// dX[t,w] = dY[t, W+w] + (next's prev-slot) + (prev's next-slot)
+template <typename T, int V>
+__global__ void seq2col_bwd_kernel(const T* __restrict__ dY, T* __restrict__ dX,
+                                   const uint8_t* __restrict__ is_start,
+                                   const uint8_t* __restrict__ is_end,
+                                   long nT, int W) {
+  const long chunks = W / V;
+  const long total = nT * chunks;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long t = i / chunks;
+    int w = (int)(i % chunks) * V;
+    const long row = 3L * W;
+#pragma unroll
+    for (int k = 0; k < V; k++) {
+      float acc = Elem<T>::ld(dY + t * row + W + w + k);
+      if (t + 1 < nT && !is_start[t + 1])
+        acc += Elem<T>::ld(dY + (t + 1) * row + 0 + w + k);
+      if (t > 0 && !is_end[t - 1])
+        acc += Elem<T>::ld(dY + (t - 1) * row + 2 * W + w + k);
+      Elem<T>::st(dX + t * (long)W + w + k, acc);
+    }
+  }
+}
+
+// ------------------------------------------------------------- maxout
+// Pieces-major: X [N, P, W] -> Y [N, W], which [N, W] (uint8).
+template <typename T, int V>
+__global__ void maxout_fwd_kernel(const T* __restrict__ X, T* __restrict__ Y,
+                                  uint8_t* __restrict__ which, long N, int P, int W) {
+  const long chunks = W / V;
+  const long total = N * chunks;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long n = i / chunks;
+    int w = (int)(i % chunks) * V;
+    float best[V];
+    uint8_t arg[V];
+    const T* base = X + n * (long)P * W + w;
+#pragma unroll
+    for (int k = 0; k < V; k++) {
+      best[k] = Elem<T>::ld(base + k);
+      arg[k] = 0;
+    }
+    for (int p = 1; p < P; p++) {
+      const T* bp = base + (long)p * W;
+#pragma unroll
+      for (int k = 0; k < V; k++) {
+        float v = Elem<T>::ld(bp + k);
+        if (v > best[k]) { best[k] = v; arg[k] = (uint8_t)p; }
+      }
+    }
+    T* out = Y + n * (long)W + w;
+    uint8_t* wh = which + n * (long)W + w;
+#pragma unroll
+    for (int k = 0; k < V; k++) {
+      Elem<T>::st(out + k, best[k]);
+      wh[k] = arg[k];
+    }
+  }
+}
+
+template <typename T, int V>
+__global__ void maxout_bwd_kernel(const T* __restrict__ dY,
+                                  const uint8_t* __restrict__ which,
+                                  T* __restrict__ dX, long N, int P, int W) {
+  const long chunks = W / V;
+  const long total = N * (long)P * chunks;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long n = i / (P * chunks);
+    long rem = i % (P * chunks);
+    int p = (int)(rem / chunks);
+    int w = (int)(rem % chunks) * V;
+    const T* dy = dY + n * (long)W + w;
+    const uint8_t* wh = which + n * (long)W + w;
+    T* out = dX + (n * (long)P + p) * W + w;
+#pragma unroll
+    for (int k = 0; k < V; k++)
+      Elem<T>::st(out + k, wh[k] == p ? Elem<T>::ld(dy + k) : 0.0f);
+  }
+}
+
+// ----------------------------------------------------------- layernorm
+// One wave per row (W <= a few hundred); 4 waves per block.
+template <typename T>
+__global__ void layernorm_fwd_kernel(const T* __restrict__ X,
+                                     const T* __restrict__ g,
+                                     const T* __restrict__ b,
+                                     T* __restrict__ Y,
+                                     float* __restrict__ mu_out,
+                                     float* __restrict__ rstd_out,
+                                     long N, int W, float eps) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  for (long n = wave; n < N; n += nwaves) {
+    const T* row = X + n * (long)W;
+    float s = 0.f, sq = 0.f;
+    for (int w = lane; w < W; w += SRX_WAVE) {
+      float v = Elem<T>::ld(row + w);
+      s += v;
+      sq += v * v;
+    }
+    s = wave_reduce_sum(s);
+    sq = wave_reduce_sum(sq);
+    float mu = s / W;
+    float var = sq / W - mu * mu;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (lane == 0) {
+      mu_out[n] = mu;
+      rstd_out[n] = rstd;
+    }
+    T* out = Y + n * (long)W;
+    for (int w = lane; w < W; w += SRX_WAVE) {
+      float v = (Elem<T>::ld(row + w) - mu) * rstd;
+      Elem<T>::st(out + w, v * Elem<T>::ld(g + w) + Elem<T>::ld(b + w));
+    }
+  }
+}
+
+// dX per row; dg/db accumulated into fp32 buffers with atomics (W is small,
+// contention is spread across W addresses x many L2 channels).
+template <typename T>
+__global__ void layernorm_bwd_kernel(const T* __restrict__ dY,
+                                     const T* __restrict__ X,
+                                     const T* __restrict__ g,
+                                     const float* __restrict__ mu,
+                                     const float* __restrict__ rstd,
+                                     T* __restrict__ dX,
+                                     float* __restrict__ dg32,
+                                     float* __restrict__ db32,
+                                     long N, int W) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  for (long n = wave; n < N; n += nwaves) {
+    const T* xrow = X + n * (long)W;
+    const T* dyrow = dY + n * (long)W;
+    float m = mu[n], r = rstd[n];
+    float s1 = 0.f, s2 = 0.f;
+    for (int w = lane; w < W; w += SRX_WAVE) {
+      float xhat = (Elem<T>::ld(xrow + w) - m) * r;
+      float dy = Elem<T>::ld(dyrow + w);
+      float dxhat = dy * Elem<T>::ld(g + w);
+      s1 += dxhat;
+      s2 += dxhat * xhat;
+      atomicAdd(dg32 + w, dy * xhat);
+      atomicAdd(db32 + w, dy);
+    }
+    s1 = wave_reduce_sum(s1) / W;
+    s2 = wave_reduce_sum(s2) / W;
+    T* out = dX + n * (long)W;
+    for (int w = lane; w < W; w += SRX_WAVE) {
+      float xhat = (Elem<T>::ld(xrow + w) - m) * r;
+      float dxhat = Elem<T>::ld(dyrow + w) * Elem<T>::ld(g + w);
+      Elem<T>::st(out + w, r * (dxhat - s1 - xhat * s2));
+    }
+  }
+}
+
+// ---------------------------------------------------------- fused Adam
+// One launch for the whole sharded step (SURVEY.md §2.5 fused_adam_sharded):
+// grad (model dtype) + fp32 master/m/v -> updated state + model-dtype param.
+// Decoupled weight decay; bias correction folded into lr/denom; clip scale
+// precomputed on host from the all-reduced global norm.
+template <typename T>
+__global__ void adam_step_kernel(const T* __restrict__ grad,
+                                 float* __restrict__ master,
+                                 float* __restrict__ m,
+                                 float* __restrict__ v,
+                                 T* __restrict__ param_out,
+                                 long n, float clip_scale, float lr,
+                                 float beta1, float beta2, float eps,
+                                 float wd, float bc1, float bc2) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float gv = Elem<T>::ld(grad + i) * clip_scale;
+    float p = master[i] * (1.0f - lr * wd);
+    float mi = beta1 * m[i] + (1.0f - beta1) * gv;
+    float vi = beta2 * v[i] + (1.0f - beta2) * gv * gv;
+    m[i] = mi;
+    v[i] = vi;
+    float denom = sqrtf(vi / bc2) + eps;
+    p -= (lr / bc1) * (mi / denom);
+    master[i] = p;
+    Elem<T>::st(param_out + i, p);
+  }
+}
+
+// grad-norm^2 partial reduce (one value per block -> atomicAdd)
+template <typename T>
+__global__ void sqnorm_kernel(const T* __restrict__ x, long n, float* __restrict__ out) {
+  float acc = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float v = Elem<T>::ld(x + i);
+    acc += v * v;
+  }
+  acc = wave_reduce_sum(acc);
+  __shared__ float wsum[16];
+  int wid = threadIdx.x / SRX_WAVE;
+  int lane = threadIdx.x & (SRX_WAVE - 1);
+  if (lane == 0) wsum[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int i = 0; i < (int)(blockDim.x / SRX_WAVE); i++) s += wsum[i];
+    atomicAdd(out, s);
+  }
+}

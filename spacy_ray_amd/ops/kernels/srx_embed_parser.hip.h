@@ -1,0 +1,123 @@
+// HashEmbed (fused murmur + 4-row gather/scatter) and the parser state
+// scorer (fused nF-row gather + sum + bias + maxout P=2).
+// Semantics: ops/torch_ref.py (hashembed_*, parser_step_score).
+#pragma once
+#include "srx_common.hip.h"
+
+// ----------------------------------------------------------- hashembed
+// One wave per token: the wave's lanes cover the W embedding columns; the
+// 4 murmur row ids are computed from the wave-uniform id (scalar-unit work)
+// and the 4 rows are gathered and summed in fp32 (SURVEY.md §2.5
+// hashembed_fwd: out[i] = sum_s E[h_s(id_i) % rows]).
+template <typename T>
+__global__ void hashembed_fwd_kernel(const T* __restrict__ table,
+                                     const uint64_t* __restrict__ ids,
+                                     T* __restrict__ Y,
+                                     int32_t* __restrict__ rows_out,
+                                     long nT, int nrows, int W, uint32_t seed) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  for (long t = wave; t < nT; t += nwaves) {
+    uint32_t h[4];
+    murmur3_hash4_u64_dev(ids[t], seed, h);
+    int32_t r0 = (int32_t)(h[0] % (uint32_t)nrows);
+    int32_t r1 = (int32_t)(h[1] % (uint32_t)nrows);
+    int32_t r2 = (int32_t)(h[2] % (uint32_t)nrows);
+    int32_t r3 = (int32_t)(h[3] % (uint32_t)nrows);
+    if (lane < 4) rows_out[t * 4 + lane] = lane == 0 ? r0 : lane == 1 ? r1 : lane == 2 ? r2 : r3;
+    const T* t0 = table + (long)r0 * W;
+    const T* t1 = table + (long)r1 * W;
+    const T* t2 = table + (long)r2 * W;
+    const T* t3 = table + (long)r3 * W;
+    T* out = Y + t * (long)W;
+    for (int w = lane; w < W; w += SRX_WAVE) {
+      float acc = Elem<T>::ld(t0 + w) + Elem<T>::ld(t1 + w) +
+                  Elem<T>::ld(t2 + w) + Elem<T>::ld(t3 + w);
+      Elem<T>::st(out + w, acc);
+    }
+  }
+}
+
+// Backward: scatter-add dY into an fp32 table-grad workspace (atomics; the
+// Zipf-hot rows are the contention worry — SURVEY.md §7 hard-part 2; fp32
+// atomics across W addresses and many L2 channels are acceptable round 1,
+// sort-by-row segmented reduction is the planned upgrade).
+template <typename T>
+__global__ void hashembed_bwd_kernel(const T* __restrict__ dY,
+                                     const int32_t* __restrict__ rows,
+                                     float* __restrict__ dT32,
+                                     long nT, int W) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  for (long t = wave; t < nT; t += nwaves) {
+    int32_t r0 = rows[t * 4 + 0], r1 = rows[t * 4 + 1];
+    int32_t r2 = rows[t * 4 + 2], r3 = rows[t * 4 + 3];
+    const T* dy = dY + t * (long)W;
+    for (int w = lane; w < W; w += SRX_WAVE) {
+      float v = Elem<T>::ld(dy + w);
+      atomicAdd(dT32 + (long)r0 * W + w, v);
+      atomicAdd(dT32 + (long)r1 * W + w, v);
+      atomicAdd(dT32 + (long)r2 * W + w, v);
+      atomicAdd(dT32 + (long)r3 * W + w, v);
+    }
+  }
+}
+
+// ------------------------------------------------- parser step scorer
+// One wave per state (SURVEY.md §2.5 parser_step_score): gather nF
+// precomputed rows (pieces-major [T+1, nF, 2*H]), sum + bias, maxout P=2.
+// H=64 fits one lane per hidden unit exactly.
+template <typename T>
+__global__ void parser_step_fwd_kernel(const T* __restrict__ pre,
+                                       const int64_t* __restrict__ feats,
+                                       const T* __restrict__ bias,
+                                       T* __restrict__ hidden,
+                                       uint8_t* __restrict__ which,
+                                       long S, int nF, int H) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  const int HP = 2 * H;
+  for (long s = wave; s < S; s += nwaves) {
+    const int64_t* fs = feats + s * nF;
+    for (int h = lane; h < H; h += SRX_WAVE) {
+      float acc0 = Elem<T>::ld(bias + h);
+      float acc1 = Elem<T>::ld(bias + H + h);
+      for (int f = 0; f < nF; f++) {
+        const T* row = pre + (fs[f] * (long)nF + f) * HP;
+        acc0 += Elem<T>::ld(row + h);
+        acc1 += Elem<T>::ld(row + H + h);
+      }
+      bool second = acc1 > acc0;
+      Elem<T>::st(hidden + s * (long)H + h, second ? acc1 : acc0);
+      which[s * (long)H + h] = (uint8_t)second;
+    }
+  }
+}
+
+// Backward: scatter dHidden into fp32 workspaces for dPre and dBias.
+template <typename T>
+__global__ void parser_step_bwd_kernel(const T* __restrict__ dHidden,
+                                       const int64_t* __restrict__ feats,
+                                       const uint8_t* __restrict__ which,
+                                       float* __restrict__ dPre32,
+                                       float* __restrict__ dBias32,
+                                       long S, int nF, int H) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  const int HP = 2 * H;
+  for (long s = wave; s < S; s += nwaves) {
+    const int64_t* fs = feats + s * nF;
+    for (int h = lane; h < H; h += SRX_WAVE) {
+      float d = Elem<T>::ld(dHidden + s * (long)H + h);
+      int slot = which[s * (long)H + h] ? H + h : h;
+      atomicAdd(dBias32 + slot, d);
+      for (int f = 0; f < nF; f++) {
+        atomicAdd(dPre32 + (fs[f] * (long)nF + f) * HP + slot, d);
+      }
+    }
+  }
+}

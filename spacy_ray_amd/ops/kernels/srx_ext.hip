@@ -1,0 +1,313 @@
+// spacy_ray_amd._srx_hip — gfx950 kernel bindings (torch extension).
+// Kernels live in the *.hip.h headers; this TU instantiates fp32 + bf16
+// variants and exposes the op surface consumed by ops/api.py and
+// parallel/engine.py.  Tested on-GPU against ops/torch_ref.py fp32.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "srx_common.hip.h"
+#include "srx_elementwise.hip.h"
+#include "srx_embed_parser.hip.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+inline int grid_for(long total, int per_thread = 1) {
+  long blocks = (total + (long)kBlock * per_thread - 1) / ((long)kBlock * per_thread);
+  return (int)std::min<long>(blocks, 16384);
+}
+
+inline void check_dev(const at::Tensor& t) {
+  TORCH_CHECK(t.is_cuda(), "expected a GPU tensor");
+  TORCH_CHECK(t.is_contiguous(), "expected contiguous");
+}
+
+#define DISPATCH_F(dtype, ...)                                        \
+  if ((dtype) == at::kFloat) {                                        \
+    using scalar_t = float;                                           \
+    constexpr int kVec = 4;                                           \
+    __VA_ARGS__;                                                      \
+  } else if ((dtype) == at::kBFloat16) {                              \
+    using scalar_t = bf16_t;                                          \
+    constexpr int kVec = 8;                                           \
+    __VA_ARGS__;                                                      \
+  } else {                                                            \
+    TORCH_CHECK(false, "unsupported dtype (need f32 or bf16)");       \
+  }
+
+std::pair<at::Tensor, at::Tensor> boundary_masks(const at::Tensor& lengths, long nT) {
+  auto opts = at::TensorOptions().dtype(at::kByte).device(lengths.device());
+  auto starts = at::zeros({nT}, opts);
+  auto ends = at::zeros({nT}, opts);
+  auto l = lengths.to(at::kLong);
+  auto offs = l.cumsum(0);
+  auto nz = l > 0;
+  auto s_idx = (offs - l).masked_select(nz);
+  auto e_idx = (offs - 1).masked_select(nz);
+  starts.index_fill_(0, s_idx, 1);
+  ends.index_fill_(0, e_idx, 1);
+  return {starts, ends};
+}
+
+// --------------------------------------------------------------- seq2col
+at::Tensor seq2col_fwd(at::Tensor X, at::Tensor lengths) {
+  check_dev(X);
+  long nT = X.size(0);
+  int W = (int)X.size(1);
+  auto Y = at::empty({nT, 3L * W}, X.options());
+  if (nT == 0) return Y;
+  auto [starts, ends] = boundary_masks(lengths, nT);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(X.scalar_type(), {
+    const int V = (W % kVec == 0) ? kVec : 1;
+    long total = nT * 3L * (W / V);
+    if (V == kVec)
+      hipLaunchKernelGGL((seq2col_fwd_kernel<scalar_t, kVec>), dim3(grid_for(total)),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)X.data_ptr(), (scalar_t*)Y.data_ptr(),
+                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(), nT, W);
+    else
+      hipLaunchKernelGGL((seq2col_fwd_kernel<scalar_t, 1>), dim3(grid_for(nT * 3L * W)),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)X.data_ptr(), (scalar_t*)Y.data_ptr(),
+                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(), nT, W);
+  });
+  return Y;
+}
+
+at::Tensor seq2col_bwd(at::Tensor dY, at::Tensor lengths) {
+  check_dev(dY);
+  long nT = dY.size(0);
+  int W3 = (int)dY.size(1);
+  int W = W3 / 3;
+  auto dX = at::empty({nT, (long)W}, dY.options());
+  if (nT == 0) return dX;
+  auto [starts, ends] = boundary_masks(lengths, nT);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(dY.scalar_type(), {
+    const int V = (W % kVec == 0) ? kVec : 1;
+    if (V == kVec)
+      hipLaunchKernelGGL((seq2col_bwd_kernel<scalar_t, kVec>),
+                         dim3(grid_for(nT * (W / kVec))), dim3(kBlock), 0, stream,
+                         (const scalar_t*)dY.data_ptr(), (scalar_t*)dX.data_ptr(),
+                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(), nT, W);
+    else
+      hipLaunchKernelGGL((seq2col_bwd_kernel<scalar_t, 1>),
+                         dim3(grid_for(nT * (long)W)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)dY.data_ptr(), (scalar_t*)dX.data_ptr(),
+                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(), nT, W);
+  });
+  return dX;
+}
+
+// ---------------------------------------------------------------- maxout
+std::vector<at::Tensor> maxout_fwd(at::Tensor X) {
+  check_dev(X);
+  int P = (int)X.size(-2);
+  int W = (int)X.size(-1);
+  long N = X.numel() / ((long)P * W);
+  auto sizes = X.sizes().vec();
+  sizes.erase(sizes.end() - 2);
+  auto Y = at::empty(sizes, X.options());
+  auto which = at::empty(sizes, X.options().dtype(at::kByte));
+  if (N == 0) return {Y, which};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(X.scalar_type(), {
+    const int V = (W % kVec == 0) ? kVec : 1;
+    if (V == kVec)
+      hipLaunchKernelGGL((maxout_fwd_kernel<scalar_t, kVec>),
+                         dim3(grid_for(N * (W / kVec))), dim3(kBlock), 0, stream,
+                         (const scalar_t*)X.data_ptr(), (scalar_t*)Y.data_ptr(),
+                         which.data_ptr<uint8_t>(), N, P, W);
+    else
+      hipLaunchKernelGGL((maxout_fwd_kernel<scalar_t, 1>),
+                         dim3(grid_for(N * (long)W)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)X.data_ptr(), (scalar_t*)Y.data_ptr(),
+                         which.data_ptr<uint8_t>(), N, P, W);
+  });
+  return {Y, which};
+}
+
+at::Tensor maxout_bwd(at::Tensor dY, at::Tensor which, long P) {
+  check_dev(dY);
+  int W = (int)dY.size(-1);
+  long N = dY.numel() / W;
+  auto sizes = dY.sizes().vec();
+  sizes.insert(sizes.end() - 1, P);
+  auto dX = at::empty(sizes, dY.options());
+  if (N == 0) return dX;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(dY.scalar_type(), {
+    const int V = (W % kVec == 0) ? kVec : 1;
+    long total = N * P * (W / (V == kVec ? kVec : 1));
+    if (V == kVec)
+      hipLaunchKernelGGL((maxout_bwd_kernel<scalar_t, kVec>), dim3(grid_for(total)),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)dY.data_ptr(), which.data_ptr<uint8_t>(),
+                         (scalar_t*)dX.data_ptr(), N, (int)P, W);
+    else
+      hipLaunchKernelGGL((maxout_bwd_kernel<scalar_t, 1>), dim3(grid_for(N * P * (long)W)),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)dY.data_ptr(), which.data_ptr<uint8_t>(),
+                         (scalar_t*)dX.data_ptr(), N, (int)P, W);
+  });
+  return dX;
+}
+
+// ------------------------------------------------------------- layernorm
+std::vector<at::Tensor> layernorm_fwd(at::Tensor X, at::Tensor g, at::Tensor b, double eps) {
+  check_dev(X);
+  int W = (int)X.size(-1);
+  long N = X.numel() / W;
+  auto Y = at::empty_like(X);
+  auto mu = at::empty({N}, X.options().dtype(at::kFloat));
+  auto rstd = at::empty({N}, X.options().dtype(at::kFloat));
+  if (N == 0) return {Y, mu, rstd};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = grid_for(N * SRX_WAVE);
+  DISPATCH_F(X.scalar_type(), {
+    hipLaunchKernelGGL((layernorm_fwd_kernel<scalar_t>), dim3(grid), dim3(kBlock), 0,
+                       stream, (const scalar_t*)X.data_ptr(),
+                       (const scalar_t*)g.data_ptr(), (const scalar_t*)b.data_ptr(),
+                       (scalar_t*)Y.data_ptr(), mu.data_ptr<float>(),
+                       rstd.data_ptr<float>(), N, W, (float)eps);
+  });
+  return {Y, mu, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
+                                      at::Tensor mu, at::Tensor rstd) {
+  check_dev(dY);
+  int W = (int)X.size(-1);
+  long N = X.numel() / W;
+  auto dX = at::empty_like(X);
+  auto dg32 = at::zeros({W}, X.options().dtype(at::kFloat));
+  auto db32 = at::zeros({W}, X.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (N > 0) {
+    DISPATCH_F(X.scalar_type(), {
+      hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t>), dim3(grid_for(N * SRX_WAVE)),
+                         dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                         (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
+                         mu.data_ptr<float>(), rstd.data_ptr<float>(),
+                         (scalar_t*)dX.data_ptr(), dg32.data_ptr<float>(),
+                         db32.data_ptr<float>(), N, W);
+    });
+  }
+  return {dX, dg32.to(X.scalar_type()), db32.to(X.scalar_type())};
+}
+
+// ------------------------------------------------------------- hashembed
+std::vector<at::Tensor> hashembed_fwd(at::Tensor table, at::Tensor ids, int64_t seed) {
+  check_dev(table);
+  check_dev(ids);
+  TORCH_CHECK(ids.scalar_type() == at::kLong, "ids must be int64 (bit-cast uint64)");
+  long nT = ids.size(0);
+  int nrows = (int)table.size(0);
+  int W = (int)table.size(1);
+  auto Y = at::empty({nT, (long)W}, table.options());
+  auto rows = at::empty({nT, 4}, table.options().dtype(at::kInt));
+  if (nT == 0) return {Y, rows};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(table.scalar_type(), {
+    hipLaunchKernelGGL((hashembed_fwd_kernel<scalar_t>), dim3(grid_for(nT * SRX_WAVE)),
+                       dim3(kBlock), 0, stream, (const scalar_t*)table.data_ptr(),
+                       (const uint64_t*)ids.data_ptr<int64_t>(), (scalar_t*)Y.data_ptr(),
+                       rows.data_ptr<int32_t>(), nT, nrows, W, (uint32_t)seed);
+  });
+  return {Y, rows};
+}
+
+at::Tensor hashembed_bwd(at::Tensor dY, at::Tensor rows, int64_t nrows) {
+  check_dev(dY);
+  long nT = dY.size(0);
+  int W = (int)dY.size(1);
+  auto dT32 = at::zeros({nrows, (long)W}, dY.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (nT > 0) {
+    DISPATCH_F(dY.scalar_type(), {
+      hipLaunchKernelGGL((hashembed_bwd_kernel<scalar_t>), dim3(grid_for(nT * SRX_WAVE)),
+                         dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                         rows.data_ptr<int32_t>(), dT32.data_ptr<float>(), nT, W);
+    });
+  }
+  return dT32.to(dY.scalar_type());
+}
+
+// ----------------------------------------------------- parser step score
+std::vector<at::Tensor> parser_step_fwd(at::Tensor pre, at::Tensor feats, at::Tensor bias) {
+  check_dev(pre);
+  check_dev(feats);
+  TORCH_CHECK(feats.scalar_type() == at::kLong, "feats must be int64");
+  long S = feats.size(0);
+  int nF = (int)feats.size(1);
+  int HP = (int)pre.size(-1);
+  int H = HP / 2;
+  auto hidden = at::empty({S, (long)H}, pre.options());
+  auto which = at::empty({S, (long)H}, pre.options().dtype(at::kByte));
+  if (S == 0) return {hidden, which};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(pre.scalar_type(), {
+    hipLaunchKernelGGL((parser_step_fwd_kernel<scalar_t>), dim3(grid_for(S * SRX_WAVE)),
+                       dim3(kBlock), 0, stream, (const scalar_t*)pre.data_ptr(),
+                       feats.data_ptr<int64_t>(), (const scalar_t*)bias.data_ptr(),
+                       (scalar_t*)hidden.data_ptr(), which.data_ptr<uint8_t>(), S, nF, H);
+  });
+  return {hidden, which};
+}
+
+std::vector<at::Tensor> parser_step_bwd(at::Tensor dHidden, at::Tensor feats,
+                                        at::Tensor which, int64_t T1, int64_t nF,
+                                        int64_t HP) {
+  check_dev(dHidden);
+  long S = feats.size(0);
+  int H = (int)HP / 2;
+  auto dPre32 = at::zeros({T1, nF, HP}, dHidden.options().dtype(at::kFloat));
+  auto dBias32 = at::zeros({HP}, dHidden.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (S > 0) {
+    DISPATCH_F(dHidden.scalar_type(), {
+      hipLaunchKernelGGL((parser_step_bwd_kernel<scalar_t>),
+                         dim3(grid_for(S * SRX_WAVE)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)dHidden.data_ptr(), feats.data_ptr<int64_t>(),
+                         which.data_ptr<uint8_t>(), dPre32.data_ptr<float>(),
+                         dBias32.data_ptr<float>(), S, (int)nF, H);
+    });
+  }
+  return {dPre32.to(dHidden.scalar_type()), dBias32.to(dHidden.scalar_type())};
+}
+
+// ------------------------------------------------------------ fused Adam
+void adam_step(at::Tensor grad, at::Tensor master, at::Tensor m, at::Tensor v,
+               at::Tensor param_out, double clip_scale, double lr, double beta1,
+               double beta2, double eps, double wd, double bc1, double bc2) {
+  check_dev(grad);
+  long n = grad.numel();
+  if (n == 0) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(grad.scalar_type(), {
+    hipLaunchKernelGGL((adam_step_kernel<scalar_t>), dim3(grid_for(n, 4)), dim3(kBlock),
+                       0, stream, (const scalar_t*)grad.data_ptr(),
+                       master.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), (scalar_t*)param_out.data_ptr(), n,
+                       (float)clip_scale, (float)lr, (float)beta1, (float)beta2,
+                       (float)eps, (float)wd, (float)bc1, (float)bc2);
+  });
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("seq2col_fwd", &seq2col_fwd);
+  m.def("seq2col_bwd", &seq2col_bwd);
+  m.def("maxout_fwd", &maxout_fwd);
+  m.def("maxout_bwd", &maxout_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("hashembed_fwd", &hashembed_fwd);
+  m.def("hashembed_bwd", &hashembed_bwd);
+  m.def("parser_step_fwd", &parser_step_fwd);
+  m.def("parser_step_bwd", &parser_step_bwd);
+  m.def("adam_step", &adam_step);
+}

@@ -54,14 +54,19 @@ def seq2col_backward(dY: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
 
 
 def maxout(X: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    """[T, W, P] -> ([T, W], argmax [T, W] int8)."""
-    best, which = X.max(dim=-1)
+    """Pieces-major maxout: [..., P, W] -> ([..., W], argmax [..., W] uint8).
+
+    Pieces-major (vs thinc's pieces-last) so that on GPU each piece is a
+    contiguous W-row — coalesced loads and vectorized max in the HIP kernel."""
+    best, which = X.max(dim=-2)
     return best, which.to(torch.uint8)
 
 
 def maxout_backward(dY: torch.Tensor, which: torch.Tensor, P: int) -> torch.Tensor:
-    dX = dY.new_zeros(dY.shape + (P,))
-    dX.scatter_(-1, which.long().unsqueeze(-1), dY.unsqueeze(-1))
+    shape = list(dY.shape)
+    shape.insert(-1, P)
+    dX = dY.new_zeros(shape)
+    dX.scatter_(-2, which.long().unsqueeze(-2), dY.unsqueeze(-2))
     return dX
 
 
@@ -108,20 +113,21 @@ def softmax_ce(scores: torch.Tensor, target: torch.Tensor, mask: Optional[torch.
 
 
 def parser_step_score(
-    precomputed: torch.Tensor,  # [T+1, nF, H*P] (row T = padding zeros)
+    precomputed: torch.Tensor,  # [T+1, nF, P*H] pieces-major (row T = pad)
     feats: torch.Tensor,        # [S, nF] int64 indices into [0, T]; T = missing
-    bias: torch.Tensor,         # [H*P]
+    bias: torch.Tensor,         # [P*H]
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Per-step parser state scorer input: gather nF precomputed rows per
-    state, sum, + bias, maxout over P.  Returns (hidden [S, H], which [S, H]).
+    state, sum, + bias, maxout over P (pieces-major).  Returns
+    (hidden [S, H], which [S, H]).
     (Contract of spaCy parser_model.pyx precompute_hiddens)"""
     S, nF = feats.shape
     HP = precomputed.shape[-1]
     slot = torch.arange(nF, device=feats.device).unsqueeze(0)
-    summed = precomputed[feats.long(), slot].sum(dim=1) + bias  # [S, H*P]
+    summed = precomputed[feats.long(), slot].sum(dim=1) + bias  # [S, P*H]
     P = 2
     H = HP // P
-    hidden, which = maxout(summed.view(S, H, P))
+    hidden, which = maxout(summed.view(S, P, H))
     return hidden, which
 
 

@@ -1,0 +1,152 @@
+"""GPU kernel parity: each gfx950 HIP kernel vs the plain-torch fp32
+reference in ops/torch_ref.py (same op, same inputs).  bf16 variants are
+checked against the fp32 reference with bf16-scale tolerances."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from spacy_ray_amd import _srx_hip
+from spacy_ray_amd.ops import torch_ref as ref
+
+need_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+
+DEV = "cuda:0"
+
+
+def _tol(dtype):
+    return dict(atol=2e-2, rtol=2e-2) if dtype == torch.bfloat16 else dict(atol=1e-5, rtol=1e-5)
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("W", [96, 100])
+def test_seq2col_gpu(dtype, W):
+    torch.manual_seed(0)
+    lengths = torch.tensor([5, 1, 9, 3], device=DEV)
+    T = int(lengths.sum())
+    X = torch.randn(T, W, device=DEV, dtype=dtype)
+    Y = _srx_hip.seq2col_fwd(X, lengths)
+    Yr = ref.seq2col(X.float(), lengths)
+    assert torch.allclose(Y.float(), Yr, **_tol(dtype))
+    dY = torch.randn(T, 3 * W, device=DEV, dtype=dtype)
+    dX = _srx_hip.seq2col_bwd(dY, lengths)
+    dXr = ref.seq2col_backward(dY.float(), lengths)
+    assert torch.allclose(dX.float(), dXr, **_tol(dtype))
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape", [(17, 3, 96), (5, 2, 128), (9, 3, 50)])
+def test_maxout_gpu(dtype, shape):
+    torch.manual_seed(1)
+    X = torch.randn(*shape, device=DEV, dtype=dtype)
+    Y, which = _srx_hip.maxout_fwd(X)
+    Yr, whichr = ref.maxout(X.float())
+    assert torch.allclose(Y.float(), Yr, **_tol(dtype))
+    assert (which.cpu() == whichr.cpu()).float().mean() > 0.99  # fp ties may differ
+    dY = torch.randn_like(Y)
+    dX = _srx_hip.maxout_bwd(dY, which, shape[-2])
+    dXr = ref.maxout_backward(dY.float(), which, shape[-2])
+    assert torch.allclose(dX.float(), dXr, **_tol(dtype))
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("W", [96, 300])
+def test_layernorm_gpu(dtype, W):
+    torch.manual_seed(2)
+    N = 257
+    X = torch.randn(N, W, device=DEV, dtype=dtype)
+    g = torch.randn(W, device=DEV, dtype=dtype)
+    b = torch.randn(W, device=DEV, dtype=dtype)
+    Y, mu, rstd = _srx_hip.layernorm_fwd(X, g, b, 1e-5)
+    Yr = torch.nn.functional.layer_norm(X.float(), (W,), g.float(), b.float(), 1e-5)
+    assert torch.allclose(Y.float(), Yr, **_tol(dtype))
+    dY = torch.randn_like(X)
+    dX, dg, db = _srx_hip.layernorm_bwd(dY, X, g, mu, rstd)
+    X2 = X.float().requires_grad_(True)
+    g2 = g.float().requires_grad_(True)
+    b2 = b.float().requires_grad_(True)
+    Y2 = torch.nn.functional.layer_norm(X2, (W,), g2, b2, 1e-5)
+    Y2.backward(dY.float())
+    tol = dict(atol=5e-2, rtol=5e-2) if dtype == torch.bfloat16 else dict(atol=1e-3, rtol=1e-3)
+    assert torch.allclose(dX.float(), X2.grad, **tol)
+    assert torch.allclose(dg.float(), g2.grad, **tol)
+    assert torch.allclose(db.float(), b2.grad, **tol)
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_hashembed_gpu_matches_cpu_hash(dtype):
+    """GPU murmur must be bit-identical to the C++ CPU murmur."""
+    torch.manual_seed(3)
+    R, W, T = 500, 96, 1000
+    table = torch.randn(R, W, device=DEV, dtype=dtype)
+    ids_np = (np.arange(T, dtype=np.uint64) * np.uint64(0x9E3779B97F4A7C15)) + np.uint64(13)
+    ids = torch.from_numpy(ids_np.view(np.int64)).to(DEV)
+    Y, rows = _srx_hip.hashembed_fwd(table, ids, 7)
+    rows_cpu = ref.hashembed_rows_cpu(ids_np, 7, R)
+    assert (rows.cpu().numpy() == rows_cpu).all(), "GPU murmur != CPU murmur"
+    Yr = ref.hashembed_forward(table.float(), torch.from_numpy(rows_cpu).long().to(DEV))
+    assert torch.allclose(Y.float(), Yr, **_tol(dtype))
+    dY = torch.randn(T, W, device=DEV, dtype=dtype)
+    dT = _srx_hip.hashembed_bwd(dY, rows, R)
+    dTr = ref.hashembed_backward(dY.float(), rows.long(), R)
+    tol = dict(atol=1e-1, rtol=5e-2) if dtype == torch.bfloat16 else dict(atol=1e-3, rtol=1e-4)
+    assert torch.allclose(dT.float(), dTr, **tol)
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("H", [64, 80])
+def test_parser_step_gpu(dtype, H):
+    torch.manual_seed(4)
+    T, nF, S = 50, 13, 37
+    P = 2
+    pre = torch.randn(T + 1, nF, P * H, device=DEV, dtype=dtype)
+    bias = torch.randn(P * H, device=DEV, dtype=dtype)
+    feats = torch.randint(0, T + 1, (S, nF), device=DEV)
+    hidden, which = _srx_hip.parser_step_fwd(pre, feats, bias)
+    hr, whichr = ref.parser_step_score(pre.float(), feats, bias.float())
+    assert torch.allclose(hidden.float(), hr, **_tol(dtype))
+    dH = torch.randn(S, H, device=DEV, dtype=dtype)
+    dPre, dBias = _srx_hip.parser_step_bwd(dH, feats, which, T + 1, nF, 2 * H)
+    # reference backward via autograd
+    pre2 = pre.float().requires_grad_(True)
+    bias2 = bias.float().requires_grad_(True)
+    slot = torch.arange(nF, device=DEV).unsqueeze(0)
+    summed = pre2[feats.long(), slot].sum(dim=1) + bias2
+    h2 = summed.view(S, P, H).max(dim=-2).values
+    h2.backward(dH.float())
+    tol = dict(atol=5e-2, rtol=5e-2) if dtype == torch.bfloat16 else dict(atol=1e-3, rtol=1e-4)
+    assert torch.allclose(dPre.float(), pre2.grad, **tol)
+    assert torch.allclose(dBias.float(), bias2.grad, **tol)
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_adam_step_gpu(dtype):
+    torch.manual_seed(5)
+    n = 10_000
+    grad = torch.randn(n, device=DEV, dtype=dtype)
+    master = torch.randn(n, device=DEV, dtype=torch.float32)
+    m = torch.zeros(n, device=DEV, dtype=torch.float32)
+    v = torch.zeros(n, device=DEV, dtype=torch.float32)
+    param_out = torch.zeros(n, device=DEV, dtype=dtype)
+    lr, b1, b2, eps, wd = 0.01, 0.9, 0.999, 1e-8, 0.01
+    ref_p = master.clone()
+    g32 = grad.float() * 0.5
+    ref_p.mul_(1 - lr * wd)
+    mr = (1 - b1) * g32
+    vr = (1 - b2) * g32 * g32
+    denom = (vr / (1 - b2)).sqrt() + eps
+    ref_p -= (lr / (1 - b1)) * (mr / denom)
+    _srx_hip.adam_step(grad, master, m, v, param_out, 0.5, lr, b1, b2, eps, wd,
+                       1 - b1, 1 - b2)
+    assert torch.allclose(master, ref_p, atol=1e-6)
+    assert torch.allclose(m, mr, atol=1e-6)
+    assert torch.allclose(v, vr, atol=1e-6)
+    assert torch.allclose(param_out.float(), ref_p.float().to(dtype).float(), atol=1e-2)

@@ -39,12 +39,13 @@ def test_seq2col_backward_matches_autograd():
 
 def test_maxout_backward_matches_autograd():
     torch.manual_seed(1)
-    X = torch.randn(7, 4, 3, requires_grad=True)
+    X = torch.randn(7, 3, 4, requires_grad=True)  # pieces-major [T, P, W]
     Y = ops.maxout(X)
+    assert Y.shape == (7, 4)
     g = torch.randn_like(Y)
     Y.backward(g)
     X2 = X.detach().clone().requires_grad_(True)
-    Y2 = X2.max(dim=-1).values
+    Y2 = X2.max(dim=-2).values
     Y2.backward(g)
     assert torch.allclose(X.grad, X2.grad)
 
@@ -102,7 +103,7 @@ def test_parser_step_score_fwd_bwd():
     bias2 = bias.detach().clone().requires_grad_(True)
     slot = torch.arange(nF).unsqueeze(0)
     summed = pre2[feats.long(), slot].sum(dim=1) + bias2
-    h2 = summed.view(S, H, P).max(dim=-1).values
+    h2 = summed.view(S, P, H).max(dim=-2).values  # pieces-major
     h2.backward(g)
     assert torch.allclose(pre.grad, pre2.grad, atol=1e-6)
     assert torch.allclose(bias.grad, bias2.grad, atol=1e-6)
