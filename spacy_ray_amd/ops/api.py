@@ -212,6 +212,55 @@ def parser_step_score(precomputed, feats, bias):
     return _ParserStepScore.apply(precomputed, feats, bias)
 
 
+class _ParserStepScoreAccum(torch.autograd.Function):
+    """Step scorer whose backward ACCUMULATES dPre into one persistent fp32
+    buffer instead of materializing a fresh [T+1, nF, HP] gradient per
+    transition step (the per-step allocation + autograd summation was the
+    dominant host cost — see the two-phase scheme in pipes._step_loop).
+    `precomputed` is passed detached; only `bias` is differentiable here."""
+
+    @staticmethod
+    def forward(ctx, precomputed, feats, bias, dPre32):
+        S, nF = feats.shape
+        HP = precomputed.shape[-1]
+        P = 2
+        H = HP // P
+        if _want_hip(precomputed):
+            hidden, which = hip_ext().parser_step_fwd(precomputed, feats.contiguous(), bias)
+        else:
+            slot = torch.arange(nF, device=feats.device).unsqueeze(0)
+            summed = precomputed[feats.long(), slot].sum(dim=1) + bias
+            hidden, which = ref.maxout(summed.view(S, P, H))
+        ctx.save_for_backward(feats, which)
+        # dPre32 is a side accumulator (mutated across steps) — stash it on
+        # ctx directly so autograd's saved-tensor version check doesn't trip.
+        ctx.dPre32 = dPre32
+        ctx.HP = HP
+        return hidden
+
+    @staticmethod
+    def backward(ctx, dHidden):
+        feats, which = ctx.saved_tensors
+        dPre32 = ctx.dPre32
+        HP = ctx.HP
+        P = 2
+        if _want_hip(dHidden):
+            dBias = hip_ext().parser_step_bwd_into(dHidden.contiguous(), feats, which, dPre32)
+        else:
+            S = dHidden.shape[0]
+            nF = feats.shape[1]
+            dSummed = ref.maxout_backward(dHidden, which, P).reshape(S, HP)
+            dBias = dSummed.sum(dim=0)
+            flat = dPre32.view(-1, HP)
+            idx = (feats.long() * nF + torch.arange(nF, device=feats.device)).reshape(-1)
+            flat.index_add_(0, idx, dSummed.float().repeat_interleave(nF, dim=0))
+        return None, None, dBias.to(dHidden.dtype), None
+
+
+def parser_step_score_accum(precomputed_detached, feats, bias, dPre32):
+    return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32)
+
+
 # ------------------------------------------------------- ragged reductions
 def reduce_mean_ragged(X, lengths):
     if _want_hip(X):

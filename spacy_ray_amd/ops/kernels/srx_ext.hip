@@ -257,6 +257,33 @@ std::vector<at::Tensor> parser_step_fwd(at::Tensor pre, at::Tensor feats, at::Te
   return {hidden, which};
 }
 
+// Accumulating variant: scatter this step's dPre gradient into ONE
+// persistent fp32 buffer (no per-step [T+1,nF,HP] allocation/zeroing — the
+// parser step loop calls this once per transition step; the caller zeroes
+// the buffer once per batch and injects dPre back into autograd via a
+// surrogate product).  Returns the per-step dBias (small, differentiable
+// path for the bias parameter).
+at::Tensor parser_step_bwd_into(at::Tensor dHidden, at::Tensor feats,
+                                at::Tensor which, at::Tensor dPre32) {
+  check_dev(dHidden);
+  long S = feats.size(0);
+  int nF = (int)feats.size(1);
+  int HP = (int)dPre32.size(-1);
+  int H = HP / 2;
+  auto dBiasStep = at::zeros({(long)HP}, dHidden.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (S > 0) {
+    DISPATCH_F(dHidden.scalar_type(), {
+      hipLaunchKernelGGL((parser_step_bwd_kernel<scalar_t>),
+                         dim3(grid_for(S * SRX_WAVE)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)dHidden.data_ptr(), feats.data_ptr<int64_t>(),
+                         which.data_ptr<uint8_t>(), dPre32.data_ptr<float>(),
+                         dBiasStep.data_ptr<float>(), S, nF, H);
+    });
+  }
+  return dBiasStep.to(dHidden.scalar_type());
+}
+
 std::vector<at::Tensor> parser_step_bwd(at::Tensor dHidden, at::Tensor feats,
                                         at::Tensor which, int64_t T1, int64_t nF,
                                         int64_t HP) {
@@ -309,5 +336,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hashembed_bwd", &hashembed_bwd);
   m.def("parser_step_fwd", &parser_step_fwd);
   m.def("parser_step_bwd", &parser_step_bwd);
+  m.def("parser_step_bwd_into", &parser_step_bwd_into);
   m.def("adam_step", &adam_step);
 }

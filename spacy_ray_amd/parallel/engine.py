@@ -121,6 +121,7 @@ class ZeRO1Engine:
         self.exp_avg_sq = torch.zeros_like(self.master)
         self.grad_shard = torch.zeros(self.shard_elems, dtype=self.dtype, device=self.device)
         self._g32 = torch.zeros_like(self.master)
+        self._param_shard: Optional[torch.Tensor] = None
 
         # ---- overlap machinery
         self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
@@ -132,6 +133,12 @@ class ZeRO1Engine:
             p.register_post_accumulate_grad_hook(self._grad_ready_hook)
 
     # ------------------------------------------------------------ internals
+    @staticmethod
+    def _hip_ext():
+        from spacy_ray_amd.ops.api import hip_ext
+
+        return hip_ext()
+
     def _grad_ready_hook(self, p: torch.nn.Parameter) -> None:
         if not (self._sync and self.overlap):
             return
@@ -176,32 +183,47 @@ class ZeRO1Engine:
             torch.cuda.current_stream().wait_stream(self.comm_stream)
 
         s = self.spec
-        self._g32.copy_(self.grad_shard)
-        # global-norm clip: one scalar all-reduce over shard norms
-        if s.grad_clip:
-            sq = self._g32.pow(2).sum()
-            self.comm.all_reduce_(sq)
-            norm = sq.sqrt()
-            scale = torch.clamp(s.grad_clip / (norm + 1e-12), max=1.0)
-            self._g32.mul_(scale)
         lr = s.lr(self.step_count)
         t = self.step_count + 1
-        if s.L2:
-            if s.L2_is_weight_decay:
-                self.master.mul_(1.0 - lr * s.L2)
-            else:
-                self._g32.add_(self.master, alpha=s.L2)
-        self.exp_avg.mul_(s.beta1).add_(self._g32, alpha=1 - s.beta1)
-        self.exp_avg_sq.mul_(s.beta2).addcmul_(self._g32, self._g32, value=1 - s.beta2)
         bc1 = 1 - s.beta1 ** t
         bc2 = 1 - s.beta2 ** t
-        denom = (self.exp_avg_sq / bc2).sqrt_().add_(s.eps)
-        self.master.addcdiv_(self.exp_avg, denom, value=-lr / bc1)
+        wd = s.L2 if (s.L2 and s.L2_is_weight_decay) else 0.0
+        # global-norm clip: one scalar all-reduce over shard norms
+        if s.grad_clip:
+            sq = self.grad_shard.float().pow(2).sum()
+            self.comm.all_reduce_(sq)
+            scale_t = torch.clamp(s.grad_clip / (sq.sqrt() + 1e-12), max=1.0)
+        else:
+            scale_t = None
+        hip = self._hip_ext() if self.is_cuda else None
+        if hip is not None and (s.L2_is_weight_decay or not s.L2):
+            # ONE fused kernel: clip-scale + decoupled wd + Adam + bf16 cast
+            # (SURVEY.md §2.5 fused_adam_sharded)
+            if self._param_shard is None:
+                self._param_shard = torch.empty_like(self.grad_shard)
+            scale = float(scale_t) if scale_t is not None else 1.0
+            hip.adam_step(self.grad_shard, self.master, self.exp_avg,
+                          self.exp_avg_sq, self._param_shard, scale, lr,
+                          s.beta1, s.beta2, s.eps, wd, bc1, bc2)
+            new_param = self._param_shard
+        else:
+            self._g32.copy_(self.grad_shard)
+            if scale_t is not None:
+                self._g32.mul_(scale_t)
+            if s.L2 and not s.L2_is_weight_decay:
+                self._g32.add_(self.master, alpha=s.L2)
+            if wd:
+                self.master.mul_(1.0 - lr * wd)
+            self.exp_avg.mul_(s.beta1).add_(self._g32, alpha=1 - s.beta1)
+            self.exp_avg_sq.mul_(s.beta2).addcmul_(self._g32, self._g32, value=1 - s.beta2)
+            denom = (self.exp_avg_sq / bc2).sqrt_().add_(s.eps)
+            self.master.addcdiv_(self.exp_avg, denom, value=-lr / bc1)
+            new_param = self.master
         # write back + republish
         r = self.comm.rank
         for bkt in self.buckets:
             dst = self.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
-            dst.copy_(self.master[bkt.shard_off : bkt.shard_off + bkt.per].to(self.dtype))
+            dst.copy_(new_param[bkt.shard_off : bkt.shard_off + bkt.per].to(self.dtype))
             bkt.ready = 0
             bkt.launched = False
         for bkt in self.buckets:

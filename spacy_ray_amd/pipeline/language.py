@@ -76,8 +76,8 @@ class Language:
         for name, pipe in self.pipeline:
             if isinstance(pipe, Tok2VecPipe) or name in self._frozen:
                 continue
-            loss, _count = pipe.get_loss(examples, t2v, batch)
-            losses[name] = losses.get(name, 0.0) + float(loss.detach())
+            loss, display = pipe.get_loss(examples, t2v, batch)
+            losses[name] = losses.get(name, 0.0) + display
             total = loss if total is None else total + loss
         if total is None:
             total = torch.zeros((), device=self.device)

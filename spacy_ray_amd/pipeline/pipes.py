@@ -34,7 +34,10 @@ class TrainablePipe:
         raise NotImplementedError
 
     def get_loss(self, examples, t2v, batch) -> Tuple[torch.Tensor, float]:
-        """-> (loss tensor in the autograd graph, float count for logging)"""
+        """-> (loss tensor for the main backward, display float for logging).
+
+        The tensor may be a SURROGATE whose value is meaningless but whose
+        gradient is exact (transition pipes); always log the display float."""
         raise NotImplementedError
 
     def predict_and_set(self, docs, t2v, batch) -> None:
@@ -130,7 +133,7 @@ class TaggerPipe(TrainablePipe):
         loss = torch.nn.functional.cross_entropy(
             scores.float(), gold, ignore_index=-1, reduction="sum"
         ) / max(1, n)
-        return loss, float(n)
+        return loss, float(loss.detach())
 
     def predict_and_set(self, docs, t2v, batch) -> None:
         with torch.no_grad():
@@ -191,9 +194,24 @@ class _TransitionPipeBase(TrainablePipe):
 
     # ---- shared machinery
     def _step_loop(self, states, t2v, train: bool):
+        """Greedy transition loop.
+
+        Training uses a TWO-PHASE backward (the per-step [T+1,nF,HP] gradient
+        tensors autograd would otherwise allocate+sum dominated host time):
+        phase 1 — the per-step losses (upper GEMM + bias path) are
+        backpropagated HERE, while each step's dPre scatters into ONE
+        persistent fp32 buffer (ops.parser_step_score_accum); phase 2 — the
+        returned loss is a surrogate (pre · dPre).sum() whose backward hands
+        exactly dPre to the precompute GEMM inside the caller's single main
+        backward, flowing on to lower_W / pad / tok2vec."""
+        from spacy_ray_amd.ops import api as _ops
+
         device = t2v.device
         T = t2v.shape[0]
         pre = self.module.precompute(t2v)
+        pre_d = pre.detach()
+        if train:
+            dPre32 = torch.zeros(pre.shape, dtype=torch.float32, device=device)
         loss_terms: List[torch.Tensor] = []
         n_states_total = 0
         max_steps = 4 * T + 16
@@ -207,7 +225,13 @@ class _TransitionPipeBase(TrainablePipe):
             feats_t = torch.from_numpy(
                 np.where(feats < 0, T, feats).astype(np.int64)
             ).to(device)
-            scores = self.module.score(pre, feats_t)  # [S, A]
+            if train:
+                hidden = _ops.parser_step_score_accum(
+                    pre_d, feats_t, self.module.lower_b, dPre32
+                )
+                scores = self.module.upper(hidden)  # [S, A]
+            else:
+                scores = self.module.score(pre_d, feats_t)
             valid_t = torch.from_numpy(valid).to(device)
             if train:
                 costs = states.costs()
@@ -231,20 +255,22 @@ class _TransitionPipeBase(TrainablePipe):
                 fallback = np.where(valid, s_np, NEG_INF)
                 choose_from = np.where(counts > 0, choose_from, fallback)
             else:
-                with torch.no_grad():
-                    s_np = scores.float().cpu().numpy()
+                s_np = scores.float().cpu().numpy()
                 choose_from = np.where(valid, s_np, NEG_INF)
             actions = choose_from.argmax(axis=1).astype(np.int32)
             actions[~active] = -1
             states.advance(actions)
-        if train:
-            total = (
-                torch.stack(loss_terms).sum() / max(1, n_states_total)
-                if loss_terms
-                else t2v.new_zeros(())
-            )
-            return total, float(n_states_total)
-        return None, 0.0
+        if not train:
+            return None, 0.0
+        if not loss_terms:
+            return t2v.new_zeros(()), 0.0
+        step_loss = torch.stack(loss_terms).sum() / max(1, n_states_total)
+        display = float(step_loss.detach())
+        step_loss.backward()  # phase 1: upper + lower_b grads; dPre32 filled
+        # dPre32 already carries the 1/n_states normalization (it was filled
+        # by the normalized step_loss backward) — no extra scaling here.
+        surrogate = (pre.float() * dPre32).sum()
+        return surrogate, display
 
     def get_loss(self, examples, t2v, batch):
         lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
@@ -255,7 +281,8 @@ class _TransitionPipeBase(TrainablePipe):
     def predict_and_set(self, docs, t2v, batch) -> None:
         lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
         states = self._make_states(lengths)
-        self._step_loop(states, t2v, train=False)
+        with torch.no_grad():
+            self._step_loop(states, t2v, train=False)
         self._annotate(docs, states)
 
 
