@@ -133,6 +133,14 @@ def main() -> None:
         dist.all_reduce(t_dev, op=dist.ReduceOp.MAX)
         elapsed = float(t_dev.item())
 
+    if rank == 0 and os.environ.get("SRX_TIMING") == "1":
+        from spacy_ray_amd.utils import timing
+
+        tt = timing.phase_times()
+        cc = timing.phase_counts()
+        for k in sorted(tt):
+            print(f"# {k}: {tt[k]:.0f} ms total, {cc[k]} calls", flush=True)
+
     if rank == 0:
         total_words = words_per_step * world * args.steps
         value = total_words / elapsed

@@ -125,6 +125,56 @@ def make_synthetic_docs(
     return docs
 
 
+@registry.readers("spacy-mi.MixedTreebankCorpus.v1")
+def create_mixed_treebank_corpus(
+    n_treebanks: int = 8,
+    docs_per_treebank: int = 500,
+    words_per_doc: int = 18,
+    vocab_size_per_treebank: int = 4000,
+    n_tags: int = 17,
+    n_deps: int = 37,
+    seed: int = 0,
+    shuffle: bool = True,
+):
+    """Multilingual UD-style stress corpus (BASELINE config #5): N disjoint
+    synthetic lexicons (one per 'treebank'/language) sharing one UPOS/dep
+    space, mixed into common batches — Zipf-hot rows from several lexicons
+    hammer the HashEmbed tables at once."""
+    cache: dict = {}
+
+    def corpus(nlp) -> Iterator[Example]:
+        if "docs" not in cache:
+            docs: List[Doc] = []
+            for tb in range(n_treebanks):
+                tb_docs = make_synthetic_docs(
+                    nlp.vocab,
+                    n_docs=docs_per_treebank,
+                    words_per_doc=words_per_doc,
+                    vocab_size=vocab_size_per_treebank,
+                    n_tags=n_tags,
+                    n_deps=n_deps,
+                    n_ent_types=1,
+                    seed=seed * 1000 + tb,
+                    world_seed=tb,  # each treebank is its own learnable world
+                )
+                # language-prefix the word forms so lexicons are disjoint
+                for d in tb_docs:
+                    docs.append(
+                        Doc(nlp.vocab, [f"l{tb}:{w}" for w in d.words],
+                            tags=d.tags, heads=d.heads, deps=d.deps, ents=d.ents)
+                    )
+            cache["docs"] = docs
+            cache["epoch"] = 0
+        docs = list(cache["docs"])
+        if shuffle:
+            random.Random(seed + cache["epoch"]).shuffle(docs)
+        cache["epoch"] += 1
+        for doc in docs:
+            yield Example.from_doc(doc)
+
+    return corpus
+
+
 @registry.readers("spacy-mi.SyntheticCorpus.v1")
 def create_synthetic_corpus(
     n_docs: int = 1000,

@@ -80,6 +80,26 @@ def make_hash_embed_cnn(
     return ModelSpec(build, width=width, kind="tok2vec")
 
 
+@registry.architectures("spacy-transformers.TransformerModel.v3")
+def make_transformer_model(
+    name: str = "roberta-base",
+    get_spans=None,
+    tokenizer_config: Optional[dict] = None,
+    transformer_config: Optional[dict] = None,
+    window: int = 128,
+    stride: int = 96,
+):
+    def build():
+        from .transformer import TransformerTok2Vec
+
+        return TransformerTok2Vec(name=name, window=window, stride=stride,
+                                  transformer_config=transformer_config)
+
+    tc = transformer_config or {}
+    width = tc.get("hidden_size", 768 if name == "roberta-base" else 64)
+    return ModelSpec(build, width=width, kind="tok2vec")
+
+
 @registry.architectures("spacy.Tok2VecListener.v1")
 def make_tok2vec_listener(width: int, upstream: str = "*"):
     return ModelSpec(lambda: None, width=width, kind="listener")

@@ -1,0 +1,106 @@
+"""Transformer tok2vec (the en_core_web_trf path, BASELINE config #4).
+
+Role of spacy-transformers' TransformerModel + listener (SURVEY.md §2.5 trf
+row, §5.7): run a roberta-class encoder over strided sub-word windows and
+pool back to one vector per spaCy token.  MI355X-first adaptations:
+
+* the encoder is a stock ``transformers`` Roberta built from a LOCAL config
+  with random-init weights (no network for checkpoints — BASELINE.md), so
+  all dense math runs through hipBLASLt in bf16 under the same ZeRO-1 flat
+  engine as the CNN path;
+* offline "tokenization": there is no downloadable BPE vocab, so each word
+  maps to a stable pseudo-subword id derived from its murmur NORM hash —
+  the architectural shape (windowing, masking, pooling, listener) is
+  identical, the learned vocabulary is synthetic-data-appropriate;
+* long docs are chopped into windows of ``window`` tokens with stride
+  ``stride`` (overlaps averaged), the spacy-transformers span strategy
+   — so the 512-position limit never binds.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from .batch import TokenBatch
+
+BOS, PAD, EOS = 0, 1, 2
+N_SPECIAL = 10
+
+
+class TransformerTok2Vec(nn.Module):
+    def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
+                 transformer_config: Optional[dict] = None):
+        super().__init__()
+        from transformers import RobertaConfig, RobertaModel
+
+        cfg_kwargs = dict(transformer_config or {})
+        if name == "roberta-base":
+            base = dict(vocab_size=50265, hidden_size=768, num_hidden_layers=12,
+                        num_attention_heads=12, intermediate_size=3072,
+                        max_position_embeddings=514)
+        else:  # local/small variants via transformer_config
+            base = dict(vocab_size=2000, hidden_size=64, num_hidden_layers=2,
+                        num_attention_heads=4, intermediate_size=128,
+                        max_position_embeddings=514)
+        base.update(cfg_kwargs)
+        config = RobertaConfig(**base)
+        self.trf = RobertaModel(config, add_pooling_layer=False)
+        self.width = config.hidden_size
+        self.vocab_size = config.vocab_size
+        self.window = min(window, config.max_position_embeddings - 4)
+        self.stride = min(stride, self.window)
+        assert 0 < self.stride <= self.window
+
+    def _windows(self, lengths: List[int]):
+        """Per doc: window (start, end) pairs over token positions."""
+        spans = []  # (doc_offset+start, doc_offset+end)
+        off = 0
+        for n in lengths:
+            s = 0
+            while True:
+                e = min(s + self.window, n)
+                spans.append((off + s, off + e))
+                if e >= n:
+                    break
+                s += self.stride
+            off += n
+        return spans
+
+    def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
+        device = batch.attr_ids.device
+        T = batch.n_tokens
+        # pseudo-subword ids from the NORM hash (stable, offline)
+        word_ids = (batch.attr_ids[:, 0].remainder(self.vocab_size - N_SPECIAL)
+                    + N_SPECIAL)
+        lengths = [len(d) for d in batch.docs]
+        spans = self._windows(lengths)
+        L = max(e - s for s, e in spans) + 2
+        nW = len(spans)
+        input_ids = torch.full((nW, L), PAD, dtype=torch.long, device=device)
+        attn = torch.zeros(nW, L, dtype=torch.long, device=device)
+        # gather map: flat positions of each window row
+        gather = torch.zeros(nW, L, dtype=torch.long, device=device)
+        valid = torch.zeros(nW, L, dtype=torch.bool, device=device)
+        for w, (s, e) in enumerate(spans):
+            n = e - s
+            input_ids[w, 0] = BOS
+            idx = torch.arange(s, e, device=device)
+            input_ids[w, 1 : 1 + n] = word_ids[idx]
+            input_ids[w, 1 + n] = EOS
+            attn[w, : n + 2] = 1
+            gather[w, 1 : 1 + n] = idx
+            valid[w, 1 : 1 + n] = True
+        out = self.trf(input_ids=input_ids, attention_mask=attn).last_hidden_state
+        # overlap-averaged scatter back to [T, width]
+        acc = out.new_zeros(T, self.width)
+        cnt = out.new_zeros(T, 1)
+        flat_idx = gather[valid]
+        acc.index_add_(0, flat_idx, out[valid])
+        cnt.index_add_(0, flat_idx, out.new_ones(flat_idx.shape[0], 1))
+        Y = acc / cnt.clamp(min=1)
+        if drop and self.training:
+            Y = torch.nn.functional.dropout(Y, drop)
+        return Y

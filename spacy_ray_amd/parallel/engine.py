@@ -169,12 +169,21 @@ class ZeRO1Engine:
     # ------------------------------------------------------------- stepper
     def accumulate(self, examples, drop: float = 0.0, losses: Optional[Dict] = None,
                    sync: bool = True) -> None:
+        from spacy_ray_amd.utils import timing
+
         self._sync = sync
         total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop)
-        total.backward()
+        with timing.phase("bwd/main"):
+            total.backward()
         self._sync = False
 
     def apply_step(self) -> None:
+        from spacy_ray_amd.utils import timing
+
+        with timing.phase("comm+opt/apply_step"):
+            self._apply_step_inner()
+
+    def _apply_step_inner(self) -> None:
         # launch any bucket the hooks didn't (grad-less params, overlap off)
         for bkt in self.buckets:
             if not bkt.launched:

@@ -10,6 +10,13 @@ def make_tok2vec_pipe(name: str, model):
     return Tok2VecPipe(name, model)
 
 
+@registry.factories("transformer")
+def make_transformer_pipe(name: str, model):
+    # the transformer IS the shared tok2vec of the pipeline (listener heads
+    # consume its [T, width] output exactly like the CNN path)
+    return Tok2VecPipe(name, model)
+
+
 @registry.factories("tagger")
 def make_tagger_pipe(name: str, model):
     return TaggerPipe(name, model)

@@ -66,17 +66,22 @@ class Language:
     def forward_loss(self, examples: Sequence[Example], losses: Optional[Dict[str, float]] = None,
                      drop: float = 0.0):
         """One forward pass over all trainable pipes -> (total_loss, losses)."""
+        from spacy_ray_amd.utils import timing
+
         if losses is None:
             losses = {}
-        docs = [eg.predicted for eg in examples]
-        batch = TokenBatch(docs, self.device)
+        with timing.phase("data/batch_build"):
+            docs = [eg.predicted for eg in examples]
+            batch = TokenBatch(docs, self.device)
         t2v_pipe = self.tok2vec
-        t2v = t2v_pipe.forward(batch, drop=drop) if t2v_pipe is not None else None
+        with timing.phase("fwd/tok2vec"):
+            t2v = t2v_pipe.forward(batch, drop=drop) if t2v_pipe is not None else None
         total = None
         for name, pipe in self.pipeline:
             if isinstance(pipe, Tok2VecPipe) or name in self._frozen:
                 continue
-            loss, display = pipe.get_loss(examples, t2v, batch)
+            with timing.phase(f"loss/{name}"):
+                loss, display = pipe.get_loss(examples, t2v, batch)
             losses[name] = losses.get(name, 0.0) + display
             total = loss if total is None else total + loss
         if total is None:

@@ -205,6 +205,7 @@ class _TransitionPipeBase(TrainablePipe):
         exactly dPre to the precompute GEMM inside the caller's single main
         backward, flowing on to lower_W / pad / tok2vec."""
         from spacy_ray_amd.ops import api as _ops
+        from spacy_ray_amd.utils import timing
 
         device = t2v.device
         T = t2v.shape[0]
@@ -216,57 +217,65 @@ class _TransitionPipeBase(TrainablePipe):
         n_states_total = 0
         max_steps = 4 * T + 16
         for _ in range(max_steps):
-            final = states.is_final()
-            if final.all():
-                break
-            active = final == 0
-            feats = states.features()
-            valid = states.valid().astype(bool)
-            feats_t = torch.from_numpy(
-                np.where(feats < 0, T, feats).astype(np.int64)
-            ).to(device)
+            with timing.span("raw/states_cpu"):
+                final = states.is_final()
+                if final.all():
+                    break
+                active = final == 0
+                feats = states.features()
+                valid = states.valid().astype(bool)
+            with timing.span("raw/score_fwd"):
+                feats_t = torch.from_numpy(
+                    np.where(feats < 0, T, feats).astype(np.int64)
+                ).to(device)
+                if train:
+                    hidden = _ops.parser_step_score_accum(
+                        pre_d, feats_t, self.module.lower_b, dPre32
+                    )
+                    scores = self.module.upper(hidden)  # [S, A]
+                else:
+                    scores = self.module.score(pre_d, feats_t)
+                valid_t = torch.from_numpy(valid).to(device)
             if train:
-                hidden = _ops.parser_step_score_accum(
-                    pre_d, feats_t, self.module.lower_b, dPre32
-                )
-                scores = self.module.upper(hidden)  # [S, A]
-            else:
-                scores = self.module.score(pre_d, feats_t)
-            valid_t = torch.from_numpy(valid).to(device)
-            if train:
-                costs = states.costs()
-                cmin = costs.min(axis=1, keepdims=True)
-                is_gold = (costs <= cmin + 1e-6) & valid
-                counts = is_gold.sum(axis=1, keepdims=True)
-                ok = (counts[:, 0] > 0) & active
-                target = is_gold.astype(np.float32) / np.maximum(counts, 1)
-                target_t = torch.from_numpy(target).to(device)
-                logp = torch.log_softmax(
-                    scores.float().masked_fill(~valid_t, NEG_INF), dim=-1
-                )
-                row_loss = -(target_t * logp).sum(dim=-1)
-                ok_t = torch.from_numpy(ok).to(device)
-                loss_terms.append(row_loss.masked_fill(~ok_t, 0).sum())
-                n_states_total += int(active.sum())
-                with torch.no_grad():
-                    s_np = scores.detach().float().cpu().numpy()
+                with timing.span("raw/oracle_cpu"):
+                    costs = states.costs()
+                    cmin = costs.min(axis=1, keepdims=True)
+                    is_gold = (costs <= cmin + 1e-6) & valid
+                    counts = is_gold.sum(axis=1, keepdims=True)
+                    ok = (counts[:, 0] > 0) & active
+                    target = is_gold.astype(np.float32) / np.maximum(counts, 1)
+                with timing.span("raw/loss_build"):
+                    target_t = torch.from_numpy(target).to(device)
+                    logp = torch.log_softmax(
+                        scores.float().masked_fill(~valid_t, NEG_INF), dim=-1
+                    )
+                    row_loss = -(target_t * logp).sum(dim=-1)
+                    ok_t = torch.from_numpy(ok).to(device)
+                    loss_terms.append(row_loss.masked_fill(~ok_t, 0).sum())
+                    n_states_total += int(active.sum())
+                with timing.span("raw/score_d2h"):
+                    with torch.no_grad():
+                        s_np = scores.detach().float().cpu().numpy()
                 choose_from = np.where(is_gold, s_np, NEG_INF)
                 # states where no valid min-cost action exists: any valid one
                 fallback = np.where(valid, s_np, NEG_INF)
                 choose_from = np.where(counts > 0, choose_from, fallback)
             else:
-                s_np = scores.float().cpu().numpy()
+                with timing.span("raw/score_d2h"):
+                    s_np = scores.float().cpu().numpy()
                 choose_from = np.where(valid, s_np, NEG_INF)
-            actions = choose_from.argmax(axis=1).astype(np.int32)
-            actions[~active] = -1
-            states.advance(actions)
+            with timing.span("raw/advance_cpu"):
+                actions = choose_from.argmax(axis=1).astype(np.int32)
+                actions[~active] = -1
+                states.advance(actions)
         if not train:
             return None, 0.0
         if not loss_terms:
             return t2v.new_zeros(()), 0.0
         step_loss = torch.stack(loss_terms).sum() / max(1, n_states_total)
         display = float(step_loss.detach())
-        step_loss.backward()  # phase 1: upper + lower_b grads; dPre32 filled
+        with timing.span("raw/phase1_bwd"):
+            step_loss.backward()  # phase 1: upper + lower_b grads; dPre32 filled
         # dPre32 already carries the 1/n_states normalization (it was filled
         # by the normalized step_loss backward) — no extra scaling here.
         surrogate = (pre.float() * dPre32).sum()

@@ -1,0 +1,123 @@
+"""Transformer tok2vec path (config #4 shape) on CPU with a tiny encoder."""
+import pytest
+import torch
+
+from spacy_ray_amd.config.config import Config, resolve, resolve_dot_names
+from spacy_ray_amd.pipeline.language import init_nlp
+from spacy_ray_amd.train.stepper import SimpleStepper
+
+TRF_CFG = """
+[nlp]
+lang = "en"
+pipeline = ["transformer", "tagger"]
+
+[components]
+
+[components.transformer]
+factory = "transformer"
+
+[components.transformer.model]
+@architectures = "spacy-transformers.TransformerModel.v3"
+name = "tiny"
+window = 16
+stride = 12
+
+[components.transformer.model.transformer_config]
+vocab_size = 512
+hidden_size = 32
+num_hidden_layers = 1
+num_attention_heads = 2
+intermediate_size = 64
+
+[components.tagger]
+factory = "tagger"
+
+[components.tagger.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.tagger.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 32
+
+[corpora]
+
+[corpora.train]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 50
+words_per_doc = 30
+vocab_size = 100
+n_tags = 5
+seed = 0
+
+[corpora.dev]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 20
+words_per_doc = 30
+vocab_size = 100
+n_tags = 5
+seed = 1
+shuffle = false
+
+[training]
+seed = 0
+max_steps = 5
+train_corpus = "corpora.train"
+dev_corpus = "corpora.dev"
+
+[training.batcher]
+@batchers = "spacy.batch_by_words.v1"
+size = 300
+
+[training.optimizer]
+@optimizers = "Adam.v1"
+learn_rate = 0.001
+
+[training.score_weights]
+tag_acc = 1.0
+"""
+
+
+def test_transformer_pipeline_trains_cpu():
+    cfg = Config.from_str(TRF_CFG)
+    nlp = init_nlp(cfg, sample_size=16)
+    icfg = cfg.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    (train_corpus,) = resolve_dot_names(icfg, [T["train_corpus"]])
+    examples = []
+    for eg in train_corpus(nlp):
+        examples.append(eg)
+        if len(examples) >= 8:
+            break
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    losses = {}
+    for _ in range(3):
+        stepper.accumulate(examples, drop=0.0, losses=losses)
+        stepper.apply_step()
+    assert losses["tagger"] > 0
+    # windowing: docs of 30 words with window 16 produce overlapping spans;
+    # output must still be one vector per token
+    from spacy_ray_amd.models.batch import TokenBatch
+
+    batch = TokenBatch([eg.predicted for eg in examples], nlp.device)
+    Y = nlp.tok2vec.forward(batch)
+    assert Y.shape == (batch.n_tokens, 32)
+    assert torch.isfinite(Y).all()
+
+
+def test_mixed_treebank_corpus():
+    from spacy_ray_amd.config.registry import registry
+
+    registry.ensure_populated()
+    corpus = registry.readers.get("spacy-mi.MixedTreebankCorpus.v1")(
+        n_treebanks=3, docs_per_treebank=5, words_per_doc=10,
+        vocab_size_per_treebank=50, n_tags=17, n_deps=37, seed=0)
+
+    class FakeNlp:
+        from spacy_ray_amd.vocab.doc import Vocab
+
+        vocab = Vocab("xx")
+
+    egs = list(corpus(FakeNlp))
+    assert len(egs) == 15
+    prefixes = {eg.reference.words[0].split(":")[0] for eg in egs}
+    assert len(prefixes) == 3  # mixed languages present
