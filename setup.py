@@ -62,4 +62,9 @@ setup(
     packages=["spacy_ray_amd"],
     ext_modules=ext_modules,
     cmdclass=cmdclass,
+    entry_points={
+        "console_scripts": [
+            "spacy-mi = spacy_ray_amd.cli.main:main",
+        ]
+    },
 )

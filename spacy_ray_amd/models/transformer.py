@@ -77,22 +77,23 @@ class TransformerTok2Vec(nn.Module):
                     + N_SPECIAL)
         lengths = [len(d) for d in batch.docs]
         spans = self._windows(lengths)
-        L = max(e - s for s, e in spans) + 2
         nW = len(spans)
-        input_ids = torch.full((nW, L), PAD, dtype=torch.long, device=device)
-        attn = torch.zeros(nW, L, dtype=torch.long, device=device)
-        # gather map: flat positions of each window row
-        gather = torch.zeros(nW, L, dtype=torch.long, device=device)
-        valid = torch.zeros(nW, L, dtype=torch.bool, device=device)
-        for w, (s, e) in enumerate(spans):
-            n = e - s
-            input_ids[w, 0] = BOS
-            idx = torch.arange(s, e, device=device)
-            input_ids[w, 1 : 1 + n] = word_ids[idx]
-            input_ids[w, 1 + n] = EOS
-            attn[w, : n + 2] = 1
-            gather[w, 1 : 1 + n] = idx
-            valid[w, 1 : 1 + n] = True
+        spans_np = np.asarray(spans, dtype=np.int64)
+        starts = torch.from_numpy(spans_np[:, 0]).to(device)
+        ns = torch.from_numpy(spans_np[:, 1] - spans_np[:, 0]).to(device)
+        L = int((spans_np[:, 1] - spans_np[:, 0]).max()) + 2
+        # fully vectorized window assembly (no per-window python loop)
+        pos = torch.arange(L, device=device)
+        tok_pos = pos.unsqueeze(0) - 1  # [1, L] word slot within window
+        valid = (tok_pos >= 0) & (tok_pos < ns.unsqueeze(1))  # [nW, L]
+        gather = (starts.unsqueeze(1) + tok_pos).clamp_(min=0)
+        gather = torch.where(valid, gather, torch.zeros_like(gather))
+        input_ids = torch.where(valid, word_ids[gather],
+                                torch.full_like(gather, PAD))
+        input_ids[:, 0] = BOS
+        eos_col = ns + 1
+        input_ids[torch.arange(nW, device=device), eos_col] = EOS
+        attn = (pos.unsqueeze(0) <= eos_col.unsqueeze(1)).long()
         out = self.trf(input_ids=input_ids, attention_mask=attn).last_hidden_state
         # overlap-averaged scatter back to [T, width]
         acc = out.new_zeros(T, self.width)

@@ -213,7 +213,12 @@ class _TransitionPipeBase(TrainablePipe):
         pre_d = pre.detach()
         if train:
             dPre32 = torch.zeros(pre.shape, dtype=torch.float32, device=device)
-        loss_terms: List[torch.Tensor] = []
+        # per-step collections; the CE loss over all steps is computed ONCE
+        # after the loop (one log_softmax on the concatenated active-state
+        # scores instead of ~2 x doc_len small launches)
+        score_chunks: List[torch.Tensor] = []
+        gold_chunks: List[np.ndarray] = []
+        valid_chunks: List[np.ndarray] = []
         n_states_total = 0
         max_steps = 4 * T + 16
         for _ in range(max_steps):
@@ -222,8 +227,9 @@ class _TransitionPipeBase(TrainablePipe):
                 if final.all():
                     break
                 active = final == 0
-                feats = states.features()
-                valid = states.valid().astype(bool)
+                act_idx = np.nonzero(active)[0]
+                feats = states.features()[act_idx]  # compacted: active only
+                valid = states.valid()[act_idx].astype(bool)
             with timing.span("raw/score_fwd"):
                 feats_t = torch.from_numpy(
                     np.where(feats < 0, T, feats).astype(np.int64)
@@ -232,27 +238,20 @@ class _TransitionPipeBase(TrainablePipe):
                     hidden = _ops.parser_step_score_accum(
                         pre_d, feats_t, self.module.lower_b, dPre32
                     )
-                    scores = self.module.upper(hidden)  # [S, A]
+                    scores = self.module.upper(hidden)  # [S_active, A]
                 else:
                     scores = self.module.score(pre_d, feats_t)
-                valid_t = torch.from_numpy(valid).to(device)
             if train:
                 with timing.span("raw/oracle_cpu"):
-                    costs = states.costs()
+                    costs = states.costs()[act_idx]
                     cmin = costs.min(axis=1, keepdims=True)
                     is_gold = (costs <= cmin + 1e-6) & valid
                     counts = is_gold.sum(axis=1, keepdims=True)
-                    ok = (counts[:, 0] > 0) & active
-                    target = is_gold.astype(np.float32) / np.maximum(counts, 1)
                 with timing.span("raw/loss_build"):
-                    target_t = torch.from_numpy(target).to(device)
-                    logp = torch.log_softmax(
-                        scores.float().masked_fill(~valid_t, NEG_INF), dim=-1
-                    )
-                    row_loss = -(target_t * logp).sum(dim=-1)
-                    ok_t = torch.from_numpy(ok).to(device)
-                    loss_terms.append(row_loss.masked_fill(~ok_t, 0).sum())
-                    n_states_total += int(active.sum())
+                    score_chunks.append(scores)
+                    gold_chunks.append(is_gold)
+                    valid_chunks.append(valid)
+                    n_states_total += len(act_idx)
                 with timing.span("raw/score_d2h"):
                     with torch.no_grad():
                         s_np = scores.detach().float().cpu().numpy()
@@ -265,14 +264,26 @@ class _TransitionPipeBase(TrainablePipe):
                     s_np = scores.float().cpu().numpy()
                 choose_from = np.where(valid, s_np, NEG_INF)
             with timing.span("raw/advance_cpu"):
-                actions = choose_from.argmax(axis=1).astype(np.int32)
-                actions[~active] = -1
+                actions_a = choose_from.argmax(axis=1).astype(np.int32)
+                actions = np.full(len(active), -1, dtype=np.int32)
+                actions[act_idx] = actions_a
                 states.advance(actions)
         if not train:
             return None, 0.0
-        if not loss_terms:
+        if not score_chunks:
             return t2v.new_zeros(()), 0.0
-        step_loss = torch.stack(loss_terms).sum() / max(1, n_states_total)
+        with timing.span("raw/loss_build"):
+            all_scores = torch.cat(score_chunks, dim=0).float()
+            all_gold = torch.from_numpy(np.concatenate(gold_chunks)).to(device)
+            all_valid = torch.from_numpy(np.concatenate(valid_chunks)).to(device)
+            counts_t = all_gold.sum(dim=-1)
+            ok_t = counts_t > 0
+            logp = torch.log_softmax(
+                all_scores.masked_fill(~all_valid, NEG_INF), dim=-1
+            )
+            target = all_gold.float() / counts_t.clamp(min=1).unsqueeze(-1)
+            row_loss = -(target * logp).sum(dim=-1)
+            step_loss = row_loss.masked_fill(~ok_t, 0).sum() / max(1, n_states_total)
         display = float(step_loss.detach())
         with timing.span("raw/phase1_bwd"):
             step_loss.backward()  # phase 1: upper + lower_b grads; dPre32 filled

@@ -1,0 +1,94 @@
+"""DocBin round-trip, Corpus.v1 reader, batchers, attrs."""
+import numpy as np
+import pytest
+
+from spacy_ray_amd.config.registry import registry
+from spacy_ray_amd.data.batcher import (configure_minibatch,
+                                        configure_minibatch_by_words)
+from spacy_ray_amd.data.docbin import DocBin
+from spacy_ray_amd.vocab.attrs import word_shape
+from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+
+def test_word_shape():
+    assert word_shape("Apple") == "Xxxxx"
+    assert word_shape("ABC123") == "XXXddd"
+    assert word_shape("aaaaaaaa") == "xxxx"
+    assert word_shape("don't") == "xxx'x"
+
+
+def test_attr_hashes_shape(vocab):
+    d = Doc(vocab, ["Hello", "world", "!"])
+    assert d.attr_hashes.shape == (3, 4)
+    assert d.attr_hashes.dtype == np.uint64
+    # same word -> same attr hashes
+    d2 = Doc(vocab, ["Hello"])
+    assert (d.attr_hashes[0] == d2.attr_hashes[0]).all()
+
+
+def test_docbin_roundtrip(tmp_path, vocab):
+    docs = [
+        Doc(vocab, ["a", "b", "c"], tags=["T1", "T2", "T1"],
+            heads=[1, -1, 1], deps=["dep0", "ROOT", "dep1"],
+            ents=["O", "U-ORG", "O"]),
+        Doc(vocab, ["x", "y"]),
+    ]
+    db = DocBin(docs)
+    p = tmp_path / "data.spacy"
+    db.to_disk(p)
+    db2 = DocBin.from_disk(p, vocab)
+    assert len(db2) == 2
+    d = db2.docs[0]
+    assert d.words == ["a", "b", "c"]
+    assert d.tags == ["T1", "T2", "T1"]
+    assert d.heads.tolist() == [1, -1, 1]
+    assert d.ents == ["O", "U-ORG", "O"]
+    assert db2.docs[1].tags is None
+
+
+def test_corpus_v1_reader(tmp_path, vocab):
+    registry.ensure_populated()
+    docs = [Doc(vocab, [f"w{i}" for i in range(n)], tags=["T"] * n)
+            for n in (3, 5, 7)]
+    DocBin(docs).to_disk(tmp_path / "train.spacy")
+    reader = registry.readers.get("spacy.Corpus.v1")(path=str(tmp_path / "train.spacy"),
+                                                     max_length=6)
+
+    class FakeNlp:
+        pass
+
+    FakeNlp.vocab = vocab
+    egs = list(reader(FakeNlp))
+    assert len(egs) == 2  # 7-word doc filtered by max_length
+    assert egs[0].reference.tags == ["T", "T", "T"]
+    assert egs[0].predicted.tags is None  # predicted side unannotated
+
+
+def test_batch_by_words():
+    batcher = configure_minibatch_by_words(size=10, tolerance=0.0)
+    items = [[0] * n for n in (4, 4, 4, 9, 20, 2)]
+    batches = list(batcher(items))
+    # 4+4 fits, 4+9 overflows -> [4,4], [4], [9], [20] oversize own batch, [2]
+    sizes = [sum(len(x) for x in b) for b in batches]
+    assert all(s <= 10 or len(b) == 1 for s, b in zip(sizes, batches))
+    assert sum(len(b) for b in batches) == 6  # nothing dropped
+
+
+def test_batch_by_words_discard_oversize():
+    batcher = configure_minibatch_by_words(size=10, tolerance=0.0, discard_oversize=True)
+    items = [[0] * n for n in (4, 20, 4)]
+    batches = list(batcher(items))
+    assert sum(len(b) for b in batches) == 2
+
+
+def test_batch_by_sequence():
+    batcher = configure_minibatch(size=2)
+    batches = list(batcher([1, 2, 3, 4, 5]))
+    assert [len(b) for b in batches] == [2, 2, 1]
+
+
+def test_schedule_compounding():
+    registry.ensure_populated()
+    sched = registry.schedules.get("compounding.v1")(start=1.0, stop=8.0, compound=2.0)
+    vals = [next(sched) for _ in range(5)]
+    assert vals == [1.0, 2.0, 4.0, 8.0, 8.0]
