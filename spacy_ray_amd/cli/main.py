@@ -65,6 +65,8 @@ def ray_train(
     """Launcher (contract of `/root/reference/spacy_ray/train_cli.py:56-91`)."""
     from spacy_ray_amd.parallel.launcher import launch_workers
 
+    if output_path:
+        Path(output_path).mkdir(parents=True, exist_ok=True)
     if n_workers <= 1 and not address:
         # single process: run in-process, no process group
         from spacy_ray_amd.train.worker import distributed_train

@@ -152,8 +152,15 @@ def distributed_train(
         annotating_components=T.get("annotating_components") or [],
         before_update=T.get("before_update"),
     )
+    # test hook: SRX_FAULT_INJECT="rank:step" kills this
+    # rank at that step so the supervisor's all-ranks-abort path is testable
+    fault = os.environ.get("SRX_FAULT_INJECT")
+    fault_rank, fault_step = (int(x) for x in fault.split(":")) if fault else (-1, -1)
+
     best_score = None
     for batch, info, is_best_checkpoint in step_iter:
+        if rank == fault_rank and info["step"] == fault_step:
+            raise SystemExit(41)
         words_cum += info["words"] * world
         final_info = info
         if rank == 0:

@@ -1,0 +1,65 @@
+"""CLI + launcher: end-to-end 2-worker gloo run, checkpoints, resume,
+fault supervision.  These spawn subprocesses (slow-ish but the real path)."""
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+CFG = REPO / "examples" / "configs" / "en_tagger_cpu.cfg"
+
+
+def _run_cli(args, env_extra=None, timeout=420):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.update(env_extra or {})
+    return subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "ray", "train", *args],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=timeout,
+    )
+
+
+def test_cli_two_workers_checkpoint_and_metrics(tmp_path):
+    out = tmp_path / "out"
+    r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
+                  "--training.max_steps", "8", "--training.eval_frequency", "4"])
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert (out / "model-last" / "config.cfg").exists()
+    assert (out / "model-last" / "tagger" / "model.safetensors").exists()
+    assert (out / "model-last" / "optim.rank0.pt").exists()
+    assert (out / "model-last" / "optim.rank1.pt").exists()
+    lines = (out / "metrics.jsonl").read_text().strip().splitlines()
+    rec = json.loads(lines[-1])
+    assert rec["step"] >= 4 and "wps" in rec
+    # model-best written at the eval checkpoint
+    assert (out / "model-best" / "meta.json").exists()
+
+
+def test_cli_single_worker_inprocess(tmp_path):
+    out = tmp_path / "out1"
+    r = _run_cli([str(CFG), "--output", str(out), "--training.max_steps", "4",
+                  "--training.eval_frequency", "2"])
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert (out / "model-last" / "config.cfg").exists()
+
+
+def test_launcher_aborts_all_on_rank_failure(tmp_path):
+    out = tmp_path / "out2"
+    r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
+                  "--training.max_steps", "50", "--training.eval_frequency", "50"],
+                 env_extra={"SRX_FAULT_INJECT": "1:2"})
+    assert r.returncode != 0  # supervisor propagates the failure
+
+
+def test_overrides_reach_workers(tmp_path):
+    out = tmp_path / "out3"
+    r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
+                  "--training.max_steps", "3", "--training.eval_frequency", "2",
+                  "--components.tok2vec.model.width", "64",
+                  "--components.tagger.model.tok2vec.width", "64"])
+    assert r.returncode == 0, r.stderr[-3000:]
+    cfg_text = (out / "model-last" / "config.cfg").read_text()
+    assert '"width": 64' in cfg_text or "width = 64" in cfg_text
