@@ -59,8 +59,11 @@ class MaxoutWindowEncoder(nn.Module):
         )
 
     def forward(self, X: torch.Tensor, lengths: torch.Tensor, drop: float = 0.0) -> torch.Tensor:
+        # doc-boundary masks computed once for all depth layers (and reused
+        # by every seq2col forward AND backward — no per-call index work)
+        starts, ends = ops.boundary_masks_u8(lengths, X.shape[0])
         for block in self.blocks:
-            Y = block(ops.seq2col(X, lengths))
+            Y = block(ops.seq2col(X, lengths, starts, ends))
             if drop and self.training:
                 Y = torch.nn.functional.dropout(Y, drop)
             X = X + Y  # residual (thinc `residual(...)` wrapper)

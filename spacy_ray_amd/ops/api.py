@@ -46,24 +46,43 @@ def _want_hip(t: torch.Tensor) -> bool:
 
 
 # --------------------------------------------------------------- seq2col
+def boundary_masks_u8(lengths: torch.Tensor, total: int):
+    """(is_start, is_end) uint8 device tensors — computed once per batch
+    WITHOUT any device sync (no nonzero/masked_select)."""
+    device = lengths.device
+    starts = torch.zeros(total, dtype=torch.uint8, device=device)
+    ends = torch.zeros(total, dtype=torch.uint8, device=device)
+    l = lengths.long()
+    offs = l.cumsum(0)
+    one = torch.ones_like(l, dtype=torch.uint8)
+    starts.scatter_(0, (offs - l).clamp_(0, max(total - 1, 0)), one)
+    ends.scatter_(0, (offs - 1).clamp_(0, max(total - 1, 0)), one)
+    return starts, ends
+
+
 class _Seq2Col(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, X: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
-        ctx.save_for_backward(lengths)
+    def forward(ctx, X: torch.Tensor, lengths: torch.Tensor,
+                starts: Optional[torch.Tensor], ends: Optional[torch.Tensor]) -> torch.Tensor:
+        if starts is None:
+            starts, ends = boundary_masks_u8(lengths, X.shape[0])
+        ctx.save_for_backward(lengths, starts, ends)
         if _want_hip(X):
-            return hip_ext().seq2col_fwd(X.contiguous(), lengths)
+            return hip_ext().seq2col_fwd(X.contiguous(), starts, ends)
         return ref.seq2col(X, lengths)
 
     @staticmethod
     def backward(ctx, dY: torch.Tensor):
-        (lengths,) = ctx.saved_tensors
+        lengths, starts, ends = ctx.saved_tensors
         if _want_hip(dY):
-            return hip_ext().seq2col_bwd(dY.contiguous(), lengths), None
-        return ref.seq2col_backward(dY, lengths), None
+            return hip_ext().seq2col_bwd(dY.contiguous(), starts, ends), None, None, None
+        return ref.seq2col_backward(dY, lengths), None, None, None
 
 
-def seq2col(X: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
-    return _Seq2Col.apply(X, lengths)
+def seq2col(X: torch.Tensor, lengths: torch.Tensor,
+            starts: Optional[torch.Tensor] = None,
+            ends: Optional[torch.Tensor] = None) -> torch.Tensor:
+    return _Seq2Col.apply(X, lengths, starts, ends)
 
 
 # ---------------------------------------------------------------- maxout

@@ -97,6 +97,50 @@ __global__ void parser_step_fwd_kernel(const T* __restrict__ pre,
   }
 }
 
+// ----------------------------------------------------- action selection
+// One wave per state: actions[s] = argmax of scores over the is_gold
+// columns if any, else over the valid columns (first-occurrence tie-break,
+// matching numpy argmax).  Replaces a [S,A] score D2H + host argmax per
+// transition step with a [S] int32 D2H.
+template <typename T>
+__global__ void action_select_kernel(const T* __restrict__ scores,
+                                     const uint8_t* __restrict__ is_gold,
+                                     const uint8_t* __restrict__ valid,
+                                     int32_t* __restrict__ actions,
+                                     long S, int A) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  for (long s = wave; s < S; s += nwaves) {
+    float bg = -1e38f, bv = -1e38f;
+    int ig = INT32_MAX, iv = INT32_MAX;
+    const T* row = scores + s * (long)A;
+    const uint8_t* grow = is_gold + s * (long)A;
+    const uint8_t* vrow = valid + s * (long)A;
+    for (int a = lane; a < A; a += SRX_WAVE) {
+      float v = Elem<T>::ld(row + a);
+      if (grow[a] && (v > bg || (v == bg && a < ig))) { bg = v; ig = a; }
+      if (vrow[a] && (v > bv || (v == bv && a < iv))) { bv = v; iv = a; }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float obg = __shfl_xor(bg, off, SRX_WAVE);
+      int oig = __shfl_xor(ig, off, SRX_WAVE);
+      if (oig != INT32_MAX && (obg > bg || (obg == bg && oig < ig) || ig == INT32_MAX)) {
+        bg = obg; ig = oig;
+      }
+      float obv = __shfl_xor(bv, off, SRX_WAVE);
+      int oiv = __shfl_xor(iv, off, SRX_WAVE);
+      if (oiv != INT32_MAX && (obv > bv || (obv == bv && oiv < iv) || iv == INT32_MAX)) {
+        bv = obv; iv = oiv;
+      }
+    }
+    if (lane == 0) {
+      actions[s] = ig != INT32_MAX ? ig : (iv != INT32_MAX ? iv : -1);
+    }
+  }
+}
+
 // Backward: scatter dHidden into fp32 workspaces for dPre and dBias.
 template <typename T>
 __global__ void parser_step_bwd_kernel(const T* __restrict__ dHidden,

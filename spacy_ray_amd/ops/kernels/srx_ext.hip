@@ -36,28 +36,16 @@ inline void check_dev(const at::Tensor& t) {
     TORCH_CHECK(false, "unsupported dtype (need f32 or bf16)");       \
   }
 
-std::pair<at::Tensor, at::Tensor> boundary_masks(const at::Tensor& lengths, long nT) {
-  auto opts = at::TensorOptions().dtype(at::kByte).device(lengths.device());
-  auto starts = at::zeros({nT}, opts);
-  auto ends = at::zeros({nT}, opts);
-  auto l = lengths.to(at::kLong);
-  auto offs = l.cumsum(0);
-  auto nz = l > 0;
-  auto s_idx = (offs - l).masked_select(nz);
-  auto e_idx = (offs - 1).masked_select(nz);
-  starts.index_fill_(0, s_idx, 1);
-  ends.index_fill_(0, e_idx, 1);
-  return {starts, ends};
-}
-
 // --------------------------------------------------------------- seq2col
-at::Tensor seq2col_fwd(at::Tensor X, at::Tensor lengths) {
+// starts/ends: per-token doc-boundary uint8 masks, computed ONCE per batch
+// on the python side (a masked_select/nonzero here would device-sync on
+// every call — measured as a 33 ms/step stall in the encoder stack).
+at::Tensor seq2col_fwd(at::Tensor X, at::Tensor starts, at::Tensor ends) {
   check_dev(X);
   long nT = X.size(0);
   int W = (int)X.size(1);
   auto Y = at::empty({nT, 3L * W}, X.options());
   if (nT == 0) return Y;
-  auto [starts, ends] = boundary_masks(lengths, nT);
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(X.scalar_type(), {
     const int V = (W % kVec == 0) ? kVec : 1;
@@ -76,14 +64,13 @@ at::Tensor seq2col_fwd(at::Tensor X, at::Tensor lengths) {
   return Y;
 }
 
-at::Tensor seq2col_bwd(at::Tensor dY, at::Tensor lengths) {
+at::Tensor seq2col_bwd(at::Tensor dY, at::Tensor starts, at::Tensor ends) {
   check_dev(dY);
   long nT = dY.size(0);
   int W3 = (int)dY.size(1);
   int W = W3 / 3;
   auto dX = at::empty({nT, (long)W}, dY.options());
   if (nT == 0) return dX;
-  auto [starts, ends] = boundary_masks(lengths, nT);
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(dY.scalar_type(), {
     const int V = (W % kVec == 0) ? kVec : 1;
@@ -185,9 +172,13 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
   auto dg32 = at::zeros({W}, X.options().dtype(at::kFloat));
   auto db32 = at::zeros({W}, X.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
+  TORCH_CHECK(W <= SRX_LN_MAX_W, "layernorm width > ", SRX_LN_MAX_W);
   if (N > 0) {
+    // cap the grid so each wave covers many rows: the dg/db column sums are
+    // register-accumulated per wave with one atomic per column at the end
+    int grid = (int)std::min<long>((N + 3) / 4, 1024);
     DISPATCH_F(X.scalar_type(), {
-      hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t>), dim3(grid_for(N * SRX_WAVE)),
+      hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t>), dim3(grid),
                          dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
                          (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
                          mu.data_ptr<float>(), rstd.data_ptr<float>(),
@@ -305,6 +296,22 @@ std::vector<at::Tensor> parser_step_bwd(at::Tensor dHidden, at::Tensor feats,
   return {dPre32.to(dHidden.scalar_type()), dBias32.to(dHidden.scalar_type())};
 }
 
+at::Tensor action_select(at::Tensor scores, at::Tensor is_gold, at::Tensor valid) {
+  check_dev(scores);
+  long S = scores.size(0);
+  int A = (int)scores.size(1);
+  auto actions = at::empty({S}, scores.options().dtype(at::kInt));
+  if (S == 0) return actions;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(scores.scalar_type(), {
+    hipLaunchKernelGGL((action_select_kernel<scalar_t>), dim3(grid_for(S * SRX_WAVE)),
+                       dim3(kBlock), 0, stream, (const scalar_t*)scores.data_ptr(),
+                       is_gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
+                       actions.data_ptr<int32_t>(), S, A);
+  });
+  return actions;
+}
+
 // ------------------------------------------------------------ fused Adam
 void adam_step(at::Tensor grad, at::Tensor master, at::Tensor m, at::Tensor v,
                at::Tensor param_out, double clip_scale, double lr, double beta1,
@@ -337,5 +344,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parser_step_fwd", &parser_step_fwd);
   m.def("parser_step_bwd", &parser_step_bwd);
   m.def("parser_step_bwd_into", &parser_step_bwd_into);
+  m.def("action_select", &action_select);
   m.def("adam_step", &adam_step);
 }

@@ -213,12 +213,13 @@ class _TransitionPipeBase(TrainablePipe):
         pre_d = pre.detach()
         if train:
             dPre32 = torch.zeros(pre.shape, dtype=torch.float32, device=device)
+        hip = _ops.hip_ext() if device.type == "cuda" else None
         # per-step collections; the CE loss over all steps is computed ONCE
         # after the loop (one log_softmax on the concatenated active-state
         # scores instead of ~2 x doc_len small launches)
         score_chunks: List[torch.Tensor] = []
-        gold_chunks: List[np.ndarray] = []
-        valid_chunks: List[np.ndarray] = []
+        gold_chunks: List[torch.Tensor] = []
+        valid_chunks: List[torch.Tensor] = []
         n_states_total = 0
         max_steps = 4 * T + 16
         for _ in range(max_steps):
@@ -229,7 +230,7 @@ class _TransitionPipeBase(TrainablePipe):
                 active = final == 0
                 act_idx = np.nonzero(active)[0]
                 feats = states.features()[act_idx]  # compacted: active only
-                valid = states.valid()[act_idx].astype(bool)
+                valid_np = states.valid()[act_idx]
             with timing.span("raw/score_fwd"):
                 feats_t = torch.from_numpy(
                     np.where(feats < 0, T, feats).astype(np.int64)
@@ -241,30 +242,36 @@ class _TransitionPipeBase(TrainablePipe):
                     scores = self.module.upper(hidden)  # [S_active, A]
                 else:
                     scores = self.module.score(pre_d, feats_t)
+                valid_t = torch.from_numpy(valid_np).to(device)
             if train:
                 with timing.span("raw/oracle_cpu"):
                     costs = states.costs()[act_idx]
                     cmin = costs.min(axis=1, keepdims=True)
-                    is_gold = (costs <= cmin + 1e-6) & valid
-                    counts = is_gold.sum(axis=1, keepdims=True)
+                    is_gold = ((costs <= cmin + 1e-6) & (valid_np > 0)).astype(np.uint8)
                 with timing.span("raw/loss_build"):
+                    gold_t = torch.from_numpy(is_gold).to(device)
                     score_chunks.append(scores)
-                    gold_chunks.append(is_gold)
-                    valid_chunks.append(valid)
+                    gold_chunks.append(gold_t)
+                    valid_chunks.append(valid_t)
                     n_states_total += len(act_idx)
-                with timing.span("raw/score_d2h"):
-                    with torch.no_grad():
-                        s_np = scores.detach().float().cpu().numpy()
-                choose_from = np.where(is_gold, s_np, NEG_INF)
-                # states where no valid min-cost action exists: any valid one
-                fallback = np.where(valid, s_np, NEG_INF)
-                choose_from = np.where(counts > 0, choose_from, fallback)
+                sel_gold, sel_valid = gold_t, valid_t
             else:
-                with timing.span("raw/score_d2h"):
-                    s_np = scores.float().cpu().numpy()
-                choose_from = np.where(valid, s_np, NEG_INF)
+                sel_gold, sel_valid = valid_t, valid_t
+            with timing.span("raw/score_d2h"):
+                if hip is not None:
+                    actions_a = (
+                        hip.action_select(scores.detach(), sel_gold, sel_valid)
+                        .cpu().numpy().astype(np.int32)
+                    )
+                else:
+                    s_np = scores.detach().float().cpu().numpy()
+                    g_np = is_gold if train else valid_np
+                    choose = np.where(g_np > 0, s_np, NEG_INF)
+                    fallback = np.where(valid_np > 0, s_np, NEG_INF)
+                    any_gold = (g_np > 0).any(axis=1, keepdims=True)
+                    choose = np.where(any_gold, choose, fallback)
+                    actions_a = choose.argmax(axis=1).astype(np.int32)
             with timing.span("raw/advance_cpu"):
-                actions_a = choose_from.argmax(axis=1).astype(np.int32)
                 actions = np.full(len(active), -1, dtype=np.int32)
                 actions[act_idx] = actions_a
                 states.advance(actions)
@@ -274,8 +281,8 @@ class _TransitionPipeBase(TrainablePipe):
             return t2v.new_zeros(()), 0.0
         with timing.span("raw/loss_build"):
             all_scores = torch.cat(score_chunks, dim=0).float()
-            all_gold = torch.from_numpy(np.concatenate(gold_chunks)).to(device)
-            all_valid = torch.from_numpy(np.concatenate(valid_chunks)).to(device)
+            all_gold = torch.cat(gold_chunks, dim=0) > 0
+            all_valid = torch.cat(valid_chunks, dim=0) > 0
             counts_t = all_gold.sum(dim=-1)
             ok_t = counts_t > 0
             logp = torch.log_softmax(

@@ -28,11 +28,14 @@ def test_seq2col_gpu(dtype, W):
     lengths = torch.tensor([5, 1, 9, 3], device=DEV)
     T = int(lengths.sum())
     X = torch.randn(T, W, device=DEV, dtype=dtype)
-    Y = _srx_hip.seq2col_fwd(X, lengths)
+    from spacy_ray_amd.ops.api import boundary_masks_u8
+
+    starts, ends = boundary_masks_u8(lengths, T)
+    Y = _srx_hip.seq2col_fwd(X, starts, ends)
     Yr = ref.seq2col(X.float(), lengths)
     assert torch.allclose(Y.float(), Yr, **_tol(dtype))
     dY = torch.randn(T, 3 * W, device=DEV, dtype=dtype)
-    dX = _srx_hip.seq2col_bwd(dY, lengths)
+    dX = _srx_hip.seq2col_bwd(dY, starts, ends)
     dXr = ref.seq2col_backward(dY.float(), lengths)
     assert torch.allclose(dX.float(), dXr, **_tol(dtype))
 
@@ -124,6 +127,26 @@ def test_parser_step_gpu(dtype, H):
     tol = dict(atol=5e-2, rtol=5e-2) if dtype == torch.bfloat16 else dict(atol=1e-3, rtol=1e-4)
     assert torch.allclose(dPre.float(), pre2.grad, **tol)
     assert torch.allclose(dBias.float(), bias2.grad, **tol)
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_action_select_gpu(dtype):
+    torch.manual_seed(6)
+    S, A = 123, 82
+    scores = torch.randn(S, A, device=DEV, dtype=dtype)
+    is_gold = (torch.rand(S, A, device=DEV) < 0.05).to(torch.uint8)
+    valid = ((torch.rand(S, A, device=DEV) < 0.4).to(torch.uint8) | is_gold)
+    is_gold[0] = 0  # state with no gold -> falls back to valid
+    acts = _srx_hip.action_select(scores, is_gold, valid).cpu().numpy()
+    s_np = scores.float().cpu().numpy()
+    g_np = is_gold.cpu().numpy()
+    v_np = valid.cpu().numpy()
+    choose = np.where(g_np > 0, s_np, -1e30)
+    fallback = np.where(v_np > 0, s_np, -1e30)
+    any_gold = (g_np > 0).any(axis=1, keepdims=True)
+    expect = np.where(any_gold, choose, fallback).argmax(axis=1)
+    assert (acts == expect).all()
 
 
 @need_gpu
