@@ -16,15 +16,33 @@ from spacy_ray_amd.vocab.doc import Doc
 
 
 class TokenBatch:
-    __slots__ = ("attr_ids", "lengths", "n_tokens", "docs")
+    __slots__ = ("attr_ids", "lengths", "n_tokens", "n_real_tokens", "docs")
 
-    def __init__(self, docs: Sequence[Doc], device: torch.device):
+    def __init__(self, docs: Sequence[Doc], device: torch.device,
+                 pad_to: int = 0):
+        """pad_to > 0: round the token count up to a multiple by appending a
+        pad pseudo-doc (zero attr ids, no annotations).  Keeps the GEMM M
+        dims on a small set of repeating shapes so hipBLASLt's solution
+        cache hits instead of re-selecting per batch.  Defaults: 2048 on
+        GPU, none on CPU."""
+        if isinstance(device, str):
+            device = torch.device(device)
+        if pad_to == 0:
+            pad_to = 2048 if device.type == "cuda" else 1
         self.docs = list(docs)
-        lens = np.array([len(d) for d in docs], dtype=np.int64)
-        if len(docs):
-            attr = np.concatenate([d.attr_hashes for d in docs], axis=0)
-        else:
-            attr = np.zeros((0, 4), dtype=np.uint64)
+        lens_list = [len(d) for d in docs]
+        real = int(sum(lens_list))
+        self.n_real_tokens = real
+        padded = -(-max(real, 1) // pad_to) * pad_to if pad_to > 1 else real
+        n_pad = padded - real
+        if n_pad > 0:
+            lens_list = lens_list + [n_pad]
+        lens = np.array(lens_list, dtype=np.int64)
+        arrs = [d.attr_hashes for d in docs]
+        if n_pad > 0:
+            arrs.append(np.zeros((n_pad, 4), dtype=np.uint64))
+        attr = (np.concatenate(arrs, axis=0) if arrs
+                else np.zeros((0, 4), dtype=np.uint64))
         self.n_tokens = int(lens.sum())
         # bit-cast uint64 -> int64 (torch has no uint64); kernels re-interpret
         self.attr_ids = torch.from_numpy(attr.view(np.int64)).to(device, non_blocking=True)

@@ -280,8 +280,97 @@ def parser_step_score_accum(precomputed_detached, feats, bias, dPre32):
     return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32)
 
 
+# ----------------------------------------------------------- softmax + CE
+class _SoftmaxCE(torch.autograd.Function):
+    """Fused softmax + cross-entropy (SURVEY.md §2.5 softmax_ce_fwd/bwd):
+    forward returns the summed NLL over rows with gold >= 0; backward hands
+    (softmax - onehot) * grad_out to the scores."""
+
+    @staticmethod
+    def forward(ctx, scores: torch.Tensor, gold: torch.Tensor):
+        loss_count, dScores = hip_ext().softmax_ce(scores.contiguous(), gold)
+        ctx.save_for_backward(dScores)
+        return loss_count[0]
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (dScores,) = ctx.saved_tensors
+        return dScores * grad_out.to(dScores.dtype), None
+
+
+def softmax_ce_loss(scores: torch.Tensor, gold: torch.Tensor) -> torch.Tensor:
+    """Summed cross-entropy; rows with gold < 0 are ignored.  GPU: fused
+    kernel.  CPU: torch cross_entropy."""
+    if scores.is_cuda and _want_hip(scores):
+        return _SoftmaxCE.apply(scores, gold)
+    return torch.nn.functional.cross_entropy(
+        scores.float(), gold, ignore_index=-1, reduction="sum"
+    )
+
+
 # ------------------------------------------------------- ragged reductions
+def _offsets_and_docof(lengths: torch.Tensor):
+    l = lengths.to(torch.int32)
+    offsets = torch.zeros(l.shape[0] + 1, dtype=torch.int32, device=l.device)
+    offsets[1:] = torch.cumsum(l, 0)
+    doc_of = torch.repeat_interleave(
+        torch.arange(l.shape[0], device=l.device, dtype=torch.int32), lengths.long()
+    )
+    return offsets, doc_of
+
+
+class _ReduceRagged(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, X, lengths, mode: int):
+        if _want_hip(X):
+            offsets, doc_of = _offsets_and_docof(lengths)
+            out, argmax = hip_ext().reduce_ragged(X.contiguous(), offsets, mode)
+            ctx.save_for_backward(lengths, offsets, doc_of, argmax)
+        else:
+            if mode == 0:
+                out = ref.reduce_sum_ragged(X, lengths)
+            elif mode == 1:
+                out = ref.reduce_mean_ragged(X, lengths)
+            else:
+                out, argmax_l = ref.reduce_max_ragged(X, lengths)
+                ctx.argmax_cpu = argmax_l
+            ctx.save_for_backward(lengths)
+        ctx.mode = mode
+        ctx.T = X.shape[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, dY):
+        mode, T = ctx.mode, ctx.T
+        if _want_hip(dY):
+            lengths, offsets, doc_of, argmax = ctx.saved_tensors
+            if mode == 2:
+                dX = hip_ext().reduce_max_bwd(dY.contiguous(), argmax, T)
+            else:
+                dX = hip_ext().reduce_ragged_bwd(dY.contiguous(), doc_of, offsets, T, mode)
+            return dX, None, None
+        (lengths,) = ctx.saved_tensors
+        if mode == 2:
+            dX = dY.new_zeros(T, dY.shape[1])
+            which = ctx.argmax_cpu
+            dX.scatter_(0, which, dY)
+            return dX, None, None
+        seg = torch.repeat_interleave(
+            torch.arange(lengths.shape[0], device=dY.device), lengths.long()
+        )
+        scale = dY
+        if mode == 1:
+            scale = dY / lengths.clamp(min=1).unsqueeze(1).to(dY.dtype)
+        return scale[seg], None, None
+
+
+def reduce_sum_ragged(X, lengths):
+    return _ReduceRagged.apply(X, lengths, 0)
+
+
 def reduce_mean_ragged(X, lengths):
-    if _want_hip(X):
-        return hip_ext().reduce_mean_ragged(X.contiguous(), lengths)
-    return ref.reduce_mean_ragged(X, lengths)
+    return _ReduceRagged.apply(X, lengths, 1)
+
+
+def reduce_max_ragged(X, lengths):
+    return _ReduceRagged.apply(X, lengths, 2)

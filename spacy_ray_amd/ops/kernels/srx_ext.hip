@@ -8,6 +8,7 @@
 #include "srx_common.hip.h"
 #include "srx_elementwise.hip.h"
 #include "srx_embed_parser.hip.h"
+#include "srx_softmax_reduce.hip.h"
 
 namespace {
 
@@ -330,6 +331,95 @@ void adam_step(at::Tensor grad, at::Tensor master, at::Tensor m, at::Tensor v,
   });
 }
 
+// ------------------------------------------------------- softmax + CE
+// returns (loss_and_count fp32 [2], dScores) — dScores = softmax - onehot
+// (unnormalized; the python wrapper divides by the valid count).
+std::vector<at::Tensor> softmax_ce(at::Tensor scores, at::Tensor gold) {
+  check_dev(scores);
+  TORCH_CHECK(gold.scalar_type() == at::kLong);
+  long N = scores.size(0);
+  int C = (int)scores.size(1);
+  auto dScores = at::empty_like(scores);
+  auto loss = at::zeros({2}, scores.options().dtype(at::kFloat));
+  if (N == 0) return {loss, dScores};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = (int)std::min<long>((N + 3) / 4, 2048);
+  DISPATCH_F(scores.scalar_type(), {
+    hipLaunchKernelGGL((softmax_ce_kernel<scalar_t>), dim3(grid), dim3(kBlock), 0,
+                       stream, (const scalar_t*)scores.data_ptr(),
+                       gold.data_ptr<int64_t>(), (scalar_t*)dScores.data_ptr(),
+                       loss.data_ptr<float>(), N, C);
+  });
+  return {loss, dScores};
+}
+
+// ---------------------------------------------- segmented reductions
+std::vector<at::Tensor> reduce_ragged(at::Tensor X, at::Tensor offsets, int64_t mode) {
+  check_dev(X);
+  long N = offsets.size(0) - 1;
+  int W = (int)X.size(1);
+  auto out = at::empty({N, (long)W}, X.options());
+  auto argmax = mode == 2 ? at::empty({N, (long)W}, X.options().dtype(at::kInt))
+                          : at::empty({0}, X.options().dtype(at::kInt));
+  if (N == 0) return {out, argmax};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = grid_for(N * SRX_WAVE);
+  DISPATCH_F(X.scalar_type(), {
+    if (mode == 0)
+      hipLaunchKernelGGL((reduce_ragged_kernel<scalar_t, 0>), dim3(grid), dim3(kBlock),
+                         0, stream, (const scalar_t*)X.data_ptr(),
+                         offsets.data_ptr<int32_t>(), (scalar_t*)out.data_ptr(),
+                         nullptr, N, W);
+    else if (mode == 1)
+      hipLaunchKernelGGL((reduce_ragged_kernel<scalar_t, 1>), dim3(grid), dim3(kBlock),
+                         0, stream, (const scalar_t*)X.data_ptr(),
+                         offsets.data_ptr<int32_t>(), (scalar_t*)out.data_ptr(),
+                         nullptr, N, W);
+    else
+      hipLaunchKernelGGL((reduce_ragged_kernel<scalar_t, 2>), dim3(grid), dim3(kBlock),
+                         0, stream, (const scalar_t*)X.data_ptr(),
+                         offsets.data_ptr<int32_t>(), (scalar_t*)out.data_ptr(),
+                         argmax.data_ptr<int32_t>(), N, W);
+  });
+  return {out, argmax};
+}
+
+at::Tensor reduce_ragged_bwd(at::Tensor dY, at::Tensor doc_of, at::Tensor offsets,
+                             int64_t Ttot, int64_t mode) {
+  check_dev(dY);
+  int W = (int)dY.size(1);
+  auto dX = at::empty({Ttot, (long)W}, dY.options());
+  if (Ttot == 0) return dX;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(dY.scalar_type(), {
+    if (mode == 0)
+      hipLaunchKernelGGL((reduce_ragged_bwd_kernel<scalar_t, 0>),
+                         dim3(grid_for(Ttot * (long)W)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)dY.data_ptr(), doc_of.data_ptr<int32_t>(),
+                         offsets.data_ptr<int32_t>(), (scalar_t*)dX.data_ptr(), Ttot, W);
+    else
+      hipLaunchKernelGGL((reduce_ragged_bwd_kernel<scalar_t, 1>),
+                         dim3(grid_for(Ttot * (long)W)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)dY.data_ptr(), doc_of.data_ptr<int32_t>(),
+                         offsets.data_ptr<int32_t>(), (scalar_t*)dX.data_ptr(), Ttot, W);
+  });
+  return dX;
+}
+
+at::Tensor reduce_max_bwd(at::Tensor dY, at::Tensor argmax, int64_t Ttot) {
+  check_dev(dY);
+  long N = dY.size(0);
+  int W = (int)dY.size(1);
+  auto dX = at::zeros({Ttot, (long)W}, dY.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(dY.scalar_type(), {
+    hipLaunchKernelGGL((reduce_max_bwd_kernel<scalar_t>), dim3(grid_for(N * (long)W)),
+                       dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                       argmax.data_ptr<int32_t>(), (scalar_t*)dX.data_ptr(), N, W);
+  });
+  return dX;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -346,4 +436,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parser_step_bwd_into", &parser_step_bwd_into);
   m.def("action_select", &action_select);
   m.def("adam_step", &adam_step);
+  m.def("softmax_ce", &softmax_ce);
+  m.def("reduce_ragged", &reduce_ragged);
+  m.def("reduce_ragged_bwd", &reduce_ragged_bwd);
+  m.def("reduce_max_bwd", &reduce_max_bwd);
 }

@@ -119,20 +119,23 @@ class TaggerPipe(TrainablePipe):
         self.label2id = {t: i for i, t in enumerate(self.labels)}
         self.module = TaggerHead(self.width, max(1, len(self.labels))).to(device)
 
-    def _gold_ids(self, examples) -> np.ndarray:
+    def _gold_ids(self, examples, n_tokens: int) -> np.ndarray:
         ids = []
         for eg in examples:
             tags = eg.reference.tags or ["" for _ in range(len(eg.reference))]
             ids.extend(self.label2id.get(t, -1) for t in tags)
+        if len(ids) < n_tokens:  # batch pad rows: ignore_index
+            ids.extend([-1] * (n_tokens - len(ids)))
         return np.asarray(ids, dtype=np.int64)
 
     def get_loss(self, examples, t2v, batch):
+        from spacy_ray_amd.ops import api as _ops
+
         scores = self.module(t2v)  # [T, nT]
-        gold = torch.from_numpy(self._gold_ids(examples)).to(scores.device)
-        n = int((gold >= 0).sum())
-        loss = torch.nn.functional.cross_entropy(
-            scores.float(), gold, ignore_index=-1, reduction="sum"
-        ) / max(1, n)
+        gold_np = self._gold_ids(examples, t2v.shape[0])
+        gold = torch.from_numpy(gold_np).to(scores.device)
+        n = int((gold_np >= 0).sum())
+        loss = _ops.softmax_ce_loss(scores, gold) / max(1, n)
         return loss, float(loss.detach())
 
     def predict_and_set(self, docs, t2v, batch) -> None:

@@ -151,6 +151,58 @@ def test_action_select_gpu(dtype):
 
 @need_gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_softmax_ce_gpu(dtype):
+    torch.manual_seed(7)
+    N, C = 513, 50
+    scores = torch.randn(N, C, device=DEV, dtype=dtype)
+    gold = torch.randint(0, C, (N,), device=DEV)
+    gold[::7] = -1
+    loss_count, d = _srx_hip.softmax_ce(scores, gold)
+    s32 = scores.float()
+    lr = torch.nn.functional.cross_entropy(s32, gold, ignore_index=-1, reduction="sum")
+    tol = dict(atol=5e-2, rtol=1e-2) if dtype == torch.bfloat16 else dict(atol=1e-2, rtol=1e-4)
+    assert torch.allclose(loss_count[0], lr, **tol)
+    assert int(loss_count[1]) == int((gold >= 0).sum())
+    s2 = s32.requires_grad_(True)
+    l2 = torch.nn.functional.cross_entropy(s2, gold, ignore_index=-1, reduction="sum")
+    l2.backward()
+    assert torch.allclose(d.float(), s2.grad, **_tol(dtype))
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("mode", [0, 1, 2])
+def test_reduce_ragged_gpu(dtype, mode):
+    torch.manual_seed(8)
+    lengths = torch.tensor([3, 1, 7, 5], device=DEV)
+    T, W = 16, 96
+    X = torch.randn(T, W, device=DEV, dtype=dtype)
+    from spacy_ray_amd.ops.api import _offsets_and_docof
+
+    offsets, doc_of = _offsets_and_docof(lengths)
+    out, argmax = _srx_hip.reduce_ragged(X, offsets, mode)
+    if mode == 0:
+        expect = ref.reduce_sum_ragged(X.float(), lengths)
+    elif mode == 1:
+        expect = ref.reduce_mean_ragged(X.float(), lengths)
+    else:
+        expect, which = ref.reduce_max_ragged(X.float(), lengths)
+    assert torch.allclose(out.float(), expect, **_tol(dtype))
+    dY = torch.randn(4, W, device=DEV, dtype=dtype)
+    if mode == 2:
+        dX = _srx_hip.reduce_max_bwd(dY, argmax, T)
+        dXr = torch.zeros(T, W, device=DEV)
+        dXr.scatter_(0, which, dY.float())
+        assert torch.allclose(dX.float(), dXr, **_tol(dtype))
+    else:
+        dX = _srx_hip.reduce_ragged_bwd(dY, doc_of, offsets, T, mode)
+        seg = torch.repeat_interleave(torch.arange(4, device=DEV), lengths)
+        scale = dY.float() / lengths.clamp(min=1).unsqueeze(1).float() if mode == 1 else dY.float()
+        assert torch.allclose(dX.float(), scale[seg], **_tol(dtype))
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_adam_step_gpu(dtype):
     torch.manual_seed(5)
     n = 10_000
