@@ -39,6 +39,7 @@ def ray_train_cli(
     address: Optional[str] = typer.Option(None, "--address", "-a", help="Rendezvous address host[:port] (multi-node)"),
     use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
     verbose: bool = typer.Option(False, "--verbose", "-V", help="Display more information"),
+    resume: bool = typer.Option(False, "--resume", help="Resume from <output>/model-last (params + per-rank optimizer shards)"),
 ):
     """Train a pipeline with N data-parallel workers over RCCL/xGMI."""
     logging.basicConfig(level=logging.DEBUG if verbose else logging.ERROR)
@@ -47,7 +48,7 @@ def ray_train_cli(
     raise SystemExit(
         ray_train(config, config_path=config_path, output_path=output_path,
                   code_path=code_path, n_workers=n_workers, address=address,
-                  use_gpu=use_gpu, overrides=overrides)
+                  use_gpu=use_gpu, overrides=overrides, resume=resume)
     )
 
 
@@ -61,6 +62,7 @@ def ray_train(
     address: Optional[str] = None,
     use_gpu: int = -1,
     overrides: Optional[dict] = None,
+    resume: bool = False,
 ) -> int:
     """Launcher (contract of `/root/reference/spacy_ray/train_cli.py:56-91`)."""
     from spacy_ray_amd.parallel.launcher import launch_workers
@@ -76,6 +78,7 @@ def ray_train(
             output_path=output_path,
             use_gpu=use_gpu,
             code_path=code_path,
+            resume=resume,
             metrics_path=(Path(output_path) / "metrics.jsonl") if output_path else None,
         )
         return 0
@@ -85,6 +88,8 @@ def ray_train(
     if code_path:
         worker_cmd += ["--code", str(code_path)]
     worker_cmd += ["--gpu-id", str(use_gpu)]
+    if resume:
+        worker_cmd += ["--resume"]
     if overrides:
         worker_cmd += ["--overrides-json", json.dumps(overrides)]
     master_addr, master_port = "127.0.0.1", None

@@ -117,7 +117,16 @@ class TaggerPipe(TrainablePipe):
         super().load_cfg(cfg, device)
         self.labels = list(cfg.get("labels", []))
         self.label2id = {t: i for i, t in enumerate(self.labels)}
-        self.module = TaggerHead(self.width, max(1, len(self.labels))).to(device)
+        n = max(1, len(self.labels))
+        if self.module is None:
+            # keep an existing module: its params may already be views into
+            # the distributed engine's flat buffer (resume path)
+            self.module = TaggerHead(self.width, n).to(device)
+        elif self.module.output.out_features != n:
+            raise ValueError(
+                f"{self.name}: checkpoint has {n} labels but the initialized "
+                f"module has {self.module.output.out_features}"
+            )
 
     def _gold_ids(self, examples, n_tokens: int) -> np.ndarray:
         ids = []

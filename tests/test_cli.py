@@ -54,6 +54,19 @@ def test_launcher_aborts_all_on_rank_failure(tmp_path):
     assert r.returncode != 0  # supervisor propagates the failure
 
 
+def test_resume_continues_from_checkpoint(tmp_path):
+    out = tmp_path / "outr"
+    r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
+                  "--training.max_steps", "6", "--training.eval_frequency", "3"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    before = (out / "model-last" / "tagger" / "model.safetensors").read_bytes()
+    r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out), "--resume",
+                  "--training.max_steps", "3", "--training.eval_frequency", "2"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    after = (out / "model-last" / "tagger" / "model.safetensors").read_bytes()
+    assert before != after  # training continued and re-saved
+
+
 def test_overrides_reach_workers(tmp_path):
     out = tmp_path / "out3"
     r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
