@@ -37,11 +37,15 @@ class MultiHashEmbed(nn.Module):
         self.mixer = Maxout(width * len(attrs), width, pieces=3, normalize=True)
 
     def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
-        outs = []
-        for i, table in enumerate(self.tables):
-            outs.append(ops.hashembed(table, batch.attr_ids[:, i], self.seeds[i]))
-        X = torch.cat(outs, dim=1)
-        Y = self.mixer(X)
+        from spacy_ray_amd.utils import timing
+
+        with timing.phase("t2v/embed_tables"):
+            outs = []
+            for i, table in enumerate(self.tables):
+                outs.append(ops.hashembed(table, batch.attr_ids[:, i], self.seeds[i]))
+            X = torch.cat(outs, dim=1)
+        with timing.phase("t2v/mixer"):
+            Y = self.mixer(X)
         if drop and self.training:
             Y = torch.nn.functional.dropout(Y, drop)
         return Y
@@ -59,14 +63,17 @@ class MaxoutWindowEncoder(nn.Module):
         )
 
     def forward(self, X: torch.Tensor, lengths: torch.Tensor, drop: float = 0.0) -> torch.Tensor:
+        from spacy_ray_amd.utils import timing
+
         # doc-boundary masks computed once for all depth layers (and reused
         # by every seq2col forward AND backward — no per-call index work)
         starts, ends = ops.boundary_masks_u8(lengths, X.shape[0])
-        for block in self.blocks:
-            Y = block(ops.seq2col(X, lengths, starts, ends))
-            if drop and self.training:
-                Y = torch.nn.functional.dropout(Y, drop)
-            X = X + Y  # residual (thinc `residual(...)` wrapper)
+        for bi, block in enumerate(self.blocks):
+            with timing.phase(f"t2v/encode{bi}"):
+                Y = block(ops.seq2col(X, lengths, starts, ends))
+                if drop and self.training:
+                    Y = torch.nn.functional.dropout(Y, drop)
+                X = X + Y  # residual (thinc `residual(...)` wrapper)
         return X
 
 

@@ -116,28 +116,30 @@ struct ArcEagerBatch {
 
   // 13 context tokens per state, as batch-flat indices (-1 = missing):
   // [S0,S1,S2, B0,B1,B2, L1(S0),L2(S0), R1(S0),R2(S0), L1(S1), R1(S1), head(S0)]
+  void fill_features(size_t i, int32_t* out) const {
+    const ParserState& st = states[i];
+    int32_t off = offsets[i];
+    int32_t f[13];
+    int32_t s0 = st.s0(), s1 = st.s1(), s2 = st.s2();
+    f[0] = s0; f[1] = s1; f[2] = s2;
+    f[3] = st.buf < st.len ? st.buf : -1;
+    f[4] = st.buf + 1 < st.len ? st.buf + 1 : -1;
+    f[5] = st.buf + 2 < st.len ? st.buf + 2 : -1;
+    f[6] = s0 >= 0 ? st.l1[s0] : -1;
+    f[7] = s0 >= 0 ? st.l2[s0] : -1;
+    f[8] = s0 >= 0 ? st.r1[s0] : -1;
+    f[9] = s0 >= 0 ? st.r2[s0] : -1;
+    f[10] = s1 >= 0 ? st.l1[s1] : -1;
+    f[11] = s1 >= 0 ? st.r1[s1] : -1;
+    f[12] = s0 >= 0 ? st.head[s0] : -1;
+    for (int k = 0; k < 13; k++) out[k] = f[k] >= 0 ? off + f[k] : -1;
+  }
+
   py::array_t<int32_t> features() const {
     py::ssize_t S = (py::ssize_t)states.size();
     py::array_t<int32_t> out({S, (py::ssize_t)13});
     auto r = out.mutable_unchecked<2>();
-    for (py::ssize_t i = 0; i < S; i++) {
-      const ParserState& st = states[i];
-      int32_t off = offsets[i];
-      int32_t f[13];
-      int32_t s0 = st.s0(), s1 = st.s1(), s2 = st.s2();
-      f[0] = s0; f[1] = s1; f[2] = s2;
-      f[3] = st.buf < st.len ? st.buf : -1;
-      f[4] = st.buf + 1 < st.len ? st.buf + 1 : -1;
-      f[5] = st.buf + 2 < st.len ? st.buf + 2 : -1;
-      f[6] = s0 >= 0 ? st.l1[s0] : -1;
-      f[7] = s0 >= 0 ? st.l2[s0] : -1;
-      f[8] = s0 >= 0 ? st.r1[s0] : -1;
-      f[9] = s0 >= 0 ? st.r2[s0] : -1;
-      f[10] = s1 >= 0 ? st.l1[s1] : -1;
-      f[11] = s1 >= 0 ? st.r1[s1] : -1;
-      f[12] = s0 >= 0 ? st.head[s0] : -1;
-      for (int k = 0; k < 13; k++) r(i, k) = f[k] >= 0 ? off + f[k] : -1;
-    }
+    for (py::ssize_t i = 0; i < S; i++) fill_features((size_t)i, r.mutable_data(i, 0));
     return out;
   }
 
@@ -168,6 +170,50 @@ struct ArcEagerBatch {
     return out;
   }
 
+  // ---- shared cost computation for one state (Goldberg&Nivre dynamic
+  // oracle); writes n_actions() floats, invalid actions get KInvalid.
+  void fill_costs(size_t i, uint8_t* v, float* r) const {
+    const ParserState& st = states[i];
+    const auto& gh = gold_head[i];
+    const auto& gl = gold_label[i];
+    fill_valid(v, st);
+    int32_t b = st.buf < st.len ? st.buf : -1;
+    int32_t s0 = st.s0();
+    auto in_stack = [&](int32_t t) {
+      for (int32_t s : st.stack)
+        if (s == t) return true;
+      return false;
+    };
+    float c_shift = 0, c_reduce = 0, c_la = 0, c_ra = 0;
+    if (b >= 0) {
+      if (gh[b] >= 0 && in_stack(gh[b])) c_shift += 1;
+      for (int32_t s : st.stack)
+        if (st.head[s] == -1 && gh[s] == b) c_shift += 1;
+    }
+    if (s0 >= 0) {
+      for (int32_t d = st.buf; d < st.len; d++)
+        if (gh[d] == s0) c_reduce += 1;
+      if (b >= 0) {
+        c_la = c_reduce;
+        if (gh[s0] >= 0 && gh[s0] > b) c_la += 1;
+        if (gh[b] >= 0 && gh[b] != s0 && (in_stack(gh[b]) || gh[b] > b)) c_ra += 1;
+        for (int32_t s : st.stack)
+          if (st.head[s] == -1 && gh[s] == b) c_ra += 1;
+      }
+    }
+    r[0] = v[0] ? c_shift : KInvalid;
+    r[1] = v[1] ? c_reduce : KInvalid;
+    for (int32_t l = 0; l < n_labels; l++) {
+      float la = c_la, ra = c_ra;
+      if (b >= 0 && s0 >= 0) {
+        if (gh[s0] == b && gl[s0] != l) la += 1;
+        if (gh[b] == s0 && gl[b] != l) ra += 1;
+      }
+      r[2 + l] = v[2 + l] ? la : KInvalid;
+      r[2 + n_labels + l] = v[2 + n_labels + l] ? ra : KInvalid;
+    }
+  }
+
   // Goldberg&Nivre dynamic-oracle costs; invalid actions get KInvalid.
   py::array_t<float> costs() const {
     if (!has_gold) throw std::runtime_error("costs() requires set_gold()");
@@ -176,60 +222,44 @@ struct ArcEagerBatch {
     auto r = out.mutable_unchecked<2>();
     std::vector<uint8_t> v((size_t)A);
     for (py::ssize_t i = 0; i < S; i++) {
-      const ParserState& st = states[i];
-      const auto& gh = gold_head[i];
-      const auto& gl = gold_label[i];
-      fill_valid(v.data(), st);
-      int32_t b = st.buf < st.len ? st.buf : -1;
-      int32_t s0 = st.s0();
-      // membership helpers
-      auto in_stack = [&](int32_t t) {
-        for (int32_t s : st.stack) if (s == t) return true;
-        return false;
-      };
-      // SHIFT cost: gold head of b in stack (recoverable only via RA now —
-      // i.e. only if gold head == s0, which SHIFT forgoes), plus headless
-      // stack items whose gold head is b.
-      float c_shift = 0, c_reduce = 0, c_la = 0, c_ra = 0;
-      if (b >= 0) {
-        if (gh[b] >= 0 && in_stack(gh[b])) c_shift += 1;
-        for (int32_t s : st.stack)
-          if (st.head[s] == -1 && gh[s] == b) c_shift += 1;
-      }
-      if (s0 >= 0) {
-        // REDUCE: lose gold dependents of s0 in buffer.
-        for (int32_t d = st.buf; d < st.len; d++)
-          if (gh[d] == s0) c_reduce += 1;
-        if (b >= 0) {
-          // LEFT-ARC(b -> s0): lose gold deps of s0 in buffer, and gold head
-          // of s0 if it is in buffer beyond b (head==b is the gold arc; a
-          // gold head already in the stack was lost earlier — additional
-          // cost only).
-          for (int32_t d = st.buf; d < st.len; d++)
-            if (gh[d] == s0) c_la += 1;
-          if (gh[s0] >= 0 && gh[s0] > b) c_la += 1;
-          // RIGHT-ARC(s0 -> b): lose gold head of b elsewhere (stack != s0 or
-          // buffer), and ALL headless stack items (s0 included) whose gold
-          // head is b — after RA, b is no longer in the buffer so none of
-          // them can take b as head via LEFT-ARC.
-          if (gh[b] >= 0 && gh[b] != s0 && (in_stack(gh[b]) || gh[b] > b)) c_ra += 1;
-          for (int32_t s : st.stack)
-            if (st.head[s] == -1 && gh[s] == b) c_ra += 1;
-        }
-      }
-      r(i, 0) = v[0] ? c_shift : KInvalid;
-      r(i, 1) = v[1] ? c_reduce : KInvalid;
-      for (int32_t l = 0; l < n_labels; l++) {
-        float la = c_la, ra = c_ra;
-        if (b >= 0 && s0 >= 0) {
-          if (gh[s0] == b && gl[s0] != l) la += 1;  // right arc, wrong label
-          if (gh[b] == s0 && gl[b] != l) ra += 1;
-        }
-        r(i, 2 + l) = v[2 + l] ? la : KInvalid;
-        r(i, 2 + n_labels + l) = v[2 + n_labels + l] ? ra : KInvalid;
-      }
+      fill_costs((size_t)i, v.data(), r.mutable_data(i, 0));
     }
     return out;
+  }
+
+  // ---- fused per-step call: ONE crossing of the pybind boundary returns
+  // (active_idx, features, valid, is_gold) for the active states only.
+  // is_gold[s,a] = (cost <= min valid cost + eps); empty when !with_gold.
+  py::tuple step_arrays(bool with_gold) const {
+    std::vector<int32_t> idx;
+    idx.reserve(states.size());
+    for (size_t i = 0; i < states.size(); i++)
+      if (!states[i].final_state()) idx.push_back((int32_t)i);
+    py::ssize_t Sa = (py::ssize_t)idx.size();
+    const py::ssize_t A = n_actions();
+    py::array_t<int32_t> act(Sa);
+    py::array_t<int32_t> feats({Sa, (py::ssize_t)13});
+    py::array_t<uint8_t> valid_a({Sa, A});
+    py::array_t<uint8_t> gold_a({with_gold ? Sa : 0, A});
+    std::copy(idx.begin(), idx.end(), act.mutable_data());
+    std::vector<float> crow((size_t)A);
+    for (py::ssize_t k = 0; k < Sa; k++) {
+      size_t i = (size_t)idx[(size_t)k];
+      fill_features(i, feats.mutable_data(k, 0));
+      uint8_t* v = valid_a.mutable_data(k, 0);
+      if (with_gold) {
+        fill_costs(i, v, crow.data());
+        float cmin = KInvalid;
+        for (py::ssize_t a = 0; a < A; a++)
+          if (v[a] && crow[(size_t)a] < cmin) cmin = crow[(size_t)a];
+        uint8_t* g = gold_a.mutable_data(k, 0);
+        for (py::ssize_t a = 0; a < A; a++)
+          g[a] = (v[a] && crow[(size_t)a] <= cmin + 1e-6f) ? 1 : 0;
+      } else {
+        fill_valid(v, states[i]);
+      }
+    }
+    return py::make_tuple(act, feats, valid_a, gold_a);
   }
 
   void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
@@ -345,19 +375,50 @@ struct BiluoBatch {
   }
 
   // 6 context tokens: [i-2, i-1, i, i+1, i+2, open_start]  (batch-flat, -1 pad)
+  void fill_features(size_t s, int32_t* out) const {
+    const NerState& st = states[s];
+    int32_t off = offsets[s];
+    int32_t f[6] = {st.i - 2, st.i - 1, st.i, st.i + 1, st.i + 2, st.open_start};
+    for (int k = 0; k < 6; k++)
+      out[k] = (f[k] >= 0 && f[k] < st.len) ? off + f[k] : -1;
+  }
+
   py::array_t<int32_t> features() const {
     py::ssize_t S = (py::ssize_t)states.size();
     py::array_t<int32_t> out({S, (py::ssize_t)6});
     auto r = out.mutable_unchecked<2>();
-    for (py::ssize_t s = 0; s < S; s++) {
-      const NerState& st = states[s];
-      int32_t off = offsets[s];
-      int32_t f[6] = {st.i - 2, st.i - 1, st.i, st.i + 1, st.i + 2, st.open_start};
-      for (int k = 0; k < 6; k++) {
-        r(s, k) = (f[k] >= 0 && f[k] < st.len) ? off + f[k] : -1;
+    for (py::ssize_t s = 0; s < S; s++) fill_features((size_t)s, r.mutable_data(s, 0));
+    return out;
+  }
+
+  // fused per-step call (same contract as ArcEagerBatch::step_arrays)
+  py::tuple step_arrays(bool with_gold) const {
+    std::vector<int32_t> idx;
+    idx.reserve(states.size());
+    for (size_t i = 0; i < states.size(); i++)
+      if (!states[i].final_state()) idx.push_back((int32_t)i);
+    py::ssize_t Sa = (py::ssize_t)idx.size();
+    const py::ssize_t A = n_actions();
+    py::array_t<int32_t> act(Sa);
+    py::array_t<int32_t> feats({Sa, (py::ssize_t)6});
+    py::array_t<uint8_t> valid_a({Sa, A});
+    py::array_t<uint8_t> gold_a({with_gold ? Sa : 0, A});
+    std::copy(idx.begin(), idx.end(), act.mutable_data());
+    for (py::ssize_t k = 0; k < Sa; k++) {
+      size_t i = (size_t)idx[(size_t)k];
+      const NerState& st = states[i];
+      fill_features(i, feats.mutable_data(k, 0));
+      uint8_t* v = valid_a.mutable_data(k, 0);
+      fill_valid(v, st);
+      if (with_gold) {
+        uint8_t* g = gold_a.mutable_data(k, 0);
+        int32_t gcode = st.final_state() ? -1 : gold[i][st.i];
+        bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
+        for (py::ssize_t a = 0; a < A; a++)
+          g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
       }
     }
-    return out;
+    return py::make_tuple(act, feats, valid_a, gold_a);
   }
 
   void fill_valid(uint8_t* v, const NerState& st) const {
@@ -450,6 +511,7 @@ void init_transitions(py::module_& m) {
       .def("valid", &ArcEagerBatch::valid)
       .def("costs", &ArcEagerBatch::costs)
       .def("advance", &ArcEagerBatch::advance)
+      .def("step_arrays", &ArcEagerBatch::step_arrays, py::arg("with_gold"))
       .def("heads", &ArcEagerBatch::heads)
       .def("labels", &ArcEagerBatch::labels);
 
@@ -464,5 +526,6 @@ void init_transitions(py::module_& m) {
       .def("valid", &BiluoBatch::valid)
       .def("costs", &BiluoBatch::costs)
       .def("advance", &BiluoBatch::advance)
+      .def("step_arrays", &BiluoBatch::step_arrays, py::arg("with_gold"))
       .def("tags", &BiluoBatch::tags);
 }

@@ -234,15 +234,14 @@ class _TransitionPipeBase(TrainablePipe):
         valid_chunks: List[torch.Tensor] = []
         n_states_total = 0
         max_steps = 4 * T + 16
+        n_states = len(states)
         for _ in range(max_steps):
             with timing.span("raw/states_cpu"):
-                final = states.is_final()
-                if final.all():
+                # ONE native call: active indices + features + valid +
+                # min-cost (is_gold) masks for the active states
+                act_idx, feats, valid_np, gold_np = states.step_arrays(train)
+                if len(act_idx) == 0:
                     break
-                active = final == 0
-                act_idx = np.nonzero(active)[0]
-                feats = states.features()[act_idx]  # compacted: active only
-                valid_np = states.valid()[act_idx]
             with timing.span("raw/score_fwd"):
                 feats_t = torch.from_numpy(
                     np.where(feats < 0, T, feats).astype(np.int64)
@@ -256,12 +255,8 @@ class _TransitionPipeBase(TrainablePipe):
                     scores = self.module.score(pre_d, feats_t)
                 valid_t = torch.from_numpy(valid_np).to(device)
             if train:
-                with timing.span("raw/oracle_cpu"):
-                    costs = states.costs()[act_idx]
-                    cmin = costs.min(axis=1, keepdims=True)
-                    is_gold = ((costs <= cmin + 1e-6) & (valid_np > 0)).astype(np.uint8)
                 with timing.span("raw/loss_build"):
-                    gold_t = torch.from_numpy(is_gold).to(device)
+                    gold_t = torch.from_numpy(gold_np).to(device)
                     score_chunks.append(scores)
                     gold_chunks.append(gold_t)
                     valid_chunks.append(valid_t)
@@ -277,14 +272,14 @@ class _TransitionPipeBase(TrainablePipe):
                     )
                 else:
                     s_np = scores.detach().float().cpu().numpy()
-                    g_np = is_gold if train else valid_np
+                    g_np = gold_np if train else valid_np
                     choose = np.where(g_np > 0, s_np, NEG_INF)
                     fallback = np.where(valid_np > 0, s_np, NEG_INF)
                     any_gold = (g_np > 0).any(axis=1, keepdims=True)
                     choose = np.where(any_gold, choose, fallback)
                     actions_a = choose.argmax(axis=1).astype(np.int32)
             with timing.span("raw/advance_cpu"):
-                actions = np.full(len(active), -1, dtype=np.int32)
+                actions = np.full(n_states, -1, dtype=np.int32)
                 actions[act_idx] = actions_a
                 states.advance(actions)
         if not train:
