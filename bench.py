@@ -141,11 +141,21 @@ def main() -> None:
         for k in sorted(tt):
             print(f"# {k}: {tt[k]:.0f} ms total, {cc[k]} calls", flush=True)
 
+    cfg_base = os.path.basename(args.config)
+    model_desc = {
+        "en_core_cnn.cfg": ("words/sec (whole node) en_core CNN tagger+parser+NER train",
+                            "en_core_web_cnn (MultiHashEmbed+MaxoutWindowEncoder w96d4 + tagger + parser + ner)"),
+        "en_core_trf.cfg": ("words/sec (whole node) en_core_web_trf (roberta-base) train",
+                            "en_core_web_trf (roberta-base random-init + tagger + parser + ner)"),
+        "xx_multilingual.cfg": ("words/sec (whole node) xx multilingual UD train",
+                                "xx multilingual UD (MultiHashEmbed w128d4 + tagger + parser, 8 treebanks)"),
+    }.get(cfg_base, ("words/sec (whole node) " + cfg_base, cfg_base))
+
     if rank == 0:
         total_words = words_per_step * world * args.steps
         value = total_words / elapsed
         out = {
-            "metric": "words/sec (whole node) en_core CNN tagger+parser+NER train",
+            "metric": model_desc[0],
             "value": value,
             "unit": "words/s",
             "n_gpus": world if use_cuda else 0,
@@ -158,7 +168,7 @@ def main() -> None:
             "dtype": "bf16" if use_cuda else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "en_core_web_cnn (MultiHashEmbed+MaxoutWindowEncoder w96d4 + tagger + parser + ner)",
+                "model": model_desc[1],
                 "global_batch": words_per_step * world,
                 "seq_len": args.words_per_doc,
                 "parallelism": f"dp{world}",

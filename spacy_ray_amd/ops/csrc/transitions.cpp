@@ -73,10 +73,10 @@ struct ArcEagerBatch {
   bool has_gold = false;
 
   ArcEagerBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
-                int32_t n_labels_)
+                int32_t n_labels_, int32_t base_offset = 0)
       : n_labels(n_labels_) {
     auto L = lengths.unchecked<1>();
-    int32_t off = 0;
+    int32_t off = base_offset;
     states.resize(L.shape(0));
     offsets.resize(L.shape(0));
     for (py::ssize_t i = 0; i < L.shape(0); i++) {
@@ -340,12 +340,12 @@ struct BiluoBatch {
   bool has_gold = false;
 
   BiluoBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
-             int32_t n_types_)
+             int32_t n_types_, int32_t base_offset = 0)
       : n_types(n_types_) {
     auto L = lengths.unchecked<1>();
     states.resize(L.shape(0));
     offsets.resize(L.shape(0));
-    int32_t off = 0;
+    int32_t off = base_offset;
     for (py::ssize_t i = 0; i < L.shape(0); i++) {
       states[i].init(L(i));
       offsets[i] = off;
@@ -501,8 +501,8 @@ struct BiluoBatch {
 
 void init_transitions(py::module_& m) {
   py::class_<ArcEagerBatch>(m, "ArcEagerBatch")
-      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t>(),
-           py::arg("lengths"), py::arg("n_labels"))
+      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t, int32_t>(),
+           py::arg("lengths"), py::arg("n_labels"), py::arg("base_offset") = 0)
       .def("set_gold", &ArcEagerBatch::set_gold, py::arg("heads"), py::arg("labels"))
       .def_property_readonly("n_actions", &ArcEagerBatch::n_actions)
       .def("__len__", &ArcEagerBatch::size)
@@ -516,8 +516,8 @@ void init_transitions(py::module_& m) {
       .def("labels", &ArcEagerBatch::labels);
 
   py::class_<BiluoBatch>(m, "BiluoBatch")
-      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t>(),
-           py::arg("lengths"), py::arg("n_types"))
+      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t, int32_t>(),
+           py::arg("lengths"), py::arg("n_types"), py::arg("base_offset") = 0)
       .def("set_gold", &BiluoBatch::set_gold, py::arg("codes"))
       .def_property_readonly("n_actions", &BiluoBatch::n_actions)
       .def("__len__", &BiluoBatch::size)

@@ -195,7 +195,7 @@ class _TransitionPipeBase(TrainablePipe):
     def _n_actions(self) -> int:
         raise NotImplementedError
 
-    def _make_states(self, lengths: np.ndarray):
+    def _make_states(self, lengths: np.ndarray, base: int = 0):
         raise NotImplementedError
 
     def _set_gold(self, states, examples) -> None:
@@ -205,17 +205,40 @@ class _TransitionPipeBase(TrainablePipe):
         raise NotImplementedError
 
     # ---- shared machinery
-    def _step_loop(self, states, t2v, train: bool):
-        """Greedy transition loop.
+    def _split_docs(self, lengths: np.ndarray, n_shards: int):
+        """Split docs into ~word-balanced contiguous shards:
+        [(doc_lo, doc_hi, token_base), ...]."""
+        if n_shards <= 1 or len(lengths) < 2 * n_shards:
+            return [(0, len(lengths), 0)]
+        csum = np.concatenate([[0], np.cumsum(lengths)])
+        total = csum[-1]
+        out = []
+        lo = 0
+        for k in range(n_shards):
+            target = total * (k + 1) / n_shards
+            hi = int(np.searchsorted(csum, target)) if k < n_shards - 1 else len(lengths)
+            hi = max(hi, lo + 1)
+            out.append((lo, hi, int(csum[lo])))
+            lo = hi
+            if lo >= len(lengths):
+                break
+        return out
 
-        Training uses a TWO-PHASE backward (the per-step [T+1,nF,HP] gradient
-        tensors autograd would otherwise allocate+sum dominated host time):
-        phase 1 — the per-step losses (upper GEMM + bias path) are
-        backpropagated HERE, while each step's dPre scatters into ONE
-        persistent fp32 buffer (ops.parser_step_score_accum); phase 2 — the
-        returned loss is a surrogate (pre · dPre).sum() whose backward hands
-        exactly dPre to the precompute GEMM inside the caller's single main
-        backward, flowing on to lower_W / pad / tok2vec."""
+    def _step_loop(self, shards, t2v, train: bool):
+        """Greedy transition loop over double-buffered state shards.
+
+        Two levels of overlap engineering here:
+        1. TWO-PHASE backward: the per-step losses (upper GEMM + bias path)
+           are backpropagated HERE while each step's dPre scatters into ONE
+           persistent fp32 buffer (ops.parser_step_score_accum); the
+           returned loss is a surrogate (pre · dPre).sum() whose backward
+           hands exactly dPre to the precompute GEMM inside the caller's
+           single main backward (lower_W / pad / tok2vec).
+        2. SHARD PIPELINING: states are split into word-balanced shards;
+           while one shard's scores travel GPU->CPU and its transitions
+           advance in C++, the other shard's scoring kernels run — the
+           CPU-advance / GPU-score double-buffering of SURVEY.md §2.2 N8.
+        """
         from spacy_ray_amd.ops import api as _ops
         from spacy_ray_amd.utils import timing
 
@@ -234,14 +257,12 @@ class _TransitionPipeBase(TrainablePipe):
         valid_chunks: List[torch.Tensor] = []
         n_states_total = 0
         max_steps = 4 * T + 16
-        n_states = len(states)
-        for _ in range(max_steps):
+
+        def launch(states):
             with timing.span("raw/states_cpu"):
-                # ONE native call: active indices + features + valid +
-                # min-cost (is_gold) masks for the active states
                 act_idx, feats, valid_np, gold_np = states.step_arrays(train)
-                if len(act_idx) == 0:
-                    break
+            if len(act_idx) == 0:
+                return None
             with timing.span("raw/score_fwd"):
                 feats_t = torch.from_numpy(
                     np.where(feats < 0, T, feats).astype(np.int64)
@@ -260,28 +281,56 @@ class _TransitionPipeBase(TrainablePipe):
                     score_chunks.append(scores)
                     gold_chunks.append(gold_t)
                     valid_chunks.append(valid_t)
-                    n_states_total += len(act_idx)
                 sel_gold, sel_valid = gold_t, valid_t
             else:
                 sel_gold, sel_valid = valid_t, valid_t
+            if hip is not None:
+                actions_dev = hip.action_select(scores.detach(), sel_gold, sel_valid)
+                return act_idx, actions_dev, None
+            s_np = scores.detach().float().cpu().numpy()
+            g_np = gold_np if train else valid_np
+            choose = np.where(g_np > 0, s_np, NEG_INF)
+            fallback = np.where(valid_np > 0, s_np, NEG_INF)
+            any_gold = (g_np > 0).any(axis=1, keepdims=True)
+            choose = np.where(any_gold, choose, fallback)
+            return act_idx, None, choose.argmax(axis=1).astype(np.int32)
+
+        def complete(states, pend):
+            act_idx, actions_dev, actions_np = pend
             with timing.span("raw/score_d2h"):
-                if hip is not None:
-                    actions_a = (
-                        hip.action_select(scores.detach(), sel_gold, sel_valid)
-                        .cpu().numpy().astype(np.int32)
-                    )
-                else:
-                    s_np = scores.detach().float().cpu().numpy()
-                    g_np = gold_np if train else valid_np
-                    choose = np.where(g_np > 0, s_np, NEG_INF)
-                    fallback = np.where(valid_np > 0, s_np, NEG_INF)
-                    any_gold = (g_np > 0).any(axis=1, keepdims=True)
-                    choose = np.where(any_gold, choose, fallback)
-                    actions_a = choose.argmax(axis=1).astype(np.int32)
+                if actions_dev is not None:
+                    actions_np = actions_dev.cpu().numpy().astype(np.int32)
             with timing.span("raw/advance_cpu"):
-                actions = np.full(n_states, -1, dtype=np.int32)
-                actions[act_idx] = actions_a
+                actions = np.full(len(states), -1, dtype=np.int32)
+                actions[act_idx] = actions_np
                 states.advance(actions)
+            if train:
+                n = len(act_idx)
+                nonlocal n_states_total
+                n_states_total += n
+
+        pend = [None] * len(shards)
+        done = [False] * len(shards)
+        for _ in range(max_steps):
+            progressed = False
+            for k, states in enumerate(shards):
+                if done[k]:
+                    continue
+                if pend[k] is not None:
+                    complete(states, pend[k])
+                    pend[k] = None
+                out = launch(states)
+                if out is None:
+                    done[k] = True
+                    continue
+                pend[k] = out
+                progressed = True
+            if not progressed and all(p is None for p in pend):
+                break
+        for k, states in enumerate(shards):  # drain
+            if pend[k] is not None:
+                complete(states, pend[k])
+                pend[k] = None
         if not train:
             return None, 0.0
         if not score_chunks:
@@ -306,18 +355,26 @@ class _TransitionPipeBase(TrainablePipe):
         surrogate = (pre.float() * dPre32).sum()
         return surrogate, display
 
+    def _n_shards(self, t2v) -> int:
+        return 2 if t2v.is_cuda else 1
+
     def get_loss(self, examples, t2v, batch):
         lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
-        states = self._make_states(lengths)
-        self._set_gold(states, examples)
-        return self._step_loop(states, t2v, train=True)
+        shards = []
+        for lo, hi, base in self._split_docs(lengths, self._n_shards(t2v)):
+            states = self._make_states(lengths[lo:hi], base)
+            self._set_gold(states, examples)  # global flat gold; offsets select
+            shards.append(states)
+        return self._step_loop(shards, t2v, train=True)
 
     def predict_and_set(self, docs, t2v, batch) -> None:
         lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
-        states = self._make_states(lengths)
+        splits = self._split_docs(lengths, self._n_shards(t2v))
+        shards = [self._make_states(lengths[lo:hi], base) for lo, hi, base in splits]
         with torch.no_grad():
-            self._step_loop(states, t2v, train=False)
-        self._annotate(docs, states)
+            self._step_loop(shards, t2v, train=False)
+        for (lo, hi, base), states in zip(splits, shards):
+            self._annotate(docs[lo:hi], states)
 
 
 class ParserPipe(_TransitionPipeBase):
@@ -337,8 +394,8 @@ class ParserPipe(_TransitionPipeBase):
     def _n_actions(self) -> int:
         return 2 + 2 * len(self.labels)
 
-    def _make_states(self, lengths):
-        return _srx_cpu.ArcEagerBatch(lengths, len(self.labels))
+    def _make_states(self, lengths, base: int = 0):
+        return _srx_cpu.ArcEagerBatch(lengths, len(self.labels), base)
 
     def _set_gold(self, states, examples) -> None:
         heads = np.concatenate([
@@ -388,8 +445,8 @@ class NerPipe(_TransitionPipeBase):
     def _n_actions(self) -> int:
         return 1 + 4 * len(self.labels)
 
-    def _make_states(self, lengths):
-        return _srx_cpu.BiluoBatch(lengths, len(self.labels))
+    def _make_states(self, lengths, base: int = 0):
+        return _srx_cpu.BiluoBatch(lengths, len(self.labels), base)
 
     def _set_gold(self, states, examples) -> None:
         codes = np.concatenate([
