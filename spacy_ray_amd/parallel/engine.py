@@ -169,19 +169,32 @@ class ZeRO1Engine:
     # ------------------------------------------------------------- stepper
     def accumulate(self, examples, drop: float = 0.0, losses: Optional[Dict] = None,
                    sync: bool = True) -> None:
+        import time as _time
+
         from spacy_ray_amd.utils import timing
 
+        t0 = _time.perf_counter()
         self._sync = sync
         total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop)
         with timing.phase("bwd/main"):
             total.backward()
         self._sync = False
+        self.last_compute_ms = (_time.perf_counter() - t0) * 1000
+
+    last_compute_ms: float = 0.0
+    last_comm_ms: float = 0.0
 
     def apply_step(self) -> None:
+        import time as _time
+
         from spacy_ray_amd.utils import timing
 
+        t0 = _time.perf_counter()
         with timing.phase("comm+opt/apply_step"):
             self._apply_step_inner()
+        # wall time of collective wait + sharded Adam + republish (approx:
+        # async GPU work not synced unless SRX_TIMING=1)
+        self.last_comm_ms = (_time.perf_counter() - t0) * 1000
 
     def _apply_step_inner(self) -> None:
         # launch any bucket the hooks didn't (grad-less params, overlap off)

@@ -175,6 +175,8 @@ def distributed_train(
                     "other_scores": info["other_scores"],
                     "words": info["words_scaled"], "words_cum": words_cum,
                     "wps": info["wps"], "time": now - t_start,
+                    "compute_ms": round(getattr(engine, "last_compute_ms", 0.0), 2),
+                    "comm_opt_ms": round(getattr(engine, "last_comm_ms", 0.0), 2),
                 }) + "\n")
                 metrics_fh.flush()
             if is_best_checkpoint is not None:
