@@ -356,7 +356,13 @@ class _TransitionPipeBase(TrainablePipe):
         return surrogate, display
 
     def _n_shards(self, t2v) -> int:
-        return 2 if t2v.is_cuda else 1
+        # Measured on MI355X: 2-way shard pipelining LOST ~25% words/s on
+        # 20-word-doc batches — each shard doubles the per-step python/launch
+        # overhead, which exceeds the hidden sync latency.  Default 1; the
+        # machinery stays for long-doc workloads (SRX_PARSER_SHARDS to tune).
+        import os
+
+        return int(os.environ.get("SRX_PARSER_SHARDS", "1")) if t2v.is_cuda else 1
 
     def get_loss(self, examples, t2v, batch):
         lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
