@@ -73,17 +73,33 @@ class Language:
         with timing.phase("data/batch_build"):
             docs = [eg.predicted for eg in examples]
             batch = TokenBatch(docs, self.device)
+        from .pipes import _TransitionPipeBase, run_transition_tasks
+
         t2v_pipe = self.tok2vec
         with timing.phase("fwd/tok2vec"):
             t2v = t2v_pipe.forward(batch, drop=drop) if t2v_pipe is not None else None
         total = None
+        # transition pipes (parser/NER) run INTERLEAVED: their per-step
+        # GPU-score / CPU-advance phases pipeline against each other
+        trans_tasks = []
         for name, pipe in self.pipeline:
             if isinstance(pipe, Tok2VecPipe) or name in self._frozen:
+                continue
+            if isinstance(pipe, _TransitionPipeBase):
+                trans_tasks.append((name, pipe, pipe.make_loss_task(examples, t2v)))
                 continue
             with timing.phase(f"loss/{name}"):
                 loss, display = pipe.get_loss(examples, t2v, batch)
             losses[name] = losses.get(name, 0.0) + display
             total = loss if total is None else total + loss
+        if trans_tasks:
+            with timing.phase("loss/transition_steps"):
+                run_transition_tasks([t for _, _, t in trans_tasks])
+            for name, pipe, task in trans_tasks:
+                with timing.phase(f"loss/{name}"):
+                    loss, display = pipe.finish_task(task)
+                losses[name] = losses.get(name, 0.0) + display
+                total = loss if total is None else total + loss
         if total is None:
             total = torch.zeros((), device=self.device)
         return total, losses
@@ -99,16 +115,28 @@ class Language:
 
     # ------------------------------------------------------------ inference
     def predict_docs(self, docs: Sequence[Doc]) -> Sequence[Doc]:
+        from .pipes import _TransitionPipeBase, run_transition_tasks
+
         if not docs:
             return docs
         with torch.no_grad():
             batch = TokenBatch(docs, self.device)
             t2v_pipe = self.tok2vec
             t2v = t2v_pipe.forward(batch) if t2v_pipe is not None else None
+            # interleave the transition pipes' decode loops
+            trans = []
+            for name, pipe in self.pipeline:
+                if isinstance(pipe, _TransitionPipeBase):
+                    trans.append((pipe,) + pipe.make_predict_task(docs, t2v))
+            if trans:
+                run_transition_tasks([t[1] for t in trans])
         for name, pipe in self.pipeline:
-            if isinstance(pipe, Tok2VecPipe):
+            if isinstance(pipe, (Tok2VecPipe, _TransitionPipeBase)):
                 continue
             pipe.predict_and_set(docs, t2v, batch)
+        for pipe, task, splits, shards in trans:
+            for (lo, hi, base), states in zip(splits, shards):
+                pipe._annotate(docs[lo:hi], states)
         return docs
 
     def evaluate(self, examples: Sequence[Example], batch_size: int = 256) -> Dict[str, float]:

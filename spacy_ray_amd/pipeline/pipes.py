@@ -157,6 +157,135 @@ class TaggerPipe(TrainablePipe):
             off += n
 
 
+class _TransitionTask:
+    """One pipe's transition-loop context: states shards, the detached
+    precompute tensor, the persistent dPre gradient accumulator and the
+    per-step loss collections.
+
+    Two levels of overlap engineering (SURVEY.md §2.2 N8 disposition):
+    1. TWO-PHASE backward: per-step losses (upper GEMM + bias) are
+       backpropagated in finish_task while each step's dPre scatters into
+       ONE persistent fp32 buffer (ops.parser_step_score_accum); the loss
+       returned upward is a surrogate (pre · dPre).sum() that hands exactly
+       dPre to the precompute GEMM in the caller's single main backward.
+    2. TASK/SHARD PIPELINING (run_transition_tasks): while one unit's
+       actions travel GPU->CPU and its C++ state machine advances, another
+       unit's scoring kernels run — parser and NER loops interleave so the
+       GPU never idles on a single pipe's sync."""
+
+    __slots__ = ("pipe", "shards", "t2v", "train", "T", "pre", "pre_d",
+                 "dPre32", "hip", "score_chunks", "gold_chunks",
+                 "valid_chunks", "n_states_total")
+
+    def __init__(self, pipe, shards, t2v, train: bool) -> None:
+        from spacy_ray_amd.ops import api as _ops
+
+        self.pipe = pipe
+        self.shards = shards
+        self.t2v = t2v
+        self.train = train
+        device = t2v.device
+        self.T = t2v.shape[0]
+        self.pre = pipe.module.precompute(t2v)
+        self.pre_d = self.pre.detach()
+        self.dPre32 = (
+            torch.zeros(self.pre.shape, dtype=torch.float32, device=device)
+            if train else None
+        )
+        self.hip = _ops.hip_ext() if device.type == "cuda" else None
+        self.score_chunks: List[torch.Tensor] = []
+        self.gold_chunks: List[torch.Tensor] = []
+        self.valid_chunks: List[torch.Tensor] = []
+        self.n_states_total = 0
+
+    def launch(self, states):
+        from spacy_ray_amd.ops import api as _ops
+        from spacy_ray_amd.utils import timing
+
+        device = self.t2v.device
+        with timing.span("raw/states_cpu"):
+            act_idx, feats, valid_np, gold_np = states.step_arrays(self.train)
+        if len(act_idx) == 0:
+            return None
+        with timing.span("raw/score_fwd"):
+            feats_t = torch.from_numpy(
+                np.where(feats < 0, self.T, feats).astype(np.int64)
+            ).to(device)
+            if self.train:
+                hidden = _ops.parser_step_score_accum(
+                    self.pre_d, feats_t, self.pipe.module.lower_b, self.dPre32
+                )
+                scores = self.pipe.module.upper(hidden)  # [S_active, A]
+            else:
+                scores = self.pipe.module.score(self.pre_d, feats_t)
+            valid_t = torch.from_numpy(valid_np).to(device)
+        if self.train:
+            with timing.span("raw/loss_build"):
+                gold_t = torch.from_numpy(gold_np).to(device)
+                self.score_chunks.append(scores)
+                self.gold_chunks.append(gold_t)
+                self.valid_chunks.append(valid_t)
+            sel_gold, sel_valid = gold_t, valid_t
+        else:
+            sel_gold, sel_valid = valid_t, valid_t
+        if self.hip is not None:
+            actions_dev = self.hip.action_select(scores.detach(), sel_gold, sel_valid)
+            return act_idx, actions_dev, None
+        s_np = scores.detach().float().cpu().numpy()
+        g_np = gold_np if self.train else valid_np
+        choose = np.where(g_np > 0, s_np, NEG_INF)
+        fallback = np.where(valid_np > 0, s_np, NEG_INF)
+        any_gold = (g_np > 0).any(axis=1, keepdims=True)
+        choose = np.where(any_gold, choose, fallback)
+        return act_idx, None, choose.argmax(axis=1).astype(np.int32)
+
+    def complete(self, states, pend) -> None:
+        from spacy_ray_amd.utils import timing
+
+        act_idx, actions_dev, actions_np = pend
+        with timing.span("raw/score_d2h"):
+            if actions_dev is not None:
+                actions_np = actions_dev.cpu().numpy().astype(np.int32)
+        with timing.span("raw/advance_cpu"):
+            actions = np.full(len(states), -1, dtype=np.int32)
+            actions[act_idx] = actions_np
+            states.advance(actions)
+        if self.train:
+            self.n_states_total += len(act_idx)
+
+
+def run_transition_tasks(tasks: List[_TransitionTask]) -> None:
+    """Interleaved driver over every (task, shard) unit: complete the
+    pending step (sync + C++ advance) then launch the next, round-robin —
+    one unit's CPU work hides under another unit's GPU work."""
+    units = [(t, s) for t in tasks for s in t.shards]
+    if not units:
+        return
+    pend = [None] * len(units)
+    done = [False] * len(units)
+    max_steps = max(4 * t.T + 16 for t in tasks)
+    for _ in range(max_steps):
+        progressed = False
+        for k, (task, states) in enumerate(units):
+            if done[k]:
+                continue
+            if pend[k] is not None:
+                task.complete(states, pend[k])
+                pend[k] = None
+            out = task.launch(states)
+            if out is None:
+                done[k] = True
+                continue
+            pend[k] = out
+            progressed = True
+        if not progressed:
+            break
+    for k, (task, states) in enumerate(units):  # drain
+        if pend[k] is not None:
+            task.complete(states, pend[k])
+            pend[k] = None
+
+
 class _TransitionPipeBase(TrainablePipe):
     """Shared greedy transition loop for parser and NER: per step, the C++
     batch object yields features/valid/costs for ALL states; the GPU scores
@@ -224,121 +353,25 @@ class _TransitionPipeBase(TrainablePipe):
                 break
         return out
 
-    def _step_loop(self, shards, t2v, train: bool):
-        """Greedy transition loop over double-buffered state shards.
+    def begin_task(self, shards, t2v, train: bool) -> "_TransitionTask":
+        """Build a step-loop task for this pipe (states + precompute +
+        gradient accumulator + loss collections)."""
+        return _TransitionTask(self, shards, t2v, train)
 
-        Two levels of overlap engineering here:
-        1. TWO-PHASE backward: the per-step losses (upper GEMM + bias path)
-           are backpropagated HERE while each step's dPre scatters into ONE
-           persistent fp32 buffer (ops.parser_step_score_accum); the
-           returned loss is a surrogate (pre · dPre).sum() whose backward
-           hands exactly dPre to the precompute GEMM inside the caller's
-           single main backward (lower_W / pad / tok2vec).
-        2. SHARD PIPELINING: states are split into word-balanced shards;
-           while one shard's scores travel GPU->CPU and its transitions
-           advance in C++, the other shard's scoring kernels run — the
-           CPU-advance / GPU-score double-buffering of SURVEY.md §2.2 N8.
-        """
-        from spacy_ray_amd.ops import api as _ops
+    def finish_task(self, task: "_TransitionTask"):
+        """Compute the one-shot CE over all collected steps, run the phase-1
+        backward, return (surrogate, display) — see _TransitionTask."""
         from spacy_ray_amd.utils import timing
 
-        device = t2v.device
-        T = t2v.shape[0]
-        pre = self.module.precompute(t2v)
-        pre_d = pre.detach()
-        if train:
-            dPre32 = torch.zeros(pre.shape, dtype=torch.float32, device=device)
-        hip = _ops.hip_ext() if device.type == "cuda" else None
-        # per-step collections; the CE loss over all steps is computed ONCE
-        # after the loop (one log_softmax on the concatenated active-state
-        # scores instead of ~2 x doc_len small launches)
-        score_chunks: List[torch.Tensor] = []
-        gold_chunks: List[torch.Tensor] = []
-        valid_chunks: List[torch.Tensor] = []
-        n_states_total = 0
-        max_steps = 4 * T + 16
-
-        def launch(states):
-            with timing.span("raw/states_cpu"):
-                act_idx, feats, valid_np, gold_np = states.step_arrays(train)
-            if len(act_idx) == 0:
-                return None
-            with timing.span("raw/score_fwd"):
-                feats_t = torch.from_numpy(
-                    np.where(feats < 0, T, feats).astype(np.int64)
-                ).to(device)
-                if train:
-                    hidden = _ops.parser_step_score_accum(
-                        pre_d, feats_t, self.module.lower_b, dPre32
-                    )
-                    scores = self.module.upper(hidden)  # [S_active, A]
-                else:
-                    scores = self.module.score(pre_d, feats_t)
-                valid_t = torch.from_numpy(valid_np).to(device)
-            if train:
-                with timing.span("raw/loss_build"):
-                    gold_t = torch.from_numpy(gold_np).to(device)
-                    score_chunks.append(scores)
-                    gold_chunks.append(gold_t)
-                    valid_chunks.append(valid_t)
-                sel_gold, sel_valid = gold_t, valid_t
-            else:
-                sel_gold, sel_valid = valid_t, valid_t
-            if hip is not None:
-                actions_dev = hip.action_select(scores.detach(), sel_gold, sel_valid)
-                return act_idx, actions_dev, None
-            s_np = scores.detach().float().cpu().numpy()
-            g_np = gold_np if train else valid_np
-            choose = np.where(g_np > 0, s_np, NEG_INF)
-            fallback = np.where(valid_np > 0, s_np, NEG_INF)
-            any_gold = (g_np > 0).any(axis=1, keepdims=True)
-            choose = np.where(any_gold, choose, fallback)
-            return act_idx, None, choose.argmax(axis=1).astype(np.int32)
-
-        def complete(states, pend):
-            act_idx, actions_dev, actions_np = pend
-            with timing.span("raw/score_d2h"):
-                if actions_dev is not None:
-                    actions_np = actions_dev.cpu().numpy().astype(np.int32)
-            with timing.span("raw/advance_cpu"):
-                actions = np.full(len(states), -1, dtype=np.int32)
-                actions[act_idx] = actions_np
-                states.advance(actions)
-            if train:
-                n = len(act_idx)
-                nonlocal n_states_total
-                n_states_total += n
-
-        pend = [None] * len(shards)
-        done = [False] * len(shards)
-        for _ in range(max_steps):
-            progressed = False
-            for k, states in enumerate(shards):
-                if done[k]:
-                    continue
-                if pend[k] is not None:
-                    complete(states, pend[k])
-                    pend[k] = None
-                out = launch(states)
-                if out is None:
-                    done[k] = True
-                    continue
-                pend[k] = out
-                progressed = True
-            if not progressed and all(p is None for p in pend):
-                break
-        for k, states in enumerate(shards):  # drain
-            if pend[k] is not None:
-                complete(states, pend[k])
-                pend[k] = None
-        if not train:
+        t2v = task.t2v
+        if not task.train:
             return None, 0.0
-        if not score_chunks:
+        if not task.score_chunks:
             return t2v.new_zeros(()), 0.0
         with timing.span("raw/loss_build"):
-            all_scores = torch.cat(score_chunks, dim=0).float()
-            all_gold = torch.cat(gold_chunks, dim=0) > 0
-            all_valid = torch.cat(valid_chunks, dim=0) > 0
+            all_scores = torch.cat(task.score_chunks, dim=0).float()
+            all_gold = torch.cat(task.gold_chunks, dim=0) > 0
+            all_valid = torch.cat(task.valid_chunks, dim=0) > 0
             counts_t = all_gold.sum(dim=-1)
             ok_t = counts_t > 0
             logp = torch.log_softmax(
@@ -346,14 +379,34 @@ class _TransitionPipeBase(TrainablePipe):
             )
             target = all_gold.float() / counts_t.clamp(min=1).unsqueeze(-1)
             row_loss = -(target * logp).sum(dim=-1)
-            step_loss = row_loss.masked_fill(~ok_t, 0).sum() / max(1, n_states_total)
+            step_loss = row_loss.masked_fill(~ok_t, 0).sum() / max(1, task.n_states_total)
         display = float(step_loss.detach())
         with timing.span("raw/phase1_bwd"):
             step_loss.backward()  # phase 1: upper + lower_b grads; dPre32 filled
         # dPre32 already carries the 1/n_states normalization (it was filled
         # by the normalized step_loss backward) — no extra scaling here.
-        surrogate = (pre.float() * dPre32).sum()
+        surrogate = (task.pre.float() * task.dPre32).sum()
         return surrogate, display
+
+    def _step_loop(self, shards, t2v, train: bool):
+        task = self.begin_task(shards, t2v, train)
+        run_transition_tasks([task])
+        return self.finish_task(task)
+
+    def make_loss_task(self, examples, t2v) -> "_TransitionTask":
+        lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
+        shards = []
+        for lo, hi, base in self._split_docs(lengths, self._n_shards(t2v)):
+            states = self._make_states(lengths[lo:hi], base)
+            self._set_gold(states, examples)  # global flat gold; offsets select
+            shards.append(states)
+        return self.begin_task(shards, t2v, train=True)
+
+    def make_predict_task(self, docs, t2v):
+        lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
+        splits = self._split_docs(lengths, self._n_shards(t2v))
+        shards = [self._make_states(lengths[lo:hi], base) for lo, hi, base in splits]
+        return self.begin_task(shards, t2v, train=False), splits, shards
 
     def _n_shards(self, t2v) -> int:
         # Measured on MI355X: 2-way shard pipelining LOST ~25% words/s on
@@ -365,20 +418,14 @@ class _TransitionPipeBase(TrainablePipe):
         return int(os.environ.get("SRX_PARSER_SHARDS", "1")) if t2v.is_cuda else 1
 
     def get_loss(self, examples, t2v, batch):
-        lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
-        shards = []
-        for lo, hi, base in self._split_docs(lengths, self._n_shards(t2v)):
-            states = self._make_states(lengths[lo:hi], base)
-            self._set_gold(states, examples)  # global flat gold; offsets select
-            shards.append(states)
-        return self._step_loop(shards, t2v, train=True)
+        task = self.make_loss_task(examples, t2v)
+        run_transition_tasks([task])
+        return self.finish_task(task)
 
     def predict_and_set(self, docs, t2v, batch) -> None:
-        lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
-        splits = self._split_docs(lengths, self._n_shards(t2v))
-        shards = [self._make_states(lengths[lo:hi], base) for lo, hi, base in splits]
+        task, splits, shards = self.make_predict_task(docs, t2v)
         with torch.no_grad():
-            self._step_loop(shards, t2v, train=False)
+            run_transition_tasks([task])
         for (lo, hi, base), states in zip(splits, shards):
             self._annotate(docs[lo:hi], states)
 
