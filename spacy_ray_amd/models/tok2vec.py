@@ -68,8 +68,24 @@ class MaxoutWindowEncoder(nn.Module):
         # doc-boundary masks computed once for all depth layers (and reused
         # by every seq2col forward AND backward — no per-call index work)
         starts, ends = ops.boundary_masks_u8(lengths, X.shape[0])
+        fused = ops.mwe_layer_available(X, self.width, self.blocks[0].pieces)
         for bi, block in enumerate(self.blocks):
             with timing.phase(f"t2v/encode{bi}"):
+                if fused:
+                    # whole block in ONE MFMA kernel (incl. dropout+residual)
+                    mask = None
+                    if drop and self.training:
+                        keep = 1.0 - drop
+                        mask = (
+                            (torch.rand_like(X, dtype=torch.float32) < keep)
+                            .to(X.dtype) / keep
+                        )
+                    X = ops.mwe_layer(
+                        X, block.weight, block.bias,
+                        block.norm.weight, block.norm.bias, starts, ends,
+                        mask, block.norm.eps,
+                    )
+                    continue
                 Y = block(ops.seq2col(X, lengths, starts, ends))
                 if drop and self.training:
                     Y = torch.nn.functional.dropout(Y, drop)

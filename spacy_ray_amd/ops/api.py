@@ -280,6 +280,63 @@ def parser_step_score_accum(precomputed_detached, feats, bias, dPre32):
     return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32)
 
 
+# ---------------------------------------------------- fused MWE layer (MFMA)
+class _MWELayer(torch.autograd.Function):
+    """One MaxoutWindowEncoder block as a single hand-written MFMA kernel:
+    Y = X + LN(maxout_3(seq2col(X) @ W^T + bias)).  Forward is the fused
+    gfx950 kernel (srx_mwe.hip.h); backward composes existing kernels:
+    layernorm_bwd on the saved maxout output -> maxout scatter -> the two
+    GEMM backwards (hipBLASLt) with seq2col recomputed -> seq2col_bwd."""
+
+    @staticmethod
+    def forward(ctx, X, weight, bias, g, b, starts, ends, dropmask, eps):
+        Y, Mout, which, mu, rstd = hip_ext().mwe_layer_fwd(
+            X.contiguous(), weight, bias, g, b, starts, ends, dropmask, eps
+        )
+        ctx.save_for_backward(X, weight, g, which, Mout, mu, rstd, starts, ends,
+                              *( [dropmask] if dropmask is not None else [] ))
+        ctx.has_mask = dropmask is not None
+        return Y
+
+    @staticmethod
+    def backward(ctx, dY):
+        if ctx.has_mask:
+            X, weight, g, which, Mout, mu, rstd, starts, ends, dropmask = ctx.saved_tensors
+        else:
+            X, weight, g, which, Mout, mu, rstd, starts, ends = ctx.saved_tensors
+            dropmask = None
+        hip = hip_ext()
+        dY = dY.contiguous()
+        dL = dY * dropmask if dropmask is not None else dY
+        # LN backward over the maxout output
+        dM, dg, db = hip.layernorm_bwd(dL.contiguous(), Mout, g, mu, rstd)
+        # maxout scatter: [T, W] -> [T, 3, W] pieces-major -> [T, 3W]
+        dPre = hip.maxout_bwd(dM, which, 3).reshape(dY.shape[0], -1)
+        dbias = dPre.sum(dim=0)
+        # GEMM backwards (pre = X3 @ W^T)
+        X3 = hip.seq2col_fwd(X, starts, ends)
+        dW = dPre.t().mm(X3)
+        dX3 = dPre.mm(weight)
+        dX = hip.seq2col_bwd(dX3.contiguous(), starts, ends)
+        dX += dY  # residual
+        return dX, dW, dbias.to(dY.dtype), dg, db, None, None, None, None
+
+
+def mwe_layer(X, weight, bias, g, b, starts, ends, dropmask=None, eps: float = 1e-5):
+    return _MWELayer.apply(X, weight, bias, g, b, starts, ends, dropmask, eps)
+
+
+def mwe_layer_available(X: torch.Tensor, width: int, pieces: int) -> bool:
+    return (
+        X.is_cuda
+        and X.dtype == torch.bfloat16
+        and pieces == 3
+        and width in (96, 128)
+        and X.shape[0] % 64 == 0
+        and hip_ext() is not None
+    )
+
+
 # ----------------------------------------------------------- softmax + CE
 class _SoftmaxCE(torch.autograd.Function):
     """Fused softmax + cross-entropy (SURVEY.md §2.5 softmax_ce_fwd/bwd):

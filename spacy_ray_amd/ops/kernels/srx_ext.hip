@@ -9,6 +9,7 @@
 #include "srx_elementwise.hip.h"
 #include "srx_embed_parser.hip.h"
 #include "srx_softmax_reduce.hip.h"
+#include "srx_mwe.hip.h"
 
 namespace {
 
@@ -420,6 +421,65 @@ at::Tensor reduce_max_bwd(at::Tensor dY, at::Tensor argmax, int64_t Ttot) {
   return dX;
 }
 
+// ------------------------------------------------ fused MWE layer (MFMA)
+template <int W>
+void launch_mwe(const at::Tensor& X, const at::Tensor& Wt, const at::Tensor& bias,
+                const at::Tensor& g, const at::Tensor& b, const at::Tensor& starts,
+                const at::Tensor& ends, const c10::optional<at::Tensor>& dropmask,
+                at::Tensor& Y, at::Tensor& Mout,
+                at::Tensor& which, at::Tensor& mu, at::Tensor& rstd, long T,
+                double eps, hipStream_t stream) {
+  constexpr int WP = W + 8;
+  constexpr int NW = W / 32;
+  size_t lds_bytes = (size_t)(3 * 64 * WP + 3 * W * WP) * sizeof(bf16_t) +
+                     (size_t)(2 * NW * 64) * sizeof(float);
+  static bool attr_set = false;
+  if (!attr_set) {
+    (void)hipFuncSetAttribute((const void*)mwe_layer_fwd_kernel<W>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds_bytes);
+    attr_set = true;
+  }
+  dim3 grid((unsigned)(T / 64));
+  hipLaunchKernelGGL((mwe_layer_fwd_kernel<W>), grid, dim3(64 * NW), lds_bytes,
+                     stream, (const bf16_t*)X.data_ptr(), (const bf16_t*)Wt.data_ptr(),
+                     (const bf16_t*)bias.data_ptr(), (const bf16_t*)g.data_ptr(),
+                     (const bf16_t*)b.data_ptr(), starts.data_ptr<uint8_t>(),
+                     ends.data_ptr<uint8_t>(),
+                     dropmask ? (const bf16_t*)dropmask->data_ptr() : nullptr,
+                     (bf16_t*)Y.data_ptr(),
+                     (bf16_t*)Mout.data_ptr(), which.data_ptr<uint8_t>(),
+                     mu.data_ptr<float>(), rstd.data_ptr<float>(), T, (float)eps);
+}
+
+std::vector<at::Tensor> mwe_layer_fwd(at::Tensor X, at::Tensor Wt, at::Tensor bias,
+                                      at::Tensor g, at::Tensor b, at::Tensor starts,
+                                      at::Tensor ends,
+                                      c10::optional<at::Tensor> dropmask, double eps) {
+  check_dev(X);
+  TORCH_CHECK(X.scalar_type() == at::kBFloat16, "mwe_layer is bf16-only");
+  long T = X.size(0);
+  int W = (int)X.size(1);
+  TORCH_CHECK(W == 96 || W == 128, "mwe_layer supports W in {96,128}");
+  TORCH_CHECK(T % 64 == 0, "mwe_layer needs T % 64 == 0 (TokenBatch pads)");
+  TORCH_CHECK(Wt.size(0) == 3 * W && Wt.size(1) == 3 * W, "weight must be [3W,3W]");
+  auto Y = at::empty_like(X);
+  auto Mout = at::empty_like(X);
+  auto which = at::empty({T, (long)W}, X.options().dtype(at::kByte));
+  auto mu = at::empty({T}, X.options().dtype(at::kFloat));
+  auto rstd = at::empty({T}, X.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (T > 0) {
+    if (W == 96)
+      launch_mwe<96>(X, Wt, bias, g, b, starts, ends, dropmask, Y, Mout, which, mu,
+                     rstd, T, eps, stream);
+    else
+      launch_mwe<128>(X, Wt, bias, g, b, starts, ends, dropmask, Y, Mout, which, mu,
+                      rstd, T, eps, stream);
+  }
+  return {Y, Mout, which, mu, rstd};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -440,4 +500,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_ragged", &reduce_ragged);
   m.def("reduce_ragged_bwd", &reduce_ragged_bwd);
   m.def("reduce_max_bwd", &reduce_max_bwd);
+  m.def("mwe_layer_fwd", &mwe_layer_fwd);
 }
