@@ -73,10 +73,13 @@ def make_synthetic_docs(
     n_ent_types: int = 4,
     seed: int = 0,
     world_seed: int = 0,
+    tree_style: str = "random",
 ) -> List[Doc]:
     """`world_seed` fixes the synthetic language itself (word->tag/dep/ent
     mappings) so train/dev corpora with different `seed`s sample different
-    docs from the SAME learnable world."""
+    docs from the SAME learnable world.  tree_style: "random" projective
+    trees, or "chain" (head = previous token; fully learnable — used by
+    convergence tests)."""
     rng = random.Random(seed)
     lexicon = [f"w{i}" for i in range(vocab_size)]
     # Zipf sampling over the lexicon
@@ -99,18 +102,24 @@ def make_synthetic_docs(
         for wid in word_ids:
             t = tag_of_word[wid] if rng.random() > 0.1 else rng.randrange(n_tags)
             tags.append(f"TAG{t}")
-        heads = _random_projective_heads(n, rng)
+        if tree_style == "chain":
+            heads = [-1] + list(range(n - 1))
+        else:
+            heads = _random_projective_heads(n, rng)
         deps = [f"dep{dep_of_word[wid]}" for wid in word_ids]
         for i, h in enumerate(heads):
             if h == -1:
                 deps[i] = "ROOT"
-        # BILUO entity spans: ~1 entity per 8 tokens, len 1-3, non-overlapping
+        # BILUO entity spans: word-DETERMINISTIC (span starts iff the word id
+        # is 0 mod 8; length and type derive from the id) so the pattern is
+        # learnable — convergence tests rely on this; ~1 entity per 8 tokens
         ents = ["O"] * n
         i = 0
         while i < n:
-            if rng.random() < 0.125:
-                length = min(rng.randint(1, 3), n - i)
-                etype = f"ENT{ent_of_word[word_ids[i]]}"
+            wid = int(word_ids[i])
+            if wid % 8 == 0:
+                length = min(1 + (wid // 8) % 3, n - i)
+                etype = f"ENT{ent_of_word[wid]}"
                 if length == 1:
                     ents[i] = f"U-{etype}"
                 else:
