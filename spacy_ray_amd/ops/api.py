@@ -138,7 +138,22 @@ class _HashEmbed(torch.autograd.Function):
     def backward(ctx, dY: torch.Tensor):
         (rows,) = ctx.saved_tensors
         if _want_hip(dY):
-            dT = hip_ext().hashembed_bwd(dY.contiguous(), rows, ctx.nrows)
+            hip = hip_ext()
+            T = dY.shape[0]
+            if T >= 4096:
+                # Zipf-hot rows serialize plain atomics (1.5 ms/call at
+                # T=128k measured) — sort by destination row + chunked
+                # segmented reduction instead (SURVEY.md §7 hard-part 2)
+                dst = rows.reshape(-1)
+                order = torch.argsort(dst)
+                dst_sorted = dst[order].contiguous().int()
+                src = (order // 4).int()
+                dT32 = torch.zeros(ctx.nrows, dY.shape[1], dtype=torch.float32,
+                                   device=dY.device)
+                hip.seg_scatter_add(dst_sorted, src, dY.contiguous(), dT32)
+                dT = dT32.to(dY.dtype)
+            else:
+                dT = hip.hashembed_bwd(dY.contiguous(), rows, ctx.nrows)
         else:
             dT = ref.hashembed_backward(dY, rows, ctx.nrows)
         return dT, None, None
@@ -232,14 +247,19 @@ def parser_step_score(precomputed, feats, bias):
 
 
 class _ParserStepScoreAccum(torch.autograd.Function):
-    """Step scorer whose backward ACCUMULATES dPre into one persistent fp32
-    buffer instead of materializing a fresh [T+1, nF, HP] gradient per
-    transition step (the per-step allocation + autograd summation was the
-    dominant host cost — see the two-phase scheme in pipes._step_loop).
-    `precomputed` is passed detached; only `bias` is differentiable here."""
+    """Step scorer whose backward ACCUMULATES dPre instead of materializing
+    a fresh [T+1, nF, HP] gradient per transition step (the per-step
+    allocation + autograd summation was the dominant host cost — see the
+    two-phase scheme in pipes).  `precomputed` is passed detached; only
+    `bias` is differentiable here.
+
+    GPU: backward computes the compact dSummed [S, HP] (no atomics) and
+    defers the scatter — finish_task sorts ALL steps' (token,f)
+    destinations once and runs the chunked segmented reduction.
+    CPU / fallback: scatter directly into the fp32 accumulator."""
 
     @staticmethod
-    def forward(ctx, precomputed, feats, bias, dPre32):
+    def forward(ctx, precomputed, feats, bias, dPre32, entries):
         S, nF = feats.shape
         HP = precomputed.shape[-1]
         P = 2
@@ -251,33 +271,60 @@ class _ParserStepScoreAccum(torch.autograd.Function):
             summed = precomputed[feats.long(), slot].sum(dim=1) + bias
             hidden, which = ref.maxout(summed.view(S, P, H))
         ctx.save_for_backward(feats, which)
-        # dPre32 is a side accumulator (mutated across steps) — stash it on
-        # ctx directly so autograd's saved-tensor version check doesn't trip.
+        # side accumulators (mutated across steps) — stashed on ctx directly
+        # so autograd's saved-tensor version check doesn't trip.
         ctx.dPre32 = dPre32
+        ctx.entries = entries
         ctx.HP = HP
         return hidden
 
     @staticmethod
     def backward(ctx, dHidden):
         feats, which = ctx.saved_tensors
-        dPre32 = ctx.dPre32
         HP = ctx.HP
         P = 2
         if _want_hip(dHidden):
-            dBias = hip_ext().parser_step_bwd_into(dHidden.contiguous(), feats, which, dPre32)
+            hip = hip_ext()
+            dSummed = hip.maxout_bwd(dHidden.contiguous(), which, P).view(-1, HP)
+            dBias = dSummed.float().sum(dim=0)
+            if ctx.entries is not None:
+                ctx.entries.append((feats, dSummed))
+            else:
+                dPre32 = ctx.dPre32
+                hip.parser_step_bwd_into(dHidden.contiguous(), feats, which, dPre32)
         else:
             S = dHidden.shape[0]
             nF = feats.shape[1]
             dSummed = ref.maxout_backward(dHidden, which, P).reshape(S, HP)
             dBias = dSummed.sum(dim=0)
-            flat = dPre32.view(-1, HP)
+            flat = ctx.dPre32.view(-1, HP)
             idx = (feats.long() * nF + torch.arange(nF, device=feats.device)).reshape(-1)
             flat.index_add_(0, idx, dSummed.float().repeat_interleave(nF, dim=0))
-        return None, None, dBias.to(dHidden.dtype), None
+        return None, None, dBias.to(dHidden.dtype), None, None
 
 
-def parser_step_score_accum(precomputed_detached, feats, bias, dPre32):
-    return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32)
+def parser_step_score_accum(precomputed_detached, feats, bias, dPre32, entries=None):
+    return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32, entries)
+
+
+def parser_scatter_entries(entries, dPre32) -> None:
+    """Batched dPre scatter: sort all steps' (token*nF+f) destinations once,
+    then one chunked segmented reduction into the fp32 accumulator."""
+    if not entries:
+        return
+    hip = hip_ext()
+    device = dPre32.device
+    nF = entries[0][0].shape[1]
+    HP = dPre32.shape[-1]
+    feats_all = torch.cat([f for f, _ in entries], dim=0)          # [SS, nF] i64
+    dS_all = torch.cat([d for _, d in entries], dim=0).contiguous()  # [SS, HP]
+    SS = feats_all.shape[0]
+    slot = torch.arange(nF, device=device)
+    dest = (feats_all * nF + slot).reshape(-1).int()
+    src = torch.arange(SS, device=device, dtype=torch.int32).repeat_interleave(nF)
+    order = torch.argsort(dest)
+    hip.seg_scatter_add(dest[order].contiguous(), src[order].contiguous(),
+                        dS_all, dPre32.view(-1, HP))
 
 
 # ---------------------------------------------------- fused MWE layer (MFMA)

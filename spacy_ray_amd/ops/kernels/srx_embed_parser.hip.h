@@ -97,6 +97,51 @@ __global__ void parser_step_fwd_kernel(const T* __restrict__ pre,
   }
 }
 
+// ------------------------------------- sorted segmented scatter-add
+// OUT[dst_sorted[e]] += SRC[src_idx[e]] for e in [0, M), where dst_sorted is
+// SORTED: each wave owns a chunk of entries, accumulates equal-dst runs in
+// registers and pushes ONE atomic per (run x chunk) — Zipf-hot rows get
+// M/chunk pushes instead of M (the sort-by-row segmented reduction planned
+// for HashEmbed backward, SURVEY.md §7 hard-part 2; also used for the
+// batched parser dPre scatter).
+template <typename T, int CHUNK>
+__global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
+                                       const int32_t* __restrict__ src_idx,
+                                       const T* __restrict__ SRC,
+                                       float* __restrict__ OUT,
+                                       long M, int W) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  const int ncols = (W + SRX_WAVE - 1) / SRX_WAVE;
+  for (long c0 = wave * CHUNK; c0 < M; c0 += nwaves * CHUNK) {
+    long e_end = min(c0 + (long)CHUNK, M);
+    int32_t cur = dst_sorted[c0];
+    float acc[8];  // ncols <= 8 supported (W <= 512)
+    for (int c = 0; c < ncols; c++) acc[c] = 0.f;
+    for (long e = c0; e < e_end; e++) {
+      int32_t d = dst_sorted[e];
+      if (d != cur) {
+        for (int c = 0; c < ncols; c++) {
+          int w = lane + c * SRX_WAVE;
+          if (w < W && acc[c] != 0.f) atomicAdd(OUT + (long)cur * W + w, acc[c]);
+          acc[c] = 0.f;
+        }
+        cur = d;
+      }
+      const T* src = SRC + (long)src_idx[e] * W;
+      for (int c = 0; c < ncols; c++) {
+        int w = lane + c * SRX_WAVE;
+        if (w < W) acc[c] += Elem<T>::ld(src + w);
+      }
+    }
+    for (int c = 0; c < ncols; c++) {
+      int w = lane + c * SRX_WAVE;
+      if (w < W && acc[c] != 0.f) atomicAdd(OUT + (long)cur * W + w, acc[c]);
+    }
+  }
+}
+
 // ----------------------------------------------------- action selection
 // One wave per state: actions[s] = argmax of scores over the is_gold
 // columns if any, else over the valid columns (first-occurrence tie-break,

@@ -298,6 +298,28 @@ std::vector<at::Tensor> parser_step_bwd(at::Tensor dHidden, at::Tensor feats,
   return {dPre32.to(dHidden.scalar_type()), dBias32.to(dHidden.scalar_type())};
 }
 
+// OUT (fp32 [No, W]) += segmented sums of SRC rows; dst_sorted must be sorted.
+void seg_scatter_add(at::Tensor dst_sorted, at::Tensor src_idx, at::Tensor SRC,
+                     at::Tensor OUT) {
+  check_dev(SRC);
+  TORCH_CHECK(OUT.scalar_type() == at::kFloat);
+  TORCH_CHECK(dst_sorted.scalar_type() == at::kInt && src_idx.scalar_type() == at::kInt);
+  long M = dst_sorted.numel();
+  int W = (int)SRC.size(-1);
+  TORCH_CHECK(W <= 512, "seg_scatter_add W <= 512");
+  if (M == 0) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  constexpr int CHUNK = 128;
+  long waves = (M + CHUNK - 1) / CHUNK;
+  int grid = (int)std::min<long>((waves * SRX_WAVE + kBlock - 1) / kBlock, 16384);
+  DISPATCH_F(SRC.scalar_type(), {
+    hipLaunchKernelGGL((seg_scatter_add_kernel<scalar_t, CHUNK>), dim3(grid),
+                       dim3(kBlock), 0, stream, dst_sorted.data_ptr<int32_t>(),
+                       src_idx.data_ptr<int32_t>(), (const scalar_t*)SRC.data_ptr(),
+                       OUT.data_ptr<float>(), M, W);
+  });
+}
+
 at::Tensor action_select(at::Tensor scores, at::Tensor is_gold, at::Tensor valid) {
   check_dev(scores);
   long S = scores.size(0);
@@ -495,6 +517,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parser_step_bwd", &parser_step_bwd);
   m.def("parser_step_bwd_into", &parser_step_bwd_into);
   m.def("action_select", &action_select);
+  m.def("seg_scatter_add", &seg_scatter_add);
   m.def("adam_step", &adam_step);
   m.def("softmax_ce", &softmax_ce);
   m.def("reduce_ragged", &reduce_ragged);

@@ -175,7 +175,7 @@ class _TransitionTask:
 
     __slots__ = ("pipe", "shards", "t2v", "train", "T", "pre", "pre_d",
                  "dPre32", "hip", "score_chunks", "gold_chunks",
-                 "valid_chunks", "n_states_total")
+                 "valid_chunks", "n_states_total", "entries")
 
     def __init__(self, pipe, shards, t2v, train: bool) -> None:
         from spacy_ray_amd.ops import api as _ops
@@ -193,6 +193,9 @@ class _TransitionTask:
             if train else None
         )
         self.hip = _ops.hip_ext() if device.type == "cuda" else None
+        # GPU train: per-step backwards append (feats, dSummed) here; the
+        # scatter happens once, batched+sorted, in finish_task
+        self.entries: Optional[List] = [] if (train and self.hip is not None) else None
         self.score_chunks: List[torch.Tensor] = []
         self.gold_chunks: List[torch.Tensor] = []
         self.valid_chunks: List[torch.Tensor] = []
@@ -213,7 +216,8 @@ class _TransitionTask:
             ).to(device)
             if self.train:
                 hidden = _ops.parser_step_score_accum(
-                    self.pre_d, feats_t, self.pipe.module.lower_b, self.dPre32
+                    self.pre_d, feats_t, self.pipe.module.lower_b,
+                    self.dPre32, self.entries,
                 )
                 scores = self.pipe.module.upper(hidden)  # [S_active, A]
             else:
@@ -382,7 +386,14 @@ class _TransitionPipeBase(TrainablePipe):
             step_loss = row_loss.masked_fill(~ok_t, 0).sum() / max(1, task.n_states_total)
         display = float(step_loss.detach())
         with timing.span("raw/phase1_bwd"):
-            step_loss.backward()  # phase 1: upper + lower_b grads; dPre32 filled
+            step_loss.backward()  # phase 1: upper + lower_b grads
+            if task.entries is not None:
+                # batched dPre scatter: one sort + segmented reduction for
+                # ALL steps (replaces per-step atomic scatters)
+                from spacy_ray_amd.ops.api import parser_scatter_entries
+
+                parser_scatter_entries(task.entries, task.dPre32)
+                task.entries.clear()
         # dPre32 already carries the 1/n_states normalization (it was filled
         # by the normalized step_loss backward) — no extra scaling here.
         surrogate = (task.pre.float() * task.dPre32).sum()

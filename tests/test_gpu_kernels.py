@@ -151,6 +151,53 @@ def test_action_select_gpu(dtype):
 
 @need_gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_seg_scatter_add_gpu(dtype):
+    """Sorted chunked segmented reduction == plain index_add (incl. a
+    Zipf-hot destination)."""
+    torch.manual_seed(9)
+    M, Ns, No, W = 5000, 700, 50, 96
+    dst = torch.randint(0, No, (M,), device=DEV, dtype=torch.int32)
+    dst[: M // 3] = 7  # hot row
+    src = torch.randint(0, Ns, (M,), device=DEV, dtype=torch.int32)
+    SRC = torch.randn(Ns, W, device=DEV, dtype=dtype)
+    order = torch.argsort(dst)
+    OUT = torch.zeros(No, W, device=DEV)
+    _srx_hip.seg_scatter_add(dst[order].contiguous(), src[order].contiguous(), SRC, OUT)
+    expect = torch.zeros(No, W, device=DEV)
+    expect.index_add_(0, dst.long(), SRC[src.long()].float())
+    tol = dict(atol=2e-1, rtol=1e-2) if dtype == torch.bfloat16 else dict(atol=1e-3, rtol=1e-4)
+    assert torch.allclose(OUT, expect, **tol), (OUT - expect).abs().max()
+
+
+@need_gpu
+def test_batched_parser_scatter_matches_atomic_path():
+    """The deferred sort+segmented dPre scatter == the per-step atomic
+    scatter kernel over several steps."""
+    from spacy_ray_amd.ops.api import parser_scatter_entries
+
+    torch.manual_seed(10)
+    T, nF, H = 300, 13, 64
+    HP = 2 * H
+    dtype = torch.bfloat16
+    dPre_atomic = torch.zeros(T + 1, nF, HP, device=DEV)
+    dPre_batched = torch.zeros(T + 1, nF, HP, device=DEV)
+    entries = []
+    for step in range(5):
+        S = 40 + step * 7
+        feats = torch.randint(0, T + 1, (S, nF), device=DEV)
+        dHidden = torch.randn(S, H, device=DEV, dtype=dtype)
+        which = (torch.rand(S, H, device=DEV) < 0.5).to(torch.uint8)
+        _srx_hip.parser_step_bwd_into(dHidden, feats, which, dPre_atomic)
+        dSummed = _srx_hip.maxout_bwd(dHidden, which, 2).view(S, HP)
+        entries.append((feats, dSummed))
+    parser_scatter_entries(entries, dPre_batched)
+    assert torch.allclose(dPre_atomic, dPre_batched, atol=5e-2, rtol=1e-2), (
+        (dPre_atomic - dPre_batched).abs().max()
+    )
+
+
+@need_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_softmax_ce_gpu(dtype):
     torch.manual_seed(7)
     N, C = 513, 50
