@@ -58,7 +58,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch-words", type=int, default=8000,
+    ap.add_argument("--batch-words", type=int, default=64000,
                     help="words per step per GPU (weak scaling)")
     ap.add_argument("--words-per-doc", type=int, default=20)
     ap.add_argument("--vocab-size", type=int, default=5000)

@@ -65,6 +65,10 @@ setup(
     entry_points={
         "console_scripts": [
             "spacy-mi = spacy_ray_amd.cli.main:main",
+            # reference-compatible command name: `spacy ray train ...`
+            # (BASELINE.json CLI-compat requirement; only the `ray` sub-app
+            # is provided)
+            "spacy = spacy_ray_amd.cli.main:main",
         ]
     },
 )

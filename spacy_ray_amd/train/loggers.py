@@ -36,7 +36,8 @@ def _make_console(nlp, with_time: bool):
             cols = (["T"] if with_time else []) + ["E", "#", "W"]
             cols += [f"Loss {c}" for c in loss_cols]
             sc = info.get("other_scores") or {}
-            score_cols[:] = sorted(k for k, v in sc.items() if isinstance(v, (int, float)))
+            score_cols[:] = sorted(k for k, v in sc.items()
+                                   if isinstance(v, (int, float)) and k != "speed")
             cols += score_cols + ["Score"]
             print("  ".join(f"{c:>10}" for c in cols))
             state["header"] = True

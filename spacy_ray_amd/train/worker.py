@@ -105,6 +105,11 @@ def distributed_train(
         if opt_state.exists():
             engine.load_state_dict(torch.load(opt_state, map_location=device))
 
+    # before_to_disk: optional @callbacks hook applied to nlp before saving
+    # (contract of create_before_to_disk_callback at
+    # /root/reference/spacy_ray/worker.py:96,222)
+    before_to_disk = T.get("before_to_disk")
+
     dev_examples = None
 
     def evaluate():
@@ -112,7 +117,10 @@ def distributed_train(
         if rank == 0:
             if dev_examples is None:
                 dev_examples = list(dev_corpus(nlp))
+            t_eval = time.time()
             scores = nlp.evaluate(dev_examples)
+            dt = max(1e-9, time.time() - t_eval)
+            scores["speed"] = sum(len(eg) for eg in dev_examples) / dt
             score = weighted_score(scores, T.get("score_weights") or {})
             payload = (score, scores)
         else:
@@ -185,13 +193,14 @@ def distributed_train(
             # all ranks participate: params are replicated post-all-gather;
             # rank 0 writes the pipeline, every rank its optimizer shard.
             if rank == 0:
-                best = info["score"]
                 nlp.meta["performance"] = info["other_scores"]
-                nlp.to_disk(Path(output_path) / "model-best")
+                to_save = before_to_disk(nlp) if before_to_disk else nlp
+                to_save.to_disk(Path(output_path) / "model-best")
             comm.barrier()
     if output_path:
         if rank == 0:
-            nlp.to_disk(Path(output_path) / "model-last")
+            to_save = before_to_disk(nlp) if before_to_disk else nlp
+            to_save.to_disk(Path(output_path) / "model-last")
         comm.barrier()
         torch.save(engine.state_dict(), Path(output_path) / "model-last" / f"optim.rank{rank}.pt")
     if rank == 0:
