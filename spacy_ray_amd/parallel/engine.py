@@ -125,6 +125,7 @@ class ZeRO1Engine:
 
         # ---- overlap machinery
         self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
+        self._next_bucket = len(self.buckets) - 1
         self._param_bucket: Dict[int, _Bucket] = {}
         for bkt in self.buckets:
             for p in bkt.params:
@@ -144,8 +145,25 @@ class ZeRO1Engine:
             return
         bkt = self._param_bucket[id(p)]
         bkt.ready += 1
-        if bkt.ready >= len(bkt.params) and not bkt.launched:
-            self._launch_bucket(bkt)
+        self._drain_ready()
+
+    def _drain_ready(self) -> None:
+        """Launch ready buckets in FIXED descending index order.
+
+        Collective order must be identical on every rank; grad-READY order is
+        not (e.g. a rank whose batch has no entities never backprops the NER
+        head, so that bucket only becomes ready at apply_step).  Draining in
+        a fixed order, gated on readiness, keeps NCCL ordering rank-uniform.
+        Descending because backward produces grads roughly from the last
+        pipeline params (parser/NER heads, phase-1) back to the embeddings —
+        so the fixed order still overlaps with backward."""
+        while self._next_bucket >= 0:
+            bkt = self.buckets[self._next_bucket]
+            if bkt.ready < len(bkt.params):
+                return
+            if not bkt.launched:
+                self._launch_bucket(bkt)
+            self._next_bucket -= 1
 
     def _launch_bucket(self, bkt: _Bucket) -> None:
         bkt.launched = True
@@ -198,9 +216,11 @@ class ZeRO1Engine:
 
     def _apply_step_inner(self) -> None:
         # launch any bucket the hooks didn't (grad-less params, overlap off)
-        for bkt in self.buckets:
+        # — same fixed descending order as _drain_ready
+        for bkt in reversed(self.buckets):
             if not bkt.launched:
                 self._launch_bucket(bkt)
+        self._next_bucket = len(self.buckets) - 1
         if self.is_cuda:
             torch.cuda.current_stream().wait_stream(self.comm_stream)
 
