@@ -33,7 +33,8 @@ ext_modules = [
             str(CSRC / "transitions.cpp"),
         ],
         include_dirs=[pybind11.get_include()],
-        extra_compile_args=["-O3", "-std=c++17", "-fvisibility=hidden"],
+        extra_compile_args=["-O3", "-std=c++17", "-fvisibility=hidden", "-fopenmp"],
+        extra_link_args=["-fopenmp"],
         language="c++",
     )
 ]

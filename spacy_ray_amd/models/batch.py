@@ -27,12 +27,17 @@ class TokenBatch:
         GPU, none on CPU."""
         if isinstance(device, str):
             device = torch.device(device)
-        if pad_to == 0:
-            pad_to = 2048 if device.type == "cuda" else 1
         self.docs = list(docs)
         lens_list = [len(d) for d in docs]
         real = int(sum(lens_list))
         self.n_real_tokens = real
+        if pad_to == 0:
+            if device.type == "cuda":
+                # coarser buckets at larger T: hipBLASLt re-runs solution
+                # selection per distinct M, so keep the distinct-M count tiny
+                pad_to = 2048 if real < 32768 else 8192
+            else:
+                pad_to = 1
         padded = -(-max(real, 1) // pad_to) * pad_to if pad_to > 1 else real
         n_pad = padded - real
         if n_pad > 0:

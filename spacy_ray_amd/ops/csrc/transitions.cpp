@@ -17,6 +17,24 @@
 #include <cstdint>
 #include <vector>
 
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+// step_arrays parallelism: bounded so 8 ranks/node don't oversubscribe
+static int srx_nthreads() {
+#ifdef _OPENMP
+  static int n = []() {
+    const char* env = getenv("SRX_CPP_THREADS");
+    if (env) return std::max(1, atoi(env));
+    return std::min(8, std::max(1, omp_get_max_threads() / 2));
+  }();
+  return n;
+#else
+  return 1;
+#endif
+}
+
 namespace py = pybind11;
 
 namespace {
@@ -75,6 +93,8 @@ struct ArcEagerBatch {
   ArcEagerBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
                 int32_t n_labels_, int32_t base_offset = 0)
       : n_labels(n_labels_) {
+    if (n_actions() > 256)
+      throw std::runtime_error("ArcEagerBatch: > 127 dep labels unsupported");
     auto L = lengths.unchecked<1>();
     int32_t off = base_offset;
     states.resize(L.shape(0));
@@ -242,13 +262,16 @@ struct ArcEagerBatch {
     py::array_t<uint8_t> valid_a({Sa, A});
     py::array_t<uint8_t> gold_a({with_gold ? Sa : 0, A});
     std::copy(idx.begin(), idx.end(), act.mutable_data());
-    std::vector<float> crow((size_t)A);
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
+#endif
     for (py::ssize_t k = 0; k < Sa; k++) {
+      float crow[256];  // A = 2 + 2*n_labels <= 256 labels supported
       size_t i = (size_t)idx[(size_t)k];
       fill_features(i, feats.mutable_data(k, 0));
       uint8_t* v = valid_a.mutable_data(k, 0);
       if (with_gold) {
-        fill_costs(i, v, crow.data());
+        fill_costs(i, v, crow);
         float cmin = KInvalid;
         for (py::ssize_t a = 0; a < A; a++)
           if (v[a] && crow[(size_t)a] < cmin) cmin = crow[(size_t)a];
@@ -404,6 +427,9 @@ struct BiluoBatch {
     py::array_t<uint8_t> valid_a({Sa, A});
     py::array_t<uint8_t> gold_a({with_gold ? Sa : 0, A});
     std::copy(idx.begin(), idx.end(), act.mutable_data());
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
+#endif
     for (py::ssize_t k = 0; k < Sa; k++) {
       size_t i = (size_t)idx[(size_t)k];
       const NerState& st = states[i];
