@@ -1,5 +1,9 @@
 """Micro-benchmarks of individual ops at bench shapes (diagnosis tool)."""
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 import torch
 

@@ -167,7 +167,6 @@ class ZeRO1Engine:
 
     def _launch_bucket(self, bkt: _Bucket) -> None:
         bkt.launched = True
-        r = self.comm.rank
         seg = self.flat_grad[bkt.start : bkt.end]
         out = self.grad_shard[bkt.shard_off : bkt.shard_off + bkt.per]
         if self.is_cuda:
