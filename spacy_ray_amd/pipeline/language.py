@@ -147,6 +147,21 @@ class Language:
             self.predict_docs([eg.predicted for eg in chunk])
         return score_examples(examples, self.pipe_names)
 
+    def pipe(self, inputs, batch_size: int = 256):
+        """Annotate a stream of Docs or texts in batches (spaCy nlp.pipe
+        contract)."""
+        from spacy_ray_amd.vocab.doc import simple_tokenize
+
+        batch: List[Doc] = []
+        for item in inputs:
+            doc = item if isinstance(item, Doc) else simple_tokenize(self.vocab, item)
+            batch.append(doc)
+            if len(batch) >= batch_size:
+                yield from self.predict_docs(batch)
+                batch = []
+        if batch:
+            yield from self.predict_docs(batch)
+
     def __call__(self, text: str) -> Doc:
         from spacy_ray_amd.vocab.doc import simple_tokenize
 

@@ -154,6 +154,35 @@ def test_parser_ner_train_step():
     assert all(np.isfinite(v) and v > 0 for v in losses.values()), losses
 
 
+def test_nlp_pipe_and_call():
+    cfg = Config.from_str(TAGGER_CFG)
+    nlp = init_nlp(cfg, sample_size=16)
+    docs = list(nlp.pipe(["w1 w2 w3", "w4 w5"], batch_size=2))
+    assert len(docs) == 2
+    assert docs[0].tags is not None and len(docs[0].tags) == 3
+    d = nlp("w1 w2")
+    assert d.tags is not None and len(d) == 2
+
+
+def test_xx_multilingual_config_trains_cpu():
+    cfg = Config.from_disk("examples/configs/xx_multilingual.cfg")
+    nlp = init_nlp(cfg, sample_size=24)
+    assert nlp.pipe_names == ["tok2vec", "tagger", "parser"]
+    icfg = cfg.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    train_corpus, _ = resolve_dot_names(icfg, [T["train_corpus"], T["dev_corpus"]])
+    examples = []
+    for eg in train_corpus(nlp):
+        examples.append(eg)
+        if len(examples) >= 12:
+            break
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    losses = {}
+    stepper.accumulate(examples, drop=0.0, losses=losses)
+    stepper.apply_step()
+    assert losses["tagger"] > 0 and losses["parser"] > 0
+
+
 def test_predict_sets_annotations():
     cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
     nlp = init_nlp(cfg, sample_size=16)
