@@ -49,9 +49,11 @@ class TokenBatch:
         attr = (np.concatenate(arrs, axis=0) if arrs
                 else np.zeros((0, 4), dtype=np.uint64))
         self.n_tokens = int(lens.sum())
+        from spacy_ray_amd.utils.pinned import to_device
+
         # bit-cast uint64 -> int64 (torch has no uint64); kernels re-interpret
-        self.attr_ids = torch.from_numpy(attr.view(np.int64)).to(device, non_blocking=True)
-        self.lengths = torch.from_numpy(lens).to(device, non_blocking=True)
+        self.attr_ids = to_device(attr.view(np.int64), device)
+        self.lengths = to_device(lens, device)
 
     def __len__(self) -> int:
         return len(self.docs)

@@ -142,7 +142,9 @@ class TaggerPipe(TrainablePipe):
 
         scores = self.module(t2v)  # [T, nT]
         gold_np = self._gold_ids(examples, t2v.shape[0])
-        gold = torch.from_numpy(gold_np).to(scores.device)
+        from spacy_ray_amd.utils.pinned import to_device
+
+        gold = to_device(gold_np, scores.device)
         n = int((gold_np >= 0).sum())
         loss = _ops.softmax_ce_loss(scores, gold) / max(1, n)
         return loss, float(loss.detach())
@@ -204,6 +206,7 @@ class _TransitionTask:
     def launch(self, states):
         from spacy_ray_amd.ops import api as _ops
         from spacy_ray_amd.utils import timing
+        from spacy_ray_amd.utils.pinned import to_device
 
         device = self.t2v.device
         with timing.span("raw/states_cpu"):
@@ -211,9 +214,9 @@ class _TransitionTask:
         if len(act_idx) == 0:
             return None
         with timing.span("raw/score_fwd"):
-            feats_t = torch.from_numpy(
-                np.where(feats < 0, self.T, feats).astype(np.int64)
-            ).to(device)
+            feats_t = to_device(
+                np.where(feats < 0, self.T, feats).astype(np.int64), device
+            )
             if self.train:
                 hidden = _ops.parser_step_score_accum(
                     self.pre_d, feats_t, self.pipe.module.lower_b,
@@ -222,10 +225,10 @@ class _TransitionTask:
                 scores = self.pipe.module.upper(hidden)  # [S_active, A]
             else:
                 scores = self.pipe.module.score(self.pre_d, feats_t)
-            valid_t = torch.from_numpy(valid_np).to(device)
+            valid_t = to_device(valid_np, device)
         if self.train:
             with timing.span("raw/loss_build"):
-                gold_t = torch.from_numpy(gold_np).to(device)
+                gold_t = to_device(gold_np, device)
                 self.score_chunks.append(scores)
                 self.gold_chunks.append(gold_t)
                 self.valid_chunks.append(valid_t)
