@@ -250,7 +250,7 @@ struct ArcEagerBatch {
   // ---- fused per-step call: ONE crossing of the pybind boundary returns
   // (active_idx, features, valid, is_gold) for the active states only.
   // is_gold[s,a] = (cost <= min valid cost + eps); empty when !with_gold.
-  py::tuple step_arrays(bool with_gold) const {
+  py::tuple step_arrays(bool with_gold, int64_t pad_row = -1) const {
     std::vector<int32_t> idx;
     idx.reserve(states.size());
     for (size_t i = 0; i < states.size(); i++)
@@ -258,7 +258,7 @@ struct ArcEagerBatch {
     py::ssize_t Sa = (py::ssize_t)idx.size();
     const py::ssize_t A = n_actions();
     py::array_t<int32_t> act(Sa);
-    py::array_t<int32_t> feats({Sa, (py::ssize_t)13});
+    py::array_t<int64_t> feats({Sa, (py::ssize_t)13});
     py::array_t<uint8_t> valid_a({Sa, A});
     py::array_t<uint8_t> gold_a({with_gold ? Sa : 0, A});
     std::copy(idx.begin(), idx.end(), act.mutable_data());
@@ -267,8 +267,11 @@ struct ArcEagerBatch {
 #endif
     for (py::ssize_t k = 0; k < Sa; k++) {
       float crow[256];  // A = 2 + 2*n_labels <= 256 labels supported
+      int32_t f32[13];
       size_t i = (size_t)idx[(size_t)k];
-      fill_features(i, feats.mutable_data(k, 0));
+      fill_features(i, f32);
+      int64_t* fo = feats.mutable_data(k, 0);
+      for (int q = 0; q < 13; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
       uint8_t* v = valid_a.mutable_data(k, 0);
       if (with_gold) {
         fill_costs(i, v, crow);
@@ -415,7 +418,7 @@ struct BiluoBatch {
   }
 
   // fused per-step call (same contract as ArcEagerBatch::step_arrays)
-  py::tuple step_arrays(bool with_gold) const {
+  py::tuple step_arrays(bool with_gold, int64_t pad_row = -1) const {
     std::vector<int32_t> idx;
     idx.reserve(states.size());
     for (size_t i = 0; i < states.size(); i++)
@@ -423,7 +426,7 @@ struct BiluoBatch {
     py::ssize_t Sa = (py::ssize_t)idx.size();
     const py::ssize_t A = n_actions();
     py::array_t<int32_t> act(Sa);
-    py::array_t<int32_t> feats({Sa, (py::ssize_t)6});
+    py::array_t<int64_t> feats({Sa, (py::ssize_t)6});
     py::array_t<uint8_t> valid_a({Sa, A});
     py::array_t<uint8_t> gold_a({with_gold ? Sa : 0, A});
     std::copy(idx.begin(), idx.end(), act.mutable_data());
@@ -433,7 +436,10 @@ struct BiluoBatch {
     for (py::ssize_t k = 0; k < Sa; k++) {
       size_t i = (size_t)idx[(size_t)k];
       const NerState& st = states[i];
-      fill_features(i, feats.mutable_data(k, 0));
+      int32_t f32[6];
+      fill_features(i, f32);
+      int64_t* fo = feats.mutable_data(k, 0);
+      for (int q = 0; q < 6; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
       uint8_t* v = valid_a.mutable_data(k, 0);
       fill_valid(v, st);
       if (with_gold) {
@@ -537,7 +543,7 @@ void init_transitions(py::module_& m) {
       .def("valid", &ArcEagerBatch::valid)
       .def("costs", &ArcEagerBatch::costs)
       .def("advance", &ArcEagerBatch::advance)
-      .def("step_arrays", &ArcEagerBatch::step_arrays, py::arg("with_gold"))
+      .def("step_arrays", &ArcEagerBatch::step_arrays, py::arg("with_gold"), py::arg("pad_row") = -1)
       .def("heads", &ArcEagerBatch::heads)
       .def("labels", &ArcEagerBatch::labels);
 
@@ -552,6 +558,6 @@ void init_transitions(py::module_& m) {
       .def("valid", &BiluoBatch::valid)
       .def("costs", &BiluoBatch::costs)
       .def("advance", &BiluoBatch::advance)
-      .def("step_arrays", &BiluoBatch::step_arrays, py::arg("with_gold"))
+      .def("step_arrays", &BiluoBatch::step_arrays, py::arg("with_gold"), py::arg("pad_row") = -1)
       .def("tags", &BiluoBatch::tags);
 }

@@ -210,13 +210,13 @@ class _TransitionTask:
 
         device = self.t2v.device
         with timing.span("raw/states_cpu"):
-            act_idx, feats, valid_np, gold_np = states.step_arrays(self.train)
+            # feats come back int64 with missing slots already remapped to
+            # the pad row (= T, the learned-pad index in precompute)
+            act_idx, feats, valid_np, gold_np = states.step_arrays(self.train, self.T)
         if len(act_idx) == 0:
             return None
         with timing.span("raw/score_fwd"):
-            feats_t = to_device(
-                np.where(feats < 0, self.T, feats).astype(np.int64), device
-            )
+            feats_t = to_device(feats, device)
             if self.train:
                 hidden = _ops.parser_step_score_accum(
                     self.pre_d, feats_t, self.pipe.module.lower_b,
