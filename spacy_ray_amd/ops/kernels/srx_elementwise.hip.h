@@ -227,9 +227,11 @@ __global__ void adam_step_kernel(const T* __restrict__ grad,
                                  float* __restrict__ m,
                                  float* __restrict__ v,
                                  T* __restrict__ param_out,
-                                 long n, float clip_scale, float lr,
+                                 long n, const float* __restrict__ clip_scale_ptr,
+                                 float lr,
                                  float beta1, float beta2, float eps,
                                  float wd, float bc1, float bc2) {
+  const float clip_scale = clip_scale_ptr ? *clip_scale_ptr : 1.0f;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
     float gv = Elem<T>::ld(grad + i) * clip_scale;

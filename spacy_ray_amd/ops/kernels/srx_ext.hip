@@ -337,19 +337,23 @@ at::Tensor action_select(at::Tensor scores, at::Tensor is_gold, at::Tensor valid
 }
 
 // ------------------------------------------------------------ fused Adam
+// clip_scale: 0-dim fp32 DEVICE tensor (or empty for no clip) — keeping the
+// scale on-device avoids a host sync per optimizer step.
 void adam_step(at::Tensor grad, at::Tensor master, at::Tensor m, at::Tensor v,
-               at::Tensor param_out, double clip_scale, double lr, double beta1,
+               at::Tensor param_out, at::Tensor clip_scale, double lr, double beta1,
                double beta2, double eps, double wd, double bc1, double bc2) {
   check_dev(grad);
   long n = grad.numel();
   if (n == 0) return;
+  const float* scale_ptr =
+      clip_scale.numel() > 0 ? clip_scale.data_ptr<float>() : nullptr;
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(grad.scalar_type(), {
     hipLaunchKernelGGL((adam_step_kernel<scalar_t>), dim3(grid_for(n, 4)), dim3(kBlock),
                        0, stream, (const scalar_t*)grad.data_ptr(),
                        master.data_ptr<float>(), m.data_ptr<float>(),
                        v.data_ptr<float>(), (scalar_t*)param_out.data_ptr(), n,
-                       (float)clip_scale, (float)lr, (float)beta1, (float)beta2,
+                       scale_ptr, (float)lr, (float)beta1, (float)beta2,
                        (float)eps, (float)wd, (float)bc1, (float)bc2);
   });
 }

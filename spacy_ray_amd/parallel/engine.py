@@ -239,12 +239,14 @@ class ZeRO1Engine:
         hip = self._hip_ext() if self.is_cuda else None
         if hip is not None and (s.L2_is_weight_decay or not s.L2):
             # ONE fused kernel: clip-scale + decoupled wd + Adam + bf16 cast
-            # (SURVEY.md §2.5 fused_adam_sharded)
+            # (SURVEY.md §2.5 fused_adam_sharded); the clip scale stays on
+            # device — no host sync in the optimizer step
             if self._param_shard is None:
                 self._param_shard = torch.empty_like(self.grad_shard)
-            scale = float(scale_t) if scale_t is not None else 1.0
+            scale_dev = (scale_t.float().reshape(1) if scale_t is not None
+                         else torch.empty(0, device=self.device))
             hip.adam_step(self.grad_shard, self.master, self.exp_avg,
-                          self.exp_avg_sq, self._param_shard, scale, lr,
+                          self.exp_avg_sq, self._param_shard, scale_dev, lr,
                           s.beta1, s.beta2, s.eps, wd, bc1, bc2)
             new_param = self._param_shard
         else:

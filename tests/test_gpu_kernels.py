@@ -266,7 +266,8 @@ def test_adam_step_gpu(dtype):
     vr = (1 - b2) * g32 * g32
     denom = (vr / (1 - b2)).sqrt() + eps
     ref_p -= (lr / (1 - b1)) * (mr / denom)
-    _srx_hip.adam_step(grad, master, m, v, param_out, 0.5, lr, b1, b2, eps, wd,
+    scale = torch.tensor([0.5], device=DEV)
+    _srx_hip.adam_step(grad, master, m, v, param_out, scale, lr, b1, b2, eps, wd,
                        1 - b1, 1 - b2)
     assert torch.allclose(master, ref_p, atol=1e-6)
     assert torch.allclose(m, mr, atol=1e-6)
