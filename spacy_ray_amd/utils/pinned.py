@@ -39,10 +39,15 @@ def _capacity(n: int) -> int:
     return c
 
 
+import os
+
+_DISABLED = os.environ.get("SRX_PINNED", "1") == "0"
+
+
 def to_device(arr: np.ndarray, device) -> torch.Tensor:
     """Upload a numpy array via a pooled pinned buffer (async H2D)."""
     src = torch.from_numpy(np.ascontiguousarray(arr))
-    if not torch.cuda.is_available():
+    if _DISABLED or not torch.cuda.is_available():
         return src.to(device)
     n = src.numel()
     if n == 0:
