@@ -56,10 +56,11 @@ def build_batches(nlp, *, batch_words: int, n_batches: int, seed: int,
 def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=12)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch-words", type=int, default=64000,
-                    help="words per step per GPU (weak scaling)")
+    ap.add_argument("--batch-words", type=int, default=0,
+                    help="words per step per GPU (weak scaling); default "
+                         "256k on GPU, 8k on CPU")
     ap.add_argument("--words-per-doc", type=int, default=20)
     ap.add_argument("--vocab-size", type=int, default=5000)
     ap.add_argument("--config", type=str, default="examples/configs/en_core_cnn.cfg")
@@ -77,6 +78,8 @@ def main() -> None:
     from spacy_ray_amd.pipeline.language import init_nlp
 
     use_cuda = torch.cuda.is_available()
+    if args.batch_words == 0:
+        args.batch_words = 256000 if use_cuda else 8000
     if use_cuda:
         torch.cuda.set_device(local_rank)
         device = f"cuda:{local_rank}"
