@@ -96,7 +96,8 @@ def main() -> None:
     torch.manual_seed(1234 + rank)
 
     batches = build_batches(
-        nlp, batch_words=args.batch_words, n_batches=8,
+        nlp, batch_words=args.batch_words,
+        n_batches=8 if args.batch_words <= 64000 else 4,
         seed=100 + rank, words_per_doc=args.words_per_doc,
         vocab_size=args.vocab_size,
     )
