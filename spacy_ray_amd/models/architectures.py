@@ -88,12 +88,14 @@ def make_transformer_model(
     transformer_config: Optional[dict] = None,
     window: int = 128,
     stride: int = 96,
+    attn_implementation: str = "eager",
 ):
     def build():
         from .transformer import TransformerTok2Vec
 
         return TransformerTok2Vec(name=name, window=window, stride=stride,
-                                  transformer_config=transformer_config)
+                                  transformer_config=transformer_config,
+                                  attn_implementation=attn_implementation)
 
     tc = transformer_config or {}
     width = tc.get("hidden_size", 768 if name == "roberta-base" else 64)

@@ -32,7 +32,12 @@ N_SPECIAL = 10
 
 class TransformerTok2Vec(nn.Module):
     def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
-                 transformer_config: Optional[dict] = None):
+                 transformer_config: Optional[dict] = None,
+                 attn_implementation: str = "eager"):
+        """attn_implementation: "eager" by default — the windows here are
+        tens of tokens, where aotriton's flash BACKWARD is degenerate
+        (measured 3.8 ms/call = ~46 ms/step at 128k words); eager bmm
+        attention is the right regime for L~32."""
         super().__init__()
         from transformers import RobertaConfig, RobertaModel
 
@@ -47,6 +52,7 @@ class TransformerTok2Vec(nn.Module):
                         max_position_embeddings=514)
         base.update(cfg_kwargs)
         config = RobertaConfig(**base)
+        config._attn_implementation = attn_implementation
         self.trf = RobertaModel(config, add_pooling_layer=False)
         self.width = config.hidden_size
         self.vocab_size = config.vocab_size
