@@ -88,7 +88,7 @@ def make_transformer_model(
     transformer_config: Optional[dict] = None,
     window: int = 128,
     stride: int = 96,
-    attn_implementation: str = "eager",
+    attn_implementation: str = "sdpa",
 ):
     def build():
         from .transformer import TransformerTok2Vec

@@ -33,11 +33,11 @@ N_SPECIAL = 10
 class TransformerTok2Vec(nn.Module):
     def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
                  transformer_config: Optional[dict] = None,
-                 attn_implementation: str = "eager"):
-        """attn_implementation: "eager" by default — the windows here are
-        tens of tokens, where aotriton's flash BACKWARD is degenerate
-        (measured 3.8 ms/call = ~46 ms/step at 128k words); eager bmm
-        attention is the right regime for L~32."""
+                 attn_implementation: str = "sdpa"):
+        """attn_implementation: "sdpa" (A/B-measured best at these window
+        sizes: eager bmm attention lost ~13% despite aotriton's flash
+        backward being slow at L~22 — the extra elementwise kernels cost
+        more than flash-bwd saves)."""
         super().__init__()
         from transformers import RobertaConfig, RobertaModel
 
