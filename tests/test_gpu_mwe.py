@@ -89,6 +89,44 @@ def test_mwe_layer_fused_matches_composed(W, with_mask):
 
 
 @need_gpu
+def test_srx_embedding_backward_matches_torch():
+    from spacy_ray_amd.models.transformer import _SrxEmbedding
+
+    torch.manual_seed(3)
+    R, W, N = 500, 64, 4000
+    emb = _SrxEmbedding(R, W, padding_idx=1).to(DEV).to(torch.bfloat16)
+    ids = torch.randint(0, R, (N,), device=DEV)
+    ids[::5] = 1    # padding rows must get no grad
+    ids[::3] = 7    # hot row
+    Y = emb(ids)
+    dY = torch.randn_like(Y)
+    Y.backward(dY)
+    ref_emb = torch.nn.Embedding(R, W, padding_idx=1).to(DEV)
+    with torch.no_grad():
+        ref_emb.weight.copy_(emb.weight.float())
+    Y2 = ref_emb(ids)
+    Y2.backward(dY.float())
+    assert torch.allclose(emb.weight.grad.float(), ref_emb.weight.grad,
+                          atol=2e-1, rtol=5e-2)
+    assert emb.weight.grad[1].abs().max() == 0  # padding_idx zeroed
+
+
+@need_gpu
+def test_srx_layernorm_module_matches_torch():
+    from spacy_ray_amd.models.transformer import _SrxLayerNorm
+
+    torch.manual_seed(4)
+    ln = _SrxLayerNorm(768, eps=1e-5).to(DEV).to(torch.bfloat16)
+    X = torch.randn(257, 768, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    Y = ln(X)
+    Yr = torch.nn.functional.layer_norm(X.float(), (768,), ln.weight.float(),
+                                        ln.bias.float(), 1e-5)
+    assert torch.allclose(Y.float(), Yr, atol=5e-2, rtol=5e-2)
+    Y.sum().backward()
+    assert torch.isfinite(X.grad.float()).all()
+
+
+@need_gpu
 def test_mwe_used_in_encoder_forward():
     """The CNN encoder must actually route through the fused kernel on GPU
     (bf16, W=96, padded T)."""
