@@ -117,7 +117,7 @@ __global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
   for (long c0 = wave * CHUNK; c0 < M; c0 += nwaves * CHUNK) {
     long e_end = min(c0 + (long)CHUNK, M);
     int32_t cur = dst_sorted[c0];
-    float acc[8];  // ncols <= 8 supported (W <= 512)
+    float acc[16];  // ncols <= 16 supported (W <= 1024)
     for (int c = 0; c < ncols; c++) acc[c] = 0.f;
     for (long e = c0; e < e_end; e++) {
       int32_t d = dst_sorted[e];

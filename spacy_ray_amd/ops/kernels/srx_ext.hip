@@ -306,7 +306,7 @@ void seg_scatter_add(at::Tensor dst_sorted, at::Tensor src_idx, at::Tensor SRC,
   TORCH_CHECK(dst_sorted.scalar_type() == at::kInt && src_idx.scalar_type() == at::kInt);
   long M = dst_sorted.numel();
   int W = (int)SRC.size(-1);
-  TORCH_CHECK(W <= 512, "seg_scatter_add W <= 512");
+  TORCH_CHECK(W <= 1024, "seg_scatter_add W <= 1024");
   if (M == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   constexpr int CHUNK = 128;
