@@ -114,7 +114,9 @@ def main() -> None:
             torch.cuda.synchronize()
         return
 
-    for i in range(args.warmup):
+    # warmup must cover EVERY distinct batch shape: hipBLASLt runs solution
+    # selection per new GEMM M, which otherwise lands in the timed region
+    for i in range(max(args.warmup, len(batches))):
         step(i)
 
     comm.barrier()

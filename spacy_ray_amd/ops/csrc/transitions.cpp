@@ -27,7 +27,9 @@ static int srx_nthreads() {
   static int n = []() {
     const char* env = getenv("SRX_CPP_THREADS");
     if (env) return std::max(1, atoi(env));
-    return std::min(8, std::max(1, omp_get_max_threads() / 2));
+    const char* ws = getenv("WORLD_SIZE");
+    int world = ws ? std::max(1, atoi(ws)) : 1;
+    return std::min(16, std::max(1, omp_get_max_threads() / (2 * world)));
   }();
   return n;
 #else
@@ -290,6 +292,9 @@ struct ArcEagerBatch {
 
   void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
     auto a = actions.unchecked<1>();
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (a.shape(0) > 2048)
+#endif
     for (py::ssize_t i = 0; i < a.shape(0); i++) {
       ParserState& st = states[i];
       if (st.final_state()) continue;
@@ -498,6 +503,9 @@ struct BiluoBatch {
 
   void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
     auto a = actions.unchecked<1>();
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (a.shape(0) > 2048)
+#endif
     for (py::ssize_t s = 0; s < a.shape(0); s++) {
       NerState& st = states[s];
       if (st.final_state()) continue;
