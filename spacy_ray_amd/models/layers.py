@@ -33,15 +33,10 @@ class Maxout(nn.Module):
         self.norm = LayerNorm(nO) if normalize else None
 
     def forward(self, X: torch.Tensor) -> torch.Tensor:
-        from spacy_ray_amd.utils import timing
-
-        with timing.phase("maxout/linear"):
-            Y = torch.nn.functional.linear(X, self.weight, self.bias)
-        with timing.phase("maxout/max"):
-            Y = ops.maxout(Y.view(*Y.shape[:-1], self.pieces, self.nO))
+        Y = torch.nn.functional.linear(X, self.weight, self.bias)
+        Y = ops.maxout(Y.view(*Y.shape[:-1], self.pieces, self.nO))
         if self.norm is not None:
-            with timing.phase("maxout/ln"):
-                Y = self.norm(Y)
+            Y = self.norm(Y)
         return Y
 
 

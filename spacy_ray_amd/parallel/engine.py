@@ -77,11 +77,15 @@ class ZeRO1Engine:
         params = [p for _, p in named]
 
         # ---- bucket construction at parameter boundaries
+        # Every param starts at a 64-element boundary: an odd element offset
+        # gives hipBLASLt a 2-byte-aligned weight pointer and it drops to a
+        # slow path (measured 16 ms for a 116-GFLOP GEMM vs 0.3 ms aligned).
         self.buckets: List[_Bucket] = []
         b = _Bucket(0, 0)
         offset = 0
         self._offsets: List[int] = []
         for p in params:
+            offset = -(-offset // 64) * 64
             n = p.numel()
             self._offsets.append(offset)
             b.params.append(p)
