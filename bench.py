@@ -60,7 +60,7 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--batch-words", type=int, default=0,
                     help="words per step per GPU (weak scaling); default "
-                         "512k on GPU, 8k on CPU")
+                         "1M on GPU, 8k on CPU")
     ap.add_argument("--words-per-doc", type=int, default=20)
     ap.add_argument("--vocab-size", type=int, default=5000)
     ap.add_argument("--config", type=str, default="examples/configs/en_core_cnn.cfg")
@@ -79,7 +79,7 @@ def main() -> None:
 
     use_cuda = torch.cuda.is_available()
     if args.batch_words == 0:
-        args.batch_words = 512000 if use_cuda else 8000
+        args.batch_words = 1000000 if use_cuda else 8000
     if use_cuda:
         torch.cuda.set_device(local_rank)
         device = f"cuda:{local_rank}"

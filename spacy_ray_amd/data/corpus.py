@@ -82,6 +82,12 @@ def make_synthetic_docs(
     convergence tests)."""
     rng = random.Random(seed)
     lexicon = [f"w{i}" for i in range(vocab_size)]
+    # per-word attr hashes computed ONCE over the lexicon; docs index into
+    # this matrix (the per-token string path was the dominant corpus-build
+    # cost at bench scales)
+    from spacy_ray_amd.vocab.attrs import extract_attr_hashes
+
+    lex_attr = extract_attr_hashes(lexicon)
     # Zipf sampling over the lexicon
     ranks = np.arange(1, vocab_size + 1, dtype=np.float64)
     probs = 1.0 / ranks
@@ -130,7 +136,8 @@ def make_synthetic_docs(
                 i += length
             else:
                 i += 1
-        docs.append(Doc(vocab, words, tags=tags, heads=heads, deps=deps, ents=ents))
+        docs.append(Doc(vocab, words, tags=tags, heads=heads, deps=deps,
+                        ents=ents, attr_hashes=lex_attr[word_ids]))
     return docs
 
 

@@ -46,11 +46,15 @@ class Doc:
         heads: Optional[Sequence[int]] = None,
         deps: Optional[Sequence[str]] = None,
         ents: Optional[Sequence[str]] = None,  # per-token BILUO strings, e.g. "B-ORG"/"O"
+        attr_hashes: Optional[np.ndarray] = None,  # precomputed (n,4) uint64
     ) -> None:
         self.vocab = vocab
         self.words = list(words)
         self.spaces = list(spaces) if spaces is not None else [True] * len(self.words)
-        self.attr_hashes = extract_attr_hashes(self.words)
+        # corpora with a known lexicon pass precomputed attr hashes (one
+        # lookup per token instead of 4 string builds + hashes)
+        self.attr_hashes = (attr_hashes if attr_hashes is not None
+                            else extract_attr_hashes(self.words))
         self.tags = list(tags) if tags is not None else None
         self.heads = np.asarray(heads, dtype=np.int32) if heads is not None else None
         self.deps = list(deps) if deps is not None else None
