@@ -307,6 +307,30 @@ def parser_step_score_accum(precomputed_detached, feats, bias, dPre32, entries=N
     return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32, entries)
 
 
+class _InjectGrad(torch.autograd.Function):
+    """Hand a precomputed gradient to a tensor through autograd: forward is
+    a zero scalar (the value is never used — losses are logged separately);
+    backward returns `grad_buf` cast to the tensor's dtype.  Replaces the
+    (pre.float() * dPre32).sum() surrogate, which cost three full passes
+    over the [T+1,nF,HP] fp32 buffer per pipe per step."""
+
+    @staticmethod
+    def forward(ctx, x, grad_buf):
+        ctx.grad_buf = grad_buf
+        ctx.dtype = x.dtype
+        return x.new_zeros(())
+
+    @staticmethod
+    def backward(ctx, g):
+        # g is the upstream scalar (1.0 from a plain loss sum); never read it
+        # on the host — a float(g) here would device-sync every step
+        return ctx.grad_buf.to(ctx.dtype) * g, None
+
+
+def inject_grad(x: torch.Tensor, grad_buf: torch.Tensor) -> torch.Tensor:
+    return _InjectGrad.apply(x, grad_buf)
+
+
 def parser_scatter_entries(entries, dPre32) -> None:
     """Batched dPre scatter: sort all steps' (token*nF+f) destinations once,
     then one chunked segmented reduction into the fp32 accumulator."""

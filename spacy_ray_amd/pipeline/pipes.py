@@ -411,8 +411,11 @@ class _TransitionPipeBase(TrainablePipe):
                 parser_scatter_entries(task.entries, task.dPre32)
                 task.entries.clear()
         # dPre32 already carries the 1/n_states normalization (it was filled
-        # by the normalized step_loss backward) — no extra scaling here.
-        surrogate = (task.pre.float() * task.dPre32).sum()
+        # by the normalized step_loss backward).  inject_grad hands it to the
+        # precompute tensor without materializing a surrogate product.
+        from spacy_ray_amd.ops.api import inject_grad
+
+        surrogate = inject_grad(task.pre, task.dPre32)
         return surrogate, display
 
     def _step_loop(self, shards, t2v, train: bool):
