@@ -290,6 +290,51 @@ struct ArcEagerBatch {
     return py::make_tuple(act, feats, valid_a, gold_a);
   }
 
+  // Packed variant: ONE buffer = [feats int64 Sa*13][valid u8 Sa*A]
+  // [gold u8 Sa*A (train only)] so the python side does a single H2D copy
+  // per transition step (each pageable upload blocks the host ~0.1 ms).
+  py::tuple step_arrays_packed(bool with_gold, int64_t pad_row = -1) const {
+    std::vector<int32_t> idx;
+    idx.reserve(states.size());
+    for (size_t i = 0; i < states.size(); i++)
+      if (!states[i].final_state()) idx.push_back((int32_t)i);
+    const py::ssize_t Sa = (py::ssize_t)idx.size();
+    const py::ssize_t A = n_actions();
+    const size_t fbytes = (size_t)Sa * 13 * 8;
+    const size_t vbytes = (size_t)Sa * A;
+    py::array_t<int32_t> act(Sa);
+    py::array_t<uint8_t> packed((py::ssize_t)(fbytes + vbytes + (with_gold ? vbytes : 0)));
+    std::copy(idx.begin(), idx.end(), act.mutable_data());
+    uint8_t* base = packed.mutable_data();
+    int64_t* feats = (int64_t*)base;
+    uint8_t* valid_a = base + fbytes;
+    uint8_t* gold_a = valid_a + vbytes;
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
+#endif
+    for (py::ssize_t k = 0; k < Sa; k++) {
+      float crow[256];
+      int32_t f32[13];
+      size_t i = (size_t)idx[(size_t)k];
+      fill_features(i, f32);
+      int64_t* fo = feats + k * 13;
+      for (int q = 0; q < 13; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
+      uint8_t* v = valid_a + k * A;
+      if (with_gold) {
+        fill_costs(i, v, crow);
+        float cmin = KInvalid;
+        for (py::ssize_t a = 0; a < A; a++)
+          if (v[a] && crow[(size_t)a] < cmin) cmin = crow[(size_t)a];
+        uint8_t* g = gold_a + k * A;
+        for (py::ssize_t a = 0; a < A; a++)
+          g[a] = (v[a] && crow[(size_t)a] <= cmin + 1e-6f) ? 1 : 0;
+      } else {
+        fill_valid(v, states[i]);
+      }
+    }
+    return py::make_tuple(act, packed, (py::ssize_t)13);
+  }
+
   void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
     auto a = actions.unchecked<1>();
 #ifdef _OPENMP
@@ -501,6 +546,45 @@ struct BiluoBatch {
     return out;
   }
 
+  py::tuple step_arrays_packed(bool with_gold, int64_t pad_row = -1) const {
+    std::vector<int32_t> idx;
+    idx.reserve(states.size());
+    for (size_t i = 0; i < states.size(); i++)
+      if (!states[i].final_state()) idx.push_back((int32_t)i);
+    const py::ssize_t Sa = (py::ssize_t)idx.size();
+    const py::ssize_t A = n_actions();
+    const size_t fbytes = (size_t)Sa * 6 * 8;
+    const size_t vbytes = (size_t)Sa * A;
+    py::array_t<int32_t> act(Sa);
+    py::array_t<uint8_t> packed((py::ssize_t)(fbytes + vbytes + (with_gold ? vbytes : 0)));
+    std::copy(idx.begin(), idx.end(), act.mutable_data());
+    uint8_t* base = packed.mutable_data();
+    int64_t* feats = (int64_t*)base;
+    uint8_t* valid_a = base + fbytes;
+    uint8_t* gold_a = valid_a + vbytes;
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
+#endif
+    for (py::ssize_t k = 0; k < Sa; k++) {
+      size_t i = (size_t)idx[(size_t)k];
+      const NerState& st = states[i];
+      int32_t f32[6];
+      fill_features(i, f32);
+      int64_t* fo = feats + k * 6;
+      for (int q = 0; q < 6; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
+      uint8_t* v = valid_a + k * A;
+      fill_valid(v, st);
+      if (with_gold) {
+        uint8_t* g = gold_a + k * A;
+        int32_t gcode = st.final_state() ? -1 : gold[i][st.i];
+        bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
+        for (py::ssize_t a = 0; a < A; a++)
+          g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
+      }
+    }
+    return py::make_tuple(act, packed, (py::ssize_t)6);
+  }
+
   void advance(py::array_t<int32_t, py::array::c_style | py::array::forcecast> actions) {
     auto a = actions.unchecked<1>();
 #ifdef _OPENMP
@@ -552,6 +636,7 @@ void init_transitions(py::module_& m) {
       .def("costs", &ArcEagerBatch::costs)
       .def("advance", &ArcEagerBatch::advance)
       .def("step_arrays", &ArcEagerBatch::step_arrays, py::arg("with_gold"), py::arg("pad_row") = -1)
+      .def("step_arrays_packed", &ArcEagerBatch::step_arrays_packed, py::arg("with_gold"), py::arg("pad_row") = -1)
       .def("heads", &ArcEagerBatch::heads)
       .def("labels", &ArcEagerBatch::labels);
 
@@ -567,5 +652,6 @@ void init_transitions(py::module_& m) {
       .def("costs", &BiluoBatch::costs)
       .def("advance", &BiluoBatch::advance)
       .def("step_arrays", &BiluoBatch::step_arrays, py::arg("with_gold"), py::arg("pad_row") = -1)
+      .def("step_arrays_packed", &BiluoBatch::step_arrays_packed, py::arg("with_gold"), py::arg("pad_row") = -1)
       .def("tags", &BiluoBatch::tags);
 }

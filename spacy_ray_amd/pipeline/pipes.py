@@ -223,13 +223,19 @@ class _TransitionTask:
 
         device = self.t2v.device
         with timing.span("raw/states_cpu"):
-            # feats come back int64 with missing slots already remapped to
-            # the pad row (= T, the learned-pad index in precompute)
-            act_idx, feats, valid_np, gold_np = states.step_arrays(self.train, self.T)
-        if len(act_idx) == 0:
+            # ONE packed buffer = [feats int64 | valid u8 | gold u8]; feats
+            # missing slots already remapped to the pad row (= T, the
+            # learned-pad index in precompute); ONE H2D copy per step
+            act_idx, packed, nF = states.step_arrays_packed(self.train, self.T)
+        Sa = len(act_idx)
+        if Sa == 0:
             return None
+        A = states.n_actions
+        fbytes = Sa * nF * 8
         with timing.span("raw/score_fwd"):
-            feats_t = to_device(feats, device)
+            dev = to_device(packed, device)
+            feats_t = dev[:fbytes].view(torch.int64).view(Sa, nF)
+            valid_t = dev[fbytes:fbytes + Sa * A].view(Sa, A)
             if self.train:
                 hidden = _ops.parser_step_score_accum(
                     self.pre_d, feats_t, self.pipe.module.lower_b,
@@ -238,10 +244,9 @@ class _TransitionTask:
                 scores = self.pipe.module.upper(hidden)  # [S_active, A]
             else:
                 scores = self.pipe.module.score(self.pre_d, feats_t)
-            valid_t = to_device(valid_np, device)
         if self.train:
             with timing.span("raw/loss_build"):
-                gold_t = to_device(gold_np, device)
+                gold_t = dev[fbytes + Sa * A:].view(Sa, A)
                 self.score_chunks.append(scores)
                 self.gold_chunks.append(gold_t)
                 self.valid_chunks.append(valid_t)
@@ -252,7 +257,8 @@ class _TransitionTask:
             actions_dev = self.hip.action_select(scores.detach(), sel_gold, sel_valid)
             return act_idx, actions_dev, None
         s_np = scores.detach().float().cpu().numpy()
-        g_np = gold_np if self.train else valid_np
+        valid_np = packed[fbytes:fbytes + Sa * A].reshape(Sa, A)
+        g_np = (packed[fbytes + Sa * A:].reshape(Sa, A) if self.train else valid_np)
         choose = np.where(g_np > 0, s_np, NEG_INF)
         fallback = np.where(valid_np > 0, s_np, NEG_INF)
         any_gold = (g_np > 0).any(axis=1, keepdims=True)
