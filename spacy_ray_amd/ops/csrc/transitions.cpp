@@ -53,7 +53,6 @@ struct ParserState {
   std::vector<int32_t> label;   // -1 = none
   // children bookkeeping for features (two leftmost / two rightmost)
   std::vector<int32_t> l1, l2, r1, r2;
-  std::vector<int32_t> n_head_in_stack_cache;  // unused; placeholder
 
   void init(int32_t n) {
     len = n;

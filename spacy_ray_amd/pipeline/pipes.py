@@ -313,12 +313,16 @@ def run_transition_tasks(tasks: List[_TransitionTask]) -> None:
 
 
 class _TransitionPipeBase(TrainablePipe):
-    """Shared greedy transition loop for parser and NER: per step, the C++
-    batch object yields features/valid/costs for ALL states; the GPU scores
-    them (fused gather+maxout + upper GEMM); training follows the
-    best-scoring min-cost action; loss = CE of softmax-over-valid against the
-    uniform min-cost target (contract of spaCy's parser loss,
-    SURVEY.md §2.2 N8)."""
+    """Shared greedy transition machinery for parser and NER.
+
+    Per step the C++ batch object returns ONE packed buffer (features +
+    valid + min-cost masks for the ACTIVE states); the GPU scores them
+    (fused gather+maxout + upper GEMM) and selects actions on-device;
+    training follows the best-scoring min-cost action; the loss is one CE
+    of softmax-over-valid against the uniform min-cost target computed over
+    all steps at once (contract of spaCy's parser loss).
+    See _TransitionTask for the two-phase backward and the task/shard
+    pipelining."""
 
     listens_to = "tok2vec"
 
