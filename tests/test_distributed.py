@@ -98,6 +98,40 @@ def test_grad_accumulation_equivalence():
     assert torch.allclose(g12, g1 + g2, atol=1e-5)
 
 
+def test_accumulate_gradient_microbatching():
+    """accumulate_gradient=2 splits each batch into 2 sub-batches whose
+    gradients accumulate before one optimizer step (the key the reference
+    reads but never wires — SURVEY.md §2.3)."""
+    from spacy_ray_amd.train.loop import train_while_improving
+
+    nlp, T, examples = _make_nlp_and_examples(8)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    calls = []
+    orig_acc, orig_apply = engine.accumulate, engine.apply_step
+
+    def acc(sub, drop=0.0, losses=None, sync=True):
+        calls.append(("acc", len(sub), sync))
+        orig_acc(sub, drop=drop, losses=losses, sync=sync)
+
+    def apply():
+        calls.append(("apply",))
+        orig_apply()
+
+    engine.accumulate, engine.apply_step = acc, apply
+    data = iter([(0, examples)])
+    it = train_while_improving(
+        nlp, engine, data, evaluate=lambda: (0.0, {}), dropout=0.0,
+        accumulate_gradient=2, max_steps=1, eval_frequency=100,
+    )
+    for _ in it:
+        pass
+    accs = [c for c in calls if c[0] == "acc"]
+    assert len(accs) == 2
+    assert accs[0][2] is False and accs[1][2] is True  # sync only on last
+    assert sum(c[1] for c in accs) == len(examples)
+    assert calls[-1] == ("apply",) or ("apply",) in calls
+
+
 _WORKER_SCRIPT = r"""
 import json, os, sys
 import numpy as np
