@@ -210,10 +210,13 @@ def resolve(section: Dict[str, Any], *, schema=None, validate: bool = True) -> A
         return node
 
     out = _resolve_node(dict(section))
-    if schema is not None and validate:
-        out = schema(**out).model_dump() if hasattr(schema, "model_dump") or hasattr(schema, "model_fields") else out
-        if hasattr(out, "items"):
-            out = dict(out)
+    if schema is not None and validate and isinstance(out, dict):
+        # pydantic-typed validation (spaCy's registry.resolve(schema=...)
+        # contract); resolved callables/objects pass through unvalidated
+        model = schema(**out)
+        validated = {k: getattr(model, k) for k in model.model_fields}
+        extra = model.model_extra or {}
+        out = {**validated, **extra}
     return out
 
 

@@ -86,7 +86,9 @@ def distributed_train(
 
     comm = init_comm_from_env()
     icfg = config.interpolate()
-    T = resolve(icfg["training"], validate=False)
+    from spacy_ray_amd.config.schemas import ConfigSchemaTraining
+
+    T = resolve(icfg["training"], schema=ConfigSchemaTraining)
 
     nlp = init_nlp(config, device=device)
     _check_param_manifest(nlp, comm)

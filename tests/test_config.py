@@ -74,3 +74,23 @@ def test_resolve_registry_block():
 def test_bad_override():
     with pytest.raises(ValueError):
         parse_config_overrides(["positional"])
+
+
+def test_training_schema_validation():
+    from spacy_ray_amd.config.schemas import ConfigSchemaTraining
+
+    out = resolve(
+        {"dropout": 0.2, "max_steps": 5, "custom_extra": 1,
+         "optimizer": {"@optimizers": "Adam.v1", "learn_rate": 0.01}},
+        schema=ConfigSchemaTraining,
+    )
+    assert out["dropout"] == 0.2
+    assert out["accumulate_gradient"] == 1  # default filled
+    assert out["custom_extra"] == 1         # extras pass through
+    from spacy_ray_amd.train.optimizer import AdamSpec
+
+    assert isinstance(out["optimizer"], AdamSpec)
+    with pytest.raises(Exception):
+        resolve({"accumulate_gradient": 0}, schema=ConfigSchemaTraining)
+    with pytest.raises(Exception):
+        resolve({"dropout": 1.5}, schema=ConfigSchemaTraining)
