@@ -214,7 +214,7 @@ def resolve(section: Dict[str, Any], *, schema=None, validate: bool = True) -> A
         # pydantic-typed validation (spaCy's registry.resolve(schema=...)
         # contract); resolved callables/objects pass through unvalidated
         model = schema(**out)
-        validated = {k: getattr(model, k) for k in model.model_fields}
+        validated = {k: getattr(model, k) for k in type(model).model_fields}
         extra = model.model_extra or {}
         out = {**validated, **extra}
     return out
