@@ -31,16 +31,45 @@ from spacy_ray_amd.train.scorer import weighted_score
 def _shard_corpus(corpus, rank: int, world: int):
     """Explicit rank::world interleave so an epoch is a true partition
     (improves on the reference, where every worker iterates the full corpus —
-    SURVEY.md §2.3 DP row)."""
+    SURVEY.md §2.3 DP row).  Every rank yields exactly ceil(n/world)
+    examples (wrapping on the remainder) so per-rank epochs stay aligned."""
     if world <= 1:
         return corpus
 
     def sharded(nlp) -> Iterator:
-        for i, eg in enumerate(corpus(nlp)):
-            if i % world == rank:
-                yield eg
+        egs = list(corpus(nlp))
+        n = len(egs)
+        if n == 0:
+            return
+        per = -(-n // world)
+        for j in range(per):
+            yield egs[(j * world + rank) % n]
 
     return sharded
+
+
+def _synced_batches(batches, comm: Comm, device) -> Iterator:
+    """Stop ALL ranks as soon as ANY rank's batch stream ends.  Word-count
+    batching can split a sharded epoch into different batch counts per rank;
+    without this agreement one rank exits the loop while the others wait in
+    the eval broadcast (finite-epoch hang).  One tiny all-reduce per step."""
+    if comm.world <= 1:
+        yield from batches
+        return
+    it = iter(batches)
+    dev = device if str(device).startswith("cuda") else "cpu"
+    while True:
+        try:
+            item = next(it)
+            has = 1.0
+        except StopIteration:
+            item = None
+            has = 0.0
+        flag = torch.tensor([has], device=dev)
+        comm.all_reduce_(flag)
+        if float(flag.item()) < comm.world:
+            return
+        yield item
 
 
 def _check_param_manifest(nlp, comm: Comm) -> None:
@@ -131,7 +160,10 @@ def distributed_train(
             payload = comm.broadcast_obj(payload, src=0)
         return payload
 
-    batches = create_train_batches(nlp, train_corpus, T["batcher"], T.get("max_epochs", 0) or 0)
+    batches = _synced_batches(
+        create_train_batches(nlp, train_corpus, T["batcher"], T.get("max_epochs", 0) or 0),
+        comm, device,
+    )
     if rank == 0:
         logger_setup = T.get("logger")
         if logger_setup is None:
@@ -167,7 +199,6 @@ def distributed_train(
     fault = os.environ.get("SRX_FAULT_INJECT")
     fault_rank, fault_step = (int(x) for x in fault.split(":")) if fault else (-1, -1)
 
-    best_score = None
     for batch, info, is_best_checkpoint in step_iter:
         if rank == fault_rank and info["step"] == fault_step:
             raise SystemExit(41)

@@ -54,6 +54,20 @@ def test_launcher_aborts_all_on_rank_failure(tmp_path):
     assert r.returncode != 0  # supervisor propagates the failure
 
 
+def test_finite_epochs_uneven_shards_no_hang(tmp_path):
+    """2 workers, odd corpus size, finite epochs: per-rank batch counts can
+    differ — the per-step agreement collective must end training cleanly on
+    every rank (regression test for the finite-epoch hang)."""
+    out = tmp_path / "outfe"
+    r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
+                  "--corpora.train.n_docs", "401",
+                  "--training.max_epochs", "2",
+                  "--training.max_steps", "0",
+                  "--training.eval_frequency", "5"], timeout=420)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert (out / "model-last" / "config.cfg").exists()
+
+
 def test_resume_continues_from_checkpoint(tmp_path):
     out = tmp_path / "outr"
     r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
