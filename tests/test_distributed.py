@@ -98,6 +98,26 @@ def test_grad_accumulation_equivalence():
     assert torch.allclose(g12, g1 + g2, atol=1e-5)
 
 
+def test_flat_buffer_layout_invariants():
+    """Every param starts on a 64-element boundary (hipBLASLt alignment),
+    buckets tile the flat buffer, and world divides every bucket."""
+    nlp, T, _ = _make_nlp_and_examples(2)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    for off, (name, p) in zip(engine._offsets,
+                              [(n, p) for n, p in engine.module.named_parameters()
+                               if p.requires_grad]):
+        assert off % 64 == 0, name
+        # the param's data must be a view into the flat buffer at `off`
+        assert p.data.data_ptr() == engine.flat_param.data_ptr() + off * engine.dtype.itemsize
+    prev_end = 0
+    for b in engine.buckets:
+        assert b.start == prev_end
+        assert (b.end - b.start) % (64 * engine.comm.world) == 0
+        prev_end = b.end
+    assert prev_end == engine.flat_param.numel()
+    assert engine.shard_elems == sum(b.per for b in engine.buckets)
+
+
 def test_accumulate_gradient_microbatching():
     """accumulate_gradient=2 splits each batch into 2 sub-batches whose
     gradients accumulate before one optimizer step (the key the reference
