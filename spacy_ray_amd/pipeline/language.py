@@ -78,6 +78,9 @@ class Language:
         t2v_pipe = self.tok2vec
         with timing.phase("fwd/tok2vec"):
             t2v = t2v_pipe.forward(batch, drop=drop) if t2v_pipe is not None else None
+        if t2v is not None and t2v_pipe.name in self._frozen:
+            # frozen tok2vec: listeners may read it but must not update it
+            t2v = t2v.detach()
         total = None
         # transition pipes (parser/NER) run INTERLEAVED: their per-step
         # GPU-score / CPU-advance phases pipeline against each other
