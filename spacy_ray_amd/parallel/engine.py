@@ -23,7 +23,6 @@ over single contiguous tensors (one fused HIP kernel in ops/kernels).
 """
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional
 
 import torch
@@ -280,6 +279,16 @@ class ZeRO1Engine:
         self.step_count += 1
 
     # ------------------------------------------------- checkpoint interface
+    def refresh_master_from_params(self) -> None:
+        """Re-snapshot the fp32 master from the (possibly just-loaded) model
+        params.  Needed when params were loaded from disk AFTER engine
+        construction without a matching optimizer-state file — otherwise the
+        stale master would overwrite the loaded weights on the first step."""
+        r = self.comm.rank
+        for bkt in self.buckets:
+            src = self.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
+            self.master[bkt.shard_off : bkt.shard_off + bkt.per].copy_(src.float())
+
     def state_dict(self) -> Dict:
         return {
             "step": self.step_count,

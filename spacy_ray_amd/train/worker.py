@@ -135,6 +135,10 @@ def distributed_train(
         opt_state = Path(output_path) / "model-last" / f"optim.rank{rank}.pt"
         if opt_state.exists():
             engine.load_state_dict(torch.load(opt_state, map_location=device))
+        else:
+            # params loaded but no optimizer shard: re-snapshot the master
+            # so the first step doesn't revert to pre-load weights
+            engine.refresh_master_from_params()
 
     # before_to_disk: optional @callbacks hook applied to nlp before saving
     # (contract of create_before_to_disk_callback at

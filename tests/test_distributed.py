@@ -98,6 +98,23 @@ def test_grad_accumulation_equivalence():
     assert torch.allclose(g12, g1 + g2, atol=1e-5)
 
 
+def test_refresh_master_after_external_param_load():
+    """Loading params into the flat views after engine construction must be
+    followed by refresh_master_from_params, or the stale master would revert
+    them on the first step (resume-without-optimizer-state path)."""
+    nlp, T, examples = _make_nlp_and_examples(4)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    # simulate an external load: overwrite one param in place
+    p = next(iter(engine.module.parameters()))
+    with torch.no_grad():
+        p.add_(1.0)
+    engine.refresh_master_from_params()
+    engine.accumulate(examples)
+    engine.apply_step()
+    # the +1 offset must have survived (master was refreshed, not stale)
+    assert float(p.detach().abs().mean()) > 0.5
+
+
 def test_flat_buffer_layout_invariants():
     """Every param starts on a 64-element boundary (hipBLASLt alignment),
     buckets tile the flat buffer, and world divides every bucket."""
