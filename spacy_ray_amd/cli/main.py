@@ -101,6 +101,33 @@ def ray_train(
                           master_port=master_port)
 
 
+@ray_app.command("evaluate")
+def ray_evaluate_cli(
+    model_path: Path = typer.Argument(..., help="Trained pipeline directory (model-best/model-last)"),
+    use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
+    corpus: Optional[str] = typer.Option(None, "--corpus", help="Dotted corpus name in the model's config (default: the training dev corpus)"),
+):
+    """Evaluate a saved pipeline on its dev corpus and print scores JSON."""
+    import torch
+
+    from spacy_ray_amd.config.config import resolve, resolve_dot_names
+    from spacy_ray_amd.pipeline.language import build_nlp
+    from spacy_ray_amd.train.scorer import weighted_score
+
+    device = f"cuda:{max(use_gpu, 0)}" if (use_gpu >= 0 and torch.cuda.is_available()) else "cpu"
+    config = Config.from_disk(model_path / "config.cfg")
+    nlp = build_nlp(config, device=device)
+    nlp.from_disk(model_path)
+    icfg = config.interpolate()
+    T = resolve(icfg["training"])
+    dot = corpus or T.get("dev_corpus", "corpora.dev")
+    (dev_corpus,) = resolve_dot_names(icfg, [dot])
+    examples = list(dev_corpus(nlp))
+    scores = nlp.evaluate(examples)
+    scores["score"] = weighted_score(scores, T.get("score_weights") or {})
+    print(json.dumps(scores, indent=2))
+
+
 def main() -> None:
     app()
 

@@ -81,6 +81,23 @@ def test_resume_continues_from_checkpoint(tmp_path):
     assert before != after  # training continued and re-saved
 
 
+def test_cli_evaluate(tmp_path):
+    out = tmp_path / "oute"
+    r = _run_cli([str(CFG), "--output", str(out), "--training.max_steps", "6",
+                  "--training.eval_frequency", "3"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    r = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "ray", "evaluate",
+         str(out / "model-last")],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    scores = json.loads(r.stdout)
+    assert "tag_acc" in scores and "score" in scores
+
+
 def test_overrides_reach_workers(tmp_path):
     out = tmp_path / "out3"
     r = _run_cli([str(CFG), "--n-workers", "2", "--output", str(out),
