@@ -102,9 +102,14 @@ def main() -> None:
         vocab_size=args.vocab_size,
     )
     words_per_step = sum(len(eg) for eg in batches[0])
+    # fixed replayed batches: precompute the device-side TokenBatch once
+    from spacy_ray_amd.models.batch import TokenBatch
+
+    token_batches = [TokenBatch([eg.predicted for eg in b], device) for b in batches]
 
     def step(i: int) -> None:
-        engine.accumulate(batches[i % len(batches)], drop=0.1)
+        k = i % len(batches)
+        engine.accumulate(batches[k], drop=0.1, token_batch=token_batches[k])
         engine.apply_step()
 
     if args.profile_steps:

@@ -188,14 +188,15 @@ class ZeRO1Engine:
 
     # ------------------------------------------------------------- stepper
     def accumulate(self, examples, drop: float = 0.0, losses: Optional[Dict] = None,
-                   sync: bool = True) -> None:
+                   sync: bool = True, token_batch=None) -> None:
         import time as _time
 
         from spacy_ray_amd.utils import timing
 
         t0 = _time.perf_counter()
         self._sync = sync
-        total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop)
+        total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop,
+                                         token_batch=token_batch)
         with timing.phase("bwd/main"):
             total.backward()
         self._sync = False

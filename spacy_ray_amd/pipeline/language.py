@@ -64,15 +64,19 @@ class Language:
 
     # ------------------------------------------------------------- training
     def forward_loss(self, examples: Sequence[Example], losses: Optional[Dict[str, float]] = None,
-                     drop: float = 0.0):
-        """One forward pass over all trainable pipes -> (total_loss, losses)."""
+                     drop: float = 0.0, token_batch: Optional[TokenBatch] = None):
+        """One forward pass over all trainable pipes -> (total_loss, losses).
+
+        `token_batch`: optional precomputed TokenBatch for these examples —
+        callers replaying fixed batches (bench) skip the per-step
+        concat+upload."""
         from spacy_ray_amd.utils import timing
 
         if losses is None:
             losses = {}
         with timing.phase("data/batch_build"):
-            docs = [eg.predicted for eg in examples]
-            batch = TokenBatch(docs, self.device)
+            batch = token_batch if token_batch is not None else TokenBatch(
+                [eg.predicted for eg in examples], self.device)
         from .pipes import _TransitionPipeBase, run_transition_tasks
 
         t2v_pipe = self.tok2vec
