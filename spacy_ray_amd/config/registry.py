@@ -8,7 +8,7 @@ of the same contract: named registries, decorator registration, and lookup.
 from __future__ import annotations
 
 import importlib
-from typing import Any, Callable, Dict
+from typing import Callable, Dict
 
 
 class Registry:

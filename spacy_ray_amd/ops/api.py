@@ -9,7 +9,7 @@ contract).  Set SRX_ALLOW_TORCH_FALLBACK=1 only for bring-up/debugging.
 from __future__ import annotations
 
 import os
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

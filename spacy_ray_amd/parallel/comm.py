@@ -16,7 +16,6 @@ from __future__ import annotations
 
 import datetime
 import os
-import pickle
 from typing import Any, Optional
 
 import torch

@@ -9,10 +9,8 @@ restart from the last checkpoint)."""
 from __future__ import annotations
 
 import os
-import signal
 import socket
 import subprocess
-import sys
 import time
 from typing import Dict, List, Optional
 

@@ -17,7 +17,7 @@ import torch.nn as nn
 from spacy_ray_amd import _srx_cpu
 from spacy_ray_amd.models.batch import TokenBatch
 from spacy_ray_amd.models.parser_model import TransitionModel
-from spacy_ray_amd.vocab.doc import Doc, Example, biluo_to_codes, codes_to_biluo
+from spacy_ray_amd.vocab.doc import Example, biluo_to_codes, codes_to_biluo
 
 NEG_INF = -1e30
 
@@ -181,8 +181,9 @@ class _TransitionTask:
     1. TWO-PHASE backward: per-step losses (upper GEMM + bias) are
        backpropagated in finish_task while each step's dPre scatters into
        ONE persistent fp32 buffer (ops.parser_step_score_accum); the loss
-       returned upward is a surrogate (pre · dPre).sum() that hands exactly
-       dPre to the precompute GEMM in the caller's single main backward.
+       returned upward is a zero-valued gradient-injection node
+       (ops.inject_grad) that hands exactly dPre to the precompute GEMM in
+       the caller's single main backward.
     2. TASK/SHARD PIPELINING (run_transition_tasks): while one unit's
        actions travel GPU->CPU and its C++ state machine advances, another
        unit's scoring kernels run — parser and NER loops interleave so the
