@@ -125,6 +125,11 @@ class ZeRO1Engine:
         self.grad_shard = torch.zeros(self.shard_elems, dtype=self.dtype, device=self.device)
         self._g32 = torch.zeros_like(self.master)
         self._param_shard: Optional[torch.Tensor] = None
+        # running parameter average (thinc Adam use_averages contract):
+        # updated per step, swapped in around evaluation via averaged_params()
+        self.avg: Optional[torch.Tensor] = (
+            self.master.clone() if getattr(spec, "use_averages", False) else None
+        )
 
         # ---- overlap machinery
         self.comm_stream = torch.cuda.Stream() if self.is_cuda else None
@@ -278,6 +283,43 @@ class ZeRO1Engine:
         self.flat_grad.zero_()
         self.grad_shard.zero_()
         self.step_count += 1
+        if self.avg is not None:
+            # running mean over steps (thinc's averaged weights for eval)
+            self.avg.add_(
+                (new_param.float() if new_param.dtype != torch.float32 else new_param)
+                - self.avg,
+                alpha=1.0 / self.step_count,
+            )
+
+    def averaged_params(self):
+        """Context manager: swap the running parameter AVERAGE into the live
+        flat buffer (all-gathered) for evaluation, restore after (thinc
+        ``use_averages`` contract).  No-op when use_averages is off."""
+        import contextlib
+
+        engine = self
+
+        @contextlib.contextmanager
+        def ctx():
+            if engine.avg is None:
+                yield
+                return
+            r = engine.comm.rank
+            for bkt in engine.buckets:
+                dst = engine.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
+                dst.copy_(engine.avg[bkt.shard_off : bkt.shard_off + bkt.per].to(engine.dtype))
+            for bkt in engine.buckets:
+                engine._gather_bucket(bkt)
+            try:
+                yield
+            finally:
+                for bkt in engine.buckets:
+                    dst = engine.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]
+                    dst.copy_(engine.master[bkt.shard_off : bkt.shard_off + bkt.per].to(engine.dtype))
+                for bkt in engine.buckets:
+                    engine._gather_bucket(bkt)
+
+        return ctx()
 
     # ------------------------------------------------- checkpoint interface
     def refresh_master_from_params(self) -> None:

@@ -149,17 +149,20 @@ def distributed_train(
 
     def evaluate():
         nonlocal dev_examples
-        if rank == 0:
-            if dev_examples is None:
-                dev_examples = list(dev_corpus(nlp))
-            t_eval = time.time()
-            scores = nlp.evaluate(dev_examples)
-            dt = max(1e-9, time.time() - t_eval)
-            scores["speed"] = sum(len(eg) for eg in dev_examples) / dt
-            score = weighted_score(scores, T.get("score_weights") or {})
-            payload = (score, scores)
-        else:
-            payload = None
+        # use_averages: evaluate with the running parameter average swapped
+        # in.  ALL ranks enter the context (the swap all-gathers).
+        with engine.averaged_params():
+            if rank == 0:
+                if dev_examples is None:
+                    dev_examples = list(dev_corpus(nlp))
+                t_eval = time.time()
+                scores = nlp.evaluate(dev_examples)
+                dt = max(1e-9, time.time() - t_eval)
+                scores["speed"] = sum(len(eg) for eg in dev_examples) / dt
+                score = weighted_score(scores, T.get("score_weights") or {})
+                payload = (score, scores)
+            else:
+                payload = None
         if world > 1:
             payload = comm.broadcast_obj(payload, src=0)
         return payload

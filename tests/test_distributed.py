@@ -115,6 +115,31 @@ def test_refresh_master_after_external_param_load():
     assert float(p.detach().abs().mean()) > 0.5
 
 
+def test_use_averages_swap():
+    """With use_averages, averaged_params() swaps the running mean into the
+    live params and restores the trained params after."""
+    nlp, T, examples = _make_nlp_and_examples(4)
+    spec = T["optimizer"]
+    spec.use_averages = True
+    engine = ZeRO1Engine(nlp, spec, LocalComm())
+    for _ in range(3):
+        engine.accumulate(examples)
+        engine.apply_step()
+    trained = engine.flat_param.clone()
+    with engine.averaged_params():
+        swapped = engine.flat_param.clone()
+        assert not torch.equal(trained, swapped)  # average != latest
+    assert torch.allclose(engine.flat_param, trained)  # restored
+    # no averaging configured -> context is a no-op
+    spec2 = T["optimizer"]
+    spec2.use_averages = False
+    nlp2, T2, _ = _make_nlp_and_examples(2)
+    engine2 = ZeRO1Engine(nlp2, T2["optimizer"], LocalComm())
+    before = engine2.flat_param.clone()
+    with engine2.averaged_params():
+        assert torch.equal(engine2.flat_param, before)
+
+
 def test_flat_buffer_layout_invariants():
     """Every param starts on a 64-element boundary (hipBLASLt alignment),
     buckets tile the flat buffer, and world divides every bucket."""
