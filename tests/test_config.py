@@ -76,6 +76,30 @@ def test_bad_override():
         parse_config_overrides(["positional"])
 
 
+def test_interpolation_cycle_detected():
+    cfg = Config.from_str("""
+[a]
+x = ${b.y}
+
+[b]
+y = ${a.x}
+""")
+    with pytest.raises(ValueError):
+        cfg.interpolate()
+
+
+def test_registry_unknown_name_errors():
+    from spacy_ray_amd.config.registry import registry
+
+    registry.ensure_populated()
+    with pytest.raises(KeyError):
+        registry.architectures.get("no.such.arch.v1")
+    with pytest.raises(KeyError):
+        registry.get_registry("nosuchregistry")
+    with pytest.raises(ValueError):
+        resolve({"@optimizers": "Adam.v1", "@schedules": "constant.v1"})
+
+
 def test_training_schema_validation():
     from spacy_ray_amd.config.schemas import ConfigSchemaTraining
 
