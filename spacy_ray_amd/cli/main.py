@@ -40,6 +40,8 @@ def ray_train_cli(
     use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
     verbose: bool = typer.Option(False, "--verbose", "-V", help="Display more information"),
     resume: bool = typer.Option(False, "--resume", help="Resume from <output>/model-last (params + per-rank optimizer shards)"),
+    nnodes: int = typer.Option(1, "--nnodes", help="Number of nodes (with --address on every node)"),
+    node_rank: int = typer.Option(0, "--node-rank", help="This node's rank in [0, nnodes)"),
 ):
     """Train a pipeline with N data-parallel workers over RCCL/xGMI."""
     logging.basicConfig(level=logging.DEBUG if verbose else logging.ERROR)
@@ -48,7 +50,8 @@ def ray_train_cli(
     raise SystemExit(
         ray_train(config, config_path=config_path, output_path=output_path,
                   code_path=code_path, n_workers=n_workers, address=address,
-                  use_gpu=use_gpu, overrides=overrides, resume=resume)
+                  use_gpu=use_gpu, overrides=overrides, resume=resume,
+                  nnodes=nnodes, node_rank=node_rank)
     )
 
 
@@ -63,13 +66,15 @@ def ray_train(
     use_gpu: int = -1,
     overrides: Optional[dict] = None,
     resume: bool = False,
+    nnodes: int = 1,
+    node_rank: int = 0,
 ) -> int:
     """Launcher (contract of `/root/reference/spacy_ray/train_cli.py:56-91`)."""
     from spacy_ray_amd.parallel.launcher import launch_workers
 
     if output_path:
         Path(output_path).mkdir(parents=True, exist_ok=True)
-    if n_workers <= 1 and not address:
+    if n_workers <= 1 and not address and nnodes <= 1:
         # single process: run in-process, no process group
         from spacy_ray_amd.train.worker import distributed_train
 
@@ -97,8 +102,11 @@ def ray_train(
         host, _, port = address.partition(":")
         master_addr = host or "127.0.0.1"
         master_port = int(port) if port else None
+    if nnodes > 1 and master_port is None:
+        raise SystemExit("--nnodes > 1 requires --address host:port (a fixed rendezvous port)")
     return launch_workers(worker_cmd, n_workers, master_addr=master_addr,
-                          master_port=master_port)
+                          master_port=master_port, nnodes=nnodes,
+                          node_rank=node_rank)
 
 
 @ray_app.command("evaluate")

@@ -27,19 +27,24 @@ def launch_workers(
     *,
     master_addr: str = "127.0.0.1",
     master_port: Optional[int] = None,
+    nnodes: int = 1,
+    node_rank: int = 0,
     env_extra: Optional[Dict[str, str]] = None,
     poll_interval: float = 1.0,
 ) -> int:
-    """Spawn n ranked copies of worker_cmd; supervise; return exit code."""
+    """Spawn n_workers ranked processes on THIS node; supervise; return exit
+    code.  Multi-node: run the same command on every node with the same
+    --address (rank-0 node's host:port) and the node's --node-rank; global
+    RANK = node_rank * n_workers + local."""
     port = master_port or find_free_port()
     procs: List[subprocess.Popen] = []
-    for rank in range(n_workers):
+    for local in range(n_workers):
         env = dict(os.environ)
         env.update(env_extra or {})
         env.update(
-            RANK=str(rank),
-            LOCAL_RANK=str(rank),
-            WORLD_SIZE=str(n_workers),
+            RANK=str(node_rank * n_workers + local),
+            LOCAL_RANK=str(local),
+            WORLD_SIZE=str(n_workers * nnodes),
             MASTER_ADDR=master_addr,
             MASTER_PORT=str(port),
         )
