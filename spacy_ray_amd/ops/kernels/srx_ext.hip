@@ -5,6 +5,11 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+#include <mutex>
+#include <unordered_map>
+#include <utility>
+#include <vector>
+
 #include "srx_common.hip.h"
 #include "srx_elementwise.hip.h"
 #include "srx_embed_parser.hip.h"
@@ -506,6 +511,92 @@ std::vector<at::Tensor> mwe_layer_fwd(at::Tensor X, at::Tensor Wt, at::Tensor bi
   return {Y, Mout, which, mu, rstd};
 }
 
+// --------------------------------------------- fused transition step (C++)
+// One python call per transition step: state scorer kernel + upper GEMM +
+// on-device action selection, with a C++ autograd node so the per-step
+// python op count collapses (~10 -> 1).  Backward stashes the compact
+// (feats, dSummed) pair into a session store keyed by task id; the python
+// side drains it for the batched sorted dPre scatter.
+// NOTE: backward runs on autograd worker threads WITHOUT the GIL — only
+// ATen ops and the mutex-guarded store are touched there.
+namespace fusedstep {
+
+struct Store {
+  std::mutex mu;
+  std::unordered_map<int64_t, std::vector<std::pair<at::Tensor, at::Tensor>>> m;
+};
+Store& store() {
+  static Store s;
+  return s;
+}
+
+class Fn : public torch::autograd::Function<Fn> {
+ public:
+  static torch::autograd::variable_list forward(
+      torch::autograd::AutogradContext* ctx, at::Tensor pre_d, at::Tensor feats,
+      at::Tensor lower_b, at::Tensor upperW, at::Tensor upperB,
+      at::Tensor is_gold, at::Tensor valid, int64_t task_id, bool train) {
+    auto hw = parser_step_fwd(pre_d, feats, lower_b);
+    auto hidden = hw[0];
+    auto which = hw[1];
+    auto scores = at::addmm(upperB, hidden, upperW.t());
+    auto actions = action_select(scores, is_gold, valid);
+    ctx->save_for_backward({feats, which, hidden, upperW});
+    ctx->saved_data["task_id"] = task_id;
+    ctx->saved_data["train"] = train;
+    ctx->mark_non_differentiable({actions});
+    return {scores, actions};
+  }
+
+  static torch::autograd::variable_list backward(
+      torch::autograd::AutogradContext* ctx, torch::autograd::variable_list grads) {
+    auto saved = ctx->get_saved_variables();
+    auto feats = saved[0];
+    auto which = saved[1];
+    auto hidden = saved[2];
+    auto upperW = saved[3];
+    auto dScores = grads[0].contiguous();
+    auto dUpperW = dScores.t().mm(hidden);
+    auto dUpperB = dScores.sum(0);
+    auto dHidden = dScores.mm(upperW).contiguous();
+    auto dSummed = maxout_bwd(dHidden, which, 2).view({dHidden.size(0), -1});
+    auto dLowerB = dSummed.to(at::kFloat).sum(0);  // mirrors the python accum path
+    int64_t task_id = ctx->saved_data["task_id"].toInt();
+    {
+      auto& s = store();
+      std::lock_guard<std::mutex> lock(s.mu);
+      s.m[task_id].emplace_back(feats, dSummed);
+    }
+    // pre_d is detached by contract (its gradient flows via the batched
+    // scatter + inject_grad); feats/is_gold/valid are integer masks.
+    return {at::Tensor(), at::Tensor(), dLowerB, dUpperW, dUpperB,
+            at::Tensor(), at::Tensor(), at::Tensor(), at::Tensor()};
+  }
+};
+
+std::vector<at::Tensor> fused_step(at::Tensor pre_d, at::Tensor feats,
+                                   at::Tensor lower_b, at::Tensor upperW,
+                                   at::Tensor upperB, at::Tensor is_gold,
+                                   at::Tensor valid, int64_t task_id, bool train) {
+  auto out = Fn::apply(pre_d, feats, lower_b, upperW, upperB, is_gold, valid,
+                       task_id, train);
+  return {out[0], out[1]};
+}
+
+std::vector<std::vector<at::Tensor>> fused_entries_take(int64_t task_id) {
+  auto& s = store();
+  std::lock_guard<std::mutex> lock(s.mu);
+  std::vector<std::vector<at::Tensor>> out;
+  auto it = s.m.find(task_id);
+  if (it != s.m.end()) {
+    for (auto& pr : it->second) out.push_back({pr.first, pr.second});
+    s.m.erase(it);
+  }
+  return out;
+}
+
+}  // namespace fusedstep
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -528,4 +619,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_ragged_bwd", &reduce_ragged_bwd);
   m.def("reduce_max_bwd", &reduce_max_bwd);
   m.def("mwe_layer_fwd", &mwe_layer_fwd);
+  m.def("fused_step", &fusedstep::fused_step);
+  m.def("fused_entries_take", &fusedstep::fused_entries_take);
 }

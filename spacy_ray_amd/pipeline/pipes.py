@@ -8,6 +8,8 @@ overlap with.
 """
 from __future__ import annotations
 
+import itertools
+import os
 from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
@@ -172,6 +174,9 @@ class TaggerPipe(TrainablePipe):
             off += n
 
 
+_next_task_id = itertools.count(1).__next__
+
+
 class _TransitionTask:
     """One pipe's transition-loop context: states shards, the detached
     precompute tensor, the persistent dPre gradient accumulator and the
@@ -191,7 +196,8 @@ class _TransitionTask:
 
     __slots__ = ("pipe", "shards", "t2v", "train", "T", "pre", "pre_d",
                  "dPre32", "hip", "score_chunks", "gold_chunks",
-                 "valid_chunks", "n_states_total", "entries")
+                 "valid_chunks", "n_states_total", "entries", "fused",
+                 "task_id")
 
     def __init__(self, pipe, shards, t2v, train: bool) -> None:
         from spacy_ray_amd.ops import api as _ops
@@ -212,6 +218,16 @@ class _TransitionTask:
         # GPU train: per-step backwards append (feats, dSummed) here; the
         # scatter happens once, batched+sorted, in finish_task
         self.entries: Optional[List] = [] if (train and self.hip is not None) else None
+        # fully-fused C++ step (scorer kernel + upper GEMM + action select in
+        # ONE python call, backward entries stashed in a C++ store); default
+        # on — GPU-validated vs the python path (test_fused_step_parity);
+        # SRX_FUSED_STEP=0 falls back.  See fusedstep in srx_ext.hip.
+        self.fused = (
+            train and self.hip is not None
+            and hasattr(self.hip, "fused_step")
+            and os.environ.get("SRX_FUSED_STEP", "1") == "1"
+        )
+        self.task_id = _next_task_id() if self.fused else 0
         self.score_chunks: List[torch.Tensor] = []
         self.gold_chunks: List[torch.Tensor] = []
         self.valid_chunks: List[torch.Tensor] = []
@@ -237,6 +253,17 @@ class _TransitionTask:
             dev = to_device(packed, device)
             feats_t = dev[:fbytes].view(torch.int64).view(Sa, nF)
             valid_t = dev[fbytes:fbytes + Sa * A].view(Sa, A)
+            if self.fused:
+                gold_t = dev[fbytes + Sa * A:].view(Sa, A)
+                mod = self.pipe.module
+                scores, actions_dev = self.hip.fused_step(
+                    self.pre_d, feats_t, mod.lower_b, mod.upper.weight,
+                    mod.upper.bias, gold_t, valid_t, self.task_id, True,
+                )
+                self.score_chunks.append(scores)
+                self.gold_chunks.append(gold_t)
+                self.valid_chunks.append(valid_t)
+                return act_idx, actions_dev, None
             if self.train:
                 hidden = _ops.parser_step_score_accum(
                     self.pre_d, feats_t, self.pipe.module.lower_b,
@@ -414,6 +441,12 @@ class _TransitionPipeBase(TrainablePipe):
         display = float(step_loss.detach())
         with timing.span("raw/phase1_bwd"):
             step_loss.backward()  # phase 1: upper + lower_b grads
+            if task.fused:
+                # the C++ autograd node stashed (feats, dSummed) per step in a
+                # GIL-free session store; drain into the batched scatter path
+                task.entries.extend(
+                    (e[0], e[1]) for e in task.hip.fused_entries_take(task.task_id)
+                )
             if task.entries is not None:
                 # batched dPre scatter: one sort + segmented reduction for
                 # ALL steps (replaces per-step atomic scatters)
