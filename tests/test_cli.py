@@ -107,3 +107,35 @@ def test_overrides_reach_workers(tmp_path):
     assert r.returncode == 0, r.stderr[-3000:]
     cfg_text = (out / "model-last" / "config.cfg").read_text()
     assert '"width": 64' in cfg_text or "width = 64" in cfg_text
+
+
+def test_multinode_emulation_two_launchers(tmp_path):
+    """Two launcher invocations on localhost emulate a 2-node run: each node
+    contributes one worker (--nnodes 2 --node-rank r --address host:port);
+    global ranks must be unique and training completes on both."""
+    import socket
+    import threading
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    out = tmp_path / "outmn"
+    results = {}
+
+    def node(rank):
+        results[rank] = _run_cli(
+            [str(CFG), "--n-workers", "1", "--output", str(out),
+             "--address", f"127.0.0.1:{port}", "--nnodes", "2",
+             "--node-rank", str(rank),
+             "--training.max_steps", "6", "--training.eval_frequency", "3"])
+
+    threads = [threading.Thread(target=node, args=(r,)) for r in (0, 1)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=420)
+    for r in (0, 1):
+        assert results[r].returncode == 0, (r, results[r].stderr[-3000:])
+    assert (out / "model-last" / "config.cfg").exists()
+    assert (out / "model-last" / "optim.rank0.pt").exists()
+    assert (out / "model-last" / "optim.rank1.pt").exists()
