@@ -139,3 +139,22 @@ def test_multinode_emulation_two_launchers(tmp_path):
     assert (out / "model-last" / "config.cfg").exists()
     assert (out / "model-last" / "optim.rank0.pt").exists()
     assert (out / "model-last" / "optim.rank1.pt").exists()
+
+
+def test_cli_evaluate_saved_model(tmp_path):
+    """`spacy-mi ray evaluate <model-dir>` loads a checkpoint and prints a
+    scores JSON including the weighted composite."""
+    out = tmp_path / "outev"
+    r = _run_cli([str(CFG), "--output", str(out), "--training.max_steps", "4",
+                  "--training.eval_frequency", "2"])
+    assert r.returncode == 0, r.stderr[-3000:]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    r2 = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "ray", "evaluate",
+         str(out / "model-last")],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert r2.returncode == 0, r2.stderr[-3000:]
+    scores = json.loads(r2.stdout)
+    assert "score" in scores and "tag_acc" in scores
