@@ -136,6 +136,20 @@ def ray_evaluate_cli(
     print(json.dumps(scores, indent=2))
 
 
+@app.command("serve")
+def serve_cli(
+    model_path: Path = typer.Argument(..., help="Trained pipeline directory (model-best/model-last)"),
+    host: str = typer.Option("127.0.0.1", "--host"),
+    port: int = typer.Option(8000, "--port", "-p"),
+    use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
+    max_batch: int = typer.Option(256, "--max-batch", help="Max docs per decode batch"),
+):
+    """Serve a trained pipeline over HTTP (POST /annotate, GET /info)."""
+    from spacy_ray_amd.serve.app import serve
+
+    serve(model_path, host=host, port=port, use_gpu=use_gpu, max_batch=max_batch)
+
+
 def main() -> None:
     app()
 

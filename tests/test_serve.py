@@ -1,0 +1,59 @@
+"""HTTP serving: build_app over a trained pipeline, annotate texts through
+the real decode path (TestClient, no sockets)."""
+import os
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture(scope="module")
+def served_nlp():
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg = Config.from_disk(REPO / "examples" / "configs" / "en_core_cnn.cfg")
+    return init_nlp(cfg, device="cpu", sample_size=16)
+
+
+def test_serve_endpoints(served_nlp):
+    from fastapi.testclient import TestClient
+
+    from spacy_ray_amd.serve.app import build_app
+
+    client = TestClient(build_app(served_nlp))
+    assert client.get("/health").json() == {"status": "ok"}
+
+    info = client.get("/info").json()
+    assert info["pipeline"] == served_nlp.pipe_names
+    assert "tagger" in info["labels"]
+
+    r = client.post("/annotate", json={"texts": ["hello brave new world", "x y"]})
+    assert r.status_code == 200, r.text
+    docs = r.json()["docs"]
+    assert len(docs) == 2
+    d0 = docs[0]
+    assert d0["words"] == ["hello", "brave", "new", "world"]
+    assert len(d0["tags"]) == 4
+    assert len(d0["heads"]) == 4
+    assert len(d0["ents"]) == 4 and all(isinstance(t, str) for t in d0["ents"])
+    assert isinstance(d0["spans"], list)
+
+    # empty batch is fine
+    r = client.post("/annotate", json={"texts": []})
+    assert r.status_code == 200 and r.json()["docs"] == []
+
+
+def test_serve_spans_match_biluo(served_nlp):
+    from fastapi.testclient import TestClient
+
+    from spacy_ray_amd.serve.app import build_app
+    from spacy_ray_amd.train.scorer import _ents_to_spans
+
+    client = TestClient(build_app(served_nlp))
+    r = client.post("/annotate", json={"texts": ["alpha beta gamma delta epsilon"]})
+    d = r.json()["docs"][0]
+    expect = sorted(_ents_to_spans(d["ents"]))
+    got = [(s["start"], s["end"], s["label"]) for s in d["spans"]]
+    assert got == expect
