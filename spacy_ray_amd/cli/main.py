@@ -136,6 +136,20 @@ def ray_evaluate_cli(
     print(json.dumps(scores, indent=2))
 
 
+@app.command("convert")
+def convert_cli(
+    input_path: Path = typer.Argument(..., help="CoNLL-U (.conllu) or IOB (.iob) file"),
+    output_path: Path = typer.Argument(..., help="Output DocBin (.spacy) path"),
+    fmt: Optional[str] = typer.Option(None, "--format", "-f", help="conllu | iob (default: infer from extension)"),
+    tag_col: str = typer.Option("upos", "--tag-col", help="CoNLL-U tag column: upos | xpos"),
+):
+    """Convert a CoNLL-U / IOB corpus to a DocBin for training."""
+    from spacy_ray_amd.data.convert import convert_file
+
+    n = convert_file(input_path, output_path, fmt=fmt, tag_col=tag_col)
+    print(f"wrote {n} docs -> {output_path}")
+
+
 @app.command("serve")
 def serve_cli(
     model_path: Path = typer.Argument(..., help="Trained pipeline directory (model-best/model-last)"),

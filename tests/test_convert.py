@@ -1,0 +1,89 @@
+"""Corpus converters: CoNLL-U / IOB -> DocBin."""
+import subprocess
+import sys
+from pathlib import Path
+
+from spacy_ray_amd.data.convert import iob_to_biluo, read_conllu, read_iob
+
+REPO = Path(__file__).resolve().parent.parent
+
+CONLLU = """\
+# sent_id = 1
+# text = The dog barks
+1\tThe\tthe\tDET\tDT\t_\t2\tdet\t_\t_
+2\tdog\tdog\tNOUN\tNN\t_\t3\tnsubj\t_\t_
+3\tbarks\tbark\tVERB\tVBZ\t_\t0\troot\t_\t_
+
+1-2\tcannot\t_\t_\t_\t_\t_\t_\t_\t_
+1\tcan\tcan\tAUX\tMD\t_\t0\troot\t_\t_
+2\tnot\tnot\tPART\tRB\t_\t1\tadvmod\t_\t_
+2.1\tghost\t_\t_\t_\t_\t_\t_\t_\t_
+"""
+
+IOB = """\
+-DOCSTART- -X- O O
+
+EU NNP I-ORG
+rejects VBZ O
+German JJ I-MISC
+call NN O
+
+Peter NNP B-PER
+Blackburn NNP I-PER
+"""
+
+
+def test_read_conllu():
+    docs = read_conllu(CONLLU)
+    assert len(docs) == 2
+    d = docs[0]
+    assert d.words == ["The", "dog", "barks"]
+    assert d.tags == ["DET", "NOUN", "VERB"]
+    assert list(d.heads) == [1, 2, -1]
+    assert d.deps == ["det", "nsubj", "root"]
+    d2 = docs[1]  # multiword range and empty node skipped
+    assert d2.words == ["can", "not"]
+    assert list(d2.heads) == [-1, 0]
+
+
+def test_read_conllu_xpos():
+    docs = read_conllu(CONLLU, tag_col="xpos")
+    assert docs[0].tags == ["DT", "NN", "VBZ"]
+
+
+def test_iob_to_biluo():
+    # IOB1: I- opens a span
+    assert iob_to_biluo(["I-ORG", "O", "I-MISC", "O"]) == ["U-ORG", "O", "U-MISC", "O"]
+    # IOB2: B- opens, I- continues
+    assert iob_to_biluo(["B-PER", "I-PER"]) == ["B-PER", "L-PER"]
+    assert iob_to_biluo(["B-PER", "I-PER", "I-PER", "O"]) == ["B-PER", "I-PER", "L-PER", "O"]
+    # adjacent spans, label change
+    assert iob_to_biluo(["B-A", "B-A"]) == ["U-A", "U-A"]
+    assert iob_to_biluo(["I-A", "I-B"]) == ["U-A", "U-B"]
+    assert iob_to_biluo([]) == []
+
+
+def test_read_iob():
+    docs = read_iob(IOB)
+    assert len(docs) == 2
+    assert docs[0].words == ["EU", "rejects", "German", "call"]
+    assert docs[0].ents == ["U-ORG", "O", "U-MISC", "O"]
+    assert docs[0].tags == ["NNP", "VBZ", "JJ", "NN"]
+    assert docs[1].ents == ["B-PER", "L-PER"]
+
+
+def test_convert_cli_roundtrip(tmp_path):
+    src = tmp_path / "sample.conllu"
+    src.write_text(CONLLU)
+    out = tmp_path / "sample.spacy"
+    r = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "convert",
+         str(src), str(out)],
+        cwd=str(REPO), capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Vocab
+
+    docs = list(DocBin.from_disk(out, Vocab()).get_docs(Vocab()))
+    assert len(docs) == 2 and docs[0].words == ["The", "dog", "barks"]
