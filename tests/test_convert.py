@@ -87,3 +87,60 @@ def test_convert_cli_roundtrip(tmp_path):
 
     docs = list(DocBin.from_disk(out, Vocab()).get_docs(Vocab()))
     assert len(docs) == 2 and docs[0].words == ["The", "dog", "barks"]
+
+
+def test_convert_then_train_end_to_end(tmp_path):
+    """The full user data path: CoNLL-U file -> spacy-mi convert -> DocBin ->
+    spacy.Corpus.v1 reader -> training runs and checkpoints."""
+    import random
+
+    rng = random.Random(0)
+    lex = [f"w{i}" for i in range(30)]
+    sents = []
+    for _ in range(60):
+        n = rng.randint(3, 8)
+        rows = []
+        for i in range(1, n + 1):
+            w = lex[rng.randrange(len(lex))]
+            tag = "NOUN" if w < "w5" else "VERB"
+            head = 0 if i == 1 else i - 1  # chain tree rooted at token 1
+            dep = "root" if head == 0 else "dep"
+            rows.append(f"{i}\t{w}\t_\t{tag}\t_\t_\t{head}\t{dep}\t_\t_")
+        sents.append("\n".join(rows))
+    src = tmp_path / "train.conllu"
+    src.write_text("\n\n".join(sents) + "\n")
+
+    from spacy_ray_amd.data.convert import convert_file
+
+    train_bin = tmp_path / "train.spacy"
+    n = convert_file(src, train_bin)
+    assert n == 60
+
+    from tests.test_pipeline import TAGGER_CFG
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.cli.main import ray_train
+
+    cfg_text = TAGGER_CFG
+    for split in ("train", "dev"):
+        cfg_text = cfg_text.replace(
+            f"""[corpora.{split}]
+@readers = "spacy-mi.SyntheticCorpus.v1\"""",
+            f"""[corpora.{split}]
+@readers = "spacy.Corpus.v1"
+path = "{train_bin}\"""",
+        )
+    # strip the synthetic-reader kwargs that the DocBin reader doesn't take
+    lines = [l for l in cfg_text.splitlines()
+             if not l.startswith(("n_docs", "words_per_doc", "vocab_size",
+                                  "n_tags", "seed", "world_seed", "shuffle"))]
+    cfg = Config.from_str("\n".join(lines))
+    cfg_path = tmp_path / "cfg.cfg"
+    cfg_path.write_text(cfg.to_str())
+    out = tmp_path / "model"
+    rc = ray_train(
+        Config.from_disk(cfg_path, overrides={"training.max_steps": 4,
+                                              "training.eval_frequency": 2}),
+        config_path=cfg_path, output_path=out, n_workers=1,
+    )
+    assert rc == 0
+    assert (out / "model-last" / "config.cfg").exists()
