@@ -129,6 +129,15 @@ def distributed_train(
         icfg, [T["train_corpus"], T["dev_corpus"]]
     )
     train_corpus = _shard_corpus(train_corpus, rank, world)
+    # frozen components: exclude their params from the flat buffer BEFORE the
+    # engine flattens (spaCy contract: frozen pipes are not updated at all —
+    # leaving them in would apply AdamW's decoupled weight decay every step
+    # even with zero gradient, silently shrinking the frozen weights)
+    for pname in T.get("frozen_components") or []:
+        pipe = dict(nlp.pipeline).get(pname)
+        if pipe is not None and getattr(pipe, "module", None) is not None:
+            for p in pipe.module.parameters():
+                p.requires_grad_(False)
     engine = ZeRO1Engine(nlp, T["optimizer"], comm)
     if resume and output_path and (Path(output_path) / "model-last").exists():
         nlp.from_disk(Path(output_path) / "model-last")

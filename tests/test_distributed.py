@@ -288,3 +288,43 @@ def test_two_rank_matches_single_rank_full_batch(tmp_path):
     two = _run_world(tmp_path, 2, 29514)[0]
     assert one["param_hash"] == pytest.approx(two["param_hash"], rel=1e-5)
     assert np.allclose(one["param"], two["param"], rtol=1e-4, atol=1e-6)
+
+
+def test_frozen_components_not_updated(tmp_path):
+    """training.frozen_components: the frozen pipe's params must be EXACTLY
+    unchanged after training — not even weight-decayed (regression: frozen
+    params left in the flat buffer got AdamW decoupled decay each step)."""
+    from spacy_ray_amd.train.worker import distributed_train
+
+    out = tmp_path / "frozen"
+    holder = {}
+
+    import spacy_ray_amd.train.worker as workermod
+    orig_init = workermod.init_nlp
+
+    def capture_init(config, **kw):
+        nlp = orig_init(config, **kw)
+        if "nlp" not in holder:
+            holder["nlp"] = nlp
+            t2v = dict(nlp.pipeline)["tok2vec"]
+            holder["before"] = [p.detach().clone() for p in t2v.module.parameters()]
+        return nlp
+
+    workermod.init_nlp = capture_init
+    try:
+        distributed_train(
+            Config.from_str(TAGGER_CFG, overrides={
+                "training.max_steps": 3, "training.eval_frequency": 2,
+                "training.frozen_components": ["tok2vec"]}),
+            output_path=out, use_gpu=-1)
+    finally:
+        workermod.init_nlp = orig_init
+    nlp = holder["nlp"]
+    t2v = dict(nlp.pipeline)["tok2vec"]
+    tagger = dict(nlp.pipeline)["tagger"]
+    after = list(t2v.module.parameters())
+    assert all(not p.requires_grad for p in after)
+    for b, a in zip(holder["before"], after):
+        assert torch.equal(b, a)  # bit-identical: no update, no decay
+    # the trainable head did move
+    assert any(p.requires_grad for p in tagger.module.parameters())
