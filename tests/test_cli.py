@@ -158,3 +158,41 @@ def test_cli_evaluate_saved_model(tmp_path):
     assert r2.returncode == 0, r2.stderr[-3000:]
     scores = json.loads(r2.stdout)
     assert "score" in scores and "tag_acc" in scores
+
+
+def test_code_injection_custom_reader(tmp_path):
+    """--code imports user code on every worker before config resolution:
+    a custom @readers function defined in the user file must resolve
+    (the reference's import_code contract, train_cli.py --code)."""
+    code = tmp_path / "user_code.py"
+    code.write_text("""
+from spacy_ray_amd.config.registry import registry
+from spacy_ray_amd.data.corpus import create_synthetic_corpus
+
+@registry.readers("user.TinyCorpus.v1")
+def tiny_corpus(seed: int = 0):
+    return create_synthetic_corpus(n_docs=40, words_per_doc=8, vocab_size=50,
+                                   n_tags=5, seed=seed)
+""")
+    cfg_text = CFG.read_text().replace(
+        '@readers = "spacy-mi.SyntheticCorpus.v1"', "@@MARK@@", 1)
+    # replace the train corpus block with the custom reader (strip its kwargs)
+    lines, out_lines, in_train = cfg_text.splitlines(), [], False
+    for l in lines:
+        if l == "@@MARK@@":
+            in_train = True
+            out_lines.append('@readers = "user.TinyCorpus.v1"')
+            continue
+        if in_train:
+            if l.startswith("[") or not l.strip():
+                in_train = False
+                out_lines.append(l)
+            continue  # drop synthetic kwargs
+        out_lines.append(l)
+    cfg_path = tmp_path / "cfg.cfg"
+    cfg_path.write_text("\n".join(out_lines))
+    out = tmp_path / "out"
+    r = _run_cli([str(cfg_path), "--code", str(code), "--output", str(out),
+                  "--training.max_steps", "2", "--training.eval_frequency", "2"])
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert (out / "model-last" / "config.cfg").exists()
