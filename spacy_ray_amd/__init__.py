@@ -14,3 +14,15 @@ Layer map (idiomatic MI355X rebuild of SURVEY.md §1):
   train     — train_while_improving-contract loop, loggers, scoring
 """
 __version__ = "0.1.0"
+
+
+def load(model_path, device: str = "cpu"):
+    """Load a trained pipeline directory (the `spacy.load` convenience):
+    ``nlp = spacy_ray_amd.load("/path/model-best", device="cuda:0")``."""
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import build_nlp
+
+    config = Config.from_disk(f"{model_path}/config.cfg")
+    nlp = build_nlp(config, device=device)
+    nlp.from_disk(model_path)
+    return nlp

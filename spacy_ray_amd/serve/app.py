@@ -76,15 +76,12 @@ def serve(model_path, host: str = "127.0.0.1", port: int = 8000,
     import torch
     import uvicorn
 
-    from spacy_ray_amd.config.config import Config
-    from spacy_ray_amd.pipeline.language import build_nlp
+    import spacy_ray_amd
 
     device = (
         f"cuda:{max(use_gpu, 0)}"
         if (use_gpu >= 0 and torch.cuda.is_available())
         else "cpu"
     )
-    config = Config.from_disk(f"{model_path}/config.cfg")
-    nlp = build_nlp(config, device=device)
-    nlp.from_disk(model_path)
+    nlp = spacy_ray_amd.load(model_path, device=device)
     uvicorn.run(build_app(nlp, max_batch=max_batch), host=host, port=port)

@@ -57,3 +57,17 @@ def test_serve_spans_match_biluo(served_nlp):
     expect = sorted(_ents_to_spans(d["ents"]))
     got = [(s["start"], s["end"], s["label"]) for s in d["spans"]]
     assert got == expect
+
+
+def test_package_load_roundtrip(tmp_path):
+    """spacy_ray_amd.load() on a saved checkpoint annotates text."""
+    import spacy_ray_amd
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg = Config.from_disk(REPO / "examples" / "configs" / "en_tagger_cpu.cfg")
+    nlp = init_nlp(cfg, device="cpu", sample_size=8)
+    nlp.to_disk(tmp_path / "model")
+    nlp2 = spacy_ray_amd.load(tmp_path / "model")
+    doc = nlp2("hello world again")
+    assert doc.tags and len(doc.tags) == 3
