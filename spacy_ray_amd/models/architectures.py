@@ -109,7 +109,11 @@ def make_tok2vec_listener(width: int, upstream: str = "*"):
 
 @registry.architectures("spacy.Tagger.v2")
 def make_tagger_model(tok2vec: ModelSpec, nO: Optional[int] = None, normalize: bool = False):
-    return ModelSpec(lambda: None, width=tok2vec.width, kind="tagger")
+    spec = ModelSpec(lambda: None, width=tok2vec.width, kind="tagger")
+    # a full tok2vec block (not a listener) => the pipe owns its encoder
+    # (spaCy's embedded-tok2vec configuration, e.g. standalone taggers)
+    spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
+    return spec
 
 
 @registry.architectures("spacy.TransitionBasedParser.v2")
@@ -130,4 +134,5 @@ def make_transition_parser_model(
     spec = ModelSpec(build, width=tok2vec.width, kind=f"transition:{state_type}")
     spec.nF = nF
     spec.hidden_width = hidden_width
+    spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
     return spec

@@ -92,11 +92,13 @@ class Language:
         for name, pipe in self.pipeline:
             if isinstance(pipe, Tok2VecPipe) or name in self._frozen:
                 continue
+            own = pipe.own_tok2vec(batch, drop=drop)
+            pt2v = own if own is not None else t2v
             if isinstance(pipe, _TransitionPipeBase):
-                trans_tasks.append((name, pipe, pipe.make_loss_task(examples, t2v)))
+                trans_tasks.append((name, pipe, pipe.make_loss_task(examples, pt2v)))
                 continue
             with timing.phase(f"loss/{name}"):
-                loss, display = pipe.get_loss(examples, t2v, batch)
+                loss, display = pipe.get_loss(examples, pt2v, batch)
             losses[name] = losses.get(name, 0.0) + display
             total = loss if total is None else total + loss
         if trans_tasks:
@@ -134,13 +136,19 @@ class Language:
             trans = []
             for name, pipe in self.pipeline:
                 if isinstance(pipe, _TransitionPipeBase):
-                    trans.append((pipe,) + pipe.make_predict_task(docs, t2v))
+                    own = pipe.own_tok2vec(batch)
+                    trans.append((pipe,) + pipe.make_predict_task(
+                        docs, own if own is not None else t2v))
             if trans:
                 run_transition_tasks([t[1] for t in trans])
-        for name, pipe in self.pipeline:
-            if isinstance(pipe, (Tok2VecPipe, _TransitionPipeBase)):
-                continue
-            pipe.predict_and_set(docs, t2v, batch)
+            heads = []
+            for name, pipe in self.pipeline:
+                if isinstance(pipe, (Tok2VecPipe, _TransitionPipeBase)):
+                    continue
+                own = pipe.own_tok2vec(batch)
+                heads.append((pipe, own if own is not None else t2v))
+        for pipe, pt2v in heads:
+            pipe.predict_and_set(docs, pt2v, batch)
         for pipe, task, splits, shards in trans:
             for (lo, hi, base), states in zip(splits, shards):
                 pipe._annotate(docs[lo:hi], states)

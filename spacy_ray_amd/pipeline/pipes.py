@@ -52,6 +52,23 @@ class TrainablePipe:
     def load_cfg(self, cfg: Dict, device) -> None:
         self.cfg = dict(cfg)
 
+    # ------- embedded (non-listener) tok2vec: a pipe whose config carries a
+    # full tok2vec block owns its encoder (spaCy's per-component tok2vec).
+    # The encoder is registered as a submodule of the head module so the
+    # flat-buffer engine and checkpoints see its params automatically.
+    embedded_spec = None  # set by subclasses from their ModelSpec
+
+    def _attach_embedded(self, device) -> None:
+        if (self.embedded_spec is not None and self.module is not None
+                and not hasattr(self.module, "embedded_t2v")):
+            self.module.embedded_t2v = self.embedded_spec.build().to(device)
+
+    def own_tok2vec(self, batch, drop: float = 0.0):
+        """This pipe's private tok2vec output, or None if it listens to the
+        shared encoder."""
+        emb = getattr(self.module, "embedded_t2v", None) if self.module is not None else None
+        return None if emb is None else emb(batch, drop=drop)
+
 
 class Tok2VecPipe(TrainablePipe):
     name = "tok2vec"
@@ -100,6 +117,7 @@ class TaggerPipe(TrainablePipe):
         super().__init__()
         self.name = name
         self.width = spec.width
+        self.embedded_spec = getattr(spec, "embedded_tok2vec", None)
         self.labels: List[str] = []
         self.label2id: Dict[str, int] = {}
 
@@ -114,6 +132,7 @@ class TaggerPipe(TrainablePipe):
         self.cfg["labels"] = self.labels
         if self.module is None:
             self.module = TaggerHead(self.width, max(1, len(self.labels))).to(device)
+        self._attach_embedded(device)
 
     def load_cfg(self, cfg, device) -> None:
         super().load_cfg(cfg, device)
@@ -129,6 +148,7 @@ class TaggerPipe(TrainablePipe):
                 f"{self.name}: checkpoint has {n} labels but the initialized "
                 f"module has {self.module.output.out_features}"
             )
+        self._attach_embedded(device)
 
     def _gold_ids(self, examples, n_tokens: int) -> np.ndarray:
         # per-doc gold-id arrays are cached on the reference Doc (dict
@@ -359,6 +379,7 @@ class _TransitionPipeBase(TrainablePipe):
         self.name = name
         self.spec = spec
         self.width = spec.width
+        self.embedded_spec = getattr(spec, "embedded_tok2vec", None)
         self.hidden_width = getattr(spec, "hidden_width", 64)
         self.nF = getattr(spec, "nF", 13)
         self.labels: List[str] = []
@@ -371,6 +392,7 @@ class _TransitionPipeBase(TrainablePipe):
             ).to(device)
             self.module.initialize_output(self._n_actions())
             self.module.to(device)
+        self._attach_embedded(device)
 
     def load_cfg(self, cfg, device) -> None:
         super().load_cfg(cfg, device)

@@ -198,3 +198,103 @@ def test_predict_sets_annotations():
         assert d.ents is not None
         # heads are in-range or -1 (root)
         assert all(-1 <= int(h) < len(d) for h in d.heads)
+
+
+EMBEDDED_TAGGER_CFG = """
+[nlp]
+lang = "en"
+pipeline = ["tagger"]
+
+[components]
+
+[components.tagger]
+factory = "tagger"
+
+[components.tagger.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.tagger.model.tok2vec]
+@architectures = "spacy.HashEmbedCNN.v2"
+width = 32
+depth = 2
+embed_size = 500
+
+[corpora]
+
+[corpora.train]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 200
+words_per_doc = 10
+vocab_size = 100
+n_tags = 5
+seed = 0
+
+[corpora.dev]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 50
+words_per_doc = 10
+vocab_size = 100
+n_tags = 5
+seed = 1
+shuffle = false
+
+[training]
+train_corpus = "corpora.train"
+dev_corpus = "corpora.dev"
+max_steps = 40
+eval_frequency = 20
+
+[training.optimizer]
+@optimizers = "Adam.v1"
+learn_rate = 0.01
+"""
+
+
+def test_embedded_tok2vec_tagger(tmp_path):
+    """A pipe with a FULL tok2vec block (not a listener) owns its encoder:
+    no shared tok2vec pipe in the pipeline, params live inside the pipe's
+    module, training learns, checkpoints roundtrip."""
+    import torch
+
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import build_nlp, init_nlp
+
+    cfg = Config.from_str(EMBEDDED_TAGGER_CFG)
+    nlp = init_nlp(cfg)
+    assert nlp.pipe_names == ["tagger"]
+    tagger = dict(nlp.pipeline)["tagger"]
+    assert hasattr(tagger.module, "embedded_t2v")
+    n_emb = sum(p.numel() for p in tagger.module.embedded_t2v.parameters())
+    assert n_emb > 0
+
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.config.config import resolve
+
+    T = resolve(cfg.interpolate()["training"], validate=False)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    # the embedded encoder's params are in the flat buffer
+    assert any("embedded_t2v" in n for n in engine.param_names)
+    from spacy_ray_amd.config.config import resolve_dot_names
+    (train_corpus,) = resolve_dot_names(cfg.interpolate(), ["corpora.train"])
+    examples = list(train_corpus(nlp))[:64]
+    first = last = None
+    for i in range(20):
+        losses = {}
+        engine.accumulate(examples, losses=losses)
+        engine.apply_step()
+        if first is None:
+            first = losses["tagger"]
+        last = losses["tagger"]
+    assert last < first, (first, last)
+
+    # predict + roundtrip
+    doc = nlp("hello world")
+    assert doc.tags and len(doc.tags) == 2
+    nlp.to_disk(tmp_path / "m")
+    nlp2 = build_nlp(Config.from_disk(tmp_path / "m" / "config.cfg"))
+    nlp2.from_disk(tmp_path / "m")
+    t2 = dict(nlp2.pipeline)["tagger"]
+    a = torch.cat([p.reshape(-1) for p in tagger.module.embedded_t2v.parameters()])
+    b = torch.cat([p.reshape(-1) for p in t2.module.embedded_t2v.parameters()])
+    assert torch.allclose(a.float(), b.float())
