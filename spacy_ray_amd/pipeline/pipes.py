@@ -243,7 +243,7 @@ class _TransitionTask:
         # on — GPU-validated vs the python path (test_fused_step_parity);
         # SRX_FUSED_STEP=0 falls back.  See fusedstep in srx_ext.hip.
         self.fused = (
-            train and self.hip is not None
+            self.hip is not None
             and hasattr(self.hip, "fused_step")
             and os.environ.get("SRX_FUSED_STEP", "1") == "1"
         )
@@ -273,7 +273,7 @@ class _TransitionTask:
             dev = to_device(packed, device)
             feats_t = dev[:fbytes].view(torch.int64).view(Sa, nF)
             valid_t = dev[fbytes:fbytes + Sa * A].view(Sa, A)
-            if self.fused:
+            if self.fused and self.train:
                 gold_t = dev[fbytes + Sa * A:].view(Sa, A)
                 mod = self.pipe.module
                 scores, actions_dev = self.hip.fused_step(
@@ -283,6 +283,17 @@ class _TransitionTask:
                 self.score_chunks.append(scores)
                 self.gold_chunks.append(gold_t)
                 self.valid_chunks.append(valid_t)
+                return act_idx, actions_dev, None
+            if self.fused:
+                # decode: same fused call under no_grad (no autograd node ->
+                # nothing lands in the C++ entries store); valid doubles as
+                # the "gold" mask so argmax-over-valid is selected
+                mod = self.pipe.module
+                with torch.no_grad():
+                    _, actions_dev = self.hip.fused_step(
+                        self.pre_d, feats_t, mod.lower_b, mod.upper.weight,
+                        mod.upper.bias, valid_t, valid_t, 0, False,
+                    )
                 return act_idx, actions_dev, None
             if self.train:
                 hidden = _ops.parser_step_score_accum(
