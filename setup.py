@@ -15,7 +15,7 @@ import os
 import sys
 from pathlib import Path
 
-from setuptools import setup, Extension
+from setuptools import Extension, find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -60,7 +60,7 @@ if hip_sources and "SRX_SKIP_HIP" not in os.environ:
 setup(
     name="spacy_ray_amd",
     version="0.1.0",
-    packages=["spacy_ray_amd"],
+    packages=find_packages(include=["spacy_ray_amd", "spacy_ray_amd.*"]),
     ext_modules=ext_modules,
     cmdclass=cmdclass,
     entry_points={
