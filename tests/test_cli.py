@@ -196,3 +196,26 @@ def tiny_corpus(seed: int = 0):
                   "--training.max_steps", "2", "--training.eval_frequency", "2"])
     assert r.returncode == 0, r.stderr[-3000:]
     assert (out / "model-last" / "config.cfg").exists()
+
+
+def test_bench_two_rank_gloo(tmp_path):
+    """The driver's SCALE run launches bench.py under torch.distributed.run
+    with N ranks; exercise that exact path on CPU/gloo with 2 ranks and
+    check rank 0 prints one valid JSON line with whole-job aggregation."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29531", str(REPO / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--batch-words", "2000"],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-2500:])
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, r.stdout[-1500:]
+    rec = json.loads(json_lines[0])
+    assert rec["steps"] == 2 and rec["scaling"] == "weak"
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
