@@ -298,3 +298,27 @@ def test_embedded_tok2vec_tagger(tmp_path):
     a = torch.cat([p.reshape(-1) for p in tagger.module.embedded_t2v.parameters()])
     b = torch.cat([p.reshape(-1) for p in t2.module.embedded_t2v.parameters()])
     assert torch.allclose(a.float(), b.float())
+
+
+def test_checkpoint_roundtrip_full_pipeline(tmp_path):
+    """en_core (tok2vec+tagger+parser+NER): saved and reloaded pipelines must
+    produce IDENTICAL annotations (tags, heads, deps, BILUO ents) — covers
+    the parser/NER label-state + transition-model checkpoint path."""
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    nlp = init_nlp(cfg, sample_size=32)
+    docs = make_synthetic_docs(nlp.vocab, n_docs=6, words_per_doc=9,
+                               vocab_size=150, n_tags=50, n_deps=40,
+                               n_ent_types=4, seed=21)
+
+    def annotate(model):
+        outs = model.predict_docs([d.copy_unannotated() for d in docs])
+        return [(list(d.tags), list(d.heads), list(d.deps), list(d.ents))
+                for d in outs]
+
+    preds1 = annotate(nlp)
+    nlp.to_disk(tmp_path / "model")
+    nlp2 = build_nlp(Config.from_disk(tmp_path / "model" / "config.cfg"))
+    nlp2.from_disk(tmp_path / "model")
+    assert annotate(nlp2) == preds1
