@@ -252,7 +252,33 @@ def build_nlp(config: Config, device: str = "cpu") -> Language:
         factory = registry.factories.get(factory_name)
         pipe = factory(name=name, model=model_spec, **comp_cfg)
         nlp.add_pipe(name, pipe)
+    _check_listener_widths(nlp)
     return nlp
+
+
+def _check_listener_widths(nlp: Language) -> None:
+    """Fail at build time (not deep inside a matmul) when a listener head's
+    width disagrees with the shared encoder, or when a listener has no
+    shared encoder to listen to."""
+    shared = nlp.tok2vec
+    for name, pipe in nlp.pipeline:
+        if isinstance(pipe, Tok2VecPipe) or pipe.listens_to is None:
+            continue
+        if getattr(pipe, "embedded_spec", None) is not None:
+            continue  # owns its encoder
+        if shared is None:
+            raise ValueError(
+                f"component {name!r} listens to a shared tok2vec but the "
+                f"pipeline {nlp.pipe_names} has none — add a tok2vec pipe or "
+                f"give {name!r} an embedded [components.{name}.model.tok2vec] "
+                f"block (a full architecture, not a Tok2VecListener)"
+            )
+        if pipe.width != shared.width:
+            raise ValueError(
+                f"component {name!r} expects width {pipe.width} but the "
+                f"shared tok2vec produces width {shared.width} — align "
+                f"[components.{name}.model.tok2vec.width] with the encoder"
+            )
 
 
 def init_nlp(config: Config, device: str = "cpu", sample_size: int = 128) -> Language:

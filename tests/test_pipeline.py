@@ -322,3 +322,21 @@ def test_checkpoint_roundtrip_full_pipeline(tmp_path):
     nlp2 = build_nlp(Config.from_disk(tmp_path / "model" / "config.cfg"))
     nlp2.from_disk(tmp_path / "model")
     assert annotate(nlp2) == preds1
+
+
+def test_listener_width_mismatch_fails_at_build():
+    cfg_text = TAGGER_CFG.replace(
+        '@architectures = "spacy.Tok2VecListener.v1"\nwidth = 32',
+        '@architectures = "spacy.Tok2VecListener.v1"\nwidth = 64')
+    assert "width = 64" in cfg_text
+    with pytest.raises(ValueError, match="width"):
+        build_nlp(Config.from_str(cfg_text))
+
+
+def test_listener_without_shared_tok2vec_fails_at_build():
+    import re
+
+    cfg_text = TAGGER_CFG.replace('pipeline = ["tok2vec", "tagger"]',
+                                  'pipeline = ["tagger"]')
+    with pytest.raises(ValueError, match="listens to a shared tok2vec"):
+        build_nlp(Config.from_str(cfg_text))
