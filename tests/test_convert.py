@@ -144,3 +144,38 @@ path = "{train_bin}\"""",
     )
     assert rc == 0
     assert (out / "model-last" / "config.cfg").exists()
+
+
+def test_iob_to_biluo_always_valid():
+    """Property: for ANY tag sequence, the output is structurally valid
+    BILUO (B opens, I continues, L closes with matching labels; U/O stand
+    alone) and marks the same token set as entity-covered for IOB2 input."""
+    from hypothesis import given, strategies as st
+
+    tag = st.one_of(
+        st.just("O"),
+        st.builds(lambda k, l: f"{k}-{l}",
+                  st.sampled_from(["B", "I"]), st.sampled_from(["PER", "ORG", "X"])),
+    )
+
+    @given(st.lists(tag, max_size=12))
+    def check(tags):
+        out = iob_to_biluo(tags)
+        assert len(out) == len(tags)
+        open_label = None
+        for t in out:
+            kind, _, lab = t.partition("-")
+            if kind == "B":
+                assert open_label is None
+                open_label = lab
+            elif kind == "I":
+                assert open_label == lab
+            elif kind == "L":
+                assert open_label == lab
+                open_label = None
+            else:
+                assert t in ("O", f"U-{lab}")
+                assert open_label is None
+        assert open_label is None  # every span closed
+
+    check()
