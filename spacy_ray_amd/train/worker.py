@@ -13,6 +13,7 @@ train_cli.py:41), cluster-scaled words logging (worker.py:308-311).
 from __future__ import annotations
 
 import json
+import logging
 import os
 import time
 from pathlib import Path
@@ -155,6 +156,7 @@ def distributed_train(
     before_to_disk = T.get("before_to_disk")
 
     dev_examples = None
+    eval_state: Dict[str, bool] = {}
 
     def evaluate():
         nonlocal dev_examples
@@ -168,7 +170,19 @@ def distributed_train(
                 scores = nlp.evaluate(dev_examples)
                 dt = max(1e-9, time.time() - t_eval)
                 scores["speed"] = sum(len(eg) for eg in dev_examples) / dt
-                score = weighted_score(scores, T.get("score_weights") or {})
+                weights = T.get("score_weights") or {}
+                missing = [k for k, w in weights.items()
+                           if w and k not in scores]
+                if missing and not eval_state.get("warned_weights"):
+                    # the reference errors here (E983, loggers.py:30-37); we
+                    # warn once so a typo'd weight key can't silently zero
+                    # the model-best selection
+                    logging.getLogger(__name__).warning(
+                        "score_weights keys %s not produced by evaluate "
+                        "(have: %s) — they contribute 0 to the composite",
+                        missing, sorted(scores))
+                    eval_state["warned_weights"] = True
+                score = weighted_score(scores, weights)
                 payload = (score, scores)
             else:
                 payload = None
