@@ -92,3 +92,15 @@ def test_schedule_compounding():
     sched = registry.schedules.get("compounding.v1")(start=1.0, stop=8.0, compound=2.0)
     vals = [next(sched) for _ in range(5)]
     assert vals == [1.0, 2.0, 4.0, 8.0, 8.0]
+
+
+def test_docbin_version_guard():
+    import msgpack
+    import pytest
+
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Vocab
+
+    bad = msgpack.packb({"version": 99, "docs": []}, use_bin_type=True)
+    with pytest.raises(ValueError, match="version"):
+        DocBin.from_bytes(bad, Vocab())
