@@ -340,3 +340,18 @@ def test_listener_without_shared_tok2vec_fails_at_build():
                                   'pipeline = ["tagger"]')
     with pytest.raises(ValueError, match="listens to a shared tok2vec"):
         build_nlp(Config.from_str(cfg_text))
+
+
+def test_nlp_pipe_streaming():
+    """nlp.pipe: streams texts/Docs in batches, yields annotated Docs in
+    order (spaCy nlp.pipe contract)."""
+    from spacy_ray_amd.vocab.doc import Doc
+
+    nlp = init_nlp(Config.from_str(TAGGER_CFG), sample_size=16)
+    texts = [f"word{i} and word{i + 1}" for i in range(7)]
+    inputs = list(texts[:5]) + [Doc(nlp.vocab, ["premade", "doc"])] + [texts[6]]
+    outs = list(nlp.pipe(inputs, batch_size=3))
+    assert len(outs) == 7
+    assert outs[5].words == ["premade", "doc"]
+    for d in outs:
+        assert d.tags is not None and len(d.tags) == len(d)
