@@ -30,6 +30,7 @@ class Language:
         self.pipeline: List[Tuple[str, TrainablePipe]] = []
         self.meta: Dict = {"lang": vocab.lang, "name": "pipeline", "version": "0.0.0"}
         self._frozen: List[str] = []
+        self._annotating: List[str] = []
 
     # ------------------------------------------------------------- pipeline
     @property
@@ -85,6 +86,13 @@ class Language:
         if t2v is not None and t2v_pipe.name in self._frozen:
             # frozen tok2vec: listeners may read it but must not update it
             t2v = t2v.detach()
+        if self._annotating:
+            # training.annotating_components (spaCy contract): the listed
+            # pipes SET their predictions on eg.predicted before any loss is
+            # computed, so downstream components can read the annotations.
+            # No built-in pipe consumes another's annotations today, but the
+            # mechanism is live for user components (--code factories).
+            self._run_annotating(examples, batch, t2v)
         total = None
         # transition pipes (parser/NER) run INTERLEAVED: their per-step
         # GPU-score / CPU-advance phases pipeline against each other
@@ -112,6 +120,25 @@ class Language:
         if total is None:
             total = torch.zeros((), device=self.device)
         return total, losses
+
+    def _run_annotating(self, examples, batch, t2v) -> None:
+        from .pipes import _TransitionPipeBase, run_transition_tasks
+
+        docs = [eg.predicted for eg in examples]
+        with torch.no_grad():
+            t2v_d = t2v.detach() if t2v is not None else None
+            for name, pipe in self.pipeline:
+                if name not in self._annotating or isinstance(pipe, Tok2VecPipe):
+                    continue
+                own = pipe.own_tok2vec(batch)
+                pt2v = own if own is not None else t2v_d
+                if isinstance(pipe, _TransitionPipeBase):
+                    _, task, splits, shards = (pipe,) + pipe.make_predict_task(docs, pt2v)
+                    run_transition_tasks([task])
+                    for (lo, hi, base), states in zip(splits, shards):
+                        pipe._annotate(docs[lo:hi], states)
+                else:
+                    pipe.predict_and_set(docs, pt2v, batch)
 
     def update(self, examples: Sequence[Example], *, sgd=None, losses=None, drop: float = 0.0):
         """Convenience single-process update (tests / CPU smoke)."""

@@ -65,6 +65,7 @@ def train_while_improving(
     losses: Dict[str, float] = {}
     words_seen = 0
     nlp._frozen = list(exclude)
+    nlp._annotating = list(annotating_components)
     for step, (epoch, batch) in enumerate(train_data):
         if before_update is not None:
             before_update(nlp, {"step": step, "epoch": epoch})
@@ -103,3 +104,4 @@ def train_while_improving(
             if (step - best_step) >= patience * eval_frequency and len(results) >= patience:
                 break
     nlp._frozen = []
+    nlp._annotating = []

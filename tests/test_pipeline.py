@@ -355,3 +355,32 @@ def test_nlp_pipe_streaming():
     assert outs[5].words == ["premade", "doc"]
     for d in outs:
         assert d.tags is not None and len(d.tags) == len(d)
+
+
+def test_annotating_components_set_predictions_before_loss():
+    """training.annotating_components: listed pipes write their predictions
+    onto eg.predicted before losses run (spaCy contract), so downstream
+    user components can consume them."""
+    from spacy_ray_amd.config.config import resolve_dot_names
+
+    cfg = Config.from_str(TAGGER_CFG)
+    nlp = init_nlp(cfg, sample_size=16)
+    icfg = cfg.interpolate()
+    (train_corpus,) = resolve_dot_names(icfg, ["corpora.train"])
+    examples = []
+    for eg in train_corpus(nlp):
+        examples.append(eg)
+        if len(examples) >= 4:
+            break
+    for eg in examples:
+        eg.predicted.tags = None
+    nlp._annotating = ["tagger"]
+    try:
+        total, losses = nlp.forward_loss(examples)
+    finally:
+        nlp._annotating = []
+    assert "tagger" in losses
+    for eg in examples:
+        assert eg.predicted.tags is not None
+        assert len(eg.predicted.tags) == len(eg.predicted)
+    total.backward()  # annotation ran under no_grad; loss graph intact
