@@ -53,6 +53,7 @@ def train_while_improving(
     exclude: Sequence[str] = (),
     annotating_components: Sequence[str] = (),
     before_update: Optional[Callable] = None,
+    initial_best: Optional[float] = None,
 ):
     """Generator of (batch, info, is_best_checkpoint).
 
@@ -61,7 +62,9 @@ def train_while_improving(
         stepper.apply_step() -> None                        (clip+opt+sync)
     `evaluate()` -> (score, other_scores).
     """
-    results = []
+    # resume: seed the best-so-far so a resumed run cannot overwrite
+    # model-best with a worse model (step -1 = "before this run")
+    results = [] if initial_best is None else [(float(initial_best), -1)]
     losses: Dict[str, float] = {}
     words_seen = 0
     nlp._frozen = list(exclude)

@@ -155,6 +155,15 @@ def distributed_train(
     # /root/reference/spacy_ray/worker.py:96,222)
     before_to_disk = T.get("before_to_disk")
 
+    # resume: previous best composite score (written into model-best's
+    # meta.json) seeds the loop so a worse resumed model never
+    # overwrites model-best
+    prev_best = None
+    if resume and output_path:
+        best_meta = Path(output_path) / "model-best" / "meta.json"
+        if best_meta.exists():
+            prev_best = json.loads(best_meta.read_text()).get("best_score")
+
     dev_examples = None
     eval_state: Dict[str, bool] = {}
 
@@ -223,6 +232,7 @@ def distributed_train(
         exclude=T.get("frozen_components") or [],
         annotating_components=T.get("annotating_components") or [],
         before_update=T.get("before_update"),
+        initial_best=prev_best,
     )
     # test hook: SRX_FAULT_INJECT="rank:step" kills this
     # rank at that step so the supervisor's all-ranks-abort path is testable
@@ -257,6 +267,7 @@ def distributed_train(
             # rank 0 writes the pipeline, every rank its optimizer shard.
             if rank == 0:
                 nlp.meta["performance"] = info["other_scores"]
+                nlp.meta["best_score"] = info["score"]
                 to_save = before_to_disk(nlp) if before_to_disk else nlp
                 to_save.to_disk(Path(output_path) / "model-best")
             comm.barrier()

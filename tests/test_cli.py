@@ -219,3 +219,26 @@ def test_bench_two_rank_gloo(tmp_path):
     assert rec["steps"] == 2 and rec["scaling"] == "weak"
     assert rec["config"]["parallelism"] == "dp2"
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
+
+
+def test_resume_does_not_overwrite_better_model_best(tmp_path):
+    """model-best's meta.json carries the composite best_score; a resumed
+    run seeded with it must not overwrite model-best unless it actually
+    beats it."""
+    out = tmp_path / "outrb"
+    r = _run_cli([str(CFG), "--output", str(out), "--training.max_steps", "6",
+                  "--training.eval_frequency", "3"])
+    assert r.returncode == 0, r.stderr[-3000:]
+    meta_path = out / "model-best" / "meta.json"
+    meta = json.loads(meta_path.read_text())
+    assert "best_score" in meta
+    # pretend the previous run reached a near-perfect score
+    meta["best_score"] = 0.999
+    meta["sentinel"] = "previous-best"
+    meta_path.write_text(json.dumps(meta))
+    r2 = _run_cli([str(CFG), "--output", str(out), "--resume",
+                   "--training.max_steps", "6",
+                   "--training.eval_frequency", "3"])
+    assert r2.returncode == 0, r2.stderr[-3000:]
+    meta2 = json.loads(meta_path.read_text())
+    assert meta2.get("sentinel") == "previous-best"  # not overwritten
