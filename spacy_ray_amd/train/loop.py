@@ -19,7 +19,13 @@ from spacy_ray_amd.vocab.doc import Example
 
 def create_train_batches(nlp, corpus, batcher, max_epochs: int):
     """Epoch-looped batch iterator (contract of spaCy create_train_batches,
-    used at `/root/reference/spacy_ray/worker.py:170-175`)."""
+    used at `/root/reference/spacy_ray/worker.py:170-175`).
+
+    Deliberate difference from spaCy: per-epoch shuffling lives in the
+    corpus reader (`SyntheticCorpus(shuffle=...)` / user readers), not
+    here — the corpus is re-streamed each epoch (spaCy's max_epochs=-1
+    behavior) instead of materialized + random.shuffle'd, which keeps
+    epoch order independent of global RNG state across ranks."""
     epoch = 0
     while max_epochs < 1 or epoch < max_epochs:
         examples = corpus(nlp)
