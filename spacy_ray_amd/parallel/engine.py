@@ -333,18 +333,29 @@ class ZeRO1Engine:
             self.master[bkt.shard_off : bkt.shard_off + bkt.per].copy_(src.float())
 
     def state_dict(self) -> Dict:
-        return {
+        out = {
             "step": self.step_count,
             "master": self.master,
             "exp_avg": self.exp_avg,
             "exp_avg_sq": self.exp_avg_sq,
         }
+        if self.avg is not None:
+            out["avg"] = self.avg
+        return out
 
     def load_state_dict(self, state: Dict) -> None:
         self.step_count = int(state["step"])
         self.master.copy_(state["master"])
         self.exp_avg.copy_(state["exp_avg"])
         self.exp_avg_sq.copy_(state["exp_avg_sq"])
+        if self.avg is not None:
+            if "avg" in state:
+                self.avg.copy_(state["avg"])
+            else:
+                # old checkpoint without the average: restart it from the
+                # restored master rather than the stale init-time clone
+                # (which would dominate the running mean at large step_count)
+                self.avg.copy_(self.master)
         r = self.comm.rank
         for bkt in self.buckets:
             dst = self.flat_param[bkt.start + r * bkt.per : bkt.start + (r + 1) * bkt.per]

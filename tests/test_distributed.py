@@ -328,3 +328,35 @@ def test_frozen_components_not_updated(tmp_path):
         assert torch.equal(b, a)  # bit-identical: no update, no decay
     # the trainable head did move
     assert any(p.requires_grad for p in tagger.module.parameters())
+
+
+def test_use_averages_survives_resume():
+    """state_dict round-trip carries the running parameter average; without
+    it the stale init-time avg would dominate the mean after resume."""
+    nlp, T, examples = _make_nlp_and_examples(4)
+    spec = T["optimizer"]
+    spec.use_averages = True
+    engine = ZeRO1Engine(nlp, spec, LocalComm())
+    for _ in range(3):
+        engine.accumulate(examples)
+        engine.apply_step()
+    saved = {k: (v.clone() if torch.is_tensor(v) else v)
+             for k, v in engine.state_dict().items()}
+    assert "avg" in saved
+
+    nlp2, T2, _ = _make_nlp_and_examples(4)
+    spec2 = T2["optimizer"]
+    spec2.use_averages = True
+    engine2 = ZeRO1Engine(nlp2, spec2, LocalComm())
+    engine2.load_state_dict(saved)
+    assert torch.equal(engine2.avg, engine.avg)
+    assert engine2.step_count == engine.step_count
+
+    # legacy checkpoint without 'avg': falls back to restored master
+    legacy = {k: v for k, v in saved.items() if k != "avg"}
+    nlp3, T3, _ = _make_nlp_and_examples(4)
+    spec3 = T3["optimizer"]
+    spec3.use_averages = True
+    engine3 = ZeRO1Engine(nlp3, spec3, LocalComm())
+    engine3.load_state_dict(legacy)
+    assert torch.equal(engine3.avg, engine3.master)
