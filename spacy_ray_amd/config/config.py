@@ -61,7 +61,6 @@ class Config(dict):
                  interpolate: bool = False) -> "Config":
         cfg = cls()
         section: Optional[Dict[str, Any]] = None
-        pending_comment = False
         for lineno, line in enumerate(text.splitlines(), 1):
             stripped = line.strip()
             if not stripped or stripped.startswith(("#", ";")):
