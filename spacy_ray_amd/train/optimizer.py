@@ -8,7 +8,7 @@ engine in parallel/ (ZeRO-1 over RCCL).
 
 Divergences from thinc, by design: grad clipping is global-norm over the
 whole flat gradient (thinc clips per tensor) — one fused kernel over the
-flat buffer; `use_averages` is not implemented in round 1.
+flat buffer; `use_averages` keeps a running parameter mean that is swapped in for eval and checkpoints (engine.averaged_params).
 """
 from __future__ import annotations
 

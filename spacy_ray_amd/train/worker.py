@@ -265,17 +265,22 @@ def distributed_train(
         if is_best_checkpoint and output_path:
             # all ranks participate: params are replicated post-all-gather;
             # rank 0 writes the pipeline, every rank its optimizer shard.
-            if rank == 0:
-                nlp.meta["performance"] = info["other_scores"]
-                nlp.meta["best_score"] = info["score"]
-                to_save = before_to_disk(nlp) if before_to_disk else nlp
-                to_save.to_disk(Path(output_path) / "model-best")
-            comm.barrier()
+            # Saved under the AVERAGED params when use_averages is on (the
+            # spaCy `with nlp.use_params(optimizer.averages)` contract) so
+            # the checkpoint reproduces the eval score; no-op otherwise.
+            with engine.averaged_params():
+                if rank == 0:
+                    nlp.meta["performance"] = info["other_scores"]
+                    nlp.meta["best_score"] = info["score"]
+                    to_save = before_to_disk(nlp) if before_to_disk else nlp
+                    to_save.to_disk(Path(output_path) / "model-best")
+                comm.barrier()
     if output_path:
-        if rank == 0:
-            to_save = before_to_disk(nlp) if before_to_disk else nlp
-            to_save.to_disk(Path(output_path) / "model-last")
-        comm.barrier()
+        with engine.averaged_params():
+            if rank == 0:
+                to_save = before_to_disk(nlp) if before_to_disk else nlp
+                to_save.to_disk(Path(output_path) / "model-last")
+            comm.barrier()
         torch.save(engine.state_dict(), Path(output_path) / "model-last" / f"optim.rank{rank}.pt")
     if rank == 0:
         finalize_logger()

@@ -242,3 +242,46 @@ def test_resume_does_not_overwrite_better_model_best(tmp_path):
     assert r2.returncode == 0, r2.stderr[-3000:]
     meta2 = json.loads(meta_path.read_text())
     assert meta2.get("sentinel") == "previous-best"  # not overwritten
+
+
+def test_use_averages_checkpoints_hold_averaged_params(tmp_path):
+    """With use_averages, saved checkpoints must contain the running
+    parameter AVERAGE (what eval scored), not the live trained params —
+    the spaCy `nlp.use_params(optimizer.averages)` save contract."""
+    import torch
+
+    import spacy_ray_amd
+    import spacy_ray_amd.train.worker as wm
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.train.worker import distributed_train
+
+    captured = {}
+    orig_engine = wm.ZeRO1Engine
+
+    class Capture(orig_engine):
+        def __init__(self, *a, **k):
+            super().__init__(*a, **k)
+            captured["engine"] = self
+
+    wm.ZeRO1Engine = Capture
+    out = tmp_path / "outavg"
+    try:
+        distributed_train(
+            Config.from_disk(CFG, overrides={
+                "training.max_steps": 6, "training.eval_frequency": 3,
+                "training.optimizer.use_averages": True}),
+            output_path=out, use_gpu=-1)
+    finally:
+        wm.ZeRO1Engine = orig_engine
+    engine = captured["engine"]
+    nlp = engine.nlp
+    nlp2 = spacy_ray_amd.load(out / "model-last")
+    live = [p.detach().clone() for p in nlp.torch_module().parameters()]
+    with engine.averaged_params():
+        avg = [p.detach().clone() for p in nlp.torch_module().parameters()]
+    saved = list(nlp2.torch_module().parameters())
+    # averages differ from live after a few steps ...
+    assert any(not torch.allclose(l, a) for l, a in zip(live, avg))
+    # ... and the checkpoint matches the AVERAGED params
+    for a, s in zip(avg, saved):
+        assert torch.allclose(a.float(), s.float(), atol=1e-6), "saved != averaged"
