@@ -109,7 +109,9 @@ def train_while_improving(
         if max_steps and step >= max_steps - 1:
             break
         if patience and results:
-            best_step = max(results, key=lambda r: r[0])[1]
+            # tuple max: ties prefer the LATER step (spaCy semantics),
+            # so a plateau doesn't count against patience from its start
+            best_step = max(results)[1]
             if (step - best_step) >= patience * eval_frequency and len(results) >= patience:
                 break
     nlp._frozen = []
