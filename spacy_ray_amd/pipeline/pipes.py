@@ -154,7 +154,10 @@ class TaggerPipe(TrainablePipe):
         # per-doc gold-id arrays are cached on the reference Doc (dict
         # lookups over every token measured ~50 ms/step at 512k words;
         # corpora recycle docs across epochs so the cache pays)
-        key = ("tag_ids", id(self.label2id))
+        # keyed by the label LIST identity-proof tuple, not id() — a
+        # freed dict's address can be reused by a rebuilt pipeline and
+        # would hit stale cached ids on shared docs
+        key = ("tag_ids", tuple(self.labels))
         parts = []
         for eg in examples:
             ref = eg.reference
