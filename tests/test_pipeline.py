@@ -384,3 +384,36 @@ def test_annotating_components_set_predictions_before_loss():
         assert eg.predicted.tags is not None
         assert len(eg.predicted.tags) == len(eg.predicted)
     total.backward()  # annotation ran under no_grad; loss graph intact
+
+
+@pytest.mark.parametrize("pipes", [["tok2vec", "parser"], ["tok2vec", "ner"]])
+def test_single_head_pipelines_train_and_predict(pipes):
+    """Unusual pipeline shapes (parser-only / NER-only heads) must train and
+    annotate — no hidden dependency on the full en_core lineup."""
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    cfg["nlp"]["pipeline"] = list(pipes)
+    for extra in ("tagger", "parser", "ner"):
+        if extra not in pipes and extra in cfg["components"]:
+            del cfg["components"][extra]
+    nlp = init_nlp(cfg, sample_size=24)
+    assert nlp.pipe_names == pipes
+    icfg = cfg.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    (train_corpus,) = resolve_dot_names(icfg, ["corpora.train"])
+    examples = []
+    for eg in train_corpus(nlp):
+        examples.append(eg)
+        if len(examples) >= 12:
+            break
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    losses = {}
+    stepper.accumulate(examples, losses=losses)
+    stepper.apply_step()
+    head = pipes[1]
+    assert head in losses and np.isfinite(losses[head])
+    docs = nlp.predict_docs([eg.predicted.copy_unannotated() for eg in examples[:3]])
+    for d in docs:
+        if head == "parser":
+            assert d.heads is not None and len(d.heads) == len(d)
+        else:
+            assert d.ents is not None and len(d.ents) == len(d)
