@@ -136,6 +136,97 @@ def ray_evaluate_cli(
     print(json.dumps(scores, indent=2))
 
 
+debug_app = typer.Typer(name="debug", no_args_is_help=True,
+                        help="Validate configs and data before training")
+app.add_typer(debug_app)
+
+
+@debug_app.command(
+    "config",
+    context_settings={"allow_extra_args": True, "ignore_unknown_options": True},
+)
+def debug_config_cli(
+    ctx: typer.Context,
+    config_path: Path = typer.Argument(..., help="Path to config file"),
+    code_path: Optional[Path] = typer.Option(None, "--code", "-c"),
+):
+    """Parse, interpolate and resolve a config; build the pipeline skeleton.
+    Exits nonzero with the offending section on any error."""
+    from spacy_ray_amd.config.config import resolve
+    from spacy_ray_amd.config.schemas import ConfigSchemaTraining
+    from spacy_ray_amd.pipeline.language import build_nlp
+
+    if code_path:
+        import importlib.util
+
+        spec = importlib.util.spec_from_file_location("srx_user_code", code_path)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+    overrides = parse_config_overrides(list(ctx.args))
+    config = Config.from_disk(config_path, overrides=overrides)
+    icfg = config.interpolate()
+    T = resolve(icfg.get("training", {}), schema=ConfigSchemaTraining)
+    nlp = build_nlp(config)
+    print(f"[+] config OK: pipeline {nlp.pipe_names}")
+    print(f"[+] training resolved: optimizer={type(T['optimizer']).__name__}, "
+          f"max_steps={T.get('max_steps')}, eval_frequency={T.get('eval_frequency')}, "
+          f"score_weights={T.get('score_weights')}")
+
+
+@debug_app.command("data")
+def debug_data_cli(
+    config_path: Path = typer.Argument(..., help="Path to config file"),
+    code_path: Optional[Path] = typer.Option(None, "--code", "-c"),
+    limit: int = typer.Option(1000, "--limit", help="Max docs to scan per corpus"),
+):
+    """Load the train/dev corpora and report doc/token counts and label
+    inventories; warns on empty corpora or train/dev label mismatches."""
+    from spacy_ray_amd.config.config import resolve, resolve_dot_names
+    from spacy_ray_amd.pipeline.language import build_nlp
+
+    if code_path:
+        import importlib.util
+
+        spec = importlib.util.spec_from_file_location("srx_user_code", code_path)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+    config = Config.from_disk(config_path)
+    icfg = config.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    nlp = build_nlp(config)
+    warned = False
+    stats = {}
+    for split, dot in (("train", T.get("train_corpus", "corpora.train")),
+                       ("dev", T.get("dev_corpus", "corpora.dev"))):
+        (corpus,) = resolve_dot_names(icfg, [dot])
+        n_docs = n_tokens = 0
+        tags, deps, ents = set(), set(), set()
+        for eg in corpus(nlp):
+            ref = eg.reference
+            n_docs += 1
+            n_tokens += len(ref)
+            if ref.tags:
+                tags.update(ref.tags)
+            if ref.deps:
+                deps.update(ref.deps)
+            if ref.ents:
+                ents.update(t.partition("-")[2] for t in ref.ents if t not in ("O", ""))
+            if n_docs >= limit:
+                break
+        stats[split] = (tags, deps, ents)
+        print(f"[+] {split}: {n_docs} docs, {n_tokens} tokens, "
+              f"{len(tags)} tags, {len(deps)} dep labels, {len(ents)} entity types")
+        if n_docs == 0:
+            print(f"[!] {split} corpus is EMPTY")
+            warned = True
+    for i, kind in enumerate(("tags", "dep labels", "entity types")):
+        only_dev = stats["dev"][i] - stats["train"][i]
+        if only_dev:
+            print(f"[!] {kind} in dev but never in train: {sorted(only_dev)[:10]}")
+            warned = True
+    raise SystemExit(1 if warned else 0)
+
+
 @app.command("convert")
 def convert_cli(
     input_path: Path = typer.Argument(..., help="CoNLL-U (.conllu) or IOB (.iob) file"),

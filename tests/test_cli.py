@@ -285,3 +285,28 @@ def test_use_averages_checkpoints_hold_averaged_params(tmp_path):
     # ... and the checkpoint matches the AVERAGED params
     for a, s in zip(avg, saved):
         assert torch.allclose(a.float(), s.float(), atol=1e-6), "saved != averaged"
+
+
+def _run_sub(args, timeout=240):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    return subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", *args],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=timeout,
+    )
+
+
+def test_debug_config_ok_and_fails_on_bad():
+    r = _run_sub(["debug", "config", str(CFG)])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "config OK" in r.stdout
+    # a broken override must exit nonzero
+    r2 = _run_sub(["debug", "config", str(CFG), "--training.dropout", "5.0"])
+    assert r2.returncode != 0
+
+
+def test_debug_data_reports_counts(tmp_path):
+    r = _run_sub(["debug", "data", str(CFG), "--limit", "50"])
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    assert "train:" in r.stdout and "dev:" in r.stdout
+    assert "tokens" in r.stdout
