@@ -136,6 +136,37 @@ def ray_evaluate_cli(
     print(json.dumps(scores, indent=2))
 
 
+init_app = typer.Typer(name="init", no_args_is_help=True,
+                       help="Generate starter configs")
+app.add_typer(init_app)
+
+
+@init_app.command("config")
+def init_config_cli(
+    output_path: Path = typer.Argument(..., help="Where to write the config (use - for stdout)"),
+    lang: str = typer.Option("en", "--lang", "-l"),
+    pipeline: str = typer.Option("tagger,parser,ner", "--pipeline", "-p",
+                                 help="Comma-separated: tagger, parser, ner"),
+    arch: str = typer.Option("cnn", "--arch", help="tok2vec architecture: cnn | trf"),
+    width: int = typer.Option(0, "--width", help="tok2vec width (default: 96 cnn / 768 trf)"),
+    gpu: bool = typer.Option(False, "--gpu", help="Tune defaults for GPU training"),
+):
+    """Generate a ready-to-train config (the `spacy init config` role)."""
+    from spacy_ray_amd.cli.templates import render_config
+
+    pipes = [p.strip() for p in pipeline.split(",") if p.strip()]
+    bad = [p for p in pipes if p not in ("tagger", "parser", "ner")]
+    if bad:
+        raise SystemExit(f"unknown pipeline components: {bad}")
+    text = render_config(lang=lang, pipes=pipes, arch=arch, width=width, gpu=gpu)
+    if str(output_path) == "-":
+        print(text)
+    else:
+        output_path.write_text(text)
+        print(f"[+] wrote {output_path} — train with: "
+              f"spacy-mi ray train {output_path} --output ./model")
+
+
 debug_app = typer.Typer(name="debug", no_args_is_help=True,
                         help="Validate configs and data before training")
 app.add_typer(debug_app)

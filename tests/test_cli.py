@@ -310,3 +310,38 @@ def test_debug_data_reports_counts(tmp_path):
     assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
     assert "train:" in r.stdout and "dev:" in r.stdout
     assert "tokens" in r.stdout
+
+
+def test_init_config_generates_trainable_config(tmp_path):
+    """init config -> debug config passes -> a short training run works when
+    pointed at a converted DocBin corpus."""
+    cfg_path = tmp_path / "gen.cfg"
+    r = _run_sub(["init", "config", str(cfg_path),
+                  "--pipeline", "tagger,ner", "--arch", "cnn", "--width", "32"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert cfg_path.exists()
+    r2 = _run_sub(["debug", "config", str(cfg_path)])
+    assert r2.returncode == 0, (r2.stdout[-500:], r2.stderr[-2000:])
+
+    # make a tiny corpus and train 2 steps through the generated config
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    v = Vocab()
+    docs = [Doc(v, [f"w{i}", "x", "y"], tags=["A", "B", "A"],
+                ents=["U-ORG", "O", "O"]) for i in range(40)]
+    bin_path = tmp_path / "train.spacy"
+    DocBin(docs).to_disk(bin_path)
+    r3 = _run_cli([str(cfg_path), "--output", str(tmp_path / "m"),
+                   "--paths.train", str(bin_path), "--paths.dev", str(bin_path),
+                   "--training.max_steps", "2", "--training.eval_frequency", "2",
+                   "--training.batcher.size", "200"])
+    assert r3.returncode == 0, r3.stderr[-3000:]
+    assert (tmp_path / "m" / "model-last" / "config.cfg").exists()
+
+
+def test_init_config_trf_stdout():
+    r = _run_sub(["init", "config", "-", "--arch", "trf", "--pipeline", "tagger"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "spacy-transformers.TransformerModel.v3" in r.stdout
+    assert "ner" not in r.stdout.split("[nlp]")[1].split("[components]")[0]
