@@ -78,7 +78,11 @@ class Doc:
         return "".join(out).rstrip()
 
     def copy_unannotated(self) -> "Doc":
-        return Doc(self.vocab, self.words, spaces=self.spaces)
+        # words are identical, so the (n,4) attr-hash array is shared (it is
+        # never mutated) — re-hashing every token per epoch/predict call was
+        # pure waste on the Example.from_doc path
+        return Doc(self.vocab, self.words, spaces=self.spaces,
+                   attr_hashes=self.attr_hashes)
 
     def to_dict(self) -> Dict:
         return {
