@@ -103,3 +103,52 @@ def test_hash4_fuzz_stable_and_matches_general(ids, seed):
         h3 = _srx_cpu.hash4(arr, 7)
         if len(set(arr.tolist())) == len(arr) and len(arr) > 2:
             assert not (h1 == h3).all()
+
+
+@given(
+    st.lists(st.integers(min_value=1, max_value=12), min_size=1, max_size=6),
+    st.integers(min_value=1, max_value=4),
+    st.integers(min_value=0, max_value=10_000),
+)
+@settings(max_examples=60, deadline=None)
+def test_biluo_oracle_fuzz_reconstructs_random_gold(lens, n_types, seed):
+    """For ANY random valid BILUO tag sequence, greedily following min-cost
+    actions (random tie-break) must reproduce the gold tags exactly."""
+    rng = np.random.RandomState(seed)
+    golds = []
+    for n in lens:
+        codes = np.zeros(n, dtype=np.int32)
+        i = 0
+        while i < n:
+            t = rng.randint(n_types)
+            if rng.rand() < 0.5:
+                codes[i] = 0  # O
+                i += 1
+            elif rng.rand() < 0.5 or i == n - 1:
+                codes[i] = 1 + 4 * t + 3  # U
+                i += 1
+            else:
+                span = rng.randint(2, min(5, n - i) + 1)
+                codes[i] = 1 + 4 * t + 0            # B
+                for j in range(1, span - 1):
+                    codes[i + j] = 1 + 4 * t + 1     # I
+                codes[i + span - 1] = 1 + 4 * t + 2  # L
+                i += span
+        golds.append(codes)
+    gold_flat = np.concatenate(golds)
+    b = _srx_cpu.BiluoBatch(np.asarray(lens, dtype=np.int32), n_types)
+    b.set_gold(gold_flat)
+    for _ in range(max(lens) + 2):
+        if b.is_final().all():
+            break
+        costs = b.costs()
+        valid = b.valid().astype(bool)
+        cmin = np.where(valid, costs, np.inf).min(axis=1, keepdims=True)
+        is_gold = (costs <= cmin + 1e-6) & valid
+        # random tie-break among min-cost actions
+        noise = rng.rand(*is_gold.shape)
+        pick = np.where(is_gold, noise, -1.0).argmax(axis=1).astype(np.int32)
+        pick[b.is_final().astype(bool)] = -1
+        b.advance(pick)
+    assert b.is_final().all()
+    assert (b.tags() == gold_flat).all()
