@@ -279,11 +279,13 @@ def serve_cli(
     port: int = typer.Option(8000, "--port", "-p"),
     use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
     max_batch: int = typer.Option(256, "--max-batch", help="Max docs per decode batch"),
+    max_wait_ms: float = typer.Option(5.0, "--max-wait-ms", help="Micro-batcher wait for coalescing concurrent requests"),
 ):
     """Serve a trained pipeline over HTTP (POST /annotate, GET /info)."""
     from spacy_ray_amd.serve.app import serve
 
-    serve(model_path, host=host, port=port, use_gpu=use_gpu, max_batch=max_batch)
+    serve(model_path, host=host, port=port, use_gpu=use_gpu,
+          max_batch=max_batch, max_wait_ms=max_wait_ms)
 
 
 def main() -> None:

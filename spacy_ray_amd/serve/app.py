@@ -148,7 +148,8 @@ def build_app(nlp, max_batch: int = 256, max_wait_ms: float = 5.0):
 
 
 def serve(model_path, host: str = "127.0.0.1", port: int = 8000,
-          use_gpu: int = -1, max_batch: int = 256) -> None:
+          use_gpu: int = -1, max_batch: int = 256,
+          max_wait_ms: float = 5.0) -> None:
     """Load a checkpoint directory and serve it (blocking)."""
     import torch
     import uvicorn
@@ -161,4 +162,5 @@ def serve(model_path, host: str = "127.0.0.1", port: int = 8000,
         else "cpu"
     )
     nlp = spacy_ray_amd.load(model_path, device=device)
-    uvicorn.run(build_app(nlp, max_batch=max_batch), host=host, port=port)
+    uvicorn.run(build_app(nlp, max_batch=max_batch, max_wait_ms=max_wait_ms),
+                host=host, port=port)
