@@ -417,3 +417,29 @@ def test_single_head_pipelines_train_and_predict(pipes):
             assert d.heads is not None and len(d.heads) == len(d)
         else:
             assert d.ents is not None and len(d.ents) == len(d)
+
+
+def test_annotating_components_transition_pipe():
+    """annotating_components with a transition pipe (parser): predictions
+    land on eg.predicted through the shard-split decode path."""
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    nlp = init_nlp(cfg, sample_size=24)
+    icfg = cfg.interpolate()
+    (train_corpus,) = resolve_dot_names(icfg, ["corpora.train"])
+    examples = []
+    for eg in train_corpus(nlp):
+        examples.append(eg)
+        if len(examples) >= 6:
+            break
+    for eg in examples:
+        eg.predicted.heads = None
+    nlp._annotating = ["parser", "ner"]
+    try:
+        total, losses = nlp.forward_loss(examples)
+    finally:
+        nlp._annotating = []
+    for eg in examples:
+        assert eg.predicted.heads is not None
+        assert len(eg.predicted.heads) == len(eg.predicted)
+        assert eg.predicted.ents is not None
+    total.backward()
