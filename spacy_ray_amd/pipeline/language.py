@@ -216,6 +216,12 @@ class Language:
         path = Path(path)
         path.mkdir(parents=True, exist_ok=True)
         self.config.to_disk(path / "config.cfg")
+        # spaCy meta shape: pipeline order + per-component labels
+        self.meta["pipeline"] = self.pipe_names
+        self.meta["labels"] = {
+            name: list(getattr(pipe, "labels", []) or [])
+            for name, pipe in self.pipeline
+        }
         (path / "meta.json").write_text(json.dumps(self.meta, indent=2))
         vocab_dir = path / "vocab"
         vocab_dir.mkdir(exist_ok=True)
