@@ -53,3 +53,24 @@ def test_batch_by_padded_size_bound():
     items = [_words(n) for n in (5, 5, 5, 9, 2)]
     for bt in b(items):
         assert max(len(x) for x in bt) * len(bt) <= 20
+
+
+def test_schedules_contracts():
+    from spacy_ray_amd.config.config import resolve
+
+    wl = resolve({"@schedules": "warmup_linear.v1", "initial_rate": 0.1,
+                  "warmup_steps": 10, "total_steps": 110})
+    assert wl(0) == 0.01                      # linear ramp
+    assert wl(9) == 0.1                       # peak at end of warmup
+    assert abs(wl(60) - 0.05) < 1e-9          # halfway down
+    assert wl(110) == 0.0 and wl(1000) == 0.0
+
+    comp = resolve({"@schedules": "compounding.v1", "start": 2.0, "stop": 16.0,
+                    "compound": 2.0})
+    assert [comp(i) for i in range(5)] == [2.0, 4.0, 8.0, 16.0, 16.0]
+    # iterator protocol (batchers call next())
+    it = iter(comp)
+    assert [next(it) for _ in range(4)] == [2.0, 4.0, 8.0, 16.0]
+
+    const = resolve({"@schedules": "constant.v1", "rate": 0.3})
+    assert const(0) == const(999) == 0.3
