@@ -14,7 +14,7 @@ spacy-mi convert dev.conllu dev.spacy
 
 # 3. pre-flight checks
 spacy-mi debug config quickstart.cfg --paths.train train.spacy --paths.dev dev.spacy
-spacy-mi debug data quickstart.cfg
+spacy-mi debug data quickstart.cfg --paths.train train.spacy --paths.dev dev.spacy
 
 # 4. train on 8 GPUs (one process per GPU, ZeRO-1 over RCCL/xGMI)
 spacy-mi ray train quickstart.cfg \

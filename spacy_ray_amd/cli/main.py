@@ -204,8 +204,12 @@ def debug_config_cli(
           f"score_weights={T.get('score_weights')}")
 
 
-@debug_app.command("data")
+@debug_app.command(
+    "data",
+    context_settings={"allow_extra_args": True, "ignore_unknown_options": True},
+)
 def debug_data_cli(
+    ctx: typer.Context,
     config_path: Path = typer.Argument(..., help="Path to config file"),
     code_path: Optional[Path] = typer.Option(None, "--code", "-c"),
     limit: int = typer.Option(1000, "--limit", help="Max docs to scan per corpus"),
@@ -221,7 +225,8 @@ def debug_data_cli(
         spec = importlib.util.spec_from_file_location("srx_user_code", code_path)
         mod = importlib.util.module_from_spec(spec)
         spec.loader.exec_module(mod)
-    config = Config.from_disk(config_path)
+    overrides = parse_config_overrides(list(ctx.args))
+    config = Config.from_disk(config_path, overrides=overrides)
     icfg = config.interpolate()
     T = resolve(icfg["training"], validate=False)
     nlp = build_nlp(config)

@@ -345,3 +345,22 @@ def test_init_config_trf_stdout():
     assert r.returncode == 0, r.stderr[-2000:]
     assert "spacy-transformers.TransformerModel.v3" in r.stdout
     assert "ner" not in r.stdout.split("[nlp]")[1].split("[components]")[0]
+
+
+def test_debug_data_with_path_overrides(tmp_path):
+    """debug data accepts --paths.* overrides like train does (the
+    quickstart flow: generated config + converted DocBin paths)."""
+    cfg_path = tmp_path / "gen.cfg"
+    assert _run_sub(["init", "config", str(cfg_path),
+                     "--pipeline", "tagger"]).returncode == 0
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    v = Vocab()
+    docs = [Doc(v, ["a", "b"], tags=["X", "Y"]) for _ in range(10)]
+    bin_path = tmp_path / "d.spacy"
+    DocBin(docs).to_disk(bin_path)
+    r = _run_sub(["debug", "data", str(cfg_path),
+                  "--paths.train", str(bin_path), "--paths.dev", str(bin_path)])
+    assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
+    assert "10 docs" in r.stdout
