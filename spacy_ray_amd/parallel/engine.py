@@ -335,6 +335,7 @@ class ZeRO1Engine:
     def state_dict(self) -> Dict:
         out = {
             "step": self.step_count,
+            "world": self.comm.world,
             "master": self.master,
             "exp_avg": self.exp_avg,
             "exp_avg_sq": self.exp_avg_sq,
@@ -344,6 +345,15 @@ class ZeRO1Engine:
         return out
 
     def load_state_dict(self, state: Dict) -> None:
+        saved_world = int(state.get("world", self.comm.world))
+        if saved_world != self.comm.world:
+            raise ValueError(
+                f"optimizer shard was saved at world_size={saved_world} but "
+                f"this run has world_size={self.comm.world} — the ZeRO-1 "
+                f"shard layout depends on world size; resume with the same "
+                f"--n-workers, or delete the optim.rank*.pt files to resume "
+                f"from params only"
+            )
         self.step_count = int(state["step"])
         self.master.copy_(state["master"])
         self.exp_avg.copy_(state["exp_avg"])

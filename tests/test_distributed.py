@@ -360,3 +360,18 @@ def test_use_averages_survives_resume():
     engine3 = ZeRO1Engine(nlp3, spec3, LocalComm())
     engine3.load_state_dict(legacy)
     assert torch.equal(engine3.avg, engine3.master)
+
+
+def test_resume_world_size_mismatch_clear_error():
+    """Loading an optimizer shard saved at a different world size must fail
+    with a clear message (the shard layout depends on world size)."""
+    nlp, T, examples = _make_nlp_and_examples(4)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    engine.accumulate(examples)
+    engine.apply_step()
+    state = engine.state_dict()
+    state["world"] = 4
+    nlp2, T2, _ = _make_nlp_and_examples(4)
+    engine2 = ZeRO1Engine(nlp2, T2["optimizer"], LocalComm())
+    with pytest.raises(ValueError, match="world_size"):
+        engine2.load_state_dict(state)
