@@ -109,6 +109,13 @@ def distributed_train(
         spec.loader.exec_module(mod)
 
     if use_gpu >= 0 and torch.cuda.is_available():
+        n_dev = torch.cuda.device_count()
+        if local_rank >= n_dev:
+            raise SystemExit(
+                f"rank {rank}: LOCAL_RANK {local_rank} needs GPU {local_rank} "
+                f"but only {n_dev} visible — --n-workers must be <= GPUs per "
+                f"node (or restrict with HIP_VISIBLE_DEVICES)"
+            )
         device = f"cuda:{local_rank}"
         torch.cuda.set_device(local_rank)
     else:
