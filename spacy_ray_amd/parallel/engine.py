@@ -203,7 +203,10 @@ class ZeRO1Engine:
         total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop,
                                          token_batch=token_batch)
         with timing.phase("bwd/main"):
-            total.backward()
+            if total.requires_grad:
+                total.backward()
+            # else: every loss-producing pipe is frozen — a valid (if odd)
+            # configuration; the step is a no-op rather than a crash
         self._sync = False
         self.last_compute_ms = (_time.perf_counter() - t0) * 1000
 

@@ -375,3 +375,16 @@ def test_resume_world_size_mismatch_clear_error():
     engine2 = ZeRO1Engine(nlp2, T2["optimizer"], LocalComm())
     with pytest.raises(ValueError, match="world_size"):
         engine2.load_state_dict(state)
+
+
+def test_all_frozen_step_is_noop_not_crash():
+    nlp, T, examples = _make_nlp_and_examples(4)
+    for _, pipe in nlp.pipeline:
+        for p in pipe.module.parameters():
+            p.requires_grad_(False)
+    nlp._frozen = [n for n, _ in nlp.pipeline]
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    assert engine.flat_param.numel() == 0
+    engine.accumulate(examples)
+    engine.apply_step()  # no params, no crash
+    nlp._frozen = []
