@@ -137,3 +137,16 @@ def test_micro_batcher_respects_max_batch():
     out = asyncio.run(main())
     assert out == [["1", "2"], ["3", "4"]]
     assert calls == [2, 2]  # two groups, not one of 4
+
+
+def test_annotate_empty_and_whitespace_texts(served_nlp):
+    from fastapi.testclient import TestClient
+
+    from spacy_ray_amd.serve.app import build_app
+
+    client = TestClient(build_app(served_nlp))
+    r = client.post("/annotate", json={"texts": ["", "  ", "ok then"]})
+    assert r.status_code == 200
+    docs = r.json()["docs"]
+    assert docs[0]["words"] == [] and docs[1]["words"] == []
+    assert docs[2]["words"] == ["ok", "then"]
