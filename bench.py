@@ -120,8 +120,21 @@ def main() -> None:
         return
 
     # warmup must cover EVERY distinct batch shape: hipBLASLt runs solution
-    # selection per new GEMM M, which otherwise lands in the timed region
-    for i in range(max(args.warmup, len(batches))):
+    # selection per new GEMM M, which otherwise lands in the timed region.
+    # The count must be RANK-UNIFORM: every step issues the same fixed
+    # sequence of collectives, so ranks running different warmup counts
+    # (len(batches) varies with the rank-seeded doc lengths) deadlock —
+    # one rank's reduce-scatter pairs with another's barrier.  Take the
+    # global max.
+    n_warm = max(args.warmup, len(batches))
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([float(n_warm)], dtype=torch.float64)
+        t_dev = t.to(device) if comm.backend == "nccl" else t
+        dist.all_reduce(t_dev, op=dist.ReduceOp.MAX)
+        n_warm = int(t_dev.item())
+    for i in range(n_warm):
         step(i)
 
     comm.barrier()

@@ -198,18 +198,21 @@ def tiny_corpus(seed: int = 0):
     assert (out / "model-last" / "config.cfg").exists()
 
 
-def test_bench_two_rank_gloo(tmp_path):
+@pytest.mark.parametrize("world", [2, 4])
+def test_bench_multirank_gloo(tmp_path, world):
     """The driver's SCALE run launches bench.py under torch.distributed.run
-    with N ranks; exercise that exact path on CPU/gloo with 2 ranks and
-    check rank 0 prints one valid JSON line with whole-job aggregation."""
+    with N ranks; exercise that exact path on CPU/gloo and check rank 0
+    prints one valid JSON line.  world=4 regresses the warmup-count
+    deadlock: per-rank len(batches) differ, so the warmup step count must
+    be the global max or ranks issue mismatched collective sequences."""
     env = dict(os.environ)
     env["PYTHONPATH"] = str(REPO)
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29531", str(REPO / "bench.py"),
-         "--gpus", "2", "--steps", "2", "--warmup", "1",
-         "--batch-words", "2000"],
+         "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+         "--master-port", str(29530 + world), str(REPO / "bench.py"),
+         "--gpus", str(world), "--steps", "2", "--warmup", "1",
+         "--batch-words", "1500"],
         cwd=str(REPO), env=env, capture_output=True, text=True, timeout=600,
     )
     assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-2500:])
@@ -217,7 +220,7 @@ def test_bench_two_rank_gloo(tmp_path):
     assert len(json_lines) == 1, r.stdout[-1500:]
     rec = json.loads(json_lines[0])
     assert rec["steps"] == 2 and rec["scaling"] == "weak"
-    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["parallelism"] == f"dp{world}"
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
 
 
