@@ -71,3 +71,29 @@ def test_hash_string_stable():
     assert h1 == h2[0]
     assert h2[0] != h2[1]
     assert _srx_cpu.hash_string("") != _srx_cpu.hash_string(" ")
+
+
+def test_word_shape_contract():
+    """spaCy word_shape semantics: case/digit classes, run truncation at 4,
+    long-string short-circuit."""
+    from spacy_ray_amd.vocab.attrs import word_shape
+
+    assert word_shape("Apple") == "Xxxxx"
+    assert word_shape("USA") == "XXX"
+    assert word_shape("C3PO") == "XdXX"
+    assert word_shape("don't") == "xxx'x"
+    assert word_shape("123456789") == "dddd"        # run capped at 4
+    assert word_shape("aaaaaaaa") == "xxxx"
+    assert word_shape("aaaaB") == "xxxxX"           # cap resets on class change
+    assert word_shape("x" * 100) == "LONG"
+    assert word_shape("") == ""
+
+
+def test_attr_strings_contract():
+    from spacy_ray_amd.vocab.attrs import attr_strings
+
+    norm, prefix, suffix, shape = attr_strings("Apple")
+    assert norm == "apple" and prefix == "A" and suffix == "ple"
+    assert shape == "Xxxxx"
+    # short words: suffix is the whole word, prefix first char
+    assert attr_strings("ab")[1:3] == ["a", "ab"]
