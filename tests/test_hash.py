@@ -97,3 +97,24 @@ def test_attr_strings_contract():
     assert shape == "Xxxxx"
     # short words: suffix is the whole word, prefix first char
     assert attr_strings("ab")[1:3] == ["a", "ab"]
+
+
+def test_stringstore_roundtrip_and_select_pipes():
+    from spacy_ray_amd.vocab.strings import StringStore
+
+    ss = StringStore(["hello", "café"])
+    h = ss["hello"]
+    assert ss[h] == "hello"
+    assert "hello" in ss and "missing" not in ss
+    hs = ss.add_batch(["a", "b", "a"])
+    assert ss[int(hs[0])] == "a" and len(hs) == 3
+
+    # select_pipes freezes components for the context duration
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from tests.test_pipeline import TAGGER_CFG
+
+    nlp = init_nlp(Config.from_str(TAGGER_CFG), sample_size=8)
+    with nlp.select_pipes(disable=["tagger"]):
+        assert nlp._frozen == ["tagger"]
+    assert nlp._frozen == []
