@@ -1,5 +1,7 @@
 #!/usr/bin/env bash
-# The full lifecycle on real data (every step is covered by tests/):
+# The full lifecycle on real data (every step is covered by tests/).
+# Without `pip install -e .`, substitute `python -m spacy_ray_amd.cli.main`
+# for `spacy-mi` (identical surface).
 set -euo pipefail
 
 # 0. build the C++ core + gfx950 HIP kernels (in-tree, no network)
