@@ -493,10 +493,16 @@ struct BiluoBatch {
       fill_valid(v, st);
       if (with_gold) {
         uint8_t* g = gold_a.mutable_data(k, 0);
-        int32_t gcode = st.final_state() ? -1 : gold[i][st.i];
-        bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
-        for (py::ssize_t a = 0; a < A; a++)
-          g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
+        int32_t gcode = st.final_state() ? -2 : gold[i][st.i];
+        if (gcode == -1) {
+          // MISSING ('-' / unannotated): all-zero gold row -> the loss masks
+          // the row out entirely (no positive or negative supervision)
+          std::fill(g, g + A, 0);
+        } else {
+          bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
+          for (py::ssize_t a = 0; a < A; a++)
+            g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
+        }
       }
     }
     return py::make_tuple(act, feats, valid_a, gold_a);
@@ -538,8 +544,9 @@ struct BiluoBatch {
       fill_valid(v.data(), st);
       for (py::ssize_t a = 0; a < A; a++) {
         if (!v[a]) { r(s, a) = KInvalid; continue; }
-        int32_t g = st.final_state() ? -1 : gold[s][st.i];
-        r(s, a) = ((int32_t)a == g) ? 0.0f : 1.0f;
+        int32_t g = st.final_state() ? -2 : gold[s][st.i];
+        // g == -1 (missing annotation): every valid action is free
+        r(s, a) = (g == -1 || (int32_t)a == g) ? 0.0f : 1.0f;
       }
     }
     return out;
@@ -575,10 +582,14 @@ struct BiluoBatch {
       fill_valid(v, st);
       if (with_gold) {
         uint8_t* g = gold_a + k * A;
-        int32_t gcode = st.final_state() ? -1 : gold[i][st.i];
-        bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
-        for (py::ssize_t a = 0; a < A; a++)
-          g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
+        int32_t gcode = st.final_state() ? -2 : gold[i][st.i];
+        if (gcode == -1) {
+          std::fill(g, g + A, 0);  // missing: row excluded from the loss
+        } else {
+          bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
+          for (py::ssize_t a = 0; a < A; a++)
+            g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
+        }
       }
     }
     return py::make_tuple(act, packed, (py::ssize_t)6);

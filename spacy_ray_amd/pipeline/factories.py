@@ -17,16 +17,25 @@ def make_transformer_pipe(name: str, model):
     return Tok2VecPipe(name, model)
 
 
+def _with_labels(pipe, labels):
+    """Explicit label list from the component config (spaCy's `labels` /
+    initialize-labels contract): skips corpus discovery entirely."""
+    if labels:
+        pipe.labels = list(labels)
+        pipe.label2id = {t: i for i, t in enumerate(pipe.labels)}
+    return pipe
+
+
 @registry.factories("tagger")
-def make_tagger_pipe(name: str, model):
-    return TaggerPipe(name, model)
+def make_tagger_pipe(name: str, model, labels=None):
+    return _with_labels(TaggerPipe(name, model), labels)
 
 
 @registry.factories("parser")
-def make_parser_pipe(name: str, model):
-    return ParserPipe(name, model)
+def make_parser_pipe(name: str, model, labels=None):
+    return _with_labels(ParserPipe(name, model), labels)
 
 
 @registry.factories("ner")
-def make_ner_pipe(name: str, model):
-    return NerPipe(name, model)
+def make_ner_pipe(name: str, model, labels=None):
+    return _with_labels(NerPipe(name, model), labels)

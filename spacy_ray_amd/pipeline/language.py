@@ -31,6 +31,9 @@ class Language:
         self.meta: Dict = {"lang": vocab.lang, "name": "pipeline", "version": "0.0.0"}
         self._frozen: List[str] = []
         self._annotating: List[str] = []
+        # select_pipes(disable=...): skipped during inference AND frozen for
+        # training (spaCy's disabled-component contract — ADVICE r1)
+        self._disabled: List[str] = []
 
     # ------------------------------------------------------------- pipeline
     @property
@@ -162,6 +165,8 @@ class Language:
             # interleave the transition pipes' decode loops
             trans = []
             for name, pipe in self.pipeline:
+                if name in self._disabled:
+                    continue
                 if isinstance(pipe, _TransitionPipeBase):
                     own = pipe.own_tok2vec(batch)
                     trans.append((pipe,) + pipe.make_predict_task(
@@ -170,7 +175,7 @@ class Language:
                 run_transition_tasks([t[1] for t in trans])
             heads = []
             for name, pipe in self.pipeline:
-                if isinstance(pipe, (Tok2VecPipe, _TransitionPipeBase)):
+                if isinstance(pipe, (Tok2VecPipe, _TransitionPipeBase)) or name in self._disabled:
                     continue
                 own = pipe.own_tok2vec(batch)
                 heads.append((pipe, own if own is not None else t2v))
@@ -256,15 +261,20 @@ class Language:
         return self
 
     def select_pipes(self, disable: Sequence[str]):
+        """Disabled pipes are skipped during inference (predict_docs /
+        __call__ / pipe) and frozen for training, matching spaCy's
+        select_pipes contract."""
         lang = self
 
         class _Ctx:
             def __enter__(self):
                 lang._frozen = list(disable)
+                lang._disabled = list(disable)
                 return lang
 
             def __exit__(self, *a):
                 lang._frozen = []
+                lang._disabled = []
 
         return _Ctx()
 
@@ -315,20 +325,56 @@ def _check_listener_widths(nlp: Language) -> None:
 
 
 def init_nlp(config: Config, device: str = "cpu", sample_size: int = 128) -> Language:
-    """Build + initialize (label discovery on a corpus sample + param init) —
-    the contract of spaCy's init_nlp at `/root/reference/spacy_ray/worker.py:91`.
+    """Build + initialize (label discovery + param init) — the contract of
+    spaCy's init_nlp at `/root/reference/spacy_ray/worker.py:91`.
     Deterministic: seeds torch RNG from [training.seed] BEFORE building so
-    every rank gets identical initial params (SURVEY.md §3.4 invariant)."""
+    every rank gets identical initial params (SURVEY.md §3.4 invariant).
+
+    Label discovery runs over the FULL training corpus (one epoch), like
+    spaCy's init-labels pass — a label first appearing late in the corpus
+    must not crash mid-training.  Components may instead pin labels
+    explicitly via `labels = [...]` in their [components.*] block, which
+    skips discovery for that pipe.  Module shape-init still uses only the
+    first `sample_size` examples."""
     cfg = config.interpolate()
     seed = int(cfg.get("training", {}).get("seed", 0) or 0)
     torch.manual_seed(seed)
     nlp = build_nlp(config, device=device)
     (train_corpus,) = resolve_dot_names(cfg, [cfg["training"]["train_corpus"]])
     sample: List[Example] = []
+    tag_labels: set = set()
+    dep_labels: set = set()
+    ent_labels: set = set()
+    discover = [
+        (n, p) for n, p in nlp.pipeline
+        if not getattr(p, "labels", None) and hasattr(p, "label2id")
+    ]
     for eg in train_corpus(nlp):
-        sample.append(eg)
-        if len(sample) >= sample_size:
-            break
+        if len(sample) < sample_size:
+            sample.append(eg)
+        if not discover:
+            if len(sample) >= sample_size:
+                break
+            continue
+        ref = eg.reference
+        if ref.tags:
+            tag_labels.update(ref.tags)
+        if ref.deps:
+            dep_labels.update(d for d in ref.deps if d != "ROOT")
+        if ref.ents:
+            for tag in ref.ents:
+                if tag not in ("O", "-", None, ""):
+                    ent_labels.add(tag.partition("-")[2])
+    from .pipes import NerPipe, ParserPipe, TaggerPipe
+
+    for name, pipe in discover:
+        if isinstance(pipe, TaggerPipe) and tag_labels:
+            pipe.labels = sorted(tag_labels)
+        elif isinstance(pipe, ParserPipe) and dep_labels:
+            pipe.labels = sorted(dep_labels)
+        elif isinstance(pipe, NerPipe) and ent_labels:
+            pipe.labels = sorted(ent_labels)
+        pipe.label2id = {t: i for i, t in enumerate(pipe.labels)}
     for name, pipe in nlp.pipeline:
         pipe.initialize(sample, nlp.device)
     return nlp

@@ -569,7 +569,20 @@ class ParserPipe(_TransitionPipeBase):
         labs = []
         for eg in examples:
             deps = eg.reference.deps or ["dep"] * len(eg.reference)
-            labs.extend(self.label2id.get(d, 0) for d in deps)
+            for d in deps:
+                lid = self.label2id.get(d)
+                if lid is None:
+                    if d == "ROOT":
+                        lid = 0  # root label never drives a labeled arc cost
+                    else:
+                        raise ValueError(
+                            f"parser gold dep label {d!r} is not in the "
+                            f"component's label set (size "
+                            f"{len(self.labels)}) — labels are discovered "
+                            f"over the full training corpus at init (or "
+                            f"pinned via the component's `labels` config)"
+                        )
+                labs.append(lid)
         states.set_gold(heads, np.asarray(labs, dtype=np.int32))
 
     def _annotate(self, docs, states) -> None:

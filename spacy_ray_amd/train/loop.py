@@ -110,9 +110,12 @@ def train_while_improving(
             break
         if patience and results:
             # tuple max: ties prefer the LATER step (spaCy semantics),
-            # so a plateau doesn't count against patience from its start
+            # so a plateau doesn't count against patience from its start.
+            # patience is in STEPS, exactly as in spaCy's train_while_improving
+            # (a patience=1600 / eval_frequency=200 config stops after 1600
+            # stagnant steps, i.e. 8 evals).
             best_step = max(results)[1]
-            if (step - best_step) >= patience * eval_frequency and len(results) >= patience:
+            if (step - best_step) >= patience:
                 break
     nlp._frozen = []
     nlp._annotating = []

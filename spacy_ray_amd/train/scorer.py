@@ -52,6 +52,11 @@ def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Di
             if gh is None or ph is None:
                 continue
             for i in range(len(eg.reference)):
+                # spaCy's Scorer excludes punctuation from dependency
+                # metrics (gold dep 'punct'/'p'), so dep_uas/dep_las are
+                # comparable to spaCy baselines
+                if gd and gd[i] and gd[i].lower() in ("punct", "p"):
+                    continue
                 total += 1
                 if int(gh[i]) == int(ph[i]):
                     uas += 1

@@ -142,17 +142,28 @@ def simple_tokenize(vocab: Vocab, text: str) -> Doc:
 
 def biluo_to_codes(ents: Optional[Sequence[str]], label2id: Dict[str, int]) -> np.ndarray:
     """Per-token BILUO strings -> int codes: 0=O; for type t: 1+4t=B, 2+4t=I,
-    3+4t=L, 4+4t=U (layout shared with ops/csrc/transitions.cpp BiluoBatch)."""
+    3+4t=L, 4+4t=U; -1 = MISSING (spaCy's '-' unannotated marker: the token
+    is excluded from supervision, not negative evidence — ADVICE r1).
+    Layout shared with ops/csrc/transitions.cpp BiluoBatch."""
     if ents is None:
         return np.zeros(0, dtype=np.int32)
     kinds = {"B": 0, "I": 1, "L": 2, "U": 3}
     out = np.zeros(len(ents), dtype=np.int32)
     for i, tag in enumerate(ents):
-        if tag in (None, "O", "-", ""):
+        if tag in (None, "-"):
+            out[i] = -1  # missing: no supervision for this token
+        elif tag in ("O", ""):
             out[i] = 0
         else:
             kind, _, label = tag.partition("-")
-            t = label2id[label]
+            t = label2id.get(label)
+            if t is None:
+                raise ValueError(
+                    f"NER gold label {label!r} is not in the component's label "
+                    f"set {sorted(label2id)} — labels are discovered over the "
+                    f"full training corpus at init (or pinned via the "
+                    f"component's `labels` config); rebuild the label set"
+                )
             out[i] = 1 + 4 * t + kinds[kind]
     return out
 

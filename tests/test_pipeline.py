@@ -443,3 +443,136 @@ def test_annotating_components_transition_pipe():
         assert len(eg.predicted.heads) == len(eg.predicted)
         assert eg.predicted.ents is not None
     total.backward()
+
+
+def test_patience_is_step_based():
+    """spaCy's patience is in STEPS: with eval_frequency=2 and
+    patience=4, a never-improving run stops ~4 steps after the first eval."""
+    from spacy_ray_amd.vocab.doc import Doc, Example, Vocab
+
+    cfg = Config.from_str(TAGGER_CFG)
+    nlp = init_nlp(cfg)
+    icfg = cfg.interpolate()
+    T = resolve(icfg["training"], validate=False)
+    train_corpus = resolve_dot_names(icfg, [T["train_corpus"]])[0]
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    batches = create_train_batches(nlp, train_corpus, T["batcher"], 0)
+    calls = {"n": 0}
+
+    def evaluate():  # strictly decreasing: never improves after the first
+        calls["n"] += 1
+        return 1.0 - 0.01 * calls["n"], {}
+
+    steps = [
+        info["step"]
+        for _, info, _ in train_while_improving(
+            nlp, stepper, batches, evaluate=evaluate, dropout=0.0,
+            max_steps=100, eval_frequency=2, patience=4,
+        )
+    ]
+    # first eval at step 2 (best); stop once step - best_step >= 4 -> step 6
+    assert steps[-1] <= 8, steps
+
+
+def test_label_discovery_sees_late_labels():
+    """An entity type first appearing after the 128-example init sample must
+    not crash mid-training: labels are discovered over the FULL
+    corpus."""
+    from spacy_ray_amd.config.registry import registry
+    from spacy_ray_amd.vocab.doc import Doc, Example
+
+    @registry.readers("test.LateLabelCorpus.v1")
+    def late_label_corpus():
+        def corpus(nlp):
+            for i in range(200):
+                ents = ["U-LATE" if i >= 150 else "O", "O"]
+                yield Example.from_doc(
+                    Doc(nlp.vocab, [f"w{i % 7}", "x"], tags=["A", "B"],
+                        ents=ents))
+        return corpus
+
+    cfg_text = """
+[nlp]
+lang = "en"
+pipeline = ["tok2vec", "ner"]
+
+[components]
+
+[components.tok2vec]
+factory = "tok2vec"
+
+[components.tok2vec.model]
+@architectures = "spacy.HashEmbedCNN.v2"
+width = 32
+depth = 1
+embed_size = 100
+
+[components.ner]
+factory = "ner"
+
+[components.ner.model]
+@architectures = "spacy.TransitionBasedParser.v2"
+state_type = "ner"
+
+[components.ner.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 32
+
+[corpora]
+
+[corpora.train]
+@readers = "test.LateLabelCorpus.v1"
+
+[training]
+seed = 0
+train_corpus = "corpora.train"
+dev_corpus = "corpora.train"
+"""
+    nlp = init_nlp(Config.from_str(cfg_text))
+    ner = nlp.get_pipe("ner")
+    assert "LATE" in ner.labels
+    # and training on a late batch works (would KeyError before the fix)
+    icfg = Config.from_str(cfg_text).interpolate()
+    train_corpus = resolve_dot_names(icfg, ["corpora.train"])[0]
+    egs = list(train_corpus(nlp))[150:160]
+    nlp.forward_loss(egs)
+
+
+def test_explicit_labels_in_component_config():
+    cfg_text = TAGGER_CFG.replace(
+        '[components.tagger]\nfactory = "tagger"',
+        '[components.tagger]\nfactory = "tagger"\nlabels = ["TAG0","TAG1","TAG2","TAG3","TAG4"]',
+    )
+    nlp = init_nlp(Config.from_str(cfg_text))
+    assert nlp.get_pipe("tagger").labels == ["TAG0", "TAG1", "TAG2", "TAG3", "TAG4"]
+
+
+def test_select_pipes_disables_inference():
+    from spacy_ray_amd.vocab.doc import Doc
+
+    cfg = Config.from_str(TAGGER_CFG)
+    nlp = init_nlp(cfg)
+    doc = Doc(nlp.vocab, ["a", "b"])
+    with nlp.select_pipes(disable=["tagger"]):
+        nlp.predict_docs([doc])
+        assert doc.tags is None  # disabled pipe did NOT annotate
+    nlp.predict_docs([doc])
+    assert doc.tags is not None  # re-enabled
+
+
+def test_missing_biluo_excluded_from_loss():
+    """'-' gold tokens produce all-zero gold rows (no supervision) instead of
+    negative O supervision."""
+    from spacy_ray_amd import _srx_cpu
+    from spacy_ray_amd.vocab.doc import biluo_to_codes
+    import numpy as np
+
+    codes = biluo_to_codes(["U-PER", "-", "O"], {"PER": 0})
+    assert codes.tolist() == [4, -1, 0]
+    batch = _srx_cpu.BiluoBatch(np.array([3], dtype=np.int32), 1, 0)
+    batch.set_gold(codes.astype(np.int32))
+    # step to token 1 (the missing one): advance with UNIT-PER then inspect
+    act, feats, valid, gold = batch.step_arrays(True)
+    batch.advance(np.array([4], dtype=np.int32))  # U-PER on token 0
+    act, feats, valid, gold = batch.step_arrays(True)
+    assert gold[0].sum() == 0  # all-zero gold row -> masked out of the CE

@@ -60,3 +60,15 @@ def test_weighted_score_skips_missing_and_none():
     s = weighted_score({"tag_acc": 0.5, "speed": 100.0, "x": None},
                        {"tag_acc": 2.0, "missing": 1.0, "x": 1.0})
     assert s == pytest.approx(1.0)
+
+
+def test_uas_las_excludes_punctuation():
+    # token 2's gold dep is punct: excluded from UAS/LAS entirely
+    eg = _example(
+        ["a", "b", "."],
+        dict(heads=[1, -1, 1], deps=["d1", "ROOT", "punct"]),
+        dict(heads=[1, -1, 0], deps=["d1", "ROOT", "punct"]),
+    )
+    s = score_examples([eg], ["parser"])
+    assert s["dep_uas"] == pytest.approx(1.0)
+    assert s["dep_las"] == pytest.approx(1.0)
