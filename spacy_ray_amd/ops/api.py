@@ -307,6 +307,58 @@ def parser_step_score_accum(precomputed_detached, feats, bias, dPre32, entries=N
     return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32, entries)
 
 
+class _TransitionLoopLoss(torch.autograd.Function):
+    """Batched loss + phase-1 backward for the C++-owned transition loop
+    (srx_steploop.hip): the loop returns arenas over ALL transition steps;
+    forward runs ONE fused CE kernel (softmax-over-valid vs uniform-min-cost
+    target, spaCy's parser loss contract), backward is 3 large GEMMs + one
+    maxout scatter + one dPre scatter — replacing ~121 per-step autograd
+    nodes (each with 2 tiny GEMMs + reductions) per pipe per batch
+    (VERDICT r1 items 1 & 4).
+
+    Inputs needing grad: pre [T+1,nF,HP] (the grad-connected precompute),
+    lower_b [HP], upperW [A,H], upperB [A].  The arena tensors are data.
+    Returns the SUMMED loss (caller divides by n_states, so the upstream
+    grad carries the normalization)."""
+
+    @staticmethod
+    def forward(ctx, pre, lower_b, upperW, upperB, scores, gold, valid,
+                feats, which, hidden):
+        hip = hip_ext()
+        loss_count, dScores = hip.transition_ce(scores, gold, valid)
+        ctx.save_for_backward(dScores, feats, which, hidden, upperW)
+        ctx.pre_shape = tuple(pre.shape)
+        ctx.pre_dtype = pre.dtype
+        return loss_count[0]
+
+    @staticmethod
+    def backward(ctx, g):
+        dScores, feats, which, hidden, upperW = ctx.saved_tensors
+        hip = hip_ext()
+        T1, nF, HP = ctx.pre_shape
+        dS = dScores * g.to(dScores.dtype)
+        dUpperW = dS.t().mm(hidden)
+        dUpperB = dS.sum(0, dtype=torch.float32).to(dS.dtype)
+        dHidden = dS.mm(upperW)
+        dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
+        dLowerB = dSummed.sum(0, dtype=torch.float32).to(dS.dtype)
+        dPre32 = torch.zeros(T1, nF, HP, dtype=torch.float32, device=dS.device)
+        # token-position destinations are near-uniform -> direct atomics;
+        # the Zipf-hot pad row (missing features) is folded in as ONE
+        # mask^T @ dSummed GEMM instead
+        hip.dpre_scatter(dSummed, feats, dPre32, T1 - 1)
+        pad_mask = (feats == (T1 - 1)).to(dSummed.dtype)
+        dPre32[T1 - 1] += pad_mask.t().mm(dSummed).float()
+        return (dPre32.to(ctx.pre_dtype), dLowerB, dUpperW, dUpperB,
+                None, None, None, None, None, None)
+
+
+def transition_loop_loss(pre, lower_b, upperW, upperB, scores, gold, valid,
+                         feats, which, hidden):
+    return _TransitionLoopLoss.apply(pre, lower_b, upperW, upperB, scores,
+                                     gold, valid, feats, which, hidden)
+
+
 class _InjectGrad(torch.autograd.Function):
     """Hand a precomputed gradient to a tensor through autograd: forward is
     a zero scalar (the value is never used — losses are logged separately);

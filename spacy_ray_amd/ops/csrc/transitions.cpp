@@ -17,6 +17,8 @@
 #include <cstdint>
 #include <vector>
 
+#include "step_iface.h"
+
 #ifdef _OPENMP
 #include <omp.h>
 #endif
@@ -83,7 +85,7 @@ struct ParserState {
   }
 };
 
-struct ArcEagerBatch {
+struct ArcEagerBatch : public srx::StepBatchIface {
   int32_t n_labels;
   std::vector<ParserState> states;
   std::vector<int32_t> offsets;               // doc start offset in flat arrays
@@ -289,32 +291,31 @@ struct ArcEagerBatch {
     return py::make_tuple(act, feats, valid_a, gold_a);
   }
 
-  // Packed variant: ONE buffer = [feats int64 Sa*13][valid u8 Sa*A]
-  // [gold u8 Sa*A (train only)] so the python side does a single H2D copy
-  // per transition step (each pageable upload blocks the host ~0.1 ms).
-  py::tuple step_arrays_packed(bool with_gold, int64_t pad_row = -1) const {
-    std::vector<int32_t> idx;
-    idx.reserve(states.size());
+  // ---- StepBatchIface (consumed by the _srx_hip C++ step-loop driver)
+  int64_t n_states() const override { return (int64_t)states.size(); }
+  int n_feats() const override { return 13; }
+  int n_acts() const override { return (int)n_actions(); }
+  int64_t max_transitions() const override {
+    // buf advances exactly len times (SHIFT/RIGHT-ARC); each push is popped
+    // at most once (REDUCE/LEFT-ARC) => <= 2*len transitions per doc.
+    int64_t total = 0;
+    for (auto& st : states) total += 2 * (int64_t)st.len;
+    return total;
+  }
+
+  int64_t pack_step(bool with_gold, int64_t pad_row, int32_t* act_idx,
+                    int64_t* feats, uint8_t* valid_a, uint8_t* gold_a) override {
+    int64_t Sa = 0;
     for (size_t i = 0; i < states.size(); i++)
-      if (!states[i].final_state()) idx.push_back((int32_t)i);
-    const py::ssize_t Sa = (py::ssize_t)idx.size();
-    const py::ssize_t A = n_actions();
-    const size_t fbytes = (size_t)Sa * 13 * 8;
-    const size_t vbytes = (size_t)Sa * A;
-    py::array_t<int32_t> act(Sa);
-    py::array_t<uint8_t> packed((py::ssize_t)(fbytes + vbytes + (with_gold ? vbytes : 0)));
-    std::copy(idx.begin(), idx.end(), act.mutable_data());
-    uint8_t* base = packed.mutable_data();
-    int64_t* feats = (int64_t*)base;
-    uint8_t* valid_a = base + fbytes;
-    uint8_t* gold_a = valid_a + vbytes;
+      if (!states[i].final_state()) act_idx[Sa++] = (int32_t)i;
+    const int64_t A = n_actions();
 #ifdef _OPENMP
 #pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
 #endif
-    for (py::ssize_t k = 0; k < Sa; k++) {
+    for (int64_t k = 0; k < Sa; k++) {
       float crow[256];
       int32_t f32[13];
-      size_t i = (size_t)idx[(size_t)k];
+      size_t i = (size_t)act_idx[k];
       fill_features(i, f32);
       int64_t* fo = feats + k * 13;
       for (int q = 0; q < 13; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
@@ -322,15 +323,67 @@ struct ArcEagerBatch {
       if (with_gold) {
         fill_costs(i, v, crow);
         float cmin = KInvalid;
-        for (py::ssize_t a = 0; a < A; a++)
+        for (int64_t a = 0; a < A; a++)
           if (v[a] && crow[(size_t)a] < cmin) cmin = crow[(size_t)a];
         uint8_t* g = gold_a + k * A;
-        for (py::ssize_t a = 0; a < A; a++)
+        for (int64_t a = 0; a < A; a++)
           g[a] = (v[a] && crow[(size_t)a] <= cmin + 1e-6f) ? 1 : 0;
       } else {
         fill_valid(v, states[i]);
       }
     }
+    return Sa;
+  }
+
+  inline void apply_action(ParserState& st, int32_t act) {
+    if (act == 0) {  // SHIFT
+      st.stack.push_back(st.buf);
+      st.buf += 1;
+    } else if (act == 1) {  // REDUCE
+      st.stack.pop_back();
+    } else if (act < 2 + n_labels) {  // LEFT-ARC
+      int32_t l = act - 2;
+      int32_t s0 = st.stack.back();
+      st.add_arc(st.buf, s0, l);
+      st.stack.pop_back();
+    } else {  // RIGHT-ARC
+      int32_t l = act - 2 - n_labels;
+      int32_t s0 = st.stack.back();
+      st.add_arc(s0, st.buf, l);
+      st.stack.push_back(st.buf);
+      st.buf += 1;
+    }
+    // Degenerate-state guard: buffer exhausted with stack >1 -> valid()
+    // forces REDUCE pops; remaining stack entries keep head -1 (root),
+    // like spaCy's unattached tokens defaulting to root.
+  }
+
+  void advance_active(const int32_t* act_idx, const int32_t* actions,
+                      int64_t n) override {
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (n > 2048)
+#endif
+    for (int64_t k = 0; k < n; k++) {
+      if (actions[k] < 0) continue;
+      apply_action(states[(size_t)act_idx[k]], actions[k]);
+    }
+  }
+
+  // Packed variant: ONE buffer = [feats int64 Sa*13][valid u8 Sa*A]
+  // [gold u8 Sa*A (train only)] so the python side does a single H2D copy
+  // per transition step (each pageable upload blocks the host ~0.1 ms).
+  py::tuple step_arrays_packed(bool with_gold, int64_t pad_row = -1) {
+    int64_t Sa0 = 0;
+    for (auto& st : states)
+      if (!st.final_state()) Sa0++;
+    const py::ssize_t A = n_actions();
+    const size_t fbytes = (size_t)Sa0 * 13 * 8;
+    const size_t vbytes = (size_t)Sa0 * A;
+    py::array_t<int32_t> act(Sa0);
+    py::array_t<uint8_t> packed((py::ssize_t)(fbytes + vbytes + (with_gold ? vbytes : 0)));
+    uint8_t* base = packed.mutable_data();
+    pack_step(with_gold, pad_row, act.mutable_data(), (int64_t*)base,
+              base + fbytes, base + fbytes + vbytes);
     return py::make_tuple(act, packed, (py::ssize_t)13);
   }
 
@@ -344,31 +397,11 @@ struct ArcEagerBatch {
       if (st.final_state()) continue;
       int32_t act = a(i);
       if (act < 0) continue;  // explicit no-op (already-final slot)
-      if (act == 0) {  // SHIFT
-        st.stack.push_back(st.buf);
-        st.buf += 1;
-      } else if (act == 1) {  // REDUCE
-        st.stack.pop_back();
-      } else if (act < 2 + n_labels) {  // LEFT-ARC
-        int32_t l = act - 2;
-        int32_t s0 = st.stack.back();
-        st.add_arc(st.buf, s0, l);
-        st.stack.pop_back();
-      } else {  // RIGHT-ARC
-        int32_t l = act - 2 - n_labels;
-        int32_t s0 = st.stack.back();
-        st.add_arc(s0, st.buf, l);
-        st.stack.push_back(st.buf);
-        st.buf += 1;
-      }
-      // Degenerate-state guard: buffer exhausted with stack >1 -> force pops.
-      if (st.buf >= st.len) {
-        // states are final when stack <=1; remaining stack entries keep
-        // head -1 (attached to nothing = root), like spaCy's unattached
-        // tokens defaulting to root.
-      }
+      apply_action(st, act);
     }
   }
+
+  int64_t handle() { return (int64_t)(intptr_t)static_cast<srx::StepBatchIface*>(this); }
 
   py::array_t<int32_t> heads() const {
     int32_t total = 0;
@@ -407,7 +440,7 @@ struct NerState {
   bool final_state() const { return i >= len; }
 };
 
-struct BiluoBatch {
+struct BiluoBatch : public srx::StepBatchIface {
   int32_t n_types;
   std::vector<NerState> states;
   std::vector<int32_t> offsets;
@@ -552,27 +585,27 @@ struct BiluoBatch {
     return out;
   }
 
-  py::tuple step_arrays_packed(bool with_gold, int64_t pad_row = -1) const {
-    std::vector<int32_t> idx;
-    idx.reserve(states.size());
+  // ---- StepBatchIface
+  int64_t n_states() const override { return (int64_t)states.size(); }
+  int n_feats() const override { return 6; }
+  int n_acts() const override { return (int)n_actions(); }
+  int64_t max_transitions() const override {
+    int64_t total = 0;
+    for (auto& st : states) total += (int64_t)st.len;  // one action per token
+    return total;
+  }
+
+  int64_t pack_step(bool with_gold, int64_t pad_row, int32_t* act_idx,
+                    int64_t* feats, uint8_t* valid_a, uint8_t* gold_a) override {
+    int64_t Sa = 0;
     for (size_t i = 0; i < states.size(); i++)
-      if (!states[i].final_state()) idx.push_back((int32_t)i);
-    const py::ssize_t Sa = (py::ssize_t)idx.size();
-    const py::ssize_t A = n_actions();
-    const size_t fbytes = (size_t)Sa * 6 * 8;
-    const size_t vbytes = (size_t)Sa * A;
-    py::array_t<int32_t> act(Sa);
-    py::array_t<uint8_t> packed((py::ssize_t)(fbytes + vbytes + (with_gold ? vbytes : 0)));
-    std::copy(idx.begin(), idx.end(), act.mutable_data());
-    uint8_t* base = packed.mutable_data();
-    int64_t* feats = (int64_t*)base;
-    uint8_t* valid_a = base + fbytes;
-    uint8_t* gold_a = valid_a + vbytes;
+      if (!states[i].final_state()) act_idx[Sa++] = (int32_t)i;
+    const int64_t A = n_actions();
 #ifdef _OPENMP
 #pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
 #endif
-    for (py::ssize_t k = 0; k < Sa; k++) {
-      size_t i = (size_t)idx[(size_t)k];
+    for (int64_t k = 0; k < Sa; k++) {
+      size_t i = (size_t)act_idx[k];
       const NerState& st = states[i];
       int32_t f32[6];
       fill_features(i, f32);
@@ -587,11 +620,51 @@ struct BiluoBatch {
           std::fill(g, g + A, 0);  // missing: row excluded from the loss
         } else {
           bool gold_valid = gcode >= 0 && gcode < (int32_t)A && v[gcode];
-          for (py::ssize_t a = 0; a < A; a++)
+          for (int64_t a = 0; a < A; a++)
             g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
         }
       }
     }
+    return Sa;
+  }
+
+  inline void apply_action(NerState& st, int32_t act) {
+    st.tags[st.i] = act;
+    if (act == 0) {
+      st.open = -1; st.open_start = -1;
+    } else {
+      int32_t t = (act - 1) / 4;
+      int32_t kind = (act - 1) % 4;  // 0=B,1=I,2=L,3=U
+      if (kind == 0) { st.open = t; st.open_start = st.i; }
+      else if (kind == 1) { /* stays open */ }
+      else { st.open = -1; st.open_start = -1; }
+    }
+    st.i += 1;
+  }
+
+  void advance_active(const int32_t* act_idx, const int32_t* actions,
+                      int64_t n) override {
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (n > 2048)
+#endif
+    for (int64_t k = 0; k < n; k++) {
+      if (actions[k] < 0) continue;
+      apply_action(states[(size_t)act_idx[k]], actions[k]);
+    }
+  }
+
+  py::tuple step_arrays_packed(bool with_gold, int64_t pad_row = -1) {
+    int64_t Sa0 = 0;
+    for (auto& st : states)
+      if (!st.final_state()) Sa0++;
+    const py::ssize_t A = n_actions();
+    const size_t fbytes = (size_t)Sa0 * 6 * 8;
+    const size_t vbytes = (size_t)Sa0 * A;
+    py::array_t<int32_t> act(Sa0);
+    py::array_t<uint8_t> packed((py::ssize_t)(fbytes + vbytes + (with_gold ? vbytes : 0)));
+    uint8_t* base = packed.mutable_data();
+    pack_step(with_gold, pad_row, act.mutable_data(), (int64_t*)base,
+              base + fbytes, base + fbytes + vbytes);
     return py::make_tuple(act, packed, (py::ssize_t)6);
   }
 
@@ -605,19 +678,11 @@ struct BiluoBatch {
       if (st.final_state()) continue;
       int32_t act = a(s);
       if (act < 0) continue;
-      st.tags[st.i] = act;
-      if (act == 0) {
-        st.open = -1; st.open_start = -1;
-      } else {
-        int32_t t = (act - 1) / 4;
-        int32_t kind = (act - 1) % 4;  // 0=B,1=I,2=L,3=U
-        if (kind == 0) { st.open = t; st.open_start = st.i; }
-        else if (kind == 1) { /* stays open */ }
-        else { st.open = -1; st.open_start = -1; }
-      }
-      st.i += 1;
+      apply_action(st, act);
     }
   }
+
+  int64_t handle() { return (int64_t)(intptr_t)static_cast<srx::StepBatchIface*>(this); }
 
   py::array_t<int32_t> tags() const {
     int32_t total = 0;
@@ -647,6 +712,7 @@ void init_transitions(py::module_& m) {
       .def("advance", &ArcEagerBatch::advance)
       .def("step_arrays", &ArcEagerBatch::step_arrays, py::arg("with_gold"), py::arg("pad_row") = -1)
       .def("step_arrays_packed", &ArcEagerBatch::step_arrays_packed, py::arg("with_gold"), py::arg("pad_row") = -1)
+      .def("handle", &ArcEagerBatch::handle)
       .def("heads", &ArcEagerBatch::heads)
       .def("labels", &ArcEagerBatch::labels);
 
@@ -663,5 +729,6 @@ void init_transitions(py::module_& m) {
       .def("advance", &BiluoBatch::advance)
       .def("step_arrays", &BiluoBatch::step_arrays, py::arg("with_gold"), py::arg("pad_row") = -1)
       .def("step_arrays_packed", &BiluoBatch::step_arrays_packed, py::arg("with_gold"), py::arg("pad_row") = -1)
+      .def("handle", &BiluoBatch::handle)
       .def("tags", &BiluoBatch::tags);
 }

@@ -368,6 +368,53 @@ void adam_step(at::Tensor grad, at::Tensor master, at::Tensor m, at::Tensor v,
   });
 }
 
+// ------------------------------------------- transition-pipe batched ops
+// Fused CE over ALL transition steps (the C++ loop's arenas): returns
+// (loss_count fp32 [2], dScores) — dScores = softmax_over_valid - target,
+// zero for invalid columns / unsupervised rows (see transition_ce_kernel).
+std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
+                                      at::Tensor valid) {
+  check_dev(scores);
+  long N = scores.size(0);
+  int A = (int)scores.size(1);
+  auto dScores = at::empty_like(scores);
+  auto loss = at::zeros({2}, scores.options().dtype(at::kFloat));
+  if (N == 0) return {loss, dScores};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(scores.scalar_type(), {
+    hipLaunchKernelGGL((transition_ce_kernel<scalar_t>),
+                       dim3(grid_for(N * SRX_WAVE)), dim3(kBlock), 0, stream,
+                       (const scalar_t*)scores.data_ptr(),
+                       gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
+                       (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
+                       N, A);
+  });
+  return {loss, dScores};
+}
+
+// Batched dPre scatter over all steps: direct fp32 atomics (token-position
+// destinations are near-uniform), pad row skipped (caller folds it in as
+// one mask GEMM — see _TransitionLoopLoss in pipes.py).
+void dpre_scatter(at::Tensor dSummed, at::Tensor feats, at::Tensor dPre32,
+                  int64_t pad_row) {
+  check_dev(dSummed);
+  TORCH_CHECK(dPre32.scalar_type() == at::kFloat);
+  TORCH_CHECK(feats.scalar_type() == at::kLong);
+  long S = feats.size(0);
+  int nF = (int)feats.size(1);
+  int HP = (int)dPre32.size(-1);
+  TORCH_CHECK(HP <= 256, "dpre_scatter HP <= 256");
+  if (S == 0) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(dSummed.scalar_type(), {
+    hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t>),
+                       dim3(grid_for(S * SRX_WAVE)), dim3(kBlock), 0, stream,
+                       (const scalar_t*)dSummed.data_ptr(),
+                       feats.data_ptr<int64_t>(), dPre32.data_ptr<float>(), S,
+                       nF, HP, pad_row);
+  });
+}
+
 // ------------------------------------------------------- softmax + CE
 // returns (loss_and_count fp32 [2], dScores) — dScores = softmax - onehot
 // (unnormalized; the python wrapper divides by the valid count).
@@ -604,6 +651,12 @@ std::vector<std::vector<at::Tensor>> fused_entries_take(int64_t task_id) {
 
 }  // namespace
 
+// srx_steploop.hip — the C++-owned transition loop (one python call/batch)
+std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
+    std::vector<std::tuple<int64_t, at::Tensor, at::Tensor, at::Tensor,
+                           at::Tensor, bool>>
+        tasks);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seq2col_fwd", &seq2col_fwd);
   m.def("seq2col_bwd", &seq2col_bwd);
@@ -626,4 +679,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mwe_layer_fwd", &mwe_layer_fwd);
   m.def("fused_step", &fusedstep::fused_step);
   m.def("fused_entries_take", &fusedstep::fused_entries_take);
+  m.def("transition_ce", &transition_ce);
+  m.def("dpre_scatter", &dpre_scatter);
+  m.def("run_transition_loop", &srx_run_transition_loop,
+        py::call_guard<py::gil_scoped_release>());
 }

@@ -49,6 +49,73 @@ __global__ void softmax_ce_kernel(const T* __restrict__ scores,
   }
 }
 
+// ------------------------------------------- transition-pipe fused CE
+// The parser/NER loss over ALL transition steps at once (batched by the C++
+// step loop): per row, softmax over the VALID actions, target = uniform
+// over the min-cost (gold) actions; dScores = p - target (0 for invalid
+// columns and for rows with no gold = excluded/missing supervision).
+// loss_out[0] += -sum(target * log p); loss_out[1] += #supervised rows.
+// Replaces the torch composition (cat + float + masked_fill + log_softmax +
+// mul/sum) that was ~5 full fp32 passes over [SS, A].
+template <typename T>
+__global__ void transition_ce_kernel(const T* __restrict__ scores,
+                                     const uint8_t* __restrict__ gold,
+                                     const uint8_t* __restrict__ valid,
+                                     T* __restrict__ dScores,
+                                     float* __restrict__ loss_out,
+                                     long N, int A) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  float loss_acc = 0.f;
+  float count_acc = 0.f;
+  for (long n = wave; n < N; n += nwaves) {
+    const T* row = scores + n * (long)A;
+    const uint8_t* grow = gold + n * (long)A;
+    const uint8_t* vrow = valid + n * (long)A;
+    T* drow = dScores + n * (long)A;
+    float cnt = 0.f;
+    float m = -1e38f;
+    for (int a = lane; a < A; a += SRX_WAVE) {
+      if (grow[a]) cnt += 1.f;
+      if (vrow[a]) m = fmaxf(m, Elem<T>::ld(row + a));
+    }
+    cnt = wave_reduce_sum(cnt);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, SRX_WAVE));
+    if (cnt == 0.f) {  // no supervision for this row
+      for (int a = lane; a < A; a += SRX_WAVE) Elem<T>::st(drow + a, 0.f);
+      continue;
+    }
+    float z = 0.f;
+    for (int a = lane; a < A; a += SRX_WAVE)
+      if (vrow[a]) z += __expf(Elem<T>::ld(row + a) - m);
+    z = wave_reduce_sum(z);
+    float logz = __logf(z) + m;
+    float tgt = 1.f / cnt;
+    float l = 0.f;
+    for (int a = lane; a < A; a += SRX_WAVE) {
+      if (!vrow[a]) {
+        Elem<T>::st(drow + a, 0.f);
+        continue;
+      }
+      float x = Elem<T>::ld(row + a) - logz;  // log p
+      float t = grow[a] ? tgt : 0.f;
+      Elem<T>::st(drow + a, __expf(x) - t);
+      if (grow[a]) l -= tgt * x;
+    }
+    l = wave_reduce_sum(l);
+    if (lane == 0) {
+      loss_acc += l;
+      count_acc += 1.f;
+    }
+  }
+  if (lane == 0 && (loss_acc != 0.f || count_acc != 0.f)) {
+    atomicAdd(loss_out + 0, loss_acc);
+    atomicAdd(loss_out + 1, count_acc);
+  }
+}
+
 // ------------------------------------------------- segmented reductions
 // X [T, W] with doc offsets [N+1] -> out [N, W].  One wave per doc
 // (grid-stride); lanes stride the W columns; rows of a doc walked serially

@@ -220,7 +220,7 @@ class _TransitionTask:
     __slots__ = ("pipe", "shards", "t2v", "train", "T", "pre", "pre_d",
                  "dPre32", "hip", "score_chunks", "gold_chunks",
                  "valid_chunks", "n_states_total", "entries", "fused",
-                 "task_id")
+                 "task_id", "cpp", "cpp_outs")
 
     def __init__(self, pipe, shards, t2v, train: bool) -> None:
         from spacy_ray_amd.ops import api as _ops
@@ -233,11 +233,20 @@ class _TransitionTask:
         self.T = t2v.shape[0]
         self.pre = pipe.module.precompute(t2v)
         self.pre_d = self.pre.detach()
+        self.hip = _ops.hip_ext() if device.type == "cuda" else None
+        # C++-owned loop (srx_steploop.hip): ONE python call runs the whole
+        # per-batch transition loop; arenas come back for the batched loss.
+        # SRX_CPP_LOOP=0 falls back to the python round-robin (parity tests).
+        self.cpp = (
+            self.hip is not None
+            and hasattr(self.hip, "run_transition_loop")
+            and os.environ.get("SRX_CPP_LOOP", "1") == "1"
+        )
+        self.cpp_outs: List = []
         self.dPre32 = (
             torch.zeros(self.pre.shape, dtype=torch.float32, device=device)
-            if train else None
+            if (train and not self.cpp) else None
         )
-        self.hip = _ops.hip_ext() if device.type == "cuda" else None
         # GPU train: per-step backwards append (feats, dSummed) here; the
         # scatter happens once, batched+sorted, in finish_task
         self.entries: Optional[List] = [] if (train and self.hip is not None) else None
@@ -345,9 +354,29 @@ class _TransitionTask:
 def run_transition_tasks(tasks: List[_TransitionTask]) -> None:
     """Interleaved driver over every (task, shard) unit: complete the
     pending step (sync + C++ advance) then launch the next, round-robin —
-    one unit's CPU work hides under another unit's GPU work."""
+    one unit's CPU work hides under another unit's GPU work.
+
+    GPU default: the whole loop runs in C++ (hip.run_transition_loop, ONE
+    python crossing per batch); the python round-robin below is the CPU /
+    SRX_CPP_LOOP=0 fallback."""
     units = [(t, s) for t in tasks for s in t.shards]
     if not units:
+        return
+    if all(t.cpp for t in tasks):
+        from spacy_ray_amd.utils import timing
+
+        args = []
+        for t, s in units:
+            mod = t.pipe.module
+            args.append((
+                s.handle(), t.pre_d.contiguous(), mod.lower_b.detach(),
+                mod.upper.weight.detach(), mod.upper.bias.detach(), t.train,
+            ))
+        with timing.span("raw/cpp_loop"):
+            outs = tasks[0].hip.run_transition_loop(args)
+        for (t, s), o in zip(units, outs):
+            if t.train:
+                t.cpp_outs.append(o)
         return
     pend = [None] * len(units)
     done = [False] * len(units)
@@ -460,6 +489,26 @@ class _TransitionPipeBase(TrainablePipe):
         t2v = task.t2v
         if not task.train:
             return None, 0.0
+        if task.cpp:
+            outs = [o for o in task.cpp_outs if o and o[0].shape[0] > 0]
+            if not outs:
+                return t2v.new_zeros(()), 0.0
+            if len(outs) == 1:
+                scores, gold, valid, feats, which, hidden = outs[0]
+            else:
+                scores, gold, valid, feats, which, hidden = (
+                    torch.cat(c, dim=0) for c in zip(*outs)
+                )
+            SS = scores.shape[0]
+            from spacy_ray_amd.ops.api import transition_loop_loss
+
+            mod = self.module
+            with timing.span("raw/loss_build"):
+                loss = transition_loop_loss(
+                    task.pre, mod.lower_b, mod.upper.weight, mod.upper.bias,
+                    scores, gold, valid, feats, which, hidden,
+                ) / SS
+            return loss, float(loss.detach())
         if not task.score_chunks:
             return t2v.new_zeros(()), 0.0
         with timing.span("raw/loss_build"):
