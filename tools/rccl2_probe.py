@@ -38,4 +38,11 @@ def main() -> None:
 
 
 if __name__ == "__main__":
-    main()
+    try:
+        main()
+    except Exception as e:  # surface the real error through torchrun
+        import traceback
+
+        print("RANK %s FAILED: %r" % (os.environ.get("RANK"), e), flush=True)
+        traceback.print_exc()
+        sys.exit(1)
