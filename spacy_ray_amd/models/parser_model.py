@@ -18,6 +18,34 @@ from spacy_ray_amd.ops import api as ops
 from .layers import glorot_uniform_
 
 
+class _PrecomputePad(torch.autograd.Function):
+    """tok2vec [T,W] -> [T+1, nF, HP] with the learned pad as row T,
+    WITHOUT torch.cat: the cat copied the whole [T, nF*HP] GEMM output
+    (3.3 GB at 1M words) just to append one row — 1.8 ms/call x 2 pipes in
+    the r2 profile.  The GEMM writes straight into rows [0, T) of the
+    preallocated buffer; backward splits dOut into the three grads."""
+
+    @staticmethod
+    def forward(ctx, X, lower_W, pad):
+        T = X.shape[0]
+        nF, HP = pad.shape
+        out = X.new_empty(T + 1, nF, HP)
+        torch.mm(X, lower_W.t(), out=out[:T].view(T, nF * HP))
+        out[T] = pad.to(out.dtype)
+        ctx.save_for_backward(X, lower_W)
+        return out
+
+    @staticmethod
+    def backward(ctx, dOut):
+        X, lower_W = ctx.saved_tensors
+        T = X.shape[0]
+        d2 = dOut[:T].reshape(T, -1)
+        dX = d2.mm(lower_W)
+        dW = d2.t().mm(X)
+        dPad = dOut[T]
+        return dX, dW, dPad
+
+
 class TransitionModel(nn.Module):
     PIECES = 2  # maxout pieces in the hidden layer (spaCy default)
 
@@ -43,10 +71,7 @@ class TransitionModel(nn.Module):
 
     def precompute(self, tok2vec: torch.Tensor) -> torch.Tensor:
         """tok2vec [T, W] -> [T+1, nF, H*P]; row T is the learned pad."""
-        T = tok2vec.shape[0]
-        HP = self.hidden_width * self.PIECES
-        pre = torch.nn.functional.linear(tok2vec, self.lower_W).view(T, self.nF, HP)
-        return torch.cat([pre, self.pad.unsqueeze(0).to(pre.dtype)], dim=0)
+        return _PrecomputePad.apply(tok2vec, self.lower_W, self.pad)
 
     def score(self, precomputed: torch.Tensor, feats: torch.Tensor) -> torch.Tensor:
         """feats [S, nF] int (missing already remapped to row T) -> [S, nA]."""

@@ -325,32 +325,30 @@ class _TransitionLoopLoss(torch.autograd.Function):
     def forward(ctx, pre, lower_b, upperW, upperB, scores, gold, valid,
                 feats, which, hidden):
         hip = hip_ext()
-        loss_count, dScores = hip.transition_ce(scores, gold, valid)
-        ctx.save_for_backward(dScores, feats, which, hidden, upperW)
+        loss_count, dScores, colsum = hip.transition_ce(scores, gold, valid)
+        ctx.save_for_backward(dScores, colsum, feats, which, hidden, upperW)
         ctx.pre_shape = tuple(pre.shape)
         ctx.pre_dtype = pre.dtype
         return loss_count[0]
 
     @staticmethod
     def backward(ctx, g):
-        dScores, feats, which, hidden, upperW = ctx.saved_tensors
+        dScores, colsum, feats, which, hidden, upperW = ctx.saved_tensors
         hip = hip_ext()
         T1, nF, HP = ctx.pre_shape
         dS = dScores * g.to(dScores.dtype)
         dUpperW = dS.t().mm(hidden)
-        dUpperB = dS.sum(0, dtype=torch.float32).to(dS.dtype)
+        # colsum came fused out of transition_ce (unscaled)
+        dUpperB = (colsum * g.float()).to(dS.dtype)
         dHidden = dS.mm(upperW)
         dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
-        dLowerB = dSummed.sum(0, dtype=torch.float32).to(dS.dtype)
         dPre32 = torch.zeros(T1, nF, HP, dtype=torch.float32, device=dS.device)
         # token-position destinations are near-uniform -> direct atomics;
-        # the Zipf-hot pad row (missing features) is folded in as ONE
-        # mask^T @ dSummed GEMM instead
-        hip.dpre_scatter(dSummed, feats, dPre32, T1 - 1)
-        pad_mask = (feats == (T1 - 1)).to(dSummed.dtype)
-        dPre32[T1 - 1] += pad_mask.t().mm(dSummed).float()
-        return (dPre32.to(ctx.pre_dtype), dLowerB, dUpperW, dUpperB,
-                None, None, None, None, None, None)
+        # the Zipf-hot pad row and the bias column-sum are register-
+        # accumulated inside the same kernel
+        dBias32 = hip.dpre_scatter(dSummed, feats, dPre32, T1 - 1)
+        return (dPre32.to(ctx.pre_dtype), dBias32.to(dS.dtype), dUpperW,
+                dUpperB, None, None, None, None, None, None)
 
 
 def transition_loop_loss(pre, lower_b, upperW, upperB, scores, gold, valid,

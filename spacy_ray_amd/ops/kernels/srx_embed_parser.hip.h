@@ -197,31 +197,58 @@ __global__ void action_select_kernel(const T* __restrict__ scores,
 // scatter in the r1 profile — VERDICT r1 item 3).
 // One wave per state: the dSummed row is loaded once into registers, then
 // scattered to the nF destinations.
+// dBias32 [HP] fp32 (zero-initialized) additionally receives
+// sum_s dSummed[s, :] — the lower-bias gradient, register-accumulated per
+// wave (one atomic per column per wave) so the backward skips a full
+// [SS, HP] column-reduce pass.  Also accumulates the PAD-row sums into
+// pad32 [nF, HP] the same way (register per (f, c), one atomic per wave) —
+// the pad row is Zipf-hot, so it stays out of the per-state atomics.
 template <typename T>
 __global__ void dpre_scatter_kernel(const T* __restrict__ dSummed,
                                     const int64_t* __restrict__ feats,
                                     float* __restrict__ dPre32,
+                                    float* __restrict__ dBias32,
                                     long S, int nF, int HP, long pad_row) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
   const int ncols = (HP + SRX_WAVE - 1) / SRX_WAVE;
+  float bias_acc[4];  // HP <= 256
+  float pad_acc[16 * 4];  // nF <= 16
+  for (int c = 0; c < ncols; c++) bias_acc[c] = 0.f;
+  for (int k = 0; k < nF * ncols; k++) pad_acc[k] = 0.f;
+  float* pad_dst = dPre32 + (pad_row * (long)nF) * HP;
   for (long s = wave; s < S; s += nwaves) {
-    float v[4];  // HP <= 256
+    float v[4];
     const T* src = dSummed + s * (long)HP;
     for (int c = 0; c < ncols; c++) {
       int w = lane + c * SRX_WAVE;
       v[c] = w < HP ? Elem<T>::ld(src + w) : 0.f;
+      bias_acc[c] += v[c];
     }
     const int64_t* fs = feats + s * nF;
     for (int f = 0; f < nF; f++) {
       int64_t t = fs[f];
-      if (t == pad_row) continue;  // hot pad row handled by the mask GEMM
+      if (t == pad_row) {
+        for (int c = 0; c < ncols; c++) pad_acc[f * ncols + c] += v[c];
+        continue;
+      }
       float* dst = dPre32 + (t * (long)nF + f) * HP;
       for (int c = 0; c < ncols; c++) {
         int w = lane + c * SRX_WAVE;
         if (w < HP) atomicAdd(dst + w, v[c]);
       }
+    }
+  }
+  for (int c = 0; c < ncols; c++) {
+    int w = lane + c * SRX_WAVE;
+    if (w < HP && bias_acc[c] != 0.f) atomicAdd(dBias32 + w, bias_acc[c]);
+  }
+  for (int f = 0; f < nF; f++) {
+    for (int c = 0; c < ncols; c++) {
+      int w = lane + c * SRX_WAVE;
+      float pv = pad_acc[f * ncols + c];
+      if (w < HP && pv != 0.f) atomicAdd(pad_dst + (long)f * HP + w, pv);
     }
   }
 }

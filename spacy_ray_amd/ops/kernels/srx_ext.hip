@@ -379,7 +379,8 @@ std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
   int A = (int)scores.size(1);
   auto dScores = at::empty_like(scores);
   auto loss = at::zeros({2}, scores.options().dtype(at::kFloat));
-  if (N == 0) return {loss, dScores};
+  auto colsum = at::zeros({(long)A}, scores.options().dtype(at::kFloat));
+  if (N == 0) return {loss, dScores, colsum};
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(scores.scalar_type(), {
     hipLaunchKernelGGL((transition_ce_kernel<scalar_t>),
@@ -387,16 +388,18 @@ std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
                        (const scalar_t*)scores.data_ptr(),
                        gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
                        (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
-                       N, A);
+                       colsum.data_ptr<float>(), N, A);
   });
-  return {loss, dScores};
+  return {loss, dScores, colsum};
 }
 
-// Batched dPre scatter over all steps: direct fp32 atomics (token-position
-// destinations are near-uniform), pad row skipped (caller folds it in as
-// one mask GEMM — see _TransitionLoopLoss in pipes.py).
-void dpre_scatter(at::Tensor dSummed, at::Tensor feats, at::Tensor dPre32,
-                  int64_t pad_row) {
+// Batched dPre scatter over all steps: direct fp32 atomics for the
+// near-uniform token destinations; the Zipf-hot PAD row and the bias
+// column-sum are register-accumulated per wave.  Returns dBias32 [HP] =
+// column sums of dSummed.  Grid deliberately modest (2048 blocks) so each
+// wave covers many states and the per-wave finalization atomics amortize.
+at::Tensor dpre_scatter(at::Tensor dSummed, at::Tensor feats, at::Tensor dPre32,
+                        int64_t pad_row) {
   check_dev(dSummed);
   TORCH_CHECK(dPre32.scalar_type() == at::kFloat);
   TORCH_CHECK(feats.scalar_type() == at::kLong);
@@ -404,15 +407,21 @@ void dpre_scatter(at::Tensor dSummed, at::Tensor feats, at::Tensor dPre32,
   int nF = (int)feats.size(1);
   int HP = (int)dPre32.size(-1);
   TORCH_CHECK(HP <= 256, "dpre_scatter HP <= 256");
-  if (S == 0) return;
+  TORCH_CHECK(nF <= 16, "dpre_scatter nF <= 16");
+  auto dBias32 = at::zeros({(long)HP}, dSummed.options().dtype(at::kFloat));
+  if (S == 0) return dBias32;
   auto stream = at::cuda::getCurrentCUDAStream();
+  long waves = (S + 127) / 128;  // >=128 states per wave target
+  int grid = (int)std::min<long>((waves * SRX_WAVE + kBlock - 1) / kBlock, 2048);
+  grid = std::max(grid, 64);
   DISPATCH_F(dSummed.scalar_type(), {
-    hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t>),
-                       dim3(grid_for(S * SRX_WAVE)), dim3(kBlock), 0, stream,
+    hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t>), dim3(grid),
+                       dim3(kBlock), 0, stream,
                        (const scalar_t*)dSummed.data_ptr(),
-                       feats.data_ptr<int64_t>(), dPre32.data_ptr<float>(), S,
-                       nF, HP, pad_row);
+                       feats.data_ptr<int64_t>(), dPre32.data_ptr<float>(),
+                       dBias32.data_ptr<float>(), S, nF, HP, pad_row);
   });
+  return dBias32;
 }
 
 // ------------------------------------------------------- softmax + CE

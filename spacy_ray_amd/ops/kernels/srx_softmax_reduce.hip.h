@@ -57,18 +57,25 @@ __global__ void softmax_ce_kernel(const T* __restrict__ scores,
 // loss_out[0] += -sum(target * log p); loss_out[1] += #supervised rows.
 // Replaces the torch composition (cat + float + masked_fill + log_softmax +
 // mul/sum) that was ~5 full fp32 passes over [SS, A].
+// colsum_out [A] fp32 (zero-initialized by caller) additionally receives
+// sum_n dScores[n, :] — the (unscaled) upper-bias gradient, fused here so
+// the backward skips a full [SS, A] column-reduce pass.
 template <typename T>
 __global__ void transition_ce_kernel(const T* __restrict__ scores,
                                      const uint8_t* __restrict__ gold,
                                      const uint8_t* __restrict__ valid,
                                      T* __restrict__ dScores,
                                      float* __restrict__ loss_out,
+                                     float* __restrict__ colsum_out,
                                      long N, int A) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  const int ncols = (A + SRX_WAVE - 1) / SRX_WAVE;
   float loss_acc = 0.f;
   float count_acc = 0.f;
+  float col_acc[4];  // A <= 256
+  for (int c = 0; c < ncols; c++) col_acc[c] = 0.f;
   for (long n = wave; n < N; n += nwaves) {
     const T* row = scores + n * (long)A;
     const uint8_t* grow = gold + n * (long)A;
@@ -94,14 +101,16 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
     float logz = __logf(z) + m;
     float tgt = 1.f / cnt;
     float l = 0.f;
-    for (int a = lane; a < A; a += SRX_WAVE) {
+    for (int a = lane, c = 0; a < A; a += SRX_WAVE, c++) {
       if (!vrow[a]) {
         Elem<T>::st(drow + a, 0.f);
         continue;
       }
       float x = Elem<T>::ld(row + a) - logz;  // log p
       float t = grow[a] ? tgt : 0.f;
-      Elem<T>::st(drow + a, __expf(x) - t);
+      float d = __expf(x) - t;
+      Elem<T>::st(drow + a, d);
+      col_acc[c] += d;
       if (grow[a]) l -= tgt * x;
     }
     l = wave_reduce_sum(l);
@@ -113,6 +122,10 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
   if (lane == 0 && (loss_acc != 0.f || count_acc != 0.f)) {
     atomicAdd(loss_out + 0, loss_acc);
     atomicAdd(loss_out + 1, count_acc);
+  }
+  for (int c = 0; c < ncols; c++) {
+    int a = lane + c * SRX_WAVE;
+    if (a < A && col_acc[c] != 0.f) atomicAdd(colsum_out + a, col_acc[c]);
   }
 }
 

@@ -74,8 +74,20 @@ class DistComm(Comm):
         self.world = dist.get_world_size()
         self.device = device
         self._fused = backend == "nccl"
+        # SRX_COLLECTIVE_CHECK=1: record every collective's (op, numel) and
+        # assert the SEQUENCE is identical across ranks at each barrier —
+        # NCCL deadlocks are ordering bugs, and the one found in round 1
+        # (rank-dependent warmup counts) was exactly a sequence divergence.
+        self._oplog: Optional[list] = (
+            [] if os.environ.get("SRX_COLLECTIVE_CHECK") == "1" else None
+        )
+
+    def _log(self, op: str, numel: int) -> None:
+        if self._oplog is not None:
+            self._oplog.append((op, int(numel)))
 
     def reduce_scatter_flat(self, flat, out_shard, async_op=False):
+        self._log("rs", flat.numel())
         if self._fused:
             work = dist.reduce_scatter_tensor(out_shard, flat, op=dist.ReduceOp.AVG,
                                               async_op=async_op)
@@ -88,6 +100,7 @@ class DistComm(Comm):
         return None
 
     def all_gather_flat(self, flat_out, shard, async_op=False):
+        self._log("ag", flat_out.numel())
         if self._fused:
             return dist.all_gather_into_tensor(flat_out, shard, async_op=async_op)
         per = shard.numel()
@@ -96,6 +109,7 @@ class DistComm(Comm):
         return None
 
     def all_reduce_(self, t):
+        self._log("ar", t.numel())
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
         return None
 
@@ -105,7 +119,24 @@ class DistComm(Comm):
         return holder[0]
 
     def barrier(self) -> None:
+        if self._oplog is not None:
+            self._check_collective_order()
         dist.barrier()
+
+    def _check_collective_order(self) -> None:
+        """Assert every rank issued the same collective sequence since the
+        last check (hash of the oplog, compared via all_gather_object)."""
+        import hashlib
+
+        h = hashlib.sha256(repr(self._oplog).encode()).hexdigest()
+        payload = [None] * self.world
+        dist.all_gather_object(payload, (h, len(self._oplog)))
+        if any(p != payload[0] for p in payload):
+            raise RuntimeError(
+                f"collective-sequence divergence across ranks: {payload}; "
+                f"rank {self.rank} issued {self._oplog[-8:]}"
+            )
+        self._oplog.clear()
 
 
 def init_comm_from_env(device: Optional[torch.device] = None) -> Comm:

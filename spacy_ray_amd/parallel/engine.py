@@ -208,6 +208,14 @@ class ZeRO1Engine:
             # else: every loss-producing pipe is frozen — a valid (if odd)
             # configuration; the step is a no-op rather than a crash
         self._sync = False
+        if losses is not None:
+            # display losses arrive as detached 0-dim device tensors (pipes
+            # defer the .item() so the forward path never host-syncs); ONE
+            # conversion point here, after backward is queued
+            with timing.phase("bwd/loss_sync"):
+                for k, v in losses.items():
+                    if torch.is_tensor(v):
+                        losses[k] = float(v)
         self.last_compute_ms = (_time.perf_counter() - t0) * 1000
 
     last_compute_ms: float = 0.0

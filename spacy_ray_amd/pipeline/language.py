@@ -150,6 +150,9 @@ class Language:
         if sgd is not None:
             sgd.step()
             sgd.zero_grad()
+        for k, v in losses.items():  # deferred display-loss tensors
+            if torch.is_tensor(v):
+                losses[k] = float(v)
         return losses
 
     # ------------------------------------------------------------ inference

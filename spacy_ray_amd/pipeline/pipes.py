@@ -185,7 +185,9 @@ class TaggerPipe(TrainablePipe):
         gold = to_device(gold_np, scores.device)
         n = int((gold_np >= 0).sum())
         loss = _ops.softmax_ce_loss(scores, gold) / max(1, n)
-        return loss, float(loss.detach())
+        # display is a detached 0-dim tensor: converting to float here would
+        # device-sync mid-forward; the engine converts once after backward
+        return loss, loss.detach()
 
     def predict_and_set(self, docs, t2v, batch) -> None:
         with torch.no_grad():
@@ -508,7 +510,7 @@ class _TransitionPipeBase(TrainablePipe):
                     task.pre, mod.lower_b, mod.upper.weight, mod.upper.bias,
                     scores, gold, valid, feats, which, hidden,
                 ) / SS
-            return loss, float(loss.detach())
+            return loss, loss.detach()
         if not task.score_chunks:
             return t2v.new_zeros(()), 0.0
         with timing.span("raw/loss_build"):
@@ -523,7 +525,7 @@ class _TransitionPipeBase(TrainablePipe):
             target = all_gold.float() / counts_t.clamp(min=1).unsqueeze(-1)
             row_loss = -(target * logp).sum(dim=-1)
             step_loss = row_loss.masked_fill(~ok_t, 0).sum() / max(1, task.n_states_total)
-        display = float(step_loss.detach())
+        display = step_loss.detach()
         with timing.span("raw/phase1_bwd"):
             step_loss.backward()  # phase 1: upper + lower_b grads
             if task.fused:
@@ -609,17 +611,23 @@ class ParserPipe(_TransitionPipeBase):
     def _make_states(self, lengths, base: int = 0):
         return _srx_cpu.ArcEagerBatch(lengths, len(self.labels), base)
 
-    def _set_gold(self, states, examples) -> None:
-        heads = np.concatenate([
-            eg.reference.heads if eg.reference.heads is not None
-            else np.full(len(eg.reference), -1, dtype=np.int32)
-            for eg in examples
-        ]).astype(np.int32)
-        labs = []
-        for eg in examples:
-            deps = eg.reference.deps or ["dep"] * len(eg.reference)
-            for d in deps:
-                lid = self.label2id.get(d)
+    def _gold_arrays(self, eg) -> Tuple[np.ndarray, np.ndarray]:
+        """Per-example (heads, label-ids), cached on the reference Doc —
+        corpora/bench replay docs across steps, and the per-token python
+        loop over 1M words cost ~100 ms/step before caching (same trick as
+        the tagger's _gold_ids)."""
+        ref = eg.reference
+        key = ("dep_gold", tuple(self.labels))
+        cached = ref.user_data.get(key)
+        if cached is None:
+            n = len(ref)
+            heads = (ref.heads.astype(np.int32) if ref.heads is not None
+                     else np.full(n, -1, dtype=np.int32))
+            deps = ref.deps or ["dep"] * n
+            labs = np.zeros(n, dtype=np.int32)
+            l2i = self.label2id
+            for i, d in enumerate(deps):
+                lid = l2i.get(d)
                 if lid is None:
                     if d == "ROOT":
                         lid = 0  # root label never drives a labeled arc cost
@@ -631,8 +639,16 @@ class ParserPipe(_TransitionPipeBase):
                             f"over the full training corpus at init (or "
                             f"pinned via the component's `labels` config)"
                         )
-                labs.append(lid)
-        states.set_gold(heads, np.asarray(labs, dtype=np.int32))
+                labs[i] = lid
+            cached = (heads, labs)
+            ref.user_data[key] = cached
+        return cached
+
+    def _set_gold(self, states, examples) -> None:
+        pairs = [self._gold_arrays(eg) for eg in examples]
+        heads = np.concatenate([p[0] for p in pairs])
+        labs = np.concatenate([p[1] for p in pairs])
+        states.set_gold(heads, labs)
 
     def _annotate(self, docs, states) -> None:
         heads = states.heads()
@@ -673,12 +689,20 @@ class NerPipe(_TransitionPipeBase):
     def _make_states(self, lengths, base: int = 0):
         return _srx_cpu.BiluoBatch(lengths, len(self.labels), base)
 
+    def _gold_codes(self, eg) -> np.ndarray:
+        # cached per reference Doc (see ParserPipe._gold_arrays)
+        ref = eg.reference
+        key = ("biluo_gold", tuple(self.labels))
+        cached = ref.user_data.get(key)
+        if cached is None:
+            cached = biluo_to_codes(
+                ref.ents or ["O"] * len(ref), self.label2id
+            ).astype(np.int32)
+            ref.user_data[key] = cached
+        return cached
+
     def _set_gold(self, states, examples) -> None:
-        codes = np.concatenate([
-            biluo_to_codes(eg.reference.ents or ["O"] * len(eg.reference), self.label2id)
-            for eg in examples
-        ]).astype(np.int32)
-        states.set_gold(codes)
+        states.set_gold(np.concatenate([self._gold_codes(eg) for eg in examples]))
 
     def _annotate(self, docs, states) -> None:
         tags = states.tags()

@@ -24,6 +24,10 @@ class SimpleStepper:
                    sync: bool = True) -> None:
         total, _ = self.nlp.forward_loss(examples, losses=losses, drop=drop)
         total.backward()
+        if losses is not None:  # deferred display-loss tensors -> floats
+            for k, v in losses.items():
+                if torch.is_tensor(v):
+                    losses[k] = float(v)
 
     def apply_step(self) -> None:
         self.opt.step()
