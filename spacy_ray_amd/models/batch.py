@@ -16,7 +16,8 @@ from spacy_ray_amd.vocab.doc import Doc
 
 
 class TokenBatch:
-    __slots__ = ("attr_ids", "lengths", "n_tokens", "n_real_tokens", "docs")
+    __slots__ = ("attr_ids", "lengths", "n_tokens", "n_real_tokens", "docs",
+                 "staged")
 
     def __init__(self, docs: Sequence[Doc], device: torch.device,
                  pad_to: int = 0):
@@ -54,6 +55,11 @@ class TokenBatch:
         # bit-cast uint64 -> int64 (torch has no uint64); kernels re-interpret
         self.attr_ids = to_device(attr.view(np.int64), device)
         self.lengths = to_device(lens, device)
+        # per-pipe staged device data (e.g. tagger gold ids), uploaded at
+        # step start while the GPU queue is empty — a pageable H2D later in
+        # the step blocks the host behind every queued kernel.  Replayed
+        # batches (bench) keep the cache across steps.
+        self.staged = {}
 
     def __len__(self) -> int:
         return len(self.docs)

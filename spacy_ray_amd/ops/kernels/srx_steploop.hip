@@ -22,6 +22,7 @@
 // .so boundary).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDACachingAllocator.h>
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
@@ -207,7 +208,22 @@ struct Unit {
   bool done = false, pending = false;
   long Sa = 0;
   int steps = 0;
+  // each unit runs on its OWN stream from torch's pool: on a shared stream
+  // unit B's tiny D2H queues behind unit A's kernels, adding A's latency to
+  // B's event — exactly the serialization the interleave is meant to hide.
+  c10::cuda::CUDAStream stream = c10::cuda::getDefaultCUDAStream();
+  hipStream_t hs = nullptr;
 };
+
+// Spin-wait: hipEventSynchronize parks the thread in the kernel driver and
+// the wakeup costs ~50-200 us — per transition step, on the critical chain.
+inline void spin_wait(hipEvent_t ev) {
+  while (hipEventQuery(ev) == hipErrorNotReady) {
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#endif
+  }
+}
 
 using TaskArg = std::tuple<int64_t, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, bool>;
@@ -219,8 +235,11 @@ using TaskArg = std::tuple<int64_t, at::Tensor, at::Tensor, at::Tensor,
 // arena slices over all steps (empty tensors for decode units).
 std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
     std::vector<TaskArg> tasks) {
-  auto stream = at::cuda::getCurrentCUDAStream();
-  hipStream_t hs = stream.stream();
+  auto cur = at::cuda::getCurrentCUDAStream();
+  // all unit streams wait for the current stream (pre/weights produced there)
+  hipEvent_t start_ev;
+  hipEventCreateWithFlags(&start_ev, hipEventDisableTiming);
+  hipEventRecord(start_ev, cur.stream());
   std::vector<Unit> units(tasks.size());
 
   for (size_t i = 0; i < tasks.size(); i++) {
@@ -265,7 +284,19 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
     u.valid_h = (uint8_t*)p;              p += (size_t)u.nst * u.A;
     u.gold_h = (uint8_t*)p;
     hipEventCreateWithFlags(&u.ev, hipEventDisableTiming);
+    u.stream = c10::cuda::getStreamFromPool(false, u.pre.get_device());
+    u.hs = u.stream.stream();
+    hipStreamWaitEvent(u.hs, start_ev, 0);
+    // caching-allocator safety: these tensors were allocated on the current
+    // stream but are touched on the unit stream
+    for (const at::Tensor* t : {&u.feats_a, &u.valid_a, &u.gold_a, &u.hidden_a,
+                                &u.which_a, &u.scores_a, &u.actions_d, &u.pre,
+                                &u.lowerB, &u.upperW, &u.upperB}) {
+      c10::cuda::CUDACachingAllocator::recordStream(t->storage().data_ptr(),
+                                                    u.stream);
+    }
   }
+  hipEventDestroy(start_ev);
 
   // interleaved driver: while one unit's GPU work + D2H drains, the others'
   // CPU phases (advance + pack) run — same pipelining as the round-1 python
@@ -279,7 +310,7 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
     for (Unit& u : units) {
       if (u.done) continue;
       if (u.pending) {
-        hipEventSynchronize(u.ev);
+        spin_wait(u.ev);
         u.b->advance_active(u.act_idx_h, u.actions_h, u.Sa);
         u.pending = false;
       }
@@ -302,29 +333,29 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
       char* scores_d = (char*)u.scores_a.data_ptr() + (size_t)off * u.A * es;
       int32_t* actions_d = (int32_t*)u.actions_d.data_ptr();
       hipMemcpyAsync(feats_d, u.feats_h, (size_t)Sa * u.nF * 8,
-                     hipMemcpyHostToDevice, hs);
+                     hipMemcpyHostToDevice, u.hs);
       hipMemcpyAsync(valid_d, u.valid_h, (size_t)Sa * u.A,
-                     hipMemcpyHostToDevice, hs);
+                     hipMemcpyHostToDevice, u.hs);
       if (u.train)
         hipMemcpyAsync(gold_d, u.gold_h, (size_t)Sa * u.A,
-                       hipMemcpyHostToDevice, hs);
+                       hipMemcpyHostToDevice, u.hs);
       const uint8_t* sel = u.train ? (const uint8_t*)gold_d : (const uint8_t*)valid_d;
       if (u.pre.scalar_type() == at::kBFloat16) {
         launch_fused_step<bf16_t>(u.pre.data_ptr(), (const int64_t*)feats_d,
                                   u.lowerB.data_ptr(), u.upperW.data_ptr(),
                                   u.upperB.data_ptr(), sel,
                                   (const uint8_t*)valid_d, hidden_d, which_d,
-                                  scores_d, actions_d, Sa, u.nF, u.H, u.A, hs);
+                                  scores_d, actions_d, Sa, u.nF, u.H, u.A, u.hs);
       } else {
         launch_fused_step<float>(u.pre.data_ptr(), (const int64_t*)feats_d,
                                  u.lowerB.data_ptr(), u.upperW.data_ptr(),
                                  u.upperB.data_ptr(), sel,
                                  (const uint8_t*)valid_d, hidden_d, which_d,
-                                 scores_d, actions_d, Sa, u.nF, u.H, u.A, hs);
+                                 scores_d, actions_d, Sa, u.nF, u.H, u.A, u.hs);
       }
       hipMemcpyAsync(u.actions_h, actions_d, (size_t)Sa * 4,
-                     hipMemcpyDeviceToHost, hs);
-      hipEventRecord(u.ev, hs);
+                     hipMemcpyDeviceToHost, u.hs);
+      hipEventRecord(u.ev, u.hs);
       u.pending = true;
       u.Sa = Sa;
       u.used += Sa;

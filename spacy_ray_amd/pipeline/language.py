@@ -83,6 +83,11 @@ class Language:
                 [eg.predicted for eg in examples], self.device)
         from .pipes import _TransitionPipeBase, run_transition_tasks
 
+        # gold staging FIRST (empty GPU queue): pageable H2D after kernels
+        # are queued blocks the host behind all of them
+        for name, pipe in self.pipeline:
+            if name not in self._frozen and hasattr(pipe, "stage_gold"):
+                pipe.stage_gold(examples, batch)
         t2v_pipe = self.tok2vec
         with timing.phase("fwd/tok2vec"):
             t2v = t2v_pipe.forward(batch, drop=drop) if t2v_pipe is not None else None
