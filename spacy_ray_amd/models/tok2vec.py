@@ -44,10 +44,15 @@ class MultiHashEmbed(nn.Module):
             for i, table in enumerate(self.tables):
                 outs.append(ops.hashembed(table, batch.attr_ids[:, i], self.seeds[i]))
             X = torch.cat(outs, dim=1)
-        with timing.phase("t2v/mixer"):
-            Y = self.mixer(X)
-        if drop and self.training:
-            Y = torch.nn.functional.dropout(Y, drop)
+        with timing.phase("t2v/mixer_gemm"):
+            Y = torch.nn.functional.linear(X, self.mixer.weight, self.mixer.bias)
+        with timing.phase("t2v/mixer_maxout"):
+            Y = ops.maxout(Y.view(*Y.shape[:-1], self.mixer.pieces, self.mixer.nO))
+        with timing.phase("t2v/mixer_ln"):
+            Y = self.mixer.norm(Y)
+        with timing.phase("t2v/mixer_drop"):
+            if drop and self.training:
+                Y = torch.nn.functional.dropout(Y, drop)
         return Y
 
 

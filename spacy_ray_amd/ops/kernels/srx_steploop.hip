@@ -225,6 +225,20 @@ inline void spin_wait(hipEvent_t ev) {
   }
 }
 
+// SRX_LOOP_STATS=1: per-call phase accounting (stderr) — wait = event spin,
+// pack = state-machine feature/cost extraction, gpu = copies+launch issue,
+// adv = state advance.
+struct LoopStats {
+  double pack_ms = 0, wait_ms = 0, gpu_ms = 0, adv_ms = 0;
+  long iters = 0;
+};
+
+inline double now_ms() {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return ts.tv_sec * 1e3 + ts.tv_nsec * 1e-6;
+}
+
 using TaskArg = std::tuple<int64_t, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, bool>;
 
@@ -304,23 +318,32 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
   const long max_iters = 1L << 30;
   long guard = 0;
   bool all_done = false;
+  static const bool loop_stats = getenv("SRX_LOOP_STATS") != nullptr;
+  LoopStats st;
+  double t0 = 0;
   while (!all_done) {
     TORCH_CHECK(guard++ < max_iters, "transition loop failed to terminate");
     all_done = true;
     for (Unit& u : units) {
       if (u.done) continue;
       if (u.pending) {
+        if (loop_stats) t0 = now_ms();
         spin_wait(u.ev);
+        if (loop_stats) { st.wait_ms += now_ms() - t0; t0 = now_ms(); }
         u.b->advance_active(u.act_idx_h, u.actions_h, u.Sa);
+        if (loop_stats) st.adv_ms += now_ms() - t0;
         u.pending = false;
       }
+      if (loop_stats) t0 = now_ms();
       long Sa = u.b->pack_step(u.train, u.T, u.act_idx_h, u.feats_h, u.valid_h,
                                u.gold_h);
+      if (loop_stats) st.pack_ms += now_ms() - t0;
       if (Sa == 0) {
         u.done = true;
         continue;
       }
       all_done = false;
+      if (loop_stats) { st.iters++; t0 = now_ms(); }
       long off = u.train ? u.used : 0;
       TORCH_CHECK(off + Sa <= u.cap, "transition arena overflow (", off, "+",
                   Sa, " > ", u.cap, ")");
@@ -356,12 +379,17 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
       hipMemcpyAsync(u.actions_h, actions_d, (size_t)Sa * 4,
                      hipMemcpyDeviceToHost, u.hs);
       hipEventRecord(u.ev, u.hs);
+      if (loop_stats) st.gpu_ms += now_ms() - t0;
       u.pending = true;
       u.Sa = Sa;
       u.used += Sa;
       u.steps += 1;
     }
   }
+  if (loop_stats)
+    fprintf(stderr,
+            "[srx loop] iters=%ld pack=%.2fms wait=%.2fms issue=%.2fms adv=%.2fms\n",
+            st.iters, st.pack_ms, st.wait_ms, st.gpu_ms, st.adv_ms);
 
   std::vector<std::vector<at::Tensor>> out;
   out.reserve(units.size());
