@@ -136,6 +136,10 @@ def main() -> None:
         n_warm = int(t_dev.item())
     for i in range(n_warm):
         step(i)
+    if os.environ.get("SRX_TIMING") == "1":
+        from spacy_ray_amd.utils import timing
+
+        timing.reset_times()  # drop warmup/first-call one-time costs
 
     comm.barrier()
     if use_cuda:

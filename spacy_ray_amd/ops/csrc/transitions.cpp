@@ -70,6 +70,12 @@ struct ArcEagerBatch : public srx::StepBatchIface {
   std::vector<int32_t> head, label, l1, l2, r1, r2;  // flat [total], doc-local ids
   std::vector<int32_t> gold_head, gold_label;        // flat [total]
   bool has_gold = false;
+  // O(1)-oracle bookkeeping (fill_costs was 28 ms/step at 1M words with the
+  // O(stack + buffer) formulation):
+  std::vector<uint8_t> on_stack;        // flat [total]
+  std::vector<int32_t> gold_kids_buf;   // [total] # gold children still in buffer
+  std::vector<int32_t> kids_off;        // CSR [total+1]: gold children lists
+  std::vector<int32_t> kids;            // CSR payload (doc-local child ids)
 
   ArcEagerBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
                 int32_t n_labels_, int32_t base_offset_ = 0)
@@ -95,6 +101,7 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     l2.assign(total, -1);
     r1.assign(total, -1);
     r2.assign(total, -1);
+    on_stack.assign(total, 0);
   }
 
   void set_gold(py::array_t<int32_t, py::array::c_style | py::array::forcecast> heads,
@@ -110,6 +117,26 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     std::memcpy(gold_head.data(), H.data(0) + base_offset, total * sizeof(int32_t));
     std::memcpy(gold_label.data(), Lb.data(0) + base_offset, total * sizeof(int32_t));
     has_gold = true;
+    // gold-children CSR + in-buffer counters (tokens start in the buffer)
+    gold_kids_buf.assign(total, 0);
+    kids_off.assign(total + 1, 0);
+    for (int64_t d = 0; d < n_docs; d++) {
+      const int64_t o = off[d];
+      for (int32_t i = 0; i < len[d]; i++) {
+        int32_t h = gold_head[o + i];
+        if (h >= 0 && h < len[d]) gold_kids_buf[o + h]++;
+      }
+    }
+    for (int64_t t = 0; t < total; t++) kids_off[t + 1] = kids_off[t] + gold_kids_buf[t];
+    kids.resize(kids_off[total]);
+    std::vector<int32_t> cursor(kids_off.begin(), kids_off.end() - 1);
+    for (int64_t d = 0; d < n_docs; d++) {
+      const int64_t o = off[d];
+      for (int32_t i = 0; i < len[d]; i++) {
+        int32_t h = gold_head[o + i];
+        if (h >= 0 && h < len[d]) kids[cursor[o + h]++] = i;
+      }
+    }
   }
 
   int32_t n_actions() const { return 2 + 2 * n_labels; }
@@ -205,40 +232,39 @@ struct ArcEagerBatch : public srx::StepBatchIface {
 
   // ---- shared cost computation for one state (Goldberg&Nivre dynamic
   // oracle); writes n_actions() floats, invalid actions get KInvalid.
+  // O(1) + O(deg(b)) per state via the on_stack bitmap, the per-head
+  // gold-children-in-buffer counters and the gold-children CSR (replaces
+  // the O(stack + buffer) scans — 28 ms/step of pack time at 1M words).
   void fill_costs(int64_t d, uint8_t* v, float* r) const {
     const int64_t o = off[d];
     const int32_t* gh = gold_head.data() + o;
     const int32_t* gl = gold_label.data() + o;
-    const int32_t* stk = stack.data() + o;
-    const int32_t ss = ssize[d];
     fill_valid(v, d);
     int32_t b = buf[d] < len[d] ? buf[d] : -1;
     int32_t v0 = s0(d);
-    auto in_stack = [&](int32_t t) {
-      for (int32_t k = 0; k < ss; k++)
-        if (stk[k] == t) return true;
-      return false;
-    };
     float c_shift = 0, c_reduce = 0, c_la = 0, c_ra = 0;
     if (b >= 0) {
-      if (gh[b] >= 0 && in_stack(gh[b])) c_shift += 1;
-      for (int32_t k = 0; k < ss; k++) {
-        int32_t s = stk[k];
-        if (head[o + s] == -1 && gh[s] == b) c_shift += 1;
+      // gold head of b sits on the stack: SHIFT/RIGHT-ARC can lose the arc
+      bool ghb_on_stack =
+          gh[b] >= 0 && gh[b] < len[d] && on_stack[o + gh[b]];
+      if (ghb_on_stack) c_shift += 1;
+      // headless stack tokens whose gold head is b (gold children of b on
+      // the stack without a head yet)
+      float stack_kids_of_b = 0;
+      for (int32_t k = kids_off[o + b]; k < kids_off[o + b + 1]; k++) {
+        int32_t c = kids[k];
+        if (on_stack[o + c] && head[o + c] == -1) stack_kids_of_b += 1;
       }
-    }
-    if (v0 >= 0) {
-      for (int32_t t = buf[d]; t < len[d]; t++)
-        if (gh[t] == v0) c_reduce += 1;
-      if (b >= 0) {
+      c_shift += stack_kids_of_b;
+      if (v0 >= 0) {
+        c_reduce = (float)gold_kids_buf[o + v0];
         c_la = c_reduce;
         if (gh[v0] >= 0 && gh[v0] > b) c_la += 1;
-        if (gh[b] >= 0 && gh[b] != v0 && (in_stack(gh[b]) || gh[b] > b)) c_ra += 1;
-        for (int32_t k = 0; k < ss; k++) {
-          int32_t s = stk[k];
-          if (head[o + s] == -1 && gh[s] == b) c_ra += 1;
-        }
+        if (gh[b] >= 0 && gh[b] != v0 && (ghb_on_stack || gh[b] > b)) c_ra += 1;
+        c_ra += stack_kids_of_b;
       }
+    } else if (v0 >= 0) {
+      c_reduce = (float)gold_kids_buf[o + v0];
     }
     r[0] = v[0] ? c_shift : KInvalid;
     r[1] = v[1] ? c_reduce : KInvalid;
@@ -306,22 +332,37 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     return Sa;
   }
 
+  inline void leave_buffer(int64_t d, int32_t tok) {
+    // token `tok` moves out of the buffer: its gold head loses one
+    // in-buffer child (the c_reduce counter)
+    if (has_gold) {
+      int32_t h = gold_head[off[d] + tok];
+      if (h >= 0 && h < len[d]) gold_kids_buf[off[d] + h]--;
+    }
+  }
+
   inline void apply_action(int64_t d, int32_t act) {
     const int64_t o = off[d];
     if (act == 0) {  // SHIFT
+      on_stack[o + buf[d]] = 1;
+      leave_buffer(d, buf[d]);
       stack[o + ssize[d]++] = buf[d];
       buf[d] += 1;
     } else if (act == 1) {  // REDUCE
+      on_stack[o + stack[o + ssize[d] - 1]] = 0;
       ssize[d] -= 1;
     } else if (act < 2 + n_labels) {  // LEFT-ARC
       int32_t l = act - 2;
       int32_t v0 = stack[o + ssize[d] - 1];
       add_arc(d, buf[d], v0, l);
+      on_stack[o + v0] = 0;
       ssize[d] -= 1;
     } else {  // RIGHT-ARC
       int32_t l = act - 2 - n_labels;
       int32_t v0 = stack[o + ssize[d] - 1];
       add_arc(d, v0, buf[d], l);
+      on_stack[o + buf[d]] = 1;
+      leave_buffer(d, buf[d]);
       stack[o + ssize[d]++] = buf[d];
       buf[d] += 1;
     }

@@ -109,3 +109,87 @@ def test_ner_open_entity_constraints():
     assert v[2] == 1 and v[3] == 1
     assert v[0] == 0 and v[1] == 0 and v[4] == 0
     assert v[6] == 0 and v[7] == 0  # type-1 IN/LAST invalid
+
+
+def test_oracle_costs_match_bruteforce_reference():
+    """Property test: the incremental O(1) oracle
+    (on-stack bitmap + gold-children-in-buffer counters) must equal a
+    brute-force Goldberg-Nivre cost computation at every state of random
+    valid-action walks over random projective-ish gold trees."""
+    import numpy as np
+    from spacy_ray_amd import _srx_cpu
+
+    rng = np.random.RandomState(7)
+    for trial in range(40):
+        n = int(rng.randint(2, 12))
+        L = int(rng.randint(1, 4))
+        heads = np.full(n, -1, dtype=np.int32)
+        for i in range(1, n):
+            heads[i] = rng.randint(0, i)  # head earlier than child (projective-ish)
+        if n > 1 and rng.rand() < 0.5:
+            heads[0] = rng.randint(1, n)
+        labels = rng.randint(0, L, size=n).astype(np.int32)
+        batch = _srx_cpu.ArcEagerBatch(np.array([n], dtype=np.int32), L, 0)
+        batch.set_gold(heads, labels)
+        A = batch.n_actions
+
+        # python mirror of the state for brute-force costs
+        stack, buf, head = [], 0, [-1] * n
+        KINV = 1e9
+
+        def brute_costs(valid):
+            gh, gl = heads, labels
+            b = buf if buf < n else -1
+            s0 = stack[-1] if stack else -1
+            c_shift = c_reduce = c_la = c_ra = 0.0
+            if b >= 0:
+                if gh[b] >= 0 and gh[b] in stack:
+                    c_shift += 1
+                c_shift += sum(1 for s in stack if head[s] == -1 and gh[s] == b)
+            if s0 >= 0:
+                c_reduce = sum(1 for t in range(buf, n) if gh[t] == s0)
+                if b >= 0:
+                    c_la = c_reduce
+                    if gh[s0] >= 0 and gh[s0] > b:
+                        c_la += 1
+                    if gh[b] >= 0 and gh[b] != s0 and (gh[b] in stack or gh[b] > b):
+                        c_ra += 1
+                    c_ra += sum(1 for s in stack if head[s] == -1 and gh[s] == b)
+            out = np.full(A, KINV, dtype=np.float32)
+            if valid[0]:
+                out[0] = c_shift
+            if valid[1]:
+                out[1] = c_reduce
+            for l in range(L):
+                la, ra = c_la, c_ra
+                if b >= 0 and s0 >= 0:
+                    if gh[s0] == b and gl[s0] != l:
+                        la += 1
+                    if gh[b] == s0 and gl[b] != l:
+                        ra += 1
+                if valid[2 + l]:
+                    out[2 + l] = la
+                if valid[2 + L + l]:
+                    out[2 + L + l] = ra
+            return out
+
+        for _ in range(4 * n):
+            fin = batch.is_final()
+            if fin[0]:
+                break
+            valid = batch.valid()[0]
+            costs = batch.costs()[0]
+            ref = brute_costs(valid)
+            assert np.allclose(costs, ref), (trial, costs, ref, stack, buf, head)
+            choices = np.nonzero(valid)[0]
+            act = int(choices[rng.randint(len(choices))])
+            batch.advance(np.array([act], dtype=np.int32))
+            # python mirror advance
+            if act == 0:
+                stack.append(buf); buf += 1
+            elif act == 1:
+                stack.pop()
+            elif act < 2 + L:
+                head[stack[-1]] = buf; stack.pop()
+            else:
+                head[buf] = stack[-1]; stack.append(buf); buf += 1
