@@ -111,7 +111,8 @@ class Language:
             own = pipe.own_tok2vec(batch, drop=drop)
             pt2v = own if own is not None else t2v
             if isinstance(pipe, _TransitionPipeBase):
-                trans_tasks.append((name, pipe, pipe.make_loss_task(examples, pt2v)))
+                with timing.phase(f"loss/make_task_{name}"):
+                    trans_tasks.append((name, pipe, pipe.make_loss_task(examples, pt2v)))
                 continue
             with timing.phase(f"loss/{name}"):
                 loss, display = pipe.get_loss(examples, pt2v, batch)
