@@ -36,9 +36,9 @@ static int srx_nthreads() {
     if (env) return std::max(1, atoi(env));
     const char* ws = getenv("WORLD_SIZE");
     int world = ws ? std::max(1, atoi(ws)) : 1;
-    // pack_step is memory-bound (feature/valid/gold writes); 32 threads
-    // measured faster than 16 on the 256-core MI355X box at world=1
-    return std::min(32, std::max(1, omp_get_max_threads() / (2 * world)));
+    // measured on the 256-core MI355X box: 16 threads beat 32/64 by ~35%
+    // end-to-end (fork/join + NUMA costs dominate past 16)
+    return std::min(16, std::max(1, omp_get_max_threads() / (2 * world)));
   }();
   return n;
 #else

@@ -112,7 +112,8 @@ class Language:
             pt2v = own if own is not None else t2v
             if isinstance(pipe, _TransitionPipeBase):
                 with timing.phase(f"loss/make_task_{name}"):
-                    trans_tasks.append((name, pipe, pipe.make_loss_task(examples, pt2v)))
+                    trans_tasks.append((name, pipe,
+                                        pipe.make_loss_task(examples, pt2v, batch)))
                 continue
             with timing.phase(f"loss/{name}"):
                 loss, display = pipe.get_loss(examples, pt2v, batch)
