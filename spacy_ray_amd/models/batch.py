@@ -17,7 +17,7 @@ from spacy_ray_amd.vocab.doc import Doc
 
 class TokenBatch:
     __slots__ = ("attr_ids", "lengths", "n_tokens", "n_real_tokens", "docs",
-                 "staged")
+                 "staged", "lengths_np")
 
     def __init__(self, docs: Sequence[Doc], device: torch.device,
                  pad_to: int = 0):
@@ -30,6 +30,10 @@ class TokenBatch:
             device = torch.device(device)
         self.docs = list(docs)
         lens_list = [len(d) for d in docs]
+        # real (pre-pad) per-doc lengths; reused by the transition pipes'
+        # state construction (a fresh 50k-doc list comprehension per pipe
+        # per step showed up in profiles)
+        self.lengths_np = np.asarray(lens_list, dtype=np.int32)
         real = int(sum(lens_list))
         self.n_real_tokens = real
         if pad_to == 0:

@@ -119,24 +119,36 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     std::memcpy(gold_head.data(), H.data(0) + base_offset, total * sizeof(int32_t));
     std::memcpy(gold_label.data(), Lb.data(0) + base_offset, total * sizeof(int32_t));
     has_gold = true;
-    // gold-children CSR + in-buffer counters (tokens start in the buffer)
+    // gold-children CSR + in-buffer counters (tokens start in the buffer):
+    // per-doc-independent count/fill passes run under OpenMP; only the 1M-add
+    // prefix sum is serial.
     gold_kids_buf.assign(total, 0);
-    kids_off.assign(total + 1, 0);
+    kids_off.resize(total + 1);
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (n_docs > 2048)
+#endif
     for (int64_t d = 0; d < n_docs; d++) {
       const int64_t o = off[d];
-      for (int32_t i = 0; i < len[d]; i++) {
+      const int32_t n = len[d];
+      for (int32_t i = 0; i < n; i++) {
         int32_t h = gold_head[o + i];
-        if (h >= 0 && h < len[d]) gold_kids_buf[o + h]++;
+        if (h >= 0 && h < n) gold_kids_buf[o + h]++;
       }
     }
-    for (int64_t t = 0; t < total; t++) kids_off[t + 1] = kids_off[t] + gold_kids_buf[t];
+    kids_off[0] = 0;
+    for (int64_t t = 0; t < total; t++)
+      kids_off[t + 1] = kids_off[t] + gold_kids_buf[t];
     kids.resize(kids_off[total]);
-    std::vector<int32_t> cursor(kids_off.begin(), kids_off.end() - 1);
+#ifdef _OPENMP
+#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (n_docs > 2048)
+#endif
     for (int64_t d = 0; d < n_docs; d++) {
       const int64_t o = off[d];
-      for (int32_t i = 0; i < len[d]; i++) {
+      const int32_t n = len[d];
+      std::vector<int32_t> cur(kids_off.begin() + o, kids_off.begin() + o + n);
+      for (int32_t i = 0; i < n; i++) {
         int32_t h = gold_head[o + i];
-        if (h >= 0 && h < len[d]) kids[cursor[o + h]++] = i;
+        if (h >= 0 && h < n) kids[cur[h]++] = i;
       }
     }
   }

@@ -591,7 +591,10 @@ class _TransitionPipeBase(TrainablePipe):
         staged = batch.staged.get(key) if batch is not None else None
         if staged is None:
             staged = self._build_gold(examples)
-        lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
+        if batch is not None and len(batch.docs) == len(examples):
+            lengths = batch.lengths_np
+        else:
+            lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
         shards = []
         for lo, hi, base in self._split_docs(lengths, self._n_shards(t2v)):
             states = self._make_states(lengths[lo:hi], base)
