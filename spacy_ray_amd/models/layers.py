@@ -33,7 +33,7 @@ class Maxout(nn.Module):
         self.norm = LayerNorm(nO) if normalize else None
 
     def forward(self, X: torch.Tensor) -> torch.Tensor:
-        Y = torch.nn.functional.linear(X, self.weight, self.bias)
+        Y = ops.linear_cdw(X, self.weight, self.bias)
         Y = ops.maxout(Y.view(*Y.shape[:-1], self.pieces, self.nO))
         if self.norm is not None:
             Y = self.norm(Y)

@@ -37,11 +37,13 @@ class _PrecomputePad(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dOut):
+        from spacy_ray_amd.ops.api import mm_dw_chunked
+
         X, lower_W = ctx.saved_tensors
         T = X.shape[0]
         d2 = dOut[:T].reshape(T, -1)
         dX = d2.mm(lower_W)
-        dW = d2.t().mm(X)
+        dW = mm_dw_chunked(d2, X)
         dPad = dOut[T]
         return dX, dW, dPad
 

@@ -45,7 +45,7 @@ class MultiHashEmbed(nn.Module):
                 outs.append(ops.hashembed(table, batch.attr_ids[:, i], self.seeds[i]))
             X = torch.cat(outs, dim=1)
         with timing.phase("t2v/mixer_gemm"):
-            Y = torch.nn.functional.linear(X, self.mixer.weight, self.mixer.bias)
+            Y = ops.linear_cdw(X, self.mixer.weight, self.mixer.bias)
         with timing.phase("t2v/mixer_maxout"):
             Y = ops.maxout(Y.view(*Y.shape[:-1], self.mixer.pieces, self.mixer.nO))
         with timing.phase("t2v/mixer_ln"):

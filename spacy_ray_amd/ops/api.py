@@ -307,6 +307,52 @@ def parser_step_score_accum(precomputed_detached, feats, bias, dPre32, entries=N
     return _ParserStepScoreAccum.apply(precomputed_detached, feats, bias, dPre32, entries)
 
 
+def mm_dw_chunked(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """a [K, M], b [K, N] -> a^T @ b, for weight-gradient GEMMs with tiny
+    M x N and huge K: hipBLASLt's tile grid is (M/64)*(N/64) workgroups —
+    e.g. [1M,288]^T @ [1M,384] runs ~30 workgroups on 256 CUs (~2 ms
+    measured).  Splitting K into 32 batched GEMMs + one fp32 sum fills the
+    chip; partials are rounded to bf16 once each, summed in fp32."""
+    K, M = a.shape
+    N = b.shape[1]
+    if (not a.is_cuda) or K < (1 << 18) or (M // 64 + 1) * (N // 64 + 1) >= 512:
+        return a.t().mm(b)
+    n = 32
+    ck = K // n
+    main = torch.bmm(a[: ck * n].view(n, ck, M).transpose(1, 2),
+                     b[: ck * n].view(n, ck, N))
+    out = main.sum(0, dtype=torch.float32)
+    if ck * n < K:
+        out += a[ck * n :].t().mm(b[ck * n :]).float()
+    return out.to(a.dtype)
+
+
+class _LinearCDW(torch.autograd.Function):
+    """F.linear whose backward computes dW via mm_dw_chunked (the M*N-small,
+    K-huge weight-grad GEMM) and db with fp32 accumulation."""
+
+    @staticmethod
+    def forward(ctx, X, W, b):
+        ctx.save_for_backward(X, W)
+        ctx.has_bias = b is not None
+        return torch.nn.functional.linear(X, W, b)
+
+    @staticmethod
+    def backward(ctx, dY):
+        X, W = ctx.saved_tensors
+        dY = dY.contiguous()
+        dW = mm_dw_chunked(dY, X)
+        dX = dY.mm(W)
+        db = dY.sum(0, dtype=torch.float32).to(dY.dtype) if ctx.has_bias else None
+        return dX, dW, db
+
+
+def linear_cdw(X, W, b=None):
+    if X.is_cuda and X.dim() == 2:
+        return _LinearCDW.apply(X, W, b)
+    return torch.nn.functional.linear(X, W, b)
+
+
 class _TransitionLoopLoss(torch.autograd.Function):
     """Batched loss + phase-1 backward for the C++-owned transition loop
     (srx_steploop.hip): the loop returns arenas over ALL transition steps;
@@ -337,17 +383,18 @@ class _TransitionLoopLoss(torch.autograd.Function):
         hip = hip_ext()
         T1, nF, HP = ctx.pre_shape
         dS = dScores * g.to(dScores.dtype)
-        dUpperW = dS.t().mm(hidden)
+        dUpperW = mm_dw_chunked(dS, hidden)
         # colsum came fused out of transition_ce (unscaled)
         dUpperB = (colsum * g.float()).to(dS.dtype)
         dHidden = dS.mm(upperW)
         dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
-        dPre32 = torch.zeros(T1, nF, HP, dtype=torch.float32, device=dS.device)
-        # token-position destinations are near-uniform -> direct atomics;
-        # the Zipf-hot pad row and the bias column-sum are register-
-        # accumulated inside the same kernel
-        dBias32 = hip.dpre_scatter(dSummed, feats, dPre32, T1 - 1)
-        return (dPre32.to(ctx.pre_dtype), dBias32.to(dS.dtype), dUpperW,
+        # dPre accumulates directly in the compute dtype (bf16: packed
+        # atomics, no fp32 buffer + convert); the Zipf-hot pad row and the
+        # bias column-sum are register-accumulated into fp32 side buffers
+        dPre = torch.zeros(T1, nF, HP, dtype=ctx.pre_dtype, device=dS.device)
+        dBias32, dPad32 = hip.dpre_scatter(dSummed, feats, dPre, T1 - 1)
+        dPre[T1 - 1] = dPad32.to(dPre.dtype)
+        return (dPre, dBias32.to(dS.dtype), dUpperW,
                 dUpperB, None, None, None, None, None, None)
 
 
@@ -436,7 +483,7 @@ class _MWELayer(torch.autograd.Function):
         dbias = dPre.sum(dim=0)
         # GEMM backwards (pre = X3 @ W^T)
         X3 = hip.seq2col_fwd(X, starts, ends)
-        dW = dPre.t().mm(X3)
+        dW = mm_dw_chunked(dPre, X3)
         dX3 = dPre.mm(weight)
         dX = hip.seq2col_bwd(dX3.contiguous(), starts, ends)
         dX += dY  # residual

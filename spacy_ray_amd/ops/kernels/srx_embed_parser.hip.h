@@ -2,6 +2,8 @@
 // scorer (fused nF-row gather + sum + bias + maxout P=2).
 // Semantics: ops/torch_ref.py (hashembed_*, parser_step_score).
 #pragma once
+#include <hip/hip_bf16.h>
+
 #include "srx_common.hip.h"
 
 // ----------------------------------------------------------- hashembed
@@ -200,55 +202,87 @@ __global__ void action_select_kernel(const T* __restrict__ scores,
 // dBias32 [HP] fp32 (zero-initialized) additionally receives
 // sum_s dSummed[s, :] — the lower-bias gradient, register-accumulated per
 // wave (one atomic per column per wave) so the backward skips a full
-// [SS, HP] column-reduce pass.  Also accumulates the PAD-row sums into
-// pad32 [nF, HP] the same way (register per (f, c), one atomic per wave) —
-// the pad row is Zipf-hot, so it stays out of the per-state atomics.
-template <typename T>
+// [SS, HP] column-reduce pass.  The PAD-row sums land in dPad32 [nF, HP]
+// the same way (register per (f, c), one fp32 atomic per wave) — the pad
+// row is Zipf-hot, so it stays out of the per-state atomics AND out of the
+// lower-precision dPre accumulation; the caller writes dPad into dPre's
+// pad row afterwards.
+//
+// OutBF16=false: dPre is fp32, plain atomicAdd.
+// OutBF16=true:  dPre is bf16 — gfx950 packed global_atomic_pk_add_bf16
+//   (unsafeAtomicAdd on __hip_bfloat162), one pair per lane.  Halves the
+//   scatter traffic AND removes the separate fp32->bf16 convert of the
+//   whole [T+1, nF, HP] buffer (6.6 GB read + 3.3 GB write per pipe per
+//   step).  Accumulation rounds per add, acceptable at ~2 contributions
+//   per (token, slot) destination (pad excluded).
+template <typename T, bool OutBF16>
 __global__ void dpre_scatter_kernel(const T* __restrict__ dSummed,
                                     const int64_t* __restrict__ feats,
-                                    float* __restrict__ dPre32,
+                                    void* __restrict__ dPre,
                                     float* __restrict__ dBias32,
+                                    float* __restrict__ dPad32,
                                     long S, int nF, int HP, long pad_row) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
-  const int ncols = (HP + SRX_WAVE - 1) / SRX_WAVE;
-  float bias_acc[4];  // HP <= 256
-  float pad_acc[16 * 4];  // nF <= 16
-  for (int c = 0; c < ncols; c++) bias_acc[c] = 0.f;
-  for (int k = 0; k < nF * ncols; k++) pad_acc[k] = 0.f;
-  float* pad_dst = dPre32 + (pad_row * (long)nF) * HP;
+  // fp32: lane covers columns lane + c*64.  bf16: lane covers the PAIR
+  // (2*lane + c*128, 2*lane+1 + c*128).
+  const int ncols = OutBF16 ? (HP / 2 + SRX_WAVE - 1) / SRX_WAVE
+                            : (HP + SRX_WAVE - 1) / SRX_WAVE;
+  const int span = OutBF16 ? 2 : 1;
+  float bias_acc[8];   // HP <= 256 (bf16 pairs: 2 floats per c)
+  float pad_acc[16 * 8];  // nF <= 16
+  for (int c = 0; c < ncols * span; c++) bias_acc[c] = 0.f;
+  for (int k = 0; k < nF * ncols * span; k++) pad_acc[k] = 0.f;
   for (long s = wave; s < S; s += nwaves) {
-    float v[4];
+    float v[8];
     const T* src = dSummed + s * (long)HP;
     for (int c = 0; c < ncols; c++) {
-      int w = lane + c * SRX_WAVE;
-      v[c] = w < HP ? Elem<T>::ld(src + w) : 0.f;
-      bias_acc[c] += v[c];
+      for (int e = 0; e < span; e++) {
+        int w = span * lane + e + c * span * SRX_WAVE;
+        float x = w < HP ? Elem<T>::ld(src + w) : 0.f;
+        v[c * span + e] = x;
+        bias_acc[c * span + e] += x;
+      }
     }
     const int64_t* fs = feats + s * nF;
     for (int f = 0; f < nF; f++) {
       int64_t t = fs[f];
       if (t == pad_row) {
-        for (int c = 0; c < ncols; c++) pad_acc[f * ncols + c] += v[c];
+        for (int c = 0; c < ncols * span; c++) pad_acc[f * ncols * span + c] += v[c];
         continue;
       }
-      float* dst = dPre32 + (t * (long)nF + f) * HP;
-      for (int c = 0; c < ncols; c++) {
-        int w = lane + c * SRX_WAVE;
-        if (w < HP) atomicAdd(dst + w, v[c]);
+      if (OutBF16) {
+#if defined(__gfx950__) || defined(__gfx942__) || defined(__gfx90a__)
+        __hip_bfloat162* dst =
+            (__hip_bfloat162*)dPre + (t * (long)nF + f) * (HP / 2);
+        for (int c = 0; c < ncols; c++) {
+          int p = lane + c * SRX_WAVE;
+          if (p < HP / 2) {
+            __hip_bfloat162 val(__float2bfloat16(v[c * 2]),
+                                __float2bfloat16(v[c * 2 + 1]));
+            unsafeAtomicAdd(dst + p, val);
+          }
+        }
+#endif
+      } else {
+        float* dst = (float*)dPre + (t * (long)nF + f) * HP;
+        for (int c = 0; c < ncols; c++) {
+          int w = lane + c * SRX_WAVE;
+          if (w < HP) atomicAdd(dst + w, v[c]);
+        }
       }
     }
   }
-  for (int c = 0; c < ncols; c++) {
-    int w = lane + c * SRX_WAVE;
+  for (int c = 0; c < ncols * span; c++) {
+    int w = span * lane + (c % span) + (c / span) * span * SRX_WAVE;
     if (w < HP && bias_acc[c] != 0.f) atomicAdd(dBias32 + w, bias_acc[c]);
   }
   for (int f = 0; f < nF; f++) {
-    for (int c = 0; c < ncols; c++) {
-      int w = lane + c * SRX_WAVE;
-      float pv = pad_acc[f * ncols + c];
-      if (w < HP && pv != 0.f) atomicAdd(pad_dst + (long)f * HP + w, pv);
+    for (int c = 0; c < ncols * span; c++) {
+      int w = span * lane + (c % span) + (c / span) * span * SRX_WAVE;
+      float pv = pad_acc[f * ncols * span + c];
+      if (w < HP && pv != 0.f) atomicAdd(dPad32 + (long)f * HP + w, pv);
     }
   }
 }

@@ -187,7 +187,11 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
     // serialize on the few dg/db addresses; wide layers (trf W=768) spread
     // the atomics over 8x the columns and need more waves to fill 256 CUs
     // at large N (measured 2.0 ms/call at N=260k W=768 under the 1024 cap).
-    long cap = W >= 256 ? 4096 : 1024;
+    // (re-measured r2: the old 1024 cap for narrow layers starved the
+    // 1M-row encoder backward of parallelism — 0.73 ms/call at 10x off
+    // bandwidth; per-wave register accumulation keeps the dg/db atomic
+    // count at waves*W regardless, so more waves are safe)
+    long cap = 4096;
     int grid = (int)std::min<long>((N + 3) / 4, cap);
     DISPATCH_F(X.scalar_type(), {
       hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t>), dim3(grid),
@@ -393,35 +397,49 @@ std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
   return {loss, dScores, colsum};
 }
 
-// Batched dPre scatter over all steps: direct fp32 atomics for the
-// near-uniform token destinations; the Zipf-hot PAD row and the bias
-// column-sum are register-accumulated per wave.  Returns dBias32 [HP] =
-// column sums of dSummed.  Grid deliberately modest (2048 blocks) so each
-// wave covers many states and the per-wave finalization atomics amortize.
-at::Tensor dpre_scatter(at::Tensor dSummed, at::Tensor feats, at::Tensor dPre32,
-                        int64_t pad_row) {
+// Batched dPre scatter over all steps: direct atomics for the near-uniform
+// token destinations (packed-bf16 atomics when dPre is bf16 — halves the
+// traffic and skips the fp32->bf16 convert of the whole buffer); the
+// Zipf-hot PAD row and the bias column-sum are register-accumulated per
+// wave into SEPARATE fp32 buffers.  Returns (dBias32 [HP], dPad32 [nF,HP]);
+// the caller writes dPad into dPre's pad row.
+std::vector<at::Tensor> dpre_scatter(at::Tensor dSummed, at::Tensor feats,
+                                     at::Tensor dPre, int64_t pad_row) {
   check_dev(dSummed);
-  TORCH_CHECK(dPre32.scalar_type() == at::kFloat);
   TORCH_CHECK(feats.scalar_type() == at::kLong);
   long S = feats.size(0);
   int nF = (int)feats.size(1);
-  int HP = (int)dPre32.size(-1);
+  int HP = (int)dPre.size(-1);
   TORCH_CHECK(HP <= 256, "dpre_scatter HP <= 256");
   TORCH_CHECK(nF <= 16, "dpre_scatter nF <= 16");
+  bool out_bf16 = dPre.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(out_bf16 || dPre.scalar_type() == at::kFloat,
+              "dPre must be bf16 or fp32");
+  TORCH_CHECK(!out_bf16 || HP % 2 == 0, "bf16 dPre needs even HP");
   auto dBias32 = at::zeros({(long)HP}, dSummed.options().dtype(at::kFloat));
-  if (S == 0) return dBias32;
+  auto dPad32 = at::zeros({(long)nF, (long)HP}, dSummed.options().dtype(at::kFloat));
+  if (S == 0) return {dBias32, dPad32};
   auto stream = at::cuda::getCurrentCUDAStream();
-  long waves = (S + 127) / 128;  // >=128 states per wave target
-  int grid = (int)std::min<long>((waves * SRX_WAVE + kBlock - 1) / kBlock, 2048);
+  long waves = (S + 63) / 64;  // >=64 states per wave target
+  int grid = (int)std::min<long>((waves * SRX_WAVE + kBlock - 1) / kBlock, 4096);
   grid = std::max(grid, 64);
   DISPATCH_F(dSummed.scalar_type(), {
-    hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t>), dim3(grid),
-                       dim3(kBlock), 0, stream,
-                       (const scalar_t*)dSummed.data_ptr(),
-                       feats.data_ptr<int64_t>(), dPre32.data_ptr<float>(),
-                       dBias32.data_ptr<float>(), S, nF, HP, pad_row);
+    if (out_bf16)
+      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, true>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)dSummed.data_ptr(),
+                         feats.data_ptr<int64_t>(), dPre.data_ptr(),
+                         dBias32.data_ptr<float>(), dPad32.data_ptr<float>(),
+                         S, nF, HP, pad_row);
+    else
+      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, false>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)dSummed.data_ptr(),
+                         feats.data_ptr<int64_t>(), dPre.data_ptr(),
+                         dBias32.data_ptr<float>(), dPad32.data_ptr<float>(),
+                         S, nF, HP, pad_row);
   });
-  return dBias32;
+  return {dBias32, dPad32};
 }
 
 // ------------------------------------------------------- softmax + CE

@@ -81,11 +81,20 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
     const uint8_t* grow = gold + n * (long)A;
     const uint8_t* vrow = valid + n * (long)A;
     T* drow = dScores + n * (long)A;
+    // load the row + masks ONCE into registers (A <= 256 -> 4 per lane);
+    // the 3 logical passes (max, sumexp, dScores) reuse them
+    float x[4];
+    bool g[4], vv[4];
     float cnt = 0.f;
     float m = -1e38f;
-    for (int a = lane; a < A; a += SRX_WAVE) {
-      if (grow[a]) cnt += 1.f;
-      if (vrow[a]) m = fmaxf(m, Elem<T>::ld(row + a));
+    for (int c = 0; c < ncols; c++) {
+      int a = lane + c * SRX_WAVE;
+      bool in = a < A;
+      g[c] = in && grow[a];
+      vv[c] = in && vrow[a];
+      x[c] = vv[c] ? Elem<T>::ld(row + a) : -1e38f;
+      if (g[c]) cnt += 1.f;
+      if (vv[c]) m = fmaxf(m, x[c]);
     }
     cnt = wave_reduce_sum(cnt);
 #pragma unroll
@@ -95,23 +104,25 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
       continue;
     }
     float z = 0.f;
-    for (int a = lane; a < A; a += SRX_WAVE)
-      if (vrow[a]) z += __expf(Elem<T>::ld(row + a) - m);
+    for (int c = 0; c < ncols; c++)
+      if (vv[c]) z += __expf(x[c] - m);
     z = wave_reduce_sum(z);
     float logz = __logf(z) + m;
     float tgt = 1.f / cnt;
     float l = 0.f;
-    for (int a = lane, c = 0; a < A; a += SRX_WAVE, c++) {
-      if (!vrow[a]) {
+    for (int c = 0; c < ncols; c++) {
+      int a = lane + c * SRX_WAVE;
+      if (a >= A) continue;
+      if (!vv[c]) {
         Elem<T>::st(drow + a, 0.f);
         continue;
       }
-      float x = Elem<T>::ld(row + a) - logz;  // log p
-      float t = grow[a] ? tgt : 0.f;
-      float d = __expf(x) - t;
+      float lp = x[c] - logz;  // log p
+      float t = g[c] ? tgt : 0.f;
+      float d = __expf(lp) - t;
       Elem<T>::st(drow + a, d);
       col_acc[c] += d;
-      if (grow[a]) l -= tgt * x;
+      if (g[c]) l -= tgt * lp;
     }
     l = wave_reduce_sum(l);
     if (lane == 0) {

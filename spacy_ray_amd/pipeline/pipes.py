@@ -106,7 +106,9 @@ class TaggerHead(nn.Module):
         nn.init.zeros_(self.output.bias)
 
     def forward(self, X):
-        return self.output(X)
+        from spacy_ray_amd.ops.api import linear_cdw
+
+        return linear_cdw(X, self.output.weight, self.output.bias)
 
 
 class TaggerPipe(TrainablePipe):

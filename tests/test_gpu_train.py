@@ -207,16 +207,29 @@ def test_dpre_scatter_matches_index_add():
     dSummed = torch.randn(SS, HP, device="cuda", dtype=torch.float32)
     feats = torch.randint(0, T1, (SS, nF), device="cuda", dtype=torch.int64)
     out = torch.zeros(T1, nF, HP, device="cuda", dtype=torch.float32)
-    dBias = hip.dpre_scatter(dSummed, feats, out, pad)
-    # reference: index_add over ALL entries (the kernel folds the pad row
-    # in via register accumulation), plus the fused bias column sum
+    dBias, dPad = hip.dpre_scatter(dSummed, feats, out, pad)
+    # reference: index_add over non-pad entries; pad sums come back in dPad
     ref = torch.zeros(T1 * nF, HP, device="cuda", dtype=torch.float32)
     slot = torch.arange(nF, device="cuda")
     dest = (feats * nF + slot).reshape(-1)
     src = dSummed.repeat_interleave(nF, dim=0)
-    ref.index_add_(0, dest, src)
+    mask = (feats != pad).reshape(-1)
+    ref.index_add_(0, dest[mask], src[mask])
     assert torch.allclose(out.view(T1 * nF, HP), ref, rtol=1e-4, atol=1e-3)
     assert torch.allclose(dBias, dSummed.sum(0), rtol=1e-4, atol=1e-3)
+    ref_pad = torch.zeros(nF, HP, device="cuda")
+    for f in range(nF):
+        ref_pad[f] = dSummed[feats[:, f] == pad].sum(0)
+    assert torch.allclose(dPad, ref_pad, rtol=1e-4, atol=1e-3)
+    # bf16 output path (packed atomics): looser tolerance, same structure
+    dS16 = dSummed.to(torch.bfloat16)
+    out16 = torch.zeros(T1, nF, HP, device="cuda", dtype=torch.bfloat16)
+    dBias16, dPad16 = hip.dpre_scatter(dS16, feats, out16, pad)
+    ref16 = torch.zeros(T1 * nF, HP, device="cuda", dtype=torch.float32)
+    ref16.index_add_(0, dest[mask], dS16.float().repeat_interleave(nF, dim=0)[mask])
+    diff = (out16.float().view(T1 * nF, HP) - ref16).abs()
+    assert float(diff.mean()) < 0.05, float(diff.mean())
+    assert torch.allclose(dBias16, dS16.float().sum(0), rtol=1e-2, atol=0.5)
 
 
 @need_gpu
