@@ -66,6 +66,20 @@ PYBIND11_MODULE(_srx_cpu, m) {
   m.doc() = "spacy_ray_amd CPU native core (murmur hashing, transition systems)";
   m.def("hash_strings", &hash_strings, "hash a list of UTF-8 strings to uint64");
   m.def("hash_string", &hash_string, "hash one UTF-8 string to uint64");
+  m.def("spacy_hash_strings",
+        [](const std::vector<std::string>& strs) {
+          // spaCy StringStore hash (MurmurHash64A seed 1) — the `.spacy`
+          // DocBin string-reference values
+          py::array_t<uint64_t> out((py::ssize_t)strs.size());
+          auto r = out.mutable_unchecked<1>();
+          for (py::ssize_t i = 0; i < (py::ssize_t)strs.size(); i++)
+            r(i) = srx::murmur2_64a(strs[i].data(), (int)strs[i].size(), 1);
+          return out;
+        },
+        "spaCy-compatible string hashes (MurmurHash64A, seed 1)");
+  m.def("spacy_hash_string", [](const std::string& s) {
+    return srx::murmur2_64a(s.data(), (int)s.size(), 1);
+  });
   m.def("hash4", &hash4, py::arg("ids"), py::arg("seed"),
         "murmur3 x86_128 of 8-byte keys -> (n,4) uint32");
   m.def("hashembed_rows", &hashembed_rows, py::arg("ids"), py::arg("seed"), py::arg("nrows"),

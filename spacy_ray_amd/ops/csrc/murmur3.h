@@ -107,4 +107,43 @@ inline uint64_t hash_utf8(const char* data, int len) {
   return out[0];
 }
 
+// MurmurHash64A (MurmurHash2, 64-bit, by Austin Appleby — public-domain
+// algorithm, implemented from its description).  spaCy's StringStore hash
+// is murmurhash.mrmr.hash64(utf8, len, seed=1) = this function; the real
+// `.spacy` DocBin format references strings by these values, so the
+// DocBin reader/writer (data/docbin.py) must match them bit-for-bit
+// (reference contract: spaCy strings.pyx hash_string via
+// /root/reference/bin/get-data.sh's `spacy convert` output).
+inline uint64_t murmur2_64a(const void* key, int len, uint64_t seed) {
+  const uint64_t m = 0xc6a4a7935bd1e995ULL;
+  const int r = 47;
+  uint64_t h = seed ^ ((uint64_t)len * m);
+  const uint8_t* data = (const uint8_t*)key;
+  const uint8_t* end = data + (len & ~7);
+  while (data != end) {
+    uint64_t k;
+    std::memcpy(&k, data, 8);
+    data += 8;
+    k *= m;
+    k ^= k >> r;
+    k *= m;
+    h ^= k;
+    h *= m;
+  }
+  switch (len & 7) {
+    case 7: h ^= (uint64_t)data[6] << 48; [[fallthrough]];
+    case 6: h ^= (uint64_t)data[5] << 40; [[fallthrough]];
+    case 5: h ^= (uint64_t)data[4] << 32; [[fallthrough]];
+    case 4: h ^= (uint64_t)data[3] << 24; [[fallthrough]];
+    case 3: h ^= (uint64_t)data[2] << 16; [[fallthrough]];
+    case 2: h ^= (uint64_t)data[1] << 8;  [[fallthrough]];
+    case 1: h ^= (uint64_t)data[0];
+            h *= m;
+  }
+  h ^= h >> r;
+  h *= m;
+  h ^= h >> r;
+  return h;
+}
+
 }  // namespace srx

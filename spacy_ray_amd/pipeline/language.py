@@ -34,6 +34,12 @@ class Language:
         # select_pipes(disable=...): skipped during inference AND frozen for
         # training (spaCy's disabled-component contract — ADVICE r1)
         self._disabled: List[str] = []
+        # rule-based tokenizer (spaCy prefix/suffix/infix algorithm) for the
+        # nlp(text) / pipe / serve surface; training from DocBin corpora
+        # never tokenizes
+        from spacy_ray_amd.vocab.tokenizer import Tokenizer
+
+        self.tokenizer = Tokenizer()
 
     # ------------------------------------------------------------- pipeline
     @property
@@ -207,11 +213,9 @@ class Language:
     def pipe(self, inputs, batch_size: int = 256):
         """Annotate a stream of Docs or texts in batches (spaCy nlp.pipe
         contract)."""
-        from spacy_ray_amd.vocab.doc import simple_tokenize
-
         batch: List[Doc] = []
         for item in inputs:
-            doc = item if isinstance(item, Doc) else simple_tokenize(self.vocab, item)
+            doc = item if isinstance(item, Doc) else self.tokenizer(self.vocab, item)
             batch.append(doc)
             if len(batch) >= batch_size:
                 yield from self.predict_docs(batch)
@@ -220,9 +224,7 @@ class Language:
             yield from self.predict_docs(batch)
 
     def __call__(self, text: str) -> Doc:
-        from spacy_ray_amd.vocab.doc import simple_tokenize
-
-        doc = simple_tokenize(self.vocab, text)
+        doc = self.tokenizer(self.vocab, text)
         self.predict_docs([doc])
         return doc
 
@@ -241,6 +243,10 @@ class Language:
         vocab_dir = path / "vocab"
         vocab_dir.mkdir(exist_ok=True)
         (vocab_dir / "strings.json").write_text(json.dumps(self.vocab.strings.to_list()))
+        # tokenizer settings file (spaCy layout has a `tokenizer` blob)
+        from spacy_ray_amd.vocab.tokenizer import tokenizer_to_bytes
+
+        (path / "tokenizer").write_bytes(tokenizer_to_bytes(getattr(self, "tokenizer", None)))
         for name, pipe in self.pipeline:
             pdir = path / name
             pdir.mkdir(exist_ok=True)
@@ -250,10 +256,21 @@ class Language:
 
                 state = {k: v.detach().cpu().contiguous() for k, v in pipe.module.state_dict().items()}
                 save_file(state, str(pdir / "model.safetensors"))
+                # Thinc-msgpack `model` bytes alongside (spaCy component
+                # layout: <component>/model; see data/thinc_serde.py)
+                from spacy_ray_amd.data.thinc_serde import (
+                    component_to_thinc_nodes, model_to_thinc_bytes)
+
+                (pdir / "model").write_bytes(
+                    model_to_thinc_bytes(component_to_thinc_nodes(pipe)))
 
     def from_disk(self, path) -> "Language":
         path = Path(path)
         self.meta = json.loads((path / "meta.json").read_text())
+        if (path / "tokenizer").exists():
+            from spacy_ray_amd.vocab.tokenizer import tokenizer_from_bytes
+
+            self.tokenizer = tokenizer_from_bytes((path / "tokenizer").read_bytes())
         strings = json.loads((path / "vocab" / "strings.json").read_text())
         for s in strings:
             self.vocab.strings.add(s)
@@ -267,6 +284,19 @@ class Language:
 
                 state = load_file(str(mfile))
                 pipe.module.load_state_dict(state)
+                pipe.module.to(self.device)
+            elif (pdir / "model").exists() and pipe.module is not None:
+                # spaCy-written checkpoint: Thinc-msgpack component bytes
+                from spacy_ray_amd.data.thinc_serde import (
+                    load_thinc_nodes_into_component)
+
+                n = load_thinc_nodes_into_component(
+                    pipe, (pdir / "model").read_bytes())
+                if n == 0:
+                    raise ValueError(
+                        f"component {name!r}: no tensors from the Thinc "
+                        f"model bytes matched this architecture"
+                    )
                 pipe.module.to(self.device)
         return self
 

