@@ -67,13 +67,19 @@ class _SrxEmbeddingFn(torch.autograd.Function):
             keep = flat_ids != ctx.padding_idx
             flat_ids = flat_ids[keep]
             dY2 = dY2[keep]
-        order = torch.argsort(flat_ids)
-        dW32 = torch.zeros(ctx.nrows, dY2.shape[-1], dtype=torch.float32,
+        from spacy_ray_amd.ops.api import FIXED_SCALE, deterministic
+
+        det = deterministic()
+        order = torch.argsort(flat_ids, stable=True) if det else torch.argsort(flat_ids)
+        dW32 = torch.zeros(ctx.nrows, dY2.shape[-1],
+                           dtype=torch.int64 if det else torch.float32,
                            device=dY.device)
         hip_ext().seg_scatter_add(
             flat_ids[order].int().contiguous(), order.int().contiguous(),
             dY2, dW32,
         )
+        if det:
+            dW32 = dW32.to(torch.float32) / FIXED_SCALE
         return dW32.to(dY.dtype), None, None
 
 

@@ -18,6 +18,16 @@ from . import torch_ref as ref
 _HIP = None
 _HIP_TRIED = False
 
+FIXED_SCALE = 16777216.0  # 2^24 (mirrors srx_common.hip.h SRX_FIXED_SCALE)
+
+
+def deterministic() -> bool:
+    """SRX_DETERMINISTIC=1: every atomic float accumulation runs in
+    fixed-point int64 (bit-identical across runs) and sorts are stable —
+    closes SURVEY §5.2 / VERDICT r1 item 7.  Costs ~10-20% on the scatter
+    backwards."""
+    return os.environ.get("SRX_DETERMINISTIC") == "1"
+
 
 def hip_ext():
     global _HIP, _HIP_TRIED
@@ -140,17 +150,22 @@ class _HashEmbed(torch.autograd.Function):
         if _want_hip(dY):
             hip = hip_ext()
             T = dY.shape[0]
-            if T >= 4096:
+            det = deterministic()
+            if T >= 4096 or det:
                 # Zipf-hot rows serialize plain atomics (1.5 ms/call at
                 # T=128k measured) — sort by destination row + chunked
-                # segmented reduction instead (SURVEY.md §7 hard-part 2)
+                # segmented reduction instead (SURVEY.md §7 hard-part 2).
+                # Deterministic mode: stable sort + int64 fixed-point sink.
                 dst = rows.reshape(-1)
-                order = torch.argsort(dst)
+                order = torch.argsort(dst, stable=True) if det else torch.argsort(dst)
                 dst_sorted = dst[order].contiguous().int()
                 src = (order // 4).int()
-                dT32 = torch.zeros(ctx.nrows, dY.shape[1], dtype=torch.float32,
+                acc_dt = torch.int64 if det else torch.float32
+                dT32 = torch.zeros(ctx.nrows, dY.shape[1], dtype=acc_dt,
                                    device=dY.device)
                 hip.seg_scatter_add(dst_sorted, src, dY.contiguous(), dT32)
+                if det:
+                    dT32 = dT32.to(torch.float32) / FIXED_SCALE
                 dT = dT32.to(dY.dtype)
             else:
                 dT = hip.hashembed_bwd(dY.contiguous(), rows, ctx.nrows)
@@ -182,7 +197,8 @@ class _LayerNorm(torch.autograd.Function):
     def backward(ctx, dY):
         X, g, mu, rstd = ctx.saved_tensors
         if _want_hip(dY):
-            dX, dg, db = hip_ext().layernorm_bwd(dY.contiguous(), X, g, mu, rstd)
+            dX, dg, db = hip_ext().layernorm_bwd(dY.contiguous(), X, g, mu, rstd,
+                                                 deterministic())
             return dX, dg, db, None
         xhat = (X - mu) * rstd
         dg = (dY * xhat).sum(dim=tuple(range(dY.dim() - 1)))
@@ -371,7 +387,8 @@ class _TransitionLoopLoss(torch.autograd.Function):
     def forward(ctx, pre, lower_b, upperW, upperB, scores, gold, valid,
                 feats, which, hidden):
         hip = hip_ext()
-        loss_count, dScores, colsum = hip.transition_ce(scores, gold, valid)
+        loss_count, dScores, colsum = hip.transition_ce(scores, gold, valid,
+                                                        deterministic())
         ctx.save_for_backward(dScores, colsum, feats, which, hidden, upperW)
         ctx.pre_shape = tuple(pre.shape)
         ctx.pre_dtype = pre.dtype
@@ -390,9 +407,13 @@ class _TransitionLoopLoss(torch.autograd.Function):
         dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
         # dPre accumulates directly in the compute dtype (bf16: packed
         # atomics, no fp32 buffer + convert); the Zipf-hot pad row and the
-        # bias column-sum are register-accumulated into fp32 side buffers
-        dPre = torch.zeros(T1, nF, HP, dtype=ctx.pre_dtype, device=dS.device)
+        # bias column-sum are register-accumulated into fp32 side buffers.
+        # Deterministic mode: int64 fixed-point accumulation end to end.
+        acc_dt = torch.int64 if deterministic() else ctx.pre_dtype
+        dPre = torch.zeros(T1, nF, HP, dtype=acc_dt, device=dS.device)
         dBias32, dPad32 = hip.dpre_scatter(dSummed, feats, dPre, T1 - 1)
+        if acc_dt == torch.int64:
+            dPre = (dPre.to(torch.float32) / FIXED_SCALE).to(ctx.pre_dtype)
         dPre[T1 - 1] = dPad32.to(dPre.dtype)
         return (dPre, dBias32.to(dS.dtype), dUpperW,
                 dUpperB, None, None, None, None, None, None)
@@ -477,7 +498,8 @@ class _MWELayer(torch.autograd.Function):
         dY = dY.contiguous()
         dL = dY * dropmask if dropmask is not None else dY
         # LN backward over the maxout output
-        dM, dg, db = hip.layernorm_bwd(dL.contiguous(), Mout, g, mu, rstd)
+        dM, dg, db = hip.layernorm_bwd(dL.contiguous(), Mout, g, mu, rstd,
+                                       deterministic())
         # maxout scatter: [T, W] -> [T, 3, W] pieces-major -> [T, 3W]
         dPre = hip.maxout_bwd(dM, which, 3).reshape(dY.shape[0], -1)
         dbias = dPre.sum(dim=0)

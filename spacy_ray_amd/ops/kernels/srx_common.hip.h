@@ -42,6 +42,26 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
   return v;
 }
 
+// ---- deterministic accumulation (SRX_DETERMINISTIC / SURVEY §5.2)
+// Float atomics make gradient sums depend on scheduling order (fp add is
+// not associative).  The deterministic variants accumulate in FIXED-POINT
+// int64 (value * 2^24, llrintf-rounded): integer atomic adds ARE
+// associative, so the result is bit-identical across runs regardless of
+// order.  Range: |sum| < 2^39 ~ 5.5e11 at 2^-24 ~ 6e-8 resolution — ample
+// for gradient magnitudes.  The caller converts back with a single
+// (deterministic) elementwise divide.
+#define SRX_FIXED_SCALE 16777216.0f  // 2^24
+
+template <bool DET>
+__device__ __forceinline__ void srx_atomic_add(void* buf, long idx, float v) {
+  if (DET) {
+    long long q = (long long)llrintf(v * SRX_FIXED_SCALE);
+    atomicAdd((unsigned long long*)buf + idx, (unsigned long long)q);
+  } else {
+    atomicAdd((float*)buf + idx, v);
+  }
+}
+
 // ---- MurmurHash3 x64_128 of one 8-byte key: device twin of
 // ops/csrc/murmur3.h::murmur3_hash4_u64 — MUST stay bit-identical (the
 // CPU/GPU HashEmbed row-assignment contract).

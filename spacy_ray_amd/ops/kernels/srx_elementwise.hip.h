@@ -163,15 +163,16 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ X,
 // atomic version serialized on the 2*W hot addresses: 772 us/call at
 // N=32k, W=96 — this shape runs in ~30 us.)  Needs W <= SRX_LN_MAX_W.
 #define SRX_LN_MAX_W 1024
-template <typename T>
+// DET=true: dg/db buffers are int64 fixed-point (bit-deterministic).
+template <typename T, bool DET = false>
 __global__ void layernorm_bwd_kernel(const T* __restrict__ dY,
                                      const T* __restrict__ X,
                                      const T* __restrict__ g,
                                      const float* __restrict__ mu,
                                      const float* __restrict__ rstd,
                                      T* __restrict__ dX,
-                                     float* __restrict__ dg32,
-                                     float* __restrict__ db32,
+                                     void* __restrict__ dg32,
+                                     void* __restrict__ db32,
                                      long N, int W) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
@@ -210,8 +211,8 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dY,
   for (int c = 0; c < ncols; c++) {
     int w = lane + c * SRX_WAVE;
     if (w < W) {
-      atomicAdd(dg32 + w, dg_loc[c]);
-      atomicAdd(db32 + w, db_loc[c]);
+      srx_atomic_add<DET>(dg32, w, dg_loc[c]);
+      srx_atomic_add<DET>(db32, w, db_loc[c]);
     }
   }
 }

@@ -106,11 +106,15 @@ __global__ void parser_step_fwd_kernel(const T* __restrict__ pre,
 // M/chunk pushes instead of M (the sort-by-row segmented reduction planned
 // for HashEmbed backward, SURVEY.md §7 hard-part 2; also used for the
 // batched parser dPre scatter).
-template <typename T, int CHUNK>
+// DET=true: OUT is an int64 fixed-point buffer (srx_atomic_add) — the
+// per-chunk register accumulation is already order-fixed (sorted entries),
+// and the cross-chunk pushes become associative integer adds, so the
+// result is bit-identical across runs (SRX_DETERMINISTIC).
+template <typename T, int CHUNK, bool DET = false>
 __global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
                                        const int32_t* __restrict__ src_idx,
                                        const T* __restrict__ SRC,
-                                       float* __restrict__ OUT,
+                                       void* __restrict__ OUT,
                                        long M, int W) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
@@ -126,7 +130,8 @@ __global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
       if (d != cur) {
         for (int c = 0; c < ncols; c++) {
           int w = lane + c * SRX_WAVE;
-          if (w < W && acc[c] != 0.f) atomicAdd(OUT + (long)cur * W + w, acc[c]);
+          if (w < W && acc[c] != 0.f)
+            srx_atomic_add<DET>(OUT, (long)cur * W + w, acc[c]);
           acc[c] = 0.f;
         }
         cur = d;
@@ -139,7 +144,8 @@ __global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
     }
     for (int c = 0; c < ncols; c++) {
       int w = lane + c * SRX_WAVE;
-      if (w < W && acc[c] != 0.f) atomicAdd(OUT + (long)cur * W + w, acc[c]);
+      if (w < W && acc[c] != 0.f)
+        srx_atomic_add<DET>(OUT, (long)cur * W + w, acc[c]);
     }
   }
 }
@@ -208,20 +214,24 @@ __global__ void action_select_kernel(const T* __restrict__ scores,
 // lower-precision dPre accumulation; the caller writes dPad into dPre's
 // pad row afterwards.
 //
-// OutBF16=false: dPre is fp32, plain atomicAdd.
-// OutBF16=true:  dPre is bf16 — gfx950 packed global_atomic_pk_add_bf16
+// MODE 0: dPre is fp32, plain atomicAdd.
+// MODE 1: dPre is bf16 — gfx950 packed global_atomic_pk_add_bf16
 //   (unsafeAtomicAdd on __hip_bfloat162), one pair per lane.  Halves the
 //   scatter traffic AND removes the separate fp32->bf16 convert of the
 //   whole [T+1, nF, HP] buffer (6.6 GB read + 3.3 GB write per pipe per
 //   step).  Accumulation rounds per add, acceptable at ~2 contributions
 //   per (token, slot) destination (pad excluded).
-template <typename T, bool OutBF16>
+// MODE 2: deterministic — dPre/dBias/dPad are int64 fixed-point
+//   (srx_atomic_add<true>); bit-identical across runs.
+template <typename T, int MODE>
 __global__ void dpre_scatter_kernel(const T* __restrict__ dSummed,
                                     const int64_t* __restrict__ feats,
                                     void* __restrict__ dPre,
-                                    float* __restrict__ dBias32,
-                                    float* __restrict__ dPad32,
+                                    void* __restrict__ dBias,
+                                    void* __restrict__ dPad,
                                     long S, int nF, int HP, long pad_row) {
+  constexpr bool DET = MODE == 2;
+  constexpr bool OutBF16 = MODE == 1;
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
@@ -266,23 +276,25 @@ __global__ void dpre_scatter_kernel(const T* __restrict__ dSummed,
         }
 #endif
       } else {
-        float* dst = (float*)dPre + (t * (long)nF + f) * HP;
+        long base = (t * (long)nF + f) * HP;
         for (int c = 0; c < ncols; c++) {
           int w = lane + c * SRX_WAVE;
-          if (w < HP) atomicAdd(dst + w, v[c]);
+          if (w < HP) srx_atomic_add<DET>(dPre, base + w, v[c]);
         }
       }
     }
   }
   for (int c = 0; c < ncols * span; c++) {
     int w = span * lane + (c % span) + (c / span) * span * SRX_WAVE;
-    if (w < HP && bias_acc[c] != 0.f) atomicAdd(dBias32 + w, bias_acc[c]);
+    if (w < HP && bias_acc[c] != 0.f)
+      srx_atomic_add<DET>(dBias, w, bias_acc[c]);
   }
   for (int f = 0; f < nF; f++) {
     for (int c = 0; c < ncols * span; c++) {
       int w = span * lane + (c % span) + (c / span) * span * SRX_WAVE;
       float pv = pad_acc[f * ncols * span + c];
-      if (w < HP && pv != 0.f) atomicAdd(dPad32 + (long)f * HP + w, pv);
+      if (w < HP && pv != 0.f)
+        srx_atomic_add<DET>(dPad, (long)f * HP + w, pv);
     }
   }
 }

@@ -171,13 +171,15 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor X, at::Tensor g, at::Tensor b, 
 }
 
 std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
-                                      at::Tensor mu, at::Tensor rstd) {
+                                      at::Tensor mu, at::Tensor rstd,
+                                      bool deterministic = false) {
   check_dev(dY);
   int W = (int)X.size(-1);
   long N = X.numel() / W;
   auto dX = at::empty_like(X);
-  auto dg32 = at::zeros({W}, X.options().dtype(at::kFloat));
-  auto db32 = at::zeros({W}, X.options().dtype(at::kFloat));
+  auto acc_dt = deterministic ? at::kLong : at::kFloat;
+  auto dg32 = at::zeros({W}, X.options().dtype(acc_dt));
+  auto db32 = at::zeros({W}, X.options().dtype(acc_dt));
   auto stream = at::cuda::getCurrentCUDAStream();
   TORCH_CHECK(W <= SRX_LN_MAX_W, "layernorm width > ", SRX_LN_MAX_W);
   if (N > 0) {
@@ -194,13 +196,27 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
     long cap = 4096;
     int grid = (int)std::min<long>((N + 3) / 4, cap);
     DISPATCH_F(X.scalar_type(), {
-      hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t>), dim3(grid),
-                         dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
-                         (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
-                         mu.data_ptr<float>(), rstd.data_ptr<float>(),
-                         (scalar_t*)dX.data_ptr(), dg32.data_ptr<float>(),
-                         db32.data_ptr<float>(), N, W);
+      if (deterministic)
+        hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t, true>), dim3(grid),
+                           dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                           (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
+                           mu.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (scalar_t*)dX.data_ptr(), dg32.data_ptr(),
+                           db32.data_ptr(), N, W);
+      else
+        hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t, false>), dim3(grid),
+                           dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                           (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
+                           mu.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (scalar_t*)dX.data_ptr(), dg32.data_ptr(),
+                           db32.data_ptr(), N, W);
     });
+  }
+  if (deterministic) {
+    // fixed-point -> float (deterministic elementwise divide)
+    auto s = 1.0 / 16777216.0;
+    return {dX, (dg32.to(at::kFloat) * s).to(X.scalar_type()),
+            (db32.to(at::kFloat) * s).to(X.scalar_type())};
   }
   return {dX, dg32.to(X.scalar_type()), db32.to(X.scalar_type())};
 }
@@ -313,10 +329,13 @@ std::vector<at::Tensor> parser_step_bwd(at::Tensor dHidden, at::Tensor feats,
 }
 
 // OUT (fp32 [No, W]) += segmented sums of SRC rows; dst_sorted must be sorted.
+// OUT fp32: plain float atomics.  OUT int64: deterministic fixed-point
+// (caller converts back with /2^24).
 void seg_scatter_add(at::Tensor dst_sorted, at::Tensor src_idx, at::Tensor SRC,
                      at::Tensor OUT) {
   check_dev(SRC);
-  TORCH_CHECK(OUT.scalar_type() == at::kFloat);
+  bool det = OUT.scalar_type() == at::kLong;
+  TORCH_CHECK(det || OUT.scalar_type() == at::kFloat);
   TORCH_CHECK(dst_sorted.scalar_type() == at::kInt && src_idx.scalar_type() == at::kInt);
   long M = dst_sorted.numel();
   int W = (int)SRC.size(-1);
@@ -327,10 +346,16 @@ void seg_scatter_add(at::Tensor dst_sorted, at::Tensor src_idx, at::Tensor SRC,
   long waves = (M + CHUNK - 1) / CHUNK;
   int grid = (int)std::min<long>((waves * SRX_WAVE + kBlock - 1) / kBlock, 16384);
   DISPATCH_F(SRC.scalar_type(), {
-    hipLaunchKernelGGL((seg_scatter_add_kernel<scalar_t, CHUNK>), dim3(grid),
-                       dim3(kBlock), 0, stream, dst_sorted.data_ptr<int32_t>(),
-                       src_idx.data_ptr<int32_t>(), (const scalar_t*)SRC.data_ptr(),
-                       OUT.data_ptr<float>(), M, W);
+    if (det)
+      hipLaunchKernelGGL((seg_scatter_add_kernel<scalar_t, CHUNK, true>), dim3(grid),
+                         dim3(kBlock), 0, stream, dst_sorted.data_ptr<int32_t>(),
+                         src_idx.data_ptr<int32_t>(), (const scalar_t*)SRC.data_ptr(),
+                         OUT.data_ptr(), M, W);
+    else
+      hipLaunchKernelGGL((seg_scatter_add_kernel<scalar_t, CHUNK, false>), dim3(grid),
+                         dim3(kBlock), 0, stream, dst_sorted.data_ptr<int32_t>(),
+                         src_idx.data_ptr<int32_t>(), (const scalar_t*)SRC.data_ptr(),
+                         OUT.data_ptr(), M, W);
   });
 }
 
@@ -377,23 +402,35 @@ void adam_step(at::Tensor grad, at::Tensor master, at::Tensor m, at::Tensor v,
 // (loss_count fp32 [2], dScores) — dScores = softmax_over_valid - target,
 // zero for invalid columns / unsupervised rows (see transition_ce_kernel).
 std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
-                                      at::Tensor valid) {
+                                      at::Tensor valid,
+                                      bool deterministic = false) {
   check_dev(scores);
   long N = scores.size(0);
   int A = (int)scores.size(1);
   auto dScores = at::empty_like(scores);
   auto loss = at::zeros({2}, scores.options().dtype(at::kFloat));
-  auto colsum = at::zeros({(long)A}, scores.options().dtype(at::kFloat));
-  if (N == 0) return {loss, dScores, colsum};
+  auto colsum = at::zeros(
+      {(long)A}, scores.options().dtype(deterministic ? at::kLong : at::kFloat));
+  if (N == 0) return {loss, dScores, colsum.to(at::kFloat)};
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(scores.scalar_type(), {
-    hipLaunchKernelGGL((transition_ce_kernel<scalar_t>),
-                       dim3(grid_for(N * SRX_WAVE)), dim3(kBlock), 0, stream,
-                       (const scalar_t*)scores.data_ptr(),
-                       gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
-                       (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
-                       colsum.data_ptr<float>(), N, A);
+    if (deterministic)
+      hipLaunchKernelGGL((transition_ce_kernel<scalar_t, true>),
+                         dim3(grid_for(N * SRX_WAVE)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)scores.data_ptr(),
+                         gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
+                         (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
+                         colsum.data_ptr(), N, A);
+    else
+      hipLaunchKernelGGL((transition_ce_kernel<scalar_t, false>),
+                         dim3(grid_for(N * SRX_WAVE)), dim3(kBlock), 0, stream,
+                         (const scalar_t*)scores.data_ptr(),
+                         gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
+                         (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
+                         colsum.data_ptr(), N, A);
   });
+  if (deterministic)
+    colsum = colsum.to(at::kFloat) * (1.0 / 16777216.0);
   return {loss, dScores, colsum};
 }
 
@@ -412,34 +449,47 @@ std::vector<at::Tensor> dpre_scatter(at::Tensor dSummed, at::Tensor feats,
   int HP = (int)dPre.size(-1);
   TORCH_CHECK(HP <= 256, "dpre_scatter HP <= 256");
   TORCH_CHECK(nF <= 16, "dpre_scatter nF <= 16");
-  bool out_bf16 = dPre.scalar_type() == at::kBFloat16;
-  TORCH_CHECK(out_bf16 || dPre.scalar_type() == at::kFloat,
-              "dPre must be bf16 or fp32");
-  TORCH_CHECK(!out_bf16 || HP % 2 == 0, "bf16 dPre needs even HP");
-  auto dBias32 = at::zeros({(long)HP}, dSummed.options().dtype(at::kFloat));
-  auto dPad32 = at::zeros({(long)nF, (long)HP}, dSummed.options().dtype(at::kFloat));
-  if (S == 0) return {dBias32, dPad32};
+  int mode = dPre.scalar_type() == at::kBFloat16 ? 1
+           : dPre.scalar_type() == at::kLong      ? 2
+                                                  : 0;
+  TORCH_CHECK(mode != 0 || dPre.scalar_type() == at::kFloat,
+              "dPre must be bf16, fp32 or int64 (deterministic)");
+  TORCH_CHECK(mode != 1 || HP % 2 == 0, "bf16 dPre needs even HP");
+  auto acc_dt = mode == 2 ? at::kLong : at::kFloat;
+  auto dBias32 = at::zeros({(long)HP}, dSummed.options().dtype(acc_dt));
+  auto dPad32 = at::zeros({(long)nF, (long)HP}, dSummed.options().dtype(acc_dt));
+  auto finish = [&](at::Tensor t) {
+    return mode == 2 ? (t.to(at::kFloat) * (1.0 / 16777216.0)) : t;
+  };
+  if (S == 0) return {finish(dBias32), finish(dPad32)};
   auto stream = at::cuda::getCurrentCUDAStream();
   long waves = (S + 63) / 64;  // >=64 states per wave target
   int grid = (int)std::min<long>((waves * SRX_WAVE + kBlock - 1) / kBlock, 4096);
   grid = std::max(grid, 64);
   DISPATCH_F(dSummed.scalar_type(), {
-    if (out_bf16)
-      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, true>), dim3(grid),
+    if (mode == 1)
+      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, 1>), dim3(grid),
                          dim3(kBlock), 0, stream,
                          (const scalar_t*)dSummed.data_ptr(),
                          feats.data_ptr<int64_t>(), dPre.data_ptr(),
-                         dBias32.data_ptr<float>(), dPad32.data_ptr<float>(),
+                         dBias32.data_ptr(), dPad32.data_ptr(),
+                         S, nF, HP, pad_row);
+    else if (mode == 2)
+      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, 2>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)dSummed.data_ptr(),
+                         feats.data_ptr<int64_t>(), dPre.data_ptr(),
+                         dBias32.data_ptr(), dPad32.data_ptr(),
                          S, nF, HP, pad_row);
     else
-      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, false>), dim3(grid),
+      hipLaunchKernelGGL((dpre_scatter_kernel<scalar_t, 0>), dim3(grid),
                          dim3(kBlock), 0, stream,
                          (const scalar_t*)dSummed.data_ptr(),
                          feats.data_ptr<int64_t>(), dPre.data_ptr(),
-                         dBias32.data_ptr<float>(), dPad32.data_ptr<float>(),
+                         dBias32.data_ptr(), dPad32.data_ptr(),
                          S, nF, HP, pad_row);
   });
-  return {dBias32, dPad32};
+  return {finish(dBias32), finish(dPad32)};
 }
 
 // ------------------------------------------------------- softmax + CE
@@ -690,7 +740,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxout_fwd", &maxout_fwd);
   m.def("maxout_bwd", &maxout_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
-  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_bwd", &layernorm_bwd, py::arg("dY"), py::arg("X"),
+        py::arg("g"), py::arg("mu"), py::arg("rstd"),
+        py::arg("deterministic") = false);
   m.def("hashembed_fwd", &hashembed_fwd);
   m.def("hashembed_bwd", &hashembed_bwd);
   m.def("parser_step_fwd", &parser_step_fwd);
@@ -706,7 +758,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mwe_layer_fwd", &mwe_layer_fwd);
   m.def("fused_step", &fusedstep::fused_step);
   m.def("fused_entries_take", &fusedstep::fused_entries_take);
-  m.def("transition_ce", &transition_ce);
+  m.def("transition_ce", &transition_ce, py::arg("scores"), py::arg("gold"),
+        py::arg("valid"), py::arg("deterministic") = false);
+  m.attr("FIXED_SCALE") = 16777216.0;
   m.def("dpre_scatter", &dpre_scatter);
   m.def("run_transition_loop", &srx_run_transition_loop,
         py::call_guard<py::gil_scoped_release>());

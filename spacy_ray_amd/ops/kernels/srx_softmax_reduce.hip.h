@@ -60,13 +60,15 @@ __global__ void softmax_ce_kernel(const T* __restrict__ scores,
 // colsum_out [A] fp32 (zero-initialized by caller) additionally receives
 // sum_n dScores[n, :] — the (unscaled) upper-bias gradient, fused here so
 // the backward skips a full [SS, A] column-reduce pass.
-template <typename T>
+// DET=true: colsum accumulates in int64 fixed-point (it feeds dUpperB; the
+// loss scalar stays a float atomic — display-only, never differentiated).
+template <typename T, bool DET = false>
 __global__ void transition_ce_kernel(const T* __restrict__ scores,
                                      const uint8_t* __restrict__ gold,
                                      const uint8_t* __restrict__ valid,
                                      T* __restrict__ dScores,
                                      float* __restrict__ loss_out,
-                                     float* __restrict__ colsum_out,
+                                     void* __restrict__ colsum_out,
                                      long N, int A) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
@@ -136,7 +138,7 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
   }
   for (int c = 0; c < ncols; c++) {
     int a = lane + c * SRX_WAVE;
-    if (a < A && col_acc[c] != 0.f) atomicAdd(colsum_out + a, col_acc[c]);
+    if (a < A && col_acc[c] != 0.f) srx_atomic_add<DET>(colsum_out, a, col_acc[c]);
   }
 }
 
