@@ -38,6 +38,10 @@ class Comm:
     def broadcast_obj(self, obj: Any, src: int = 0) -> Any:
         raise NotImplementedError
 
+    def all_gather_obj(self, obj: Any) -> list:
+        """Every rank's object, in rank order (sharded-eval merge)."""
+        return [obj]
+
     def barrier(self) -> None:
         pass
 
@@ -117,6 +121,11 @@ class DistComm(Comm):
         holder = [obj if self.rank == src else None]
         dist.broadcast_object_list(holder, src=src)
         return holder[0]
+
+    def all_gather_obj(self, obj):
+        out = [None] * self.world
+        dist.all_gather_object(out, obj)
+        return out
 
     def barrier(self) -> None:
         if self._oplog is not None:

@@ -32,8 +32,11 @@ def _ents_to_spans(ents: List[str]) -> Set[Tuple[int, int, str]]:
     return spans
 
 
-def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict[str, float]:
-    scores: Dict[str, float] = {}
+def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict[str, int]:
+    """Additive per-example counts — mergeable across rank shards (the
+    distributed eval all-gathers these instead of idling N-1 GPUs while
+    rank 0 scores the whole dev set)."""
+    c: Dict[str, int] = {}
     if "tagger" in pipe_names:
         correct = total = 0
         for eg in examples:
@@ -43,7 +46,7 @@ def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Di
             for g, p in zip(gold, pred):
                 total += 1
                 correct += int(g == p)
-        scores["tag_acc"] = correct / total if total else 0.0
+        c["tag_correct"], c["tag_total"] = correct, total
     if "parser" in pipe_names:
         uas = las = total = 0
         for eg in examples:
@@ -62,8 +65,7 @@ def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Di
                     uas += 1
                     if gd and pd and gd[i] == pd[i]:
                         las += 1
-        scores["dep_uas"] = uas / total if total else 0.0
-        scores["dep_las"] = las / total if total else 0.0
+        c["dep_uas_c"], c["dep_las_c"], c["dep_total"] = uas, las, total
     if "ner" in pipe_names:
         tp = fp = fn = 0
         for eg in examples:
@@ -72,12 +74,38 @@ def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Di
             tp += len(gold & pred)
             fp += len(pred - gold)
             fn += len(gold - pred)
+        c["ner_tp"], c["ner_fp"], c["ner_fn"] = tp, fp, fn
+    return c
+
+
+def merge_counts(parts: Sequence[Dict[str, int]]) -> Dict[str, int]:
+    out: Dict[str, int] = {}
+    for part in parts:
+        for k, v in (part or {}).items():
+            out[k] = out.get(k, 0) + v
+    return out
+
+
+def counts_to_scores(c: Dict[str, int]) -> Dict[str, float]:
+    scores: Dict[str, float] = {}
+    if "tag_total" in c:
+        scores["tag_acc"] = c["tag_correct"] / c["tag_total"] if c["tag_total"] else 0.0
+    if "dep_total" in c:
+        t = c["dep_total"]
+        scores["dep_uas"] = c["dep_uas_c"] / t if t else 0.0
+        scores["dep_las"] = c["dep_las_c"] / t if t else 0.0
+    if "ner_tp" in c:
+        tp, fp, fn = c["ner_tp"], c["ner_fp"], c["ner_fn"]
         p = tp / (tp + fp) if tp + fp else 0.0
         r = tp / (tp + fn) if tp + fn else 0.0
         scores["ents_p"] = p
         scores["ents_r"] = r
         scores["ents_f"] = 2 * p * r / (p + r) if p + r else 0.0
     return scores
+
+
+def score_examples(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict[str, float]:
+    return counts_to_scores(score_counts(examples, pipe_names))
 
 
 def weighted_score(scores: Dict[str, float], weights: Dict[str, float]) -> float:
