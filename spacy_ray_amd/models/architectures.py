@@ -89,13 +89,20 @@ def make_transformer_model(
     window: int = 128,
     stride: int = 96,
     attn_implementation: str = "sdpa",
+    subwords: str = "bpe",
+    bpe_vocab_size: int = 8000,
 ):
+    tk = tokenizer_config or {}
+
     def build():
         from .transformer import TransformerTok2Vec
 
         return TransformerTok2Vec(name=name, window=window, stride=stride,
                                   transformer_config=transformer_config,
-                                  attn_implementation=attn_implementation)
+                                  attn_implementation=attn_implementation,
+                                  subwords=tk.get("subwords", subwords),
+                                  bpe_vocab_size=tk.get("bpe_vocab_size", bpe_vocab_size),
+                                  tokenizer_path=tk.get("tokenizer_path"))
 
     tc = transformer_config or {}
     width = tc.get("hidden_size", 768 if name == "roberta-base" else 64)

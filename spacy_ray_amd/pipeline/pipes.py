@@ -82,14 +82,31 @@ class Tok2VecPipe(TrainablePipe):
     def initialize(self, examples, device) -> None:
         if self.module is None:
             self.module = self.spec.build().to(device)
+        if hasattr(self.module, "init_bpe"):
+            # transformer tok2vec: train/load the byte-level BPE on the
+            # corpus sample (offline — no pretrained vocab available)
+            self.module.init_bpe(examples)
 
     def load_cfg(self, cfg, device) -> None:
         super().load_cfg(cfg, device)
         if self.module is None:
             self.module = self.spec.build().to(device)
+        if hasattr(self.module, "init_bpe") and "bpe_vocab" in cfg:
+            # checkpoint restore: rebuild the trained tokenizer from its
+            # serialized JSON (saved in state_cfg)
+            from tokenizers import Tokenizer as HFTokenizer
+
+            self.module.bpe.tok = HFTokenizer.from_str(cfg["bpe_vocab"])
 
     def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
         return self.module(batch, drop=drop)
+
+    def state_cfg(self) -> Dict:
+        cfg = dict(self.cfg)
+        bpe = getattr(self.module, "bpe", None) if self.module is not None else None
+        if bpe is not None and bpe.tok is not None:
+            cfg["bpe_vocab"] = bpe.tok.to_str()  # tokenizer.json content
+        return cfg
 
     def get_loss(self, examples, t2v, batch):
         return t2v.new_zeros(()), 0.0

@@ -160,3 +160,48 @@ def test_transformer_long_doc_window_stitching():
     assert covered == set(range(20))
     # stride < window => interior tokens appear in overlapping windows
     assert len(spans) == 4  # starts 0,5,10,15 with window 8 over 20
+
+
+def test_bpe_subwords_alignment_and_checkpoint(tmp_path):
+    """Default subwords='bpe': real byte-level BPE trained at init; output
+    stays [n_words, width] via word-alignment pooling; the trained
+    tokenizer round-trips through the checkpoint."""
+    from spacy_ray_amd.vocab.doc import Doc, Example
+
+    cfg = Config.from_str(TRF_CFG)
+    nlp = init_nlp(cfg)
+    trf = nlp.get_pipe("transformer").module
+    assert trf.bpe is not None and trf.bpe.tok is not None
+    # a word the trainer never saw still tokenizes (byte-level has no OOV)
+    doc = Doc(nlp.vocab, ["w1", "zzzzunseenzzzz", "w3"])
+    ids, wid = trf.bpe.encode_doc(doc)
+    assert len(ids) >= 3 and wid.max() == 2
+    from spacy_ray_amd.models.batch import TokenBatch
+
+    tb = TokenBatch([doc], torch.device("cpu"))
+    out = trf(tb)
+    assert out.shape == (tb.n_tokens, trf.width)
+    # checkpoint: the serialized tokenizer restores identical segmentation
+    out_dir = tmp_path / "m"
+    nlp.to_disk(out_dir)
+    nlp2 = init_nlp(cfg)
+    nlp2.from_disk(out_dir)
+    trf2 = nlp2.get_pipe("transformer").module
+    d2 = Doc(nlp2.vocab, ["w1", "zzzzunseenzzzz", "w3"])
+    ids2, wid2 = trf2.bpe.encode_doc(d2)
+    assert ids2.tolist() == ids.tolist()
+    assert wid2.tolist() == wid.tolist()
+
+
+def test_hash_subwords_fallback():
+    cfg_text = TRF_CFG.replace(
+        'stride = 12', 'stride = 12\nsubwords = "hash"')
+    nlp = init_nlp(Config.from_str(cfg_text))
+    trf = nlp.get_pipe("transformer").module
+    assert trf.bpe is None
+    from spacy_ray_amd.models.batch import TokenBatch
+    from spacy_ray_amd.vocab.doc import Doc
+
+    doc = Doc(nlp.vocab, ["a", "b", "c", "d"])
+    out = trf(TokenBatch([doc], torch.device("cpu")))
+    assert out.shape[0] == 4
