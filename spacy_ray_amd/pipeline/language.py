@@ -381,6 +381,11 @@ def init_nlp(config: Config, device: str = "cpu", sample_size: int = 128) -> Lan
     torch.manual_seed(seed)
     nlp = build_nlp(config, device=device)
     (train_corpus,) = resolve_dot_names(cfg, [cfg["training"]["train_corpus"]])
+    if "transformer" in nlp.pipe_names:
+        # the transformer trains its byte-level BPE on this sample at init;
+        # a tiny sample under-merges and every word fragments into many
+        # subwords (3-4x sequence inflation = 3-4x activation memory)
+        sample_size = max(sample_size, 1024)
     sample: List[Example] = []
     tag_labels: set = set()
     dep_labels: set = set()
