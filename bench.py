@@ -23,7 +23,8 @@ import torch
 
 
 def build_batches(nlp, *, batch_words: int, n_batches: int, seed: int,
-                  words_per_doc: int, vocab_size: int):
+                  words_per_doc: int, vocab_size: int,
+                  n_tags: int = 50, n_deps: int = 40, n_ent_types: int = 4):
     from spacy_ray_amd.data.corpus import make_synthetic_docs
     from spacy_ray_amd.vocab.doc import Example
 
@@ -32,9 +33,9 @@ def build_batches(nlp, *, batch_words: int, n_batches: int, seed: int,
         n_docs=max(64, (batch_words * n_batches) // max(1, words_per_doc)),
         words_per_doc=words_per_doc,
         vocab_size=vocab_size,
-        n_tags=50,
-        n_deps=40,
-        n_ent_types=4,
+        n_tags=n_tags,
+        n_deps=n_deps,
+        n_ent_types=n_ent_types,
         seed=seed,
         world_seed=0,
     )
@@ -95,11 +96,16 @@ def main() -> None:
     engine = ZeRO1Engine(nlp, T["optimizer"], comm)
     torch.manual_seed(1234 + rank)
 
+    # label spaces must match the config's training corpus (the parser/NER
+    # gold mapping fails loudly on labels outside the discovered set)
+    label_space = {
+        "xx_multilingual.cfg": dict(n_tags=17, n_deps=37, n_ent_types=1),
+    }.get(os.path.basename(args.config), {})
     batches = build_batches(
         nlp, batch_words=args.batch_words,
         n_batches=8 if args.batch_words <= 64000 else 4,
         seed=100 + rank, words_per_doc=args.words_per_doc,
-        vocab_size=args.vocab_size,
+        vocab_size=args.vocab_size, **label_space,
     )
     words_per_step = sum(len(eg) for eg in batches[0])
     # fixed replayed batches: precompute the device-side TokenBatch once
