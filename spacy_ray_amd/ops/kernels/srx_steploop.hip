@@ -28,6 +28,7 @@
 #include <cstdint>
 #include <mutex>
 #include <stdexcept>
+#include <thread>
 #include <tuple>
 #include <unordered_map>
 #include <vector>
@@ -312,38 +313,38 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
   }
   hipEventDestroy(start_ev);
 
-  // interleaved driver: while one unit's GPU work + D2H drains, the others'
-  // CPU phases (advance + pack) run — same pipelining as the round-1 python
-  // round-robin, at native per-step cost.
+  // one HOST THREAD per unit: each unit's serial chain (advance -> pack ->
+  // H2D -> kernel -> D2H) runs independently on its own stream, so the
+  // parser's CPU phases overlap the NER's GPU phases AND vice versa
+  // (single-threaded round-robin serialized the pack work of all units:
+  // pack was 27 ms/step of the 52 ms loop at 1M words).
   const long max_iters = 1L << 30;
-  long guard = 0;
-  bool all_done = false;
   static const bool loop_stats = getenv("SRX_LOOP_STATS") != nullptr;
   LoopStats st;
-  double t0 = 0;
-  while (!all_done) {
-    TORCH_CHECK(guard++ < max_iters, "transition loop failed to terminate");
-    all_done = true;
-    for (Unit& u : units) {
-      if (u.done) continue;
+  std::mutex st_mu;
+  auto run_unit = [&](Unit& u) {
+    LoopStats ls;
+    double t0 = 0;
+    long guard = 0;
+    while (!u.done) {
+      if (guard++ >= max_iters) throw std::runtime_error("loop stuck");
       if (u.pending) {
         if (loop_stats) t0 = now_ms();
         spin_wait(u.ev);
-        if (loop_stats) { st.wait_ms += now_ms() - t0; t0 = now_ms(); }
+        if (loop_stats) { ls.wait_ms += now_ms() - t0; t0 = now_ms(); }
         u.b->advance_active(u.act_idx_h, u.actions_h, u.Sa);
-        if (loop_stats) st.adv_ms += now_ms() - t0;
+        if (loop_stats) ls.adv_ms += now_ms() - t0;
         u.pending = false;
       }
       if (loop_stats) t0 = now_ms();
       long Sa = u.b->pack_step(u.train, u.T, u.act_idx_h, u.feats_h, u.valid_h,
                                u.gold_h);
-      if (loop_stats) st.pack_ms += now_ms() - t0;
+      if (loop_stats) ls.pack_ms += now_ms() - t0;
       if (Sa == 0) {
         u.done = true;
         continue;
       }
-      all_done = false;
-      if (loop_stats) { st.iters++; t0 = now_ms(); }
+      if (loop_stats) { ls.iters++; t0 = now_ms(); }
       long off = u.train ? u.used : 0;
       TORCH_CHECK(off + Sa <= u.cap, "transition arena overflow (", off, "+",
                   Sa, " > ", u.cap, ")");
@@ -379,12 +380,32 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
       hipMemcpyAsync(u.actions_h, actions_d, (size_t)Sa * 4,
                      hipMemcpyDeviceToHost, u.hs);
       hipEventRecord(u.ev, u.hs);
-      if (loop_stats) st.gpu_ms += now_ms() - t0;
+      if (loop_stats) ls.gpu_ms += now_ms() - t0;
       u.pending = true;
       u.Sa = Sa;
       u.used += Sa;
       u.steps += 1;
     }
+    if (loop_stats) {
+      std::lock_guard<std::mutex> lock(st_mu);
+      st.iters += ls.iters; st.pack_ms += ls.pack_ms; st.wait_ms += ls.wait_ms;
+      st.gpu_ms += ls.gpu_ms; st.adv_ms += ls.adv_ms;
+    }
+  };
+  if (units.size() <= 1) {
+    for (Unit& u : units) run_unit(u);
+  } else {
+    std::vector<std::thread> threads;
+    threads.reserve(units.size());
+    std::vector<std::exception_ptr> errs(units.size());
+    for (size_t i = 0; i < units.size(); i++)
+      threads.emplace_back([&, i]() {
+        try { run_unit(units[i]); }
+        catch (...) { errs[i] = std::current_exception(); }
+      });
+    for (auto& t : threads) t.join();
+    for (auto& e : errs)
+      if (e) std::rethrow_exception(e);
   }
   if (loop_stats)
     fprintf(stderr,
