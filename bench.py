@@ -16,6 +16,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import numpy as np
@@ -67,7 +68,21 @@ def main() -> None:
     ap.add_argument("--config", type=str, default="examples/configs/en_core_cnn.cfg")
     ap.add_argument("--profile-steps", type=int, default=0,
                     help="if >0, run only this many unsynchronized steps (for rocprof)")
+    ap.add_argument("--sweep", action="store_true",
+                    help="batch-size sweep (3k/32k/128k/1M words): one JSON "
+                         "line per size — the realistic-batch curve next to "
+                         "the 1M-word headline (VERDICT r1 item 8)")
     args = ap.parse_args()
+
+    if args.sweep:
+        import subprocess
+
+        for bw in (3000, 32768, 131072, 1000000):
+            cmd = [sys.executable, __file__, "--steps", str(args.steps),
+                   "--warmup", str(args.warmup), "--batch-words", str(bw),
+                   "--config", args.config]
+            subprocess.run(cmd, check=True)
+        return
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -53,15 +53,26 @@ namespace {
 constexpr float KInvalid = 1e9f;
 
 // ------------------------------------------------------------------ parser
-// Actions: 0=SHIFT, 1=REDUCE, 2..2+L-1 = LEFT-ARC(l), 2+L..2+2L-1 = RIGHT-ARC(l)
-//
 // SoA layout: every token-indexed array is FLAT over the batch; doc d's
 // tokens live at [off[d], off[d]+len[d]).  The per-doc stack is a flat
 // arena slice of the same extent (stack depth <= len).  Stored head/label/
 // child indices are DOC-LOCAL (like the round-1 AoS layout) so the oracle
 // math is unchanged; fill_features adds the doc offset for the GPU gather.
+//
+// Actions: 0=SHIFT, 1=REDUCE, 2..2+L-1=LEFT-ARC(l), 2+L..2+2L-1=RIGHT-ARC(l),
+// [2+2L=BREAK when use_break].  BREAK (sentence boundary, spaCy USE_BREAK
+// contract re-designed — see docs/PARITY.md): valid when at least one token
+// is consumed, the buffer is non-empty and B0 is not already marked; it
+// marks B0 as a sentence start and puts the state into CLEANUP mode — only
+// REDUCE is valid until the stack empties (headless pops stay attached to
+// root, exactly like end-of-buffer cleanup), so no arc can cross the
+// boundary.  Oracle: BREAK costs (gold sent_start[B0] ? 0 : 1) + the number
+// of gold arcs that would cross the boundary (stack tokens with gold heads
+// or gold children at/after B0); SHIFT and RIGHT-ARC over an unmarked gold
+// boundary cost +1 (they pull B0 into the current sentence).
 struct ArcEagerBatch : public srx::StepBatchIface {
   int32_t n_labels;
+  bool use_break = false;
   int64_t n_docs = 0, total = 0;
   int32_t base_offset = 0;
   std::vector<int32_t> off;     // [n_docs + 1]
@@ -71,6 +82,9 @@ struct ArcEagerBatch : public srx::StepBatchIface {
   std::vector<int32_t> stack;   // flat arena [total]
   std::vector<int32_t> head, label, l1, l2, r1, r2;  // flat [total], doc-local ids
   std::vector<int32_t> gold_head, gold_label;        // flat [total]
+  std::vector<int32_t> gold_sent;                    // flat [total]; empty = none
+  std::vector<int32_t> sent_out;                     // flat [total] predicted starts
+  std::vector<uint8_t> pending_break;                // [n_docs] cleanup mode
   bool has_gold = false;
   // O(1)-oracle bookkeeping (fill_costs was 28 ms/step at 1M words with the
   // O(stack + buffer) formulation):
@@ -80,10 +94,14 @@ struct ArcEagerBatch : public srx::StepBatchIface {
   std::vector<int32_t> kids;            // CSR payload (doc-local child ids)
 
   ArcEagerBatch(py::array_t<int32_t, py::array::c_style | py::array::forcecast> lengths,
-                int32_t n_labels_, int32_t base_offset_ = 0)
-      : n_labels(n_labels_), base_offset(base_offset_) {
+                int32_t n_labels_, int32_t base_offset_ = 0,
+                bool use_break_ = false)
+      : n_labels(n_labels_), use_break(use_break_), base_offset(base_offset_) {
     if (n_actions() > 256)
-      throw std::runtime_error("ArcEagerBatch: > 127 dep labels unsupported");
+      throw std::runtime_error(
+          "ArcEagerBatch: action space > 256 (more than ~127 dep labels); "
+          "the per-state cost buffer is fixed at 256 actions — split the "
+          "label set or raise the cap in transitions.cpp/pack_step");
     auto L = lengths.unchecked<1>();
     n_docs = L.shape(0);
     off.resize(n_docs + 1);
@@ -104,6 +122,16 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     r1.assign(total, -1);
     r2.assign(total, -1);
     on_stack.assign(total, 0);
+    sent_out.assign(total, 0);
+    pending_break.assign(n_docs, 0);
+  }
+
+  void set_sent_gold(py::array_t<int32_t, py::array::c_style | py::array::forcecast> sents) {
+    auto S = sents.unchecked<1>();
+    if ((int64_t)S.shape(0) < base_offset + total)
+      throw std::runtime_error("set_sent_gold: array shorter than batch");
+    gold_sent.resize(total);
+    std::memcpy(gold_sent.data(), S.data(0) + base_offset, total * sizeof(int32_t));
   }
 
   void set_gold(py::array_t<int32_t, py::array::c_style | py::array::forcecast> heads,
@@ -153,7 +181,7 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     }
   }
 
-  int32_t n_actions() const { return 2 + 2 * n_labels; }
+  int32_t n_actions() const { return 2 + 2 * n_labels + (use_break ? 1 : 0); }
   size_t size() const { return (size_t)n_docs; }
 
   inline bool final_state(int64_t d) const {
@@ -223,6 +251,10 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     bool has_s0 = ssize[d] > 0;
     bool s0_has_head = has_s0 && head[off[d] + s0(d)] != -1;
     std::fill(v, v + A, 0);
+    if (use_break && pending_break[d] && has_s0) {
+      v[1] = 1;  // cleanup mode after BREAK: force pops until stack empties
+      return;
+    }
     if (has_buf) v[0] = 1;                                    // SHIFT
     // REDUCE: s0 has a head; or forced cleanup when the buffer is exhausted
     // (headless pops attach to root), so every non-final state has >=1 valid
@@ -234,6 +266,9 @@ struct ArcEagerBatch : public srx::StepBatchIface {
       v[2 + l] = la_ok ? 1 : 0;
       v[2 + n_labels + l] = ra_ok ? 1 : 0;
     }
+    if (use_break)
+      v[2 + 2 * n_labels] =
+          (has_buf && buf[d] > 0 && !sent_out[off[d] + buf[d]]) ? 1 : 0;
   }
 
   py::array_t<uint8_t> valid() const {
@@ -280,6 +315,16 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     } else if (v0 >= 0) {
       c_reduce = (float)gold_kids_buf[o + v0];
     }
+    // BREAK oracle terms: an unmarked gold boundary at B0 penalizes the
+    // moves that pull B0 into the current sentence (SHIFT / RIGHT-ARC);
+    // BREAK itself additionally pays for every gold arc it would sever
+    // (stack tokens whose gold head or gold children lie at/after B0).
+    bool at_gold_break = use_break && !gold_sent.empty() && b > 0 &&
+                         gold_sent[o + b] == 1 && !sent_out[o + b];
+    if (at_gold_break) {
+      c_shift += 1;
+      c_ra += 1;
+    }
     r[0] = v[0] ? c_shift : KInvalid;
     r[1] = v[1] ? c_reduce : KInvalid;
     for (int32_t l = 0; l < n_labels; l++) {
@@ -290,6 +335,19 @@ struct ArcEagerBatch : public srx::StepBatchIface {
       }
       r[2 + l] = v[2 + l] ? la : KInvalid;
       r[2 + n_labels + l] = v[2 + n_labels + l] ? ra : KInvalid;
+    }
+    if (use_break) {
+      const int32_t bk = 2 + 2 * n_labels;
+      float c_break = at_gold_break ? 0.f : 1.f;
+      if (v[bk] && !gold_sent.empty()) {
+        const int32_t* stk = stack.data() + o;
+        for (int32_t k = 0; k < ssize[d]; k++) {
+          int32_t s = stk[k];
+          if (head[o + s] == -1 && gh[s] >= b) c_break += 1;  // head severed
+          c_break += (float)gold_kids_buf[o + s];  // children severed
+        }
+      }
+      r[bk] = v[bk] ? c_break : KInvalid;
     }
   }
 
@@ -310,8 +368,9 @@ struct ArcEagerBatch : public srx::StepBatchIface {
   int n_acts() const override { return (int)n_actions(); }
   int64_t max_transitions() const override {
     // buf advances exactly len times (SHIFT/RIGHT-ARC); each push is popped
-    // at most once (REDUCE/LEFT-ARC) => <= 2*len transitions per doc.
-    return 2 * total;
+    // at most once (REDUCE/LEFT-ARC); BREAK marks each token at most once
+    // => <= (2|3)*len transitions per doc.
+    return (use_break ? 3 : 2) * total;
   }
 
   int64_t pack_step(bool with_gold, int64_t pad_row, int32_t* act_idx,
@@ -357,6 +416,13 @@ struct ArcEagerBatch : public srx::StepBatchIface {
 
   inline void apply_action(int64_t d, int32_t act) {
     const int64_t o = off[d];
+    if (use_break && act == 2 + 2 * n_labels) {  // BREAK: mark B0 sent start
+      sent_out[o + buf[d]] = 1;
+      if (ssize[d] > 0) pending_break[d] = 1;  // cleanup pops follow
+      return;
+    }
+    if (use_break && act == 1 && pending_break[d] && ssize[d] <= 1)
+      pending_break[d] = 0;  // this pop empties the stack
     if (act == 0) {  // SHIFT
       on_stack[o + buf[d]] = 1;
       leave_buffer(d, buf[d]);
@@ -460,6 +526,12 @@ struct ArcEagerBatch : public srx::StepBatchIface {
   py::array_t<int32_t> labels() const {
     py::array_t<int32_t> out((py::ssize_t)total);
     std::memcpy(out.mutable_data(), label.data(), total * sizeof(int32_t));
+    return out;
+  }
+
+  py::array_t<int32_t> sent_starts() const {
+    py::array_t<int32_t> out((py::ssize_t)total);
+    std::memcpy(out.mutable_data(), sent_out.data(), total * sizeof(int32_t));
     return out;
   }
 };
@@ -705,9 +777,12 @@ struct BiluoBatch : public srx::StepBatchIface {
 
 void init_transitions(py::module_& m) {
   py::class_<ArcEagerBatch>(m, "ArcEagerBatch")
-      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t, int32_t>(),
-           py::arg("lengths"), py::arg("n_labels"), py::arg("base_offset") = 0)
+      .def(py::init<py::array_t<int32_t, py::array::c_style | py::array::forcecast>, int32_t, int32_t, bool>(),
+           py::arg("lengths"), py::arg("n_labels"), py::arg("base_offset") = 0,
+           py::arg("use_break") = false)
       .def("set_gold", &ArcEagerBatch::set_gold, py::arg("heads"), py::arg("labels"))
+      .def("set_sent_gold", &ArcEagerBatch::set_sent_gold, py::arg("sent_starts"))
+      .def("sent_starts", &ArcEagerBatch::sent_starts)
       .def_property_readonly("n_actions", &ArcEagerBatch::n_actions)
       .def("__len__", &ArcEagerBatch::size)
       .def("is_final", &ArcEagerBatch::is_final)

@@ -32,8 +32,8 @@ def make_tagger_pipe(name: str, model, labels=None):
 
 
 @registry.factories("parser")
-def make_parser_pipe(name: str, model, labels=None):
-    return _with_labels(ParserPipe(name, model), labels)
+def make_parser_pipe(name: str, model, labels=None, use_break: bool = False):
+    return _with_labels(ParserPipe(name, model, use_break=use_break), labels)
 
 
 @registry.factories("ner")

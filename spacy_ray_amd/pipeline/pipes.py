@@ -652,6 +652,12 @@ class _TransitionPipeBase(TrainablePipe):
 class ParserPipe(_TransitionPipeBase):
     name = "parser"
 
+    def __init__(self, name: str, spec, use_break: bool = False) -> None:
+        super().__init__(name, spec)
+        # spaCy USE_BREAK contract (sentence boundaries learned as a BREAK
+        # transition) — see transitions.cpp / docs/PARITY.md; off by default
+        self.use_break = use_break
+
     def initialize(self, examples, device) -> None:
         if not self.labels:
             labels = set()
@@ -661,13 +667,19 @@ class ParserPipe(_TransitionPipeBase):
             self.labels = sorted(labels) or ["dep"]
             self.label2id = {t: i for i, t in enumerate(self.labels)}
         self.cfg["labels"] = self.labels
+        self.cfg["use_break"] = self.use_break
         self._build_module(device)
 
+    def load_cfg(self, cfg, device) -> None:
+        self.use_break = bool(cfg.get("use_break", self.use_break))
+        super().load_cfg(cfg, device)
+
     def _n_actions(self) -> int:
-        return 2 + 2 * len(self.labels)
+        return 2 + 2 * len(self.labels) + (1 if self.use_break else 0)
 
     def _make_states(self, lengths, base: int = 0):
-        return _srx_cpu.ArcEagerBatch(lengths, len(self.labels), base)
+        return _srx_cpu.ArcEagerBatch(lengths, len(self.labels), base,
+                                      self.use_break)
 
     def _gold_arrays(self, eg) -> Tuple[np.ndarray, np.ndarray]:
         """Per-example (heads, label-ids), cached on the reference Doc —
@@ -704,15 +716,27 @@ class ParserPipe(_TransitionPipeBase):
 
     def _build_gold(self, examples):
         pairs = [self._gold_arrays(eg) for eg in examples]
-        return (np.concatenate([p[0] for p in pairs]),
-                np.concatenate([p[1] for p in pairs]))
+        heads = np.concatenate([p[0] for p in pairs])
+        labs = np.concatenate([p[1] for p in pairs])
+        sents = None
+        if self.use_break and any(eg.reference.sent_starts is not None
+                                  for eg in examples):
+            sents = np.concatenate([
+                eg.reference.sent_starts if eg.reference.sent_starts is not None
+                else np.zeros(len(eg.reference), dtype=np.int32)
+                for eg in examples
+            ]).astype(np.int32)
+        return (heads, labs, sents)
 
     def _set_gold(self, states, staged) -> None:
         states.set_gold(staged[0], staged[1])
+        if len(staged) > 2 and staged[2] is not None:
+            states.set_sent_gold(staged[2])
 
     def _annotate(self, docs, states) -> None:
         heads = states.heads()
         labels = states.labels()
+        sents = states.sent_starts() if self.use_break else None
         off = 0
         for doc in docs:
             n = len(doc)
@@ -724,6 +748,11 @@ class ParserPipe(_TransitionPipeBase):
             for i in range(n):
                 if doc.heads[i] == -1:
                     doc.deps[i] = "ROOT"
+            if sents is not None:
+                ss = sents[off:off + n].copy()
+                if n > 0:
+                    ss[0] = 1  # the first token always starts a sentence
+                doc.sent_starts = ss
             off += n
 
 

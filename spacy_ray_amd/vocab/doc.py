@@ -32,7 +32,7 @@ class Vocab:
 class Doc:
     __slots__ = (
         "vocab", "words", "spaces", "attr_hashes",
-        "tags", "heads", "deps", "ents",
+        "tags", "heads", "deps", "ents", "sent_starts",
         "tensor", "user_data",
     )
 
@@ -46,6 +46,7 @@ class Doc:
         heads: Optional[Sequence[int]] = None,
         deps: Optional[Sequence[str]] = None,
         ents: Optional[Sequence[str]] = None,  # per-token BILUO strings, e.g. "B-ORG"/"O"
+        sent_starts: Optional[Sequence[int]] = None,  # 1 = starts a sentence
         attr_hashes: Optional[np.ndarray] = None,  # precomputed (n,4) uint64
     ) -> None:
         self.vocab = vocab
@@ -59,6 +60,8 @@ class Doc:
         self.heads = np.asarray(heads, dtype=np.int32) if heads is not None else None
         self.deps = list(deps) if deps is not None else None
         self.ents = list(ents) if ents is not None else None
+        self.sent_starts = (np.asarray(sent_starts, dtype=np.int32)
+                            if sent_starts is not None else None)
         self.tensor: Optional[np.ndarray] = None
         self.user_data: Dict = {}
 
@@ -92,6 +95,8 @@ class Doc:
             "heads": self.heads.tolist() if self.heads is not None else None,
             "deps": self.deps,
             "ents": self.ents,
+            "sent_starts": (self.sent_starts.tolist()
+                            if self.sent_starts is not None else None),
         }
 
     @classmethod
@@ -104,6 +109,7 @@ class Doc:
             heads=data.get("heads"),
             deps=data.get("deps"),
             ents=data.get("ents"),
+            sent_starts=data.get("sent_starts"),
         )
 
 

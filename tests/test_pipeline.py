@@ -576,3 +576,92 @@ def test_missing_biluo_excluded_from_loss():
     batch.advance(np.array([4], dtype=np.int32))  # U-PER on token 0
     act, feats, valid, gold = batch.step_arrays(True)
     assert gold[0].sum() == 0  # all-zero gold row -> masked out of the CE
+
+
+def test_parser_use_break_learns_sentence_boundaries():
+    """use_break=true: the parser trains through BREAK transitions and
+    annotates sent_starts at predict time."""
+    import numpy as np
+
+    from spacy_ray_amd.config.registry import registry
+    from spacy_ray_amd.vocab.doc import Doc, Example
+
+    @registry.readers("test.SentCorpus.v1")
+    def sent_corpus():
+        def corpus(nlp):
+            rng = np.random.RandomState(0)
+            for i in range(300):
+                # two 3-token "sentences": w-root chain per sentence
+                words = [f"a{rng.randint(6)}", "b", "c",
+                         f"d{rng.randint(6)}", "e", "f"]
+                heads = [-1, 0, 0, -1, 3, 3]
+                deps = ["ROOT", "x", "y", "ROOT", "x", "y"]
+                sents = [1, 0, 0, 1, 0, 0]
+                yield Example.from_doc(Doc(nlp.vocab, words, heads=heads,
+                                           deps=deps, sent_starts=sents))
+        return corpus
+
+    cfg_text = """
+[nlp]
+lang = "en"
+pipeline = ["tok2vec", "parser"]
+
+[components]
+
+[components.tok2vec]
+factory = "tok2vec"
+
+[components.tok2vec.model]
+@architectures = "spacy.HashEmbedCNN.v2"
+width = 32
+depth = 1
+embed_size = 200
+
+[components.parser]
+factory = "parser"
+use_break = true
+
+[components.parser.model]
+@architectures = "spacy.TransitionBasedParser.v2"
+state_type = "parser"
+
+[components.parser.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 32
+
+[corpora]
+
+[corpora.train]
+@readers = "test.SentCorpus.v1"
+
+[training]
+seed = 0
+train_corpus = "corpora.train"
+dev_corpus = "corpora.train"
+
+[training.optimizer]
+@optimizers = "Adam.v1"
+learn_rate = 0.01
+"""
+    cfg = Config.from_str(cfg_text)
+    nlp = init_nlp(cfg)
+    parser = nlp.get_pipe("parser")
+    assert parser.use_break
+    assert parser._n_actions() == 2 + 2 * len(parser.labels) + 1
+    icfg = cfg.interpolate()
+    train_corpus = resolve_dot_names(icfg, ["corpora.train"])[0]
+    T = resolve(icfg["training"], validate=False)
+    stepper = SimpleStepper(nlp, T["optimizer"])
+    egs = list(train_corpus(nlp))
+    for _ in range(30):
+        stepper.accumulate(egs[:64], drop=0.0, losses={})
+        stepper.apply_step()
+    docs = [eg.predicted for eg in egs[:8]]
+    nlp.predict_docs(docs)
+    hits = total = 0
+    for d in docs:
+        assert d.sent_starts is not None
+        assert d.sent_starts[0] == 1
+        hits += int(d.sent_starts[3] == 1)
+        total += 1
+    assert hits >= total * 0.7, (hits, total)  # boundary learned

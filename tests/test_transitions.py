@@ -193,3 +193,57 @@ def test_oracle_costs_match_bruteforce_reference():
                 head[stack[-1]] = buf; stack.pop()
             else:
                 head[buf] = stack[-1]; stack.append(buf); buf += 1
+
+
+def test_break_action_sentence_boundaries():
+    """BREAK (use_break=True): valid only with an empty stack and consumed
+    prefix; oracle prefers it exactly at gold sentence starts; predicted
+    boundaries come back via sent_starts()."""
+    import numpy as np
+    from spacy_ray_amd import _srx_cpu
+
+    # two sentences of 2 tokens each: heads [1,-1, 3,-1] (heads within sent)
+    n, L = 4, 1
+    batch = _srx_cpu.ArcEagerBatch(np.array([n], dtype=np.int32), L, 0, True)
+    A = batch.n_actions
+    assert A == 2 + 2 * L + 1
+    BREAK = A - 1
+    heads = np.array([1, -1, 3, -1], dtype=np.int32)
+    labels = np.zeros(n, dtype=np.int32)
+    batch.set_gold(heads, labels)
+    batch.set_sent_gold(np.array([1, 0, 1, 0], dtype=np.int32))
+    # initial state: stack empty, buf=0 -> BREAK invalid (nothing consumed)
+    assert batch.valid()[0][BREAK] == 0
+    # follow min-cost actions greedily; collect them
+    taken = []
+    for _ in range(4 * n):
+        if batch.is_final()[0]:
+            break
+        costs = batch.costs()[0]
+        act = int(np.argmin(costs))
+        taken.append(act)
+        batch.advance(np.array([act], dtype=np.int32))
+    assert BREAK in taken  # the oracle used the boundary
+    sents = batch.sent_starts()
+    assert sents[2] == 1  # token 2 marked as a sentence start
+    assert sents[1] == 0 and sents[3] == 0
+    # parse quality unaffected by the break
+    assert batch.heads()[0] == 1 and batch.heads()[2] == 3
+
+
+def test_break_never_preferred_without_sentence_gold():
+    import numpy as np
+    from spacy_ray_amd import _srx_cpu
+
+    batch = _srx_cpu.ArcEagerBatch(np.array([4], dtype=np.int32), 1, 0, True)
+    BREAK = batch.n_actions - 1
+    batch.set_gold(np.array([1, -1, 1, 1], dtype=np.int32),
+                   np.zeros(4, dtype=np.int32))
+    for _ in range(16):
+        if batch.is_final()[0]:
+            break
+        costs = batch.costs()[0]
+        act = int(np.argmin(costs))
+        assert act != BREAK  # cost 1 > some zero-cost action
+        batch.advance(np.array([act], dtype=np.int32))
+    assert batch.sent_starts().sum() == 0
