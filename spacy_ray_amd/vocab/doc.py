@@ -126,6 +126,25 @@ class Example:
     def from_doc(cls, doc: Doc) -> "Example":
         return cls(doc.copy_unannotated(), doc)
 
+    @classmethod
+    def from_docs(cls, predicted: Doc, reference: Doc) -> "Example":
+        """Pair a predicted Doc with a DIFFERENTLY-tokenized reference: the
+        reference annotations are char-aligned and projected onto the
+        predicted tokenization (spaCy Example(predicted, reference)
+        contract; unalignable tokens get the missing marker).  Same
+        tokenization short-circuits to a direct pair."""
+        if predicted.words == reference.words:
+            return cls(predicted, reference)
+        from .align import project_reference
+
+        tags, ents, sents, heads, deps = project_reference(predicted, reference)
+        projected = Doc(
+            predicted.vocab, predicted.words, spaces=predicted.spaces,
+            tags=tags, heads=heads, deps=deps, ents=ents, sent_starts=sents,
+            attr_hashes=predicted.attr_hashes,
+        )
+        return cls(predicted, projected)
+
     def __len__(self) -> int:
         return len(self.reference)
 
