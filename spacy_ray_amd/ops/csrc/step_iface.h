@@ -37,14 +37,6 @@ struct StepBatchIface {
   virtual int64_t pack_step(bool with_gold, int64_t pad_row, int32_t* act_idx,
                             int64_t* feats, uint8_t* valid, uint8_t* gold) = 0;
 
-  // Bit-packed variant (the GPU step loop's format): valid/gold are
-  // uint64 bitmask words, ceil(A/64) per state, bit a = action a.  Mask
-  // write traffic is A/8 bytes instead of A — the per-step pack runs on
-  // the loop's critical chain.
-  virtual int64_t pack_step_bits(bool with_gold, int64_t pad_row,
-                                 int32_t* act_idx, int64_t* feats,
-                                 uint64_t* valid, uint64_t* gold) = 0;
-
   // Apply actions[k] to states[act_idx[k]] for k in [0, n).
   virtual void advance_active(const int32_t* act_idx, const int32_t* actions,
                               int64_t n) = 0;

@@ -405,49 +405,6 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     return Sa;
   }
 
-  int64_t pack_step_bits(bool with_gold, int64_t pad_row, int32_t* act_idx,
-                         int64_t* feats, uint64_t* valid_b,
-                         uint64_t* gold_b) override {
-    int64_t Sa = 0;
-    for (int64_t i = 0; i < n_docs; i++)
-      if (!final_state(i)) act_idx[Sa++] = (int32_t)i;
-    const int64_t A = n_actions();
-    const int64_t nAW = (A + 63) / 64;
-#ifdef _OPENMP
-#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
-#endif
-    for (int64_t k = 0; k < Sa; k++) {
-      float crow[256];
-      uint8_t v[256];
-      int32_t f32[13];
-      int64_t i = act_idx[k];
-      fill_features(i, f32);
-      int64_t* fo = feats + k * 13;
-      for (int q = 0; q < 13; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
-      uint64_t vb[4] = {0, 0, 0, 0};
-      if (with_gold) {
-        fill_costs(i, v, crow);
-        float cmin = KInvalid;
-        for (int64_t a = 0; a < A; a++)
-          if (v[a] && crow[(size_t)a] < cmin) cmin = crow[(size_t)a];
-        uint64_t gb[4] = {0, 0, 0, 0};
-        for (int64_t a = 0; a < A; a++) {
-          if (v[a]) {
-            vb[a >> 6] |= 1ull << (a & 63);
-            if (crow[(size_t)a] <= cmin + 1e-6f) gb[a >> 6] |= 1ull << (a & 63);
-          }
-        }
-        for (int64_t w = 0; w < nAW; w++) gold_b[k * nAW + w] = gb[w];
-      } else {
-        fill_valid(v, i);
-        for (int64_t a = 0; a < A; a++)
-          if (v[a]) vb[a >> 6] |= 1ull << (a & 63);
-      }
-      for (int64_t w = 0; w < nAW; w++) valid_b[k * nAW + w] = vb[w];
-    }
-    return Sa;
-  }
-
   inline void leave_buffer(int64_t d, int32_t tok) {
     // token `tok` moves out of the buffer: its gold head loses one
     // in-buffer child (the c_reduce counter)
@@ -729,45 +686,6 @@ struct BiluoBatch : public srx::StepBatchIface {
           for (int64_t a = 0; a < A; a++)
             g[a] = gold_valid ? (a == gcode ? 1 : 0) : v[a];
         }
-      }
-    }
-    return Sa;
-  }
-
-  int64_t pack_step_bits(bool with_gold, int64_t pad_row, int32_t* act_idx,
-                         int64_t* feats, uint64_t* valid_b,
-                         uint64_t* gold_b) override {
-    int64_t Sa = 0;
-    for (int64_t i = 0; i < n_docs; i++)
-      if (!final_state(i)) act_idx[Sa++] = (int32_t)i;
-    const int64_t A = n_actions();
-    const int64_t nAW = (A + 63) / 64;
-#ifdef _OPENMP
-#pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
-#endif
-    for (int64_t k = 0; k < Sa; k++) {
-      int64_t i = act_idx[k];
-      uint8_t v[256];
-      int32_t f32[6];
-      fill_features(i, f32);
-      int64_t* fo = feats + k * 6;
-      for (int q = 0; q < 6; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
-      fill_valid(v, i);
-      uint64_t vb[4] = {0, 0, 0, 0};
-      for (int64_t a = 0; a < A; a++)
-        if (v[a]) vb[a >> 6] |= 1ull << (a & 63);
-      for (int64_t w = 0; w < nAW; w++) valid_b[k * nAW + w] = vb[w];
-      if (with_gold) {
-        uint64_t gb[4] = {0, 0, 0, 0};
-        int32_t gcode = final_state(i) ? -2 : gold[off[i] + cur[i]];
-        if (gcode == -1) {
-          // missing: all-zero gold row (excluded from the loss)
-        } else if (gcode >= 0 && gcode < (int32_t)A && v[gcode]) {
-          gb[gcode >> 6] |= 1ull << (gcode & 63);
-        } else {
-          for (int64_t w = 0; w < nAW; w++) gb[w] = vb[w];
-        }
-        for (int64_t w = 0; w < nAW; w++) gold_b[k * nAW + w] = gb[w];
       }
     }
     return Sa;

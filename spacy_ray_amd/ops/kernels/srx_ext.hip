@@ -407,10 +407,6 @@ std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
   check_dev(scores);
   long N = scores.size(0);
   int A = (int)scores.size(1);
-  // gold/valid: int64 bitmask tensors [N, ceil(A/64)]
-  TORCH_CHECK(gold.scalar_type() == at::kLong && valid.scalar_type() == at::kLong,
-              "transition_ce masks must be int64 bitmasks");
-  TORCH_CHECK(gold.size(-1) == (A + 63) / 64, "gold bitmask width mismatch");
   auto dScores = at::empty_like(scores);
   auto loss = at::zeros({2}, scores.options().dtype(at::kFloat));
   auto colsum = at::zeros(
@@ -422,16 +418,14 @@ std::vector<at::Tensor> transition_ce(at::Tensor scores, at::Tensor gold,
       hipLaunchKernelGGL((transition_ce_kernel<scalar_t, true>),
                          dim3(grid_for(N * SRX_WAVE)), dim3(kBlock), 0, stream,
                          (const scalar_t*)scores.data_ptr(),
-                         (const uint64_t*)gold.data_ptr<int64_t>(),
-                         (const uint64_t*)valid.data_ptr<int64_t>(),
+                         gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
                          (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
                          colsum.data_ptr(), N, A);
     else
       hipLaunchKernelGGL((transition_ce_kernel<scalar_t, false>),
                          dim3(grid_for(N * SRX_WAVE)), dim3(kBlock), 0, stream,
                          (const scalar_t*)scores.data_ptr(),
-                         (const uint64_t*)gold.data_ptr<int64_t>(),
-                         (const uint64_t*)valid.data_ptr<int64_t>(),
+                         gold.data_ptr<uint8_t>(), valid.data_ptr<uint8_t>(),
                          (scalar_t*)dScores.data_ptr(), loss.data_ptr<float>(),
                          colsum.data_ptr(), N, A);
   });

@@ -62,12 +62,10 @@ __global__ void softmax_ce_kernel(const T* __restrict__ scores,
 // the backward skips a full [SS, A] column-reduce pass.
 // DET=true: colsum accumulates in int64 fixed-point (it feeds dUpperB; the
 // loss scalar stays a float atomic — display-only, never differentiated).
-// gold/valid are BITMASKS (ceil(A/64) uint64 words per row, bit a =
-// action a) — the layout the C++ step loop's arenas use.
 template <typename T, bool DET = false>
 __global__ void transition_ce_kernel(const T* __restrict__ scores,
-                                     const uint64_t* __restrict__ gold,
-                                     const uint64_t* __restrict__ valid,
+                                     const uint8_t* __restrict__ gold,
+                                     const uint8_t* __restrict__ valid,
                                      T* __restrict__ dScores,
                                      float* __restrict__ loss_out,
                                      void* __restrict__ colsum_out,
@@ -76,15 +74,14 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
   const int ncols = (A + SRX_WAVE - 1) / SRX_WAVE;
-  const int nAW = (A + 63) >> 6;
   float loss_acc = 0.f;
   float count_acc = 0.f;
   float col_acc[4];  // A <= 256
   for (int c = 0; c < ncols; c++) col_acc[c] = 0.f;
   for (long n = wave; n < N; n += nwaves) {
     const T* row = scores + n * (long)A;
-    const uint64_t* grow = gold + n * (long)nAW;
-    const uint64_t* vrow = valid + n * (long)nAW;
+    const uint8_t* grow = gold + n * (long)A;
+    const uint8_t* vrow = valid + n * (long)A;
     T* drow = dScores + n * (long)A;
     // load the row + masks ONCE into registers (A <= 256 -> 4 per lane);
     // the 3 logical passes (max, sumexp, dScores) reuse them
@@ -95,8 +92,8 @@ __global__ void transition_ce_kernel(const T* __restrict__ scores,
     for (int c = 0; c < ncols; c++) {
       int a = lane + c * SRX_WAVE;
       bool in = a < A;
-      g[c] = in && ((grow[a >> 6] >> (a & 63)) & 1);
-      vv[c] = in && ((vrow[a >> 6] >> (a & 63)) & 1);
+      g[c] = in && grow[a];
+      vv[c] = in && vrow[a];
       x[c] = vv[c] ? Elem<T>::ld(row + a) : -1e38f;
       if (g[c]) cnt += 1.f;
       if (vv[c]) m = fmaxf(m, x[c]);

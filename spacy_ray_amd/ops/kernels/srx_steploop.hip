@@ -87,17 +87,14 @@ PinnedPool& pool() {
 //      (written to arena + this wave's LDS slot)
 //   2. scores[a] = upperB[a] + dot(hidden, upperW[a,:]) from LDS broadcast
 //   3. masked argmax (sel_mask first, valid fallback) -> actions[s]
-// sel_mask/valid are BITMASKS: ceil(A/64) uint64 words per state (bit a =
-// action a) — the CPU pack writes A/8 mask bytes instead of A per state.
 template <typename T>
 __global__ void fused_step_all_kernel(
     const T* __restrict__ pre, const int64_t* __restrict__ feats,
     const T* __restrict__ lowerB, const T* __restrict__ upperW,
-    const T* __restrict__ upperB, const uint64_t* __restrict__ sel_mask,
-    const uint64_t* __restrict__ valid, T* __restrict__ hidden_out,
+    const T* __restrict__ upperB, const uint8_t* __restrict__ sel_mask,
+    const uint8_t* __restrict__ valid, T* __restrict__ hidden_out,
     uint8_t* __restrict__ which_out, T* __restrict__ scores_out,
     int32_t* __restrict__ actions_out, long S, int nF, int H, int A) {
-  const int nAW = (A + 63) >> 6;
   extern __shared__ char smem[];
   T* Wlds = (T*)smem;                       // [H][A] transposed
   float* Blds = (float*)(Wlds + (size_t)H * A);  // [A]
@@ -138,17 +135,15 @@ __global__ void fused_step_all_kernel(
     // phase 2 + 3: scores and masked argmax
     float bg = -1e38f, bv = -1e38f;
     int ig = INT32_MAX, iv = INT32_MAX;
-    const uint64_t* grow = sel_mask + s * (long)nAW;
-    const uint64_t* vrow = valid + s * (long)nAW;
+    const uint8_t* grow = sel_mask + s * (long)A;
+    const uint8_t* vrow = valid + s * (long)A;
     for (int a = lane; a < A; a += SRX_WAVE) {
       float acc = Blds[a];
       for (int h = 0; h < H; h++)
         acc += my_hid[h] * Elem<T>::ld(Wlds + (size_t)h * A + a);
       Elem<T>::st(scores_out + s * (long)A + a, acc);
-      bool gbit = (grow[a >> 6] >> (a & 63)) & 1;
-      bool vbit = (vrow[a >> 6] >> (a & 63)) & 1;
-      if (gbit && (acc > bg || (acc == bg && a < ig))) { bg = acc; ig = a; }
-      if (vbit && (acc > bv || (acc == bv && a < iv))) { bv = acc; iv = a; }
+      if (grow[a] && (acc > bg || (acc == bg && a < ig))) { bg = acc; ig = a; }
+      if (vrow[a] && (acc > bv || (acc == bv && a < iv))) { bv = acc; iv = a; }
     }
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) {
@@ -173,7 +168,7 @@ constexpr int kBlockThreads = 256;  // 4 waves
 template <typename T>
 void launch_fused_step(const void* pre, const int64_t* feats, const void* lowerB,
                        const void* upperW, const void* upperB,
-                       const uint64_t* sel, const uint64_t* valid, void* hidden,
+                       const uint8_t* sel, const uint8_t* valid, void* hidden,
                        uint8_t* which, void* scores, int32_t* actions, long S,
                        int nF, int H, int A, hipStream_t stream) {
   size_t lds = (size_t)H * A * sizeof(T) + A * sizeof(float) +
@@ -205,10 +200,9 @@ struct Unit {
   // pinned staging
   int32_t* act_idx_h = nullptr;
   int64_t* feats_h = nullptr;
-  uint64_t* valid_h = nullptr;
-  uint64_t* gold_h = nullptr;
+  uint8_t* valid_h = nullptr;
+  uint8_t* gold_h = nullptr;
   int32_t* actions_h = nullptr;
-  int nAW = 1;
   size_t staging_bytes = 0;
   void* staging = nullptr;
   hipEvent_t ev = nullptr;
@@ -285,28 +279,25 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
     TORCH_CHECK((size_t)u.H * u.A * u.pre.element_size() <= 96 * 1024,
                 "upper layer too large for LDS staging (H*A)");
     u.cap = u.train ? u.b->max_transitions() + 1 : u.nst;
-    u.nAW = (u.A + 63) / 64;
     auto opt = u.pre.options();
     auto optb = opt.dtype(at::kByte);
     u.feats_a = at::empty({u.cap, (long)u.nF}, opt.dtype(at::kLong));
-    // valid/gold are BITMASK arenas: ceil(A/64) uint64 words per state
-    u.valid_a = at::empty({u.cap, (long)u.nAW}, opt.dtype(at::kLong));
-    u.gold_a = u.train ? at::empty({u.cap, (long)u.nAW}, opt.dtype(at::kLong))
-                       : u.valid_a;
+    u.valid_a = at::empty({u.cap, (long)u.A}, optb);
+    u.gold_a = u.train ? at::empty({u.cap, (long)u.A}, optb) : u.valid_a;
     u.hidden_a = at::empty({u.cap, (long)u.H}, opt);
     u.which_a = at::empty({u.cap, (long)u.H}, optb);
     u.scores_a = at::empty({u.cap, (long)u.A}, opt);
     u.actions_d = at::empty({u.nst}, opt.dtype(at::kInt));
     // pinned staging layout: [act_idx i32][actions i32][feats i64][valid][gold]
-    size_t bytes = (size_t)u.nst * (4 + 4 + (size_t)u.nF * 8 + 2 * (size_t)u.nAW * 8) + 64;
+    size_t bytes = (size_t)u.nst * (4 + 4 + (size_t)u.nF * 8 + 2 * (size_t)u.A) + 64;
     u.staging_bytes = bytes;
     u.staging = pool().get(bytes);
     char* p = (char*)u.staging;
     u.act_idx_h = (int32_t*)p;            p += (size_t)u.nst * 4;
     u.actions_h = (int32_t*)p;            p += (size_t)u.nst * 4;
     u.feats_h = (int64_t*)p;              p += (size_t)u.nst * u.nF * 8;
-    u.valid_h = (uint64_t*)p;             p += (size_t)u.nst * u.nAW * 8;
-    u.gold_h = (uint64_t*)p;
+    u.valid_h = (uint8_t*)p;              p += (size_t)u.nst * u.A;
+    u.gold_h = (uint8_t*)p;
     hipEventCreateWithFlags(&u.ev, hipEventDisableTiming);
     u.stream = c10::cuda::getStreamFromPool(false, u.pre.get_device());
     u.hs = u.stream.stream();
@@ -346,8 +337,8 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
         u.pending = false;
       }
       if (loop_stats) t0 = now_ms();
-      long Sa = u.b->pack_step_bits(u.train, u.T, u.act_idx_h, u.feats_h,
-                                    u.valid_h, u.gold_h);
+      long Sa = u.b->pack_step(u.train, u.T, u.act_idx_h, u.feats_h, u.valid_h,
+                               u.gold_h);
       if (loop_stats) ls.pack_ms += now_ms() - t0;
       if (Sa == 0) {
         u.done = true;
@@ -359,31 +350,31 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
                   Sa, " > ", u.cap, ")");
       const size_t es = u.pre.element_size();
       char* feats_d = (char*)u.feats_a.data_ptr() + (size_t)off * u.nF * 8;
-      char* valid_d = (char*)u.valid_a.data_ptr() + (size_t)off * u.nAW * 8;
-      char* gold_d = (char*)u.gold_a.data_ptr() + (size_t)off * u.nAW * 8;
+      char* valid_d = (char*)u.valid_a.data_ptr() + (size_t)off * u.A;
+      char* gold_d = (char*)u.gold_a.data_ptr() + (size_t)off * u.A;
       char* hidden_d = (char*)u.hidden_a.data_ptr() + (size_t)off * u.H * es;
       uint8_t* which_d = (uint8_t*)u.which_a.data_ptr() + (size_t)off * u.H;
       char* scores_d = (char*)u.scores_a.data_ptr() + (size_t)off * u.A * es;
       int32_t* actions_d = (int32_t*)u.actions_d.data_ptr();
       hipMemcpyAsync(feats_d, u.feats_h, (size_t)Sa * u.nF * 8,
                      hipMemcpyHostToDevice, u.hs);
-      hipMemcpyAsync(valid_d, u.valid_h, (size_t)Sa * u.nAW * 8,
+      hipMemcpyAsync(valid_d, u.valid_h, (size_t)Sa * u.A,
                      hipMemcpyHostToDevice, u.hs);
       if (u.train)
-        hipMemcpyAsync(gold_d, u.gold_h, (size_t)Sa * u.nAW * 8,
+        hipMemcpyAsync(gold_d, u.gold_h, (size_t)Sa * u.A,
                        hipMemcpyHostToDevice, u.hs);
-      const uint64_t* sel = (const uint64_t*)(u.train ? gold_d : valid_d);
+      const uint8_t* sel = u.train ? (const uint8_t*)gold_d : (const uint8_t*)valid_d;
       if (u.pre.scalar_type() == at::kBFloat16) {
         launch_fused_step<bf16_t>(u.pre.data_ptr(), (const int64_t*)feats_d,
                                   u.lowerB.data_ptr(), u.upperW.data_ptr(),
                                   u.upperB.data_ptr(), sel,
-                                  (const uint64_t*)valid_d, hidden_d, which_d,
+                                  (const uint8_t*)valid_d, hidden_d, which_d,
                                   scores_d, actions_d, Sa, u.nF, u.H, u.A, u.hs);
       } else {
         launch_fused_step<float>(u.pre.data_ptr(), (const int64_t*)feats_d,
                                  u.lowerB.data_ptr(), u.upperW.data_ptr(),
                                  u.upperB.data_ptr(), sel,
-                                 (const uint64_t*)valid_d, hidden_d, which_d,
+                                 (const uint8_t*)valid_d, hidden_d, which_d,
                                  scores_d, actions_d, Sa, u.nF, u.H, u.A, u.hs);
       }
       hipMemcpyAsync(u.actions_h, actions_d, (size_t)Sa * 4,

@@ -177,17 +177,7 @@ def test_transition_ce_kernel_matches_torch():
     valid[:, 0] = 1  # every row has a valid action
     gold = ((torch.rand(SS, A, device="cuda") < 0.2).to(torch.uint8) & valid)
     gold[::7] = 0  # some unsupervised rows
-
-    def pack_bits(mask):  # [N, A] u8 -> [N, ceil(A/64)] int64 bitmask
-        N, Aa = mask.shape
-        nAW = (Aa + 63) // 64
-        out = torch.zeros(N, nAW, dtype=torch.int64, device=mask.device)
-        for a in range(Aa):
-            out[:, a // 64] |= mask[:, a].long() << (a % 64)
-        return out
-
-    loss_count, dScores, colsum = hip.transition_ce(
-        scores, pack_bits(gold), pack_bits(valid))
+    loss_count, dScores, colsum = hip.transition_ce(scores, gold, valid)
     # torch reference
     NEG_INF = -1e30
     g = gold > 0
