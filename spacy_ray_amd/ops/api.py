@@ -496,20 +496,18 @@ class _MWELayer(torch.autograd.Function):
             dropmask = None
         hip = hip_ext()
         dY = dY.contiguous()
-        dL = dY * dropmask if dropmask is not None else dY
-        # LN backward over the maxout output
-        dM, dg, db = hip.layernorm_bwd(dL.contiguous(), Mout, g, mu, rstd,
-                                       deterministic())
-        # maxout scatter: [T, W] -> [T, 3, W] pieces-major -> [T, 3W]
-        dPre = hip.maxout_bwd(dM, which, 3).reshape(dY.shape[0], -1)
-        dbias = dPre.sum(dim=0)
+        # stage 1 in ONE kernel: dropout mask x LN backward x maxout scatter
+        # + the dg/db/dbias column sums (was 3 kernels + 2 reduce passes)
+        dPre, dg32, db32, dbias32 = hip.mwe_bwd_stage1(
+            dY, dropmask, Mout, g, mu, rstd, which, deterministic())
         # GEMM backwards (pre = X3 @ W^T)
         X3 = hip.seq2col_fwd(X, starts, ends)
         dW = mm_dw_chunked(dPre, X3)
         dX3 = dPre.mm(weight)
-        dX = hip.seq2col_bwd(dX3.contiguous(), starts, ends)
-        dX += dY  # residual
-        return dX, dW, dbias.to(dY.dtype), dg, db, None, None, None, None
+        # seq2col backward with the residual dY folded in
+        dX = hip.seq2col_bwd(dX3.contiguous(), starts, ends, dY)
+        return (dX, dW, dbias32.to(dY.dtype), dg32.to(dY.dtype),
+                db32.to(dY.dtype), None, None, None, None)
 
 
 def mwe_layer(X, weight, bias, g, b, starts, ends, dropmask=None, eps: float = 1e-5):

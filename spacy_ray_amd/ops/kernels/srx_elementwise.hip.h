@@ -38,10 +38,13 @@ __global__ void seq2col_fwd_kernel(const T* __restrict__ X, T* __restrict__ Y,
 }
 
 // dX[t,w] = dY[t, W+w] + (next's prev-slot) + (prev's next-slot)
+//           [+ residual[t,w] when given — fuses the encoder block's
+//            `dX += dY_upstream` residual add]
 template <typename T, int V>
 __global__ void seq2col_bwd_kernel(const T* __restrict__ dY, T* __restrict__ dX,
                                    const uint8_t* __restrict__ is_start,
                                    const uint8_t* __restrict__ is_end,
+                                   const T* __restrict__ residual,
                                    long nT, int W) {
   const long chunks = W / V;
   const long total = nT * chunks;
@@ -57,6 +60,7 @@ __global__ void seq2col_bwd_kernel(const T* __restrict__ dY, T* __restrict__ dX,
         acc += Elem<T>::ld(dY + (t + 1) * row + 0 + w + k);
       if (t > 0 && !is_end[t - 1])
         acc += Elem<T>::ld(dY + (t - 1) * row + 2 * W + w + k);
+      if (residual) acc += Elem<T>::ld(residual + t * (long)W + w + k);
       Elem<T>::st(dX + t * (long)W + w + k, acc);
     }
   }

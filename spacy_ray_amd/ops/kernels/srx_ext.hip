@@ -71,7 +71,8 @@ at::Tensor seq2col_fwd(at::Tensor X, at::Tensor starts, at::Tensor ends) {
   return Y;
 }
 
-at::Tensor seq2col_bwd(at::Tensor dY, at::Tensor starts, at::Tensor ends) {
+at::Tensor seq2col_bwd(at::Tensor dY, at::Tensor starts, at::Tensor ends,
+                       c10::optional<at::Tensor> residual = c10::nullopt) {
   check_dev(dY);
   long nT = dY.size(0);
   int W3 = (int)dY.size(1);
@@ -80,17 +81,21 @@ at::Tensor seq2col_bwd(at::Tensor dY, at::Tensor starts, at::Tensor ends) {
   if (nT == 0) return dX;
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(dY.scalar_type(), {
+    const scalar_t* res =
+        residual ? (const scalar_t*)residual->data_ptr() : nullptr;
     const int V = (W % kVec == 0) ? kVec : 1;
     if (V == kVec)
       hipLaunchKernelGGL((seq2col_bwd_kernel<scalar_t, kVec>),
                          dim3(grid_for(nT * (W / kVec))), dim3(kBlock), 0, stream,
                          (const scalar_t*)dY.data_ptr(), (scalar_t*)dX.data_ptr(),
-                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(), nT, W);
+                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(),
+                         res, nT, W);
     else
       hipLaunchKernelGGL((seq2col_bwd_kernel<scalar_t, 1>),
                          dim3(grid_for(nT * (long)W)), dim3(kBlock), 0, stream,
                          (const scalar_t*)dY.data_ptr(), (scalar_t*)dX.data_ptr(),
-                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(), nT, W);
+                         starts.data_ptr<uint8_t>(), ends.data_ptr<uint8_t>(),
+                         res, nT, W);
   });
   return dX;
 }
@@ -581,6 +586,58 @@ at::Tensor reduce_max_bwd(at::Tensor dY, at::Tensor argmax, int64_t Ttot) {
   return dX;
 }
 
+// ----------------------------------------- fused MWE backward stage 1
+// (dropmask x dY -> LN bwd -> maxout scatter) + dg/db/dbias column sums in
+// ONE kernel; see mwe_bwd_stage1_kernel.
+std::vector<at::Tensor> mwe_bwd_stage1(at::Tensor dY,
+                                       c10::optional<at::Tensor> dropmask,
+                                       at::Tensor Mout, at::Tensor g,
+                                       at::Tensor mu, at::Tensor rstd,
+                                       at::Tensor which,
+                                       bool deterministic = false) {
+  check_dev(dY);
+  long N = dY.size(0);
+  int W = (int)dY.size(1);
+  TORCH_CHECK(W <= 256, "mwe_bwd_stage1 W <= 256");
+  auto dPre = at::empty({N, 3L * W}, dY.options());
+  auto acc_dt = deterministic ? at::kLong : at::kFloat;
+  auto dg32 = at::zeros({W}, dY.options().dtype(acc_dt));
+  auto db32 = at::zeros({W}, dY.options().dtype(acc_dt));
+  auto dbias32 = at::zeros({3L * W}, dY.options().dtype(acc_dt));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (N > 0) {
+    int grid = (int)std::min<long>((N + 3) / 4, 4096);
+    DISPATCH_F(dY.scalar_type(), {
+      const scalar_t* mask =
+          dropmask ? (const scalar_t*)dropmask->data_ptr() : nullptr;
+      if (deterministic)
+        hipLaunchKernelGGL((mwe_bwd_stage1_kernel<scalar_t, true>), dim3(grid),
+                           dim3(kBlock), 0, stream,
+                           (const scalar_t*)dY.data_ptr(), mask,
+                           (const scalar_t*)Mout.data_ptr(),
+                           (const scalar_t*)g.data_ptr(), mu.data_ptr<float>(),
+                           rstd.data_ptr<float>(), which.data_ptr<uint8_t>(),
+                           (scalar_t*)dPre.data_ptr(), dg32.data_ptr(),
+                           db32.data_ptr(), dbias32.data_ptr(), N, W);
+      else
+        hipLaunchKernelGGL((mwe_bwd_stage1_kernel<scalar_t, false>), dim3(grid),
+                           dim3(kBlock), 0, stream,
+                           (const scalar_t*)dY.data_ptr(), mask,
+                           (const scalar_t*)Mout.data_ptr(),
+                           (const scalar_t*)g.data_ptr(), mu.data_ptr<float>(),
+                           rstd.data_ptr<float>(), which.data_ptr<uint8_t>(),
+                           (scalar_t*)dPre.data_ptr(), dg32.data_ptr(),
+                           db32.data_ptr(), dbias32.data_ptr(), N, W);
+    });
+  }
+  if (deterministic) {
+    double s = 1.0 / 16777216.0;
+    return {dPre, (dg32.to(at::kFloat) * s), (db32.to(at::kFloat) * s),
+            (dbias32.to(at::kFloat) * s)};
+  }
+  return {dPre, dg32, db32, dbias32};
+}
+
 // ------------------------------------------------ fused MWE layer (MFMA)
 template <int W>
 void launch_mwe(const at::Tensor& X, const at::Tensor& Wt, const at::Tensor& bias,
@@ -736,7 +793,11 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seq2col_fwd", &seq2col_fwd);
-  m.def("seq2col_bwd", &seq2col_bwd);
+  m.def("seq2col_bwd", &seq2col_bwd, py::arg("dY"), py::arg("starts"),
+        py::arg("ends"), py::arg("residual") = py::none());
+  m.def("mwe_bwd_stage1", &mwe_bwd_stage1, py::arg("dY"), py::arg("dropmask"),
+        py::arg("Mout"), py::arg("g"), py::arg("mu"), py::arg("rstd"),
+        py::arg("which"), py::arg("deterministic") = false);
   m.def("maxout_fwd", &maxout_fwd);
   m.def("maxout_bwd", &maxout_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
