@@ -194,6 +194,8 @@ def main() -> None:
     model_desc = {
         "en_core_cnn.cfg": ("words/sec (whole node) en_core CNN tagger+parser+NER train",
                             "en_core_web_cnn (MultiHashEmbed+MaxoutWindowEncoder w96d4 + tagger + parser + ner)"),
+        "en_core_trf_hash.cfg": ("words/sec (whole node) en_core_web_trf (roberta-base, hash subwords) train",
+                                 "en_core_web_trf shape (roberta-base random-init, 1 subword/word)"),
         "en_core_trf.cfg": ("words/sec (whole node) en_core_web_trf (roberta-base) train",
                             "en_core_web_trf (roberta-base random-init + tagger + parser + ner)"),
         "xx_multilingual.cfg": ("words/sec (whole node) xx multilingual UD train",

@@ -80,11 +80,7 @@ class MaxoutWindowEncoder(nn.Module):
                     # whole block in ONE MFMA kernel (incl. dropout+residual)
                     mask = None
                     if drop and self.training:
-                        keep = 1.0 - drop
-                        mask = (
-                            (torch.rand_like(X, dtype=torch.float32) < keep)
-                            .to(X.dtype) / keep
-                        )
+                        mask = ops.dropout_mask_like(X, drop)
                     X = ops.mwe_layer(
                         X, block.weight, block.bias,
                         block.norm.weight, block.norm.bias, starts, ends,

@@ -525,6 +525,23 @@ def mwe_layer_available(X: torch.Tensor, width: int, pieces: int) -> bool:
     )
 
 
+# ------------------------------------------------------------ dropout mask
+_dropout_calls = 0
+
+
+def dropout_mask_like(X: torch.Tensor, p: float) -> torch.Tensor:
+    """Scaled keep-mask in X's dtype via the Philox kernel (SURVEY §2.5
+    dropout_mask).  Reproducible under torch.manual_seed: the Philox
+    (seed, offset) derive from torch's initial seed + a call counter."""
+    global _dropout_calls
+    _dropout_calls += 1
+    if X.is_cuda and _want_hip(X):
+        return hip_ext().dropout_mask(X, p, torch.initial_seed() & 0x7FFFFFFF,
+                                      _dropout_calls)
+    keep = 1.0 - p
+    return (torch.rand_like(X, dtype=torch.float32) < keep).to(X.dtype) / keep
+
+
 # ----------------------------------------------------------- softmax + CE
 class _SoftmaxCE(torch.autograd.Function):
     """Fused softmax + cross-entropy (SURVEY.md §2.5 softmax_ce_fwd/bwd):

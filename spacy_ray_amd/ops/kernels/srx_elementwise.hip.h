@@ -221,6 +221,32 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dY,
   }
 }
 
+// -------------------------------------------------------- dropout mask
+// Philox4x32-10 dropout mask (SURVEY §2.5 dropout_mask: hiprand device
+// API): one kernel writes the scaled keep-mask directly in the compute
+// dtype — replaces the torch chain rand_like(fp32) -> compare -> scale ->
+// cast (3 kernels + a full fp32 tensor) used for the fused-MWE mask.
+// Reproducible: (seed, offset) are the Philox counter; the python side
+// derives them from torch's manual seed + a call counter.
+#include <hiprand/hiprand_kernel.h>
+
+template <typename T>
+__global__ void dropout_mask_kernel(T* __restrict__ out, long n, float keep,
+                                    float scale, unsigned long long seed,
+                                    unsigned long long offset) {
+  const long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  hiprandStatePhilox4_32_10_t st;
+  hiprand_init(seed, (unsigned long long)tid, offset, &st);
+  for (long j = tid * 4; j < n; j += stride * 4) {
+    float4 r = hiprand_uniform4(&st);
+    float v[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int k = 0; k < 4; k++)
+      if (j + k < n) Elem<T>::st(out + j + k, v[k] < keep ? scale : 0.f);
+  }
+}
+
 // ---------------------------------------------------------- fused Adam
 // One launch for the whole sharded step (SURVEY.md §2.5 fused_adam_sharded):
 // grad (model dtype) + fp32 master/m/v -> updated state + model-dtype param.

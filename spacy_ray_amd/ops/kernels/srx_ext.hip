@@ -497,6 +497,27 @@ std::vector<at::Tensor> dpre_scatter(at::Tensor dSummed, at::Tensor feats,
   return {finish(dBias32), finish(dPad32)};
 }
 
+// ----------------------------------------------------- dropout mask
+at::Tensor dropout_mask(at::Tensor like, double p, int64_t seed, int64_t offset) {
+  check_dev(like);
+  auto out = at::empty_like(like);
+  long n = out.numel();
+  if (n == 0 || p <= 0.0) {
+    out.fill_(1.0);
+    return out;
+  }
+  float keep = 1.0f - (float)p;
+  float scale = 1.0f / keep;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(like.scalar_type(), {
+    hipLaunchKernelGGL((dropout_mask_kernel<scalar_t>), dim3(grid_for(n, 4)),
+                       dim3(kBlock), 0, stream, (scalar_t*)out.data_ptr(), n,
+                       keep, scale, (unsigned long long)seed,
+                       (unsigned long long)offset);
+  });
+  return out;
+}
+
 // ------------------------------------------------------- softmax + CE
 // returns (loss_and_count fp32 [2], dScores) — dScores = softmax - onehot
 // (unnormalized; the python wrapper divides by the valid count).
@@ -812,6 +833,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("action_select", &action_select);
   m.def("seg_scatter_add", &seg_scatter_add);
   m.def("adam_step", &adam_step);
+  m.def("dropout_mask", &dropout_mask);
   m.def("softmax_ce", &softmax_ce);
   m.def("reduce_ragged", &reduce_ragged);
   m.def("reduce_ragged_bwd", &reduce_ragged_bwd);
