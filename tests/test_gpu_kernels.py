@@ -273,3 +273,25 @@ def test_adam_step_gpu(dtype):
     assert torch.allclose(m, mr, atol=1e-6)
     assert torch.allclose(v, vr, atol=1e-6)
     assert torch.allclose(param_out.float(), ref_p.float().to(dtype).float(), atol=1e-2)
+
+
+@need_gpu
+def test_dropout_mask_philox_statistics_and_reproducibility():
+    """Philox dropout mask (SURVEY §2.5 dropout_mask): keep fraction ~= 1-p,
+    kept entries scaled by 1/(1-p), same (seed, offset) -> identical mask."""
+    from spacy_ray_amd.ops.api import hip_ext
+
+    hip = hip_ext()
+    X = torch.empty(1 << 20, device="cuda", dtype=torch.bfloat16)
+    p = 0.3
+    m1 = hip.dropout_mask(X, p, 1234, 7)
+    m2 = hip.dropout_mask(X, p, 1234, 7)
+    m3 = hip.dropout_mask(X, p, 1234, 8)
+    assert torch.equal(m1, m2)
+    assert not torch.equal(m1, m3)
+    keep_frac = float((m1 > 0).float().mean())
+    assert abs(keep_frac - (1 - p)) < 0.01, keep_frac
+    kept = m1[m1 > 0].float()
+    assert torch.allclose(kept, torch.full_like(kept, 1 / (1 - p)), atol=1e-2)
+    # the mean of the mask is ~1 (unbiased dropout scaling)
+    assert abs(float(m1.float().mean()) - 1.0) < 0.02
