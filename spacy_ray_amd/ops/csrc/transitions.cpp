@@ -262,10 +262,8 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     if (has_s0 && (s0_has_head || !has_buf)) v[1] = 1;
     bool la_ok = has_s0 && has_buf && !s0_has_head;
     bool ra_ok = has_s0 && has_buf;
-    for (int32_t l = 0; l < n_labels; l++) {
-      v[2 + l] = la_ok ? 1 : 0;
-      v[2 + n_labels + l] = ra_ok ? 1 : 0;
-    }
+    if (la_ok) std::memset(v + 2, 1, (size_t)n_labels);
+    if (ra_ok) std::memset(v + 2 + n_labels, 1, (size_t)n_labels);
     if (use_break)
       v[2 + 2 * n_labels] =
           (has_buf && buf[d] > 0 && !sent_out[off[d] + buf[d]]) ? 1 : 0;
@@ -379,30 +377,115 @@ struct ArcEagerBatch : public srx::StepBatchIface {
     for (int64_t i = 0; i < n_docs; i++)
       if (!final_state(i)) act_idx[Sa++] = (int32_t)i;
     const int64_t A = n_actions();
+    const int32_t L = n_labels;
 #ifdef _OPENMP
 #pragma omp parallel for num_threads(srx_nthreads()) schedule(static) if (Sa > 512)
 #endif
     for (int64_t k = 0; k < Sa; k++) {
-      float crow[256];
       int32_t f32[13];
       int64_t i = act_idx[k];
       fill_features(i, f32);
       int64_t* fo = feats + k * 13;
       for (int q = 0; q < 13; q++) fo[q] = f32[q] < 0 ? pad_row : (int64_t)f32[q];
       uint8_t* v = valid_a + k * A;
-      if (with_gold) {
-        fill_costs(i, v, crow);
-        float cmin = KInvalid;
-        for (int64_t a = 0; a < A; a++)
-          if (v[a] && crow[(size_t)a] < cmin) cmin = crow[(size_t)a];
-        uint8_t* g = gold_a + k * A;
-        for (int64_t a = 0; a < A; a++)
-          g[a] = (v[a] && crow[(size_t)a] <= cmin + 1e-6f) ? 1 : 0;
-      } else {
-        fill_valid(v, i);
-      }
+      fill_valid(v, i);
+      if (!with_gold) continue;
+      // Min-cost gold mask WITHOUT materializing the per-action cost row:
+      // all labeled LEFT-ARCs share cost c_la except the gold label (+1
+      // for a wrong label only when the arc itself matches gold), same
+      // for RIGHT-ARC — so 5 scalars + the gold labels determine the mask
+      // and the label ranges become memsets (the 82-float crow + two
+      // 82-wide scans were ~half the pack cost at 1M words).
+      uint8_t* g = gold_a + k * A;
+      scalar_costs(i, v, g, A, L);
     }
     return Sa;
+  }
+
+  // Shared scalar-cost core for pack_step: computes c_shift/c_reduce/
+  // c_la/c_ra(/c_break) exactly as fill_costs does, then writes the
+  // min-cost mask g[0..A) directly.
+  inline void scalar_costs(int64_t d, const uint8_t* v, uint8_t* g,
+                           int64_t A, int32_t L) const {
+    const int64_t o = off[d];
+    const int32_t* gh = gold_head.data() + o;
+    const int32_t* gl = gold_label.data() + o;
+    int32_t b = buf[d] < len[d] ? buf[d] : -1;
+    int32_t v0 = s0(d);
+    float c_shift = 0, c_reduce = 0, c_la = 0, c_ra = 0;
+    if (b >= 0) {
+      bool ghb_on_stack = gh[b] >= 0 && gh[b] < len[d] && on_stack[o + gh[b]];
+      if (ghb_on_stack) c_shift += 1;
+      float stack_kids_of_b = 0;
+      for (int32_t k = kids_off[o + b]; k < kids_off[o + b + 1]; k++) {
+        int32_t c = kids[k];
+        if (on_stack[o + c] && head[o + c] == -1) stack_kids_of_b += 1;
+      }
+      c_shift += stack_kids_of_b;
+      if (v0 >= 0) {
+        c_reduce = (float)gold_kids_buf[o + v0];
+        c_la = c_reduce;
+        if (gh[v0] >= 0 && gh[v0] > b) c_la += 1;
+        if (gh[b] >= 0 && gh[b] != v0 && (ghb_on_stack || gh[b] > b)) c_ra += 1;
+        c_ra += stack_kids_of_b;
+      }
+    } else if (v0 >= 0) {
+      c_reduce = (float)gold_kids_buf[o + v0];
+    }
+    bool at_gold_break = use_break && !gold_sent.empty() && b > 0 &&
+                         gold_sent[o + b] == 1 && !sent_out[o + b];
+    if (at_gold_break) {
+      c_shift += 1;
+      c_ra += 1;
+    }
+    // gold label deltas: LA(l) costs c_la + (arc matches gold but l wrong);
+    // the +1 applies only when gh[v0] == b (resp. gh[b] == v0)
+    int32_t la_gold = (b >= 0 && v0 >= 0 && gh[v0] == b) ? gl[v0] : -1;
+    int32_t ra_gold = (b >= 0 && v0 >= 0 && gh[b] == v0) ? gl[b] : -1;
+    const int32_t bk = 2 + 2 * L;
+    float c_break = KInvalid;
+    if (use_break && v[bk]) {
+      c_break = at_gold_break ? 0.f : 1.f;
+      if (!gold_sent.empty()) {
+        const int32_t* stk = stack.data() + o;
+        for (int32_t k2 = 0; k2 < ssize[d]; k2++) {
+          int32_t s = stk[k2];
+          if (head[o + s] == -1 && gh[s] >= (b >= 0 ? b : len[d])) c_break += 1;
+          c_break += (float)gold_kids_buf[o + s];
+        }
+      }
+    }
+    float cmin = KInvalid;
+    if (v[0]) cmin = std::min(cmin, c_shift);
+    if (v[1]) cmin = std::min(cmin, c_reduce);
+    if (v[2]) cmin = std::min(cmin, c_la);          // min over LA labels = c_la
+    if (v[2 + L]) cmin = std::min(cmin, c_ra);      // min over RA labels = c_ra
+    if (use_break && v[bk]) cmin = std::min(cmin, c_break);
+    const float eps = 1e-6f;
+    std::memset(g, 0, (size_t)A);
+    if (v[0] && c_shift <= cmin + eps) g[0] = 1;
+    if (v[1] && c_reduce <= cmin + eps) g[1] = 1;
+    if (v[2]) {  // LA labels valid as a block (fill_valid sets all-or-none)
+      if (c_la <= cmin + eps) {
+        if (la_gold >= 0) {
+          if (c_la + 1 <= cmin + eps) std::memset(g + 2, 1, (size_t)L);
+          g[2 + la_gold] = 1;
+        } else {
+          std::memset(g + 2, 1, (size_t)L);
+        }
+      }
+    }
+    if (v[2 + L]) {
+      if (c_ra <= cmin + eps) {
+        if (ra_gold >= 0) {
+          if (c_ra + 1 <= cmin + eps) std::memset(g + 2 + L, 1, (size_t)L);
+          g[2 + L + ra_gold] = 1;
+        } else {
+          std::memset(g + 2 + L, 1, (size_t)L);
+        }
+      }
+    }
+    if (use_break && v[bk] && c_break <= cmin + eps) g[bk] = 1;
   }
 
   inline void leave_buffer(int64_t d, int32_t tok) {

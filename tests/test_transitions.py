@@ -247,3 +247,42 @@ def test_break_never_preferred_without_sentence_gold():
         assert act != BREAK  # cost 1 > some zero-cost action
         batch.advance(np.array([act], dtype=np.int32))
     assert batch.sent_starts().sum() == 0
+
+
+def test_pack_step_gold_mask_matches_bruteforce_costs():
+    """pack_step's scalar-cost min-cost mask must equal the mask derived
+    from the full costs() row at every state of random walks (with and
+    without BREAK)."""
+    import numpy as np
+    from spacy_ray_amd import _srx_cpu
+
+    rng = np.random.RandomState(11)
+    for trial in range(30):
+        n = int(rng.randint(2, 12))
+        L = int(rng.randint(1, 5))
+        use_break = bool(trial % 2)
+        heads = np.full(n, -1, dtype=np.int32)
+        for i in range(1, n):
+            heads[i] = rng.randint(0, i)
+        labels = rng.randint(0, L, size=n).astype(np.int32)
+        batch = _srx_cpu.ArcEagerBatch(np.array([n], dtype=np.int32), L, 0,
+                                       use_break)
+        batch.set_gold(heads, labels)
+        if use_break:
+            sents = (rng.rand(n) < 0.3).astype(np.int32)
+            sents[0] = 1
+            batch.set_sent_gold(sents)
+        for _ in range(4 * n):
+            if batch.is_final()[0]:
+                break
+            act_idx, feats, valid, gold = batch.step_arrays(True)
+            costs = batch.costs()[0]
+            vmask = batch.valid()[0]
+            cmin = costs[vmask > 0].min()
+            ref_gold = ((vmask > 0) & (costs <= cmin + 1e-6)).astype(np.uint8)
+            assert np.array_equal(valid[0], vmask), (trial, valid[0], vmask)
+            assert np.array_equal(gold[0], ref_gold), (
+                trial, gold[0], ref_gold, costs)
+            choices = np.nonzero(ref_gold)[0]
+            act = int(choices[rng.randint(len(choices))])
+            batch.advance(np.array([act], dtype=np.int32))
