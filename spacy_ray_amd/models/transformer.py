@@ -94,10 +94,32 @@ class _SrxEmbedding(nn.Embedding):
         return super().forward(ids)
 
 
+def _fused_output_forward(self, hidden_states, input_tensor):
+    """RobertaSelfOutput/RobertaOutput forward with the residual add fused
+    into our layernorm kernels (ops.add_layernorm): dense -> dropout ->
+    LN(hidden + input) loses one full elementwise pass each way."""
+    from spacy_ray_amd.ops import api as ops
+
+    hidden_states = self.dense(hidden_states)
+    hidden_states = self.dropout(hidden_states)
+    if hidden_states.is_cuda and hidden_states.shape[-1] <= 1024:
+        shape = hidden_states.shape
+        return ops.add_layernorm(
+            hidden_states.reshape(-1, shape[-1]).contiguous(),
+            input_tensor.reshape(-1, shape[-1]).contiguous(),
+            self.LayerNorm.weight, self.LayerNorm.bias, self.LayerNorm.eps,
+        ).view(shape)
+    return self.LayerNorm(hidden_states + input_tensor)
+
+
 def _srx_optimize_roberta(trf: nn.Module) -> None:
     """Swap LayerNorm/Embedding modules for kernel-backed drop-ins (same
-    attribute paths + param names: checkpoints unaffected)."""
+    attribute paths + param names: checkpoints unaffected), and fuse the
+    residual+LayerNorm in the attention/MLP output blocks."""
     for mod in trf.modules():
+        cls_name = type(mod).__name__
+        if cls_name in ("RobertaSelfOutput", "RobertaOutput"):
+            mod.forward = _fused_output_forward.__get__(mod)
         for name, child in list(mod.named_children()):
             if type(child) is nn.LayerNorm:
                 new = _SrxLayerNorm(child.normalized_shape, eps=child.eps)
@@ -271,7 +293,19 @@ class TransformerTok2Vec(nn.Module):
         eos_col = ns + 1
         input_ids[torch.arange(nW, device=device), eos_col] = EOS
         attn = (pos.unsqueeze(0) <= eos_col.unsqueeze(1)).long()
-        out = self.trf(input_ids=input_ids, attention_mask=attn).last_hidden_state
+        import os
+
+        if os.environ.get("SRX_SDPA") == "math" and input_ids.is_cuda:
+            # A/B knob: the math SDPA backward is plain GEMMs — aotriton's
+            # flash backward measured 5.8x its forward at these short
+            # windows (profiles/trf262k_kernel_stats.csv)
+            from torch.nn.attention import SDPBackend, sdpa_kernel
+
+            with sdpa_kernel([SDPBackend.MATH]):
+                out = self.trf(input_ids=input_ids,
+                               attention_mask=attn).last_hidden_state
+        else:
+            out = self.trf(input_ids=input_ids, attention_mask=attn).last_hidden_state
         # overlap- and subword-averaged scatter back to [T, width]
         acc = out.new_zeros(T, self.width)
         cnt = out.new_zeros(T, 1)

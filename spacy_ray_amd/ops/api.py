@@ -525,6 +525,50 @@ def mwe_layer_available(X: torch.Tensor, width: int, pieces: int) -> bool:
     )
 
 
+class _AddLayerNorm(torch.autograd.Function):
+    """LN(X + R) with the residual add fused into the layernorm kernels'
+    data passes (saves a full elementwise pass each way on the roberta
+    blocks' `LayerNorm(hidden + input)` pattern).  The sum S is
+    materialized once for the backward (the LN backward needs its input);
+    dX = dR = LN-input grad."""
+
+    @staticmethod
+    def forward(ctx, X, R, g, b, eps: float):
+        S = X + R
+        if _want_hip(S):
+            Y, mu, rstd = hip_ext().layernorm_fwd(S.contiguous(), g, b, eps)
+            ctx.save_for_backward(S, g, mu, rstd)
+            ctx.eps = eps
+            return Y
+        mu = S.mean(dim=-1, keepdim=True)
+        var = S.var(dim=-1, unbiased=False, keepdim=True)
+        rstd = torch.rsqrt(var + eps)
+        ctx.save_for_backward(S, g, mu, rstd)
+        return (S - mu) * rstd * g + b
+
+    @staticmethod
+    def backward(ctx, dY):
+        S, g, mu, rstd = ctx.saved_tensors
+        if _want_hip(dY):
+            dS, dg, db = hip_ext().layernorm_bwd(dY.contiguous(), S, g, mu, rstd,
+                                                 deterministic())
+            return dS, dS, dg, db, None
+        xhat = (S - mu) * rstd
+        dg = (dY * xhat).sum(dim=tuple(range(dY.dim() - 1)))
+        db = dY.sum(dim=tuple(range(dY.dim() - 1)))
+        dxhat = dY * g
+        dS = rstd * (
+            dxhat
+            - dxhat.mean(dim=-1, keepdim=True)
+            - xhat * (dxhat * xhat).mean(dim=-1, keepdim=True)
+        )
+        return dS, dS, dg, db, None
+
+
+def add_layernorm(X, R, g, b, eps: float = 1e-5):
+    return _AddLayerNorm.apply(X, R, g, b, eps)
+
+
 # ------------------------------------------------------------ dropout mask
 _dropout_calls = 0
 
