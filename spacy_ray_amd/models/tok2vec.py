@@ -40,10 +40,7 @@ class MultiHashEmbed(nn.Module):
         from spacy_ray_amd.utils import timing
 
         with timing.phase("t2v/embed_tables"):
-            outs = []
-            for i, table in enumerate(self.tables):
-                outs.append(ops.hashembed(table, batch.attr_ids[:, i], self.seeds[i]))
-            X = torch.cat(outs, dim=1)
+            X = ops.multi_hashembed(batch.attr_ids, self.seeds, list(self.tables))
         with timing.phase("t2v/mixer_gemm"):
             Y = ops.linear_cdw(X, self.mixer.weight, self.mixer.bias)
         with timing.phase("t2v/mixer_maxout"):

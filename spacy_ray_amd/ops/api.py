@@ -178,6 +178,61 @@ def hashembed(table: torch.Tensor, ids: torch.Tensor, seed: int) -> torch.Tensor
     return _HashEmbed.apply(table, ids, seed)
 
 
+class _MultiHashEmbed(torch.autograd.Function):
+    """All 4 attr tables in one autograd node producing the concatenated
+    [T, 4W] matrix: forward writes each table's output straight into its
+    column block (no torch.cat — a 0.77 GB copy per step at 1M words);
+    backward runs the sorted segmented reduction per table on COLUMN VIEWS
+    of dX (strided seg_scatter source — no .contiguous() copies)."""
+
+    @staticmethod
+    def forward(ctx, ids4, seeds, *tables):
+        hip = hip_ext()
+        T = ids4.shape[0]
+        W = tables[0].shape[1]
+        X = tables[0].new_empty(T, W * len(tables))
+        rows_all = []
+        for i, table in enumerate(tables):
+            _, rows = hip.hashembed_fwd(table, ids4[:, i].contiguous(),
+                                        int(seeds[i]), X, i * W)
+            rows_all.append(rows)
+        ctx.save_for_backward(*rows_all)
+        ctx.nrows = [t.shape[0] for t in tables]
+        ctx.W = W
+        return X
+
+    @staticmethod
+    def backward(ctx, dX):
+        hip = hip_ext()
+        rows_all = ctx.saved_tensors
+        W = ctx.W
+        det = deterministic()
+        grads = []
+        for i, rows in enumerate(rows_all):
+            dY = dX[:, i * W : (i + 1) * W]  # strided view, no copy
+            dst = rows.reshape(-1)
+            order = torch.argsort(dst, stable=True) if det else torch.argsort(dst)
+            dst_sorted = dst[order].contiguous().int()
+            src = (order // 4).int()
+            acc_dt = torch.int64 if det else torch.float32
+            dT32 = torch.zeros(ctx.nrows[i], W, dtype=acc_dt, device=dX.device)
+            hip.seg_scatter_add(dst_sorted, src, dY, dT32)
+            if det:
+                dT32 = dT32.to(torch.float32) / FIXED_SCALE
+            grads.append(dT32.to(dX.dtype))
+        return (None, None, *grads)
+
+
+def multi_hashembed(ids4: torch.Tensor, seeds, tables) -> torch.Tensor:
+    """GPU fast path for MultiHashEmbed; CPU falls back to per-table
+    hashembed + cat."""
+    if ids4.is_cuda and _want_hip(tables[0]):
+        return _MultiHashEmbed.apply(ids4, list(seeds), *tables)
+    outs = [hashembed(t, ids4[:, i].contiguous(), int(seeds[i]))
+            for i, t in enumerate(tables)]
+    return torch.cat(outs, dim=1)
+
+
 # ------------------------------------------------------------- layernorm
 class _LayerNorm(torch.autograd.Function):
     @staticmethod

@@ -11,12 +11,16 @@
 // 4 murmur row ids are computed from the wave-uniform id (scalar-unit work)
 // and the 4 rows are gathered and summed in fp32 (SURVEY.md §2.5
 // hashembed_fwd: out[i] = sum_s E[h_s(id_i) % rows]).
+// ldY: row stride of Y — lets the 4 attr tables write straight into their
+// column block of the concatenated [T, 4W] embed matrix (the separate
+// torch.cat copied 0.77 GB per step at 1M words).
 template <typename T>
 __global__ void hashembed_fwd_kernel(const T* __restrict__ table,
                                      const uint64_t* __restrict__ ids,
                                      T* __restrict__ Y,
                                      int32_t* __restrict__ rows_out,
-                                     long nT, int nrows, int W, uint32_t seed) {
+                                     long nT, int nrows, int W, long ldY,
+                                     uint32_t seed) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
@@ -32,7 +36,7 @@ __global__ void hashembed_fwd_kernel(const T* __restrict__ table,
     const T* t1 = table + (long)r1 * W;
     const T* t2 = table + (long)r2 * W;
     const T* t3 = table + (long)r3 * W;
-    T* out = Y + t * (long)W;
+    T* out = Y + t * ldY;
     for (int w = lane; w < W; w += SRX_WAVE) {
       float acc = Elem<T>::ld(t0 + w) + Elem<T>::ld(t1 + w) +
                   Elem<T>::ld(t2 + w) + Elem<T>::ld(t3 + w);
@@ -110,12 +114,14 @@ __global__ void parser_step_fwd_kernel(const T* __restrict__ pre,
 // per-chunk register accumulation is already order-fixed (sorted entries),
 // and the cross-chunk pushes become associative integer adds, so the
 // result is bit-identical across runs (SRX_DETERMINISTIC).
+// ldSRC: row stride of SRC (lets callers pass column-block VIEWS of a
+// wider matrix, e.g. one attr's slice of the concatenated embed grad).
 template <typename T, int CHUNK, bool DET = false>
 __global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
                                        const int32_t* __restrict__ src_idx,
                                        const T* __restrict__ SRC,
                                        void* __restrict__ OUT,
-                                       long M, int W) {
+                                       long M, int W, long ldSRC) {
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
   const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
@@ -136,7 +142,7 @@ __global__ void seg_scatter_add_kernel(const int32_t* __restrict__ dst_sorted,
         }
         cur = d;
       }
-      const T* src = SRC + (long)src_idx[e] * W;
+      const T* src = SRC + (long)src_idx[e] * ldSRC;
       for (int c = 0; c < ncols; c++) {
         int w = lane + c * SRX_WAVE;
         if (w < W) acc[c] += Elem<T>::ld(src + w);

@@ -227,22 +227,38 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
 }
 
 // ------------------------------------------------------------- hashembed
-std::vector<at::Tensor> hashembed_fwd(at::Tensor table, at::Tensor ids, int64_t seed) {
+// out_/col_off: optional preallocated [nT, ldY] destination + column
+// offset — the 4 attr tables write straight into their block of the
+// concatenated embed matrix (no separate cat copy).
+std::vector<at::Tensor> hashembed_fwd(at::Tensor table, at::Tensor ids, int64_t seed,
+                                      c10::optional<at::Tensor> out_ = c10::nullopt,
+                                      int64_t col_off = 0) {
   check_dev(table);
   check_dev(ids);
   TORCH_CHECK(ids.scalar_type() == at::kLong, "ids must be int64 (bit-cast uint64)");
   long nT = ids.size(0);
   int nrows = (int)table.size(0);
   int W = (int)table.size(1);
-  auto Y = at::empty({nT, (long)W}, table.options());
+  at::Tensor Y;
+  long ldY = W;
+  char* yptr = nullptr;
+  if (out_) {
+    Y = *out_;
+    TORCH_CHECK(Y.size(0) == nT && Y.scalar_type() == table.scalar_type());
+    ldY = Y.size(1);
+    yptr = (char*)Y.data_ptr() + col_off * Y.element_size();
+  } else {
+    Y = at::empty({nT, (long)W}, table.options());
+    yptr = (char*)Y.data_ptr();
+  }
   auto rows = at::empty({nT, 4}, table.options().dtype(at::kInt));
   if (nT == 0) return {Y, rows};
   auto stream = at::cuda::getCurrentCUDAStream();
   DISPATCH_F(table.scalar_type(), {
     hipLaunchKernelGGL((hashembed_fwd_kernel<scalar_t>), dim3(grid_for(nT * SRX_WAVE)),
                        dim3(kBlock), 0, stream, (const scalar_t*)table.data_ptr(),
-                       (const uint64_t*)ids.data_ptr<int64_t>(), (scalar_t*)Y.data_ptr(),
-                       rows.data_ptr<int32_t>(), nT, nrows, W, (uint32_t)seed);
+                       (const uint64_t*)ids.data_ptr<int64_t>(), (scalar_t*)yptr,
+                       rows.data_ptr<int32_t>(), nT, nrows, W, ldY, (uint32_t)seed);
   });
   return {Y, rows};
 }
@@ -338,12 +354,14 @@ std::vector<at::Tensor> parser_step_bwd(at::Tensor dHidden, at::Tensor feats,
 // (caller converts back with /2^24).
 void seg_scatter_add(at::Tensor dst_sorted, at::Tensor src_idx, at::Tensor SRC,
                      at::Tensor OUT) {
-  check_dev(SRC);
+  TORCH_CHECK(SRC.is_cuda() && SRC.dim() == 2 && SRC.stride(1) == 1,
+              "SRC must be a CUDA row-view (innermost stride 1)");
   bool det = OUT.scalar_type() == at::kLong;
   TORCH_CHECK(det || OUT.scalar_type() == at::kFloat);
   TORCH_CHECK(dst_sorted.scalar_type() == at::kInt && src_idx.scalar_type() == at::kInt);
   long M = dst_sorted.numel();
   int W = (int)SRC.size(-1);
+  long ldSRC = SRC.stride(0);
   TORCH_CHECK(W <= 1024, "seg_scatter_add W <= 1024");
   if (M == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -355,12 +373,12 @@ void seg_scatter_add(at::Tensor dst_sorted, at::Tensor src_idx, at::Tensor SRC,
       hipLaunchKernelGGL((seg_scatter_add_kernel<scalar_t, CHUNK, true>), dim3(grid),
                          dim3(kBlock), 0, stream, dst_sorted.data_ptr<int32_t>(),
                          src_idx.data_ptr<int32_t>(), (const scalar_t*)SRC.data_ptr(),
-                         OUT.data_ptr(), M, W);
+                         OUT.data_ptr(), M, W, ldSRC);
     else
       hipLaunchKernelGGL((seg_scatter_add_kernel<scalar_t, CHUNK, false>), dim3(grid),
                          dim3(kBlock), 0, stream, dst_sorted.data_ptr<int32_t>(),
                          src_idx.data_ptr<int32_t>(), (const scalar_t*)SRC.data_ptr(),
-                         OUT.data_ptr(), M, W);
+                         OUT.data_ptr(), M, W, ldSRC);
   });
 }
 
@@ -825,7 +843,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd, py::arg("dY"), py::arg("X"),
         py::arg("g"), py::arg("mu"), py::arg("rstd"),
         py::arg("deterministic") = false);
-  m.def("hashembed_fwd", &hashembed_fwd);
+  m.def("hashembed_fwd", &hashembed_fwd, py::arg("table"), py::arg("ids"),
+        py::arg("seed"), py::arg("out") = py::none(), py::arg("col_off") = 0);
   m.def("hashembed_bwd", &hashembed_bwd);
   m.def("parser_step_fwd", &parser_step_fwd);
   m.def("parser_step_bwd", &parser_step_bwd);
