@@ -207,11 +207,13 @@ class DocBin:
             heads = None
             if "HEAD" in col:
                 rel = seg[:, col["HEAD"]].view(np.int64)
-                if rel.any():
+                # all-zero rel with no DEP annotation = unannotated (every
+                # token self-headed is not a tree anyone writes); all-zero
+                # WITH deps (e.g. a one-token "ROOT" doc) is a real parse
+                has_deps = "DEP" in col and bool(seg[:, col["DEP"]].any())
+                if rel.any() or has_deps:
                     idx = np.arange(n)
                     heads = np.where(rel == 0, -1, idx + rel).astype(np.int32)
-                # all-zero rel = no dependency annotation (every token
-                # self-headed is not a tree spaCy would write)
             deps = None
             if "DEP" in col and seg[:, col["DEP"]].any():
                 deps = [id2str.get(int(h), "") if h else "" for h in seg[:, col["DEP"]]]

@@ -180,3 +180,97 @@ def test_config_roundtrip_fuzz(data):
                 for k, v in d.items()
                 if not (isinstance(v, dict) and not prune(v))}
     assert prune(dict(cfg2)) == prune(dict(cfg))
+
+
+def test_fuzz_tokenizer_space_roundtrip_and_stability():
+    """Random texts: tokenize -> (words, spaces) reconstructs the text
+    (single-space normalized); serialization round-trip preserves output."""
+    import random
+
+    from spacy_ray_amd.vocab.tokenizer import Tokenizer
+
+    rng = random.Random(3)
+    tok = Tokenizer()
+    tok2 = Tokenizer.from_bytes(tok.to_bytes())
+    pieces = ["Hello", "don't", "U.S.", "3.5", "state-of-the-art", "(x)",
+              '"quote"', "a@b.com", "https://x.io/y", "...", "word,", "!?",
+              "£5", "e.g.", "I'm", "Dr.", "c'est"]
+    for _ in range(60):
+        text = " ".join(rng.choice(pieces) for _ in range(rng.randint(1, 12)))
+        words, spaces = tok.tokenize(text)
+        rebuilt = "".join(w + (" " if s else "") for w, s in zip(words, spaces))
+        assert rebuilt == text, (text, words, spaces)
+        assert tok2.tokenize(text) == (words, spaces)
+
+
+def test_fuzz_alignment_projection_consistency():
+    """Random re-tokenizations (random merges of adjacent tokens): the
+    a2b/b2a maps are mutually consistent and cover every non-empty token."""
+    import random
+
+    from spacy_ray_amd.vocab.align import get_alignment
+
+    rng = random.Random(5)
+    for _ in range(60):
+        n = rng.randint(1, 15)
+        b = [f"w{rng.randint(0, 9)}" for _ in range(n)]
+        # random merge of adjacent reference tokens -> predicted tokens
+        a = []
+        i = 0
+        while i < n:
+            j = min(n, i + rng.randint(1, 3))
+            a.append("".join(b[i:j]))
+            i = j
+        al = get_alignment(a, b)
+        for ai, bs in enumerate(al.a2b):
+            assert bs, (a, b, ai)
+            for bi in bs:
+                assert ai in al.b2a[bi]
+        for bi, as_ in enumerate(al.b2a):
+            assert as_, (a, b, bi)
+
+
+def test_fuzz_spacy_docbin_roundtrip():
+    """Random annotated docs survive the real `.spacy` wire format."""
+    import random
+
+    import numpy as np
+
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    rng = random.Random(7)
+    v = Vocab()
+    docs = []
+    for _ in range(20):
+        n = rng.randint(1, 12)
+        words = [f"tok{rng.randint(0, 30)}" for _ in range(n)]
+        tags = [f"T{rng.randint(0, 5)}" for _ in range(n)]
+        heads = [-1] + [rng.randint(0, i) for i in range(1, n)] if n > 1 else [-1]
+        deps = ["ROOT"] + [f"d{rng.randint(0, 4)}" for _ in range(n - 1)]
+        ents = ["O"] * n
+        i = 0
+        while i < n:
+            if rng.random() < 0.25:
+                ln = min(rng.randint(1, 3), n - i)
+                lab = f"E{rng.randint(0, 2)}"
+                if ln == 1:
+                    ents[i] = f"U-{lab}"
+                else:
+                    ents[i] = f"B-{lab}"
+                    for k in range(i + 1, i + ln - 1):
+                        ents[k] = f"I-{lab}"
+                    ents[i + ln - 1] = f"L-{lab}"
+                i += ln
+            else:
+                i += 1
+        docs.append(Doc(v, words, tags=tags, heads=heads, deps=deps, ents=ents))
+    data = DocBin(docs).to_bytes()
+    out = DocBin.from_bytes(data, Vocab()).docs
+    assert len(out) == len(docs)
+    for d0, d1 in zip(docs, out):
+        assert d1.words == d0.words
+        assert d1.tags == d0.tags
+        assert d1.heads.tolist() == list(d0.heads)
+        assert d1.deps == d0.deps
+        assert d1.ents == d0.ents
