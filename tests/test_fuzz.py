@@ -246,7 +246,9 @@ def test_fuzz_spacy_docbin_roundtrip():
         n = rng.randint(1, 12)
         words = [f"tok{rng.randint(0, 30)}" for _ in range(n)]
         tags = [f"T{rng.randint(0, 5)}" for _ in range(n)]
-        heads = [-1] + [rng.randint(0, i) for i in range(1, n)] if n > 1 else [-1]
+        # head strictly earlier than the child (self-head means ROOT in the
+        # spaCy wire convention; our Doc uses -1 for roots)
+        heads = [-1] + [rng.randint(0, i - 1) for i in range(1, n)] if n > 1 else [-1]
         deps = ["ROOT"] + [f"d{rng.randint(0, 4)}" for _ in range(n - 1)]
         ents = ["O"] * n
         i = 0
