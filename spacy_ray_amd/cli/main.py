@@ -277,6 +277,56 @@ def convert_cli(
     print(f"wrote {n} docs -> {output_path}")
 
 
+@app.command("package")
+def package_cli(
+    model_path: Path = typer.Argument(..., help="Trained pipeline directory (model-best/model-last)"),
+    output_dir: Path = typer.Argument(..., help="Directory to create the package in"),
+    name: str = typer.Option("pipeline", "--name", "-n", help="Package name (lang prefix added from meta)"),
+    version: str = typer.Option("0.0.0", "--version", "-v"),
+):
+    """Wrap a trained pipeline into a pip-installable package (spaCy
+    `package` role): <lang>_<name>-<version>/ with setup.py, a loader
+    module and the model data."""
+    import json
+    import shutil
+
+    meta = json.loads((model_path / "meta.json").read_text())
+    lang = meta.get("lang", "xx")
+    pkg = f"{lang}_{name}"
+    root = output_dir / f"{pkg}-{version}"
+    mod_dir = root / pkg
+    data_dir = mod_dir / f"{pkg}-{version}"
+    if data_dir.exists():
+        shutil.rmtree(data_dir)
+    data_dir.parent.mkdir(parents=True, exist_ok=True)
+    shutil.copytree(model_path, data_dir)
+    meta["name"] = name
+    meta["version"] = version
+    (data_dir / "meta.json").write_text(json.dumps(meta, indent=2))
+    (mod_dir / "__init__.py").write_text(
+        '"""Auto-generated pipeline package (spacy-mi package)."""\n'
+        "from pathlib import Path\n\n"
+        f"__version__ = {version!r}\n\n\n"
+        "def load(device: str = \"cpu\"):\n"
+        f"    from spacy_ray_amd import load as _load\n\n"
+        f"    return _load(Path(__file__).parent / '{pkg}-{version}',"
+        " device=device)\n"
+    )
+    (root / "setup.py").write_text(
+        "from setuptools import setup\n\n"
+        "setup(\n"
+        f"    name={pkg!r},\n"
+        f"    version={version!r},\n"
+        f"    packages=[{pkg!r}],\n"
+        "    include_package_data=True,\n"
+        f"    package_data={{{pkg!r}: ['{pkg}-{version}/**/*', '{pkg}-{version}/*']}},\n"
+        "    install_requires=['spacy_ray_amd'],\n"
+        ")\n"
+    )
+    (root / "MANIFEST.in").write_text(f"recursive-include {pkg} *\n")
+    print(f"package written -> {root}")
+
+
 @app.command("serve")
 def serve_cli(
     model_path: Path = typer.Argument(..., help="Trained pipeline directory (model-best/model-last)"),

@@ -367,3 +367,37 @@ def test_debug_data_with_path_overrides(tmp_path):
                   "--paths.train", str(bin_path), "--paths.dev", str(bin_path)])
     assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
     assert "10 docs" in r.stdout
+
+
+def test_cli_package_wraps_trained_model(tmp_path):
+    """spacy-mi package: trained dir -> pip-shaped package whose load()
+    restores a working pipeline."""
+    import subprocess
+    import sys
+
+    out = tmp_path / "model"
+    r = _run_cli([str(CFG), "--output", str(out), "--training.max_steps", "2",
+                  "--training.eval_frequency", "2"])
+    assert r.returncode == 0, r.stderr[-2000:]
+    pkg_out = tmp_path / "pkg"
+    r2 = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "package",
+         str(out / "model-last"), str(pkg_out), "--name", "demo",
+         "--version", "1.2.3"],
+        cwd=str(REPO), capture_output=True, text=True, timeout=300,
+    )
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    root = pkg_out / "en_demo-1.2.3"
+    assert (root / "setup.py").exists()
+    assert (root / "en_demo" / "__init__.py").exists()
+    assert (root / "en_demo" / "en_demo-1.2.3" / "meta.json").exists()
+    # import the generated module (no pip: path injection) and load()
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location(
+        "en_demo", root / "en_demo" / "__init__.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    nlp = mod.load()
+    doc = nlp("Don't panic!")
+    assert doc.tags is not None and len(doc.tags) == len(doc)
