@@ -271,27 +271,30 @@ class TransformerTok2Vec(nn.Module):
         return (word_ids[:total],
                 torch.arange(total, device=device, dtype=torch.int64), lens)
 
-    def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
-        device = batch.attr_ids.device
-        T = batch.n_tokens
-        seq_ids, seq2word, lengths = self._sequence(batch, device)
-        spans = self._windows(lengths)
-        nW = len(spans)
-        spans_np = np.asarray(spans, dtype=np.int64)
-        starts = torch.from_numpy(spans_np[:, 0]).to(device)
-        ns = torch.from_numpy(spans_np[:, 1] - spans_np[:, 0]).to(device)
-        L = int((spans_np[:, 1] - spans_np[:, 0]).max()) + 2
-        # fully vectorized window assembly (no per-window python loop)
+    # fixed length-bucket boundaries: every window pads only to its
+    # bucket's L instead of the batch-global max (one long doc used to pad
+    # EVERY window to ~130 positions while typical docs are ~46 subtokens
+    # — linear-in-L GEMM work ~2.8x down, quadratic attention ~8x down,
+    # identical outputs since padding is masked).  Fixed boundaries keep
+    # hipBLASLt's solution cache on a small set of shapes.
+    BUCKETS = (32, 48, 64, 96)
+
+    def _run_windows(self, seq_ids, ns_np, starts_np, L, device):
+        """Run the encoder over the given windows padded to length L;
+        returns (out [n, L, width], gather, valid)."""
+        n = len(ns_np)
+        starts = torch.from_numpy(starts_np).to(device)
+        ns = torch.from_numpy(ns_np).to(device)
         pos = torch.arange(L, device=device)
         tok_pos = pos.unsqueeze(0) - 1  # [1, L] sequence slot within window
-        valid = (tok_pos >= 0) & (tok_pos < ns.unsqueeze(1))  # [nW, L]
+        valid = (tok_pos >= 0) & (tok_pos < ns.unsqueeze(1))  # [n, L]
         gather = (starts.unsqueeze(1) + tok_pos).clamp_(min=0)
         gather = torch.where(valid, gather, torch.zeros_like(gather))
         input_ids = torch.where(valid, seq_ids[gather],
                                 torch.full_like(gather, PAD))
         input_ids[:, 0] = BOS
         eos_col = ns + 1
-        input_ids[torch.arange(nW, device=device), eos_col] = EOS
+        input_ids[torch.arange(n, device=device), eos_col] = EOS
         attn = (pos.unsqueeze(0) <= eos_col.unsqueeze(1)).long()
         import os
 
@@ -306,12 +309,31 @@ class TransformerTok2Vec(nn.Module):
                                attention_mask=attn).last_hidden_state
         else:
             out = self.trf(input_ids=input_ids, attention_mask=attn).last_hidden_state
-        # overlap- and subword-averaged scatter back to [T, width]
-        acc = out.new_zeros(T, self.width)
-        cnt = out.new_zeros(T, 1)
-        flat_idx = seq2word[gather[valid]]
-        acc.index_add_(0, flat_idx, out[valid])
-        cnt.index_add_(0, flat_idx, out.new_ones(flat_idx.shape[0], 1))
+        return out, gather, valid
+
+    def forward(self, batch: TokenBatch, drop: float = 0.0) -> torch.Tensor:
+        device = batch.attr_ids.device
+        T = batch.n_tokens
+        seq_ids, seq2word, lengths = self._sequence(batch, device)
+        spans = self._windows(lengths)
+        spans_np = np.asarray(spans, dtype=np.int64)
+        ns_all = (spans_np[:, 1] - spans_np[:, 0]).astype(np.int64)
+        starts_all = spans_np[:, 0]
+        acc = torch.zeros(T, self.width, device=device,
+                          dtype=next(self.trf.parameters()).dtype)
+        cnt = acc.new_zeros(T, 1)
+        bounds = [b for b in self.BUCKETS if b < self.window + 2] + [self.window + 2]
+        prev = 0
+        for L in bounds:
+            sel = np.nonzero((ns_all + 2 > prev) & (ns_all + 2 <= L))[0]
+            prev = L
+            if len(sel) == 0:
+                continue
+            out, gather, valid = self._run_windows(
+                seq_ids, ns_all[sel], starts_all[sel], L, device)
+            flat_idx = seq2word[gather[valid]]
+            acc.index_add_(0, flat_idx, out[valid])
+            cnt.index_add_(0, flat_idx, out.new_ones(flat_idx.shape[0], 1))
         Y = acc / cnt.clamp(min=1)
         if drop and self.training:
             Y = torch.nn.functional.dropout(Y, drop)

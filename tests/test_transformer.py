@@ -205,3 +205,33 @@ def test_hash_subwords_fallback():
     doc = Doc(nlp.vocab, ["a", "b", "c", "d"])
     out = trf(TokenBatch([doc], torch.device("cpu")))
     assert out.shape[0] == 4
+
+
+def test_window_length_bucketing_equivalent_outputs():
+    """Length-bucketed window batching (each window pads to its bucket's L
+    instead of the batch-global max) must produce the same outputs as a
+    single global pad length — padding is fully masked."""
+    import re
+
+    import numpy as np
+
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.models.batch import TokenBatch
+    from spacy_ray_amd.vocab.doc import Doc
+
+    cfg_text = TRF_CFG.replace("window = 16", "window = 128").replace(
+        "stride = 12", "stride = 96")
+    torch.manual_seed(0)
+    nlp = init_nlp(Config.from_str(cfg_text))
+    trf = nlp.get_pipe("transformer").module
+    trf.eval()
+    docs = make_synthetic_docs(nlp.vocab, n_docs=20, words_per_doc=12,
+                               vocab_size=200, n_tags=5, n_deps=5,
+                               n_ent_types=2, seed=2)
+    docs.append(Doc(nlp.vocab, [f"w{i % 50}" for i in range(70)]))
+    tb = TokenBatch(docs, torch.device("cpu"))
+    with torch.no_grad():
+        y_bucketed = trf(tb)
+        trf.BUCKETS = ()  # single global bucket = pre-bucketing behavior
+        y_single = trf(tb)
+    assert torch.allclose(y_bucketed, y_single, atol=2e-5)
