@@ -23,6 +23,11 @@ __device__ __forceinline__ bf16_t f2bf(float f) {
 template <typename T>
 struct Elem;
 
+// paired loads/stores (2 contiguous elements in one transaction — bf16
+// pairs move as one dword instead of two 2-byte accesses)
+template <typename T>
+struct Elem2;
+
 template <>
 struct Elem<float> {
   static __device__ __forceinline__ float ld(const float* p) { return *p; }
@@ -33,6 +38,30 @@ template <>
 struct Elem<bf16_t> {
   static __device__ __forceinline__ float ld(const bf16_t* p) { return bf2f(*p); }
   static __device__ __forceinline__ void st(bf16_t* p, float v) { *p = f2bf(v); }
+};
+
+template <>
+struct Elem2<float> {
+  static __device__ __forceinline__ void ld(const float* p, float* v) {
+    float2 x = *(const float2*)p;
+    v[0] = x.x; v[1] = x.y;
+  }
+  static __device__ __forceinline__ void st(float* p, const float* v) {
+    *(float2*)p = make_float2(v[0], v[1]);
+  }
+};
+
+template <>
+struct Elem2<bf16_t> {
+  static __device__ __forceinline__ void ld(const bf16_t* p, float* v) {
+    uint32_t u = *(const uint32_t*)p;
+    v[0] = bf2f((bf16_t)(u & 0xffffu));
+    v[1] = bf2f((bf16_t)(u >> 16));
+  }
+  static __device__ __forceinline__ void st(bf16_t* p, const float* v) {
+    uint32_t u = (uint32_t)f2bf(v[0]) | ((uint32_t)f2bf(v[1]) << 16);
+    *(uint32_t*)p = u;
+  }
 };
 
 // Wave-wide (64-lane) sum reduction; result valid in all lanes.

@@ -162,11 +162,128 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ X,
   }
 }
 
+#define SRX_LN_MAX_W 1024
+
+// Wide-row variants (W % 128 == 0, e.g. the trf's 768): each lane owns
+// CONTIGUOUS PAIRS so every access is a full dword (the scalar kernels
+// move bf16 2 bytes at a time — layernorm was 11%% of trf GPU time).
+template <typename T>
+__global__ void layernorm_fwd_v2_kernel(const T* __restrict__ X,
+                                        const T* __restrict__ g,
+                                        const T* __restrict__ b,
+                                        T* __restrict__ Y,
+                                        float* __restrict__ mu_out,
+                                        float* __restrict__ rstd_out,
+                                        long N, int W, float eps) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  const int npairs = W / 2;
+  for (long n = wave; n < N; n += nwaves) {
+    const T* row = X + n * (long)W;
+    float s = 0.f, sq = 0.f;
+    for (int p = lane; p < npairs; p += SRX_WAVE) {
+      float v[2];
+      Elem2<T>::ld(row + 2 * p, v);
+      s += v[0] + v[1];
+      sq += v[0] * v[0] + v[1] * v[1];
+    }
+    s = wave_reduce_sum(s);
+    sq = wave_reduce_sum(sq);
+    float mu = s / W;
+    float var = sq / W - mu * mu;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (lane == 0) {
+      mu_out[n] = mu;
+      rstd_out[n] = rstd;
+    }
+    T* out = Y + n * (long)W;
+    for (int p = lane; p < npairs; p += SRX_WAVE) {
+      float v[2], gg[2], bb[2], o[2];
+      Elem2<T>::ld(row + 2 * p, v);
+      Elem2<T>::ld(g + 2 * p, gg);
+      Elem2<T>::ld(b + 2 * p, bb);
+      o[0] = (v[0] - mu) * rstd * gg[0] + bb[0];
+      o[1] = (v[1] - mu) * rstd * gg[1] + bb[1];
+      Elem2<T>::st(out + 2 * p, o);
+    }
+  }
+}
+
+template <typename T, bool DET = false>
+__global__ void layernorm_bwd_v2_kernel(const T* __restrict__ dY,
+                                        const T* __restrict__ X,
+                                        const T* __restrict__ g,
+                                        const float* __restrict__ mu,
+                                        const float* __restrict__ rstd,
+                                        T* __restrict__ dX,
+                                        void* __restrict__ dg32,
+                                        void* __restrict__ db32,
+                                        long N, int W) {
+  const int lane = threadIdx.x & (SRX_WAVE - 1);
+  const long wave = ((long)blockIdx.x * blockDim.x + threadIdx.x) / SRX_WAVE;
+  const long nwaves = ((long)gridDim.x * blockDim.x) / SRX_WAVE;
+  const int npairs = W / 2;
+  const int ncols = (npairs + SRX_WAVE - 1) / SRX_WAVE;
+  float dg_loc[SRX_LN_MAX_W / SRX_WAVE], db_loc[SRX_LN_MAX_W / SRX_WAVE];
+  for (int c = 0; c < 2 * ncols; c++) { dg_loc[c] = 0.f; db_loc[c] = 0.f; }
+  for (long n = wave; n < N; n += nwaves) {
+    const T* xrow = X + n * (long)W;
+    const T* dyrow = dY + n * (long)W;
+    float m = mu[n], r = rstd[n];
+    float s1 = 0.f, s2 = 0.f;
+    for (int c = 0; c < ncols; c++) {
+      int p = lane + c * SRX_WAVE;
+      if (p >= npairs) break;
+      float x[2], dy[2], gg[2];
+      Elem2<T>::ld(xrow + 2 * p, x);
+      Elem2<T>::ld(dyrow + 2 * p, dy);
+      Elem2<T>::ld(g + 2 * p, gg);
+#pragma unroll
+      for (int e = 0; e < 2; e++) {
+        float xhat = (x[e] - m) * r;
+        float dxhat = dy[e] * gg[e];
+        s1 += dxhat;
+        s2 += dxhat * xhat;
+        dg_loc[2 * c + e] += dy[e] * xhat;
+        db_loc[2 * c + e] += dy[e];
+      }
+    }
+    s1 = wave_reduce_sum(s1) / W;
+    s2 = wave_reduce_sum(s2) / W;
+    T* out = dX + n * (long)W;
+    for (int c = 0; c < ncols; c++) {
+      int p = lane + c * SRX_WAVE;
+      if (p >= npairs) break;
+      float x[2], dy[2], gg[2], o[2];
+      Elem2<T>::ld(xrow + 2 * p, x);
+      Elem2<T>::ld(dyrow + 2 * p, dy);
+      Elem2<T>::ld(g + 2 * p, gg);
+#pragma unroll
+      for (int e = 0; e < 2; e++) {
+        float xhat = (x[e] - m) * r;
+        float dxhat = dy[e] * gg[e];
+        o[e] = r * (dxhat - s1 - xhat * s2);
+      }
+      Elem2<T>::st(out + 2 * p, o);
+    }
+  }
+  for (int c = 0; c < ncols; c++) {
+    int p = lane + c * SRX_WAVE;
+    if (p >= npairs) continue;
+#pragma unroll
+    for (int e = 0; e < 2; e++) {
+      int w = 2 * p + e;
+      if (dg_loc[2 * c + e] != 0.f) srx_atomic_add<DET>(dg32, w, dg_loc[2 * c + e]);
+      if (db_loc[2 * c + e] != 0.f) srx_atomic_add<DET>(db32, w, db_loc[2 * c + e]);
+    }
+  }
+}
+
 // dX per row; dg/db: per-wave REGISTER accumulation over all the wave's
 // rows, ONE atomicAdd per column per wave at the end.  (A per-row-per-elem
 // atomic version serialized on the 2*W hot addresses: 772 us/call at
 // N=32k, W=96 — this shape runs in ~30 us.)  Needs W <= SRX_LN_MAX_W.
-#define SRX_LN_MAX_W 1024
 // DET=true: dg/db buffers are int64 fixed-point (bit-deterministic).
 template <typename T, bool DET = false>
 __global__ void layernorm_bwd_kernel(const T* __restrict__ dY,

@@ -166,11 +166,18 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor X, at::Tensor g, at::Tensor b, 
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = grid_for(N * SRX_WAVE);
   DISPATCH_F(X.scalar_type(), {
-    hipLaunchKernelGGL((layernorm_fwd_kernel<scalar_t>), dim3(grid), dim3(kBlock), 0,
-                       stream, (const scalar_t*)X.data_ptr(),
-                       (const scalar_t*)g.data_ptr(), (const scalar_t*)b.data_ptr(),
-                       (scalar_t*)Y.data_ptr(), mu.data_ptr<float>(),
-                       rstd.data_ptr<float>(), N, W, (float)eps);
+    if (W % (2 * SRX_WAVE) == 0)
+      hipLaunchKernelGGL((layernorm_fwd_v2_kernel<scalar_t>), dim3(grid), dim3(kBlock), 0,
+                         stream, (const scalar_t*)X.data_ptr(),
+                         (const scalar_t*)g.data_ptr(), (const scalar_t*)b.data_ptr(),
+                         (scalar_t*)Y.data_ptr(), mu.data_ptr<float>(),
+                         rstd.data_ptr<float>(), N, W, (float)eps);
+    else
+      hipLaunchKernelGGL((layernorm_fwd_kernel<scalar_t>), dim3(grid), dim3(kBlock), 0,
+                         stream, (const scalar_t*)X.data_ptr(),
+                         (const scalar_t*)g.data_ptr(), (const scalar_t*)b.data_ptr(),
+                         (scalar_t*)Y.data_ptr(), mu.data_ptr<float>(),
+                         rstd.data_ptr<float>(), N, W, (float)eps);
   });
   return {Y, mu, rstd};
 }
@@ -200,8 +207,23 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dY, at::Tensor X, at::Tensor g,
     // count at waves*W regardless, so more waves are safe)
     long cap = 4096;
     int grid = (int)std::min<long>((N + 3) / 4, cap);
+    bool v2 = W % (2 * SRX_WAVE) == 0;
     DISPATCH_F(X.scalar_type(), {
-      if (deterministic)
+      if (v2 && deterministic)
+        hipLaunchKernelGGL((layernorm_bwd_v2_kernel<scalar_t, true>), dim3(grid),
+                           dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                           (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
+                           mu.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (scalar_t*)dX.data_ptr(), dg32.data_ptr(),
+                           db32.data_ptr(), N, W);
+      else if (v2)
+        hipLaunchKernelGGL((layernorm_bwd_v2_kernel<scalar_t, false>), dim3(grid),
+                           dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
+                           (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
+                           mu.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (scalar_t*)dX.data_ptr(), dg32.data_ptr(),
+                           db32.data_ptr(), N, W);
+      else if (deterministic)
         hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t, true>), dim3(grid),
                            dim3(kBlock), 0, stream, (const scalar_t*)dY.data_ptr(),
                            (const scalar_t*)X.data_ptr(), (const scalar_t*)g.data_ptr(),
