@@ -852,6 +852,17 @@ std::vector<std::vector<at::Tensor>> srx_run_transition_loop(
                            at::Tensor, bool>>
         tasks);
 
+// srx_gpustate.hip — fully GPU-resident state machines (one wave per doc)
+std::vector<at::Tensor> srx_gpu_arceager(
+    at::Tensor pre, at::Tensor off, at::Tensor lens, at::Tensor gh,
+    at::Tensor gl, at::Tensor kids_off, at::Tensor kids, at::Tensor lowerB,
+    at::Tensor upperW, at::Tensor upperB, int64_t total, int64_t n_labels,
+    bool train);
+std::vector<at::Tensor> srx_gpu_biluo(
+    at::Tensor pre, at::Tensor off, at::Tensor lens, at::Tensor gold,
+    at::Tensor lowerB, at::Tensor upperW, at::Tensor upperB, int64_t total,
+    int64_t n_types, bool train);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seq2col_fwd", &seq2col_fwd);
   m.def("seq2col_bwd", &seq2col_bwd, py::arg("dY"), py::arg("starts"),
@@ -888,4 +899,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dpre_scatter", &dpre_scatter);
   m.def("run_transition_loop", &srx_run_transition_loop,
         py::call_guard<py::gil_scoped_release>());
+  m.def("gpu_arceager", &srx_gpu_arceager, py::arg("pre"), py::arg("off"),
+        py::arg("lens"), py::arg("gh"), py::arg("gl"), py::arg("kids_off"),
+        py::arg("kids"), py::arg("lowerB"), py::arg("upperW"),
+        py::arg("upperB"), py::arg("total"), py::arg("n_labels"),
+        py::arg("train"));
+  m.def("gpu_biluo", &srx_gpu_biluo, py::arg("pre"), py::arg("off"),
+        py::arg("lens"), py::arg("gold"), py::arg("lowerB"), py::arg("upperW"),
+        py::arg("upperB"), py::arg("total"), py::arg("n_types"),
+        py::arg("train"));
+  m.attr("GPU_STATE_MAXLEN") = 128;
 }

@@ -259,7 +259,8 @@ class _TransitionTask:
     __slots__ = ("pipe", "shards", "t2v", "train", "T", "pre", "pre_d",
                  "dPre32", "hip", "score_chunks", "gold_chunks",
                  "valid_chunks", "n_states_total", "entries", "fused",
-                 "task_id", "cpp", "cpp_outs")
+                 "task_id", "cpp", "cpp_outs", "gpu", "gpu_decode",
+                 "steps_dev")
 
     def __init__(self, pipe, shards, t2v, train: bool) -> None:
         from spacy_ray_amd.ops import api as _ops
@@ -303,6 +304,12 @@ class _TransitionTask:
         self.gold_chunks: List[torch.Tensor] = []
         self.valid_chunks: List[torch.Tensor] = []
         self.n_states_total = 0
+        # GPU-resident state machine (srx_gpustate.hip): set by
+        # make_loss_task/make_predict_task to the device-gold dict; the
+        # whole transition loop then runs in ONE kernel launch.
+        self.gpu: Optional[Dict] = None
+        self.gpu_decode = None
+        self.steps_dev: Optional[torch.Tensor] = None
 
     def launch(self, states):
         from spacy_ray_amd.ops import api as _ops
@@ -398,6 +405,16 @@ def run_transition_tasks(tasks: List[_TransitionTask]) -> None:
     GPU default: the whole loop runs in C++ (hip.run_transition_loop, ONE
     python crossing per batch); the python round-robin below is the CPU /
     SRX_CPP_LOOP=0 fallback."""
+    # GPU-resident state machines first: each is ONE async kernel launch on
+    # the current stream, so any remaining host-loop tasks below overlap it.
+    gpu_tasks = [t for t in tasks if t.gpu is not None]
+    if gpu_tasks:
+        from spacy_ray_amd.utils import timing
+
+        with timing.span("raw/gpu_state"):
+            for t in gpu_tasks:
+                t.pipe._gpu_launch(t)
+        tasks = [t for t in tasks if t.gpu is None]
     units = [(t, s) for t in tasks for s in t.shards]
     if not units:
         return
@@ -528,7 +545,7 @@ class _TransitionPipeBase(TrainablePipe):
         t2v = task.t2v
         if not task.train:
             return None, 0.0
-        if task.cpp:
+        if task.cpp or task.gpu is not None:
             outs = [o for o in task.cpp_outs if o and o[0].shape[0] > 0]
             if not outs:
                 return t2v.new_zeros(()), 0.0
@@ -538,7 +555,6 @@ class _TransitionPipeBase(TrainablePipe):
                 scores, gold, valid, feats, which, hidden = (
                     torch.cat(c, dim=0) for c in zip(*outs)
                 )
-            SS = scores.shape[0]
             from spacy_ray_amd.ops.api import transition_loop_loss
 
             mod = self.module
@@ -546,7 +562,15 @@ class _TransitionPipeBase(TrainablePipe):
                 loss = transition_loop_loss(
                     task.pre, mod.lower_b, mod.upper.weight, mod.upper.bias,
                     scores, gold, valid, feats, which, hidden,
-                ) / SS
+                )
+                if task.steps_dev is not None:
+                    # GPU state machine: arenas are CAPACITY-sized (unused
+                    # rows masked out of the CE); normalize by the REAL
+                    # transition count the kernel accumulated — a device
+                    # scalar, so no host sync
+                    loss = loss / task.steps_dev.to(loss.dtype).clamp(min=1)[0]
+                else:
+                    loss = loss / scores.shape[0]
             return loss, loss.detach()
         if not task.score_chunks:
             return t2v.new_zeros(()), 0.0
@@ -605,6 +629,51 @@ class _TransitionPipeBase(TrainablePipe):
     def _build_gold(self, examples):
         raise NotImplementedError
 
+    # ---- GPU-resident state machine (srx_gpustate.hip): the ENTIRE greedy
+    # transition loop for a batch runs in ONE kernel launch (one wave64 per
+    # doc — state, dynamic oracle, scorer and argmax all in LDS/registers),
+    # replacing the per-step host round trips of the C++ loop.  Train
+    # returns the same arenas the host loop produces, so the batched CE
+    # backward is shared; decode returns heads/labels (or BILUO tags)
+    # directly.  SRX_GPU_STATES=0 reverts to the host loop.
+    def _gpu_ready(self, lengths, t2v) -> bool:
+        if not t2v.is_cuda or os.environ.get("SRX_GPU_STATES", "1") != "1":
+            return False
+        from spacy_ray_amd.ops import api as _ops
+
+        hip = _ops.hip_ext()
+        if hip is None or not hasattr(hip, "gpu_arceager"):
+            return False
+        if self.hidden_width > 64:
+            return False
+        return self._gpu_supported(lengths, hip)
+
+    def _gpu_supported(self, lengths, hip) -> bool:
+        return False
+
+    def _gold_to_device(self, staged, lengths, device):
+        raise NotImplementedError
+
+    def _gpu_launch(self, task) -> None:
+        raise NotImplementedError
+
+    def _maybe_gpu_task(self, staged, lengths, t2v, batch, train):
+        """Build a GPU-state-machine task if supported, else None."""
+        if not self._gpu_ready(lengths, t2v):
+            return None
+        if train:
+            dev_key = (self.name + "_gold_dev", tuple(self.labels))
+            gdev = batch.staged.get(dev_key) if batch is not None else None
+            if gdev is None:
+                gdev = self._gold_to_device(staged, lengths, t2v.device)
+                if batch is not None:
+                    batch.staged[dev_key] = gdev
+        else:
+            gdev = self._gold_to_device(None, lengths, t2v.device)
+        task = self.begin_task([], t2v, train=train)
+        task.gpu = gdev
+        return task
+
     def make_loss_task(self, examples, t2v, batch=None) -> "_TransitionTask":
         key = (self.name + "_gold", tuple(self.labels))
         staged = batch.staged.get(key) if batch is not None else None
@@ -614,6 +683,9 @@ class _TransitionPipeBase(TrainablePipe):
             lengths = batch.lengths_np
         else:
             lengths = np.asarray([len(eg.reference) for eg in examples], dtype=np.int32)
+        task = self._maybe_gpu_task(staged, lengths, t2v, batch, train=True)
+        if task is not None:
+            return task
         shards = []
         for lo, hi, base in self._split_docs(lengths, self._n_shards(t2v)):
             states = self._make_states(lengths[lo:hi], base)
@@ -623,6 +695,9 @@ class _TransitionPipeBase(TrainablePipe):
 
     def make_predict_task(self, docs, t2v):
         lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
+        task = self._maybe_gpu_task(None, lengths, t2v, None, train=False)
+        if task is not None:
+            return task, None, None
         splits = self._split_docs(lengths, self._n_shards(t2v))
         shards = [self._make_states(lengths[lo:hi], base) for lo, hi, base in splits]
         return self.begin_task(shards, t2v, train=False), splits, shards
@@ -645,8 +720,14 @@ class _TransitionPipeBase(TrainablePipe):
         task, splits, shards = self.make_predict_task(docs, t2v)
         with torch.no_grad():
             run_transition_tasks([task])
+        if task.gpu is not None:
+            self._annotate_gpu(docs, task.gpu_decode)
+            return
         for (lo, hi, base), states in zip(splits, shards):
             self._annotate(docs[lo:hi], states)
+
+    def _annotate_gpu(self, docs, decode) -> None:
+        raise NotImplementedError
 
 
 class ParserPipe(_TransitionPipeBase):
@@ -734,9 +815,10 @@ class ParserPipe(_TransitionPipeBase):
             states.set_sent_gold(staged[2])
 
     def _annotate(self, docs, states) -> None:
-        heads = states.heads()
-        labels = states.labels()
         sents = states.sent_starts() if self.use_break else None
+        self._annotate_arrays(docs, states.heads(), states.labels(), sents)
+
+    def _annotate_arrays(self, docs, heads, labels, sents=None) -> None:
         off = 0
         for doc in docs:
             n = len(doc)
@@ -754,6 +836,70 @@ class ParserPipe(_TransitionPipeBase):
                     ss[0] = 1  # the first token always starts a sentence
                 doc.sent_starts = ss
             off += n
+
+    # ---- GPU state machine hooks (srx_gpustate.hip::gpu_arceager_kernel)
+    def _gpu_supported(self, lengths, hip) -> bool:
+        # BREAK needs the sentence-gold machinery (host loop); stack/arc
+        # arrays live in per-wave LDS sized for GPU_STATE_MAXLEN tokens
+        return (not self.use_break and
+                (len(lengths) == 0 or int(lengths.max()) <= hip.GPU_STATE_MAXLEN))
+
+    def _gold_to_device(self, staged, lengths, device):
+        lengths = np.asarray(lengths, dtype=np.int64)
+        n_docs = len(lengths)
+        off = np.zeros(n_docs, dtype=np.int64)
+        if n_docs > 1:
+            np.cumsum(lengths[:-1], out=off[1:])
+        total = int(lengths.sum())
+
+        def to(a, dt):
+            return torch.from_numpy(np.ascontiguousarray(a, dtype=dt)).to(device)
+
+        g = {
+            "off": to(off, np.int32), "lens": to(lengths, np.int32),
+            "total": total,
+        }
+        empty = torch.empty(0, dtype=torch.int32, device=device)
+        if staged is None:  # decode: the oracle inputs are never read
+            g.update(gh=empty, gl=empty, kids_off=empty, kids=empty)
+            return g
+        heads, labs = staged[0], staged[1]
+        # gold-children CSR over batch-global parent ids with DOC-LOCAL
+        # child ids (mirrors ArcEagerBatch::set_gold's kids/kids_off)
+        tok_off = np.repeat(off, lengths)
+        tok_len = np.repeat(lengths, lengths)
+        local = np.arange(total, dtype=np.int64) - tok_off
+        gh = heads.astype(np.int64)
+        ok = (gh >= 0) & (gh < tok_len)
+        parent = tok_off[ok] + gh[ok]
+        order = np.argsort(parent, kind="stable")
+        kids = local[ok][order]
+        kids_off = np.zeros(total + 1, dtype=np.int64)
+        np.cumsum(np.bincount(parent, minlength=total), out=kids_off[1:])
+        g.update(gh=to(heads, np.int32), gl=to(labs, np.int32),
+                 kids_off=to(kids_off, np.int32), kids=to(kids, np.int32))
+        return g
+
+    def _gpu_launch(self, task) -> None:
+        g = task.gpu
+        mod = self.module
+        outs = task.hip.gpu_arceager(
+            task.pre_d.contiguous(), g["off"], g["lens"], g["gh"], g["gl"],
+            g["kids_off"], g["kids"], mod.lower_b.detach(),
+            mod.upper.weight.detach(), mod.upper.bias.detach(), g["total"],
+            len(self.labels), task.train,
+        )
+        if task.train:
+            task.cpp_outs.append(list(outs[:6]))
+            task.steps_dev = outs[8]
+        else:
+            task.gpu_decode = (outs[6], outs[7])
+
+    def _annotate_gpu(self, docs, decode) -> None:
+        heads_d, labels_d = decode
+        self._annotate_arrays(
+            docs, heads_d.cpu().numpy(), labels_d.cpu().numpy()
+        )
 
 
 class NerPipe(_TransitionPipeBase):
@@ -797,9 +943,47 @@ class NerPipe(_TransitionPipeBase):
         states.set_gold(staged)
 
     def _annotate(self, docs, states) -> None:
-        tags = states.tags()
+        self._annotate_tags(docs, states.tags())
+
+    def _annotate_tags(self, docs, tags) -> None:
         off = 0
         for doc in docs:
             n = len(doc)
             doc.ents = codes_to_biluo(tags[off:off + n], self.labels)
             off += n
+
+    # ---- GPU state machine hooks (srx_gpustate.hip::gpu_biluo_kernel)
+    def _gpu_supported(self, lengths, hip) -> bool:
+        return True  # (open, open_start) state is register-resident: no len cap
+
+    def _gold_to_device(self, staged, lengths, device):
+        lengths = np.asarray(lengths, dtype=np.int64)
+        n_docs = len(lengths)
+        off = np.zeros(n_docs, dtype=np.int64)
+        if n_docs > 1:
+            np.cumsum(lengths[:-1], out=off[1:])
+
+        def to(a, dt):
+            return torch.from_numpy(np.ascontiguousarray(a, dtype=dt)).to(device)
+
+        gold = (to(staged, np.int32) if staged is not None
+                else torch.empty(0, dtype=torch.int32, device=device))
+        return {"off": to(off, np.int32), "lens": to(lengths, np.int32),
+                "total": int(lengths.sum()), "gold": gold}
+
+    def _gpu_launch(self, task) -> None:
+        g = task.gpu
+        mod = self.module
+        outs = task.hip.gpu_biluo(
+            task.pre_d.contiguous(), g["off"], g["lens"], g["gold"],
+            mod.lower_b.detach(), mod.upper.weight.detach(),
+            mod.upper.bias.detach(), g["total"], len(self.labels), task.train,
+        )
+        if task.train:
+            task.cpp_outs.append(list(outs[:6]))
+            task.steps_dev = outs[7]
+        else:
+            task.gpu_decode = outs[6]
+
+    def _annotate_gpu(self, docs, decode) -> None:
+        self._annotate_tags(docs, decode.cpu().numpy())

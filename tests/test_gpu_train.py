@@ -88,6 +88,7 @@ def test_fused_step_parity(monkeypatch):
                                         "examples", "configs", "en_core_cnn.cfg"))
 
     def run(fused):
+        monkeypatch.setenv("SRX_GPU_STATES", "0")  # exercise the step loop
         monkeypatch.setenv("SRX_FUSED_STEP", "1" if fused else "0")
         torch.manual_seed(0)
         np.random.seed(0)
@@ -130,6 +131,7 @@ def test_cpp_loop_parity(monkeypatch):
                                         "examples", "configs", "en_core_cnn.cfg"))
 
     def run(cpp):
+        monkeypatch.setenv("SRX_GPU_STATES", "0")  # exercise the host loops
         monkeypatch.setenv("SRX_CPP_LOOP", "1" if cpp else "0")
         torch.manual_seed(0)
         np.random.seed(0)
@@ -161,6 +163,92 @@ def test_cpp_loop_parity(monkeypatch):
     # greedy decode: near-ties under bf16 rounding may flip a few actions
     assert (heads_cpp == heads_ref).mean() > 0.97
     assert (ents_cpp == ents_ref).mean() > 0.97
+
+
+@need_gpu
+def test_gpu_state_machine_parity(monkeypatch):
+    """SRX_GPU_STATES=1 (fully GPU-resident transition loop: one wave per
+    doc runs the whole greedy parse/NER in-kernel, srx_gpustate.hip) must
+    match the C++ host loop: same losses, same gradients, same decode
+    annotations (same state machine + oracle + scorer math re-derived
+    in-kernel; tolerance covers bf16 accumulation-order jitter)."""
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.vocab.doc import Example
+
+    cfg = Config.from_disk(os.path.join(os.path.dirname(__file__), "..",
+                                        "examples", "configs", "en_core_cnn.cfg"))
+
+    def run(gpu_states):
+        monkeypatch.setenv("SRX_GPU_STATES", "1" if gpu_states else "0")
+        torch.manual_seed(0)
+        np.random.seed(0)
+        nlp = init_nlp(cfg, device="cuda:0", sample_size=32)
+        T = resolve(cfg.interpolate()["training"], validate=False)
+        engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+        docs = make_synthetic_docs(nlp.vocab, n_docs=64, words_per_doc=14,
+                                   vocab_size=400, n_tags=50, n_deps=40,
+                                   n_ent_types=4, seed=13)
+        examples = [Example.from_doc(d) for d in docs]
+        losses = {}
+        engine.accumulate(examples, drop=0.0, losses=losses)
+        torch.cuda.synchronize()
+        pred = [d.copy_unannotated() for d in docs]
+        nlp.predict_docs(pred)
+        heads = np.concatenate([d.heads for d in pred])
+        deps = np.concatenate([np.asarray(d.deps, dtype=object) for d in pred])
+        ents = np.concatenate([np.asarray(d.ents, dtype=object) for d in pred])
+        return dict(losses), engine.grad_shard.float().clone(), heads, deps, ents
+
+    losses_ref, grad_ref, heads_ref, deps_ref, ents_ref = run(False)
+    losses_gpu, grad_gpu, heads_gpu, deps_gpu, ents_gpu = run(True)
+    for k in losses_ref:
+        assert abs(losses_gpu[k] - losses_ref[k]) <= 1e-3 + 0.02 * abs(losses_ref[k]), (
+            k, losses_ref[k], losses_gpu[k])
+    denom = grad_ref.abs().mean().clamp(min=1e-8)
+    rel = (grad_gpu - grad_ref).abs().mean() / denom
+    assert float(rel) < 0.05, float(rel)
+    # greedy decode: near-ties under bf16 rounding may flip a few actions
+    assert (heads_gpu == heads_ref).mean() > 0.97
+    assert (deps_gpu == deps_ref).mean() > 0.97
+    assert (ents_gpu == ents_ref).mean() > 0.97
+
+
+@need_gpu
+def test_gpu_state_machine_long_doc_fallback(monkeypatch):
+    """Docs above GPU_STATE_MAXLEN must fall back to the host loop (mixed
+    batch still trains and decodes)."""
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.vocab.doc import Example
+
+    monkeypatch.setenv("SRX_GPU_STATES", "1")
+    cfg = Config.from_disk(os.path.join(os.path.dirname(__file__), "..",
+                                        "examples", "configs", "en_core_cnn.cfg"))
+    nlp = init_nlp(cfg, device="cuda:0", sample_size=16)
+    T = resolve(cfg.interpolate()["training"], validate=False)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    docs = make_synthetic_docs(nlp.vocab, n_docs=8, words_per_doc=200,
+                               vocab_size=300, n_tags=50, n_deps=40,
+                               n_ent_types=4, seed=3)
+    assert max(len(d) for d in docs) > 128  # forces the parser fallback
+    losses = {}
+    engine.accumulate([Example.from_doc(d) for d in docs], drop=0.0,
+                      losses=losses)
+    engine.apply_step()
+    torch.cuda.synchronize()
+    assert all(np.isfinite(v) for v in losses.values()), losses
+    pred = [d.copy_unannotated() for d in docs]
+    nlp.predict_docs(pred)
+    for d in pred:
+        assert d.heads is not None and len(d.heads) == len(d)
+        assert d.ents is not None
 
 
 @need_gpu
