@@ -198,6 +198,9 @@ class Language:
         for pipe, pt2v in heads:
             pipe.predict_and_set(docs, pt2v, batch)
         for pipe, task, splits, shards in trans:
+            if task.gpu is not None:  # GPU state machine: decode tensors
+                pipe._annotate_gpu(docs, task.gpu_decode)
+                continue
             for (lo, hi, base), states in zip(splits, shards):
                 pipe._annotate(docs[lo:hi], states)
         return docs
