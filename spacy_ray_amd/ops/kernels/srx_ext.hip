@@ -709,7 +709,9 @@ void launch_mwe(const at::Tensor& X, const at::Tensor& Wt, const at::Tensor& bia
                 double eps, hipStream_t stream) {
   constexpr int WP = W + 8;
   constexpr int NW = W / 32;
-  size_t lds_bytes = (size_t)(3 * 64 * WP + 3 * W * WP) * sizeof(bf16_t) +
+  // per-section staging: [64][WP] A + [3W][WP] B (73 KB at W=96 -> two
+  // blocks per CU; the old all-sections layout was 101 KB -> one block)
+  size_t lds_bytes = (size_t)(64 * WP + 3 * W * WP) * sizeof(bf16_t) +
                      (size_t)(2 * NW * 64) * sizeof(float);
   static bool attr_set = false;
   if (!attr_set) {

@@ -12,14 +12,19 @@
 //   (global n-tile p*(W/32)+w): the P=3 maxout is an in-register
 //   elementwise max over the wave's own 3 accumulators, and the LayerNorm
 //   column span of a wave is exactly within-piece cols [32w, 32w+32);
-// * A_s tiles staged in LDS [64][W+8] (+8 bf16 row pad: the natural
-//   48/64-dword stride would multi-way-conflict ds_read_b128; 52/68 dwords
-//   pad to 2-way);
+// * A and B are staged PER SECTION ([64][W+8] + [3W][W+8] at a time, not
+//   all three sections at once): at W=96 that is 73 KB of LDS instead of
+//   101 KB, which fits TWO blocks per CU (6 waves) instead of one — MFMA
+//   and staging latency of one block hides under the other (the
+//   single-block version measured 1.85 ms at T=1M; occupancy was the
+//   bound, not bandwidth: the 166 KB weight re-read per block stays in
+//   L2).  +8 bf16 row pad: the natural 48/64-dword stride would
+//   multi-way-conflict ds_read_b128; 52/68 dwords pad to 2-way;
 // * B section staged TRANSPOSED in LDS [3W][W+8] so a lane's 8 consecutive
 //   k-elements are contiguous (ds_read_b128);
 // * LayerNorm row stats: 5-step __shfl_xor column reduce per wave -> LDS
-//   partials -> combine across waves; epilogue adds the residual from the
-//   LDS-resident A_1 (= X itself).
+//   partials -> combine across waves; epilogue adds the residual re-read
+//   from global X (hot in L2).
 //
 // Saved for the composed backward: maxout output M, argmax, mu, rstd
 // (backward = layernorm_bwd + maxout scatter + two GEMMs + seq2col_bwd,
@@ -49,8 +54,8 @@ __global__ __launch_bounds__(2 * W) void mwe_layer_fwd_kernel(
   constexpr int WP = W + 8;       // padded LDS row stride (bf16 elements)
   constexpr int NW = W / 32;      // waves per block = within-piece tiles
   extern __shared__ bf16_t lds[];
-  bf16_t* ldsA = lds;                          // 3 * 64 * WP
-  bf16_t* ldsB = lds + 3 * 64 * WP;            // 3W * WP
+  bf16_t* ldsA = lds;                          // 64 * WP (current section)
+  bf16_t* ldsB = lds + 64 * WP;                // 3W * WP (current section)
   float* ldsP = (float*)(ldsB + 3 * W * WP);   // LN partials [2][NW][64]
 
   const int tid = threadIdx.x;
@@ -58,25 +63,6 @@ __global__ __launch_bounds__(2 * W) void mwe_layer_fwd_kernel(
   const int wave = tid / SRX_WAVE;  // 0..NW-1 = within-piece tile position
   const int lane = tid % SRX_WAVE;
   const long t0 = (long)blockIdx.x * 64;
-
-  // ---- stage A sections: A_s[r] = X[t0+r-1+s] (zeroed across doc bounds)
-  for (int row = tid; row < 3 * 64; row += nthreads) {
-    int s = row / 64;
-    int r = row % 64;
-    long t = t0 + r;
-    long src = t + s - 1;
-    bool zero = (s == 0 && (t == 0 || is_start[t])) ||
-                (s == 2 && (t == T - 1 || is_end[t])) || src < 0 || src >= T;
-    bf16_t* dst = ldsA + (s * 64 + r) * WP;
-    const bf16_t* srcp = X + src * W;
-    for (int c = 0; c < W; c += 8) {
-      if (zero) {
-        *(srx_bf16x8*)(dst + c) = srx_bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      } else {
-        *(srx_bf16x8*)(dst + c) = *(const srx_bf16x8*)(srcp + c);
-      }
-    }
-  }
 
   srx_f32x16 acc[3][2];  // [piece][m-tile]
 #pragma unroll
@@ -89,6 +75,22 @@ __global__ __launch_bounds__(2 * W) void mwe_layer_fwd_kernel(
   // ---- K loop: 3 sections x (W/16) MFMA K-steps
   for (int s = 0; s < 3; s++) {
     __syncthreads();
+    // stage A_s: A_s[r] = X[t0+r-1+s] (zeroed across doc bounds)
+    for (int r = tid; r < 64; r += nthreads) {
+      long t = t0 + r;
+      long src = t + s - 1;
+      bool zero = (s == 0 && (t == 0 || is_start[t])) ||
+                  (s == 2 && (t == T - 1 || is_end[t])) || src < 0 || src >= T;
+      bf16_t* dst = ldsA + r * WP;
+      const bf16_t* srcp = X + src * W;
+      for (int c = 0; c < W; c += 8) {
+        if (zero) {
+          *(srx_bf16x8*)(dst + c) = srx_bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        } else {
+          *(srx_bf16x8*)(dst + c) = *(const srx_bf16x8*)(srcp + c);
+        }
+      }
+    }
     // stage B_s transposed: ldsB[j][k] = Wt[j][s*W + k]
     for (int j = tid; j < 3 * W; j += nthreads) {
       const bf16_t* srcp = Wt + (long)j * (3 * W) + s * W;
@@ -105,7 +107,7 @@ __global__ __launch_bounds__(2 * W) void mwe_layer_fwd_kernel(
 #pragma unroll
       for (int mi = 0; mi < 2; mi++) {
         const bf16_t* ap =
-            ldsA + (s * 64 + 32 * mi + (lane & 31)) * WP + k + 8 * (lane >> 5);
+            ldsA + (32 * mi + (lane & 31)) * WP + k + 8 * (lane >> 5);
         afrag[mi] = *(const srx_bf16x8*)ap;
       }
 #pragma unroll
@@ -187,7 +189,7 @@ __global__ __launch_bounds__(2 * W) void mwe_layer_fwd_kernel(
       float v = mx[mi][rr];
       float y = (v - mu) * rstd * bf2f(g[wpcol]) + bf2f(b[wpcol]);
       if (dropmask) y *= bf2f(dropmask[t * W + wpcol]);
-      y += bf2f(ldsA[(64 + r) * WP + wpcol]);  // residual: X row from A_1
+      y += bf2f(X[t * W + wpcol]);  // residual (L2-hot re-read)
       Y[t * W + wpcol] = f2bf(y);
       Mout[t * W + wpcol] = f2bf(v);
       which[t * W + wpcol] = arg[mi][rr];
