@@ -3,8 +3,11 @@
 Usage (on a GPU box):  python tools/kbench.py [names...]
 Names default to all.  Prints ms/call over 20 timed iterations.
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
