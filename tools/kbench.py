@@ -60,7 +60,7 @@ def bench_dpre():
     feats = torch.randint(0, T + 1, (SS, nF), device=dev, dtype=torch.int64)
     dPre = torch.zeros(T + 1, nF, HP, device=dev, dtype=torch.bfloat16)
     timeit("dpre_scatter bf16 (1.9M rows)",
-           lambda: hip.dpre_scatter(dS, feats, dPre))
+           lambda: hip.dpre_scatter(dS, feats, dPre, T))
 
 
 def bench_ce():
