@@ -38,6 +38,45 @@
 namespace {
 
 constexpr float GS_KINV = 1e9f;
+constexpr int GS_HPAD = 72;  // H(<=64) rounded up: 16B-aligned LDS rows,
+                             // zero-padded tail so the dot can run 8-wide
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short gs_bf16x8;
+
+// dot(hid[0..64), wrow[0..64)) — both LDS-resident, H<=64 zero-padded.
+// b128 vector reads: 8 LDS ops for the row instead of 64 scalar ones (the
+// scores phase was the latency bound of the state-machine loop); hid is
+// wave-uniform (same address in every lane -> LDS broadcast).
+template <typename T>
+__device__ __forceinline__ float gs_rowdot64(const T* wrow, const T* hid);
+
+template <>
+__device__ __forceinline__ float gs_rowdot64<bf16_t>(const bf16_t* wrow,
+                                                     const bf16_t* hid) {
+  float acc = 0.f;
+#pragma unroll
+  for (int h0 = 0; h0 < 64; h0 += 8) {
+    gs_bf16x8 wf = *(const gs_bf16x8*)(wrow + h0);
+    gs_bf16x8 hf = *(const gs_bf16x8*)(hid + h0);
+#pragma unroll
+    for (int e = 0; e < 8; e++)
+      acc += bf2f(((const bf16_t*)&wf)[e]) * bf2f(((const bf16_t*)&hf)[e]);
+  }
+  return acc;
+}
+
+template <>
+__device__ __forceinline__ float gs_rowdot64<float>(const float* wrow,
+                                                    const float* hid) {
+  float acc = 0.f;
+#pragma unroll
+  for (int h0 = 0; h0 < 64; h0 += 4) {
+    float4 wf = *(const float4*)(wrow + h0);
+    float4 hf = *(const float4*)(hid + h0);
+    acc += wf.x * hf.x + wf.y * hf.y + wf.z * hf.z + wf.w * hf.w;
+  }
+  return acc;
+}
 
 // ---------------------------------------------------------------- parser
 // Actions: 0=SHIFT, 1=REDUCE, 2..2+L-1=LEFT-ARC(l), 2+L..2+2L-1=RIGHT-ARC(l)
@@ -62,16 +101,18 @@ __global__ __launch_bounds__(256) void gpu_arceager_kernel(
     long n_docs, long pad_row, int nF, int H, int A, int L) {
   extern __shared__ char smem[];
   const int HP = 2 * H;
-  T* Wlds = (T*)smem;                           // [H][A] transposed
-  float* Blds = (float*)(Wlds + (size_t)H * A);  // [A]
+  T* Wlds = (T*)smem;                           // [A][GS_HPAD] row-major
+  float* Blds = (float*)(Wlds + (size_t)A * GS_HPAD);  // [A]
   // per-wave state block
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const int wslot = threadIdx.x / SRX_WAVE;
   const int waves_per_block = blockDim.x / SRX_WAVE;
-  char* wbase = (char*)(Blds + A);
-  // per-wave slot: 7 byte-arrays of MAXLEN + 64 hidden floats + MAXLEN gkb
-  // (host allocates the same: 8*MAXLEN + 256 bytes per wave)
-  const size_t per_wave = (size_t)SRX_GS_MAXLEN * 8 + 64 * 4;
+  // bias region padded to 16 B so every wave slot stays b128-aligned
+  char* wbase = (char*)Blds + (((size_t)A * sizeof(float) + 15) & ~(size_t)15);
+  // per-wave slot: 7 byte-arrays of MAXLEN + GS_HPAD hidden (T) + MAXLEN
+  // gkb (host allocates the same)
+  const size_t per_wave =
+      (size_t)SRX_GS_MAXLEN * 8 + GS_HPAD * sizeof(T);
   char* my = wbase + (size_t)wslot * per_wave;
   uint8_t* stk = (uint8_t*)my;                       // [MAXLEN]
   int8_t* head = (int8_t*)(stk + SRX_GS_MAXLEN);     // [MAXLEN] -1 = none
@@ -80,16 +121,18 @@ __global__ __launch_bounds__(256) void gpu_arceager_kernel(
   int8_t* l2 = l1 + SRX_GS_MAXLEN;
   int8_t* r1 = l2 + SRX_GS_MAXLEN;
   int8_t* r2 = r1 + SRX_GS_MAXLEN;
-  float* my_hid = (float*)(my + (size_t)SRX_GS_MAXLEN * 7);  // [H<=64]
-  uint8_t* gkb = (uint8_t*)(my_hid + 64);            // [MAXLEN]
+  T* my_hid = (T*)(my + (size_t)SRX_GS_MAXLEN * 7);  // [GS_HPAD]
+  uint8_t* gkb = (uint8_t*)my_hid + GS_HPAD * sizeof(T);  // [MAXLEN]
 
-  // block prologue: stage upper weights transposed + bias
-  for (int idx = threadIdx.x; idx < A * H; idx += blockDim.x) {
-    int a = idx / H, h = idx % H;
-    Wlds[(size_t)h * A + a] = upperW[idx];
+  // block prologue: stage upper weights row-major (zero-padded to GS_HPAD
+  // so the scores dot reads b128 vectors) + bias; zero the hidden pad tail
+  for (int idx = threadIdx.x; idx < A * GS_HPAD; idx += blockDim.x) {
+    int a = idx / GS_HPAD, h = idx % GS_HPAD;
+    Wlds[idx] = h < H ? upperW[(size_t)a * H + h] : (T)0;
   }
   for (int a = threadIdx.x; a < A; a += blockDim.x)
     Blds[a] = Elem<T>::ld(upperB + a);
+  for (int h = lane; h < GS_HPAD; h += SRX_WAVE) my_hid[h] = (T)0;
   __syncthreads();
 
   const long wave_id = ((long)blockIdx.x * waves_per_block) + wslot;
@@ -193,7 +236,7 @@ __global__ __launch_bounds__(256) void gpu_arceager_kernel(
         }
         bool second = acc1 > acc0;
         float hv = second ? acc1 : acc0;
-        my_hid[h] = hv;
+        Elem<T>::st(my_hid + h, hv);  // T-rounded: matches the arena copy
         if (TRAIN) {
           Elem<T>::st(hidden_a + arow * H + h, hv);
           which_a[arow * H + h] = (uint8_t)second;
@@ -208,9 +251,7 @@ __global__ __launch_bounds__(256) void gpu_arceager_kernel(
       float bg = -1e38f, bv = -1e38f;
       int ig = INT32_MAX, iv = INT32_MAX;
       for (int a = lane; a < A; a += SRX_WAVE) {
-        float acc = Blds[a];
-        for (int h = 0; h < H; h++)
-          acc += my_hid[h] * Elem<T>::ld(Wlds + (size_t)h * A + a);
+        float acc = Blds[a] + gs_rowdot64<T>(Wlds + (size_t)a * GS_HPAD, my_hid);
         bool valid, gold = false;
         if (a == 0) valid = v_shift;
         else if (a == 1) valid = v_reduce;
@@ -326,19 +367,21 @@ __global__ __launch_bounds__(256) void gpu_biluo_kernel(
     long n_docs, long pad_row, int nF, int H, int A, int NT) {
   extern __shared__ char smem[];
   const int HP = 2 * H;
-  T* Wlds = (T*)smem;
-  float* Blds = (float*)(Wlds + (size_t)H * A);
+  T* Wlds = (T*)smem;                           // [A][GS_HPAD] row-major
+  float* Blds = (float*)(Wlds + (size_t)A * GS_HPAD);
   const int lane = threadIdx.x & (SRX_WAVE - 1);
   const int wslot = threadIdx.x / SRX_WAVE;
   const int waves_per_block = blockDim.x / SRX_WAVE;
-  float* hid_lds = Blds + A;
-  float* my_hid = hid_lds + (size_t)wslot * 64;
-  for (int idx = threadIdx.x; idx < A * H; idx += blockDim.x) {
-    int a = idx / H, h = idx % H;
-    Wlds[(size_t)h * A + a] = upperW[idx];
+  T* hid_lds =
+      (T*)((char*)Blds + (((size_t)A * sizeof(float) + 15) & ~(size_t)15));
+  T* my_hid = hid_lds + (size_t)wslot * GS_HPAD;
+  for (int idx = threadIdx.x; idx < A * GS_HPAD; idx += blockDim.x) {
+    int a = idx / GS_HPAD, h = idx % GS_HPAD;
+    Wlds[idx] = h < H ? upperW[(size_t)a * H + h] : (T)0;
   }
   for (int a = threadIdx.x; a < A; a += blockDim.x)
     Blds[a] = Elem<T>::ld(upperB + a);
+  for (int h = lane; h < GS_HPAD; h += SRX_WAVE) my_hid[h] = (T)0;
   __syncthreads();
 
   const long wave_id = ((long)blockIdx.x * waves_per_block) + wslot;
@@ -366,7 +409,7 @@ __global__ __launch_bounds__(256) void gpu_biluo_kernel(
         }
         bool second = acc1 > acc0;
         float hv = second ? acc1 : acc0;
-        my_hid[h] = hv;
+        Elem<T>::st(my_hid + h, hv);  // T-rounded: matches the arena copy
         if (TRAIN) {
           Elem<T>::st(hidden_a + arow * H + h, hv);
           which_a[arow * H + h] = (uint8_t)second;
@@ -380,9 +423,7 @@ __global__ __launch_bounds__(256) void gpu_biluo_kernel(
       float bg = -1e38f, bv = -1e38f;
       int ig = INT32_MAX, iv = INT32_MAX;
       for (int a = lane; a < A; a += SRX_WAVE) {
-        float acc = Blds[a];
-        for (int h = 0; h < H; h++)
-          acc += my_hid[h] * Elem<T>::ld(Wlds + (size_t)h * A + a);
+        float acc = Blds[a] + gs_rowdot64<T>(Wlds + (size_t)a * GS_HPAD, my_hid);
         bool valid;
         if (open < 0) {
           if (a == 0) valid = true;                       // OUT
@@ -464,7 +505,7 @@ std::vector<at::Tensor> srx_gpu_arceager(
   int A = 2 + 2 * (int)n_labels;
   long pad_row = pre.size(0) - 1;
   TORCH_CHECK(H <= 64, "gpu state machine supports H <= 64");
-  TORCH_CHECK((size_t)H * A * pre.element_size() <= 96 * 1024, "upper too large");
+  TORCH_CHECK((size_t)A * 72 * pre.element_size() <= 96 * 1024, "upper too large");
   auto opt = pre.options();
   auto optb = opt.dtype(at::kByte);
   auto opti = opt.dtype(at::kInt);
@@ -493,9 +534,11 @@ std::vector<at::Tensor> srx_gpu_arceager(
   auto stream = at::cuda::getCurrentCUDAStream();
   const int block = 256;
   const int waves_per_block = block / SRX_WAVE;
-  // per-wave LDS: 7 arrays of MAXLEN bytes + 64 hidden floats + MAXLEN gkb
-  size_t per_wave = (size_t)SRX_GS_MAXLEN * 7 + 64 * 4 + SRX_GS_MAXLEN;
-  size_t lds = (size_t)H * A * pre.element_size() + A * sizeof(float) +
+  // per-wave LDS: 7 arrays of MAXLEN bytes + GS_HPAD hidden (T) + MAXLEN gkb
+  size_t per_wave =
+      (size_t)SRX_GS_MAXLEN * 8 + 72 /*GS_HPAD*/ * pre.element_size();
+  size_t lds = (size_t)A * 72 * pre.element_size() +
+               (((size_t)A * sizeof(float) + 15) & ~(size_t)15) +
                waves_per_block * per_wave;
   int grid = (int)std::min<long>((n_docs + waves_per_block - 1) / waves_per_block,
                                  16384);
@@ -550,7 +593,7 @@ std::vector<at::Tensor> srx_gpu_biluo(
   int A = 1 + 4 * (int)n_types;
   long pad_row = pre.size(0) - 1;
   TORCH_CHECK(H <= 64, "gpu state machine supports H <= 64");
-  TORCH_CHECK((size_t)H * A * pre.element_size() <= 96 * 1024, "upper too large");
+  TORCH_CHECK((size_t)A * 72 * pre.element_size() <= 96 * 1024, "upper too large");
   auto opt = pre.options();
   auto optb = opt.dtype(at::kByte);
   auto opti = opt.dtype(at::kInt);
@@ -577,8 +620,9 @@ std::vector<at::Tensor> srx_gpu_biluo(
   auto stream = at::cuda::getCurrentCUDAStream();
   const int block = 256;
   const int waves_per_block = block / SRX_WAVE;
-  size_t lds = (size_t)H * A * pre.element_size() + A * sizeof(float) +
-               waves_per_block * 64 * sizeof(float);
+  size_t lds = (size_t)A * 72 /*GS_HPAD*/ * pre.element_size() +
+               (((size_t)A * sizeof(float) + 15) & ~(size_t)15) +
+               waves_per_block * 72 * pre.element_size();
   int grid = (int)std::min<long>((n_docs + waves_per_block - 1) / waves_per_block,
                                  16384);
   grid = std::max(grid, 1);
