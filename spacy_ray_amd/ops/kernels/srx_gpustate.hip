@@ -41,12 +41,16 @@ constexpr float GS_KINV = 1e9f;
 constexpr int GS_HPAD = 72;  // H(<=64) rounded up: 16B-aligned LDS rows,
                              // zero-padded tail so the dot can run 8-wide
 
-typedef __attribute__((__vector_size__(8 * sizeof(short)))) short gs_bf16x8;
+typedef __attribute__((__vector_size__(2 * sizeof(__bf16)))) __bf16 gs_bf16x2;
+typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int
+    gs_u32x4;
 
 // dot(hid[0..64), wrow[0..64)) — both LDS-resident, H<=64 zero-padded.
-// b128 vector reads: 8 LDS ops for the row instead of 64 scalar ones (the
-// scores phase was the latency bound of the state-machine loop); hid is
-// wave-uniform (same address in every lane -> LDS broadcast).
+// The scores phase is the ISSUE-bound inner loop of the state machine
+// (~380 of ~700 wave-instructions per transition as a scalar loop): b128
+// vector reads (8 LDS ops per 64-row) + v_dot2c_f32_bf16 (1 instruction
+// per 2 elements) cut it to ~45.  hid is wave-uniform (same address in
+// every lane -> LDS broadcast).
 template <typename T>
 __device__ __forceinline__ float gs_rowdot64(const T* wrow, const T* hid);
 
@@ -56,11 +60,13 @@ __device__ __forceinline__ float gs_rowdot64<bf16_t>(const bf16_t* wrow,
   float acc = 0.f;
 #pragma unroll
   for (int h0 = 0; h0 < 64; h0 += 8) {
-    gs_bf16x8 wf = *(const gs_bf16x8*)(wrow + h0);
-    gs_bf16x8 hf = *(const gs_bf16x8*)(hid + h0);
+    gs_u32x4 wf = *(const gs_u32x4*)(wrow + h0);
+    gs_u32x4 hf = *(const gs_u32x4*)(hid + h0);
 #pragma unroll
-    for (int e = 0; e < 8; e++)
-      acc += bf2f(((const bf16_t*)&wf)[e]) * bf2f(((const bf16_t*)&hf)[e]);
+    for (int e = 0; e < 4; e++)
+      acc = __builtin_amdgcn_fdot2_f32_bf16(
+          __builtin_bit_cast(gs_bf16x2, wf[e]),
+          __builtin_bit_cast(gs_bf16x2, hf[e]), acc, false);
   }
   return acc;
 }
