@@ -440,13 +440,14 @@ class _TransitionLoopLoss(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, pre, lower_b, upperW, upperB, scores, gold, valid,
-                feats, which, hidden):
+                feats, which, hidden, doc_off, doc_lens, cap_mult, maxlen):
         hip = hip_ext()
         loss_count, dScores, colsum = hip.transition_ce(scores, gold, valid,
                                                         deterministic())
         ctx.save_for_backward(dScores, colsum, feats, which, hidden, upperW)
         ctx.pre_shape = tuple(pre.shape)
         ctx.pre_dtype = pre.dtype
+        ctx.doc_layout = (doc_off, doc_lens, cap_mult, maxlen)
         return loss_count[0]
 
     @staticmethod
@@ -460,24 +461,41 @@ class _TransitionLoopLoss(torch.autograd.Function):
         dUpperB = (colsum * g.float()).to(dS.dtype)
         dHidden = dS.mm(upperW)
         dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
-        # dPre accumulates directly in the compute dtype (bf16: packed
-        # atomics, no fp32 buffer + convert); the Zipf-hot pad row and the
-        # bias column-sum are register-accumulated into fp32 side buffers.
-        # Deterministic mode: int64 fixed-point accumulation end to end.
-        acc_dt = torch.int64 if deterministic() else ctx.pre_dtype
-        dPre = torch.zeros(T1, nF, HP, dtype=acc_dt, device=dS.device)
-        dBias32, dPad32 = hip.dpre_scatter(dSummed, feats, dPre, T1 - 1)
-        if acc_dt == torch.int64:
-            dPre = (dPre.to(torch.float32) / FIXED_SCALE).to(ctx.pre_dtype)
+        doc_off, doc_lens, cap_mult, maxlen = ctx.doc_layout
+        if (doc_off is not None and not deterministic()
+                and HP % 64 == 0 and 0 < maxlen and maxlen * HP * 4 <= 65536
+                and hasattr(hip, "dpre_scatter_docmajor")):
+            # GPU-state-machine arenas are doc-major: (token, slot) rows are
+            # exclusive to one (doc, slot) stream, so the scatter runs
+            # atomic-free (LDS per-doc accumulation + plain stores) into an
+            # UNINITIALIZED buffer — the direct kernel below is bound by
+            # packed-atomic op rate, not bandwidth.
+            dPre = torch.empty(T1, nF, HP, dtype=ctx.pre_dtype,
+                               device=dS.device)
+            dBias32, dPad32 = hip.dpre_scatter_docmajor(
+                dSummed, feats, dPre, doc_off, doc_lens, T1 - 1, cap_mult,
+                maxlen)
+        else:
+            # dPre accumulates directly in the compute dtype (bf16: packed
+            # atomics, no fp32 buffer + convert); the Zipf-hot pad row and
+            # the bias column-sum are register-accumulated into fp32 side
+            # buffers.  Deterministic mode: int64 fixed-point end to end.
+            acc_dt = torch.int64 if deterministic() else ctx.pre_dtype
+            dPre = torch.zeros(T1, nF, HP, dtype=acc_dt, device=dS.device)
+            dBias32, dPad32 = hip.dpre_scatter(dSummed, feats, dPre, T1 - 1)
+            if acc_dt == torch.int64:
+                dPre = (dPre.to(torch.float32) / FIXED_SCALE).to(ctx.pre_dtype)
         dPre[T1 - 1] = dPad32.to(dPre.dtype)
-        return (dPre, dBias32.to(dS.dtype), dUpperW,
-                dUpperB, None, None, None, None, None, None)
+        return (dPre, dBias32.to(dS.dtype), dUpperW, dUpperB,
+                None, None, None, None, None, None, None, None, None, None)
 
 
 def transition_loop_loss(pre, lower_b, upperW, upperB, scores, gold, valid,
-                         feats, which, hidden):
+                         feats, which, hidden, doc_off=None, doc_lens=None,
+                         cap_mult=1, maxlen=0):
     return _TransitionLoopLoss.apply(pre, lower_b, upperW, upperB, scores,
-                                     gold, valid, feats, which, hidden)
+                                     gold, valid, feats, which, hidden,
+                                     doc_off, doc_lens, cap_mult, maxlen)
 
 
 class _InjectGrad(torch.autograd.Function):

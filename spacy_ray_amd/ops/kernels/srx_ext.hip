@@ -537,6 +537,44 @@ std::vector<at::Tensor> dpre_scatter(at::Tensor dSummed, at::Tensor feats,
   return {finish(dBias32), finish(dPad32)};
 }
 
+// Doc-major atomic-free variant (GPU-state-machine arenas): one block per
+// doc, LDS accumulation, plain stores into an UNINITIALIZED dPre — see
+// dpre_docmajor_kernel.  Same return contract as dpre_scatter.
+std::vector<at::Tensor> dpre_scatter_docmajor(at::Tensor dSummed,
+                                              at::Tensor feats, at::Tensor dPre,
+                                              at::Tensor off, at::Tensor lens,
+                                              int64_t pad_row, int64_t cap_mult,
+                                              int64_t maxlen) {
+  check_dev(dSummed);
+  TORCH_CHECK(feats.scalar_type() == at::kLong);
+  TORCH_CHECK(dPre.scalar_type() == dSummed.scalar_type(),
+              "docmajor dPre accumulates in the compute dtype");
+  int nF = (int)feats.size(1);
+  int HP = (int)dPre.size(-1);
+  long n_docs = off.size(0);
+  TORCH_CHECK(HP % SRX_WAVE == 0 && HP <= 128, "docmajor needs HP in {64,128}");
+  TORCH_CHECK(nF <= 16, "docmajor nF <= 16");
+  size_t lds = (size_t)maxlen * HP * sizeof(float);
+  TORCH_CHECK(lds <= 64 * 1024, "docmajor maxlen*HP too large for LDS");
+  auto dBias32 = at::zeros({(long)HP}, dSummed.options().dtype(at::kFloat));
+  auto dPad32 =
+      at::zeros({(long)nF, (long)HP}, dSummed.options().dtype(at::kFloat));
+  if (n_docs == 0) return {dBias32, dPad32};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = (int)std::min<long>(n_docs, 65535);
+  DISPATCH_F(dSummed.scalar_type(), {
+    hipLaunchKernelGGL((dpre_docmajor_kernel<scalar_t>), dim3(grid),
+                       dim3(256), lds, stream,
+                       (const scalar_t*)dSummed.data_ptr(),
+                       feats.data_ptr<int64_t>(),
+                       (scalar_t*)dPre.data_ptr(),
+                       dBias32.data_ptr<float>(), dPad32.data_ptr<float>(),
+                       off.data_ptr<int32_t>(), lens.data_ptr<int32_t>(),
+                       n_docs, pad_row, nF, HP, (int)cap_mult);
+  });
+  return {dBias32, dPad32};
+}
+
 // ----------------------------------------------------- dropout mask
 at::Tensor dropout_mask(at::Tensor like, double p, int64_t seed, int64_t offset) {
   check_dev(like);
@@ -899,6 +937,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("valid"), py::arg("deterministic") = false);
   m.attr("FIXED_SCALE") = 16777216.0;
   m.def("dpre_scatter", &dpre_scatter);
+  m.def("dpre_scatter_docmajor", &dpre_scatter_docmajor);
   m.def("run_transition_loop", &srx_run_transition_loop,
         py::call_guard<py::gil_scoped_release>());
   m.def("gpu_arceager", &srx_gpu_arceager, py::arg("pre"), py::arg("off"),

@@ -558,10 +558,15 @@ class _TransitionPipeBase(TrainablePipe):
             from spacy_ray_amd.ops.api import transition_loop_loss
 
             mod = self.module
+            lkw = {}
+            if task.gpu is not None:  # doc-major arenas: atomic-free scatter
+                lkw = dict(doc_off=task.gpu["off"], doc_lens=task.gpu["lens"],
+                           cap_mult=task.gpu["cap_mult"],
+                           maxlen=task.gpu["maxlen"])
             with timing.span("raw/loss_build"):
                 loss = transition_loop_loss(
                     task.pre, mod.lower_b, mod.upper.weight, mod.upper.bias,
-                    scores, gold, valid, feats, which, hidden,
+                    scores, gold, valid, feats, which, hidden, **lkw,
                 )
                 if task.steps_dev is not None:
                     # GPU state machine: arenas are CAPACITY-sized (unused
@@ -857,7 +862,8 @@ class ParserPipe(_TransitionPipeBase):
 
         g = {
             "off": to(off, np.int32), "lens": to(lengths, np.int32),
-            "total": total,
+            "total": total, "cap_mult": 2,
+            "maxlen": int(lengths.max()) if n_docs else 0,
         }
         empty = torch.empty(0, dtype=torch.int32, device=device)
         if staged is None:  # decode: the oracle inputs are never read
@@ -969,7 +975,8 @@ class NerPipe(_TransitionPipeBase):
         gold = (to(staged, np.int32) if staged is not None
                 else torch.empty(0, dtype=torch.int32, device=device))
         return {"off": to(off, np.int32), "lens": to(lengths, np.int32),
-                "total": int(lengths.sum()), "gold": gold}
+                "total": int(lengths.sum()), "gold": gold, "cap_mult": 1,
+                "maxlen": int(lengths.max()) if n_docs else 0}
 
     def _gpu_launch(self, task) -> None:
         g = task.gpu
