@@ -440,14 +440,15 @@ class _TransitionLoopLoss(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, pre, lower_b, upperW, upperB, scores, gold, valid,
-                feats, which, hidden, doc_off, doc_lens, cap_mult, maxlen):
+                feats, which, hidden, doc_off, doc_lens, cap_mult, maxlen,
+                doc_total):
         hip = hip_ext()
         loss_count, dScores, colsum = hip.transition_ce(scores, gold, valid,
                                                         deterministic())
         ctx.save_for_backward(dScores, colsum, feats, which, hidden, upperW)
         ctx.pre_shape = tuple(pre.shape)
         ctx.pre_dtype = pre.dtype
-        ctx.doc_layout = (doc_off, doc_lens, cap_mult, maxlen)
+        ctx.doc_layout = (doc_off, doc_lens, cap_mult, maxlen, doc_total)
         return loss_count[0]
 
     @staticmethod
@@ -461,7 +462,7 @@ class _TransitionLoopLoss(torch.autograd.Function):
         dUpperB = (colsum * g.float()).to(dS.dtype)
         dHidden = dS.mm(upperW)
         dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
-        doc_off, doc_lens, cap_mult, maxlen = ctx.doc_layout
+        doc_off, doc_lens, cap_mult, maxlen, doc_total = ctx.doc_layout
         if (doc_off is not None and not deterministic()
                 and HP % 64 == 0 and 0 < maxlen and maxlen * HP * 4 <= 65536
                 and hasattr(hip, "dpre_scatter_docmajor")):
@@ -472,6 +473,11 @@ class _TransitionLoopLoss(torch.autograd.Function):
             # packed-atomic op rate, not bandwidth.
             dPre = torch.empty(T1, nF, HP, dtype=ctx.pre_dtype,
                                device=dS.device)
+            # the kernel writes every (token, slot) row of the DOCS; batch
+            # PADDING tokens [doc_total, T1-1) get no contributions — zero
+            # them explicitly (the pad row T1-1 is set from dPad below)
+            if doc_total < T1 - 1:
+                dPre[doc_total:T1 - 1].zero_()
             dBias32, dPad32 = hip.dpre_scatter_docmajor(
                 dSummed, feats, dPre, doc_off, doc_lens, T1 - 1, cap_mult,
                 maxlen)
@@ -487,15 +493,17 @@ class _TransitionLoopLoss(torch.autograd.Function):
                 dPre = (dPre.to(torch.float32) / FIXED_SCALE).to(ctx.pre_dtype)
         dPre[T1 - 1] = dPad32.to(dPre.dtype)
         return (dPre, dBias32.to(dS.dtype), dUpperW, dUpperB,
-                None, None, None, None, None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None, None, None,
+                None)
 
 
 def transition_loop_loss(pre, lower_b, upperW, upperB, scores, gold, valid,
                          feats, which, hidden, doc_off=None, doc_lens=None,
-                         cap_mult=1, maxlen=0):
+                         cap_mult=1, maxlen=0, doc_total=0):
     return _TransitionLoopLoss.apply(pre, lower_b, upperW, upperB, scores,
                                      gold, valid, feats, which, hidden,
-                                     doc_off, doc_lens, cap_mult, maxlen)
+                                     doc_off, doc_lens, cap_mult, maxlen,
+                                     doc_total)
 
 
 class _InjectGrad(torch.autograd.Function):

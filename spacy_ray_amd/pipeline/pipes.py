@@ -562,7 +562,8 @@ class _TransitionPipeBase(TrainablePipe):
             if task.gpu is not None:  # doc-major arenas: atomic-free scatter
                 lkw = dict(doc_off=task.gpu["off"], doc_lens=task.gpu["lens"],
                            cap_mult=task.gpu["cap_mult"],
-                           maxlen=task.gpu["maxlen"])
+                           maxlen=task.gpu["maxlen"],
+                           doc_total=task.gpu["total"])
             with timing.span("raw/loss_build"):
                 loss = transition_loop_loss(
                     task.pre, mod.lower_b, mod.upper.weight, mod.upper.bias,
