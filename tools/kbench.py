@@ -55,12 +55,30 @@ def bench_mwe():
 
 
 def bench_dpre():
-    SS, nF, HP, T = 1900032, 13, 128, 1000064
+    import numpy as np
+
+    nF, HP, L = 13, 128, 20
+    n_docs = 50000
+    T = n_docs * L
+    SS = 2 * T
+    rng = np.random.default_rng(0)
+    # doc-major feats: row r of doc d draws from doc d's tokens or pad
+    doc = np.repeat(np.arange(n_docs, dtype=np.int64), 2 * L)
+    base = doc * L
+    f = base[:, None] + rng.integers(0, L, (SS, nF))
+    pad = rng.random((SS, nF)) < 0.25
+    f[pad] = T
+    feats = torch.from_numpy(f).to(dev)
     dS = torch.randn(SS, HP, device=dev, dtype=torch.bfloat16)
-    feats = torch.randint(0, T + 1, (SS, nF), device=dev, dtype=torch.int64)
     dPre = torch.zeros(T + 1, nF, HP, device=dev, dtype=torch.bfloat16)
     timeit("dpre_scatter bf16 (1.9M rows)",
            lambda: hip.dpre_scatter(dS, feats, dPre, T))
+    off = torch.arange(n_docs, device=dev, dtype=torch.int32) * L
+    lens = torch.full((n_docs,), L, device=dev, dtype=torch.int32)
+    dPre2 = torch.empty(T + 1, nF, HP, device=dev, dtype=torch.bfloat16)
+    timeit("dpre_docmajor bf16 (1.9M rows)",
+           lambda: hip.dpre_scatter_docmajor(dS, feats, dPre2, off, lens,
+                                             T, 2, L))
 
 
 def bench_ce():
