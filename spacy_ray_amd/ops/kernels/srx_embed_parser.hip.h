@@ -357,16 +357,20 @@ __global__ __launch_bounds__(256) void dpre_docmajor_kernel(
     float* __restrict__ dPad,           // [nF, HP]
     const int32_t* __restrict__ off, const int32_t* __restrict__ lens,
     long n_docs, long pad_row, int nF, int HP, int cap_mult) {
+  // 2D grid: block (d, q) owns slot q of doc d — enough independent blocks
+  // (n_docs * nF) that each block's short serial row chain hides under
+  // others (the doc-per-block, q-serial version was latency-bound at
+  // 12.8 ms vs the atomic kernel's 5.3; this shape measures by block
+  // parallelism instead)
   extern __shared__ float acc[];  // [maxlen][HP]
   const int tid = threadIdx.x;
   const int lane = tid & (SRX_WAVE - 1);
   const int wslot = tid / SRX_WAVE;
   const int nw = blockDim.x / SRX_WAVE;
   const int ncols = HP / SRX_WAVE;  // HP % 64 == 0, HP <= 128
+  const int q = blockIdx.y;
   float bias_acc[2] = {0.f, 0.f};
-  float pad_acc[16][2];
-  for (int f = 0; f < nF; f++)
-    for (int c = 0; c < 2; c++) pad_acc[f][c] = 0.f;
+  float pad_acc[2] = {0.f, 0.f};
   for (long d = blockIdx.x; d < n_docs; d += gridDim.x) {
     const int n = lens[d];
     const long o = off[d];
@@ -374,38 +378,34 @@ __global__ __launch_bounds__(256) void dpre_docmajor_kernel(
     const int rows = cap_mult * n;
     for (int i = tid; i < n * HP; i += blockDim.x) acc[i] = 0.f;
     __syncthreads();
-    for (int q = 0; q < nF; q++) {
-      for (int r = wslot; r < rows; r += nw) {
-        const long s = rbase + r;
-        const long t = feats[s * (long)nF + q];
-        const T* src = dSummed + s * (long)HP;
-        float x[2];
+    for (int r = wslot; r < rows; r += nw) {
+      const long s = rbase + r;
+      const long t = feats[s * (long)nF + q];
+      const T* src = dSummed + s * (long)HP;
+      float x[2];
+      for (int c = 0; c < ncols; c++)
+        x[c] = Elem<T>::ld(src + lane + c * SRX_WAVE);
+      if (q == 0)  // bias = rowsum over ALL rows; q==0 blocks count once
+        for (int c = 0; c < ncols; c++) bias_acc[c] += x[c];
+      if (t == pad_row) {
+        for (int c = 0; c < ncols; c++) pad_acc[c] += x[c];
+      } else {
+        float* a = acc + (size_t)(t - o) * HP;
         for (int c = 0; c < ncols; c++)
-          x[c] = Elem<T>::ld(src + lane + c * SRX_WAVE);
-        if (q == 0)  // bias = rowsum over ALL rows; count each row once
-          for (int c = 0; c < ncols; c++) bias_acc[c] += x[c];
-        if (t == pad_row) {
-          for (int c = 0; c < ncols; c++) pad_acc[q][c] += x[c];
-        } else {
-          float* a = acc + (size_t)(t - o) * HP;
-          for (int c = 0; c < ncols; c++)
-            atomicAdd(a + lane + c * SRX_WAVE, x[c]);
-        }
+          atomicAdd(a + lane + c * SRX_WAVE, x[c]);
       }
-      __syncthreads();
-      for (int i = tid; i < n * HP; i += blockDim.x) {
-        int local = i / HP, w = i - local * HP;
-        Elem<T>::st(dPre + ((o + local) * (long)nF + q) * HP + w, acc[i]);
-        acc[i] = 0.f;
-      }
-      __syncthreads();
     }
+    __syncthreads();
+    for (int i = tid; i < n * HP; i += blockDim.x) {
+      int local = i / HP, w = i - local * HP;
+      Elem<T>::st(dPre + ((o + local) * (long)nF + q) * HP + w, acc[i]);
+    }
+    __syncthreads();
   }
   for (int c = 0; c < ncols; c++) {
-    if (bias_acc[c] != 0.f)
+    if (q == 0 && bias_acc[c] != 0.f)
       atomicAdd(dBias + lane + c * SRX_WAVE, bias_acc[c]);
-    for (int f = 0; f < nF; f++)
-      if (pad_acc[f][c] != 0.f)
-        atomicAdd(dPad + (long)f * HP + lane + c * SRX_WAVE, pad_acc[f][c]);
+    if (pad_acc[c] != 0.f)
+      atomicAdd(dPad + (long)q * HP + lane + c * SRX_WAVE, pad_acc[c]);
   }
 }

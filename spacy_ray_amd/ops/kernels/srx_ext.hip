@@ -563,8 +563,8 @@ std::vector<at::Tensor> dpre_scatter_docmajor(at::Tensor dSummed,
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = (int)std::min<long>(n_docs, 65535);
   DISPATCH_F(dSummed.scalar_type(), {
-    hipLaunchKernelGGL((dpre_docmajor_kernel<scalar_t>), dim3(grid),
-                       dim3(256), lds, stream,
+    hipLaunchKernelGGL((dpre_docmajor_kernel<scalar_t>),
+                       dim3(grid, (unsigned)nF), dim3(256), lds, stream,
                        (const scalar_t*)dSummed.data_ptr(),
                        feats.data_ptr<int64_t>(),
                        (scalar_t*)dPre.data_ptr(),
