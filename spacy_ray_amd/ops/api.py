@@ -463,7 +463,13 @@ class _TransitionLoopLoss(torch.autograd.Function):
         dHidden = dS.mm(upperW)
         dSummed = hip.maxout_bwd(dHidden.contiguous(), which, 2).view(-1, HP)
         doc_off, doc_lens, cap_mult, maxlen, doc_total = ctx.doc_layout
+        # Doc-major atomic-free scatter: measured SLOWER than the direct
+        # packed-bf16 atomic kernel on MI355X (12.8-23.6 ms vs 5.3 at 1.9M
+        # rows — the per-(doc,slot) LDS accumulation re-reads dSummed up to
+        # nF times and thrashes L2); kept opt-in for long-doc workloads
+        # where per-destination contention would bite the atomics.
         if (doc_off is not None and not deterministic()
+                and os.environ.get("SRX_DPRE_DOCMAJOR", "0") == "1"
                 and HP % 64 == 0 and 0 < maxlen and maxlen * HP * 4 <= 65536
                 and hasattr(hip, "dpre_scatter_docmajor")):
             # GPU-state-machine arenas are doc-major: (token, slot) rows are

@@ -79,6 +79,16 @@ def bench_dpre():
     timeit("dpre_docmajor bf16 (1.9M rows)",
            lambda: hip.dpre_scatter_docmajor(dS, feats, dPre2, off, lens,
                                              T, 2, L))
+    # diagnostics: atomic-op-rate hypothesis — nF=1 should be ~1/13 the
+    # time if op-bound; fp32 (2x the ops of packed bf16) should be ~2x
+    feats1 = feats[:, :1].contiguous()
+    dPre1 = torch.zeros(T + 1, 1, HP, device=dev, dtype=torch.bfloat16)
+    timeit("dpre_scatter bf16 nF=1",
+           lambda: hip.dpre_scatter(dS, feats1, dPre1, T))
+    dS32 = dS.float()
+    dPre32 = torch.zeros(T + 1, nF, HP, device=dev, dtype=torch.float32)
+    timeit("dpre_scatter fp32 (2x ops)",
+           lambda: hip.dpre_scatter(dS32, feats, dPre32, T))
 
 
 def bench_ce():
