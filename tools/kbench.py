@@ -70,6 +70,10 @@ def bench_dpre():
     f[pad] = T
     feats = torch.from_numpy(f).to(dev)
     dS = torch.randn(SS, HP, device=dev, dtype=torch.bfloat16)
+    # realistic sparsity: dSummed comes out of maxout_bwd — exactly one of
+    # each piece pair (h, H+h) is nonzero
+    pick = torch.randint(0, 2, (SS, 1, HP // 2), device=dev)
+    dS.view(SS, 2, HP // 2).scatter_(1, pick, 0.0)
     dPre = torch.zeros(T + 1, nF, HP, device=dev, dtype=torch.bfloat16)
     timeit("dpre_scatter bf16 (1.9M rows)",
            lambda: hip.dpre_scatter(dS, feats, dPre, T))
