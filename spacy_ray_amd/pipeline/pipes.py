@@ -229,10 +229,14 @@ class TaggerPipe(TrainablePipe):
     def predict_and_set(self, docs, t2v, batch) -> None:
         with torch.no_grad():
             pred = self.module(t2v).argmax(dim=-1).cpu().numpy()
+        # vectorized id->tag lookup over the whole batch
+        table = (np.asarray(self.labels, dtype=object) if self.labels
+                 else np.asarray([""], dtype=object))
+        tag_strs = table[np.clip(pred, 0, len(table) - 1)]
         off = 0
         for doc in docs:
             n = len(doc)
-            doc.tags = [self.labels[i] if self.labels else "" for i in pred[off:off + n]]
+            doc.tags = tag_strs[off:off + n].tolist()
             off += n
 
 
@@ -825,17 +829,18 @@ class ParserPipe(_TransitionPipeBase):
         self._annotate_arrays(docs, states.heads(), states.labels(), sents)
 
     def _annotate_arrays(self, docs, heads, labels, sents=None) -> None:
+        # vectorized label-string lookup over the WHOLE batch (the
+        # per-token python loop dominated decode latency at 200k words)
+        L = len(self.labels)
+        table = np.asarray(list(self.labels) + ["ROOT"], dtype=object)
+        labels = np.asarray(labels)
+        safe = np.where((labels >= 0) & (labels < L), labels, L)
+        dep_strs = np.where(np.asarray(heads) == -1, "ROOT", table[safe])
         off = 0
         for doc in docs:
             n = len(doc)
             doc.heads = heads[off:off + n].copy()
-            doc.deps = [
-                self.labels[l] if 0 <= l < len(self.labels) else "ROOT"
-                for l in labels[off:off + n]
-            ]
-            for i in range(n):
-                if doc.heads[i] == -1:
-                    doc.deps[i] = "ROOT"
+            doc.deps = dep_strs[off:off + n].tolist()
             if sents is not None:
                 ss = sents[off:off + n].copy()
                 if n > 0:
@@ -953,10 +958,17 @@ class NerPipe(_TransitionPipeBase):
         self._annotate_tags(docs, states.tags())
 
     def _annotate_tags(self, docs, tags) -> None:
+        # one vectorized code->string lookup for the whole batch
+        from spacy_ray_amd.vocab.doc import biluo_string_table
+
+        table = biluo_string_table(self.labels)
+        tags = np.asarray(tags)
+        safe = np.where((tags > 0) & (tags < len(table)), tags, 0)
+        ent_strs = table[safe]
         off = 0
         for doc in docs:
             n = len(doc)
-            doc.ents = codes_to_biluo(tags[off:off + n], self.labels)
+            doc.ents = ent_strs[off:off + n].tolist()
             off += n
 
     # ---- GPU state machine hooks (srx_gpustate.hip::gpu_biluo_kernel)
