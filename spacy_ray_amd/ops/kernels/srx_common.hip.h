@@ -64,6 +64,68 @@ struct Elem2<bf16_t> {
   }
 };
 
+// ---- true vector load/store of V elements as ONE 16-byte instruction.
+// The per-element Elem<T> loops compile to scalar global_load_ushort
+// streams (hipcc does not merge them): seq2col_bwd measured 56 memory
+// instructions per 8-element item instead of ~6, 8x off its bandwidth
+// floor.  Pointers must be 16B-aligned (W and row strides are multiples
+// of 8 elements everywhere these are used).
+typedef __attribute__((__vector_size__(16))) uint32_t srx_u32x4_t;
+
+template <typename T, int V>
+struct ElemV;
+
+template <>
+struct ElemV<bf16_t, 8> {
+  static __device__ __forceinline__ void ld(const bf16_t* p, float* v) {
+    srx_u32x4_t u = *(const srx_u32x4_t*)p;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      v[2 * i] = bf2f((bf16_t)(u[i] & 0xffffu));
+      v[2 * i + 1] = bf2f((bf16_t)(u[i] >> 16));
+    }
+  }
+  static __device__ __forceinline__ void st(bf16_t* p, const float* v) {
+    srx_u32x4_t u;
+#pragma unroll
+    for (int i = 0; i < 4; i++)
+      u[i] = (uint32_t)f2bf(v[2 * i]) | ((uint32_t)f2bf(v[2 * i + 1]) << 16);
+    *(srx_u32x4_t*)p = u;
+  }
+};
+
+template <>
+struct ElemV<float, 4> {
+  static __device__ __forceinline__ void ld(const float* p, float* v) {
+    float4 x = *(const float4*)p;
+    v[0] = x.x; v[1] = x.y; v[2] = x.z; v[3] = x.w;
+  }
+  static __device__ __forceinline__ void st(float* p, const float* v) {
+    *(float4*)p = make_float4(v[0], v[1], v[2], v[3]);
+  }
+};
+
+// scalar fallbacks so the V=1 tail instantiations compile
+template <>
+struct ElemV<bf16_t, 1> {
+  static __device__ __forceinline__ void ld(const bf16_t* p, float* v) {
+    v[0] = bf2f(*p);
+  }
+  static __device__ __forceinline__ void st(bf16_t* p, const float* v) {
+    *p = f2bf(v[0]);
+  }
+};
+
+template <>
+struct ElemV<float, 1> {
+  static __device__ __forceinline__ void ld(const float* p, float* v) {
+    v[0] = *p;
+  }
+  static __device__ __forceinline__ void st(float* p, const float* v) {
+    *p = v[0];
+  }
+};
+
 // Wave-wide (64-lane) sum reduction; result valid in all lanes.
 __device__ __forceinline__ float wave_reduce_sum(float v) {
 #pragma unroll

@@ -26,14 +26,14 @@ __global__ void seq2col_fwd_kernel(const T* __restrict__ X, T* __restrict__ Y,
     bool zero = (s == 0 && (t == 0 || is_start[t])) ||
                 (s == 2 && (t == nT - 1 || is_end[t]));
     T* out = Y + (t * 3 + s) * (long)W + w;
+    float v[V];
     if (zero || src_t < 0 || src_t >= nT) {
 #pragma unroll
-      for (int k = 0; k < V; k++) Elem<T>::st(out + k, 0.0f);
+      for (int k = 0; k < V; k++) v[k] = 0.0f;
     } else {
-      const T* in = X + src_t * (long)W + w;
-#pragma unroll
-      for (int k = 0; k < V; k++) Elem<T>::st(out + k, Elem<T>::ld(in + k));
+      ElemV<T, V>::ld(X + src_t * (long)W + w, v);
     }
+    ElemV<T, V>::st(out, v);
   }
 }
 
@@ -53,16 +53,24 @@ __global__ void seq2col_bwd_kernel(const T* __restrict__ dY, T* __restrict__ dX,
     long t = i / chunks;
     int w = (int)(i % chunks) * V;
     const long row = 3L * W;
+    float acc[V], tmp[V];
+    ElemV<T, V>::ld(dY + t * row + W + w, acc);
+    if (t + 1 < nT && !is_start[t + 1]) {
+      ElemV<T, V>::ld(dY + (t + 1) * row + w, tmp);
 #pragma unroll
-    for (int k = 0; k < V; k++) {
-      float acc = Elem<T>::ld(dY + t * row + W + w + k);
-      if (t + 1 < nT && !is_start[t + 1])
-        acc += Elem<T>::ld(dY + (t + 1) * row + 0 + w + k);
-      if (t > 0 && !is_end[t - 1])
-        acc += Elem<T>::ld(dY + (t - 1) * row + 2 * W + w + k);
-      if (residual) acc += Elem<T>::ld(residual + t * (long)W + w + k);
-      Elem<T>::st(dX + t * (long)W + w + k, acc);
+      for (int k = 0; k < V; k++) acc[k] += tmp[k];
     }
+    if (t > 0 && !is_end[t - 1]) {
+      ElemV<T, V>::ld(dY + (t - 1) * row + 2 * W + w, tmp);
+#pragma unroll
+      for (int k = 0; k < V; k++) acc[k] += tmp[k];
+    }
+    if (residual) {
+      ElemV<T, V>::ld(residual + t * (long)W + w, tmp);
+#pragma unroll
+      for (int k = 0; k < V; k++) acc[k] += tmp[k];
+    }
+    ElemV<T, V>::st(dX + t * (long)W + w, acc);
   }
 }
 
@@ -77,28 +85,32 @@ __global__ void maxout_fwd_kernel(const T* __restrict__ X, T* __restrict__ Y,
        i += (long)gridDim.x * blockDim.x) {
     long n = i / chunks;
     int w = (int)(i % chunks) * V;
-    float best[V];
+    float best[V], cand[V];
     uint8_t arg[V];
     const T* base = X + n * (long)P * W + w;
+    ElemV<T, V>::ld(base, best);
 #pragma unroll
-    for (int k = 0; k < V; k++) {
-      best[k] = Elem<T>::ld(base + k);
-      arg[k] = 0;
-    }
+    for (int k = 0; k < V; k++) arg[k] = 0;
     for (int p = 1; p < P; p++) {
-      const T* bp = base + (long)p * W;
+      ElemV<T, V>::ld(base + (long)p * W, cand);
 #pragma unroll
       for (int k = 0; k < V; k++) {
-        float v = Elem<T>::ld(bp + k);
-        if (v > best[k]) { best[k] = v; arg[k] = (uint8_t)p; }
+        if (cand[k] > best[k]) { best[k] = cand[k]; arg[k] = (uint8_t)p; }
       }
     }
-    T* out = Y + n * (long)W + w;
+    ElemV<T, V>::st(Y + n * (long)W + w, best);
     uint8_t* wh = which + n * (long)W + w;
+    if (V == 8) {  // one 8-byte store for the argmax bytes
+      uint32_t lo = 0, hi = 0;
 #pragma unroll
-    for (int k = 0; k < V; k++) {
-      Elem<T>::st(out + k, best[k]);
-      wh[k] = arg[k];
+      for (int k = 0; k < 4 && k < V; k++) lo |= (uint32_t)arg[k] << (8 * k);
+#pragma unroll
+      for (int k = 4; k < V; k++) hi |= (uint32_t)arg[k] << (8 * (k - 4));
+      ((uint32_t*)wh)[0] = lo;
+      ((uint32_t*)wh)[1] = hi;
+    } else {
+#pragma unroll
+      for (int k = 0; k < V; k++) wh[k] = arg[k];
     }
   }
 }
@@ -115,12 +127,22 @@ __global__ void maxout_bwd_kernel(const T* __restrict__ dY,
     long rem = i % (P * chunks);
     int p = (int)(rem / chunks);
     int w = (int)(rem % chunks) * V;
-    const T* dy = dY + n * (long)W + w;
     const uint8_t* wh = which + n * (long)W + w;
-    T* out = dX + (n * (long)P + p) * W + w;
+    float v[V];
+    ElemV<T, V>::ld(dY + n * (long)W + w, v);
+    if (V == 8) {
+      uint32_t lo = ((const uint32_t*)wh)[0], hi = ((const uint32_t*)wh)[1];
 #pragma unroll
-    for (int k = 0; k < V; k++)
-      Elem<T>::st(out + k, wh[k] == p ? Elem<T>::ld(dy + k) : 0.0f);
+      for (int k = 0; k < V; k++) {
+        uint8_t a = (uint8_t)(((k < 4 ? lo : hi) >> (8 * (k & 3))) & 0xff);
+        if (a != p) v[k] = 0.0f;
+      }
+    } else {
+#pragma unroll
+      for (int k = 0; k < V; k++)
+        if (wh[k] != p) v[k] = 0.0f;
+    }
+    ElemV<T, V>::st(dX + (n * (long)P + p) * W + w, v);
   }
 }
 
