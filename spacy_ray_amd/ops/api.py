@@ -767,3 +767,61 @@ def reduce_mean_ragged(X, lengths):
 
 def reduce_max_ragged(X, lengths):
     return _ReduceRagged.apply(X, lengths, 2)
+
+
+# ---- elementwise activations (Thinc kernel-surface parity: mish, swish,
+# gelu, clipped_linear and its relu/hard_* family — SURVEY.md §2.2 N1).
+ACT_MISH, ACT_SWISH, ACT_GELU, ACT_CLIPPED_LINEAR = 0, 1, 2, 3
+
+
+class _Activation(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, X, op, slope, offset, lo, hi):
+        ctx.save_for_backward(X)
+        ctx.act = (op, slope, offset, lo, hi)
+        if _want_hip(X):
+            return hip_ext().act_fwd(X, op, slope, offset, lo, hi)
+        return ref.act_forward(X, op, slope, offset, lo, hi)
+
+    @staticmethod
+    def backward(ctx, dY):
+        (X,) = ctx.saved_tensors
+        op, slope, offset, lo, hi = ctx.act
+        if _want_hip(X):
+            dX = hip_ext().act_bwd(dY, X, op, slope, offset, lo, hi)
+        else:
+            dX = ref.act_backward(dY, X, op, slope, offset, lo, hi)
+        return dX, None, None, None, None, None
+
+
+def _act(X, op, slope=1.0, offset=0.0, lo=float("-inf"), hi=float("inf")):
+    return _Activation.apply(X, op, slope, offset, lo, hi)
+
+
+def mish(X: torch.Tensor) -> torch.Tensor:
+    return _act(X, ACT_MISH)
+
+
+def swish(X: torch.Tensor) -> torch.Tensor:
+    return _act(X, ACT_SWISH)
+
+
+def gelu(X: torch.Tensor) -> torch.Tensor:
+    return _act(X, ACT_GELU)
+
+
+def clipped_linear(X, slope=1.0, offset=0.0, min_val=float("-inf"),
+                   max_val=float("inf")):
+    return _act(X, ACT_CLIPPED_LINEAR, slope, offset, min_val, max_val)
+
+
+def relu(X: torch.Tensor) -> torch.Tensor:
+    return clipped_linear(X, 1.0, 0.0, 0.0, float("inf"))
+
+
+def hard_sigmoid(X: torch.Tensor) -> torch.Tensor:
+    return clipped_linear(X, 0.2, 0.5, 0.0, 1.0)
+
+
+def hard_tanh(X: torch.Tensor) -> torch.Tensor:
+    return clipped_linear(X, 1.0, 0.0, -1.0, 1.0)

@@ -15,6 +15,7 @@
 #include "srx_embed_parser.hip.h"
 #include "srx_softmax_reduce.hip.h"
 #include "srx_mwe.hip.h"
+#include "srx_activations.hip.h"
 
 namespace {
 
@@ -575,6 +576,82 @@ std::vector<at::Tensor> dpre_scatter_docmajor(at::Tensor dSummed,
   return {dBias32, dPad32};
 }
 
+// ----------------------------------------------------- activations
+// Thinc's elementwise activation surface (mish/swish/gelu/clipped_linear
+// — srx_activations.hip.h); OP codes shared with ops/api.py.
+template <int OP>
+static void act_launch(bool bwd, const at::Tensor& A, const at::Tensor& B,
+                       at::Tensor& out, double slope, double offset, double lo,
+                       double hi) {
+  long n = out.numel();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_F(out.scalar_type(), {
+    constexpr int V = sizeof(scalar_t) == 2 ? 8 : 4;
+    if (n % V == 0) {
+      long chunks = n / V;
+      if (bwd)
+        hipLaunchKernelGGL((act_bwd_kernel<scalar_t, V, OP>),
+                           dim3(grid_for(chunks)), dim3(kBlock), 0, stream,
+                           (const scalar_t*)A.data_ptr(),
+                           (const scalar_t*)B.data_ptr(),
+                           (scalar_t*)out.data_ptr(), chunks, (float)slope,
+                           (float)offset, (float)lo, (float)hi);
+      else
+        hipLaunchKernelGGL((act_fwd_kernel<scalar_t, V, OP>),
+                           dim3(grid_for(chunks)), dim3(kBlock), 0, stream,
+                           (const scalar_t*)A.data_ptr(),
+                           (scalar_t*)out.data_ptr(), chunks, (float)slope,
+                           (float)offset, (float)lo, (float)hi);
+    } else {
+      if (bwd)
+        hipLaunchKernelGGL((act_bwd_kernel<scalar_t, 1, OP>),
+                           dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                           (const scalar_t*)A.data_ptr(),
+                           (const scalar_t*)B.data_ptr(),
+                           (scalar_t*)out.data_ptr(), n, (float)slope,
+                           (float)offset, (float)lo, (float)hi);
+      else
+        hipLaunchKernelGGL((act_fwd_kernel<scalar_t, 1, OP>),
+                           dim3(grid_for(n)), dim3(kBlock), 0, stream,
+                           (const scalar_t*)A.data_ptr(),
+                           (scalar_t*)out.data_ptr(), n, (float)slope,
+                           (float)offset, (float)lo, (float)hi);
+    }
+  });
+}
+
+at::Tensor act_fwd(at::Tensor X, int64_t op, double slope, double offset,
+                   double lo, double hi) {
+  check_dev(X);
+  auto Xc = X.contiguous();
+  auto Y = at::empty_like(Xc);
+  if (Y.numel() == 0) return Y;
+  at::Tensor dummy;
+  switch (op) {
+    case SRX_ACT_MISH: act_launch<SRX_ACT_MISH>(false, Xc, dummy, Y, slope, offset, lo, hi); break;
+    case SRX_ACT_SWISH: act_launch<SRX_ACT_SWISH>(false, Xc, dummy, Y, slope, offset, lo, hi); break;
+    case SRX_ACT_GELU: act_launch<SRX_ACT_GELU>(false, Xc, dummy, Y, slope, offset, lo, hi); break;
+    default: act_launch<SRX_ACT_CLIPPED_LINEAR>(false, Xc, dummy, Y, slope, offset, lo, hi); break;
+  }
+  return Y;
+}
+
+at::Tensor act_bwd(at::Tensor dY, at::Tensor X, int64_t op, double slope,
+                   double offset, double lo, double hi) {
+  check_dev(dY);
+  auto dYc = dY.contiguous();
+  auto Xc = X.contiguous();
+  auto dX = at::empty_like(Xc);
+  if (dX.numel() == 0) return dX;
+  switch (op) {
+    case SRX_ACT_MISH: act_launch<SRX_ACT_MISH>(true, dYc, Xc, dX, slope, offset, lo, hi); break;
+    case SRX_ACT_SWISH: act_launch<SRX_ACT_SWISH>(true, dYc, Xc, dX, slope, offset, lo, hi); break;
+    case SRX_ACT_GELU: act_launch<SRX_ACT_GELU>(true, dYc, Xc, dX, slope, offset, lo, hi); break;
+    default: act_launch<SRX_ACT_CLIPPED_LINEAR>(true, dYc, Xc, dX, slope, offset, lo, hi); break;
+  }
+  return dX;
+}
+
 // ----------------------------------------------------- dropout mask
 at::Tensor dropout_mask(at::Tensor like, double p, int64_t seed, int64_t offset) {
   check_dev(like);
@@ -926,6 +1003,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seg_scatter_add", &seg_scatter_add);
   m.def("adam_step", &adam_step);
   m.def("dropout_mask", &dropout_mask);
+  m.def("act_fwd", &act_fwd);
+  m.def("act_bwd", &act_bwd);
   m.def("softmax_ce", &softmax_ce);
   m.def("reduce_ragged", &reduce_ragged);
   m.def("reduce_ragged_bwd", &reduce_ragged_bwd);

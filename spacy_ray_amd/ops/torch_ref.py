@@ -157,3 +157,40 @@ def reduce_max_ragged(X: torch.Tensor, lengths: torch.Tensor) -> Tuple[torch.Ten
             which[i] = idx + off
         off += n
     return out, which
+
+
+# ---- activations (Thinc's elementwise surface: mish/swish/gelu/
+# clipped_linear — SURVEY.md §2.2 N1); fp32 torch compositions used as the
+# CPU path and the GPU numerics reference.
+def act_forward(X: torch.Tensor, op: int, slope: float = 1.0,
+                offset: float = 0.0, lo: float = float("-inf"),
+                hi: float = float("inf")) -> torch.Tensor:
+    if op == 0:  # mish
+        return X * torch.tanh(torch.nn.functional.softplus(X))
+    if op == 1:  # swish / silu
+        return X * torch.sigmoid(X)
+    if op == 2:  # gelu (erf form, matching thinc's gaussian gelu)
+        return 0.5 * X * (1.0 + torch.erf(X * 0.7071067811865475))
+    return torch.clamp(slope * X + offset, min=lo, max=hi)
+
+
+def act_backward(dY: torch.Tensor, X: torch.Tensor, op: int,
+                 slope: float = 1.0, offset: float = 0.0,
+                 lo: float = float("-inf"),
+                 hi: float = float("inf")) -> torch.Tensor:
+    if op == 0:
+        sp = torch.nn.functional.softplus(X)
+        t = torch.tanh(sp)
+        sig = torch.sigmoid(X)
+        return dY * (t + X * sig * (1 - t * t))
+    if op == 1:
+        sig = torch.sigmoid(X)
+        return dY * (sig * (1 + X * (1 - sig)))
+    if op == 2:
+        phi = 0.5 * (1.0 + torch.erf(X * 0.7071067811865475))
+        pdf = 0.3989422804014327 * torch.exp(-0.5 * X * X)
+        return dY * (phi + X * pdf)
+    pre = slope * X + offset
+    return dY * torch.where((pre > lo) & (pre < hi),
+                            torch.full_like(X, slope),
+                            torch.zeros_like(X))

@@ -295,3 +295,29 @@ def test_dropout_mask_philox_statistics_and_reproducibility():
     assert torch.allclose(kept, torch.full_like(kept, 1 / (1 - p)), atol=1e-2)
     # the mean of the mask is ~1 (unbiased dropout scaling)
     assert abs(float(m1.float().mean()) - 1.0) < 0.02
+
+
+@need_gpu
+def test_activation_kernels_match_fp32_ref():
+    """act_fwd/act_bwd HIP kernels vs the fp32 torch reference (mish,
+    swish, gelu, clipped_linear), bf16 and fp32, vector and tail paths."""
+    from spacy_ray_amd.ops import api, torch_ref as ref
+
+    hip = api.hip_ext()
+    torch.manual_seed(4)
+    cases = [(0, 1.0, 0.0, float("-inf"), float("inf")),
+             (1, 1.0, 0.0, float("-inf"), float("inf")),
+             (2, 1.0, 0.0, float("-inf"), float("inf")),
+             (3, 0.2, 0.5, 0.0, 1.0)]
+    for n in (4096, 4099):  # vector path and scalar-tail path
+        for dt, tol in ((torch.float32, 1e-5), (torch.bfloat16, 2e-2)):
+            X = torch.randn(n, device="cuda", dtype=dt)
+            dY = torch.randn(n, device="cuda", dtype=dt)
+            for op, slope, offset, lo, hi in cases:
+                y = hip.act_fwd(X, op, slope, offset, lo, hi).float()
+                y_ref = ref.act_forward(X.float(), op, slope, offset, lo, hi)
+                assert torch.allclose(y, y_ref, atol=tol, rtol=tol), (op, dt)
+                dx = hip.act_bwd(dY, X, op, slope, offset, lo, hi).float()
+                dx_ref = ref.act_backward(dY.float(), X.float(), op, slope,
+                                          offset, lo, hi)
+                assert torch.allclose(dx, dx_ref, atol=tol, rtol=tol), (op, dt)

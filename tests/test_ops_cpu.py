@@ -119,3 +119,27 @@ def test_ragged_reductions():
     assert torch.allclose(m[2], X[3:].mean(dim=0))
     mx, which = ref.reduce_max_ragged(X, lengths)
     assert torch.allclose(mx[2], X[3:].max(dim=0).values)
+
+
+def test_activations_match_torch_builtins():
+    """mish/swish/gelu equal torch's own implementations; clipped_linear
+    family obeys its closed form (Thinc kernel-surface parity)."""
+    import torch
+    import torch.nn.functional as F
+
+    from spacy_ray_amd.ops import api
+
+    x = torch.randn(257, 19)
+    assert torch.allclose(api.swish(x), F.silu(x), atol=1e-6)
+    assert torch.allclose(api.mish(x), F.mish(x), atol=1e-6)
+    assert torch.allclose(api.gelu(x), F.gelu(x), atol=1e-5)  # erf form
+    assert torch.allclose(api.relu(x), F.relu(x))
+    assert torch.allclose(api.hard_tanh(x), torch.clamp(x, -1, 1))
+    hs = torch.clamp(0.2 * x + 0.5, 0.0, 1.0)
+    assert torch.allclose(api.hard_sigmoid(x), hs)
+    # backward parity against autograd on the torch builtins
+    xa = x.clone().requires_grad_(True)
+    xb = x.clone().requires_grad_(True)
+    api.mish(xa).sum().backward()
+    F.mish(xb).sum().backward()
+    assert torch.allclose(xa.grad, xb.grad, atol=1e-5)
