@@ -112,6 +112,58 @@ def _fused_output_forward(self, hidden_states, input_tensor):
     return self.LayerNorm(hidden_states + input_tensor)
 
 
+def _srx_attention_interface(module, query, key, value, attention_mask,
+                             dropout: float = 0.0, scaling=None, **kwargs):
+    """transformers AttentionInterface "srx_window": two hipBLASLt bmm
+    GEMMs + the fused masked-softmax(+philox dropout) kernels
+    (ops/kernels/srx_attn.hip.h) — replaces aotriton flash, whose
+    backward measured 5.7x its forward at these window lengths, and the
+    eager math chain (~6 elementwise kernels/call).  Falls back to SDPA
+    off-GPU or for shapes the kernel does not cover."""
+    import os
+
+    from spacy_ray_amd.ops import api as _ops
+
+    B, H, L, D = query.shape
+    if (query.is_cuda and _ops.hip_ext() is not None and L <= 256
+            and key.shape[2] == L
+            and os.environ.get("SRX_ATTN", "1") == "1"):
+        if attention_mask is None:
+            lens = torch.full((B,), L, dtype=torch.int32, device=query.device)
+        else:
+            row = attention_mask[:, 0, 0, :].reshape(B, -1)[:, :L]
+            if row.dtype == torch.bool:
+                valid = row
+            else:
+                valid = row > -1.0  # additive mask: 0 = attend
+            # windows are PREFIX-masked by construction (_run_windows)
+            lens = valid.sum(dim=-1, dtype=torch.int32)
+        scale = scaling if scaling is not None else D ** -0.5
+        out = _ops.window_attention(query, key, value, lens, scale, dropout)
+        return out.transpose(1, 2).contiguous(), None
+    from transformers.integrations.sdpa_attention import sdpa_attention_forward
+
+    return sdpa_attention_forward(module, query, key, value, attention_mask,
+                                  dropout=dropout, scaling=scaling, **kwargs)
+
+
+def _register_srx_attention() -> str:
+    try:
+        from transformers import AttentionInterface
+
+        try:
+            AttentionInterface.register("srx_window", _srx_attention_interface)
+        except Exception:
+            pass  # already registered
+        from transformers.modeling_utils import ALL_ATTENTION_FUNCTIONS
+
+        if ALL_ATTENTION_FUNCTIONS.get_interface("srx_window", None) is None:
+            return "sdpa"
+        return "srx_window"
+    except Exception:
+        return "sdpa"
+
+
 def _srx_optimize_roberta(trf: nn.Module) -> None:
     """Swap LayerNorm/Embedding modules for kernel-backed drop-ins (same
     attribute paths + param names: checkpoints unaffected), and fuse the
@@ -181,7 +233,7 @@ class SubwordBPE:
 class TransformerTok2Vec(nn.Module):
     def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
                  transformer_config: Optional[dict] = None,
-                 attn_implementation: str = "sdpa",
+                 attn_implementation: str = "srx_window",
                  subwords: str = "bpe", bpe_vocab_size: int = 8000,
                  tokenizer_path: Optional[str] = None):
         """attn_implementation: "sdpa" (A/B-measured best at these window
@@ -204,6 +256,8 @@ class TransformerTok2Vec(nn.Module):
                         max_position_embeddings=514)
         base.update(cfg_kwargs)
         config = RobertaConfig(**base)
+        if attn_implementation == "srx_window":
+            attn_implementation = _register_srx_attention()
         config._attn_implementation = attn_implementation
         self.trf = RobertaModel(config, add_pooling_layer=False)
         _srx_optimize_roberta(self.trf)

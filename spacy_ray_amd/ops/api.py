@@ -825,3 +825,49 @@ def hard_sigmoid(X: torch.Tensor) -> torch.Tensor:
 
 def hard_tanh(X: torch.Tensor) -> torch.Tensor:
     return clipped_linear(X, 1.0, 0.0, -1.0, 1.0)
+
+
+# ---- windowed attention: two hipBLASLt bmm GEMMs + the fused masked
+# softmax(+dropout) kernels (srx_attn.hip.h).  Replaces SDPA inside the
+# transformer's short windows: aotriton's flash backward measured 5.7x
+# its forward at L<=96 and the eager math path costs ~6 elementwise
+# kernels per call (profiles/trf262k_r2_kernel_stats.csv).
+class _WindowAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, lens, scale, drop_p):
+        hip = hip_ext()
+        B, H, L, D = q.shape
+        qf = q.reshape(B * H, L, D)
+        kf = k.reshape(B * H, L, D)
+        vf = v.reshape(B * H, L, D)
+        S = torch.bmm(qf, kf.transpose(1, 2))
+        seed = (int(torch.randint(0, 2**62, (1,), device="cpu").item())
+                if drop_p > 0 else 0)
+        P, lse = hip.attn_softmax_fwd(S, lens, H, scale, drop_p, seed)
+        O = torch.bmm(P, vf)
+        ctx.save_for_backward(S, lse, P, qf, kf, vf, lens)
+        ctx.attn_meta = (H, scale, drop_p, seed)
+        return O.view(B, H, L, D)
+
+    @staticmethod
+    def backward(ctx, dO):
+        S, lse, P, qf, kf, vf, lens = ctx.saved_tensors
+        H, scale, drop_p, seed = ctx.attn_meta
+        hip = hip_ext()
+        BH, L, D = qf.shape
+        dOf = dO.reshape(BH, L, D)
+        if not dOf.is_contiguous():
+            dOf = dOf.contiguous()
+        dPt = torch.bmm(dOf, vf.transpose(1, 2))
+        dS = hip.attn_softmax_bwd(S, lse, dPt, lens, H, scale, drop_p, seed)
+        dQ = torch.bmm(dS, kf)
+        dK = torch.bmm(dS.transpose(1, 2), qf)
+        dV = torch.bmm(P.transpose(1, 2), dOf)
+        shp = dO.shape
+        return (dQ.view(shp), dK.view(shp), dV.view(shp), None, None, None)
+
+
+def window_attention(q, k, v, lens, scale: float, drop_p: float = 0.0):
+    """q/k/v: [B, H, L, D]; lens: [B] int32 valid prefix per window.
+    Returns [B, H, L, D]."""
+    return _WindowAttention.apply(q, k, v, lens, scale, drop_p)

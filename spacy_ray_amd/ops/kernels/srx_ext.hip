@@ -16,6 +16,7 @@
 #include "srx_softmax_reduce.hip.h"
 #include "srx_mwe.hip.h"
 #include "srx_activations.hip.h"
+#include "srx_attn.hip.h"
 
 namespace {
 
@@ -652,6 +653,75 @@ at::Tensor act_bwd(at::Tensor dY, at::Tensor X, int64_t op, double slope,
   return dX;
 }
 
+// ----------------------------------------------------- attention softmax
+// Fused masked softmax (+dropout) between the two attention bmm GEMMs —
+// srx_attn.hip.h.  S/P/dS: [NH, L, L]; lens: [NH/heads] int32.
+std::vector<at::Tensor> attn_softmax_fwd(at::Tensor S, at::Tensor lens,
+                                         int64_t heads, double scale,
+                                         double drop_p, int64_t seed) {
+  check_dev(S);
+  long NH = S.size(0);
+  int L = (int)S.size(-1);
+  TORCH_CHECK(L <= SRX_ATTN_MAX_L, "attn softmax: L <= 256");
+  auto P = at::empty_like(S);
+  auto lse = at::empty({NH, (long)L}, S.options().dtype(at::kFloat));
+  long n_rows = NH * L;
+  if (n_rows == 0) return {P, lse};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = grid_for(n_rows * SRX_WAVE);
+  float keep = 1.0f - (float)drop_p;
+  DISPATCH_F(S.scalar_type(), {
+    if (drop_p > 0.0)
+      hipLaunchKernelGGL((attn_softmax_fwd_kernel<scalar_t, true>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)S.data_ptr(), (scalar_t*)P.data_ptr(),
+                         lse.data_ptr<float>(), lens.data_ptr<int32_t>(),
+                         n_rows, L, (int)heads, (float)scale, keep,
+                         (unsigned long long)seed);
+    else
+      hipLaunchKernelGGL((attn_softmax_fwd_kernel<scalar_t, false>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)S.data_ptr(), (scalar_t*)P.data_ptr(),
+                         lse.data_ptr<float>(), lens.data_ptr<int32_t>(),
+                         n_rows, L, (int)heads, (float)scale, keep,
+                         (unsigned long long)seed);
+  });
+  return {P, lse};
+}
+
+at::Tensor attn_softmax_bwd(at::Tensor S, at::Tensor lse, at::Tensor dPt,
+                            at::Tensor lens, int64_t heads, double scale,
+                            double drop_p, int64_t seed) {
+  check_dev(S);
+  long NH = S.size(0);
+  int L = (int)S.size(-1);
+  auto dS = at::empty_like(S);
+  long n_rows = NH * L;
+  if (n_rows == 0) return dS;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = grid_for(n_rows * SRX_WAVE);
+  float keep = 1.0f - (float)drop_p;
+  DISPATCH_F(S.scalar_type(), {
+    if (drop_p > 0.0)
+      hipLaunchKernelGGL((attn_softmax_bwd_kernel<scalar_t, true>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)S.data_ptr(), lse.data_ptr<float>(),
+                         (const scalar_t*)dPt.data_ptr(),
+                         (scalar_t*)dS.data_ptr(), lens.data_ptr<int32_t>(),
+                         n_rows, L, (int)heads, (float)scale, keep,
+                         (unsigned long long)seed);
+    else
+      hipLaunchKernelGGL((attn_softmax_bwd_kernel<scalar_t, false>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const scalar_t*)S.data_ptr(), lse.data_ptr<float>(),
+                         (const scalar_t*)dPt.data_ptr(),
+                         (scalar_t*)dS.data_ptr(), lens.data_ptr<int32_t>(),
+                         n_rows, L, (int)heads, (float)scale, keep,
+                         (unsigned long long)seed);
+  });
+  return dS;
+}
+
 // ----------------------------------------------------- dropout mask
 at::Tensor dropout_mask(at::Tensor like, double p, int64_t seed, int64_t offset) {
   check_dev(like);
@@ -1004,6 +1074,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step);
   m.def("dropout_mask", &dropout_mask);
   m.def("act_fwd", &act_fwd);
+  m.def("attn_softmax_fwd", &attn_softmax_fwd);
+  m.def("attn_softmax_bwd", &attn_softmax_bwd);
   m.def("act_bwd", &act_bwd);
   m.def("softmax_ce", &softmax_ce);
   m.def("reduce_ragged", &reduce_ragged);
