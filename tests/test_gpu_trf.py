@@ -129,29 +129,31 @@ def test_window_attention_matches_sdpa():
         k = torch.randn(B, H, L, D, device="cuda", dtype=dt, requires_grad=True)
         v = torch.randn(B, H, L, D, device="cuda", dtype=dt, requires_grad=True)
         scale = D ** -0.5
+        valid_q = (torch.arange(L, device="cuda")[None, None, :, None] <
+                   lens[:, None, None, None])
         out = api.window_attention(q, k, v, lens, scale, 0.0)
-        dO = torch.randn_like(out)
+        dO = torch.randn_like(out) * valid_q  # no upstream grad on pad rows
         out.backward(dO)
         gq, gk, gv = q.grad.clone(), k.grad.clone(), v.grad.clone()
         q.grad = k.grad = v.grad = None
-        # reference: SDPA math with an explicit bool prefix mask
-        mask = (torch.arange(L, device="cuda")[None, :] <
-                lens[:, None]).view(B, 1, 1, L)
+        # reference mask: valid queries see the prefix; PAD queries attend
+        # position 0 only (well-defined softmax — no NaN rows; with zero
+        # upstream grad they contribute nothing, matching our zeroed rows)
+        j = torch.arange(L, device="cuda")
+        qvalid = (j[:, None] < lens[:, None, None])  # [B, L, 1]
+        kvalid = (j[None, None, :] < lens[:, None, None])  # [B, 1, L]
+        mask2d = torch.where(qvalid, kvalid,
+                             (j[None, None, :] == 0)).unsqueeze(1)
         from torch.nn.attention import SDPBackend, sdpa_kernel
 
         with sdpa_kernel([SDPBackend.MATH]):
             ref = torch.nn.functional.scaled_dot_product_attention(
-                q, k, v, attn_mask=mask, scale=scale)
+                q, k, v, attn_mask=mask2d, scale=scale)
         ref.backward(dO)
-        valid_q = (torch.arange(L, device="cuda")[None, None, :, None] <
-                   lens[:, None, None, None])
-        # compare only valid query rows (padding rows: ours are 0, sdpa NaN)
-        d = ((out - ref).abs() * valid_q).max()
+        d = ((out - ref).abs() * valid_q).max()  # ours: pad query rows = 0
         assert float(d) < tol, (dt, float(d))
         for ours, theirs in ((gq, q.grad), (gk, k.grad), (gv, v.grad)):
-            theirs = torch.nan_to_num(theirs * valid_q, 0.0)
-            ours_m = ours * valid_q if theirs.shape == ours.shape else ours
-            dg = (ours_m - theirs).abs().max()
+            dg = (ours - theirs).abs().max()
             assert float(dg) < tol * 3, (dt, float(dg))
 
 
