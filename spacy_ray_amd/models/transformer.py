@@ -155,6 +155,16 @@ def _register_srx_attention() -> str:
             AttentionInterface.register("srx_window", _srx_attention_interface)
         except Exception:
             pass  # already registered
+        try:
+            # transformers builds the 4D mask through a PER-IMPLEMENTATION
+            # registry; without this, custom interfaces receive mask=None
+            # and padding tokens would attend
+            from transformers import AttentionMaskInterface
+            from transformers.masking_utils import sdpa_mask
+
+            AttentionMaskInterface.register("srx_window", sdpa_mask)
+        except Exception:
+            pass
         from transformers.modeling_utils import ALL_ATTENTION_FUNCTIONS
 
         if ALL_ATTENTION_FUNCTIONS.get_interface("srx_window", None) is None:
