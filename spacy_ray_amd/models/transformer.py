@@ -116,10 +116,12 @@ def _srx_attention_interface(module, query, key, value, attention_mask,
                              dropout: float = 0.0, scaling=None, **kwargs):
     """transformers AttentionInterface "srx_window": two hipBLASLt bmm
     GEMMs + the fused masked-softmax(+philox dropout) kernels
-    (ops/kernels/srx_attn.hip.h) — replaces aotriton flash, whose
-    backward measured 5.7x its forward at these window lengths, and the
-    eager math chain (~6 elementwise kernels/call).  Falls back to SDPA
-    off-GPU or for shapes the kernel does not cover."""
+    (ops/kernels/srx_attn.hip.h).  A/B at 262k words: 236k words/s vs
+    294k for aotriton flash — materializing S/P through HBM costs more
+    than flash's slow backward saves, so flash stays the default and
+    this is the opt-in parity/fallback implementation (select with
+    attn_implementation = "srx_window").  Falls back to SDPA off-GPU or
+    for shapes the kernels do not cover."""
     import os
 
     from spacy_ray_amd.ops import api as _ops
@@ -243,13 +245,16 @@ class SubwordBPE:
 class TransformerTok2Vec(nn.Module):
     def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
                  transformer_config: Optional[dict] = None,
-                 attn_implementation: str = "srx_window",
+                 attn_implementation: str = "sdpa",
                  subwords: str = "bpe", bpe_vocab_size: int = 8000,
                  tokenizer_path: Optional[str] = None):
-        """attn_implementation: "sdpa" (A/B-measured best at these window
-        sizes: eager bmm attention lost ~13% despite aotriton's flash
-        backward being slow at L~22 — the extra elementwise kernels cost
-        more than flash-bwd saves).
+        """attn_implementation: "sdpa" (aotriton flash; A/B-measured best
+        at these window sizes).  "srx_window" = our bmm + fused
+        masked-softmax/philox-dropout kernels (srx_attn.hip.h): measured
+        236k vs 294k words/s end-to-end at 262k words — materializing
+        S/P through HBM costs more than aotriton's slow backward, same
+        lesson as the r2 eager-bmm A/B.  Kept as the parity/fallback
+        implementation and for future true-fused work (docs/ROADMAP.md).
         subwords: "bpe" (real byte-level BPE trained/loaded offline, the
         default) or "hash" (murmur pseudo-subwords, 1 word = 1 position)."""
         super().__init__()
