@@ -833,6 +833,13 @@ def hard_tanh(X: torch.Tensor) -> torch.Tensor:
 # its forward at L<=96 and the eager math path costs ~6 elementwise
 # kernels per call (profiles/trf262k_r2_kernel_stats.csv).
 class _WindowAttention(torch.autograd.Function):
+    """Two execution paths:
+    * FUSED (bf16, D=64, L<=96): the flash-style MFMA kernels
+      (attn_fused_fwd/bwd) — Q/K/V/dO tiles and the S x S probabilities
+      stay in LDS, only lse is saved, the backward recomputes P.
+    * bmm fallback (fp32, D!=64 or L>96): hipBLASLt bmm GEMMs + the
+      fused masked-softmax kernels (S/P materialized in HBM)."""
+
     @staticmethod
     def forward(ctx, q, k, v, lens, scale, drop_p):
         hip = hip_ext()
@@ -840,20 +847,38 @@ class _WindowAttention(torch.autograd.Function):
         qf = q.reshape(B * H, L, D)
         kf = k.reshape(B * H, L, D)
         vf = v.reshape(B * H, L, D)
-        S = torch.bmm(qf, kf.transpose(1, 2))
         seed = (int(torch.randint(0, 2**62, (1,), device="cpu").item())
                 if drop_p > 0 else 0)
+        fused = (L <= 96 and D == 64 and q.dtype == torch.bfloat16
+                 and hasattr(hip, "attn_fused_fwd"))
+        ctx.attn_meta = (H, scale, drop_p, seed, fused)
+        if fused:
+            O, lse = hip.attn_fused_fwd(qf, kf, vf, lens, H, scale, drop_p,
+                                        seed)
+            ctx.save_for_backward(qf, kf, vf, lens, lse)
+            return O.view(B, H, L, D)
+        S = torch.bmm(qf, kf.transpose(1, 2))
         P, lse = hip.attn_softmax_fwd(S, lens, H, scale, drop_p, seed)
         O = torch.bmm(P, vf)
         ctx.save_for_backward(S, lse, P, qf, kf, vf, lens)
-        ctx.attn_meta = (H, scale, drop_p, seed)
         return O.view(B, H, L, D)
 
     @staticmethod
     def backward(ctx, dO):
-        S, lse, P, qf, kf, vf, lens = ctx.saved_tensors
-        H, scale, drop_p, seed = ctx.attn_meta
+        H, scale, drop_p, seed, fused = ctx.attn_meta
         hip = hip_ext()
+        if fused:
+            qf, kf, vf, lens, lse = ctx.saved_tensors
+            BH, L, D = qf.shape
+            dOf = dO.reshape(BH, L, D)
+            if not dOf.is_contiguous():
+                dOf = dOf.contiguous()
+            dQ, dK, dV = hip.attn_fused_bwd(qf, kf, vf, dOf, lse, lens, H,
+                                            scale, drop_p, seed)
+            shp = dO.shape
+            return (dQ.view(shp), dK.view(shp), dV.view(shp),
+                    None, None, None)
+        S, lse, P, qf, kf, vf, lens = ctx.saved_tensors
         BH, L, D = qf.shape
         dOf = dO.reshape(BH, L, D)
         if not dOf.is_contiguous():

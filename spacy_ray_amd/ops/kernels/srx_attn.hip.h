@@ -141,3 +141,365 @@ __global__ void attn_softmax_bwd_kernel(
         Elem<T>::st(drow + j0 + e, scale * p[e] * (dp[e] - delta));
   }
 }
+
+// ===================== fused flash-style attention (MFMA) ==============
+// One workgroup per (window, head); S <= 96, D = 64: Q/K/V (+dO) tiles
+// and the S x S probabilities live ENTIRELY in LDS — no HBM-materialized
+// S/P (the bmm+softmax formulation above measured 236k vs aotriton
+// flash's 294k words/s: the S/P round trips cost more than flash's slow
+// backward).  v_mfma_f32_32x32x16_bf16 tiles, one wave per 32-row
+// M-tile (the mwe_layer_fwd fragment/acc mapping: A row = lane&31,
+// k-half = lane>>5; acc element rr -> row (rr&3)+8*(rr>>2)+4*(lane>>5),
+// col lane&31).  Forward saves ONLY lse; backward recomputes P.
+// Dropout: philox per element (counter (nh*L+i)*L+j), regenerated in
+// the backward — no stored mask.
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short srx_attn_bf16x8;
+typedef __attribute__((__vector_size__(16 * sizeof(float)))) float srx_attn_f32x16;
+
+#define SRX_ATTN_FUSED_MAX_L 96
+#define SRX_ATTN_LDQ 72    // row-major [Lp][72] (D=64 + 16B-align pad)
+#define SRX_ATTN_LDT 104   // transposed / prob tiles [.][104]
+
+__device__ __forceinline__ srx_attn_bf16x8
+srx_attn_frag(const bf16_t* base, int ld, int row, int k0, int lane) {
+  return *(const srx_attn_bf16x8*)(base + (size_t)(row + (lane & 31)) * ld +
+                                   k0 + 8 * (lane >> 5));
+}
+
+// strided fragment: A^T — element e of the fragment comes from
+// src[(k0 + 8*(lane>>5) + e)][row + (lane&31)] of an [.][ld] tile
+__device__ __forceinline__ srx_attn_bf16x8
+srx_attn_frag_t(const bf16_t* base, int ld, int row, int k0, int lane) {
+  union {
+    srx_attn_bf16x8 v;
+    bf16_t u[8];
+  } f;
+  const int k = k0 + 8 * (lane >> 5);
+  const int c = row + (lane & 31);
+#pragma unroll
+  for (int e = 0; e < 8; e++) f.u[e] = base[(size_t)(k + e) * ld + c];
+  return f.v;
+}
+
+__device__ __forceinline__ float srx_attn_row_red_max(float v) {
+#pragma unroll
+  for (int off = 1; off < 32; off <<= 1) v = fmaxf(v, __shfl_xor(v, off, SRX_WAVE));
+  return v;
+}
+
+__device__ __forceinline__ float srx_attn_row_red_sum(float v) {
+#pragma unroll
+  for (int off = 1; off < 32; off <<= 1) v += __shfl_xor(v, off, SRX_WAVE);
+  return v;
+}
+
+__device__ __forceinline__ float srx_attn_drop_u(unsigned long long seed,
+                                                 unsigned long long idx) {
+  hiprandStatePhilox4_32_10_t st;
+  hiprand_init(seed, idx, 0, &st);
+  return hiprand_uniform(&st);
+}
+
+// ------------------------------------------------------------- forward
+// Q/K/V: [NH, L, 64] (compute dtype = bf16 only); O same; lse [NH, L].
+template <bool DROP>
+__global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
+    const bf16_t* __restrict__ V, const int32_t* __restrict__ lens,
+    bf16_t* __restrict__ O, float* __restrict__ lse, long NH, int L,
+    int heads, float scale, float keep, unsigned long long seed) {
+  const int D = 64;
+  const int Lp = (L + 31) & ~31;
+  const int NT = Lp / 32;
+  extern __shared__ char smem[];
+  bf16_t* Qt = (bf16_t*)smem;                       // [Lp][72]
+  bf16_t* Kt = Qt + (size_t)Lp * SRX_ATTN_LDQ;      // [Lp][72]
+  bf16_t* VtT = Kt + (size_t)Lp * SRX_ATTN_LDQ;     // [64][104] (V^T)
+  bf16_t* Pt = VtT + (size_t)64 * SRX_ATTN_LDT;     // [Lp][104]
+  const int tid = threadIdx.x;
+  const int lane = tid & (SRX_WAVE - 1);
+  const int w = tid / SRX_WAVE;  // wave = M-tile
+  const int nthr = blockDim.x;
+  for (long nh = blockIdx.x; nh < NH; nh += gridDim.x) {
+    const int len = lens[nh / heads];
+    const bf16_t* Qg = Q + nh * (size_t)L * D;
+    const bf16_t* Kg = K + nh * (size_t)L * D;
+    const bf16_t* Vg = V + nh * (size_t)L * D;
+    __syncthreads();  // previous iteration's LDS reads complete
+    for (int idx = tid; idx < Lp * D; idx += nthr) {
+      int r = idx / D, d = idx - r * D;
+      bf16_t qv = 0, kv = 0, vv = 0;
+      if (r < L) {
+        qv = Qg[(size_t)r * D + d];
+        kv = Kg[(size_t)r * D + d];
+        vv = Vg[(size_t)r * D + d];
+      }
+      Qt[(size_t)r * SRX_ATTN_LDQ + d] = qv;
+      Kt[(size_t)r * SRX_ATTN_LDQ + d] = kv;
+      VtT[(size_t)d * SRX_ATTN_LDT + r] = vv;
+    }
+    __syncthreads();
+    const int m0 = 32 * w;  // this wave's M-tile row base
+    if (m0 < Lp) {
+      // ---- S = scale * Q@K^T over NT column tiles
+      srx_attn_f32x16 sacc[3];
+#pragma unroll
+      for (int t = 0; t < 3; t++)
+#pragma unroll
+        for (int rr = 0; rr < 16; rr++) sacc[t][rr] = 0.f;
+      for (int t = 0; t < NT; t++)
+#pragma unroll
+        for (int k0 = 0; k0 < 64; k0 += 16) {
+          srx_attn_bf16x8 a = srx_attn_frag(Qt, SRX_ATTN_LDQ, m0, k0, lane);
+          srx_attn_bf16x8 b = srx_attn_frag(Kt, SRX_ATTN_LDQ, 32 * t, k0, lane);
+          sacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, sacc[t], 0, 0, 0);
+        }
+      // ---- rowwise masked softmax across the NT tiles (rows stay
+      // within one 32-lane half: xor<32 reductions)
+      const int col = lane & 31;
+      float p[3][16];
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        float mx = -1e30f;
+        for (int t = 0; t < NT; t++) {
+          int j = 32 * t + col;
+          float s = (j < len && r < len) ? scale * sacc[t][rr] : -1e30f;
+          p[t][rr] = s;
+          mx = fmaxf(mx, s);
+        }
+        mx = srx_attn_row_red_max(mx);
+        float sum = 0.f;
+        for (int t = 0; t < NT; t++) {
+          float e = (p[t][rr] > -1e29f) ? __expf(p[t][rr] - mx) : 0.f;
+          p[t][rr] = e;
+          sum += e;
+        }
+        sum = srx_attn_row_red_sum(sum);
+        float inv = sum > 0.f ? 1.0f / sum : 0.f;
+        for (int t = 0; t < NT; t++) {
+          float v = p[t][rr] * inv;
+          if (DROP && v != 0.f) {
+            int j = 32 * t + col;
+            float u = srx_attn_drop_u(
+                seed, ((unsigned long long)(nh * L + r)) * L + j);
+            v = u < keep ? v / keep : 0.f;
+          }
+          p[t][rr] = v;
+        }
+        if (col == 0 && r < L) lse[nh * (size_t)L + r] = (r < len) ? mx + __logf(sum) : 0.f;
+      }
+      // stage P (dropped+normalized) for the second GEMM
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        for (int t = 0; t < NT; t++)
+          Pt[(size_t)r * SRX_ATTN_LDT + 32 * t + col] = f2bf(p[t][rr]);
+      }
+      // ---- O = P@V (A = Pt rows, B = V^T rows), N = 64 -> 2 tiles
+      srx_attn_f32x16 oacc[2];
+#pragma unroll
+      for (int t = 0; t < 2; t++)
+#pragma unroll
+        for (int rr = 0; rr < 16; rr++) oacc[t][rr] = 0.f;
+      for (int k0 = 0; k0 < Lp; k0 += 16)
+#pragma unroll
+        for (int t = 0; t < 2; t++) {
+          srx_attn_bf16x8 a = srx_attn_frag(Pt, SRX_ATTN_LDT, m0, k0, lane);
+          srx_attn_bf16x8 b = srx_attn_frag(VtT, SRX_ATTN_LDT, 32 * t, k0, lane);
+          oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, oacc[t], 0, 0, 0);
+        }
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        if (r < L) {
+          for (int t = 0; t < 2; t++)
+            O[nh * (size_t)L * D + (size_t)r * D + 32 * t + col] =
+                f2bf(r < len ? oacc[t][rr] : 0.f);
+        }
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------ backward
+// Recomputes P from (Q, K, lse); LDS holds row-major Q/K/V/dO, transposed
+// Q^T/K^T/dO^T (for the dQ/dK/dV GEMMs' B operands) and the P / dS
+// matrices.  GEMM chain per (window, head):
+//   (1) S = Q@K^T -> P (regs) and P~ -> Pt (LDS, dropout applied)
+//   (2) dP~ = dO@V^T;  dP = dP~*mask;  delta = rowsum(P*dP)
+//       dS = scale*P*(dP - delta) -> DSt (LDS)
+//   (3) dQ = dS@K      (B = K^T)
+//   (4) dK = dS^T@Q    (A = DSt strided, B = Q^T)
+//   (5) dV = P~^T@dO   (A = Pt strided,  B = dO^T)
+template <bool DROP>
+__global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
+    const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
+    const float* __restrict__ lse, const int32_t* __restrict__ lens,
+    bf16_t* __restrict__ dQ, bf16_t* __restrict__ dK,
+    bf16_t* __restrict__ dV, long NH, int L, int heads, float scale,
+    float keep, unsigned long long seed) {
+  const int D = 64;
+  const int Lp = (L + 31) & ~31;
+  const int NT = Lp / 32;
+  extern __shared__ char smem[];
+  bf16_t* Qt = (bf16_t*)smem;                        // [Lp][72]
+  bf16_t* Kt = Qt + (size_t)Lp * SRX_ATTN_LDQ;       // [Lp][72]
+  bf16_t* Vt = Kt + (size_t)Lp * SRX_ATTN_LDQ;       // [Lp][72]
+  bf16_t* dOt = Vt + (size_t)Lp * SRX_ATTN_LDQ;      // [Lp][72]
+  bf16_t* QtT = dOt + (size_t)Lp * SRX_ATTN_LDQ;     // [64][104]
+  bf16_t* KtT = QtT + (size_t)64 * SRX_ATTN_LDT;     // [64][104]
+  bf16_t* dOtT = KtT + (size_t)64 * SRX_ATTN_LDT;    // [64][104]
+  bf16_t* Pt = dOtT + (size_t)64 * SRX_ATTN_LDT;     // [Lp][104]
+  bf16_t* DSt = Pt + (size_t)Lp * SRX_ATTN_LDT;      // [Lp][104]
+  const int tid = threadIdx.x;
+  const int lane = tid & (SRX_WAVE - 1);
+  const int w = tid / SRX_WAVE;
+  const int nthr = blockDim.x;
+  for (long nh = blockIdx.x; nh < NH; nh += gridDim.x) {
+    const int len = lens[nh / heads];
+    const size_t gbase = nh * (size_t)L * D;
+    __syncthreads();
+    for (int idx = tid; idx < Lp * D; idx += nthr) {
+      int r = idx / D, d = idx - r * D;
+      bf16_t qv = 0, kv = 0, vv = 0, dv = 0;
+      if (r < L) {
+        qv = Q[gbase + (size_t)r * D + d];
+        kv = K[gbase + (size_t)r * D + d];
+        vv = V[gbase + (size_t)r * D + d];
+        dv = dO[gbase + (size_t)r * D + d];
+      }
+      Qt[(size_t)r * SRX_ATTN_LDQ + d] = qv;
+      Kt[(size_t)r * SRX_ATTN_LDQ + d] = kv;
+      Vt[(size_t)r * SRX_ATTN_LDQ + d] = vv;
+      dOt[(size_t)r * SRX_ATTN_LDQ + d] = dv;
+      QtT[(size_t)d * SRX_ATTN_LDT + r] = qv;
+      KtT[(size_t)d * SRX_ATTN_LDT + r] = kv;
+      dOtT[(size_t)d * SRX_ATTN_LDT + r] = dv;
+    }
+    __syncthreads();
+    const int m0 = 32 * w;
+    const int col = lane & 31;
+    if (m0 < Lp) {
+      // ---- (1) recompute P rows of this M-tile
+      srx_attn_f32x16 sacc[3];
+#pragma unroll
+      for (int t = 0; t < 3; t++)
+#pragma unroll
+        for (int rr = 0; rr < 16; rr++) sacc[t][rr] = 0.f;
+      for (int t = 0; t < NT; t++)
+#pragma unroll
+        for (int k0 = 0; k0 < 64; k0 += 16) {
+          srx_attn_bf16x8 a = srx_attn_frag(Qt, SRX_ATTN_LDQ, m0, k0, lane);
+          srx_attn_bf16x8 b = srx_attn_frag(Kt, SRX_ATTN_LDQ, 32 * t, k0, lane);
+          sacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, sacc[t], 0, 0, 0);
+        }
+      float p[3][16], pd[3][16];  // undropped P; dropped P~
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        const float l = (r < len) ? lse[nh * (size_t)L + r] : 0.f;
+        for (int t = 0; t < NT; t++) {
+          int j = 32 * t + col;
+          float v = (r < len && j < len) ? __expf(scale * sacc[t][rr] - l) : 0.f;
+          p[t][rr] = v;
+          if (DROP && v != 0.f) {
+            float u = srx_attn_drop_u(
+                seed, ((unsigned long long)(nh * L + r)) * L + j);
+            v = u < keep ? v / keep : 0.f;
+          }
+          pd[t][rr] = v;
+          Pt[(size_t)r * SRX_ATTN_LDT + j] = f2bf(v);
+        }
+      }
+      // ---- (2) dP~ = dO@V^T; dS
+      srx_attn_f32x16 dpacc[3];
+#pragma unroll
+      for (int t = 0; t < 3; t++)
+#pragma unroll
+        for (int rr = 0; rr < 16; rr++) dpacc[t][rr] = 0.f;
+      for (int t = 0; t < NT; t++)
+#pragma unroll
+        for (int k0 = 0; k0 < 64; k0 += 16) {
+          srx_attn_bf16x8 a = srx_attn_frag(dOt, SRX_ATTN_LDQ, m0, k0, lane);
+          srx_attn_bf16x8 b = srx_attn_frag(Vt, SRX_ATTN_LDQ, 32 * t, k0, lane);
+          dpacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, dpacc[t], 0, 0, 0);
+        }
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        float delta = 0.f;
+        for (int t = 0; t < NT; t++) {
+          float dp = dpacc[t][rr];
+          if (DROP) {
+            int j = 32 * t + col;
+            if (p[t][rr] != 0.f) {
+              float u = srx_attn_drop_u(
+                  seed, ((unsigned long long)(nh * L + r)) * L + j);
+              dp = u < keep ? dp / keep : 0.f;
+            } else {
+              dp = 0.f;
+            }
+          }
+          dpacc[t][rr] = dp;
+          delta += p[t][rr] * dp;
+        }
+        delta = srx_attn_row_red_sum(delta);
+        for (int t = 0; t < NT; t++) {
+          float ds = scale * p[t][rr] * (dpacc[t][rr] - delta);
+          DSt[(size_t)r * SRX_ATTN_LDT + 32 * t + col] = f2bf(ds);
+        }
+      }
+    }
+    __syncthreads();  // Pt / DSt complete across all waves
+    if (m0 < Lp) {
+      // ---- (3) dQ = dS@K  (A = DSt rows, B = K^T rows)
+      srx_attn_f32x16 qacc[2];
+#pragma unroll
+      for (int t = 0; t < 2; t++)
+#pragma unroll
+        for (int rr = 0; rr < 16; rr++) qacc[t][rr] = 0.f;
+      for (int k0 = 0; k0 < Lp; k0 += 16)
+#pragma unroll
+        for (int t = 0; t < 2; t++) {
+          srx_attn_bf16x8 a = srx_attn_frag(DSt, SRX_ATTN_LDT, m0, k0, lane);
+          srx_attn_bf16x8 b = srx_attn_frag(KtT, SRX_ATTN_LDT, 32 * t, k0, lane);
+          qacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, qacc[t], 0, 0, 0);
+        }
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        if (r < L)
+          for (int t = 0; t < 2; t++)
+            dQ[gbase + (size_t)r * D + 32 * t + col] = f2bf(qacc[t][rr]);
+      }
+      // ---- (4) dK = dS^T@Q  and (5) dV = P~^T@dO  (j-tile = this wave)
+      srx_attn_f32x16 kacc[2], vacc[2];
+#pragma unroll
+      for (int t = 0; t < 2; t++)
+#pragma unroll
+        for (int rr = 0; rr < 16; rr++) { kacc[t][rr] = 0.f; vacc[t][rr] = 0.f; }
+      for (int k0 = 0; k0 < Lp; k0 += 16) {
+        srx_attn_bf16x8 at_ds = srx_attn_frag_t(DSt, SRX_ATTN_LDT, m0, k0, lane);
+        srx_attn_bf16x8 at_p = srx_attn_frag_t(Pt, SRX_ATTN_LDT, m0, k0, lane);
+#pragma unroll
+        for (int t = 0; t < 2; t++) {
+          srx_attn_bf16x8 bq = srx_attn_frag(QtT, SRX_ATTN_LDT, 32 * t, k0, lane);
+          srx_attn_bf16x8 bo = srx_attn_frag(dOtT, SRX_ATTN_LDT, 32 * t, k0, lane);
+          kacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(at_ds, bq, kacc[t], 0, 0, 0);
+          vacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(at_p, bo, vacc[t], 0, 0, 0);
+        }
+      }
+#pragma unroll
+      for (int rr = 0; rr < 16; rr++) {
+        const int j = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+        if (j < L)
+          for (int t = 0; t < 2; t++) {
+            dK[gbase + (size_t)j * D + 32 * t + col] = f2bf(kacc[t][rr]);
+            dV[gbase + (size_t)j * D + 32 * t + col] = f2bf(vacc[t][rr]);
+          }
+      }
+    }
+  }
+}

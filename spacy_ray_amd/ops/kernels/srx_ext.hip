@@ -722,6 +722,90 @@ at::Tensor attn_softmax_bwd(at::Tensor S, at::Tensor lse, at::Tensor dPt,
   return dS;
 }
 
+// ------------------------------------------- fused flash-style attention
+std::vector<at::Tensor> attn_fused_fwd(at::Tensor Q, at::Tensor K,
+                                       at::Tensor V, at::Tensor lens,
+                                       int64_t heads, double scale,
+                                       double drop_p, int64_t seed) {
+  check_dev(Q);
+  TORCH_CHECK(Q.scalar_type() == at::kBFloat16, "fused attention is bf16-only");
+  long NH = Q.size(0);
+  int L = (int)Q.size(1);
+  TORCH_CHECK((int)Q.size(2) == 64, "fused attention needs D == 64");
+  TORCH_CHECK(L <= SRX_ATTN_FUSED_MAX_L, "fused attention: L <= 96");
+  auto O = at::empty_like(Q);
+  auto lse = at::empty({NH, (long)L}, Q.options().dtype(at::kFloat));
+  if (NH == 0) return {O, lse};
+  int Lp = (L + 31) & ~31;
+  int NT = Lp / 32;
+  size_t lds =
+      ((size_t)2 * Lp * SRX_ATTN_LDQ + 64 * SRX_ATTN_LDT + (size_t)Lp * SRX_ATTN_LDT) * 2;
+  static std::once_flag attr_f;
+  std::call_once(attr_f, []() {
+    (void)hipFuncSetAttribute((const void*)attn_fused_fwd_kernel<true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              100 * 1024);
+    (void)hipFuncSetAttribute((const void*)attn_fused_fwd_kernel<false>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              100 * 1024);
+  });
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = (int)std::min<long>(NH, 65535);
+  float keep = 1.0f - (float)drop_p;
+#define LAUNCH_AF(DR)                                                         \
+  hipLaunchKernelGGL((attn_fused_fwd_kernel<DR>), dim3(grid),                 \
+                     dim3(64 * NT), lds, stream,                              \
+                     (const bf16_t*)Q.data_ptr(), (const bf16_t*)K.data_ptr(),\
+                     (const bf16_t*)V.data_ptr(), lens.data_ptr<int32_t>(),   \
+                     (bf16_t*)O.data_ptr(), lse.data_ptr<float>(), NH, L,     \
+                     (int)heads, (float)scale, keep, (unsigned long long)seed)
+  if (drop_p > 0.0) LAUNCH_AF(true); else LAUNCH_AF(false);
+#undef LAUNCH_AF
+  return {O, lse};
+}
+
+std::vector<at::Tensor> attn_fused_bwd(at::Tensor Q, at::Tensor K,
+                                       at::Tensor V, at::Tensor dO,
+                                       at::Tensor lse, at::Tensor lens,
+                                       int64_t heads, double scale,
+                                       double drop_p, int64_t seed) {
+  check_dev(Q);
+  long NH = Q.size(0);
+  int L = (int)Q.size(1);
+  auto dQ = at::empty_like(Q);
+  auto dK = at::empty_like(Q);
+  auto dV = at::empty_like(Q);
+  if (NH == 0) return {dQ, dK, dV};
+  int Lp = (L + 31) & ~31;
+  int NT = Lp / 32;
+  size_t lds = ((size_t)4 * Lp * SRX_ATTN_LDQ + (size_t)3 * 64 * SRX_ATTN_LDT +
+                (size_t)2 * Lp * SRX_ATTN_LDT) * 2;
+  static std::once_flag attr_b;
+  std::call_once(attr_b, []() {
+    (void)hipFuncSetAttribute((const void*)attn_fused_bwd_kernel<true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              144 * 1024);
+    (void)hipFuncSetAttribute((const void*)attn_fused_bwd_kernel<false>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              144 * 1024);
+  });
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = (int)std::min<long>(NH, 65535);
+  float keep = 1.0f - (float)drop_p;
+#define LAUNCH_AB(DR)                                                         \
+  hipLaunchKernelGGL((attn_fused_bwd_kernel<DR>), dim3(grid),                 \
+                     dim3(64 * NT), lds, stream,                              \
+                     (const bf16_t*)Q.data_ptr(), (const bf16_t*)K.data_ptr(),\
+                     (const bf16_t*)V.data_ptr(), (const bf16_t*)dO.data_ptr(),\
+                     lse.data_ptr<float>(), lens.data_ptr<int32_t>(),         \
+                     (bf16_t*)dQ.data_ptr(), (bf16_t*)dK.data_ptr(),          \
+                     (bf16_t*)dV.data_ptr(), NH, L, (int)heads,               \
+                     (float)scale, keep, (unsigned long long)seed)
+  if (drop_p > 0.0) LAUNCH_AB(true); else LAUNCH_AB(false);
+#undef LAUNCH_AB
+  return {dQ, dK, dV};
+}
+
 // ----------------------------------------------------- dropout mask
 at::Tensor dropout_mask(at::Tensor like, double p, int64_t seed, int64_t offset) {
   check_dev(like);
@@ -1075,6 +1159,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_mask", &dropout_mask);
   m.def("act_fwd", &act_fwd);
   m.def("attn_softmax_fwd", &attn_softmax_fwd);
+  m.def("attn_fused_fwd", &attn_fused_fwd);
+  m.def("attn_fused_bwd", &attn_fused_bwd);
   m.def("attn_softmax_bwd", &attn_softmax_bwd);
   m.def("act_bwd", &act_bwd);
   m.def("softmax_ce", &softmax_ce);

@@ -166,8 +166,8 @@ def test_window_attention_dropout_statistics():
     torch.manual_seed(1)
     B, H, L, D = 4, 4, 64, 64
     lens = torch.full((B,), L, dtype=torch.int32, device="cuda")
-    q = torch.randn(B, H, L, D, device="cuda", dtype=torch.float32,
-                    requires_grad=True)
+    q = torch.randn(B, H, L, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)  # bf16 -> exercises the FUSED path
     k = torch.randn_like(q, requires_grad=True)
     v = torch.randn_like(q, requires_grad=True)
     out = api.window_attention(q, k, v, lens, D ** -0.5, 0.3)
