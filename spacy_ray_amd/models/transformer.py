@@ -245,7 +245,7 @@ class SubwordBPE:
 class TransformerTok2Vec(nn.Module):
     def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
                  transformer_config: Optional[dict] = None,
-                 attn_implementation: str = "sdpa",
+                 attn_implementation: str = "srx_window",
                  subwords: str = "bpe", bpe_vocab_size: int = 8000,
                  tokenizer_path: Optional[str] = None):
         """attn_implementation: "sdpa" (aotriton flash; A/B-measured best
