@@ -200,6 +200,37 @@ __device__ __forceinline__ float srx_attn_drop_u(unsigned long long seed,
   return hiprand_uniform(&st);
 }
 
+// Batched dropout multipliers for the MFMA acc layout: a lane's 16 acc
+// elements per tile are 4 groups of 4 CONSECUTIVE rows at one column, so
+// ONE philox draw (uniform4, counter (nh*L + j)*ceil(L/4) + r0/4) covers
+// a group — 12 inits per lane instead of 48/96 (per-element philox made
+// the first fused version SLOWER than aotriton under training dropout).
+// dmul[t][rr] = 0 (dropped) or 1/keep; identical in fwd and bwd.
+template <int MAXT>
+__device__ __forceinline__ void srx_attn_drop_mul(
+    float dmul[MAXT][16], unsigned long long seed, long nh, int L, int NT,
+    int m0, int lane, float keep) {
+  const int col = lane & 31;
+  const int nq4 = (L + 3) >> 2;
+  const float inv_keep = 1.0f / keep;
+#pragma unroll
+  for (int g = 0; g < 4; g++) {
+    const int r0 = m0 + 8 * g + 4 * (lane >> 5);
+    for (int t = 0; t < NT; t++) {
+      const int j = 32 * t + col;
+      hiprandStatePhilox4_32_10_t st;
+      hiprand_init(seed,
+                   ((unsigned long long)(nh * (long)L + j)) * nq4 + (r0 >> 2),
+                   0, &st);
+      float4 u = hiprand_uniform4(&st);
+      const float uu[4] = {u.x, u.y, u.z, u.w};
+#pragma unroll
+      for (int c = 0; c < 4; c++)
+        dmul[t][(g << 2) | c] = uu[c] < keep ? inv_keep : 0.f;
+    }
+  }
+}
+
 // ------------------------------------------------------------- forward
 // Q/K/V: [NH, L, 64] (compute dtype = bf16 only); O same; lse [NH, L].
 template <bool DROP>
@@ -258,6 +289,9 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
       // within one 32-lane half: xor<32 reductions)
       const int col = lane & 31;
       float p[3][16];
+      float dmul[3][16];
+      if (DROP)
+        srx_attn_drop_mul<3>(dmul, seed, nh, L, NT, m0, lane, keep);
 #pragma unroll
       for (int rr = 0; rr < 16; rr++) {
         const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
@@ -279,12 +313,7 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
         float inv = sum > 0.f ? 1.0f / sum : 0.f;
         for (int t = 0; t < NT; t++) {
           float v = p[t][rr] * inv;
-          if (DROP && v != 0.f) {
-            int j = 32 * t + col;
-            float u = srx_attn_drop_u(
-                seed, ((unsigned long long)(nh * L + r)) * L + j);
-            v = u < keep ? v / keep : 0.f;
-          }
+          if (DROP) v *= dmul[t][rr];
           p[t][rr] = v;
         }
         if (col == 0 && r < L) lse[nh * (size_t)L + r] = (r < len) ? mx + __logf(sum) : 0.f;
@@ -395,7 +424,10 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
           srx_attn_bf16x8 b = srx_attn_frag(Kt, SRX_ATTN_LDQ, 32 * t, k0, lane);
           sacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, sacc[t], 0, 0, 0);
         }
-      float p[3][16], pd[3][16];  // undropped P; dropped P~
+      float p[3][16];    // undropped P
+      float dmul[3][16];  // dropout multipliers (regenerated, == fwd's)
+      if (DROP)
+        srx_attn_drop_mul<3>(dmul, seed, nh, L, NT, m0, lane, keep);
 #pragma unroll
       for (int rr = 0; rr < 16; rr++) {
         const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
@@ -404,13 +436,8 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
           int j = 32 * t + col;
           float v = (r < len && j < len) ? __expf(scale * sacc[t][rr] - l) : 0.f;
           p[t][rr] = v;
-          if (DROP && v != 0.f) {
-            float u = srx_attn_drop_u(
-                seed, ((unsigned long long)(nh * L + r)) * L + j);
-            v = u < keep ? v / keep : 0.f;
-          }
-          pd[t][rr] = v;
-          Pt[(size_t)r * SRX_ATTN_LDT + j] = f2bf(v);
+          if (DROP) v *= dmul[t][rr];
+          Pt[(size_t)r * SRX_ATTN_LDT + j] = f2bf(v);  // P~ (for dV)
         }
       }
       // ---- (2) dP~ = dO@V^T; dS
@@ -432,16 +459,7 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
         float delta = 0.f;
         for (int t = 0; t < NT; t++) {
           float dp = dpacc[t][rr];
-          if (DROP) {
-            int j = 32 * t + col;
-            if (p[t][rr] != 0.f) {
-              float u = srx_attn_drop_u(
-                  seed, ((unsigned long long)(nh * L + r)) * L + j);
-              dp = u < keep ? dp / keep : 0.f;
-            } else {
-              dp = 0.f;
-            }
-          }
+          if (DROP) dp *= dmul[t][rr];
           dpacc[t][rr] = dp;
           delta += p[t][rr] * dp;
         }
