@@ -206,9 +206,9 @@ __device__ __forceinline__ float srx_attn_drop_u(unsigned long long seed,
 // a group — 12 inits per lane instead of 48/96 (per-element philox made
 // the first fused version SLOWER than aotriton under training dropout).
 // dmul[t][rr] = 0 (dropped) or 1/keep; identical in fwd and bwd.
-template <int MAXT>
+template <int NT>
 __device__ __forceinline__ void srx_attn_drop_mul(
-    float dmul[MAXT][16], unsigned long long seed, long nh, int L, int NT,
+    float dmul[NT][16], unsigned long long seed, long nh, int L,
     int m0, int lane, float keep) {
   const int col = lane & 31;
   const int nq4 = (L + 3) >> 2;
@@ -216,6 +216,7 @@ __device__ __forceinline__ void srx_attn_drop_mul(
 #pragma unroll
   for (int g = 0; g < 4; g++) {
     const int r0 = m0 + 8 * g + 4 * (lane >> 5);
+#pragma unroll
     for (int t = 0; t < NT; t++) {
       const int j = 32 * t + col;
       hiprandStatePhilox4_32_10_t st;
@@ -233,15 +234,16 @@ __device__ __forceinline__ void srx_attn_drop_mul(
 
 // ------------------------------------------------------------- forward
 // Q/K/V: [NH, L, 64] (compute dtype = bf16 only); O same; lse [NH, L].
-template <bool DROP>
+template <bool DROP, int NT>
 __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const int32_t* __restrict__ lens,
     bf16_t* __restrict__ O, float* __restrict__ lse, long NH, int L,
     int heads, float scale, float keep, unsigned long long seed) {
   const int D = 64;
-  const int Lp = (L + 31) & ~31;
-  const int NT = Lp / 32;
+  constexpr int Lp = 32 * NT;  // compile-time: keeps p/dmul/acc arrays in
+                               // REGISTERS (runtime-NT indexing lowered
+                               // them to 640 B/lane of global scratch)
   extern __shared__ char smem[];
   bf16_t* Qt = (bf16_t*)smem;                       // [Lp][72]
   bf16_t* Kt = Qt + (size_t)Lp * SRX_ATTN_LDQ;      // [Lp][72]
@@ -273,11 +275,12 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
     const int m0 = 32 * w;  // this wave's M-tile row base
     if (m0 < Lp) {
       // ---- S = scale * Q@K^T over NT column tiles
-      srx_attn_f32x16 sacc[3];
+      srx_attn_f32x16 sacc[NT];
 #pragma unroll
-      for (int t = 0; t < 3; t++)
+      for (int t = 0; t < NT; t++)
 #pragma unroll
         for (int rr = 0; rr < 16; rr++) sacc[t][rr] = 0.f;
+#pragma unroll
       for (int t = 0; t < NT; t++)
 #pragma unroll
         for (int k0 = 0; k0 < 64; k0 += 16) {
@@ -288,14 +291,15 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
       // ---- rowwise masked softmax across the NT tiles (rows stay
       // within one 32-lane half: xor<32 reductions)
       const int col = lane & 31;
-      float p[3][16];
-      float dmul[3][16];
+      float p[NT][16];
+      float dmul[DROP ? NT : 1][16];
       if (DROP)
-        srx_attn_drop_mul<3>(dmul, seed, nh, L, NT, m0, lane, keep);
+        srx_attn_drop_mul<DROP ? NT : 1>(dmul, seed, nh, L, m0, lane, keep);
 #pragma unroll
       for (int rr = 0; rr < 16; rr++) {
         const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
         float mx = -1e30f;
+#pragma unroll
         for (int t = 0; t < NT; t++) {
           int j = 32 * t + col;
           float s = (j < len && r < len) ? scale * sacc[t][rr] : -1e30f;
@@ -304,6 +308,7 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
         }
         mx = srx_attn_row_red_max(mx);
         float sum = 0.f;
+#pragma unroll
         for (int t = 0; t < NT; t++) {
           float e = (p[t][rr] > -1e29f) ? __expf(p[t][rr] - mx) : 0.f;
           p[t][rr] = e;
@@ -311,9 +316,10 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
         }
         sum = srx_attn_row_red_sum(sum);
         float inv = sum > 0.f ? 1.0f / sum : 0.f;
+#pragma unroll
         for (int t = 0; t < NT; t++) {
           float v = p[t][rr] * inv;
-          if (DROP) v *= dmul[t][rr];
+          if (DROP) v *= dmul[DROP ? t : 0][rr];
           p[t][rr] = v;
         }
         if (col == 0 && r < L) lse[nh * (size_t)L + r] = (r < len) ? mx + __logf(sum) : 0.f;
@@ -322,6 +328,7 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
 #pragma unroll
       for (int rr = 0; rr < 16; rr++) {
         const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
+#pragma unroll
         for (int t = 0; t < NT; t++)
           Pt[(size_t)r * SRX_ATTN_LDT + 32 * t + col] = f2bf(p[t][rr]);
       }
@@ -361,7 +368,7 @@ __global__ __launch_bounds__(192) void attn_fused_fwd_kernel(
 //   (3) dQ = dS@K      (B = K^T)
 //   (4) dK = dS^T@Q    (A = DSt strided, B = Q^T)
 //   (5) dV = P~^T@dO   (A = Pt strided,  B = dO^T)
-template <bool DROP>
+template <bool DROP, int NT>
 __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ K,
     const bf16_t* __restrict__ V, const bf16_t* __restrict__ dO,
@@ -370,8 +377,7 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
     bf16_t* __restrict__ dV, long NH, int L, int heads, float scale,
     float keep, unsigned long long seed) {
   const int D = 64;
-  const int Lp = (L + 31) & ~31;
-  const int NT = Lp / 32;
+  constexpr int Lp = 32 * NT;
   extern __shared__ char smem[];
   bf16_t* Qt = (bf16_t*)smem;                        // [Lp][72]
   bf16_t* Kt = Qt + (size_t)Lp * SRX_ATTN_LDQ;       // [Lp][72]
@@ -412,11 +418,12 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
     const int col = lane & 31;
     if (m0 < Lp) {
       // ---- (1) recompute P rows of this M-tile
-      srx_attn_f32x16 sacc[3];
+      srx_attn_f32x16 sacc[NT];
 #pragma unroll
-      for (int t = 0; t < 3; t++)
+      for (int t = 0; t < NT; t++)
 #pragma unroll
         for (int rr = 0; rr < 16; rr++) sacc[t][rr] = 0.f;
+#pragma unroll
       for (int t = 0; t < NT; t++)
 #pragma unroll
         for (int k0 = 0; k0 < 64; k0 += 16) {
@@ -424,28 +431,30 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
           srx_attn_bf16x8 b = srx_attn_frag(Kt, SRX_ATTN_LDQ, 32 * t, k0, lane);
           sacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, sacc[t], 0, 0, 0);
         }
-      float p[3][16];    // undropped P
-      float dmul[3][16];  // dropout multipliers (regenerated, == fwd's)
+      float p[NT][16];    // undropped P
+      float dmul[DROP ? NT : 1][16];  // dropout multipliers (== fwd's)
       if (DROP)
-        srx_attn_drop_mul<3>(dmul, seed, nh, L, NT, m0, lane, keep);
+        srx_attn_drop_mul<DROP ? NT : 1>(dmul, seed, nh, L, m0, lane, keep);
 #pragma unroll
       for (int rr = 0; rr < 16; rr++) {
         const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
         const float l = (r < len) ? lse[nh * (size_t)L + r] : 0.f;
+#pragma unroll
         for (int t = 0; t < NT; t++) {
           int j = 32 * t + col;
           float v = (r < len && j < len) ? __expf(scale * sacc[t][rr] - l) : 0.f;
           p[t][rr] = v;
-          if (DROP) v *= dmul[t][rr];
+          if (DROP) v *= dmul[DROP ? t : 0][rr];
           Pt[(size_t)r * SRX_ATTN_LDT + j] = f2bf(v);  // P~ (for dV)
         }
       }
       // ---- (2) dP~ = dO@V^T; dS
-      srx_attn_f32x16 dpacc[3];
+      srx_attn_f32x16 dpacc[NT];
 #pragma unroll
-      for (int t = 0; t < 3; t++)
+      for (int t = 0; t < NT; t++)
 #pragma unroll
         for (int rr = 0; rr < 16; rr++) dpacc[t][rr] = 0.f;
+#pragma unroll
       for (int t = 0; t < NT; t++)
 #pragma unroll
         for (int k0 = 0; k0 < 64; k0 += 16) {
@@ -457,13 +466,15 @@ __global__ __launch_bounds__(192) void attn_fused_bwd_kernel(
       for (int rr = 0; rr < 16; rr++) {
         const int r = m0 + (rr & 3) + 8 * (rr >> 2) + 4 * (lane >> 5);
         float delta = 0.f;
+#pragma unroll
         for (int t = 0; t < NT; t++) {
           float dp = dpacc[t][rr];
-          if (DROP) dp *= dmul[t][rr];
+          if (DROP) dp *= dmul[DROP ? t : 0][rr];
           dpacc[t][rr] = dp;
           delta += p[t][rr] * dp;
         }
         delta = srx_attn_row_red_sum(delta);
+#pragma unroll
         for (int t = 0; t < NT; t++) {
           float ds = scale * p[t][rr] * (dpacc[t][rr] - delta);
           DSt[(size_t)r * SRX_ATTN_LDT + 32 * t + col] = f2bf(ds);

@@ -742,24 +742,29 @@ std::vector<at::Tensor> attn_fused_fwd(at::Tensor Q, at::Tensor K,
       ((size_t)2 * Lp * SRX_ATTN_LDQ + 64 * SRX_ATTN_LDT + (size_t)Lp * SRX_ATTN_LDT) * 2;
   static std::once_flag attr_f;
   std::call_once(attr_f, []() {
-    (void)hipFuncSetAttribute((const void*)attn_fused_fwd_kernel<true>,
-                              hipFuncAttributeMaxDynamicSharedMemorySize,
-                              100 * 1024);
-    (void)hipFuncSetAttribute((const void*)attn_fused_fwd_kernel<false>,
-                              hipFuncAttributeMaxDynamicSharedMemorySize,
-                              100 * 1024);
+#define AF_ATTR(DR, NTV)                                                      \
+    (void)hipFuncSetAttribute((const void*)attn_fused_fwd_kernel<DR, NTV>,    \
+                              hipFuncAttributeMaxDynamicSharedMemorySize,     \
+                              100 * 1024)
+    AF_ATTR(true, 1); AF_ATTR(true, 2); AF_ATTR(true, 3);
+    AF_ATTR(false, 1); AF_ATTR(false, 2); AF_ATTR(false, 3);
+#undef AF_ATTR
   });
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = (int)std::min<long>(NH, 65535);
   float keep = 1.0f - (float)drop_p;
-#define LAUNCH_AF(DR)                                                         \
-  hipLaunchKernelGGL((attn_fused_fwd_kernel<DR>), dim3(grid),                 \
-                     dim3(64 * NT), lds, stream,                              \
+#define LAUNCH_AF(DR, NTV)                                                    \
+  hipLaunchKernelGGL((attn_fused_fwd_kernel<DR, NTV>), dim3(grid),            \
+                     dim3(64 * NTV), lds, stream,                             \
                      (const bf16_t*)Q.data_ptr(), (const bf16_t*)K.data_ptr(),\
                      (const bf16_t*)V.data_ptr(), lens.data_ptr<int32_t>(),   \
                      (bf16_t*)O.data_ptr(), lse.data_ptr<float>(), NH, L,     \
                      (int)heads, (float)scale, keep, (unsigned long long)seed)
-  if (drop_p > 0.0) LAUNCH_AF(true); else LAUNCH_AF(false);
+  if (drop_p > 0.0) {
+    if (NT == 1) LAUNCH_AF(true, 1); else if (NT == 2) LAUNCH_AF(true, 2); else LAUNCH_AF(true, 3);
+  } else {
+    if (NT == 1) LAUNCH_AF(false, 1); else if (NT == 2) LAUNCH_AF(false, 2); else LAUNCH_AF(false, 3);
+  }
 #undef LAUNCH_AF
   return {O, lse};
 }
@@ -782,26 +787,31 @@ std::vector<at::Tensor> attn_fused_bwd(at::Tensor Q, at::Tensor K,
                 (size_t)2 * Lp * SRX_ATTN_LDT) * 2;
   static std::once_flag attr_b;
   std::call_once(attr_b, []() {
-    (void)hipFuncSetAttribute((const void*)attn_fused_bwd_kernel<true>,
-                              hipFuncAttributeMaxDynamicSharedMemorySize,
-                              144 * 1024);
-    (void)hipFuncSetAttribute((const void*)attn_fused_bwd_kernel<false>,
-                              hipFuncAttributeMaxDynamicSharedMemorySize,
-                              144 * 1024);
+#define AB_ATTR(DR, NTV)                                                      \
+    (void)hipFuncSetAttribute((const void*)attn_fused_bwd_kernel<DR, NTV>,    \
+                              hipFuncAttributeMaxDynamicSharedMemorySize,     \
+                              144 * 1024)
+    AB_ATTR(true, 1); AB_ATTR(true, 2); AB_ATTR(true, 3);
+    AB_ATTR(false, 1); AB_ATTR(false, 2); AB_ATTR(false, 3);
+#undef AB_ATTR
   });
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = (int)std::min<long>(NH, 65535);
   float keep = 1.0f - (float)drop_p;
-#define LAUNCH_AB(DR)                                                         \
-  hipLaunchKernelGGL((attn_fused_bwd_kernel<DR>), dim3(grid),                 \
-                     dim3(64 * NT), lds, stream,                              \
+#define LAUNCH_AB(DR, NTV)                                                    \
+  hipLaunchKernelGGL((attn_fused_bwd_kernel<DR, NTV>), dim3(grid),            \
+                     dim3(64 * NTV), lds, stream,                             \
                      (const bf16_t*)Q.data_ptr(), (const bf16_t*)K.data_ptr(),\
                      (const bf16_t*)V.data_ptr(), (const bf16_t*)dO.data_ptr(),\
                      lse.data_ptr<float>(), lens.data_ptr<int32_t>(),         \
                      (bf16_t*)dQ.data_ptr(), (bf16_t*)dK.data_ptr(),          \
                      (bf16_t*)dV.data_ptr(), NH, L, (int)heads,               \
                      (float)scale, keep, (unsigned long long)seed)
-  if (drop_p > 0.0) LAUNCH_AB(true); else LAUNCH_AB(false);
+  if (drop_p > 0.0) {
+    if (NT == 1) LAUNCH_AB(true, 1); else if (NT == 2) LAUNCH_AB(true, 2); else LAUNCH_AB(true, 3);
+  } else {
+    if (NT == 1) LAUNCH_AB(false, 1); else if (NT == 2) LAUNCH_AB(false, 2); else LAUNCH_AB(false, 3);
+  }
 #undef LAUNCH_AB
   return {dQ, dK, dV};
 }
