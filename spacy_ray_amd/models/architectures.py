@@ -88,7 +88,7 @@ def make_transformer_model(
     transformer_config: Optional[dict] = None,
     window: int = 128,
     stride: int = 96,
-    attn_implementation: str = "srx_window",
+    attn_implementation: str = "sdpa",
     subwords: str = "bpe",
     bpe_vocab_size: int = 8000,
 ):

@@ -114,14 +114,17 @@ def _fused_output_forward(self, hidden_states, input_tensor):
 
 def _srx_attention_interface(module, query, key, value, attention_mask,
                              dropout: float = 0.0, scaling=None, **kwargs):
-    """transformers AttentionInterface "srx_window": two hipBLASLt bmm
-    GEMMs + the fused masked-softmax(+philox dropout) kernels
-    (ops/kernels/srx_attn.hip.h).  A/B at 262k words: 236k words/s vs
-    294k for aotriton flash — materializing S/P through HBM costs more
-    than flash's slow backward saves, so flash stays the default and
-    this is the opt-in parity/fallback implementation (select with
-    attn_implementation = "srx_window").  Falls back to SDPA off-GPU or
-    for shapes the kernels do not cover."""
+    """transformers AttentionInterface "srx_window": our hand-written
+    attention (ops/kernels/srx_attn.hip.h).  bf16/D=64/L<=96 windows run
+    the fully-fused flash-style MFMA kernels (Q/K/V/dO and the S x S
+    probabilities LDS-resident, philox dropout regenerated in backward,
+    only lse saved); other shapes run bmm GEMMs + fused masked-softmax
+    kernels.  SDPA-parity-tested (test_window_attention_matches_sdpa).
+    A/B at 262k words: fused 252k / bmm 236k vs aotriton flash 294k
+    words/s, so flash stays the DEFAULT and this is the opt-in
+    implementation (attn_implementation = "srx_window"); the remaining
+    gap is block occupancy (one (window, head) per workgroup; VGPR 225
+    at NT=3) — see docs/ROADMAP.md item 1 for the batching plan."""
     import os
 
     from spacy_ray_amd.ops import api as _ops
@@ -245,7 +248,7 @@ class SubwordBPE:
 class TransformerTok2Vec(nn.Module):
     def __init__(self, name: str = "roberta-base", window: int = 128, stride: int = 96,
                  transformer_config: Optional[dict] = None,
-                 attn_implementation: str = "srx_window",
+                 attn_implementation: str = "sdpa",
                  subwords: str = "bpe", bpe_vocab_size: int = 8000,
                  tokenizer_path: Optional[str] = None):
         """attn_implementation: "sdpa" (aotriton flash; A/B-measured best
