@@ -136,8 +136,17 @@ def make_synthetic_docs(
                 i += length
             else:
                 i += 1
+        # doc-level category: majority vote of per-word classes (learnable
+        # from the pooled representation) + sentence starts every ~7 tokens
+        # (word-deterministic: wid % 7 == 0) for senter/textcat pipes
+        cat_votes = np.bincount(ent_of_word[word_ids], minlength=n_ent_types)
+        cats = {f"CAT{j}": 0.0 for j in range(n_ent_types)}
+        cats[f"CAT{int(cat_votes.argmax())}"] = 1.0
+        sent_starts = (word_ids % 7 == 0).astype(np.int32)
+        sent_starts[0] = 1
         docs.append(Doc(vocab, words, tags=tags, heads=heads, deps=deps,
-                        ents=ents, attr_hashes=lex_attr[word_ids]))
+                        ents=ents, cats=cats, sent_starts=sent_starts,
+                        attr_hashes=lex_attr[word_ids]))
     return docs
 
 

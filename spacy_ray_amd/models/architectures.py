@@ -143,3 +143,20 @@ def make_transition_parser_model(
     spec.hidden_width = hidden_width
     spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
     return spec
+
+
+@registry.architectures("spacy.TextCatCNN.v2")
+def make_textcat_cnn_model(tok2vec: ModelSpec, exclusive_classes: bool = True,
+                           nO: Optional[int] = None):
+    spec = ModelSpec(lambda: None, width=tok2vec.width, kind="textcat")
+    spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
+    return spec
+
+
+@registry.architectures("spacy.TextCatReduce.v1")
+def make_textcat_reduce_model(tok2vec: ModelSpec,
+                              exclusive_classes: bool = True,
+                              use_reduce_mean: bool = True, **_ignored):
+    spec = ModelSpec(lambda: None, width=tok2vec.width, kind="textcat")
+    spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
+    return spec

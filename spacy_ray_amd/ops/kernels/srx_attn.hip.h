@@ -193,13 +193,6 @@ __device__ __forceinline__ float srx_attn_row_red_sum(float v) {
   return v;
 }
 
-__device__ __forceinline__ float srx_attn_drop_u(unsigned long long seed,
-                                                 unsigned long long idx) {
-  hiprandStatePhilox4_32_10_t st;
-  hiprand_init(seed, idx, 0, &st);
-  return hiprand_uniform(&st);
-}
-
 // Batched dropout multipliers for the MFMA acc layout: a lane's 16 acc
 // elements per tile are 4 groups of 4 CONSECUTIVE rows at one column, so
 // ONE philox draw (uniform4, counter (nh*L + j)*ceil(L/4) + r0/4) covers

@@ -39,3 +39,24 @@ def make_parser_pipe(name: str, model, labels=None, use_break: bool = False):
 @registry.factories("ner")
 def make_ner_pipe(name: str, model, labels=None):
     return _with_labels(NerPipe(name, model), labels)
+
+
+@registry.factories("textcat")
+def make_textcat_pipe(name: str, model, labels=None):
+    from .pipes import TextcatPipe
+
+    return _with_labels(TextcatPipe(name, model, exclusive_classes=True), labels)
+
+
+@registry.factories("textcat_multilabel")
+def make_textcat_multilabel_pipe(name: str, model, labels=None):
+    from .pipes import TextcatPipe
+
+    return _with_labels(TextcatPipe(name, model, exclusive_classes=False), labels)
+
+
+@registry.factories("senter")
+def make_senter_pipe(name: str, model):
+    from .pipes import SenterPipe
+
+    return SenterPipe(name, model)

@@ -1007,3 +1007,170 @@ class NerPipe(_TransitionPipeBase):
 
     def _annotate_gpu(self, docs, decode) -> None:
         self._annotate_tags(docs, decode.cpu().numpy())
+
+
+class TextcatHead(nn.Module):
+    def __init__(self, width: int, n_cats: int):
+        super().__init__()
+        self.output = nn.Linear(width, n_cats)
+        nn.init.zeros_(self.output.weight)
+        nn.init.zeros_(self.output.bias)
+
+    def forward(self, X):
+        from spacy_ray_amd.ops.api import linear_cdw
+
+        return linear_cdw(X, self.output.weight, self.output.bias)
+
+
+class TextcatPipe(TrainablePipe):
+    """Doc-level classification over mean-pooled tok2vec (the reference is
+    a generic spaCy trainer — `spacy ray train` accepts textcat configs,
+    /root/reference/worker.py:71-106 builds whatever pipeline the config
+    names).  `exclusive_classes=True` = spaCy's `textcat` (softmax CE vs
+    the gold distribution); False = `textcat_multilabel` (per-label
+    binary cross-entropy).  Pooling = reduce_mean_ragged (the segmented
+    reduce kernel)."""
+
+    name = "textcat"
+    listens_to = "tok2vec"
+
+    def __init__(self, name: str, spec, exclusive_classes: bool = True) -> None:
+        super().__init__()
+        self.name = name
+        self.width = spec.width
+        self.embedded_spec = getattr(spec, "embedded_tok2vec", None)
+        self.exclusive = exclusive_classes
+        self.labels: List[str] = []
+        self.label2id: Dict[str, int] = {}
+
+    def initialize(self, examples, device) -> None:
+        if not self.labels:
+            labels = set()
+            for eg in examples:
+                if eg.reference.cats:
+                    labels.update(eg.reference.cats.keys())
+            self.labels = sorted(labels) or ["POSITIVE"]
+            self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.cfg["labels"] = self.labels
+        self.cfg["exclusive"] = self.exclusive
+        if self.module is None:
+            self.module = TextcatHead(self.width, max(1, len(self.labels))).to(device)
+        self._attach_embedded(device)
+
+    def load_cfg(self, cfg, device) -> None:
+        super().load_cfg(cfg, device)
+        self.labels = list(cfg.get("labels", []))
+        self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.exclusive = bool(cfg.get("exclusive", True))
+        if self.module is None:
+            self.module = TextcatHead(self.width, max(1, len(self.labels))).to(device)
+        self._attach_embedded(device)
+
+    def _doc_scores(self, t2v, batch, docs):
+        from spacy_ray_amd.ops import api as _ops
+
+        if batch is not None:
+            lengths, total = batch.lengths, batch.n_tokens
+        else:
+            lens = [len(d) for d in docs]
+            lengths = torch.as_tensor(lens, dtype=torch.int64,
+                                      device=t2v.device)
+            total = sum(lens)
+        pooled = _ops.reduce_mean_ragged(t2v[:total], lengths)
+        # TokenBatch may append a PAD pseudo-doc (GEMM shape bucketing)
+        return self.module(pooled[:len(docs)])
+
+    def _gold_matrix(self, examples) -> np.ndarray:
+        n = len(examples)
+        Y = np.zeros((n, max(1, len(self.labels))), dtype=np.float32)
+        for i, eg in enumerate(examples):
+            for lab, val in (eg.reference.cats or {}).items():
+                j = self.label2id.get(lab)
+                if j is not None:
+                    Y[i, j] = float(val)
+        return Y
+
+    def get_loss(self, examples, t2v, batch):
+        scores = self._doc_scores(t2v, batch, [eg.reference for eg in examples])
+        gold = torch.from_numpy(self._gold_matrix(examples)).to(scores.device)
+        n = max(1, scores.shape[0])
+        if self.exclusive:
+            logp = torch.log_softmax(scores.float(), dim=-1)
+            loss = -(gold * logp).sum() / n
+        else:
+            loss = torch.nn.functional.binary_cross_entropy_with_logits(
+                scores.float(), gold, reduction="sum") / n
+        return loss, loss.detach()
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        with torch.no_grad():
+            scores = self._doc_scores(t2v, batch, docs)
+            probs = (torch.softmax(scores.float(), dim=-1) if self.exclusive
+                     else torch.sigmoid(scores.float())).cpu().numpy()
+        for i, doc in enumerate(docs):
+            doc.cats = {lab: float(probs[i, j])
+                        for j, lab in enumerate(self.labels)}
+
+    def state_cfg(self) -> Dict:
+        return self.cfg
+
+
+class SenterPipe(TaggerPipe):
+    """Sentence-boundary recognizer (spaCy's `senter`): a 2-class per-token
+    head over sent_starts — reuses the tagger machinery with gold derived
+    from `Doc.sent_starts` and predictions written back there."""
+
+    name = "senter"
+
+    def initialize(self, examples, device) -> None:
+        self.labels = ["I", "S"]  # S = sentence start
+        self.label2id = {"I": 0, "S": 1}
+        self.cfg["labels"] = self.labels
+        if self.module is None:
+            self.module = TaggerHead(self.width, 2).to(device)
+        self._attach_embedded(device)
+
+    def _gold_ids(self, examples, n_tokens: int) -> np.ndarray:
+        key = ("senter_ids", self.name)
+        parts = []
+        for eg in examples:
+            ref = eg.reference
+            cached = ref.user_data.get(key)
+            if cached is None:
+                n = len(ref)
+                if ref.sent_starts is not None:
+                    cached = (ref.sent_starts > 0).astype(np.int64)
+                    if n > 0:
+                        cached = cached.copy()
+                        cached[0] = 1  # the first token always starts one
+                else:
+                    cached = np.full(n, -1, dtype=np.int64)  # unannotated
+                ref.user_data[key] = cached
+            parts.append(cached)
+        ids = np.concatenate(parts) if parts else np.zeros(0, dtype=np.int64)
+        if len(ids) < n_tokens:
+            ids = np.concatenate([ids, np.full(n_tokens - len(ids), -1,
+                                               dtype=np.int64)])
+        return ids
+
+    def stage_gold(self, examples, batch) -> None:
+        key = ("tagger_gold", self.name)
+        if batch is None or key in batch.staged:
+            return
+        from spacy_ray_amd.utils.pinned import to_device
+
+        gold_np = self._gold_ids(examples, batch.n_tokens)
+        gold = to_device(gold_np, batch.attr_ids.device)
+        batch.staged[key] = (gold, int((gold_np >= 0).sum()))
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        with torch.no_grad():
+            pred = self.module(t2v).argmax(dim=-1).cpu().numpy()
+        off = 0
+        for doc in docs:
+            n = len(doc)
+            ss = pred[off:off + n].astype(np.int32)
+            if n > 0:
+                ss[0] = 1
+            doc.sent_starts = ss
+            off += n

@@ -66,6 +66,30 @@ def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict
                     if gd and pd and gd[i] == pd[i]:
                         las += 1
         c["dep_uas_c"], c["dep_las_c"], c["dep_total"] = uas, las, total
+    if any(n.startswith("textcat") for n in pipe_names):
+        correct = total = 0
+        for eg in examples:
+            gold, pred = eg.reference.cats, eg.predicted.cats
+            if not gold or not pred:
+                continue
+            total += 1
+            g_best = max(gold, key=gold.get)
+            p_best = max(pred, key=pred.get)
+            correct += int(g_best == p_best)
+        c["cats_correct"], c["cats_total"] = correct, total
+    if "senter" in pipe_names:
+        tp = fp = fn = 0
+        for eg in examples:
+            gs, ps = eg.reference.sent_starts, eg.predicted.sent_starts
+            if gs is None or ps is None:
+                continue
+            for i in range(1, len(eg.reference)):  # position 0 is trivial
+                g = int(gs[i]) > 0
+                p = int(ps[i]) > 0
+                tp += int(g and p)
+                fp += int(p and not g)
+                fn += int(g and not p)
+        c["sent_tp"], c["sent_fp"], c["sent_fn"] = tp, fp, fn
     if "ner" in pipe_names:
         tp = fp = fn = 0
         for eg in examples:
@@ -94,6 +118,15 @@ def counts_to_scores(c: Dict[str, int]) -> Dict[str, float]:
         t = c["dep_total"]
         scores["dep_uas"] = c["dep_uas_c"] / t if t else 0.0
         scores["dep_las"] = c["dep_las_c"] / t if t else 0.0
+    if "cats_total" in c:
+        scores["cats_macro_acc"] = (c["cats_correct"] / c["cats_total"]
+                                    if c["cats_total"] else 0.0)
+    if "sent_tp" in c:
+        tp, fp, fn = c["sent_tp"], c["sent_fp"], c["sent_fn"]
+        p = tp / (tp + fp) if tp + fp else 0.0
+        r = tp / (tp + fn) if tp + fn else 0.0
+        scores["sents_p"], scores["sents_r"] = p, r
+        scores["sents_f"] = 2 * p * r / (p + r) if p + r else 0.0
     if "ner_tp" in c:
         tp, fp, fn = c["ner_tp"], c["ner_fp"], c["ner_fn"]
         p = tp / (tp + fp) if tp + fp else 0.0

@@ -32,7 +32,7 @@ class Vocab:
 class Doc:
     __slots__ = (
         "vocab", "words", "spaces", "attr_hashes",
-        "tags", "heads", "deps", "ents", "sent_starts",
+        "tags", "heads", "deps", "ents", "sent_starts", "cats",
         "tensor", "user_data",
     )
 
@@ -47,6 +47,7 @@ class Doc:
         deps: Optional[Sequence[str]] = None,
         ents: Optional[Sequence[str]] = None,  # per-token BILUO strings, e.g. "B-ORG"/"O"
         sent_starts: Optional[Sequence[int]] = None,  # 1 = starts a sentence
+        cats: Optional[Dict[str, float]] = None,  # doc-level categories
         attr_hashes: Optional[np.ndarray] = None,  # precomputed (n,4) uint64
     ) -> None:
         self.vocab = vocab
@@ -62,6 +63,8 @@ class Doc:
         self.ents = list(ents) if ents is not None else None
         self.sent_starts = (np.asarray(sent_starts, dtype=np.int32)
                             if sent_starts is not None else None)
+        self.cats: Optional[Dict[str, float]] = (dict(cats) if cats is not None
+                                                 else None)
         self.tensor: Optional[np.ndarray] = None
         self.user_data: Dict = {}
 
@@ -97,6 +100,7 @@ class Doc:
             "ents": self.ents,
             "sent_starts": (self.sent_starts.tolist()
                             if self.sent_starts is not None else None),
+            "cats": self.cats,
         }
 
     @classmethod
@@ -110,6 +114,7 @@ class Doc:
             deps=data.get("deps"),
             ents=data.get("ents"),
             sent_starts=data.get("sent_starts"),
+            cats=data.get("cats"),
         )
 
 
