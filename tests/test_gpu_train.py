@@ -405,3 +405,42 @@ def test_deterministic_mode_bit_identical_grads(monkeypatch):
     g3 = run()
     rel = (g3.float() - g1.float()).abs().mean() / g1.float().abs().mean().clamp(min=1e-8)
     assert float(rel) < 0.02
+
+
+@need_gpu
+def test_textcat_senter_gpu_train():
+    """textcat + senter pipes on GPU: kernels behind linear_cdw /
+    reduce_mean_ragged carry doc classification and sentence recognition;
+    losses fall."""
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.vocab.doc import Example
+
+    cfg = Config.from_disk(os.path.join(os.path.dirname(__file__), "..",
+                                        "examples", "configs",
+                                        "en_textcat.cfg"))
+    nlp = init_nlp(cfg, device="cuda:0", sample_size=64)
+    T = resolve(cfg.interpolate()["training"], validate=False)
+    engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+    docs = make_synthetic_docs(nlp.vocab, n_docs=96, words_per_doc=16,
+                               vocab_size=400, n_tags=50, n_deps=40,
+                               n_ent_types=4, seed=9)
+    examples = [Example.from_doc(d) for d in docs]
+    first = last = None
+    for i in range(12):
+        losses = {}
+        engine.accumulate(examples, drop=0.0, losses=losses)
+        engine.apply_step()
+        if i == 0:
+            first = dict(losses)
+        last = dict(losses)
+    torch.cuda.synchronize()
+    assert last["textcat"] < first["textcat"], (first, last)
+    assert last["senter"] < first["senter"], (first, last)
+    pred = [d.copy_unannotated() for d in docs[:8]]
+    nlp.predict_docs(pred)
+    for d in pred:
+        assert d.cats and d.sent_starts is not None
