@@ -779,3 +779,28 @@ def test_textcat_and_senter_train_and_predict():
     for eg in examples[:3]:
         assert eg.predicted.cats and len(eg.predicted.cats) >= 3
         assert eg.predicted.sent_starts is not None
+
+
+def test_textcat_senter_checkpoint_roundtrip(tmp_path):
+    """to_disk/load for the textcat+senter pipeline: predictions identical
+    after the roundtrip (generic component serialization covers new pipes)."""
+    import numpy as np
+
+    import spacy_ray_amd
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg = Config.from_disk("examples/configs/en_textcat.cfg")
+    nlp = init_nlp(cfg, device="cpu", sample_size=32)
+    docs = make_synthetic_docs(nlp.vocab, n_docs=8, words_per_doc=12,
+                               vocab_size=120, n_tags=10, n_deps=10,
+                               n_ent_types=3, seed=5)
+    nlp.to_disk(str(tmp_path))
+    nlp2 = spacy_ray_amd.load(str(tmp_path), device="cpu")
+    outs1 = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
+    for o1, o2 in zip(outs1, outs2):
+        for k in o1.cats:
+            assert abs(o1.cats[k] - o2.cats[k]) < 1e-5
+        assert (o1.sent_starts == o2.sent_starts).all()
