@@ -155,7 +155,9 @@ def init_config_cli(
     from spacy_ray_amd.cli.templates import render_config
 
     pipes = [p.strip() for p in pipeline.split(",") if p.strip()]
-    bad = [p for p in pipes if p not in ("tagger", "parser", "ner")]
+    bad = [p for p in pipes if p not in
+           ("tagger", "parser", "ner", "textcat",
+            "textcat_multilabel", "senter")]
     if bad:
         raise SystemExit(f"unknown pipeline components: {bad}")
     text = render_config(lang=lang, pipes=pipes, arch=arch, width=width, gpu=gpu)
