@@ -154,9 +154,39 @@ def bench_gpustate():
 
 
 ALL = {"mwe": bench_mwe, "dpre": bench_dpre, "ce": bench_ce,
-       "gpustate": bench_gpustate}
+       "gpustate": bench_gpustate, "attn": bench_attn}
 
 if __name__ == "__main__":
     names = sys.argv[1:] or list(ALL)
     for n in names:
         ALL[n]()
+
+
+def bench_attn():
+    """Fused flash-style MFMA attention vs SDPA flash at trf window shapes."""
+    from spacy_ray_amd.ops.api import window_attention
+
+    B, H, L, D = 1024, 12, 96, 64  # ~one bucket's worth of windows
+    lens = torch.randint(20, L + 1, (B,), device=dev, dtype=torch.int32)
+    q = torch.randn(B, H, L, D, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    scale = D ** -0.5
+
+    def fused_fb():
+        out = window_attention(q, k, v, lens, scale, 0.1)
+        out.sum().backward()
+        q.grad = k.grad = v.grad = None
+
+    timeit("attn fused fwd+bwd (1k win)", fused_fb)
+
+    mask = (torch.arange(L, device=dev)[None, :] < lens[:, None]).view(B, 1, 1, L)
+
+    def sdpa_fb():
+        out = torch.nn.functional.scaled_dot_product_attention(
+            q, k, v, attn_mask=mask, dropout_p=0.1, scale=scale)
+        out.sum().backward()
+        q.grad = k.grad = v.grad = None
+
+    timeit("attn sdpa  fwd+bwd (1k win)", sdpa_fb)
