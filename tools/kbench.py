@@ -153,15 +153,6 @@ def bench_gpustate():
                                  T, 18, True))
 
 
-ALL = {"mwe": bench_mwe, "dpre": bench_dpre, "ce": bench_ce,
-       "gpustate": bench_gpustate, "attn": bench_attn}
-
-if __name__ == "__main__":
-    names = sys.argv[1:] or list(ALL)
-    for n in names:
-        ALL[n]()
-
-
 def bench_attn():
     """Fused flash-style MFMA attention vs SDPA flash at trf window shapes."""
     from spacy_ray_amd.ops.api import window_attention
@@ -190,3 +181,12 @@ def bench_attn():
         q.grad = k.grad = v.grad = None
 
     timeit("attn sdpa  fwd+bwd (1k win)", sdpa_fb)
+
+
+ALL = {"mwe": bench_mwe, "dpre": bench_dpre, "ce": bench_ce,
+       "gpustate": bench_gpustate, "attn": bench_attn}
+
+if __name__ == "__main__":
+    names = sys.argv[1:] or list(ALL)
+    for n in names:
+        ALL[n]()
