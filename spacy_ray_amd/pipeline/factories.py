@@ -60,3 +60,12 @@ def make_senter_pipe(name: str, model):
     from .pipes import SenterPipe
 
     return SenterPipe(name, model)
+
+
+@registry.factories("entity_ruler")
+def make_entity_ruler_pipe(name: str, model=None, overwrite_ents: bool = False,
+                           patterns=None):
+    from .ruler import EntityRulerPipe
+
+    return EntityRulerPipe(name, model, overwrite_ents=overwrite_ents,
+                           patterns=patterns)

@@ -112,7 +112,8 @@ class Language:
         # GPU-score / CPU-advance phases pipeline against each other
         trans_tasks = []
         for name, pipe in self.pipeline:
-            if isinstance(pipe, Tok2VecPipe) or name in self._frozen:
+            if (isinstance(pipe, Tok2VecPipe) or name in self._frozen
+                    or not getattr(pipe, "trainable", True)):
                 continue
             own = pipe.own_tok2vec(batch, drop=drop)
             pt2v = own if own is not None else t2v

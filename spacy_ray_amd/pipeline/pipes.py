@@ -27,6 +27,7 @@ NEG_INF = -1e30
 class TrainablePipe:
     name: str
     listens_to: Optional[str] = None
+    trainable: bool = True  # rule-based pipes (entity_ruler) set False
 
     def __init__(self) -> None:
         self.module: Optional[nn.Module] = None

@@ -90,7 +90,7 @@ def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict
                 fp += int(p and not g)
                 fn += int(g and not p)
         c["sent_tp"], c["sent_fp"], c["sent_fn"] = tp, fp, fn
-    if "ner" in pipe_names:
+    if "ner" in pipe_names or "entity_ruler" in pipe_names:
         tp = fp = fn = 0
         for eg in examples:
             gold = _ents_to_spans(eg.reference.ents or [])

@@ -804,3 +804,83 @@ def test_textcat_senter_checkpoint_roundtrip(tmp_path):
         for k in o1.cats:
             assert abs(o1.cats[k] - o2.cats[k]) < 1e-5
         assert (o1.sent_starts == o2.sent_starts).all()
+
+
+def test_entity_ruler_patterns_and_pipeline():
+    """entity_ruler: phrase + token-spec patterns, longest-match,
+    existing-entity preservation, checkpoint roundtrip via cfg.json."""
+    from spacy_ray_amd.pipeline.ruler import EntityRulerPipe
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    vocab = Vocab()
+    ruler = EntityRulerPipe("entity_ruler")
+    ruler.add_patterns([
+        {"label": "ORG", "pattern": "Acme Corp"},
+        {"label": "GPE", "pattern": [{"LOWER": "san"}, {"LOWER": "francisco"}]},
+        {"label": "CARDINAL", "pattern": [{"IS_DIGIT": True}]},
+        {"label": "ORG", "pattern": "Acme"},  # shorter: longest must win
+    ])
+    doc = Doc(vocab, ["Acme", "Corp", "opened", "in", "San", "Francisco",
+                      "in", "1999"])
+    ruler([doc])
+    assert doc.ents == ["B-ORG", "L-ORG", "O", "O", "B-GPE", "L-GPE", "O",
+                        "U-CARDINAL"]
+    # existing entities preserved by default
+    doc2 = Doc(vocab, ["Acme", "rocks"], ents=["U-PRODUCT", "O"])
+    ruler([doc2])
+    assert doc2.ents == ["U-PRODUCT", "O"]
+    # overwrite_ents=True replaces
+    ruler2 = EntityRulerPipe("entity_ruler", overwrite_ents=True,
+                             patterns=[{"label": "ORG", "pattern": "Acme"}])
+    ruler2([doc2])
+    assert doc2.ents == ["U-ORG", "O"]
+    # cfg roundtrip
+    cfg = ruler.state_cfg()
+    r3 = EntityRulerPipe("entity_ruler")
+    r3.load_cfg(cfg, "cpu")
+    doc3 = Doc(vocab, ["San", "Francisco"])
+    r3([doc3])
+    assert doc3.ents == ["B-GPE", "L-GPE"]
+    # unsupported keys fail loudly
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        ruler.add_patterns([{"label": "X", "pattern": [{"REGEX": "a+"}]}])
+
+
+def test_entity_ruler_in_language_pipeline():
+    """entity_ruler inside a config-built pipeline: runs in predict,
+    skipped in training, survives to_disk/load."""
+    import spacy_ray_amd
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg_text = TEXTCAT_CFG.replace(
+        'pipeline = ["tok2vec", "textcat", "senter"]',
+        'pipeline = ["tok2vec", "textcat", "senter", "entity_ruler"]',
+    ).replace(
+        "[corpora]",
+        """[components.entity_ruler]
+factory = "entity_ruler"
+
+[corpora]""",
+    )
+    cfg = Config.from_str(cfg_text)
+    nlp = init_nlp(cfg, device="cpu", sample_size=16)
+    nlp.get_pipe("entity_ruler").add_patterns(
+        [{"label": "WORD", "pattern": "w1"}])
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+
+    docs = make_synthetic_docs(nlp.vocab, n_docs=4, words_per_doc=10,
+                               vocab_size=30, n_tags=5, n_deps=5,
+                               n_ent_types=2, seed=2)
+    outs = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    hits = sum(e == "U-WORD" for d in outs for e in (d.ents or []))
+    assert hits > 0  # w1 is frequent under the Zipf lexicon
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as td:
+        nlp.to_disk(td)
+        nlp2 = spacy_ray_amd.load(td, device="cpu")
+        outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
+        assert [d.ents for d in outs2] == [d.ents for d in outs]
