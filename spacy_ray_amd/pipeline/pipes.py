@@ -19,7 +19,7 @@ import torch.nn as nn
 from spacy_ray_amd import _srx_cpu
 from spacy_ray_amd.models.batch import TokenBatch
 from spacy_ray_amd.models.parser_model import TransitionModel
-from spacy_ray_amd.vocab.doc import Example, biluo_to_codes, codes_to_biluo
+from spacy_ray_amd.vocab.doc import Example, biluo_to_codes
 
 NEG_INF = -1e30
 
