@@ -25,8 +25,9 @@ from typing import List, Optional, Tuple
 from spacy_ray_amd.vocab.doc import Doc, Vocab
 
 
-def _finish_conllu_sentence(vocab, words, tags, heads, deps):
-    return Doc(vocab, words, tags=tags, heads=heads, deps=deps)
+def _finish_conllu_sentence(vocab, words, tags, heads, deps, morphs):
+    return Doc(vocab, words, tags=tags, heads=heads, deps=deps,
+               morphs=morphs)
 
 
 def read_conllu(text: str, vocab: Optional[Vocab] = None,
@@ -39,12 +40,14 @@ def read_conllu(text: str, vocab: Optional[Vocab] = None,
     tags: List[str] = []
     heads: List[int] = []
     deps: List[str] = []
+    morphs: List[str] = []
     for raw in text.splitlines():
         line = raw.rstrip("\n")
         if not line.strip():
             if words:
-                docs.append(_finish_conllu_sentence(vocab, words, tags, heads, deps))
-                words, tags, heads, deps = [], [], [], []
+                docs.append(_finish_conllu_sentence(vocab, words, tags, heads,
+                                                    deps, morphs))
+                words, tags, heads, deps, morphs = [], [], [], [], []
             continue
         if line.startswith("#"):
             continue
@@ -59,8 +62,10 @@ def read_conllu(text: str, vocab: Optional[Vocab] = None,
         head = int(parts[6]) if parts[6] != "_" else 0
         heads.append(head - 1)  # 1-based with 0=root -> index with -1=root
         deps.append(parts[7] if parts[7] != "_" else "dep")
+        morphs.append(parts[5] if parts[5] != "_" else "")
     if words:
-        docs.append(_finish_conllu_sentence(vocab, words, tags, heads, deps))
+        docs.append(_finish_conllu_sentence(vocab, words, tags, heads, deps,
+                                            morphs))
     return docs
 
 

@@ -69,3 +69,10 @@ def make_entity_ruler_pipe(name: str, model=None, overwrite_ents: bool = False,
 
     return EntityRulerPipe(name, model, overwrite_ents=overwrite_ents,
                            patterns=patterns)
+
+
+@registry.factories("morphologizer")
+def make_morphologizer_pipe(name: str, model, labels=None):
+    from .pipes import MorphologizerPipe
+
+    return _with_labels(MorphologizerPipe(name, model), labels)

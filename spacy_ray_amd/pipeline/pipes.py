@@ -1175,3 +1175,68 @@ class SenterPipe(TaggerPipe):
                 ss[0] = 1
             doc.sent_starts = ss
             off += n
+
+
+class MorphologizerPipe(TaggerPipe):
+    """Morphological feature prediction (spaCy's `morphologizer`): a
+    per-token classifier over the observed UD FEATS strings (Doc.morphs;
+    populated by `spacy-mi convert` from CoNLL-U column 6).  Reuses the
+    tagger head/loss machinery; predictions land back in Doc.morphs and
+    score as morph_acc."""
+
+    name = "morphologizer"
+
+    def initialize(self, examples, device) -> None:
+        if not self.labels:
+            labels = set()
+            for eg in examples:
+                if eg.reference.morphs:
+                    labels.update(m for m in eg.reference.morphs if m)
+            self.labels = sorted(labels) or ["_"]
+            self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.cfg["labels"] = self.labels
+        if self.module is None:
+            self.module = TaggerHead(self.width, max(1, len(self.labels))).to(device)
+        self._attach_embedded(device)
+
+    def _gold_ids(self, examples, n_tokens: int) -> np.ndarray:
+        key = ("morph_ids", tuple(self.labels))
+        parts = []
+        for eg in examples:
+            ref = eg.reference
+            cached = ref.user_data.get(key)
+            if cached is None:
+                morphs = ref.morphs or ["" for _ in range(len(ref))]
+                cached = np.fromiter(
+                    (self.label2id.get(m, -1) for m in morphs),
+                    dtype=np.int64, count=len(morphs),
+                )
+                ref.user_data[key] = cached
+            parts.append(cached)
+        ids = np.concatenate(parts) if parts else np.zeros(0, dtype=np.int64)
+        if len(ids) < n_tokens:
+            ids = np.concatenate([ids, np.full(n_tokens - len(ids), -1,
+                                               dtype=np.int64)])
+        return ids
+
+    def stage_gold(self, examples, batch) -> None:
+        key = ("tagger_gold", self.name)
+        if batch is None or key in batch.staged:
+            return
+        from spacy_ray_amd.utils.pinned import to_device
+
+        gold_np = self._gold_ids(examples, batch.n_tokens)
+        gold = to_device(gold_np, batch.attr_ids.device)
+        batch.staged[key] = (gold, int((gold_np >= 0).sum()))
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        with torch.no_grad():
+            pred = self.module(t2v).argmax(dim=-1).cpu().numpy()
+        table = (np.asarray(self.labels, dtype=object) if self.labels
+                 else np.asarray([""], dtype=object))
+        morph_strs = table[np.clip(pred, 0, len(table) - 1)]
+        off = 0
+        for doc in docs:
+            n = len(doc)
+            doc.morphs = morph_strs[off:off + n].tolist()
+            off += n

@@ -66,6 +66,18 @@ def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict
                     if gd and pd and gd[i] == pd[i]:
                         las += 1
         c["dep_uas_c"], c["dep_las_c"], c["dep_total"] = uas, las, total
+    if "morphologizer" in pipe_names:
+        correct = total = 0
+        for eg in examples:
+            gold, pred = eg.reference.morphs, eg.predicted.morphs
+            if gold is None or pred is None:
+                continue
+            for g, p in zip(gold, pred):
+                if not g:
+                    continue  # missing annotation
+                total += 1
+                correct += int(g == p)
+        c["morph_correct"], c["morph_total"] = correct, total
     if any(n.startswith("textcat") for n in pipe_names):
         correct = total = 0
         for eg in examples:
@@ -118,6 +130,9 @@ def counts_to_scores(c: Dict[str, int]) -> Dict[str, float]:
         t = c["dep_total"]
         scores["dep_uas"] = c["dep_uas_c"] / t if t else 0.0
         scores["dep_las"] = c["dep_las_c"] / t if t else 0.0
+    if "morph_total" in c:
+        scores["morph_acc"] = (c["morph_correct"] / c["morph_total"]
+                               if c["morph_total"] else 0.0)
     if "cats_total" in c:
         scores["cats_macro_acc"] = (c["cats_correct"] / c["cats_total"]
                                     if c["cats_total"] else 0.0)

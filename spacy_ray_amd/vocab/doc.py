@@ -32,7 +32,7 @@ class Vocab:
 class Doc:
     __slots__ = (
         "vocab", "words", "spaces", "attr_hashes",
-        "tags", "heads", "deps", "ents", "sent_starts", "cats",
+        "tags", "heads", "deps", "ents", "sent_starts", "cats", "morphs",
         "tensor", "user_data",
     )
 
@@ -48,6 +48,7 @@ class Doc:
         ents: Optional[Sequence[str]] = None,  # per-token BILUO strings, e.g. "B-ORG"/"O"
         sent_starts: Optional[Sequence[int]] = None,  # 1 = starts a sentence
         cats: Optional[Dict[str, float]] = None,  # doc-level categories
+        morphs: Optional[Sequence[str]] = None,  # UD FEATS strings per token
         attr_hashes: Optional[np.ndarray] = None,  # precomputed (n,4) uint64
     ) -> None:
         self.vocab = vocab
@@ -65,6 +66,7 @@ class Doc:
                             if sent_starts is not None else None)
         self.cats: Optional[Dict[str, float]] = (dict(cats) if cats is not None
                                                  else None)
+        self.morphs = list(morphs) if morphs is not None else None
         self.tensor: Optional[np.ndarray] = None
         self.user_data: Dict = {}
 
@@ -101,6 +103,7 @@ class Doc:
             "sent_starts": (self.sent_starts.tolist()
                             if self.sent_starts is not None else None),
             "cats": self.cats,
+            "morphs": self.morphs,
         }
 
     @classmethod
@@ -115,6 +118,7 @@ class Doc:
             ents=data.get("ents"),
             sent_starts=data.get("sent_starts"),
             cats=data.get("cats"),
+            morphs=data.get("morphs"),
         )
 
 

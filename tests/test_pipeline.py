@@ -884,3 +884,83 @@ factory = "entity_ruler"
         nlp2 = spacy_ray_amd.load(td, device="cpu")
         outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
         assert [d.ents for d in outs2] == [d.ents for d in outs]
+
+
+def test_morphologizer_trains_and_predicts():
+    """morphologizer over CoNLL-U-style FEATS strings: learns a
+    word-deterministic morph signal, predicts into Doc.morphs, scores as
+    morph_acc; missing FEATS ('' after convert) are excluded from loss."""
+    import numpy as np
+    import torch
+
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.train.scorer import score_examples
+    from spacy_ray_amd.vocab.doc import Example
+
+    cfg_text = TEXTCAT_CFG.replace(
+        'pipeline = ["tok2vec", "textcat", "senter"]',
+        'pipeline = ["tok2vec", "morphologizer"]',
+    ).replace("""[components.textcat]
+factory = "textcat"
+
+[components.textcat.model]
+@architectures = "spacy.TextCatReduce.v1"
+exclusive_classes = true
+
+[components.textcat.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64
+
+[components.senter]
+factory = "senter"
+
+[components.senter.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.senter.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64""",
+"""[components.morphologizer]
+factory = "morphologizer"
+
+[components.morphologizer.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.morphologizer.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64""")
+    torch.manual_seed(0)
+    cfg = Config.from_str(cfg_text)
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+
+    # init_nlp discovers labels from the corpus sample: attach morphs to a
+    # synthetic world where morph = f(tag) (fully learnable)
+    nlp = init_nlp(cfg, device="cpu", sample_size=8)
+    docs = make_synthetic_docs(nlp.vocab, n_docs=64, words_per_doc=12,
+                               vocab_size=100, n_tags=8, n_deps=5,
+                               n_ent_types=2, seed=3)
+    for d in docs:
+        d.morphs = [f"Feat={t}" for t in d.tags]
+    # re-initialize the pipe with morph-bearing examples for label discovery
+    pipe = nlp.get_pipe("morphologizer")
+    pipe.labels = []
+    examples = [Example.from_doc(d) for d in docs]
+    pipe.initialize(examples, "cpu")
+    opt = {"@optimizers": "Adam.v1", "learn_rate": 0.01}
+    engine = ZeRO1Engine(nlp, resolve({"o": opt}, validate=False)["o"],
+                         LocalComm())
+    first = last = None
+    for i in range(25):
+        losses = {}
+        engine.accumulate(examples, drop=0.0, losses=losses)
+        engine.apply_step()
+        if i == 0:
+            first = dict(losses)
+        last = dict(losses)
+    assert last["morphologizer"] < first["morphologizer"]
+    nlp.predict_docs([eg.predicted for eg in examples])
+    scores = score_examples(examples, ["morphologizer"])
+    assert scores["morph_acc"] > 0.5, scores
