@@ -160,3 +160,10 @@ def make_textcat_reduce_model(tok2vec: ModelSpec,
     spec = ModelSpec(lambda: None, width=tok2vec.width, kind="textcat")
     spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
     return spec
+
+
+@registry.architectures("spacy.SpanCategorizer.v1")
+def make_spancat_model(tok2vec: ModelSpec, **_ignored):
+    spec = ModelSpec(lambda: None, width=tok2vec.width, kind="spancat")
+    spec.embedded_tok2vec = tok2vec if tok2vec.kind == "tok2vec" else None
+    return spec

@@ -76,3 +76,14 @@ def make_morphologizer_pipe(name: str, model, labels=None):
     from .pipes import MorphologizerPipe
 
     return _with_labels(MorphologizerPipe(name, model), labels)
+
+
+@registry.factories("spancat")
+def make_spancat_pipe(name: str, model, spans_key: str = "sc",
+                      max_ngram: int = 3, threshold: float = 0.5,
+                      labels=None):
+    from .pipes import SpancatPipe
+
+    return _with_labels(
+        SpancatPipe(name, model, spans_key=spans_key, max_ngram=max_ngram,
+                    threshold=threshold), labels)

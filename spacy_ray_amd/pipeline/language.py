@@ -391,13 +391,15 @@ def init_nlp(config: Config, device: str = "cpu", sample_size: int = 128) -> Lan
         # subwords (3-4x sequence inflation = 3-4x activation memory)
         sample_size = max(sample_size, 1024)
     sample: List[Example] = []
-    tag_labels: set = set()
-    dep_labels: set = set()
-    ent_labels: set = set()
+    # generic full-corpus discovery: each pipe declares what it reads from
+    # a reference Doc (labels_from) — subclass-safe (morphologizer IS a
+    # TaggerPipe but reads morphs, senter has fixed labels)
     discover = [
         (n, p) for n, p in nlp.pipeline
         if not getattr(p, "labels", None) and hasattr(p, "label2id")
+        and hasattr(p, "labels_from")
     ]
+    label_sets: Dict[str, set] = {n: set() for n, _ in discover}
     for eg in train_corpus(nlp):
         if len(sample) < sample_size:
             sample.append(eg)
@@ -406,23 +408,11 @@ def init_nlp(config: Config, device: str = "cpu", sample_size: int = 128) -> Lan
                 break
             continue
         ref = eg.reference
-        if ref.tags:
-            tag_labels.update(ref.tags)
-        if ref.deps:
-            dep_labels.update(d for d in ref.deps if d != "ROOT")
-        if ref.ents:
-            for tag in ref.ents:
-                if tag not in ("O", "-", None, ""):
-                    ent_labels.add(tag.partition("-")[2])
-    from .pipes import NerPipe, ParserPipe, TaggerPipe
-
+        for n, p in discover:
+            label_sets[n].update(p.labels_from(ref))
     for name, pipe in discover:
-        if isinstance(pipe, TaggerPipe) and tag_labels:
-            pipe.labels = sorted(tag_labels)
-        elif isinstance(pipe, ParserPipe) and dep_labels:
-            pipe.labels = sorted(dep_labels)
-        elif isinstance(pipe, NerPipe) and ent_labels:
-            pipe.labels = sorted(ent_labels)
+        if label_sets[name]:
+            pipe.labels = sorted(label_sets[name])
         pipe.label2id = {t: i for i, t in enumerate(pipe.labels)}
     for name, pipe in nlp.pipeline:
         pipe.initialize(sample, nlp.device)

@@ -133,6 +133,12 @@ class TaggerPipe(TrainablePipe):
     name = "tagger"
     listens_to = "tok2vec"
 
+    @staticmethod
+    def labels_from(ref):
+        """Labels this pipe reads from a reference Doc (init_nlp's
+        full-corpus discovery hook; subclasses override)."""
+        return ref.tags or ()
+
     def __init__(self, name: str, spec) -> None:
         super().__init__()
         self.name = name
@@ -744,6 +750,10 @@ class _TransitionPipeBase(TrainablePipe):
 class ParserPipe(_TransitionPipeBase):
     name = "parser"
 
+    @staticmethod
+    def labels_from(ref):
+        return (d for d in ref.deps if d != "ROOT") if ref.deps else ()
+
     def __init__(self, name: str, spec, use_break: bool = False) -> None:
         super().__init__(name, spec)
         # spaCy USE_BREAK contract (sentence boundaries learned as a BREAK
@@ -918,6 +928,13 @@ class ParserPipe(_TransitionPipeBase):
 class NerPipe(_TransitionPipeBase):
     name = "ner"
 
+    @staticmethod
+    def labels_from(ref):
+        if not ref.ents:
+            return ()
+        return (t.partition("-")[2] for t in ref.ents
+                if t not in ("O", "-", None, ""))
+
     def initialize(self, examples, device) -> None:
         if not self.labels:
             labels = set()
@@ -1035,6 +1052,10 @@ class TextcatPipe(TrainablePipe):
     name = "textcat"
     listens_to = "tok2vec"
 
+    @staticmethod
+    def labels_from(ref):
+        return ref.cats.keys() if ref.cats else ()
+
     def __init__(self, name: str, spec, exclusive_classes: bool = True) -> None:
         super().__init__()
         self.name = name
@@ -1123,6 +1144,10 @@ class SenterPipe(TaggerPipe):
 
     name = "senter"
 
+    @staticmethod
+    def labels_from(ref):
+        return ("I", "S")  # fixed label set
+
     def initialize(self, examples, device) -> None:
         self.labels = ["I", "S"]  # S = sentence start
         self.label2id = {"I": 0, "S": 1}
@@ -1186,6 +1211,10 @@ class MorphologizerPipe(TaggerPipe):
 
     name = "morphologizer"
 
+    @staticmethod
+    def labels_from(ref):
+        return (m for m in ref.morphs if m) if ref.morphs else ()
+
     def initialize(self, examples, device) -> None:
         if not self.labels:
             labels = set()
@@ -1240,3 +1269,143 @@ class MorphologizerPipe(TaggerPipe):
             n = len(doc)
             doc.morphs = morph_strs[off:off + n].tolist()
             off += n
+
+
+class SpancatHead(nn.Module):
+    def __init__(self, width: int, n_labels: int):
+        super().__init__()
+        self.output = nn.Linear(width, n_labels)
+        nn.init.zeros_(self.output.weight)
+        nn.init.zeros_(self.output.bias)
+
+    def forward(self, X):
+        from spacy_ray_amd.ops.api import linear_cdw
+
+        return linear_cdw(X, self.output.weight, self.output.bias)
+
+
+class SpancatPipe(TrainablePipe):
+    """Span categorizer (spaCy's `spancat`): an ngram suggester (sizes
+    1..max_ngram) over each doc plus a multi-label classifier on pooled
+    span representations.  Overlapping spans allowed (the point of spancat
+    vs NER).  Span pooling is mean-over-tokens computed from ONE prefix
+    sum of the tok2vec matrix — every candidate is two gathers and a
+    divide, no per-span loops.  Gold/predictions live in
+    `Doc.spans[spans_key]` as (start, end, label) triples; scored as
+    spans_sc_p/r/f."""
+
+    name = "spancat"
+    listens_to = "tok2vec"
+
+    def labels_from(self, ref):
+        return (lab for (_s, _e, lab) in ref.spans.get(self.spans_key, ()))
+
+    def __init__(self, name: str, spec, spans_key: str = "sc",
+                 max_ngram: int = 3, threshold: float = 0.5) -> None:
+        super().__init__()
+        self.name = name
+        self.width = spec.width
+        self.embedded_spec = getattr(spec, "embedded_tok2vec", None)
+        self.spans_key = spans_key
+        self.max_ngram = int(max_ngram)
+        self.threshold = float(threshold)
+        self.labels: List[str] = []
+        self.label2id: Dict[str, int] = {}
+
+    def initialize(self, examples, device) -> None:
+        if not self.labels:
+            labels = set()
+            for eg in examples:
+                for (s, e, lab) in eg.reference.spans.get(self.spans_key, []):
+                    labels.add(lab)
+            self.labels = sorted(labels) or ["SPAN"]
+            self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.cfg.update(labels=self.labels, spans_key=self.spans_key,
+                        max_ngram=self.max_ngram, threshold=self.threshold)
+        if self.module is None:
+            self.module = SpancatHead(self.width, max(1, len(self.labels))).to(device)
+        self._attach_embedded(device)
+
+    def load_cfg(self, cfg, device) -> None:
+        super().load_cfg(cfg, device)
+        self.labels = list(cfg.get("labels", []))
+        self.label2id = {t: i for i, t in enumerate(self.labels)}
+        self.spans_key = cfg.get("spans_key", "sc")
+        self.max_ngram = int(cfg.get("max_ngram", 3))
+        self.threshold = float(cfg.get("threshold", 0.5))
+        if self.module is None:
+            self.module = SpancatHead(self.width, max(1, len(self.labels))).to(device)
+        self._attach_embedded(device)
+
+    # ---------------------------------------------------------- candidates
+    def _suggest(self, lengths: np.ndarray):
+        """ngram candidates -> (doc_idx, start, end) arrays + global token
+        offsets; deterministic order (doc-major, start-major, short-first)."""
+        di, ss, ee = [], [], []
+        off = 0
+        for d, n in enumerate(lengths.tolist()):
+            for k in range(1, self.max_ngram + 1):
+                for s in range(0, n - k + 1):
+                    di.append(d)
+                    ss.append(off + s)
+                    ee.append(off + s + k)
+            off += n
+        return (np.asarray(di, dtype=np.int64), np.asarray(ss, dtype=np.int64),
+                np.asarray(ee, dtype=np.int64))
+
+    def _span_scores(self, t2v, lengths: np.ndarray):
+        device = t2v.device
+        di, ss, ee = self._suggest(lengths)
+        total = int(lengths.sum())
+        csum = torch.cumsum(t2v[:total].float(), dim=0)
+        P = torch.cat([csum.new_zeros(1, csum.shape[1]), csum], dim=0)
+        s_t = torch.from_numpy(ss).to(device)
+        e_t = torch.from_numpy(ee).to(device)
+        pooled = (P[e_t] - P[s_t]) / (e_t - s_t).unsqueeze(1).float()
+        return self.module(pooled.to(t2v.dtype)), di, ss, ee
+
+    def get_loss(self, examples, t2v, batch):
+        lengths = (batch.lengths_np if batch is not None
+                   and len(batch.docs) == len(examples)
+                   else np.asarray([len(eg.reference) for eg in examples],
+                                   dtype=np.int32))
+        scores, di, ss, ee = self._span_scores(t2v, lengths)
+        # gold target matrix over candidates
+        offs = np.zeros(len(lengths), dtype=np.int64)
+        np.cumsum(lengths[:-1], out=offs[1:]) if len(lengths) > 1 else None
+        gold_sets = [
+            {(int(s), int(e), lab)
+             for (s, e, lab) in eg.reference.spans.get(self.spans_key, [])}
+            for eg in examples
+        ]
+        Y = np.zeros((len(di), max(1, len(self.labels))), dtype=np.float32)
+        for c in range(len(di)):
+            d = int(di[c])
+            if d >= len(examples):
+                continue  # pad pseudo-doc
+            s_loc = int(ss[c] - offs[d])
+            e_loc = int(ee[c] - offs[d])
+            for lab, j in self.label2id.items():
+                if (s_loc, e_loc, lab) in gold_sets[d]:
+                    Y[c, j] = 1.0
+        target = torch.from_numpy(Y).to(scores.device)
+        n = max(1, scores.shape[0])
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            scores.float(), target, reduction="sum") / n
+        return loss, loss.detach()
+
+    def predict_and_set(self, docs, t2v, batch) -> None:
+        lengths = np.asarray([len(d) for d in docs], dtype=np.int32)
+        with torch.no_grad():
+            scores, di, ss, ee = self._span_scores(t2v, lengths)
+            probs = torch.sigmoid(scores.float()).cpu().numpy()
+        offs = np.zeros(len(lengths), dtype=np.int64)
+        np.cumsum(lengths[:-1], out=offs[1:]) if len(lengths) > 1 else None
+        found = [[] for _ in docs]
+        hit_c, hit_j = np.nonzero(probs > self.threshold)
+        for c, j in zip(hit_c.tolist(), hit_j.tolist()):
+            d = int(di[c])
+            found[d].append((int(ss[c] - offs[d]), int(ee[c] - offs[d]),
+                             self.labels[j]))
+        for d, doc in enumerate(docs):
+            doc.spans[self.spans_key] = found[d]

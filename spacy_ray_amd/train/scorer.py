@@ -66,6 +66,16 @@ def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict
                     if gd and pd and gd[i] == pd[i]:
                         las += 1
         c["dep_uas_c"], c["dep_las_c"], c["dep_total"] = uas, las, total
+    if "spancat" in pipe_names:
+        tp = fp = fn = 0
+        for eg in examples:
+            for key in (set(eg.reference.spans) | set(eg.predicted.spans)):
+                gold = {tuple(sp) for sp in eg.reference.spans.get(key, [])}
+                pred = {tuple(sp) for sp in eg.predicted.spans.get(key, [])}
+                tp += len(gold & pred)
+                fp += len(pred - gold)
+                fn += len(gold - pred)
+        c["spans_tp"], c["spans_fp"], c["spans_fn"] = tp, fp, fn
     if "morphologizer" in pipe_names:
         correct = total = 0
         for eg in examples:
@@ -130,6 +140,12 @@ def counts_to_scores(c: Dict[str, int]) -> Dict[str, float]:
         t = c["dep_total"]
         scores["dep_uas"] = c["dep_uas_c"] / t if t else 0.0
         scores["dep_las"] = c["dep_las_c"] / t if t else 0.0
+    if "spans_tp" in c:
+        tp, fp, fn = c["spans_tp"], c["spans_fp"], c["spans_fn"]
+        p = tp / (tp + fp) if tp + fp else 0.0
+        r = tp / (tp + fn) if tp + fn else 0.0
+        scores["spans_sc_p"], scores["spans_sc_r"] = p, r
+        scores["spans_sc_f"] = 2 * p * r / (p + r) if p + r else 0.0
     if "morph_total" in c:
         scores["morph_acc"] = (c["morph_correct"] / c["morph_total"]
                                if c["morph_total"] else 0.0)

@@ -33,7 +33,7 @@ class Doc:
     __slots__ = (
         "vocab", "words", "spaces", "attr_hashes",
         "tags", "heads", "deps", "ents", "sent_starts", "cats", "morphs",
-        "tensor", "user_data",
+        "spans", "tensor", "user_data",
     )
 
     def __init__(
@@ -49,6 +49,7 @@ class Doc:
         sent_starts: Optional[Sequence[int]] = None,  # 1 = starts a sentence
         cats: Optional[Dict[str, float]] = None,  # doc-level categories
         morphs: Optional[Sequence[str]] = None,  # UD FEATS strings per token
+        spans: Optional[Dict] = None,  # spans groups: key -> [(start, end, label)]
         attr_hashes: Optional[np.ndarray] = None,  # precomputed (n,4) uint64
     ) -> None:
         self.vocab = vocab
@@ -67,6 +68,9 @@ class Doc:
         self.cats: Optional[Dict[str, float]] = (dict(cats) if cats is not None
                                                  else None)
         self.morphs = list(morphs) if morphs is not None else None
+        self.spans: Dict[str, list] = (
+            {k: [tuple(sp) for sp in v] for k, v in spans.items()}
+            if spans else {})
         self.tensor: Optional[np.ndarray] = None
         self.user_data: Dict = {}
 
@@ -104,6 +108,8 @@ class Doc:
                             if self.sent_starts is not None else None),
             "cats": self.cats,
             "morphs": self.morphs,
+            "spans": {k: [list(sp) for sp in v]
+                      for k, v in self.spans.items()} or None,
         }
 
     @classmethod
@@ -119,6 +125,7 @@ class Doc:
             sent_starts=data.get("sent_starts"),
             cats=data.get("cats"),
             morphs=data.get("morphs"),
+            spans=data.get("spans"),
         )
 
 

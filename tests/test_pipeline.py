@@ -947,6 +947,7 @@ width = 64""")
     # re-initialize the pipe with morph-bearing examples for label discovery
     pipe = nlp.get_pipe("morphologizer")
     pipe.labels = []
+    pipe.module = None  # rebuild the head for the discovered label count
     examples = [Example.from_doc(d) for d in docs]
     pipe.initialize(examples, "cpu")
     opt = {"@optimizers": "Adam.v1", "learn_rate": 0.01}
@@ -964,3 +965,99 @@ width = 64""")
     nlp.predict_docs([eg.predicted for eg in examples])
     scores = score_examples(examples, ["morphologizer"])
     assert scores["morph_acc"] > 0.5, scores
+
+
+def test_spancat_trains_and_predicts():
+    """spancat: ngram suggester + multilabel span classifier over prefix-sum
+    pooling; learns word-deterministic (possibly overlapping) spans and
+    writes Doc.spans[spans_key]; scored as spans_sc_f."""
+    import numpy as np
+    import torch
+
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.train.scorer import score_examples
+    from spacy_ray_amd.vocab.doc import Example
+
+    cfg_text = TEXTCAT_CFG.replace(
+        'pipeline = ["tok2vec", "textcat", "senter"]',
+        'pipeline = ["tok2vec", "spancat"]',
+    ).replace("""[components.textcat]
+factory = "textcat"
+
+[components.textcat.model]
+@architectures = "spacy.TextCatReduce.v1"
+exclusive_classes = true
+
+[components.textcat.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64
+
+[components.senter]
+factory = "senter"
+
+[components.senter.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.senter.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64""",
+"""[components.spancat]
+factory = "spancat"
+spans_key = "sc"
+max_ngram = 2
+
+[components.spancat.model]
+@architectures = "spacy.SpanCategorizer.v1"
+
+[components.spancat.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64""")
+    torch.manual_seed(0)
+    cfg = Config.from_str(cfg_text)
+    nlp = init_nlp(cfg, device="cpu", sample_size=8)
+    docs = make_synthetic_docs(nlp.vocab, n_docs=64, words_per_doc=10,
+                               vocab_size=60, n_tags=6, n_deps=5,
+                               n_ent_types=2, seed=7)
+    # word-deterministic spans: every token whose id ends in 0 is a LONE
+    # span; every bigram starting at an even word id is a PAIR span
+    for d in docs:
+        spans = []
+        for i, w in enumerate(d.words):
+            wid = int(w[1:])
+            if wid % 10 == 0:
+                spans.append((i, i + 1, "LONE"))
+            if wid % 2 == 0 and i + 2 <= len(d):
+                spans.append((i, i + 2, "PAIR"))
+        d.spans["sc"] = spans
+    pipe = nlp.get_pipe("spancat")
+    pipe.labels = []
+    pipe.module = None  # rebuild the head for the discovered label count
+    examples = [Example.from_doc(d) for d in docs]
+    pipe.initialize(examples, "cpu")
+    assert pipe.labels == ["LONE", "PAIR"]
+    opt = {"@optimizers": "Adam.v1", "learn_rate": 0.02}
+    engine = ZeRO1Engine(nlp, resolve({"o": opt}, validate=False)["o"],
+                         LocalComm())
+    first = last = None
+    for i in range(40):
+        losses = {}
+        engine.accumulate(examples, drop=0.0, losses=losses)
+        engine.apply_step()
+        if i == 0:
+            first = dict(losses)
+        last = dict(losses)
+    assert last["spancat"] < first["spancat"]
+    nlp.predict_docs([eg.predicted for eg in examples])
+    scores = score_examples(examples, ["spancat"])
+    assert scores["spans_sc_f"] > 0.5, scores
+    # overlapping spans survive (the point of spancat vs NER)
+    any_overlap = any(
+        s1 != s2 and s1[0] < s2[1] and s2[0] < s1[1]
+        for eg in examples for s1 in eg.predicted.spans.get("sc", [])
+        for s2 in eg.predicted.spans.get("sc", [])
+    )
+    assert any_overlap
