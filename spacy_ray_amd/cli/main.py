@@ -157,7 +157,8 @@ def init_config_cli(
     pipes = [p.strip() for p in pipeline.split(",") if p.strip()]
     bad = [p for p in pipes if p not in
            ("tagger", "parser", "ner", "textcat",
-            "textcat_multilabel", "senter", "morphologizer")]
+            "textcat_multilabel", "senter", "morphologizer",
+            "spancat")]
     if bad:
         raise SystemExit(f"unknown pipeline components: {bad}")
     text = render_config(lang=lang, pipes=pipes, arch=arch, width=width, gpu=gpu)
