@@ -103,6 +103,9 @@ class MicroBatcher:
 
 def _doc_json(d) -> dict:
     rec = d.to_dict()
+    # Doc.spans groups (spancat output) under their own key; "spans" stays
+    # the entity-span view derived from BILUO tags (response compat)
+    rec["span_groups"] = rec.pop("spans", None)
     rec["spans"] = [
         {"start": s, "end": e, "label": lab}
         for (s, e, lab) in sorted(_ents_to_spans(d.ents or []))
