@@ -32,13 +32,16 @@ def make_tagger_pipe(name: str, model, labels=None):
 
 
 @registry.factories("parser")
-def make_parser_pipe(name: str, model, labels=None, use_break: bool = False):
-    return _with_labels(ParserPipe(name, model, use_break=use_break), labels)
+def make_parser_pipe(name: str, model, labels=None, use_break: bool = False,
+                     beam_width: int = 1):
+    return _with_labels(
+        ParserPipe(name, model, use_break=use_break, beam_width=beam_width),
+        labels)
 
 
 @registry.factories("ner")
-def make_ner_pipe(name: str, model, labels=None):
-    return _with_labels(NerPipe(name, model), labels)
+def make_ner_pipe(name: str, model, labels=None, beam_width: int = 1):
+    return _with_labels(NerPipe(name, model, beam_width=beam_width), labels)
 
 
 @registry.factories("textcat")

@@ -184,7 +184,8 @@ class Language:
             for name, pipe in self.pipeline:
                 if name in self._disabled:
                     continue
-                if isinstance(pipe, _TransitionPipeBase):
+                if (isinstance(pipe, _TransitionPipeBase)
+                        and getattr(pipe, "beam_width", 1) <= 1):
                     own = pipe.own_tok2vec(batch)
                     trans.append((pipe,) + pipe.make_predict_task(
                         docs, own if own is not None else t2v))
@@ -192,7 +193,9 @@ class Language:
                 run_transition_tasks([t[1] for t in trans])
             heads = []
             for name, pipe in self.pipeline:
-                if isinstance(pipe, (Tok2VecPipe, _TransitionPipeBase)) or name in self._disabled:
+                if (isinstance(pipe, Tok2VecPipe) or name in self._disabled
+                        or (isinstance(pipe, _TransitionPipeBase)
+                            and getattr(pipe, "beam_width", 1) <= 1)):
                     continue
                 own = pipe.own_tok2vec(batch)
                 heads.append((pipe, own if own is not None else t2v))

@@ -734,6 +734,9 @@ class _TransitionPipeBase(TrainablePipe):
         return self.finish_task(task)
 
     def predict_and_set(self, docs, t2v, batch) -> None:
+        if getattr(self, "beam_width", 1) > 1:
+            self._beam_annotate(docs, t2v)
+            return
         task, splits, shards = self.make_predict_task(docs, t2v)
         with torch.no_grad():
             run_transition_tasks([task])
@@ -742,6 +745,9 @@ class _TransitionPipeBase(TrainablePipe):
             return
         for (lo, hi, base), states in zip(splits, shards):
             self._annotate(docs[lo:hi], states)
+
+    def _beam_annotate(self, docs, t2v) -> None:
+        raise NotImplementedError
 
     def _annotate_gpu(self, docs, decode) -> None:
         raise NotImplementedError
@@ -754,11 +760,22 @@ class ParserPipe(_TransitionPipeBase):
     def labels_from(ref):
         return (d for d in ref.deps if d != "ROOT") if ref.deps else ()
 
-    def __init__(self, name: str, spec, use_break: bool = False) -> None:
+    def __init__(self, name: str, spec, use_break: bool = False,
+                 beam_width: int = 1) -> None:
         super().__init__(name, spec)
         # spaCy USE_BREAK contract (sentence boundaries learned as a BREAK
         # transition) — see transitions.cpp / docs/PARITY.md; off by default
         self.use_break = use_break
+        # decode-time beam search (spaCy's beam parser role; training is
+        # greedy like the reference) — pipeline/beam.py
+        self.beam_width = int(beam_width)
+        if self.beam_width > 1 and use_break:
+            raise ValueError("beam decoding does not support use_break")
+
+    def _beam_annotate(self, docs, t2v) -> None:
+        from .beam import beam_parse_annotate
+
+        beam_parse_annotate(self, docs, t2v, self.beam_width)
 
     def initialize(self, examples, device) -> None:
         if not self.labels:
@@ -934,6 +951,15 @@ class NerPipe(_TransitionPipeBase):
             return ()
         return (t.partition("-")[2] for t in ref.ents
                 if t not in ("O", "-", None, ""))
+
+    def __init__(self, name: str, spec, beam_width: int = 1) -> None:
+        super().__init__(name, spec)
+        self.beam_width = int(beam_width)
+
+    def _beam_annotate(self, docs, t2v) -> None:
+        from .beam import beam_ner_annotate
+
+        beam_ner_annotate(self, docs, t2v, self.beam_width)
 
     def initialize(self, examples, device) -> None:
         if not self.labels:

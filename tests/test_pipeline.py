@@ -1061,3 +1061,52 @@ width = 64""")
         for s2 in eg.predicted.spans.get("sc", [])
     )
     assert any_overlap
+
+
+def test_beam_decode_parser_and_ner():
+    """Beam decoding (decode-only, spaCy's beam parser/NER role): width-1
+    beam equals the greedy decoder exactly (same model, same tie-breaks up
+    to float order), larger widths return valid structures with total
+    log-prob >= the width-1 path's."""
+    import numpy as np
+    import torch
+
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.pipeline.beam import beam_decode, _ArcEagerState
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    torch.manual_seed(0)
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    nlp = init_nlp(cfg, device="cpu", sample_size=24)
+    docs = make_synthetic_docs(nlp.vocab, n_docs=6, words_per_doc=9,
+                               vocab_size=80, n_tags=50, n_deps=40,
+                               n_ent_types=4, seed=4)
+    greedy = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    parser = nlp.get_pipe("parser")
+    ner = nlp.get_pipe("ner")
+    from spacy_ray_amd.models.batch import TokenBatch
+
+    for width in (1, 4):
+        parser.beam_width = width
+        ner.beam_width = width
+        outs = nlp.predict_docs([d.copy_unannotated() for d in docs])
+        for d in outs:
+            assert len(d.heads) == len(d)
+            assert all(-1 <= int(h) < len(d) for h in d.heads)
+            assert len(d.ents) == len(d)
+        if width == 1:
+            for g, b in zip(greedy, outs):
+                assert (np.asarray(g.heads) == np.asarray(b.heads)).all()
+                assert g.ents == b.ents
+    # beam total log-prob is monotone in width (same scoring model)
+    batch = TokenBatch([d.copy_unannotated() for d in docs],
+                       torch.device("cpu"))
+    with torch.no_grad():
+        t2v = nlp.tok2vec.forward(batch)
+    s1 = beam_decode(parser, docs, t2v, 1, _ArcEagerState)
+    s4 = beam_decode(parser, docs, t2v, 4, _ArcEagerState)
+    for a, b in zip(s1, s4):
+        assert b.score >= a.score - 1e-4
+    parser.beam_width = 1
+    ner.beam_width = 1
