@@ -90,3 +90,19 @@ def make_spancat_pipe(name: str, model, spans_key: str = "sc",
     return _with_labels(
         SpancatPipe(name, model, spans_key=spans_key, max_ngram=max_ngram,
                     threshold=threshold), labels)
+
+
+@registry.factories("attribute_ruler")
+def make_attribute_ruler_pipe(name: str, model=None, patterns=None):
+    from .attr_ruler import AttributeRulerPipe
+
+    return AttributeRulerPipe(name, model, patterns=patterns)
+
+
+@registry.factories("lemmatizer")
+def make_lemmatizer_pipe(name: str, model=None, mode: str = "rule",
+                         lookups=None, overwrite: bool = False):
+    from .lemmatizer import LemmatizerPipe
+
+    return LemmatizerPipe(name, model, mode=mode, lookups=lookups,
+                          overwrite=overwrite)

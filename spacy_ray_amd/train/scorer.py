@@ -66,6 +66,18 @@ def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict
                     if gd and pd and gd[i] == pd[i]:
                         las += 1
         c["dep_uas_c"], c["dep_las_c"], c["dep_total"] = uas, las, total
+    if "lemmatizer" in pipe_names:
+        correct = total = 0
+        for eg in examples:
+            gold, pred = eg.reference.lemmas, eg.predicted.lemmas
+            if gold is None or pred is None:
+                continue
+            for g, p in zip(gold, pred):
+                if not g:
+                    continue
+                total += 1
+                correct += int(g == p)
+        c["lemma_correct"], c["lemma_total"] = correct, total
     if "spancat" in pipe_names:
         tp = fp = fn = 0
         for eg in examples:
@@ -140,6 +152,9 @@ def counts_to_scores(c: Dict[str, int]) -> Dict[str, float]:
         t = c["dep_total"]
         scores["dep_uas"] = c["dep_uas_c"] / t if t else 0.0
         scores["dep_las"] = c["dep_las_c"] / t if t else 0.0
+    if "lemma_total" in c:
+        scores["lemma_acc"] = (c["lemma_correct"] / c["lemma_total"]
+                               if c["lemma_total"] else 0.0)
     if "spans_tp" in c:
         tp, fp, fn = c["spans_tp"], c["spans_fp"], c["spans_fn"]
         p = tp / (tp + fp) if tp + fp else 0.0

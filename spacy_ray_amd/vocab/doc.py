@@ -33,7 +33,7 @@ class Doc:
     __slots__ = (
         "vocab", "words", "spaces", "attr_hashes",
         "tags", "heads", "deps", "ents", "sent_starts", "cats", "morphs",
-        "spans", "tensor", "user_data",
+        "lemmas", "spans", "tensor", "user_data",
     )
 
     def __init__(
@@ -50,6 +50,7 @@ class Doc:
         cats: Optional[Dict[str, float]] = None,  # doc-level categories
         morphs: Optional[Sequence[str]] = None,  # UD FEATS strings per token
         spans: Optional[Dict] = None,  # spans groups: key -> [(start, end, label)]
+        lemmas: Optional[Sequence[str]] = None,
         attr_hashes: Optional[np.ndarray] = None,  # precomputed (n,4) uint64
     ) -> None:
         self.vocab = vocab
@@ -71,6 +72,7 @@ class Doc:
         self.spans: Dict[str, list] = (
             {k: [tuple(sp) for sp in v] for k, v in spans.items()}
             if spans else {})
+        self.lemmas = list(lemmas) if lemmas is not None else None
         self.tensor: Optional[np.ndarray] = None
         self.user_data: Dict = {}
 
@@ -110,6 +112,7 @@ class Doc:
             "morphs": self.morphs,
             "spans": {k: [list(sp) for sp in v]
                       for k, v in self.spans.items()} or None,
+            "lemmas": self.lemmas,
         }
 
     @classmethod
@@ -126,6 +129,7 @@ class Doc:
             cats=data.get("cats"),
             morphs=data.get("morphs"),
             spans=data.get("spans"),
+            lemmas=data.get("lemmas"),
         )
 
 

@@ -1110,3 +1110,184 @@ def test_beam_decode_parser_and_ner():
         assert b.score >= a.score - 1e-4
     parser.beam_width = 1
     ner.beam_width = 1
+
+
+def test_lemmatizer_and_attribute_ruler():
+    """Rule lemmatizer (built-in English rules + irregulars, lookup mode)
+    and attribute_ruler (pattern-driven TAG/LEMMA/MORPH overrides) — the
+    two rule components completing the en_core_web_sm lineup."""
+    from spacy_ray_amd.pipeline.attr_ruler import AttributeRulerPipe
+    from spacy_ray_amd.pipeline.lemmatizer import LemmatizerPipe
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    vocab = Vocab()
+    lem = LemmatizerPipe("lemmatizer")
+    doc = Doc(vocab, ["The", "children", "were", "running", "and",
+                      "stopped", "near", "better", "cities"],
+              tags=["DT", "NNS", "VBD", "VBG", "CC", "VBD", "IN", "JJR",
+                    "NNS"])
+    lem([doc])
+    assert doc.lemmas == ["the", "child", "be", "run", "and", "stop",
+                          "near", "good", "city"]
+    # lookup mode
+    lem2 = LemmatizerPipe("lemmatizer", mode="lookup",
+                          lookups={"went": "go"})
+    doc2 = Doc(vocab, ["went", "home"])
+    lem2([doc2])
+    assert doc2.lemmas == ["go", "home"]
+    # attribute ruler overrides the tagger + feeds the lemmatizer
+    ar = AttributeRulerPipe("attribute_ruler")
+    ar.add_patterns([
+        {"patterns": [[{"LOWER": "wo"}, {"LOWER": "n't"}]],
+         "attrs": {"LEMMA": "will"}, "index": 0},
+        {"patterns": [[{"LOWER": "wo"}, {"LOWER": "n't"}]],
+         "attrs": {"LEMMA": "not", "TAG": "RB"}, "index": 1},
+    ])
+    doc3 = Doc(vocab, ["I", "wo", "n't", "go"],
+               tags=["PRP", "MD", "RB", "VB"])
+    ar([doc3])
+    lem([doc3])
+    assert doc3.lemmas[1] == "will" and doc3.lemmas[2] == "not"
+    assert doc3.tags[2] == "RB"
+    # roundtrip through cfg
+    cfg = ar.state_cfg()
+    ar2 = AttributeRulerPipe("attribute_ruler")
+    ar2.load_cfg(cfg, "cpu")
+    doc4 = Doc(vocab, ["wo", "n't"], tags=["MD", "RB"])
+    ar2([doc4])
+    assert doc4.lemmas[0] == "will"
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        ar.add_patterns([{"patterns": [[{"LOWER": "x"}]],
+                          "attrs": {"DEP": "x"}}])
+    # scorer integration
+    from spacy_ray_amd.train.scorer import score_examples
+    from spacy_ray_amd.vocab.doc import Example
+
+    ref = Doc(vocab, ["children"], tags=["NNS"], lemmas=["child"])
+    eg = Example.from_doc(ref)
+    eg.predicted.tags = ["NNS"]
+    lem([eg.predicted])
+    scores = score_examples([eg], ["lemmatizer"])
+    assert scores["lemma_acc"] == 1.0
+
+
+def test_en_core_web_sm_shaped_pipeline():
+    """The REAL en_core_web_sm component lineup — tok2vec, tagger, parser,
+    ner, attribute_ruler, lemmatizer — builds from a config, trains the
+    statistical components, runs the rule components in predict order
+    (attribute_ruler AFTER the tagger, lemmatizer last), and round-trips
+    through to_disk/load."""
+    import tempfile
+
+    import torch
+
+    import spacy_ray_amd
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg_text = """
+[nlp]
+lang = "en"
+pipeline = ["tok2vec", "tagger", "parser", "ner", "attribute_ruler", "lemmatizer"]
+
+[components]
+
+[components.tok2vec]
+factory = "tok2vec"
+
+[components.tok2vec.model]
+@architectures = "spacy.HashEmbedCNN.v2"
+width = 64
+depth = 2
+embed_size = 500
+window_size = 1
+maxout_pieces = 3
+subword_features = true
+pretrained_vectors = null
+
+[components.tagger]
+factory = "tagger"
+
+[components.tagger.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.tagger.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64
+
+[components.parser]
+factory = "parser"
+
+[components.parser.model]
+@architectures = "spacy.TransitionBasedParser.v2"
+state_type = "parser"
+
+[components.parser.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64
+
+[components.ner]
+factory = "ner"
+
+[components.ner.model]
+@architectures = "spacy.TransitionBasedParser.v2"
+state_type = "ner"
+
+[components.ner.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64
+
+[components.attribute_ruler]
+factory = "attribute_ruler"
+
+[components.lemmatizer]
+factory = "lemmatizer"
+mode = "rule"
+
+[corpora]
+
+[corpora.train]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 64
+words_per_doc = 12
+vocab_size = 150
+n_tags = 10
+seed = 0
+
+[corpora.dev]
+@readers = "spacy-mi.SyntheticCorpus.v1"
+n_docs = 16
+seed = 1
+
+[training]
+seed = 0
+train_corpus = "corpora.train"
+dev_corpus = "corpora.dev"
+
+[training.optimizer]
+@optimizers = "Adam.v1"
+learn_rate = 0.001
+"""
+    torch.manual_seed(0)
+    nlp = init_nlp(Config.from_str(cfg_text), device="cpu", sample_size=16)
+    nlp.get_pipe("attribute_ruler").add_patterns(
+        [{"patterns": [[{"ORTH": "w1"}]], "attrs": {"TAG": "SPECIAL"}}])
+    docs = make_synthetic_docs(nlp.vocab, n_docs=6, words_per_doc=10,
+                               vocab_size=40, n_tags=10, n_deps=5,
+                               n_ent_types=2, seed=2)
+    outs = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    for d in outs:
+        assert d.tags and d.heads is not None and d.ents is not None
+        assert d.lemmas is not None and all(d.lemmas)
+        for w, t in zip(d.words, d.tags):
+            if w == "w1":
+                assert t == "SPECIAL"  # ruler ran after the tagger
+    with tempfile.TemporaryDirectory() as td:
+        nlp.to_disk(td)
+        nlp2 = spacy_ray_amd.load(td, device="cpu")
+        outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
+        assert [d.lemmas for d in outs2] == [d.lemmas for d in outs]
+        assert [d.tags for d in outs2] == [d.tags for d in outs]
