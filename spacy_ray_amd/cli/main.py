@@ -158,7 +158,7 @@ def init_config_cli(
     bad = [p for p in pipes if p not in
            ("tagger", "parser", "ner", "textcat",
             "textcat_multilabel", "senter", "morphologizer",
-            "spancat")]
+            "spancat", "entity_ruler", "attribute_ruler", "lemmatizer")]
     if bad:
         raise SystemExit(f"unknown pipeline components: {bad}")
     text = render_config(lang=lang, pipes=pipes, arch=arch, width=width, gpu=gpu)

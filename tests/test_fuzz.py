@@ -276,3 +276,54 @@ def test_fuzz_spacy_docbin_roundtrip():
         assert d1.heads.tolist() == list(d0.heads)
         assert d1.deps == d0.deps
         assert d1.ents == d0.ents
+
+
+def test_fuzz_rule_components():
+    """Property fuzz: entity_ruler/attribute_ruler/lemmatizer never crash
+    or emit malformed annotations on random docs + random patterns."""
+    import random
+
+    from spacy_ray_amd.pipeline.attr_ruler import AttributeRulerPipe
+    from spacy_ray_amd.pipeline.lemmatizer import LemmatizerPipe
+    from spacy_ray_amd.pipeline.ruler import EntityRulerPipe
+    from spacy_ray_amd.train.scorer import _ents_to_spans
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    rng = random.Random(0)
+    vocab = Vocab()
+    lexicon = [f"w{i}" for i in range(30)] + ["42", "7", "Acme"]
+    for trial in range(50):
+        n_pat = rng.randint(1, 5)
+        patterns = []
+        for _ in range(n_pat):
+            if rng.random() < 0.5:
+                patterns.append({"label": "X",
+                                 "pattern": " ".join(rng.choices(lexicon,
+                                                     k=rng.randint(1, 3)))})
+            else:
+                toks = [rng.choice([{"ORTH": rng.choice(lexicon)},
+                                    {"LOWER": rng.choice(lexicon)},
+                                    {"IS_DIGIT": rng.random() < 0.5}])
+                        for _ in range(rng.randint(1, 3))]
+                patterns.append({"label": "Y", "pattern": toks})
+        ruler = EntityRulerPipe("entity_ruler",
+                                overwrite_ents=rng.random() < 0.5,
+                                patterns=patterns)
+        ar = AttributeRulerPipe("attribute_ruler", patterns=[
+            {"patterns": [p["pattern"]] if not isinstance(p["pattern"], str)
+             else [p["pattern"]],
+             "attrs": {"TAG": "T"}, "index": rng.choice([0, -1])}
+            for p in patterns[:2]
+        ])
+        lem = LemmatizerPipe("lemmatizer")
+        n = rng.randint(0, 12)
+        words = rng.choices(lexicon, k=n)
+        ents = (None if rng.random() < 0.5 else
+                ["O"] * n)
+        doc = Doc(vocab, words, ents=ents)
+        ruler([doc])
+        ar([doc])
+        lem([doc])
+        assert len(doc.ents) == n
+        _ents_to_spans(doc.ents)  # BILUO sequence must be consumable
+        assert doc.lemmas is not None and len(doc.lemmas) == n
