@@ -388,3 +388,62 @@ def test_all_frozen_step_is_noop_not_crash():
     engine.accumulate(examples)
     engine.apply_step()  # no params, no crash
     nlp._frozen = []
+
+
+_WORKER_SCRIPT_BREADTH = r"""
+import json, os, sys
+import torch
+sys.path.insert(0, "@@REPO@@")
+from spacy_ray_amd.config.config import Config, resolve, resolve_dot_names
+from spacy_ray_amd.data.corpus import make_synthetic_docs
+from spacy_ray_amd.parallel.comm import init_comm_from_env
+from spacy_ray_amd.parallel.engine import ZeRO1Engine
+from spacy_ray_amd.pipeline.language import init_nlp
+from spacy_ray_amd.vocab.doc import Example
+from tests.test_pipeline import TEXTCAT_CFG
+
+rank = int(os.environ["RANK"]); world = int(os.environ["WORLD_SIZE"])
+torch.manual_seed(0)
+cfg = Config.from_str(TEXTCAT_CFG)
+nlp = init_nlp(cfg, sample_size=32)
+icfg = cfg.interpolate()
+T = resolve(icfg["training"], validate=False)
+docs = make_synthetic_docs(nlp.vocab, n_docs=8, words_per_doc=12,
+                           vocab_size=100, n_tags=10, n_deps=5,
+                           n_ent_types=3, seed=5)
+docs = [d for d in docs if len(d) == 12][:4] or docs[:4]
+examples = [Example.from_doc(d) for d in docs]
+comm = init_comm_from_env()
+engine = ZeRO1Engine(nlp, T["optimizer"], comm)
+per = len(examples) // world
+mine = examples[rank * per:(rank + 1) * per]
+for _ in range(2):
+    engine.accumulate(mine)
+    engine.apply_step()
+out = {"param_hash": float(engine.flat_param.double().abs().sum())}
+print("RESULT" + json.dumps(out))
+"""
+
+
+def test_two_rank_gloo_textcat_senter(tmp_path):
+    """world=2 gloo with the textcat+senter pipeline: the new heads share
+    the flat-param layout across ranks and land on identical params."""
+    script = tmp_path / "worker_breadth.py"
+    script.write_text(_WORKER_SCRIPT_BREADTH.replace("@@REPO@@", str(REPO)))
+    procs = []
+    port = 29533
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(rank), LOCAL_RANK=str(rank), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env, cwd=str(REPO),
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    results = []
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, f"worker failed:\n{err[-2000:]}"
+        line = [l for l in out.splitlines() if l.startswith("RESULT")][0]
+        results.append(json.loads(line[len("RESULT"):]))
+    assert results[0]["param_hash"] == pytest.approx(
+        results[1]["param_hash"], rel=1e-9)
