@@ -132,6 +132,13 @@ def score_counts(examples: Sequence[Example], pipe_names: Sequence[str]) -> Dict
             tp += len(gold & pred)
             fp += len(pred - gold)
             fn += len(gold - pred)
+            # per-type counts (spaCy's ents_per_type), additive/mergeable
+            for (_s, _e, lab) in gold & pred:
+                c[f"ner_tp::{lab}"] = c.get(f"ner_tp::{lab}", 0) + 1
+            for (_s, _e, lab) in pred - gold:
+                c[f"ner_fp::{lab}"] = c.get(f"ner_fp::{lab}", 0) + 1
+            for (_s, _e, lab) in gold - pred:
+                c[f"ner_fn::{lab}"] = c.get(f"ner_fn::{lab}", 0) + 1
         c["ner_tp"], c["ner_fp"], c["ner_fn"] = tp, fp, fn
     return c
 
@@ -180,6 +187,19 @@ def counts_to_scores(c: Dict[str, int]) -> Dict[str, float]:
         scores["ents_p"] = p
         scores["ents_r"] = r
         scores["ents_f"] = 2 * p * r / (p + r) if p + r else 0.0
+        labels = {k.split("::", 1)[1] for k in c
+                  if "::" in k and k.startswith("ner_")}
+        if labels:
+            per_type = {}
+            for lab in sorted(labels):
+                ltp = c.get(f"ner_tp::{lab}", 0)
+                lfp = c.get(f"ner_fp::{lab}", 0)
+                lfn = c.get(f"ner_fn::{lab}", 0)
+                lp = ltp / (ltp + lfp) if ltp + lfp else 0.0
+                lr = ltp / (ltp + lfn) if ltp + lfn else 0.0
+                per_type[lab] = {"p": lp, "r": lr,
+                                 "f": 2 * lp * lr / (lp + lr) if lp + lr else 0.0}
+            scores["ents_per_type"] = per_type
     return scores
 
 

@@ -444,3 +444,66 @@ def test_textcat_senter_gpu_train():
     nlp.predict_docs(pred)
     for d in pred:
         assert d.cats and d.sent_starts is not None
+
+
+@need_gpu
+def test_gpu_state_machine_exact_maxlen_boundary(monkeypatch):
+    """Docs of exactly GPU_STATE_MAXLEN (128) tokens run on the GPU state
+    machine (the LDS arrays/bitmaps are sized for exactly this bound) and
+    match the host loop; 129-token docs fall back."""
+    from spacy_ray_amd.config.config import Config, resolve
+    from spacy_ray_amd.parallel.comm import LocalComm
+    from spacy_ray_amd.parallel.engine import ZeRO1Engine
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.vocab.doc import Doc, Example
+
+    cfg = Config.from_disk(os.path.join(os.path.dirname(__file__), "..",
+                                        "examples", "configs", "en_core_cnn.cfg"))
+
+    def make_docs(nlp):
+        base = make_synthetic_docs(nlp.vocab, n_docs=8, words_per_doc=40,
+                                   vocab_size=300, n_tags=50, n_deps=40,
+                                   n_ent_types=4, seed=21)
+        docs = []
+        for want in (127, 128, 128, 1, 2):
+            d0 = base[len(docs) % len(base)]
+            reps = (want + len(d0) - 1) // len(d0)
+            words = (d0.words * reps)[:want]
+            tags = (d0.tags * reps)[:want]
+            heads = [max(0, i - 1) for i in range(want)]
+            heads[0] = -1
+            deps = ["dep0"] * want
+            deps[0] = "ROOT"
+            ents = (d0.ents * reps)[:want]
+            # re-open truncated spans safely: strip a leading I/L
+            for i in (0,):
+                if ents[i][0] in "IL":
+                    ents[i] = "O"
+            docs.append(Doc(nlp.vocab, words, tags=tags, heads=heads,
+                            deps=deps, ents=["O"] * want))
+        return docs
+
+    def run(gpu_states):
+        monkeypatch.setenv("SRX_GPU_STATES", "1" if gpu_states else "0")
+        torch.manual_seed(0)
+        np.random.seed(0)
+        nlp = init_nlp(cfg, device="cuda:0", sample_size=16)
+        T = resolve(cfg.interpolate()["training"], validate=False)
+        engine = ZeRO1Engine(nlp, T["optimizer"], LocalComm())
+        docs = make_docs(nlp)
+        examples = [Example.from_doc(d) for d in docs]
+        losses = {}
+        engine.accumulate(examples, drop=0.0, losses=losses)
+        torch.cuda.synchronize()
+        pred = [d.copy_unannotated() for d in docs]
+        nlp.predict_docs(pred)
+        heads = np.concatenate([d.heads for d in pred])
+        return dict(losses), heads
+
+    losses_ref, heads_ref = run(False)
+    losses_gpu, heads_gpu = run(True)
+    for k in losses_ref:
+        assert abs(losses_gpu[k] - losses_ref[k]) <= 1e-3 + 0.03 * abs(losses_ref[k]), (
+            k, losses_ref[k], losses_gpu[k])
+    assert (heads_gpu == heads_ref).mean() > 0.97
