@@ -198,15 +198,27 @@ class Language:
                             and getattr(pipe, "beam_width", 1) <= 1)):
                     continue
                 own = pipe.own_tok2vec(batch)
-                heads.append((pipe, own if own is not None else t2v))
-        for pipe, pt2v in heads:
-            pipe.predict_and_set(docs, pt2v, batch)
-        for pipe, task, splits, shards in trans:
-            if task.gpu is not None:  # GPU state machine: decode tensors
-                pipe._annotate_gpu(docs, task.gpu_decode)
-                continue
-            for (lo, hi, base), states in zip(splits, shards):
-                pipe._annotate(docs[lo:hi], states)
+                heads.append((name, pipe, own if own is not None else t2v))
+        # annotations land in PIPELINE ORDER (spaCy contract): a later
+        # component sees / may respect an earlier one's output — e.g.
+        # entity_ruler BEFORE ner keeps ruler entities only if ner runs
+        # later by position, ner-then-entity_ruler lets the ruler fill O
+        # tokens.  (The decode LOOPS above still ran interleaved; only the
+        # doc writes are ordered here.)
+        head_by_name = {name: (pipe, pt2v) for name, pipe, pt2v in heads}
+        trans_by_name = {pipe.name: (pipe, task, splits, shards)
+                         for pipe, task, splits, shards in trans}
+        for name, _ in self.pipeline:
+            if name in head_by_name:
+                pipe, pt2v = head_by_name[name]
+                pipe.predict_and_set(docs, pt2v, batch)
+            elif name in trans_by_name:
+                pipe, task, splits, shards = trans_by_name[name]
+                if task.gpu is not None:  # GPU state machine: decode tensors
+                    pipe._annotate_gpu(docs, task.gpu_decode)
+                    continue
+                for (lo, hi, base), states in zip(splits, shards):
+                    pipe._annotate(docs[lo:hi], states)
         return docs
 
     def evaluate(self, examples: Sequence[Example], batch_size: int = 256) -> Dict[str, float]:

@@ -72,6 +72,7 @@ class EntityRulerPipe(TrainablePipe):
         self.embedded_spec = None
         self.overwrite_ents = bool(overwrite_ents)
         self.patterns: List[Dict] = []
+        self._compiled: List = []
         if patterns:
             self.add_patterns(patterns)
 

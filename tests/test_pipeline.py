@@ -1291,3 +1291,42 @@ learn_rate = 0.001
         outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
         assert [d.lemmas for d in outs2] == [d.lemmas for d in outs]
         assert [d.tags for d in outs2] == [d.tags for d in outs]
+
+
+def test_entity_ruler_after_ner_annotation_order():
+    """Annotations land in pipeline order: an entity_ruler placed AFTER
+    ner fills O tokens without being clobbered by ner's decode writes
+    (the spaCy ruler-after-ner pattern)."""
+    import torch
+
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    torch.manual_seed(0)
+    nlp = init_nlp(cfg, device="cpu", sample_size=16)
+    # append a ruler AFTER ner
+    from spacy_ray_amd.pipeline.ruler import EntityRulerPipe
+
+    ruler = EntityRulerPipe("entity_ruler")
+    nlp.add_pipe("entity_ruler", ruler)
+    docs = make_synthetic_docs(nlp.vocab, n_docs=4, words_per_doc=10,
+                               vocab_size=50, n_tags=10, n_deps=5,
+                               n_ent_types=2, seed=6)
+    # pick a word the (untrained) ner labels O in at least one doc
+    base = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    target = None
+    for d in base:
+        for w, e in zip(d.words, d.ents):
+            if e == "O":
+                target = w
+                break
+        if target:
+            break
+    assert target is not None
+    ruler.add_patterns([{"label": "RULED", "pattern": target}])
+    outs = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    ruled = [e for d in outs for w, e in zip(d.words, d.ents)
+             if w == target]
+    assert any(e == "U-RULED" for e in ruled), ruled
