@@ -240,6 +240,8 @@ def debug_data_cli(
         (corpus,) = resolve_dot_names(icfg, [dot])
         n_docs = n_tokens = 0
         tags, deps, ents = set(), set(), set()
+        cats, morphs, span_labels = set(), set(), set()
+        n_sents = n_spans = 0
         for eg in corpus(nlp):
             ref = eg.reference
             n_docs += 1
@@ -250,15 +252,36 @@ def debug_data_cli(
                 deps.update(ref.deps)
             if ref.ents:
                 ents.update(t.partition("-")[2] for t in ref.ents if t not in ("O", ""))
+            if ref.cats:
+                cats.update(ref.cats.keys())
+            if ref.morphs:
+                morphs.update(m for m in ref.morphs if m)
+            if ref.sent_starts is not None:
+                n_sents += int((ref.sent_starts > 0).sum())
+            for group in ref.spans.values():
+                n_spans += len(group)
+                span_labels.update(lab for (_s, _e, lab) in group)
             if n_docs >= limit:
                 break
-        stats[split] = (tags, deps, ents)
+        stats[split] = (tags, deps, ents, cats, morphs, span_labels)
+        extras = []
+        if cats:
+            extras.append(f"{len(cats)} cats")
+        if morphs:
+            extras.append(f"{len(morphs)} morph classes")
+        if n_sents:
+            extras.append(f"{n_sents} sentence starts")
+        if n_spans:
+            extras.append(f"{n_spans} spans ({len(span_labels)} labels)")
+        extra = (", " + ", ".join(extras)) if extras else ""
         print(f"[+] {split}: {n_docs} docs, {n_tokens} tokens, "
-              f"{len(tags)} tags, {len(deps)} dep labels, {len(ents)} entity types")
+              f"{len(tags)} tags, {len(deps)} dep labels, "
+              f"{len(ents)} entity types{extra}")
         if n_docs == 0:
             print(f"[!] {split} corpus is EMPTY")
             warned = True
-    for i, kind in enumerate(("tags", "dep labels", "entity types")):
+    for i, kind in enumerate(("tags", "dep labels", "entity types", "cats",
+                              "morph classes", "span labels")):
         only_dev = stats["dev"][i] - stats["train"][i]
         if only_dev:
             print(f"[!] {kind} in dev but never in train: {sorted(only_dev)[:10]}")
