@@ -16,7 +16,7 @@ lower-cased form (spaCy's fallback).  Runs after the tagger (rule mode
 reads PTB-style tags when present).  Scored as lemma_acc."""
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 from .pipes import TrainablePipe
 
