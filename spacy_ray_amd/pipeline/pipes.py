@@ -1399,24 +1399,26 @@ class SpancatPipe(TrainablePipe):
                    else np.asarray([len(eg.reference) for eg in examples],
                                    dtype=np.int32))
         scores, di, ss, ee = self._span_scores(t2v, lengths)
-        # gold target matrix over candidates
-        offs = np.zeros(len(lengths), dtype=np.int64)
-        np.cumsum(lengths[:-1], out=offs[1:]) if len(lengths) > 1 else None
-        gold_sets = [
-            {(int(s), int(e), lab)
-             for (s, e, lab) in eg.reference.spans.get(self.spans_key, [])}
-            for eg in examples
-        ]
+        # gold target: O(#gold spans) — a candidate's index is closed-form
+        # (doc base + ngram-size block offset + start), so each gold span
+        # hits exactly one row instead of scanning all candidates x labels
         Y = np.zeros((len(di), max(1, len(self.labels))), dtype=np.float32)
-        for c in range(len(di)):
-            d = int(di[c])
-            if d >= len(examples):
-                continue  # pad pseudo-doc
-            s_loc = int(ss[c] - offs[d])
-            e_loc = int(ee[c] - offs[d])
-            for lab, j in self.label2id.items():
-                if (s_loc, e_loc, lab) in gold_sets[d]:
-                    Y[c, j] = 1.0
+        n_per_doc = lengths.astype(np.int64)
+        cand_per_doc = np.zeros(len(lengths), dtype=np.int64)
+        for k in range(1, self.max_ngram + 1):
+            cand_per_doc += np.maximum(0, n_per_doc - k + 1)
+        doc_base = np.zeros(len(lengths), dtype=np.int64)
+        if len(lengths) > 1:
+            np.cumsum(cand_per_doc[:-1], out=doc_base[1:])
+        for d, eg in enumerate(examples):
+            n = int(n_per_doc[d])
+            for (s0, e0, lab) in eg.reference.spans.get(self.spans_key, []):
+                k = int(e0) - int(s0)
+                j = self.label2id.get(lab)
+                if j is None or k < 1 or k > self.max_ngram or int(e0) > n:
+                    continue
+                block = sum(max(0, n - kk + 1) for kk in range(1, k))
+                Y[doc_base[d] + block + int(s0), j] = 1.0
         target = torch.from_numpy(Y).to(scores.device)
         n = max(1, scores.shape[0])
         loss = torch.nn.functional.binary_cross_entropy_with_logits(
