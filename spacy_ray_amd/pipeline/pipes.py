@@ -1369,7 +1369,13 @@ class SpancatPipe(TrainablePipe):
     # ---------------------------------------------------------- candidates
     def _suggest(self, lengths: np.ndarray):
         """ngram candidates -> (doc_idx, start, end) arrays + global token
-        offsets; deterministic order (doc-major, start-major, short-first)."""
+        offsets; deterministic order (doc-major, then ngram-size blocks,
+        start-major) — the gold indexer's closed form depends on it.
+        Cached by the length vector (training replays fixed batches)."""
+        key = lengths.tobytes()
+        cached = getattr(self, "_sugg_cache", None)
+        if cached is not None and cached[0] == key:
+            return cached[1]
         di, ss, ee = [], [], []
         off = 0
         for d, n in enumerate(lengths.tolist()):
@@ -1379,8 +1385,10 @@ class SpancatPipe(TrainablePipe):
                     ss.append(off + s)
                     ee.append(off + s + k)
             off += n
-        return (np.asarray(di, dtype=np.int64), np.asarray(ss, dtype=np.int64),
-                np.asarray(ee, dtype=np.int64))
+        out = (np.asarray(di, dtype=np.int64), np.asarray(ss, dtype=np.int64),
+               np.asarray(ee, dtype=np.int64))
+        self._sugg_cache = (key, out)
+        return out
 
     def _span_scores(self, t2v, lengths: np.ndarray):
         device = t2v.device
