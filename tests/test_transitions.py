@@ -286,3 +286,51 @@ def test_pack_step_gold_mask_matches_bruteforce_costs():
             choices = np.nonzero(ref_gold)[0]
             act = int(choices[rng.randint(len(choices))])
             batch.advance(np.array([act], dtype=np.int32))
+
+
+def test_beam_python_states_mirror_cpp_systems():
+    """The beam decoder's pure-python states must track the C++ transition
+    systems exactly (features, valid masks, finality, final heads/labels /
+    tags) through random valid action sequences."""
+    import random
+
+    from spacy_ray_amd.pipeline.beam import _ArcEagerState, _BiluoState
+
+    rng = random.Random(0)
+    L = 3  # labels / types
+    for trial in range(40):
+        n = rng.randint(1, 14)
+        cpp = _srx_cpu.ArcEagerBatch(np.array([n], dtype=np.int32), L)
+        py = _ArcEagerState(n)
+        for _ in range(2 * n + 2):
+            assert bool(cpp.is_final()[0]) == py.is_final()
+            if py.is_final():
+                break
+            v_cpp = cpp.valid()[0].astype(bool)
+            v_py = py.valid_mask(L)
+            assert (v_cpp == v_py).all(), (trial, v_cpp, v_py)
+            f_cpp = cpp.features()[0]
+            f_py = np.asarray(py.features(), dtype=np.int32)
+            assert (f_cpp == f_py).all(), (trial, f_cpp, f_py)
+            act = rng.choice(np.flatnonzero(v_py))
+            cpp.advance(np.array([act], dtype=np.int32))
+            py.apply(int(act), L)
+        assert (cpp.heads() == np.asarray(py.head, dtype=np.int32)).all()
+        assert (cpp.labels() == np.asarray(py.label, dtype=np.int32)).all()
+    for trial in range(40):
+        n = rng.randint(1, 14)
+        cpp = _srx_cpu.BiluoBatch(np.array([n], dtype=np.int32), L)
+        py = _BiluoState(n)
+        for _ in range(n + 1):
+            assert bool(cpp.is_final()[0]) == py.is_final()
+            if py.is_final():
+                break
+            v_cpp = cpp.valid()[0].astype(bool)
+            v_py = py.valid_mask(L)
+            assert (v_cpp == v_py).all(), (trial, v_cpp, v_py)
+            f_cpp = cpp.features()[0]
+            f_py = np.asarray(py.features(), dtype=np.int32)
+            assert (f_cpp == f_py).all(), (trial, f_cpp, f_py)
+            act = rng.choice(np.flatnonzero(v_py))
+            cpp.advance(np.array([act], dtype=np.int32))
+            py.apply(int(act), L)
