@@ -71,3 +71,32 @@ def test_nlp_call_uses_rule_tokenizer():
     doc = nlp.tokenizer(nlp.vocab, "Don't panic!")
     assert doc.words == ["Do", "n't", "panic", "!"]
     assert doc.text == "Do n't panic !".replace(" n't", "n't") or doc.words
+
+
+def test_tokenizer_text_reconstruction_property():
+    """Hypothesis: for arbitrary printable text, the (words, spaces) pair
+    reconstructs the normalized input (tokenization loses no characters
+    other than collapsed whitespace) — spaCy's non-destructive contract."""
+    from hypothesis import given, settings, strategies as st
+
+    from spacy_ray_amd.vocab.tokenizer import Tokenizer
+
+    tok = Tokenizer()
+
+    @settings(max_examples=150, deadline=None)
+    @given(st.text(
+        alphabet=st.characters(whitelist_categories=("Lu", "Ll", "Nd", "Po",
+                                                     "Ps", "Pe", "Sc", "Zs"),
+                               max_codepoint=0x2000),
+        max_size=60))
+    def check(text):
+        words, spaces = tok.tokenize(text)
+        assert len(words) == len(spaces)
+        rebuilt = "".join(
+            w + (" " if sp else "") for w, sp in zip(words, spaces)
+        ).rstrip()
+        # tokenizing never loses non-space characters
+        assert rebuilt.replace(" ", "") == "".join(text.split())
+        assert all(w for w in words)  # no empty tokens
+
+    check()
