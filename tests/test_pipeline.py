@@ -1330,3 +1330,69 @@ def test_entity_ruler_after_ner_annotation_order():
     ruled = [e for d in outs for w, e in zip(d.words, d.ents)
              if w == target]
     assert any(e == "U-RULED" for e in ruled), ruled
+
+
+def test_spancat_checkpoint_roundtrip(tmp_path):
+    """spancat labels/spans_key/threshold survive to_disk/load and
+    predictions are identical."""
+    import numpy as np
+    import torch
+
+    import spacy_ray_amd
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    cfg_text = TEXTCAT_CFG.replace(
+        'pipeline = ["tok2vec", "textcat", "senter"]',
+        'pipeline = ["tok2vec", "spancat"]',
+    ).replace("""[components.textcat]
+factory = "textcat"
+
+[components.textcat.model]
+@architectures = "spacy.TextCatReduce.v1"
+exclusive_classes = true
+
+[components.textcat.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64
+
+[components.senter]
+factory = "senter"
+
+[components.senter.model]
+@architectures = "spacy.Tagger.v2"
+
+[components.senter.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64""",
+"""[components.spancat]
+factory = "spancat"
+spans_key = "myspans"
+max_ngram = 2
+threshold = 0.4
+
+[components.spancat.model]
+@architectures = "spacy.SpanCategorizer.v1"
+
+[components.spancat.model.tok2vec]
+@architectures = "spacy.Tok2VecListener.v1"
+width = 64""")
+    torch.manual_seed(0)
+    nlp = init_nlp(Config.from_str(cfg_text), device="cpu", sample_size=8)
+    pipe = nlp.get_pipe("spancat")
+    pipe.labels = ["A", "B"]
+    pipe.label2id = {"A": 0, "B": 1}
+    pipe.module = None
+    pipe.initialize([], "cpu")
+    docs = make_synthetic_docs(nlp.vocab, n_docs=4, words_per_doc=8,
+                               vocab_size=40, n_tags=5, n_deps=5,
+                               n_ent_types=2, seed=8)
+    outs1 = nlp.predict_docs([d.copy_unannotated() for d in docs])
+    nlp.to_disk(str(tmp_path))
+    nlp2 = spacy_ray_amd.load(str(tmp_path), device="cpu")
+    p2 = nlp2.get_pipe("spancat")
+    assert p2.labels == ["A", "B"] and p2.spans_key == "myspans"
+    assert p2.max_ngram == 2 and abs(p2.threshold - 0.4) < 1e-9
+    outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
+    assert [d.spans for d in outs2] == [d.spans for d in outs1]
