@@ -149,9 +149,13 @@ class Language:
                     continue
                 own = pipe.own_tok2vec(batch)
                 pt2v = own if own is not None else t2v_d
-                if isinstance(pipe, _TransitionPipeBase):
-                    _, task, splits, shards = (pipe,) + pipe.make_predict_task(docs, pt2v)
+                if (isinstance(pipe, _TransitionPipeBase)
+                        and getattr(pipe, "beam_width", 1) <= 1):
+                    task, splits, shards = pipe.make_predict_task(docs, pt2v)
                     run_transition_tasks([task])
+                    if task.gpu is not None:  # GPU state machine decode
+                        pipe._annotate_gpu(docs, task.gpu_decode)
+                        continue
                     for (lo, hi, base), states in zip(splits, shards):
                         pipe._annotate(docs[lo:hi], states)
                 else:
