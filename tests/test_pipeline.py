@@ -1396,3 +1396,33 @@ width = 64""")
     assert p2.max_ngram == 2 and abs(p2.threshold - 0.4) < 1e-9
     outs2 = nlp2.predict_docs([d.copy_unannotated() for d in docs])
     assert [d.spans for d in outs2] == [d.spans for d in outs1]
+
+
+def test_annotating_components_with_transition_and_rule_pipes():
+    """training.annotating_components: listed pipes set predictions on
+    eg.predicted before losses are computed — covers the transition-pipe
+    branch (greedy + beam) and rule pipes."""
+    import torch
+
+    from spacy_ray_amd.config.config import Config
+    from spacy_ray_amd.data.corpus import make_synthetic_docs
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.vocab.doc import Example
+
+    cfg = Config.from_disk("examples/configs/en_core_cnn.cfg")
+    torch.manual_seed(0)
+    nlp = init_nlp(cfg, device="cpu", sample_size=16)
+    nlp._annotating = ["parser", "ner"]
+    docs = make_synthetic_docs(nlp.vocab, n_docs=6, words_per_doc=8,
+                               vocab_size=50, n_tags=10, n_deps=5,
+                               n_ent_types=2, seed=9)
+    examples = [Example.from_doc(d) for d in docs]
+    total, losses = nlp.forward_loss(examples, drop=0.0)
+    assert all(eg.predicted.heads is not None for eg in examples)
+    assert all(eg.predicted.ents is not None for eg in examples)
+    # beam branch of the annotating path
+    nlp.get_pipe("parser").beam_width = 2
+    examples2 = [Example.from_doc(d) for d in docs]
+    nlp.forward_loss(examples2, drop=0.0)
+    assert all(eg.predicted.heads is not None for eg in examples2)
+    nlp.get_pipe("parser").beam_width = 1
