@@ -12,8 +12,9 @@
 //      u64 registers, gold-children-in-buffer counters in LDS, gold-kids
 //      CSR in global memory),
 //   3. hidden = maxout2(sum of nF precomputed rows + bias)  (lanes = H),
-//   4. scores = upperB + hidden @ upperW^T  (upperW transposed in LDS,
-//      staged once per block; lanes = A),
+//   4. scores = upperB + hidden @ upperW^T  (upperW staged row-major
+//      [A][GS_HPAD] in LDS once per block; the dot runs as
+//      v_dot2c_f32_bf16 over b128 vector reads; lanes = A),
 //   5. masked argmax (min-cost mask first, valid fallback) via shuffles,
 //   6. state advance (scalar updates mirrored in every lane).
 // Training writes per-step rows (scores/gold/valid/feats/hidden/which)
