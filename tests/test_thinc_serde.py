@@ -60,3 +60,36 @@ def test_component_thinc_bytes_written_and_reloadable(tmp_path):
     h1 = nlp.get_pipe("tagger").module
     h2 = nlp2.get_pipe("tagger").module
     assert torch.allclose(h1.output.weight, h2.output.weight)
+
+
+def test_thinc_bytes_roundtrip_new_pipe_types():
+    """Thinc-msgpack component bytes restore params for every new
+    trainable pipe type (textcat/senter/morphologizer/spancat)."""
+    import torch
+
+    from spacy_ray_amd.data.thinc_serde import (
+        component_to_thinc_nodes, load_thinc_nodes_into_component,
+        model_to_thinc_bytes)
+    from spacy_ray_amd.pipeline.pipes import (MorphologizerPipe, SenterPipe,
+                                              SpancatPipe, TextcatPipe)
+
+    class Spec:
+        width = 32
+        embedded_tok2vec = None
+
+    for cls in (TextcatPipe, SenterPipe, MorphologizerPipe, SpancatPipe):
+        p = cls("x", Spec())
+        p.labels = ["A", "B"]
+        p.label2id = {"A": 0, "B": 1}
+        p.initialize([], "cpu")
+        with torch.no_grad():
+            for t in p.module.parameters():
+                t.add_(torch.randn_like(t))
+        blob = model_to_thinc_bytes(component_to_thinc_nodes(p))
+        ref = {k: v.clone() for k, v in p.module.state_dict().items()}
+        with torch.no_grad():
+            for t in p.module.parameters():
+                t.zero_()
+        assert load_thinc_nodes_into_component(p, blob) > 0
+        for k, v in p.module.state_dict().items():
+            assert torch.allclose(v, ref[k]), (cls.__name__, k)
