@@ -193,3 +193,25 @@ def test_legacy_native_docbin_still_readable():
     out = DocBin.from_bytes(legacy, Vocab())
     assert out.docs[0].words == ["x", "y"]
     assert out.docs[0].tags == ["A", "B"]
+
+
+def test_docbin_native_schema_carries_new_doc_fields():
+    """The native DocBin schema round-trips cats/morphs/lemmas/spans/
+    sent_starts (the `.spacy` wire format stays the documented
+    SPACY_ATTRS subset)."""
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    v = Vocab()
+    d = Doc(v, ["cats", "ran"], tags=["NNS", "VBD"], cats={"X": 1.0},
+            morphs=["Number=Plur", ""], lemmas=["cat", "run"],
+            spans={"sc": [(0, 2, "S")]}, sent_starts=[1, 0])
+    db = DocBin([d])
+    d2 = DocBin._from_native_bytes(db.to_native_bytes(), v).docs[0]
+    assert d2.cats == {"X": 1.0}
+    assert d2.morphs == ["Number=Plur", ""]
+    assert d2.lemmas == ["cat", "run"]
+    assert d2.spans == {"sc": [(0, 2, "S")]}
+    assert list(d2.sent_starts) == [1, 0]
+    d3 = DocBin.from_bytes(db.to_bytes(), v).docs[0]
+    assert d3.tags == ["NNS", "VBD"]
