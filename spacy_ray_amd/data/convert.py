@@ -20,7 +20,7 @@ IOB (CoNLL-02/03 style):
 from __future__ import annotations
 
 from pathlib import Path
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 from spacy_ray_amd.vocab.doc import Doc, Vocab
 

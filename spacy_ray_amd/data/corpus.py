@@ -12,7 +12,6 @@ entity spans, so models have learnable structure for convergence tests.
 from __future__ import annotations
 
 import random
-from pathlib import Path
 from typing import Iterator, List, Optional
 
 import numpy as np

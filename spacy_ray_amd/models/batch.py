@@ -7,7 +7,7 @@ Python objects anywhere near the hot path.
 """
 from __future__ import annotations
 
-from typing import List, Sequence
+from typing import Sequence
 
 import numpy as np
 import torch

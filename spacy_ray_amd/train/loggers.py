@@ -12,7 +12,6 @@ from __future__ import annotations
 
 import json
 import time
-from pathlib import Path
 from typing import Optional
 
 from spacy_ray_amd.config.registry import registry

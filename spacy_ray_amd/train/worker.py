@@ -22,7 +22,7 @@ from typing import Dict, Iterator, Optional
 import torch
 
 from spacy_ray_amd.config.config import Config, resolve, resolve_dot_names
-from spacy_ray_amd.parallel.comm import Comm, LocalComm, init_comm_from_env
+from spacy_ray_amd.parallel.comm import Comm, init_comm_from_env
 from spacy_ray_amd.parallel.engine import ZeRO1Engine
 from spacy_ray_amd.pipeline.language import init_nlp
 from spacy_ray_amd.train.loop import create_train_batches, train_while_improving
