@@ -100,3 +100,29 @@ def test_tokenizer_text_reconstruction_property():
         assert all(w for w in words)  # no empty tokens
 
     check()
+
+
+def test_matcher_and_phrase_matcher():
+    """Public Matcher/PhraseMatcher API (the rulers' token-spec dialect)."""
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+    from spacy_ray_amd.vocab.matcher import Matcher, PhraseMatcher
+
+    vocab = Vocab()
+    doc = Doc(vocab, ["Acme", "Corp", "bought", "acme", "for", "42"])
+    m = Matcher()
+    m.add("ORG", [[{"ORTH": "Acme"}, {"ORTH": "Corp"}]])
+    m.add("ACME_ANY", [[{"LOWER": "acme"}]])
+    m.add("NUM", [[{"IS_DIGIT": True}]])
+    got = m(doc)
+    assert ("ORG", 0, 2) in got
+    assert ("ACME_ANY", 0, 1) in got and ("ACME_ANY", 3, 4) in got
+    assert ("NUM", 5, 6) in got
+    assert got == sorted(got, key=lambda t: (t[1], t[2]))
+    pm = PhraseMatcher()
+    pm.add("P", ["Acme Corp", "for"])
+    got2 = pm(doc)
+    assert ("P", 0, 2) in got2 and ("P", 4, 5) in got2
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        m.add("BAD", [[{"REGEX": "x"}]])
