@@ -40,12 +40,15 @@ def ray_train_cli(
     use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
     verbose: bool = typer.Option(False, "--verbose", "-V", help="Display more information"),
     resume: bool = typer.Option(False, "--resume", help="Resume from <output>/model-last (params + per-rank optimizer shards)"),
+    init_tok2vec: Optional[Path] = typer.Option(None, "--init-tok2vec", help="Pretrained tok2vec weights (spacy-mi pretrain output dir or .safetensors)"),
     nnodes: int = typer.Option(1, "--nnodes", help="Number of nodes (with --address on every node)"),
     node_rank: int = typer.Option(0, "--node-rank", help="This node's rank in [0, nnodes)"),
 ):
     """Train a pipeline with N data-parallel workers over RCCL/xGMI."""
     logging.basicConfig(level=logging.DEBUG if verbose else logging.ERROR)
     overrides = parse_config_overrides(list(ctx.args))
+    if init_tok2vec is not None:
+        overrides["training.init_tok2vec"] = str(init_tok2vec)
     config = Config.from_disk(config_path, overrides=overrides)
     raise SystemExit(
         ray_train(config, config_path=config_path, output_path=output_path,
@@ -301,6 +304,40 @@ def convert_cli(
 
     n = convert_file(input_path, output_path, fmt=fmt, tag_col=tag_col)
     print(f"wrote {n} docs -> {output_path}")
+
+
+@app.command("pretrain")
+def pretrain_cli(
+    ctx: typer.Context,
+    config_path: Path = typer.Argument(..., help="Path to config file (its tok2vec + train corpus are used)"),
+    output_path: Path = typer.Argument(..., help="Output directory for tok2vec.safetensors"),
+    steps: int = typer.Option(1000, "--steps", help="Pretraining steps"),
+    batch_docs: int = typer.Option(64, "--batch-docs", help="Docs per step"),
+    n_buckets: int = typer.Option(4096, "--buckets", help="Masked-token target buckets"),
+    mask_rate: float = typer.Option(0.15, "--mask-rate"),
+    learn_rate: float = typer.Option(1e-3, "--lr"),
+    use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
+):
+    """Pretrain the config's tok2vec with a masked-token objective on the
+    raw training corpus (the `spacy pretrain` role); load the result at
+    training time with `--init-tok2vec` / `training.init_tok2vec`."""
+    from spacy_ray_amd.config.config import resolve_dot_names
+    from spacy_ray_amd.pipeline.language import init_nlp
+    from spacy_ray_amd.train.pretrain import pretrain_tok2vec, save_tok2vec
+
+    overrides = parse_config_overrides(list(ctx.args))
+    config = Config.from_disk(config_path, overrides=overrides)
+    device = f"cuda:{use_gpu}" if use_gpu >= 0 else "cpu"
+    nlp = init_nlp(config, device=device)
+    icfg = config.interpolate()
+    dot = icfg.get("training", {}).get("train_corpus", "corpora.train")
+    (corpus,) = resolve_dot_names(icfg, [dot])
+    losses = pretrain_tok2vec(nlp, corpus, steps=steps, batch_docs=batch_docs,
+                              n_buckets=n_buckets, mask_rate=mask_rate,
+                              learn_rate=learn_rate)
+    path = save_tok2vec(nlp, output_path)
+    print(f"[+] wrote {path} (final loss {losses[-1]:.4f}) — train with "
+          f"--init-tok2vec {output_path}")
 
 
 @app.command("package")

@@ -128,6 +128,19 @@ def distributed_train(
     T = resolve(icfg["training"], schema=ConfigSchemaTraining)
 
     nlp = init_nlp(config, device=device)
+    init_t2v = icfg.get("training", {}).get("init_tok2vec")
+    if init_t2v and not resume:
+        # spaCy's [initialize] init_tok2vec contract: load pretrained
+        # encoder weights (spacy-mi pretrain output) before the engine
+        # snapshots the fp32 master.  Same file on every rank => params
+        # stay rank-identical.
+        from spacy_ray_amd.train.pretrain import load_init_tok2vec
+
+        n_loaded = load_init_tok2vec(nlp, init_t2v)
+        if rank == 0:
+            logging.getLogger("spacy_ray_amd").info(
+                "loaded %d pretrained tok2vec tensors from %s",
+                n_loaded, init_t2v)
     _check_param_manifest(nlp, comm)
     # params are now identical on all ranks; diverge the RNG for dropout
     seed = int(icfg.get("training", {}).get("seed", 0) or 0)
