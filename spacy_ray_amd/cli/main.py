@@ -306,6 +306,52 @@ def convert_cli(
     print(f"wrote {n} docs -> {output_path}")
 
 
+@app.command("assemble")
+def assemble_cli(
+    ctx: typer.Context,
+    config_path: Path = typer.Argument(..., help="Path to config file"),
+    output_path: Path = typer.Argument(..., help="Output pipeline directory"),
+):
+    """Build and initialize a pipeline from a config WITHOUT training and
+    save it (spaCy's `assemble` role — e.g. rule-only pipelines, or a
+    random-init starting point)."""
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    overrides = parse_config_overrides(list(ctx.args))
+    config = Config.from_disk(config_path, overrides=overrides)
+    nlp = init_nlp(config, device="cpu")
+    nlp.to_disk(output_path)
+    print(f"[+] assembled pipeline {nlp.pipe_names} -> {output_path}")
+
+
+@app.command("apply")
+def apply_cli(
+    model_path: Path = typer.Argument(..., help="Trained pipeline directory"),
+    input_path: Path = typer.Argument(..., help="Input DocBin (.spacy) or text file (one doc per line)"),
+    output_path: Path = typer.Argument(..., help="Output DocBin (.spacy) with predictions"),
+    use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
+    batch_size: int = typer.Option(256, "--batch-size", "-b"),
+):
+    """Annotate a corpus with a trained pipeline and write the predictions
+    as a DocBin (spaCy's `apply` role)."""
+    import spacy_ray_amd
+    from spacy_ray_amd.data.docbin import DocBin
+
+    device = f"cuda:{use_gpu}" if use_gpu >= 0 else "cpu"
+    nlp = spacy_ray_amd.load(str(model_path), device=device)
+    if str(input_path).endswith(".spacy"):
+        docs_in = list(DocBin.from_disk(input_path, nlp.vocab).get_docs(nlp.vocab))
+        docs = [d.copy_unannotated() for d in docs_in]
+    else:
+        lines = [l for l in Path(input_path).read_text().splitlines() if l.strip()]
+        docs = [nlp.tokenizer(nlp.vocab, l) for l in lines]
+    out = []
+    for i in range(0, len(docs), batch_size):
+        out.extend(nlp.predict_docs(docs[i:i + batch_size]))
+    DocBin(out).to_disk(output_path)
+    print(f"[+] annotated {len(out)} docs -> {output_path}")
+
+
 @app.command("pretrain")
 def pretrain_cli(
     ctx: typer.Context,

@@ -401,3 +401,34 @@ def test_cli_package_wraps_trained_model(tmp_path):
     nlp = mod.load()
     doc = nlp("Don't panic!")
     assert doc.tags is not None and len(doc.tags) == len(doc)
+
+
+def test_assemble_and_apply_cli(tmp_path):
+    """`assemble` builds+saves an untrained pipeline; `apply` annotates a
+    DocBin or a text file with a saved pipeline (spaCy CLI roles)."""
+    import subprocess
+    import sys
+
+    from spacy_ray_amd.data.docbin import DocBin
+    from spacy_ray_amd.vocab.doc import Doc, Vocab
+
+    asm = tmp_path / "asm"
+    r = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "assemble",
+         "examples/configs/en_tagger_cpu.cfg", str(asm)],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert (asm / "config.cfg").exists() and (asm / "tagger").exists()
+    # apply over a DocBin
+    v = Vocab()
+    din = tmp_path / "in.spacy"
+    DocBin([Doc(v, ["hello", "world"]), Doc(v, ["again"])]).to_disk(din)
+    dout = tmp_path / "out.spacy"
+    r = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "apply",
+         str(asm), str(din), str(dout)],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-800:]
+    docs = list(DocBin.from_disk(dout, v).get_docs(v))
+    assert len(docs) == 2
+    assert docs[0].tags and len(docs[0].tags) == 2
