@@ -144,6 +144,35 @@ init_app = typer.Typer(name="init", no_args_is_help=True,
 app.add_typer(init_app)
 
 
+@init_app.command("fill-config")
+def init_fill_config_cli(
+    base_path: Path = typer.Argument(..., help="Partial config to fill"),
+    output_path: Path = typer.Argument("-", help="Output path ('-' = stdout)"),
+):
+    """Fill a partial config with schema defaults (the `spacy init
+    fill-config` role): every [training] key the schema knows gets its
+    default materialized; existing values are kept."""
+    import json as _json
+
+    from spacy_ray_amd.config.schemas import ConfigSchemaTraining
+
+    config = Config.from_disk(base_path)
+    training = dict(config.get("training", {}))
+    defaults = ConfigSchemaTraining().model_dump()
+    filled = 0
+    for key, val in defaults.items():
+        if key not in training and val is not None and key != "score_weights":
+            training[key] = val
+            filled += 1
+    config["training"] = training
+    text = config.to_str()
+    if str(output_path) == "-":
+        print(text)
+    else:
+        output_path.write_text(text)
+        print(f"[+] wrote {output_path} ({filled} defaults filled)")
+
+
 @init_app.command("config")
 def init_config_cli(
     output_path: Path = typer.Argument(..., help="Where to write the config (use - for stdout)"),

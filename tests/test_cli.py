@@ -432,3 +432,27 @@ def test_assemble_and_apply_cli(tmp_path):
     docs = list(DocBin.from_disk(dout, v).get_docs(v))
     assert len(docs) == 2
     assert docs[0].tags and len(docs[0].tags) == 2
+
+
+def test_init_fill_config_cli(tmp_path):
+    """`init fill-config` materializes [training] schema defaults while
+    preserving existing values."""
+    import subprocess
+    import sys
+
+    base = tmp_path / "partial.cfg"
+    base.write_text('[nlp]\nlang = "en"\npipeline = ["tok2vec","tagger"]\n\n'
+                    '[training]\nmax_steps = 500\n')
+    out = tmp_path / "full.cfg"
+    r = subprocess.run(
+        [sys.executable, "-m", "spacy_ray_amd.cli.main", "init",
+         "fill-config", str(base), str(out)],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-800:]
+    from spacy_ray_amd.config.config import Config
+
+    cfg = Config.from_disk(out)
+    t = cfg["training"]
+    assert t["max_steps"] == 500          # preserved
+    assert t["dropout"] == 0.1            # defaulted
+    assert t["eval_frequency"] == 200     # defaulted
