@@ -381,6 +381,48 @@ def apply_cli(
     print(f"[+] annotated {len(out)} docs -> {output_path}")
 
 
+@app.command("find-threshold")
+def find_threshold_cli(
+    model_path: Path = typer.Argument(..., help="Trained pipeline directory"),
+    pipe_name: str = typer.Argument(..., help="Component to tune (spancat / textcat_multilabel)"),
+    scores_key: str = typer.Option("", "--scores-key", help="Score to maximize (default: spans_sc_f for spancat, cats_macro_acc otherwise)"),
+    n_trials: int = typer.Option(11, "--n-trials", help="Thresholds to try in (0, 1)"),
+    use_gpu: int = typer.Option(-1, "--gpu-id", "-g"),
+    corpus_dot: str = typer.Option("corpora.dev", "--corpus"),
+):
+    """Sweep a prediction threshold on dev data and report the best
+    (spaCy's `find-threshold` role for spancat / multilabel textcat)."""
+    import spacy_ray_amd
+    from spacy_ray_amd.config.config import resolve_dot_names
+    from spacy_ray_amd.train.scorer import score_examples
+
+    device = f"cuda:{use_gpu}" if use_gpu >= 0 else "cpu"
+    nlp = spacy_ray_amd.load(str(model_path), device=device)
+    pipe = nlp.get_pipe(pipe_name)
+    if not hasattr(pipe, "threshold"):
+        raise SystemExit(f"{pipe_name} has no threshold to tune")
+    icfg = nlp.config.interpolate()
+    (corpus,) = resolve_dot_names(icfg, [corpus_dot])
+    examples = list(corpus(nlp))
+    if not examples:
+        raise SystemExit(f"{corpus_dot} is empty")
+    key = scores_key or ("spans_sc_f" if pipe_name == "spancat"
+                         else "cats_macro_acc")
+    best = (None, -1.0)
+    for i in range(1, n_trials + 1):
+        thr = i / (n_trials + 1)
+        pipe.threshold = thr
+        for eg in examples:
+            eg.predicted = eg.predicted.copy_unannotated()
+        nlp.predict_docs([eg.predicted for eg in examples])
+        sc = score_examples(examples, [pipe_name]).get(key, 0.0)
+        print(f"  threshold {thr:.3f}  {key} {sc:.4f}")
+        if sc > best[1]:
+            best = (thr, sc)
+    print(f"[+] best threshold {best[0]:.3f} ({key} {best[1]:.4f}) — set "
+          f"`threshold = {best[0]:.3f}` on [components.{pipe_name}]")
+
+
 @app.command("pretrain")
 def pretrain_cli(
     ctx: typer.Context,
