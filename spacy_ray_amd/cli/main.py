@@ -117,6 +117,7 @@ def ray_evaluate_cli(
     model_path: Path = typer.Argument(..., help="Trained pipeline directory (model-best/model-last)"),
     use_gpu: int = typer.Option(-1, "--gpu-id", "-g", help="GPU ID or -1 for CPU"),
     corpus: Optional[str] = typer.Option(None, "--corpus", help="Dotted corpus name in the model's config (default: the training dev corpus)"),
+    output: Optional[Path] = typer.Option(None, "--output", "-o", help="Also write the scores JSON to this file"),
 ):
     """Evaluate a saved pipeline on its dev corpus and print scores JSON."""
     import torch
@@ -136,7 +137,10 @@ def ray_evaluate_cli(
     examples = list(dev_corpus(nlp))
     scores = nlp.evaluate(examples)
     scores["score"] = weighted_score(scores, T.get("score_weights") or {})
-    print(json.dumps(scores, indent=2))
+    text = json.dumps(scores, indent=2)
+    if output is not None:
+        output.write_text(text)
+    print(text)
 
 
 init_app = typer.Typer(name="init", no_args_is_help=True,
@@ -379,6 +383,34 @@ def apply_cli(
         out.extend(nlp.predict_docs(docs[i:i + batch_size]))
     DocBin(out).to_disk(output_path)
     print(f"[+] annotated {len(out)} docs -> {output_path}")
+
+
+@debug_app.command("model")
+def debug_model_cli(
+    ctx: typer.Context,
+    config_path: Path = typer.Argument(..., help="Path to config file"),
+    component: str = typer.Argument("", help="Component name (default: all)"),
+):
+    """Build the pipeline skeleton and print each component's parameter
+    table (name, shape, count) — spaCy's `debug model` role."""
+    from spacy_ray_amd.pipeline.language import init_nlp
+
+    overrides = parse_config_overrides(list(ctx.args))
+    config = Config.from_disk(config_path, overrides=overrides)
+    nlp = init_nlp(config, device="cpu", sample_size=8)
+    total = 0
+    for name, pipe in nlp.pipeline:
+        if component and name != component:
+            continue
+        if pipe.module is None:
+            print(f"[{name}] (no parameters — rule component)")
+            continue
+        print(f"[{name}]")
+        for pname, p in pipe.module.named_parameters():
+            n = p.numel()
+            total += n
+            print(f"  {pname:48s} {str(tuple(p.shape)):>18s} {n:>10,d}")
+    print(f"total parameters: {total:,d}")
 
 
 @app.command("find-threshold")
